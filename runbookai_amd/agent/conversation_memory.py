@@ -1,0 +1,133 @@
+"""Chat-mode conversation memory with auto-compression.
+
+Parity with reference src/agent/conversation-memory.ts (555 LoC): message
+ring + investigation summaries; auto-compress after 16 msgs (maybe_compress
+L422), get_context_for_prompt token-budgeted (L249-292), search + related-
+context recall (L228-246, L539-552), JSON round-trip (L474-516).
+"""
+from __future__ import annotations
+
+import json
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+from ..utils.tokens import estimate_tokens
+from .scratchpad import jaccard
+from .types import now_ms
+
+
+@dataclass
+class Message:
+    role: str  # user | assistant
+    content: str
+    timestamp: int = field(default_factory=now_ms)
+
+    def to_dict(self) -> dict[str, Any]:
+        return {"role": self.role, "content": self.content, "timestamp": self.timestamp}
+
+
+@dataclass
+class InvestigationSummary:
+    query: str
+    answer_digest: str
+    services: list[str] = field(default_factory=list)
+    timestamp: int = field(default_factory=now_ms)
+
+    def to_dict(self) -> dict[str, Any]:
+        return {"query": self.query, "answerDigest": self.answer_digest,
+                "services": self.services, "timestamp": self.timestamp}
+
+
+class ConversationMemory:
+    def __init__(self, summarize_after_messages: int = 16, llm: Any = None) -> None:
+        self.summarize_after = summarize_after_messages
+        self.llm = llm
+        self.messages: list[Message] = []
+        self.compressed_summary: str = ""
+        self.investigations: list[InvestigationSummary] = []
+
+    def add_message(self, role: str, content: str) -> None:
+        self.messages.append(Message(role=role, content=content))
+        self.maybe_compress()
+
+    def add_investigation(self, query: str, answer: str, services: Optional[list[str]] = None) -> None:
+        self.investigations.append(
+            InvestigationSummary(query=query, answer_digest=answer[:400], services=list(services or []))
+        )
+
+    # -- compression (reference L422) ----------------------------------------
+
+    def maybe_compress(self) -> bool:
+        if len(self.messages) <= self.summarize_after:
+            return False
+        old = self.messages[: -self.summarize_after // 2]
+        self.messages = self.messages[-self.summarize_after // 2:]
+        digest_lines = [f"{m.role}: {m.content[:140]}" for m in old]
+        if self.llm is not None:
+            try:
+                self.compressed_summary = self.llm.complete(
+                    "Summarize this conversation in <=8 bullet points, keeping service "
+                    "names, symptoms and conclusions:\n" + "\n".join(digest_lines)
+                )[:1500]
+                return True
+            except Exception:  # noqa: BLE001
+                pass
+        prefix = (self.compressed_summary + "\n") if self.compressed_summary else ""
+        self.compressed_summary = (prefix + "\n".join(digest_lines))[-2000:]
+        return True
+
+    # -- recall (reference L228-246, L539-552) -------------------------------
+
+    def search(self, query: str, limit: int = 3) -> list[Message]:
+        scored = [(jaccard(m.content, query), m) for m in self.messages]
+        scored = [t for t in scored if t[0] > 0.05]
+        scored.sort(key=lambda t: t[0], reverse=True)
+        return [m for _, m in scored[:limit]]
+
+    def related_investigations(self, query: str, limit: int = 2) -> list[InvestigationSummary]:
+        scored = [(jaccard(f"{s.query} {s.answer_digest}", query), s) for s in self.investigations]
+        scored = [t for t in scored if t[0] > 0.05]
+        scored.sort(key=lambda t: t[0], reverse=True)
+        return [s for _, s in scored[:limit]]
+
+    # -- prompt context (reference L249-292) ---------------------------------
+
+    def get_context_for_prompt(self, token_budget: int = 2000) -> str:
+        parts: list[str] = []
+        if self.compressed_summary:
+            parts.append("## Earlier conversation (compressed)\n" + self.compressed_summary)
+        recent: list[str] = []
+        used = estimate_tokens("\n".join(parts))
+        for m in reversed(self.messages):
+            cost = estimate_tokens(m.content) + 4
+            if used + cost > token_budget:
+                break
+            recent.append(f"{m.role}: {m.content}")
+            used += cost
+        if recent:
+            parts.append("## Recent messages\n" + "\n".join(reversed(recent)))
+        return "\n\n".join(parts)
+
+    # -- round-trip (reference L474-516) -------------------------------------
+
+    def to_json(self) -> str:
+        return json.dumps({
+            "summarizeAfter": self.summarize_after,
+            "messages": [m.to_dict() for m in self.messages],
+            "compressedSummary": self.compressed_summary,
+            "investigations": [s.to_dict() for s in self.investigations],
+        })
+
+    @classmethod
+    def from_json(cls, raw: str) -> "ConversationMemory":
+        data = json.loads(raw)
+        mem = cls(summarize_after_messages=data.get("summarizeAfter", 16))
+        mem.messages = [Message(role=m["role"], content=m["content"], timestamp=m.get("timestamp", 0))
+                        for m in data.get("messages", [])]
+        mem.compressed_summary = data.get("compressedSummary", "")
+        mem.investigations = [
+            InvestigationSummary(query=s["query"], answer_digest=s.get("answerDigest", ""),
+                                 services=s.get("services", []), timestamp=s.get("timestamp", 0))
+            for s in data.get("investigations", [])
+        ]
+        return mem

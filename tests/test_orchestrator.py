@@ -1,0 +1,208 @@
+"""Orchestrator tests with scripted LLM + mock tool executor (parity with
+reference agent/__tests__/investigation-orchestrator.test.ts:14-120 —
+the whole investigation loop runs against a deterministic fake model)."""
+import json
+
+from runbookai_amd.agent.orchestrator import InvestigationOrchestrator
+from runbookai_amd.model.client import MockLLMClient
+
+
+class MockToolExecutor:
+    """Canned telemetry per tool."""
+
+    def __init__(self, overrides=None):
+        self.calls = []
+        self.overrides = overrides or {}
+
+    def execute(self, tool_name, params):
+        self.calls.append((tool_name, params))
+        if tool_name in self.overrides:
+            value = self.overrides[tool_name]
+            if isinstance(value, Exception):
+                raise value
+            return value
+        if tool_name == "pagerduty_get_incident":
+            return {"incident": {"id": params.get("incidentId"), "title": "checkout latency spike",
+                                 "status": "triggered", "service": "checkout-api"}}
+        if tool_name == "cloudwatch_alarms":
+            return {"alarms": [{"name": "redis-conns", "state": "ALARM",
+                                "reason": "connections > 950", "service": "redis"}]}
+        if tool_name == "cloudwatch_logs":
+            return {"events": [{"message": "redis: connection pool exhausted"},
+                               {"message": "i/o timeout to redis"}]}
+        if tool_name == "search_knowledge":
+            return {"results": [{"title": "Redis connection exhaustion runbook",
+                                 "type": "runbook",
+                                 "content": "raise pool size; restart workers"}]}
+        if tool_name == "datadog":
+            return {"series": [], "items": []}
+        return {"items": []}
+
+
+def scripted_llm():
+    """Per-phase canned JSON: triage -> hypotheses -> evaluate confirm -> conclusion -> remediation."""
+    llm = MockLLMClient()
+    llm.on(r"triaging a production incident", json.dumps({
+        "summary": "checkout-api latency caused by redis issues",
+        "symptoms": ["latency spike", "redis timeouts"],
+        "affectedServices": ["checkout-api", "redis"],
+        "severity": "high", "timeline": "09:10-09:40",
+    }))
+    llm.on(r"generating root-cause hypotheses", json.dumps({
+        "hypotheses": [
+            {"statement": "redis connection pool exhaustion", "rationale": "pool errors in logs",
+             "priority": 1, "affectedServices": ["redis", "checkout-api"]},
+            {"statement": "network partition to redis", "rationale": "i/o timeouts",
+             "priority": 2, "affectedServices": ["redis"]},
+        ],
+    }))
+    llm.on(r"evaluating evidence", json.dumps({
+        "action": "confirm", "confidence": 0.9,
+        "reasoning": "pool exhausted messages + alarm",
+        "evidence": [{"description": "connection pool exhausted in logs", "supports": True,
+                      "source": "cloudwatch_logs"}],
+    }))
+    llm.on(r"writing the conclusion", json.dumps({
+        "rootCause": "redis connection pool exhaustion in checkout-api",
+        "confidence": "high",
+        "summary": "pool capped at 100; spike exceeded it",
+        "affectedServices": ["checkout-api", "redis"],
+        "evidence": ["pool exhausted log lines", "redis-conns alarm"],
+    }))
+    llm.on(r"planning remediation", json.dumps({
+        "summary": "raise pool size and restart",
+        "steps": [
+            {"description": "increase redis pool max to 500", "risk": "medium",
+             "requiresApproval": False},
+            {"description": "rolling restart checkout-api", "risk": "high", "requiresApproval": True},
+        ],
+        "rollback": "revert pool config",
+    }))
+    return llm
+
+
+def test_full_investigation_flow():
+    llm = scripted_llm()
+    tools = MockToolExecutor()
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=tools)
+    result = orch.investigate("checkout-api latency spiked, redis timeouts", incident_id="PD-1")
+
+    assert result.success
+    assert "redis connection pool exhaustion" in result.root_cause
+    assert result.confidence == "high"
+    assert "checkout-api" in result.affected_services
+    assert result.remediation_plan is not None
+    assert len(result.remediation_plan["steps"]) == 2
+    assert "triage" in result.phases_visited
+    assert "conclude" in result.phases_visited
+    assert "complete" in result.phases_visited
+    # incident-provider seed was queried first
+    assert tools.calls[0][0] == "pagerduty_get_incident"
+
+
+def test_prune_then_conclude_on_exhausted_hypotheses():
+    llm = MockLLMClient()
+    llm.on(r"triaging", json.dumps({"summary": "s", "symptoms": [], "affectedServices": [],
+                                    "severity": "low"}))
+    llm.on(r"generating root-cause", json.dumps({"hypotheses": [
+        {"statement": "bad deploy", "rationale": "", "priority": 1}]}))
+    llm.on(r"evaluating evidence", json.dumps({
+        "action": "prune", "confidence": 0.1, "reasoning": "no deploys happened"}))
+    llm.on(r"writing the conclusion", json.dumps({
+        "rootCause": "inconclusive — all hypotheses pruned", "confidence": "low", "summary": ""}))
+    llm.on(r"planning remediation", json.dumps({"summary": "none", "steps": []}))
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=MockToolExecutor())
+    result = orch.investigate("mystery incident")
+    assert result.success
+    assert result.confidence == "low"
+    assert result.hypotheses[0]["status"] == "pruned"
+
+
+def test_branching_creates_subhypotheses():
+    llm = MockLLMClient()
+    llm.on(r"triaging", json.dumps({"summary": "s", "symptoms": ["slow"],
+                                    "affectedServices": ["db"], "severity": "medium"}))
+    llm.on(r"generating root-cause", json.dumps({"hypotheses": [
+        {"statement": "database problem", "rationale": "", "priority": 1}]}))
+    responses = iter([
+        json.dumps({"action": "branch", "confidence": 0.5, "reasoning": "too generic",
+                    "subHypotheses": [
+                        {"statement": "db connection exhaustion", "rationale": "", "priority": 1},
+                        {"statement": "slow query regression", "rationale": "", "priority": 2}]}),
+        json.dumps({"action": "confirm", "confidence": 0.85, "reasoning": "conn errors"}),
+    ])
+    llm.on(r"evaluating evidence", lambda p: next(responses))
+    llm.on(r"writing the conclusion", json.dumps({
+        "rootCause": "db connection exhaustion", "confidence": "high", "summary": ""}))
+    llm.on(r"planning remediation", json.dumps({"summary": "scale", "steps": [
+        {"description": "scale db pool", "risk": "low"}]}))
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=MockToolExecutor())
+    result = orch.investigate("db slow")
+    assert result.success
+    statuses = {h["statement"]: h["status"] for h in result.hypotheses}
+    assert statuses["database problem"] == "branched"
+    assert statuses["db connection exhaustion"] == "confirmed"
+
+
+def test_tool_fallback_when_unavailable():
+    llm = scripted_llm()
+    tools = MockToolExecutor()
+    # datadog not available -> orchestrator must adapt queries to cloudwatch
+    orch = InvestigationOrchestrator(
+        llm=llm, tool_executor=tools,
+        available_tools={"cloudwatch_alarms", "cloudwatch_logs", "search_knowledge",
+                         "pagerduty_get_incident", "aws_query"},
+    )
+    result = orch.investigate("high latency in checkout", incident_id="PD-2")
+    assert result.success
+    called = {name for name, _ in tools.calls}
+    assert "datadog" not in called
+
+
+def test_unparseable_llm_output_falls_back():
+    llm = MockLLMClient()
+    llm.on(r"triaging", "I think something is wrong with redis???")
+    llm.on(r"generating root-cause", "not json either")
+    llm.on(r"evaluating evidence", "nope")
+    llm.on(r"writing the conclusion", "still not json")
+    llm.on(r"planning remediation", "nada")
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=MockToolExecutor(), max_iterations=3)
+    result = orch.investigate("redis acting up")
+    # graceful degradation: still completes with fallback hypothesis/conclusion
+    assert result.success
+    assert result.root_cause != ""
+    assert result.confidence == "low"
+
+
+def test_tool_errors_are_evidence_not_crashes():
+    llm = scripted_llm()
+    tools = MockToolExecutor(overrides={"cloudwatch_logs": RuntimeError("access denied")})
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=tools)
+    result = orch.investigate("latency spike", incident_id="PD-3")
+    assert result.success
+
+
+def test_auto_remediate_respects_approval():
+    llm = scripted_llm()
+    tools = MockToolExecutor()
+    denied = []
+    orch = InvestigationOrchestrator(
+        llm=llm, tool_executor=tools, auto_remediate=True,
+        approval_callback=lambda step: denied.append(step) and False,
+    )
+    result = orch.investigate("latency", incident_id="PD-4")
+    assert result.success
+    # the high-risk step needed approval and was denied
+    assert len(denied) == 1
+    assert denied[0]["risk"] == "high"
+
+
+def test_event_stream():
+    llm = scripted_llm()
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=MockToolExecutor())
+    events = []
+    orch.on(lambda e: events.append(e.type))
+    orch.investigate("latency", incident_id="PD-5")
+    assert "phase" in events
+    assert "hypothesis" in events
+    assert "conclusion" in events
