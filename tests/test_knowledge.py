@@ -1,0 +1,260 @@
+"""Knowledge layer tests: FTS5 store, chunking, vector store, hybrid RRF.
+(Parity: reference temp-dir real-SQLite test pattern — hook-handlers/mcp
+tests create real knowledge DBs; graph-store.test.ts 594 LoC.)"""
+import os
+
+import numpy as np
+import pytest
+
+from runbookai_amd.knowledge.indexer.embedder import (
+    HashEmbedder,
+    cosine_similarity,
+    find_most_similar,
+)
+from runbookai_amd.knowledge.retriever.default import KnowledgeRetriever, create_retriever
+from runbookai_amd.knowledge.retriever.hybrid import HybridRetriever, reciprocal_rank_fusion
+from runbookai_amd.knowledge.sources.filesystem import (
+    chunk_markdown,
+    infer_doc_type,
+    load_from_filesystem,
+    load_markdown,
+    parse_frontmatter,
+)
+from runbookai_amd.knowledge.store.graph_store import ServiceGraph
+from runbookai_amd.knowledge.store.sqlite_store import KnowledgeStore
+from runbookai_amd.knowledge.store.vector_store import VectorStore
+from runbookai_amd.knowledge.types import SearchHit, SourceConfig
+
+EXAMPLES = os.path.join(os.path.dirname(__file__), "..", "examples", "runbooks")
+
+
+class TestFilesystemSource:
+    def test_frontmatter(self):
+        meta, body = parse_frontmatter("---\ntitle: T\nservices: [a, b]\n---\n# Hi\nbody")
+        assert meta["title"] == "T"
+        assert meta["services"] == ["a", "b"]
+        assert body.startswith("# Hi")
+
+    def test_no_frontmatter(self):
+        meta, body = parse_frontmatter("# Just markdown")
+        assert meta == {}
+
+    def test_header_chunking_line_ranges(self):
+        body = "# Title\nintro\n\n## Symptoms\n- a\n- b\n\n## Mitigation\n1. do x\n"
+        chunks = chunk_markdown("d1", body)
+        sections = [c.section for c in chunks]
+        assert "Symptoms" in sections
+        assert "Mitigation" in sections
+        mit = next(c for c in chunks if c.section == "Mitigation")
+        assert mit.chunk_type == "procedure"
+        assert mit.start_line > 0 and mit.end_line >= mit.start_line
+
+    def test_infer_doc_type(self):
+        assert infer_doc_type("docs/postmortems/x.md", "") == "postmortem"
+        assert infer_doc_type("runbooks/y.md", "") == "runbook"
+        assert infer_doc_type("misc.md", "this is a postmortem of the outage") == "postmortem"
+
+    def test_load_examples(self):
+        docs = load_from_filesystem(EXAMPLES)
+        assert len(docs) >= 3
+        redis = next(d for d in docs if "Redis" in d.title)
+        assert redis.doc_type == "runbook"
+        assert "redis" in redis.services
+        assert len(redis.chunks) >= 4
+
+
+class TestKnowledgeStore:
+    def _store_with_examples(self):
+        store = KnowledgeStore(":memory:")
+        for doc in load_from_filesystem(EXAMPLES):
+            store.upsert_document(doc)
+        return store
+
+    def test_fts_search(self):
+        store = self._store_with_examples()
+        hits = store.search("connection pool exhausted")
+        assert hits
+        assert "Redis" in hits[0].title
+
+    def test_type_filter(self):
+        store = self._store_with_examples()
+        hits = store.search("redis outage", doc_type="postmortem")
+        assert hits
+        assert all(h.doc_type == "postmortem" for h in hits)
+
+    def test_service_filter(self):
+        store = self._store_with_examples()
+        hits = store.search("timeout", service="api-gateway")
+        assert all("api-gateway" in h.services for h in hits)
+
+    def test_upsert_replaces(self):
+        store = self._store_with_examples()
+        docs = load_from_filesystem(EXAMPLES)
+        for d in docs:
+            store.upsert_document(d)  # second upsert must not duplicate
+        s = store.stats()
+        assert s["documents"] == len(docs)
+
+    def test_stats_by_type(self):
+        store = self._store_with_examples()
+        s = store.stats()
+        assert s["byType"]["runbook"] >= 2
+        assert s["byType"]["postmortem"] >= 1
+
+
+class TestEmbedder:
+    def test_deterministic(self):
+        e = HashEmbedder()
+        v1 = e.embed_text("redis connection pool")
+        v2 = HashEmbedder().embed_text("redis connection pool")
+        assert np.allclose(v1, v2)
+
+    def test_similar_texts_closer(self):
+        e = HashEmbedder()
+        a = e.embed_text("redis connection pool exhausted in checkout")
+        b = e.embed_text("checkout redis pool exhaustion errors")
+        c = e.embed_text("kubernetes node disk pressure eviction")
+        assert cosine_similarity(a, b) > cosine_similarity(a, c)
+
+    def test_cache(self):
+        e = HashEmbedder()
+        e.embed_text("x")
+        e.embed_text("x")
+        assert e.cache_stats()["hits"] == 1
+
+    def test_find_most_similar_reference(self):
+        e = HashEmbedder()
+        matrix = e.embed_texts(["redis pool", "gateway 5xx", "disk full"])
+        q = e.embed_text("redis pool issues")
+        top = find_most_similar(q, matrix, top_k=2)
+        assert top[0][0] == 0
+
+
+class TestVectorStore:
+    def _vs(self):
+        vs = VectorStore(":memory:", embedder=HashEmbedder())
+        vs.add_chunks([
+            {"chunkId": "c1", "docId": "d1", "title": "Redis runbook", "section": "Symptoms",
+             "services": ["redis"], "content": "connection pool exhausted i/o timeout",
+             "type": "runbook"},
+            {"chunkId": "c2", "docId": "d2", "title": "Gateway runbook", "section": "Symptoms",
+             "services": ["api-gateway"], "content": "gateway timeout upstream unavailable 5xx",
+             "type": "runbook"},
+        ])
+        return vs
+
+    def test_search_ranks_relevant_first(self):
+        vs = self._vs()
+        hits = vs.search("redis connection pool exhausted", limit=2, min_score=0.0)
+        assert hits
+        assert hits[0].chunk_id == "c1"
+
+    def test_min_score_filters(self):
+        vs = self._vs()
+        hits = vs.search("zzz qqq completely unrelated xyzzy", limit=2, min_score=0.9)
+        assert hits == []
+
+    def test_persistence_roundtrip(self, tmp_path):
+        db = str(tmp_path / "vec.db")
+        vs = VectorStore(db, embedder=HashEmbedder())
+        vs.add_chunks([{"chunkId": "c1", "docId": "d1", "title": "t", "section": "",
+                        "services": [], "content": "redis pool", "type": "runbook"}])
+        vs.close()
+        vs2 = VectorStore(db, embedder=HashEmbedder())
+        assert vs2.count() == 1
+
+
+class TestHybridRRF:
+    def _hit(self, cid, title="t"):
+        return SearchHit(doc_id="d", chunk_id=cid, title=title, content="", doc_type="runbook",
+                         score=1.0)
+
+    def test_rrf_weighting(self):
+        # c1 ranks 1st in vector (weight .6), c2 ranks 1st in fts (weight .4)
+        fused = reciprocal_rank_fusion([
+            (0.4, [self._hit("c2"), self._hit("c1")]),
+            (0.6, [self._hit("c1"), self._hit("c2")]),
+        ])
+        assert fused[0].chunk_id == "c1"
+
+    def test_hybrid_search_end_to_end(self):
+        store = KnowledgeStore(":memory:")
+        for doc in load_from_filesystem(EXAMPLES):
+            store.upsert_document(doc)
+        embedder = HashEmbedder()
+        vs = VectorStore(":memory:", embedder=embedder)
+        for cid, did, title, content in store.all_chunks():
+            vs.add_chunks([{"chunkId": cid, "docId": did, "title": title, "section": "",
+                            "services": [], "content": content, "type": "runbook"}])
+        h = HybridRetriever(store, vs)
+        hits = h.search("redis connection pool exhaustion", limit=3)
+        assert hits
+        assert "Redis" in hits[0].title
+
+    def test_degrades_to_fts_without_vectors(self):
+        store = KnowledgeStore(":memory:")
+        for doc in load_from_filesystem(EXAMPLES):
+            store.upsert_document(doc)
+        h = HybridRetriever(store, None)
+        assert h._effective_mode() == "fts"
+        assert h.search("redis pool", limit=2)
+
+
+class TestRetriever:
+    def test_sync_and_retrieve(self):
+        r = KnowledgeRetriever(
+            sources=[SourceConfig(kind="filesystem", path=EXAMPLES)],
+            embedder=HashEmbedder(),
+        )
+        counts = r.sync()
+        assert counts["documents"] >= 3
+        rk = r.retrieve({"query": "redis connection pool exhausted checkout"})
+        assert not rk.is_empty()
+        assert rk.runbooks or rk.postmortems
+
+    def test_search_lazy_init(self):
+        r = KnowledgeRetriever(sources=[SourceConfig(kind="filesystem", path=EXAMPLES)])
+        results = r.search("gateway 5xx deploy")
+        assert results
+        assert "type" in results[0]
+
+
+class TestServiceGraph:
+    def _graph(self):
+        g = ServiceGraph()
+        g.add_dependency("checkout-api", "cart-service")
+        g.add_dependency("checkout-api", "payment-service")
+        g.add_dependency("cart-service", "redis", critical=True)
+        g.add_dependency("payment-service", "postgres")
+        return g
+
+    def test_upstream_downstream(self):
+        g = self._graph()
+        assert "redis" in g.upstream("checkout-api")
+        assert "checkout-api" in g.downstream("redis")
+
+    def test_blast_radius_depth(self):
+        g = self._graph()
+        assert g.downstream("redis", max_depth=1) == ["cart-service"]
+        assert set(g.downstream("redis", max_depth=2)) == {"cart-service", "checkout-api"}
+
+    def test_path(self):
+        g = self._graph()
+        assert g.find_path("checkout-api", "redis") == ["checkout-api", "cart-service", "redis"]
+        assert g.find_path("redis", "checkout-api") == []
+
+    def test_edge_attr(self):
+        g = self._graph()
+        assert g.edge_attr("cart-service", "redis", "critical") is True
+
+    def test_load_services_config(self):
+        g = ServiceGraph()
+        g.load_services_config([
+            {"name": "a", "dependsOn": ["b"], "owner": "team-x"},
+            {"name": "b", "dependsOn": []},
+        ])
+        assert g.dependencies_of("a") == ["b"]
+        assert g.node("a")["owner"] == "team-x"
+
+    def test_stats(self):
+        assert self._graph().stats() == {"nodes": 5, "edges": 4}
