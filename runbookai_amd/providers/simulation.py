@@ -1,0 +1,213 @@
+"""Simulated incident environment backing every provider tool.
+
+The reference's providers call live AWS/Datadog/PagerDuty/... HTTP APIs.
+This environment has no egress, and BASELINE.json specifies measurement on
+a "simulated incident set" — so the provider layer reads from a
+SimScenario: a deterministic in-memory world (services, alarms, logs,
+metrics, incidents, deployments, pods) that tools query exactly as they
+would query the real APIs. Scenarios are either built-in (demo), generated
+from eval fixtures, or loaded from YAML.
+
+This mirrors the role of the reference's scripts/simulate/setup-incidents.sh
+(which provisions REAL AWS failures for manual testing) as an in-process,
+hermetic equivalent.
+"""
+from __future__ import annotations
+
+import random
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+
+@dataclass
+class SimScenario:
+    name: str = "default"
+    incident: dict[str, Any] = field(default_factory=dict)
+    services: list[dict[str, Any]] = field(default_factory=list)
+    alarms: list[dict[str, Any]] = field(default_factory=list)
+    log_events: list[dict[str, Any]] = field(default_factory=list)
+    metrics: dict[str, list[float]] = field(default_factory=dict)
+    deployments: list[dict[str, Any]] = field(default_factory=list)
+    pods: list[dict[str, Any]] = field(default_factory=list)
+    monitors: list[dict[str, Any]] = field(default_factory=list)
+    resources: dict[str, list[dict[str, Any]]] = field(default_factory=dict)  # aws service -> items
+    notes: list[dict[str, Any]] = field(default_factory=list)  # incident notes added by tools
+    slack_messages: list[dict[str, Any]] = field(default_factory=list)
+    mutations: list[dict[str, Any]] = field(default_factory=list)
+
+    # ------------------------------------------------------------------ built-ins
+
+    @classmethod
+    def redis_exhaustion(cls) -> "SimScenario":
+        """The flagship demo scenario (reference demo/demo-data.ts: scripted
+        Redis-connection-exhaustion investigation)."""
+        s = cls(name="redis-conn-exhaustion")
+        s.incident = {
+            "id": "PD-EXAMPLE-001",
+            "title": "checkout-api latency spiked and redis timeouts increased",
+            "status": "triggered",
+            "urgency": "high",
+            "service": "checkout-api",
+            "createdAt": "2026-02-10T09:10:00Z",
+        }
+        s.services = [
+            {"name": "checkout-api", "status": "degraded", "type": "ecs"},
+            {"name": "cart-service", "status": "degraded", "type": "ecs"},
+            {"name": "redis", "status": "saturated", "type": "elasticache"},
+            {"name": "payment-service", "status": "healthy", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "checkout-api-p99-latency", "state": "ALARM",
+             "reason": "p99 > 2000ms for 15 minutes", "service": "checkout-api"},
+            {"name": "redis-connected-clients", "state": "ALARM",
+             "reason": "connected_clients > 950 (maxclients 1000)", "service": "redis"},
+            {"name": "payment-success-rate", "state": "OK", "reason": "", "service": "payment-service"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-10T09:12:03Z", "service": "checkout-api", "level": "ERROR",
+             "message": "redis: connection pool exhausted (100/100 in use), waited 5000ms"},
+            {"timestamp": "2026-02-10T09:12:09Z", "service": "cart-service", "level": "ERROR",
+             "message": "dial tcp 10.0.3.12:6379: i/o timeout to redis"},
+            {"timestamp": "2026-02-10T09:13:41Z", "service": "checkout-api", "level": "ERROR",
+             "message": "redis: connection pool exhausted (100/100 in use), waited 5000ms"},
+            {"timestamp": "2026-02-10T09:14:02Z", "service": "cart-service", "level": "WARN",
+             "message": "retrying GET cart:u-8812 after timeout (attempt 3)"},
+            {"timestamp": "2026-02-10T09:15:00Z", "service": "checkout-api", "level": "ERROR",
+             "message": "checkout failed: upstream cart-service returned 503"},
+        ]
+        s.metrics = {
+            "redis.net.clients": [420, 455, 512, 778, 943, 991, 998, 1000],
+            "checkout-api.latency.p99": [180, 210, 240, 890, 1900, 2400, 2600, 2550],
+            "cart-service.error_rate": [0.1, 0.2, 0.1, 4.5, 12.0, 18.2, 22.0, 19.8],
+        }
+        s.deployments = [
+            {"service": "cart-service", "version": "v2026.02.10-1", "at": "2026-02-10T09:02:00Z",
+             "change": "config: redis pool size 200 -> 100"},
+        ]
+        s.pods = [
+            {"name": "checkout-api-7f9c", "namespace": "prod", "status": "Running", "restarts": 0,
+             "cpu": "240m", "memory": "512Mi"},
+            {"name": "cart-service-1b2d", "namespace": "prod", "status": "Running", "restarts": 3,
+             "cpu": "180m", "memory": "420Mi"},
+        ]
+        s.monitors = [
+            {"name": "checkout latency", "status": "Alert", "query": "avg:checkout.latency{*} > 2000"},
+            {"name": "redis clients", "status": "Alert", "query": "avg:redis.net.clients{*} > 900"},
+        ]
+        s.resources = {
+            "elasticache": [{"id": "redis-prod-001", "engine": "redis", "status": "available",
+                             "nodes": 3, "maxclients": 1000}],
+            "ecs": [{"name": "checkout-api", "desiredCount": 6, "runningCount": 6,
+                     "taskDefinition": "checkout-api:118"},
+                    {"name": "cart-service", "desiredCount": 4, "runningCount": 4,
+                     "taskDefinition": "cart-service:201"}],
+        }
+        return s
+
+    @classmethod
+    def gateway_5xx(cls) -> "SimScenario":
+        s = cls(name="api-gateway-5xx")
+        s.incident = {
+            "id": "PD-EXAMPLE-002",
+            "title": "API gateway started returning 5xx after deploy",
+            "status": "triggered", "urgency": "high", "service": "api-gateway",
+            "createdAt": "2026-02-11T14:05:00Z",
+        }
+        s.services = [
+            {"name": "api-gateway", "status": "degraded", "type": "ecs"},
+            {"name": "user-service", "status": "unhealthy", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "gateway-5xx-rate", "state": "ALARM",
+             "reason": "5xx rate > 5% for 10 minutes", "service": "api-gateway"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-11T14:06:10Z", "service": "api-gateway", "level": "ERROR",
+             "message": "upstream unavailable: user-service timed out after 10s (gateway timeout)"},
+            {"timestamp": "2026-02-11T14:06:30Z", "service": "user-service", "level": "ERROR",
+             "message": "panic: nil pointer dereference in handler v3.4.0 (deploy 14:03)"},
+            {"timestamp": "2026-02-11T14:07:00Z", "service": "api-gateway", "level": "ERROR",
+             "message": "504 gateway timeout routing /v1/users"},
+        ]
+        s.metrics = {
+            "api-gateway.5xx_rate": [0.1, 0.1, 0.2, 6.5, 9.1, 11.2],
+            "user-service.availability": [100, 100, 100, 42, 18, 11],
+        }
+        s.deployments = [
+            {"service": "user-service", "version": "v3.4.0", "at": "2026-02-11T14:03:00Z",
+             "change": "release v3.4.0: new profile handler"},
+        ]
+        s.monitors = [{"name": "gateway 5xx", "status": "Alert", "query": "sum:gateway.5xx{*} > 100"}]
+        s.resources = {"ecs": [{"name": "user-service", "desiredCount": 4, "runningCount": 2,
+                                "taskDefinition": "user-service:77"}]}
+        return s
+
+    @classmethod
+    def from_fixture(cls, case: dict[str, Any]) -> "SimScenario":
+        """Generate a scenario from an eval fixture case: the telemetry
+        reflects the expected root cause so a competent agent can find it."""
+        expected = case.get("expected", {})
+        keywords = expected.get("rootCauseKeywords", ["error"])
+        services = expected.get("affectedServices", ["service-a"])
+        rng = random.Random(case.get("id", "seed"))
+        s = cls(name=case.get("id", "generated"))
+        s.incident = {
+            "id": case.get("incidentId", "PD-GEN-001"),
+            "title": case.get("query", "incident"),
+            "status": "triggered", "urgency": "high",
+            "service": services[0] if services else "unknown",
+            "createdAt": "2026-02-12T10:00:00Z",
+        }
+        s.services = [{"name": svc, "status": "degraded" if i < 2 else "healthy", "type": "ecs"}
+                      for i, svc in enumerate(services)]
+        phrase = " ".join(keywords)
+        for i, svc in enumerate(services[:3]):
+            s.alarms.append({
+                "name": f"{svc}-health", "state": "ALARM",
+                "reason": f"{phrase} detected on {svc}", "service": svc,
+            })
+            for j in range(3):
+                s.log_events.append({
+                    "timestamp": f"2026-02-12T10:{i:02d}:{j * 7:02d}Z", "service": svc,
+                    "level": "ERROR",
+                    "message": f"{svc}: {phrase} (occurrence {j + 1})",
+                })
+        base = rng.uniform(50, 200)
+        s.metrics = {f"{services[0]}.error_rate": [round(base * (1 + 0.5 * i), 1) for i in range(6)]}
+        s.monitors = [{"name": f"{services[0]} errors", "status": "Alert",
+                       "query": f"avg:{services[0]}.errors > 10"}]
+        s.resources = {"ecs": [{"name": svc, "desiredCount": 3, "runningCount": 3,
+                                "taskDefinition": f"{svc}:1"} for svc in services]}
+        context = case.get("context", "")
+        if context:
+            s.log_events.append({"timestamp": "2026-02-12T10:05:00Z",
+                                 "service": services[0] if services else "unknown",
+                                 "level": "ERROR", "message": context})
+        return s
+
+
+_SCENARIOS = {
+    "redis-conn-exhaustion": SimScenario.redis_exhaustion,
+    "api-gateway-5xx": SimScenario.gateway_5xx,
+}
+
+_current: Optional[SimScenario] = None
+
+
+def set_scenario(scenario: Optional[SimScenario]) -> None:
+    global _current
+    _current = scenario
+
+
+def get_scenario() -> SimScenario:
+    global _current
+    if _current is None:
+        _current = SimScenario.redis_exhaustion()
+    return _current
+
+
+def load_scenario(name: str) -> SimScenario:
+    factory = _SCENARIOS.get(name)
+    if factory is None:
+        raise KeyError(f"unknown scenario '{name}' (have: {sorted(_SCENARIOS)})")
+    return factory()
