@@ -1,0 +1,604 @@
+"""`runbook` CLI.
+
+Parity with reference src/cli.tsx (2464 LoC) command surface: ask @1104,
+chat @1119, investigate @1133 (--verbose/--auto-remediate/--learn/
+--apply-runbook-updates, structured run @586-989 with event rendering and
+free-form fallback @1169-1189), status @1194, init @1208, demo @1240,
+knowledge sync/search/add/validate/stats @1250-1553, deploy @1556,
+config @1587, integrations claude @1666-1995, webhook @1998,
+slack-gateway @2056, mcp serve/tools @2182-2206, operability @2209-2350,
+checkpoint list/show/delete @2353-2461. Ink UI is replaced by plain
+ANSI terminal rendering.
+"""
+from __future__ import annotations
+
+import json
+import os
+import sys
+from typing import Any, Optional
+
+import click
+
+from .config.schema import Config, load_config, set_config_value, validate_config
+from .model.client import create_llm_client
+
+BOLD = "\033[1m"
+DIM = "\033[2m"
+RESET = "\033[0m"
+GREEN = "\033[32m"
+YELLOW = "\033[33m"
+RED = "\033[31m"
+CYAN = "\033[36m"
+
+
+def _echo(s: str = "") -> None:
+    click.echo(s)
+
+
+def _build_runtime(config: Config, scenario: Optional[str] = None,
+                   provider_override: Optional[str] = None) -> dict[str, Any]:
+    """Build retriever + tools + llm (reference createRuntimeAgent cli.tsx:87-109)."""
+    from .knowledge.indexer.embedder import create_embedder
+    from .knowledge.retriever.default import create_retriever
+    from .providers.simulation import load_scenario, set_scenario
+    from .skills.registry import get_skill_registry
+    from .tools.registry import ToolRegistry, get_runtime_tools
+
+    if scenario:
+        set_scenario(load_scenario(scenario))
+    retriever = create_retriever(embedder=create_embedder({"backend": "auto"}))
+    skills = get_skill_registry()
+    skills.load_user_skills()
+    llm_cfg = config.llm.model_dump(by_alias=False)
+    if provider_override:
+        llm_cfg["provider"] = provider_override
+    llm = create_llm_client(llm_cfg)
+    registry = ToolRegistry(knowledge_retriever=retriever, skill_registry=skills, llm=llm)
+    tools = get_runtime_tools(registry, config.providers.model_dump())
+    return {"retriever": retriever, "registry": registry, "tools": tools, "llm": llm,
+            "skills": skills}
+
+
+@click.group()
+@click.option("--config", "config_path", default=None, help="Path to config.yaml")
+@click.pass_context
+def cli(ctx: click.Context, config_path: Optional[str]) -> None:
+    """Runbook — MI355X-native AI SRE investigation agent."""
+    ctx.ensure_object(dict)
+    ctx.obj["config"] = load_config(config_path)
+
+
+# -- ask / chat ----------------------------------------------------------------
+
+@cli.command()
+@click.argument("query")
+@click.option("--provider", default=None, help="LLM provider override (local/mock)")
+@click.option("--scenario", default=None, help="Simulated scenario name")
+@click.pass_context
+def ask(ctx: click.Context, query: str, provider: Optional[str], scenario: Optional[str]) -> None:
+    """One-shot free-form investigation of a question."""
+    from .agent.agent import Agent
+    from .agent.types import AgentConfig, EventType
+
+    rt = _build_runtime(ctx.obj["config"], scenario, provider)
+    agent = Agent(llm=rt["llm"], tools=rt["tools"], knowledge_retriever=rt["retriever"],
+                  config=AgentConfig(), scratchpad_dir=".runbook/scratchpad")
+    for event in agent.run(query):
+        _render_agent_event(event)
+
+
+def _render_agent_event(event: Any) -> None:
+    from .agent.types import EventType
+
+    t = event.type
+    if t == EventType.THINKING:
+        _echo(f"{DIM}💭 {event.data.get('text', '')[:200]}{RESET}")
+    elif t == EventType.TOOL_START:
+        _echo(f"{CYAN}🔧 {event.data.get('tool')}{RESET} {DIM}{json.dumps(event.data.get('args', {}))[:120]}{RESET}")
+    elif t == EventType.TOOL_END:
+        _echo(f"   {GREEN}✓{RESET} {event.data.get('summary', '')[:160]} "
+              f"{DIM}[{event.data.get('resultId', '')}]{RESET}")
+    elif t == EventType.TOOL_ERROR:
+        _echo(f"   {RED}✗ {event.data.get('error', '')[:160]}{RESET}")
+    elif t == EventType.TOOL_LIMIT:
+        _echo(f"   {YELLOW}⚠ {event.data.get('reason', '')[:160]}{RESET}")
+    elif t == EventType.KNOWLEDGE_RETRIEVED:
+        _echo(f"{DIM}📚 retrieved {event.data.get('count')} knowledge docs{RESET}")
+    elif t == EventType.CONTEXT_CLEARED:
+        _echo(f"{DIM}🧹 compacted context ({event.data.get('cleared')} results cleared){RESET}")
+    elif t == EventType.ANSWER_FINAL:
+        _echo("\n" + event.data.get("text", ""))
+    elif t == EventType.DONE:
+        _echo(f"\n{DIM}done in {event.data.get('iterations')} iterations{RESET}")
+
+
+@cli.command()
+@click.option("--provider", default=None)
+@click.pass_context
+def chat(ctx: click.Context, provider: Optional[str]) -> None:
+    """Interactive chat REPL with conversation memory."""
+    from .agent.agent import Agent
+    from .agent.conversation_memory import ConversationMemory
+    from .agent.types import AgentConfig
+
+    rt = _build_runtime(ctx.obj["config"], None, provider)
+    memory = ConversationMemory(summarize_after_messages=16, llm=rt["llm"])
+    agent = Agent(llm=rt["llm"], tools=rt["tools"], knowledge_retriever=rt["retriever"],
+                  config=AgentConfig(), scratchpad_dir=".runbook/scratchpad",
+                  conversation_memory=memory)
+    _echo(f"{BOLD}runbook chat{RESET} — type 'exit' to quit")
+    while True:
+        try:
+            query = input(f"{BOLD}> {RESET}").strip()
+        except (EOFError, KeyboardInterrupt):
+            break
+        if query.lower() in ("exit", "quit", ""):
+            if query:
+                break
+            continue
+        for event in agent.run(query):
+            _render_agent_event(event)
+
+
+# -- investigate ----------------------------------------------------------------
+
+@cli.command()
+@click.argument("incident_id")
+@click.option("--verbose", is_flag=True)
+@click.option("--auto-remediate", is_flag=True)
+@click.option("--learn", is_flag=True, help="Run the learning loop afterwards")
+@click.option("--apply-runbook-updates", is_flag=True)
+@click.option("--provider", default=None)
+@click.option("--scenario", default=None)
+@click.option("--checkpoint/--no-checkpoint", "do_checkpoint", default=True)
+@click.pass_context
+def investigate(ctx: click.Context, incident_id: str, verbose: bool, auto_remediate: bool,
+                learn: bool, apply_runbook_updates: bool, provider: Optional[str],
+                scenario: Optional[str], do_checkpoint: bool) -> None:
+    """Structured hypothesis-driven investigation of an incident."""
+    from .agent.orchestrator import InvestigationOrchestrator
+    from .session.checkpoint import CheckpointStore, checkpoint_from_machine
+
+    config: Config = ctx.obj["config"]
+    rt = _build_runtime(config, scenario or "redis-conn-exhaustion", provider)
+
+    def render(e: Any) -> None:
+        d = e.data
+        if e.type == "phase":
+            _echo(f"\n{BOLD}▶ {d['to'].upper()}{RESET}")
+        elif e.type == "hypothesis":
+            _echo(f"  💡 [{d.get('priority')}] {d.get('statement')}")
+        elif e.type == "query":
+            mark = GREEN + "✓" + RESET if d.get("ok") else RED + "✗" + RESET
+            _echo(f"  {mark} {d.get('tool')} {DIM}({d.get('purpose', '')}){RESET}")
+        elif e.type == "evaluated":
+            _echo(f"  ⚖ {d.get('action')} (confidence {d.get('confidence'):.2f}) "
+                  f"{DIM}{d.get('hypothesis', '')[:80]}{RESET}")
+        elif e.type == "conclusion":
+            _echo(f"\n{GREEN}✅ Root cause:{RESET} {d.get('rootCause')} "
+                  f"{DIM}[{d.get('confidence')}]{RESET}")
+        elif e.type == "remediation_plan":
+            _echo(f"  🛠 plan: {d.get('summary')} ({d.get('steps')} steps)")
+        elif verbose:
+            _echo(f"{DIM}  {e.type}: {json.dumps(d, default=str)[:160]}{RESET}")
+
+    try:
+        orch = InvestigationOrchestrator(
+            llm=rt["llm"], tool_executor=rt["registry"],
+            available_tools=set(t.name for t in rt["tools"]),
+            knowledge_retriever=rt["retriever"],
+            max_iterations=ctx.obj["config"].agent.max_iterations * 2,
+            auto_remediate=auto_remediate,
+            approval_callback=lambda step: click.confirm(
+                f"Approve {step.get('risk')}-risk step: {step.get('description')}?", default=False),
+        )
+        orch.on(render)
+        result = orch.investigate(f"Investigate incident {incident_id}", incident_id=incident_id)
+        if do_checkpoint:
+            store = CheckpointStore()
+            cp = checkpoint_from_machine(orch.machine, label="final")
+            store.save(cp)
+            _echo(f"{DIM}checkpoint saved: {cp.checkpoint_id}{RESET}")
+        _echo("\n" + result.summary)
+        _echo(f"\n{DIM}duration: {result.duration_ms} ms · LLM calls: "
+              f"{orch.stats['llm_calls']} · tool calls: {orch.stats['tool_calls']}{RESET}")
+        if learn:
+            from .learning.loop import run_learning_loop
+
+            _echo(f"\n{BOLD}▶ LEARNING{RESET}")
+            artifacts = run_learning_loop(rt["llm"], result.to_dict(),
+                                          retriever=rt["retriever"],
+                                          apply_updates=apply_runbook_updates)
+            _echo(f"  postmortem: {artifacts['postmortemPath']}")
+            _echo(f"  suggestions: {len(artifacts['suggestions'])} "
+                  f"(applied {len(artifacts['applied'])}, proposed {len(artifacts['proposed'])})")
+    except Exception as e:  # noqa: BLE001 — fall back to free-form agent (reference cli.tsx:1169-1189)
+        _echo(f"{YELLOW}structured investigation failed ({e}); falling back to free-form agent{RESET}")
+        from .agent.agent import Agent
+        from .agent.types import AgentConfig
+
+        agent = Agent(llm=rt["llm"], tools=rt["tools"], knowledge_retriever=rt["retriever"],
+                      config=AgentConfig(), scratchpad_dir=".runbook/scratchpad")
+        for event in agent.run(f"Investigate incident {incident_id}", incident_id=incident_id):
+            _render_agent_event(event)
+
+
+# -- demo / status / init --------------------------------------------------------
+
+@cli.command()
+@click.option("--fast", is_flag=True, help="3x speed")
+def demo(fast: bool) -> None:
+    """Scripted Redis-connection-exhaustion investigation (no GPU, no keys)."""
+    from .demo.runner import format_step, run_demo
+
+    for step in run_demo(fast=fast):
+        _echo(format_step(step))
+
+
+@cli.command()
+@click.pass_context
+def status(ctx: click.Context) -> None:
+    """Show config, providers, knowledge and GPU status."""
+    config: Config = ctx.obj["config"]
+    problems = validate_config(config)
+    _echo(f"{BOLD}runbook status{RESET}")
+    _echo(f"  config: {'valid' if not problems else 'INVALID'}")
+    for p in problems:
+        _echo(f"    {RED}✗ {p}{RESET}")
+    _echo(f"  llm: {config.llm.provider}/{config.llm.model} "
+          f"TP={config.llm.tensor_parallel} dtype={config.llm.dtype}")
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            name = torch.cuda.get_device_name(0)
+            count = torch.cuda.device_count()
+            free, total = torch.cuda.mem_get_info(0)
+            _echo(f"  gpu: {count}× {name} ({free / 2**30:.0f}/{total / 2**30:.0f} GiB free)")
+        else:
+            _echo("  gpu: none visible (CPU mode)")
+    except Exception:  # noqa: BLE001
+        _echo("  gpu: torch unavailable")
+    if os.path.isdir(".runbook"):
+        from .knowledge.retriever.default import create_retriever
+
+        r = create_retriever(in_memory=False)
+        _echo(f"  knowledge: {json.dumps(r.stats().get('byType', {}))}")
+    else:
+        _echo("  knowledge: not initialized (run `runbook init`)")
+
+
+@cli.command()
+@click.option("--template", type=click.Choice(["ecs-rds", "serverless", "enterprise"]),
+              default="ecs-rds")
+def init(template: str) -> None:
+    """Initialize .runbook/ with a config template."""
+    from .config.onboarding import quick_setup
+
+    paths = quick_setup(template)
+    for p in paths:
+        _echo(f"  wrote {p}")
+    _echo(f"{GREEN}initialized .runbook/ ({template}){RESET}")
+
+
+# -- knowledge -------------------------------------------------------------------
+
+@cli.group()
+def knowledge() -> None:
+    """Knowledge base commands."""
+
+
+@knowledge.command("sync")
+@click.option("--since", type=float, default=None)
+def knowledge_sync(since: Optional[float]) -> None:
+    from .knowledge.indexer.embedder import create_embedder
+    from .knowledge.retriever.default import create_retriever
+
+    r = create_retriever(embedder=create_embedder())
+    counts = r.sync(since=since)
+    _echo(f"synced {counts['documents']} documents / {counts['chunks']} chunks")
+
+
+@knowledge.command("search")
+@click.argument("query")
+@click.option("--limit", default=5)
+@click.option("--type", "doc_type", default=None)
+def knowledge_search(query: str, limit: int, doc_type: Optional[str]) -> None:
+    from .knowledge.indexer.embedder import create_embedder
+    from .knowledge.retriever.default import create_retriever
+
+    r = create_retriever(embedder=create_embedder())
+    for hit in r.search(query, limit=limit, doc_type=doc_type):
+        _echo(f"{BOLD}{hit['title']}{RESET} {DIM}({hit['type']}, score {hit['score']:.3f}){RESET}")
+        _echo(f"  {hit['content'][:160]}")
+
+
+@knowledge.command("add")
+@click.argument("path")
+def knowledge_add(path: str) -> None:
+    from .knowledge.retriever.default import create_retriever
+    from .knowledge.sources.filesystem import load_markdown
+
+    r = create_retriever()
+    doc = load_markdown(path)
+    r.store.upsert_document(doc)
+    _echo(f"added '{doc.title}' ({len(doc.chunks)} chunks)")
+
+
+@knowledge.command("validate")
+def knowledge_validate() -> None:
+    from .knowledge.retriever.default import create_retriever
+
+    r = create_retriever()
+    r.ensure_initialized()
+    stats = r.stats()
+    _echo(f"{GREEN}knowledge base valid{RESET}: {json.dumps(stats)}")
+
+
+@knowledge.command("stats")
+def knowledge_stats() -> None:
+    from .knowledge.retriever.default import create_retriever
+
+    r = create_retriever()
+    r.ensure_initialized()
+    _echo(json.dumps(r.stats(), indent=1))
+
+
+# -- config ----------------------------------------------------------------------
+
+@cli.command("config")
+@click.option("--set", "set_kv", default=None, help="a.b.c=value dotted write")
+@click.option("--show", is_flag=True)
+@click.pass_context
+def config_cmd(ctx: click.Context, set_kv: Optional[str], show: bool) -> None:
+    """Show or modify configuration."""
+    if set_kv:
+        key, _, value = set_kv.partition("=")
+        set_config_value(".runbook/config.yaml", key.strip(), value.strip())
+        _echo(f"set {key.strip()} = {value.strip()}")
+        return
+    config: Config = ctx.obj["config"]
+    _echo(json.dumps(config.model_dump(), indent=1, default=str))
+
+
+# -- eval ------------------------------------------------------------------------
+
+@cli.command("eval")
+@click.option("--fixtures", default="examples/evals/investigation-fixtures.sample.json")
+@click.option("--offline", is_flag=True, help="Score fixture mockResults only")
+@click.option("--concurrency", default=1)
+@click.option("--provider", default=None)
+@click.option("--report", "report_path", default=None)
+@click.pass_context
+def eval_cmd(ctx: click.Context, fixtures: str, offline: bool, concurrency: int,
+             provider: Optional[str], report_path: Optional[str]) -> None:
+    """Run the investigation benchmark."""
+    from .evals.benchmark import load_fixtures, run_benchmark
+
+    fx = load_fixtures(fixtures)
+    config: Config = ctx.obj["config"]
+    llm_cfg = config.llm.model_dump(by_alias=False)
+    if provider:
+        llm_cfg["provider"] = provider
+
+    def llm_factory() -> Any:
+        return create_llm_client(llm_cfg)
+
+    report = run_benchmark(fx, llm_factory=None if offline else llm_factory,
+                           offline=offline, concurrency=concurrency)
+    for case in report["cases"]:
+        mark = GREEN + "PASS" + RESET if case["passed"] else RED + "FAIL" + RESET
+        _echo(f"[{mark}] {case['id']}: {case['score']['overall']:.2f} "
+              f"{DIM}({case['durationMs']} ms){RESET}")
+    _echo(f"\npass rate {report['passRate']:.0%} · avg score {report['averageOverallScore']:.2f}"
+          f" · wall {report['wallMs']} ms")
+    if report_path:
+        with open(report_path, "w", encoding="utf-8") as f:
+            json.dump(report, f, indent=1)
+        _echo(f"{DIM}report → {report_path}{RESET}")
+    sys.exit(0 if report["failed"] == 0 else 1)
+
+
+# -- checkpoint ------------------------------------------------------------------
+
+@cli.group()
+def checkpoint() -> None:
+    """Investigation checkpoint management."""
+
+
+@checkpoint.command("list")
+@click.argument("investigation_id", required=False)
+def checkpoint_list(investigation_id: Optional[str]) -> None:
+    from .session.checkpoint import CheckpointStore
+
+    store = CheckpointStore()
+    if investigation_id is None:
+        for inv in store.list_investigations():
+            cps = store.list(inv)
+            _echo(f"{inv}: {len(cps)} checkpoints")
+        return
+    for cp in store.list(investigation_id):
+        _echo(cp.format())
+
+
+@checkpoint.command("show")
+@click.argument("investigation_id")
+@click.argument("checkpoint_id", required=False)
+def checkpoint_show(investigation_id: str, checkpoint_id: Optional[str]) -> None:
+    from .session.checkpoint import CheckpointStore
+
+    store = CheckpointStore()
+    cp = (store.load(investigation_id, checkpoint_id) if checkpoint_id
+          else store.load_latest(investigation_id))
+    if cp is None:
+        _echo(f"{RED}no checkpoint found{RESET}")
+        sys.exit(1)
+    _echo(cp.format())
+
+
+@checkpoint.command("delete")
+@click.argument("investigation_id")
+@click.argument("checkpoint_id", required=False)
+def checkpoint_delete(investigation_id: str, checkpoint_id: Optional[str]) -> None:
+    from .session.checkpoint import CheckpointStore
+
+    removed = CheckpointStore().delete(investigation_id, checkpoint_id)
+    _echo(f"deleted {removed} checkpoint(s)")
+
+
+# -- mcp -------------------------------------------------------------------------
+
+@cli.group()
+def mcp() -> None:
+    """MCP server (stdio JSON-RPC)."""
+
+
+@mcp.command("serve")
+def mcp_serve() -> None:
+    from .mcp.server import run_stdio_server
+
+    run_stdio_server()
+
+
+@mcp.command("tools")
+def mcp_tools() -> None:
+    from .mcp.server import MCPServer
+
+    server = MCPServer()
+    for t in server.tool_specs():
+        _echo(f"{BOLD}{t['name']}{RESET}: {t['description']}")
+
+
+# -- surfaces: slack gateway / webhook / integrations / operability ---------------
+
+@cli.command("slack-gateway")
+@click.option("--port", default=3030)
+@click.pass_context
+def slack_gateway(ctx: click.Context, port: int) -> None:
+    """Run the Slack events gateway (HTTP)."""
+    from .slack.gateway import SlackGateway
+
+    rt = _build_runtime(ctx.obj["config"])
+    gw = SlackGateway(config=ctx.obj["config"].incident.slack, runtime=rt)
+    _echo(f"slack gateway listening on :{port}")
+    gw.serve(port=port)
+
+
+@cli.command("webhook")
+@click.option("--port", default=3031)
+def webhook(port: int) -> None:
+    """Run the Slack approval-button webhook server."""
+    from .webhooks.slack_webhook import ApprovalWebhook
+
+    _echo(f"approval webhook listening on :{port}")
+    ApprovalWebhook().serve(port=port)
+
+
+@cli.group()
+def integrations() -> None:
+    """Integration management (claude hooks, ...)."""
+
+
+@integrations.group()
+def claude() -> None:
+    """Claude Code hook integration."""
+
+
+@claude.command("enable")
+@click.option("--scope", type=click.Choice(["project", "user"]), default="project")
+def claude_enable(scope: str) -> None:
+    from .integrations.claude_hooks import install_hooks
+
+    path = install_hooks(scope)
+    _echo(f"{GREEN}claude hooks installed{RESET} → {path}")
+
+
+@claude.command("status")
+def claude_status() -> None:
+    from .integrations.claude_hooks import hooks_status
+
+    _echo(json.dumps(hooks_status(), indent=1))
+
+
+@claude.command("disable")
+@click.option("--scope", type=click.Choice(["project", "user"]), default="project")
+def claude_disable(scope: str) -> None:
+    from .integrations.claude_hooks import uninstall_hooks
+
+    uninstall_hooks(scope)
+    _echo("claude hooks removed")
+
+
+@claude.command("hook")
+def claude_hook() -> None:
+    """Hook entrypoint: reads a hook event from stdin, writes a response."""
+    from .integrations.hook_handlers import handle_stdin
+
+    handle_stdin()
+
+
+@claude.command("learn")
+@click.argument("session_id")
+@click.pass_context
+def claude_learn(ctx: click.Context, session_id: str) -> None:
+    from .integrations.session_store import SessionStore
+    from .learning.claude_session_ingestion import ingest_session
+
+    rt = _build_runtime(ctx.obj["config"])
+    result = ingest_session(SessionStore(), session_id, rt["llm"], rt["retriever"])
+    _echo(json.dumps({k: v for k, v in result.items() if k != "postmortem"}, indent=1,
+                     default=str))
+
+
+@cli.group()
+def operability() -> None:
+    """Operability-context ingestion (agent change claims)."""
+
+
+@operability.command("ingest")
+@click.argument("phase", type=click.Choice(["start", "checkpoint", "end"]))
+@click.option("--summary", default="")
+@click.option("--files", default="")
+@click.option("--services", default="")
+def operability_ingest(phase: str, summary: str, files: str, services: str) -> None:
+    from .integrations.operability_ingestion import ingest_claim
+
+    claim = ingest_claim(phase, summary=summary,
+                         files=[f for f in files.split(",") if f],
+                         services=[s for s in services.split(",") if s])
+    _echo(json.dumps(claim, indent=1, default=str))
+
+
+@operability.command("replay")
+def operability_replay() -> None:
+    from .integrations.operability_ingestion import replay_spool
+
+    n = replay_spool()
+    _echo(f"replayed {n} spooled claims")
+
+
+@operability.command("status")
+def operability_status() -> None:
+    from .integrations.operability_ingestion import spool_status
+
+    _echo(json.dumps(spool_status(), indent=1))
+
+
+@cli.command()
+@click.argument("service")
+@click.option("--version", default="latest")
+@click.pass_context
+def deploy(ctx: click.Context, service: str, version: str) -> None:
+    """Deploy a service via the deploy-service skill (approval-gated)."""
+    rt = _build_runtime(ctx.obj["config"])
+    out = rt["registry"].execute("skill", {"action": "execute", "name": "deploy-service",
+                                           "params": {"service": service, "version": version}})
+    _echo(json.dumps(out, indent=1, default=str))
+
+
+def main() -> None:
+    cli(obj={})
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,133 @@
+"""Claude Code hook event handlers.
+
+Parity with reference src/integrations/hook-handlers.ts (508 LoC):
+SessionStart knowledge-stats banner @244-282; UserPromptSubmit context
+injection — service/symptom extraction -> knowledge search ->
+systemMessage @288-374; PreToolUse dangerous-command blocking via regex
+list returning {continue: false, stopReason} @380-417; PostToolUse/Stop
+session-state tracking @423-455; stdin dispatcher @455-508.
+"""
+from __future__ import annotations
+
+import json
+import re
+import sys
+from typing import Any, Optional
+
+# Dangerous-command regex list (reference @380-417).
+DANGEROUS_PATTERNS = [
+    re.compile(r"\brm\s+-rf\s+/(?:\s|$)"),
+    re.compile(r"\bkubectl\s+delete\s+(deployment|pod|service|namespace)\b"),
+    re.compile(r"\baws\s+(ec2|ecs|rds)\s+\S*(terminate|delete|stop)\S*"),
+    re.compile(r"\bdocker\s+(rm|stop|kill)\s+.*-f"),
+    re.compile(r"\bdrop\s+(database|table)\b", re.IGNORECASE),
+    re.compile(r"\bmkfs\.\w+\s"),
+]
+
+_SERVICE_RE = re.compile(r"\b([a-z][a-z0-9]*(?:-[a-z0-9]+)+)\b")
+_SYMPTOM_WORDS = ("latency", "timeout", "error", "5xx", "oom", "crash", "exhausted",
+                  "deadlock", "throttl", "unavailable", "degraded")
+
+
+def handle_session_start(payload: dict[str, Any], retriever: Any = None) -> dict[str, Any]:
+    """Knowledge-stats banner (reference @244-282)."""
+    stats: dict[str, Any] = {}
+    if retriever is not None:
+        try:
+            stats = retriever.stats()
+        except Exception:  # noqa: BLE001
+            stats = {}
+    if not stats:
+        return {"continue": True}
+    banner = (f"Runbook knowledge base: {stats.get('documents', 0)} docs "
+              f"({json.dumps(stats.get('byType', {}))}) — "
+              "search with the runbook MCP tools.")
+    return {"continue": True, "systemMessage": banner}
+
+
+def handle_user_prompt_submit(payload: dict[str, Any], retriever: Any = None) -> dict[str, Any]:
+    """Context injection (reference @288-374)."""
+    prompt = str(payload.get("prompt", payload.get("user_prompt", "")))
+    services = list(dict.fromkeys(_SERVICE_RE.findall(prompt)))[:4]
+    symptoms = [w for w in _SYMPTOM_WORDS if w in prompt.lower()][:4]
+    if retriever is None or not (services or symptoms):
+        return {"continue": True}
+    query = " ".join(services + symptoms)
+    try:
+        hits = retriever.search(query, limit=3)
+    except Exception:  # noqa: BLE001
+        return {"continue": True}
+    if not hits:
+        return {"continue": True}
+    lines = ["Relevant operational knowledge:"]
+    for h in hits:
+        lines.append(f"- {h['title']}: {str(h['content'])[:160]}")
+    return {"continue": True, "systemMessage": "\n".join(lines)}
+
+
+def handle_pre_tool_use(payload: dict[str, Any]) -> dict[str, Any]:
+    """Dangerous-command blocking (reference @380-417)."""
+    tool_input = payload.get("tool_input", {}) or {}
+    command = str(tool_input.get("command", ""))
+    for pattern in DANGEROUS_PATTERNS:
+        if pattern.search(command):
+            return {"continue": False,
+                    "stopReason": f"runbook safety hook: blocked dangerous command "
+                                  f"matching /{pattern.pattern}/"}
+    return {"continue": True}
+
+
+def handle_post_tool_use(payload: dict[str, Any], store: Any = None) -> dict[str, Any]:
+    if store is not None:
+        try:
+            store.append_event(payload.get("session_id", "unknown"),
+                               {"kind": "tool_use", **payload})
+        except Exception:  # noqa: BLE001
+            pass
+    return {"continue": True}
+
+
+def handle_stop(payload: dict[str, Any], store: Any = None) -> dict[str, Any]:
+    if store is not None:
+        try:
+            store.append_event(payload.get("session_id", "unknown"),
+                               {"kind": "stop", **payload})
+        except Exception:  # noqa: BLE001
+            pass
+    return {"continue": True}
+
+
+def dispatch(payload: dict[str, Any], retriever: Any = None, store: Any = None) -> dict[str, Any]:
+    """Stdin dispatcher (reference @455-508)."""
+    event = payload.get("hook_event_name", payload.get("event", ""))
+    if event == "SessionStart":
+        return handle_session_start(payload, retriever)
+    if event == "UserPromptSubmit":
+        return handle_user_prompt_submit(payload, retriever)
+    if event == "PreToolUse":
+        return handle_pre_tool_use(payload)
+    if event == "PostToolUse":
+        return handle_post_tool_use(payload, store)
+    if event == "Stop":
+        return handle_stop(payload, store)
+    return {"continue": True}
+
+
+def handle_stdin() -> None:
+    try:
+        payload = json.load(sys.stdin)
+    except json.JSONDecodeError:
+        print(json.dumps({"continue": True}))
+        return
+    retriever = None
+    store = None
+    try:
+        from ..knowledge.retriever.default import create_retriever
+
+        retriever = create_retriever()
+        from .session_store import SessionStore
+
+        store = SessionStore()
+    except Exception:  # noqa: BLE001
+        pass
+    print(json.dumps(dispatch(payload, retriever, store), default=str))

@@ -1,0 +1,174 @@
+"""Slack events gateway: @mention command parsing -> full agent run.
+
+Parity with reference src/slack/gateway.ts (561 LoC): @mention command
+parser infra/knowledge/deploy/investigate (L95-121), channel/user
+allow-lists + signature verification (L190-258), HTTP Events API
+(L384-530), runs a full Agent per request (L288-324), event dedupe cache
+(L70). Socket Mode has no equivalent without egress; the HTTP Events API
+endpoint is served with the stdlib http.server.
+"""
+from __future__ import annotations
+
+import hashlib
+import hmac
+import json
+import re
+import time
+from http.server import BaseHTTPRequestHandler, HTTPServer
+from typing import Any, Optional
+
+_MENTION_RE = re.compile(r"<@[A-Z0-9]+>\s*", re.IGNORECASE)
+
+
+def parse_command(text: str) -> dict[str, Any]:
+    """Command parser (reference L95-121): infra/knowledge/deploy/investigate/help."""
+    clean = _MENTION_RE.sub("", text or "").strip()
+    lowered = clean.lower()
+    for cmd in ("investigate", "knowledge", "infra", "deploy", "status", "help"):
+        if lowered.startswith(cmd):
+            return {"command": cmd, "args": clean[len(cmd):].strip(), "raw": clean}
+    return {"command": "ask", "args": clean, "raw": clean}
+
+
+def verify_signature(signing_secret: str, timestamp: str, body: bytes,
+                     signature: str, max_age_s: float = 300.0) -> bool:
+    """Slack v0 signature verification (reference L190-258)."""
+    try:
+        if abs(time.time() - float(timestamp)) > max_age_s:
+            return False
+    except (TypeError, ValueError):
+        return False
+    base = f"v0:{timestamp}:{body.decode('utf-8', 'replace')}".encode()
+    expected = "v0=" + hmac.new(signing_secret.encode(), base, hashlib.sha256).hexdigest()
+    return hmac.compare_digest(expected, signature or "")
+
+
+class SlackGateway:
+    def __init__(self, config: Optional[dict[str, Any]] = None,
+                 runtime: Optional[dict[str, Any]] = None) -> None:
+        self.config = config or {}
+        self.runtime = runtime or {}
+        self.allowed_channels = set(self.config.get("allowedChannels", []) or [])
+        self.allowed_users = set(self.config.get("allowedUsers", []) or [])
+        self.signing_secret = self.config.get("signingSecret", "")
+        self._seen_events: dict[str, float] = {}  # dedupe cache (reference L70)
+        self.replies: list[dict[str, Any]] = []
+
+    # -- auth ----------------------------------------------------------------
+
+    def allowed(self, channel: str, user: str) -> bool:
+        if self.allowed_channels and channel not in self.allowed_channels:
+            return False
+        if self.allowed_users and user not in self.allowed_users:
+            return False
+        return True
+
+    def dedupe(self, event_id: str, ttl_s: float = 600.0) -> bool:
+        """True if this event was already seen."""
+        now = time.time()
+        for k, t in list(self._seen_events.items()):
+            if now - t > ttl_s:
+                del self._seen_events[k]
+        if event_id in self._seen_events:
+            return True
+        self._seen_events[event_id] = now
+        return False
+
+    # -- event handling (reference L288-324: full agent run) -------------------
+
+    def handle_event(self, event: dict[str, Any]) -> dict[str, Any]:
+        channel = event.get("channel", "")
+        user = event.get("user", "")
+        if not self.allowed(channel, user):
+            return {"ok": False, "error": "channel or user not allowed"}
+        event_id = event.get("event_ts", event.get("ts", str(time.time())))
+        if self.dedupe(event_id):
+            return {"ok": True, "deduped": True}
+        parsed = parse_command(event.get("text", ""))
+        reply = self._run_command(parsed)
+        record = {"channel": channel, "threadTs": event.get("ts", ""), "text": reply}
+        self.replies.append(record)
+        return {"ok": True, "command": parsed["command"], "reply": reply}
+
+    def _run_command(self, parsed: dict[str, Any]) -> str:
+        cmd, args = parsed["command"], parsed["args"]
+        if cmd == "help":
+            return ("Commands: investigate <incident>, knowledge <query>, infra, "
+                    "deploy <service>, status, or just ask a question.")
+        if cmd == "knowledge":
+            retriever = self.runtime.get("retriever")
+            if retriever is None:
+                return "knowledge base unavailable"
+            hits = retriever.search(args, limit=3)
+            if not hits:
+                return f"no knowledge found for '{args}'"
+            return "\n".join(f"• {h['title']}: {h['content'][:120]}" for h in hits)
+        if cmd == "infra":
+            registry = self.runtime.get("registry")
+            if registry is None:
+                return "infra tools unavailable"
+            alarms = registry.execute("cloudwatch_alarms", {"state": "ALARM"})
+            return f"{alarms['count']} alarms firing: " + \
+                ", ".join(a["name"] for a in alarms["alarms"][:5])
+        if cmd == "status":
+            return "runbook gateway up"
+        if cmd == "deploy":
+            return f"deploys require approval — run `runbook deploy {args}` from a terminal"
+        # investigate / ask: full agent run
+        llm = self.runtime.get("llm")
+        tools = self.runtime.get("tools", [])
+        retriever = self.runtime.get("retriever")
+        if llm is None:
+            return "agent unavailable (no LLM configured)"
+        from ..agent.agent import Agent
+        from ..agent.types import AgentConfig, EventType
+
+        agent = Agent(llm=llm, tools=tools, knowledge_retriever=retriever,
+                      config=AgentConfig(max_iterations=5))
+        answer = ""
+        for ev in agent.run(args or parsed["raw"]):
+            if ev.type == EventType.ANSWER_FINAL:
+                answer = ev.data.get("text", "")
+        return answer[:2800] or "investigation produced no answer"
+
+    # -- HTTP server (reference L384-530) ----------------------------------------
+
+    def serve(self, port: int = 3030) -> None:
+        gateway = self
+
+        class Handler(BaseHTTPRequestHandler):
+            def do_POST(self) -> None:  # noqa: N802
+                length = int(self.headers.get("Content-Length", 0))
+                body = self.rfile.read(length)
+                if gateway.signing_secret:
+                    ok = verify_signature(
+                        gateway.signing_secret,
+                        self.headers.get("X-Slack-Request-Timestamp", ""),
+                        body,
+                        self.headers.get("X-Slack-Signature", ""),
+                    )
+                    if not ok:
+                        self.send_response(401)
+                        self.end_headers()
+                        return
+                try:
+                    payload = json.loads(body)
+                except json.JSONDecodeError:
+                    self.send_response(400)
+                    self.end_headers()
+                    return
+                if payload.get("type") == "url_verification":
+                    out = json.dumps({"challenge": payload.get("challenge", "")}).encode()
+                else:
+                    event = payload.get("event", {})
+                    result = gateway.handle_event(event)
+                    out = json.dumps(result, default=str).encode()
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.end_headers()
+                self.wfile.write(out)
+
+            def log_message(self, *args: Any) -> None:
+                pass
+
+        HTTPServer(("127.0.0.1", port), Handler).serve_forever()

@@ -1,0 +1,338 @@
+"""Surface tests: CLI (demo/knowledge/eval offline), MCP server, slack
+gateway parsing/auth, webhook approvals, hooks, checkpoints, learning,
+operability ingestion."""
+import json
+import os
+
+import pytest
+from click.testing import CliRunner
+
+from runbookai_amd.cli import cli
+
+
+@pytest.fixture
+def runner(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    # examples dir is used by create_retriever default sources; copy one runbook
+    os.makedirs("examples/runbooks", exist_ok=True)
+    with open("examples/runbooks/redis.md", "w") as f:
+        f.write("---\ntitle: Redis runbook\ntype: runbook\nservices: [redis]\n---\n"
+                "# Redis runbook\n\n## Mitigation\n1. raise pool size\n")
+    return CliRunner()
+
+
+class TestCLI:
+    def test_demo_fast(self, runner, monkeypatch):
+        import runbookai_amd.demo.runner as dr
+
+        monkeypatch.setattr(dr.time, "sleep", lambda s: None)
+        result = runner.invoke(cli, ["demo", "--fast"], obj={})
+        assert result.exit_code == 0
+        assert "Redis" in result.output or "redis" in result.output
+        assert "CONFIRMED" in result.output
+
+    def test_knowledge_sync_and_search(self, runner):
+        result = runner.invoke(cli, ["knowledge", "sync"], obj={})
+        assert result.exit_code == 0, result.output
+        assert "synced" in result.output
+        result = runner.invoke(cli, ["knowledge", "search", "redis pool"], obj={})
+        assert result.exit_code == 0
+        assert "Redis runbook" in result.output
+
+    def test_init_and_status(self, runner):
+        result = runner.invoke(cli, ["init", "--template", "ecs-rds"], obj={})
+        assert result.exit_code == 0, result.output
+        assert os.path.exists(".runbook/config.yaml")
+        result = runner.invoke(cli, ["status"], obj={})
+        assert result.exit_code == 0
+        assert "llm:" in result.output
+
+    def test_config_set(self, runner):
+        result = runner.invoke(cli, ["config", "--set", "llm.model=llama3-70b"], obj={})
+        assert result.exit_code == 0
+        import yaml
+
+        with open(".runbook/config.yaml") as f:
+            data = yaml.safe_load(f)
+        assert data["llm"]["model"] == "llama3-70b"
+
+    def test_eval_offline(self, runner, tmp_path):
+        fixtures = {
+            "version": "1.0", "passThreshold": 0.7,
+            "cases": [{"id": "c", "query": "q", "expected": {"rootCauseKeywords": ["x"]},
+                       "mockResult": {"rootCause": "x happened"}}],
+        }
+        fp = tmp_path / "fx.json"
+        fp.write_text(json.dumps(fixtures))
+        result = runner.invoke(cli, ["eval", "--fixtures", str(fp), "--offline"], obj={})
+        assert result.exit_code == 0, result.output
+        assert "PASS" in result.output
+
+    def test_investigate_with_mock_provider(self, runner):
+        # mock LLM yields unparseable output -> graceful fallback path completes
+        result = runner.invoke(cli, ["investigate", "PD-EXAMPLE-001", "--provider", "mock",
+                                     "--no-checkpoint"], obj={})
+        assert result.exit_code == 0, result.output
+        assert "Root cause" in result.output or "Investigation" in result.output
+
+
+class TestMCP:
+    def _server(self):
+        from runbookai_amd.knowledge.indexer.embedder import HashEmbedder
+        from runbookai_amd.knowledge.retriever.default import KnowledgeRetriever
+        from runbookai_amd.knowledge.types import SourceConfig
+        from runbookai_amd.mcp.server import MCPServer
+
+        examples = os.path.join(os.path.dirname(__file__), "..", "examples", "runbooks")
+        r = KnowledgeRetriever(sources=[SourceConfig(kind="filesystem", path=examples)],
+                               embedder=HashEmbedder())
+        return MCPServer(retriever=r)
+
+    def test_initialize_and_list(self):
+        s = self._server()
+        resp = s.handle({"jsonrpc": "2.0", "id": 1, "method": "initialize", "params": {}})
+        assert resp["result"]["serverInfo"]["name"] == "runbook-knowledge"
+        resp = s.handle({"jsonrpc": "2.0", "id": 2, "method": "tools/list"})
+        assert len(resp["result"]["tools"]) == 5
+
+    def test_tools_call_search(self):
+        s = self._server()
+        resp = s.handle({"jsonrpc": "2.0", "id": 3, "method": "tools/call",
+                         "params": {"name": "search_runbooks",
+                                    "arguments": {"query": "redis pool"}}})
+        text = resp["result"]["content"][0]["text"]
+        assert "Redis" in text
+
+    def test_unknown_method(self):
+        s = self._server()
+        resp = s.handle({"jsonrpc": "2.0", "id": 4, "method": "bogus/method"})
+        assert resp["error"]["code"] == -32601
+
+
+class TestSlackGateway:
+    def test_command_parsing(self):
+        from runbookai_amd.slack.gateway import parse_command
+
+        assert parse_command("<@U123> investigate PD-1")["command"] == "investigate"
+        assert parse_command("knowledge redis pool")["command"] == "knowledge"
+        assert parse_command("what is going on?")["command"] == "ask"
+
+    def test_signature_verification(self):
+        import hashlib
+        import hmac
+        import time as _t
+
+        from runbookai_amd.slack.gateway import verify_signature
+
+        secret = "s3cret"
+        ts = str(_t.time())
+        body = b'{"type":"event_callback"}'
+        base = f"v0:{ts}:{body.decode()}".encode()
+        sig = "v0=" + hmac.new(secret.encode(), base, hashlib.sha256).hexdigest()
+        assert verify_signature(secret, ts, body, sig)
+        assert not verify_signature(secret, ts, body, "v0=bad")
+        assert not verify_signature(secret, "123", body, sig)  # too old
+
+    def test_allow_list_and_dedupe(self):
+        from runbookai_amd.slack.gateway import SlackGateway
+
+        gw = SlackGateway(config={"allowedChannels": ["C1"]})
+        out = gw.handle_event({"channel": "C2", "user": "U1", "text": "help", "ts": "1"})
+        assert out["ok"] is False
+        out = gw.handle_event({"channel": "C1", "user": "U1", "text": "help", "ts": "1"})
+        assert out["ok"] is True
+        out2 = gw.handle_event({"channel": "C1", "user": "U1", "text": "help", "ts": "1"})
+        assert out2.get("deduped")
+
+
+class TestWebhook:
+    def test_approval_roundtrip(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import ApprovalWebhook, PendingApprovalStore
+
+        store = PendingApprovalStore(str(tmp_path / "pending"))
+        aid = store.create({"operation": "restart", "resource": "cart-service"})
+        assert store.list_pending()
+        hook = ApprovalWebhook(store)
+        out = hook.handle_interaction({"user": {"username": "alice"},
+                                       "actions": [{"value": f"approve:{aid}"}]})
+        assert out["status"] == "approved"
+        assert store.get(aid)["approver"] == "alice"
+        assert not store.list_pending()
+
+
+class TestHooks:
+    def test_dangerous_command_blocked(self):
+        from runbookai_amd.integrations.hook_handlers import handle_pre_tool_use
+
+        out = handle_pre_tool_use({"tool_input": {"command": "kubectl delete deployment api"}})
+        assert out["continue"] is False
+        assert "blocked" in out["stopReason"]
+
+    def test_safe_command_passes(self):
+        from runbookai_amd.integrations.hook_handlers import handle_pre_tool_use
+
+        assert handle_pre_tool_use({"tool_input": {"command": "kubectl get pods"}})["continue"]
+
+    def test_prompt_context_injection(self):
+        from runbookai_amd.integrations.hook_handlers import handle_user_prompt_submit
+
+        class R:
+            def search(self, q, limit=3):
+                return [{"title": "Redis runbook", "content": "raise pool"}]
+
+        out = handle_user_prompt_submit({"prompt": "checkout-api has latency issues"}, R())
+        assert "Redis runbook" in out.get("systemMessage", "")
+
+    def test_install_status_uninstall(self, tmp_path, monkeypatch):
+        monkeypatch.chdir(tmp_path)
+        from runbookai_amd.integrations import claude_hooks
+
+        path = claude_hooks.install_hooks("project")
+        assert os.path.exists(path)
+        status = claude_hooks.hooks_status()
+        assert status["project"]["enabled"]
+        claude_hooks.uninstall_hooks("project")
+        assert not claude_hooks.hooks_status()["project"]["enabled"]
+
+
+class TestCheckpoints:
+    def test_save_load_latest_delete(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore, InvestigationCheckpoint
+
+        store = CheckpointStore(str(tmp_path / "cp"))
+        for i in range(3):
+            cp = InvestigationCheckpoint(
+                checkpoint_id=CheckpointStore.new_id(), investigation_id="inv-1",
+                phase="investigate", root_cause=f"cause-{i}", created_at=100.0 + i)
+            store.save(cp)
+        assert len(store.list("inv-1")) == 3
+        assert store.load_latest("inv-1").root_cause == "cause-2"
+        assert store.list_investigations() == ["inv-1"]
+        assert store.delete("inv-1") == 3
+
+    def test_cap_50(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore, InvestigationCheckpoint
+
+        store = CheckpointStore(str(tmp_path / "cp"))
+        for i in range(55):
+            store.save(InvestigationCheckpoint(
+                checkpoint_id=CheckpointStore.new_id(), investigation_id="inv-2",
+                created_at=float(i)))
+        assert len(store.list("inv-2")) == 50
+
+    def test_from_machine(self):
+        from runbookai_amd.agent.state_machine import Conclusion, InvestigationStateMachine
+        from runbookai_amd.session.checkpoint import checkpoint_from_machine
+
+        m = InvestigationStateMachine()
+        m.start()
+        m.add_hypothesis("h1")
+        m.transition.__self__.set_conclusion(Conclusion(root_cause="rc", confidence="high",
+                                                        summary=""))
+        cp = checkpoint_from_machine(m, label="test")
+        assert cp.root_cause == "rc"
+        assert len(cp.hypotheses) == 1
+
+
+class TestLearning:
+    def test_learning_loop_with_mock(self, tmp_path):
+        from runbookai_amd.learning.loop import run_learning_loop
+        from runbookai_amd.model.client import MockLLMClient
+
+        llm = MockLLMClient([json.dumps({
+            "postmortem": {"title": "PM: redis outage", "summary": "pool exhausted",
+                           "rootCause": "redis pool exhaustion",
+                           "timeline": ["09:02 deploy"], "impact": "checkout down",
+                           "actionItems": ["alert on pool"]},
+            "knowledgeSuggestions": [
+                {"kind": "new_runbook", "title": "Pool sizing", "content": "size pools",
+                 "services": ["redis"]}],
+        })])
+        result = {"investigationId": "inv-9", "rootCause": "redis pool exhaustion",
+                  "summary": "s", "affectedServices": ["redis"]}
+        out = run_learning_loop(llm, result, runbook_dir=str(tmp_path / ".runbook"))
+        assert os.path.exists(out["postmortemPath"])
+        content = open(out["postmortemPath"]).read()
+        assert "type: postmortem" in content
+        assert "redis pool exhaustion" in content
+        assert len(out["proposed"]) == 1  # apply_updates=False -> proposals
+
+    def test_fallback_draft_on_garbage(self, tmp_path):
+        from runbookai_amd.learning.loop import run_learning_loop
+        from runbookai_amd.model.client import MockLLMClient
+
+        llm = MockLLMClient(["not json at all"])
+        out = run_learning_loop(llm, {"investigationId": "inv-10", "rootCause": "rc"},
+                                runbook_dir=str(tmp_path / ".runbook"))
+        assert out["fallback"]
+        assert os.path.exists(out["postmortemPath"])
+
+    def test_apply_updates_appends_to_matching_runbook(self, tmp_path):
+        from runbookai_amd.knowledge.indexer.embedder import HashEmbedder
+        from runbookai_amd.knowledge.retriever.default import KnowledgeRetriever
+        from runbookai_amd.knowledge.types import SourceConfig
+        from runbookai_amd.learning.loop import run_learning_loop
+        from runbookai_amd.model.client import MockLLMClient
+
+        rb_dir = tmp_path / ".runbook" / "runbooks"
+        rb_dir.mkdir(parents=True)
+        rb = rb_dir / "redis-pool.md"
+        rb.write_text("---\ntitle: Redis pool sizing\ntype: runbook\nservices: [redis]\n---\n"
+                      "# Redis pool sizing\n")
+        retriever = KnowledgeRetriever(
+            sources=[SourceConfig(kind="filesystem", path=str(rb_dir))],
+            embedder=HashEmbedder())
+        retriever.sync()
+        llm = MockLLMClient([json.dumps({
+            "postmortem": {"title": "PM", "summary": "s", "rootCause": "rc"},
+            "knowledgeSuggestions": [
+                {"kind": "update_runbook", "title": "Redis pool sizing update",
+                 "targetRunbook": "Redis pool sizing", "content": "new guidance",
+                 "services": ["redis"]}],
+        })])
+        out = run_learning_loop(llm, {"investigationId": "inv-11", "rootCause": "rc"},
+                                runbook_dir=str(tmp_path / ".runbook"), retriever=retriever,
+                                apply_updates=True)
+        assert out["applied"] == [str(rb)]
+        assert "Learned update" in rb.read_text()
+
+
+class TestOperability:
+    def test_reconcile_and_trust(self):
+        from runbookai_amd.providers.operability_context import (
+            AgentChangeClaim,
+            VerifiedChangeFact,
+            reconcile_claims,
+            trust_score,
+        )
+
+        claim = AgentChangeClaim(claim_id="c1", agent="cc", repo="org/app",
+                                 files=["a.py", "b.py"], timestamp=1000.0)
+        fact = VerifiedChangeFact(fact_id="f1", source="git", repo="org/app",
+                                  files=["a.py", "b.py"], timestamp=1200.0)
+        rec = reconcile_claims([claim], [fact])
+        assert rec[0]["status"] == "verified"
+        assert trust_score(rec) == 1.0
+        rec2 = reconcile_claims([claim], [])
+        assert rec2[0]["status"] == "unverified"
+
+    def test_ingest_spool_replay(self, tmp_path):
+        from runbookai_amd.integrations.operability_ingestion import (
+            ingest_claim,
+            replay_spool,
+            spool_status,
+        )
+        from runbookai_amd.providers.operability_context.factory import UnavailableAdapter
+
+        spool = str(tmp_path / "spool" / "claims.jsonl")
+        out = ingest_claim("start", adapter=UnavailableAdapter("http"), spool_path=spool,
+                           summary="did things")
+        assert out["spooled"]
+        assert spool_status(spool)["spooled"] == 1
+
+        from runbookai_amd.providers.operability_context.factory import FileSpoolAdapter
+
+        n = replay_spool(adapter=FileSpoolAdapter(str(tmp_path / "delivered.jsonl")),
+                         spool_path=spool)
+        assert n == 1
+        assert spool_status(spool)["spooled"] == 0
