@@ -1,0 +1,154 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: simulated `runbook investigate` on the local MI355X
+engine (BASELINE config 3/4 shape).
+
+One STEP = one complete structured investigation (TRIAGE→…→REMEDIATE)
+driven end-to-end through the Llama-3-8B (bf16, random-init) engine with
+grammar-constrained decoding against the simulated incident set, scored
+with the reference scorer. Data is synthetic (simulated telemetry +
+fixture-derived incidents); weights are random-init (no network for
+checkpoints) — constrained decoding keeps every phase schema-valid so the
+full agent/tool/LLM path executes identically to a trained checkpoint.
+
+Scaling is WEAK data-parallelism: each rank runs its own TP=1 engine
+replica and `--steps` investigations (`--concurrency` of them in flight
+sharing the continuous-batching engine); the whole-job value aggregates
+over ranks.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N --steps K --warmup W
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from concurrent.futures import ThreadPoolExecutor
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=2,
+                        help="timed investigations per rank")
+    parser.add_argument("--warmup", type=int, default=1)
+    parser.add_argument("--concurrency", type=int, default=4,
+                        help="investigations in flight per rank (continuous batching)")
+    parser.add_argument("--model", default=None, help="tiny | llama3-8b | llama3-70b")
+    parser.add_argument("--tp", type=int, default=1)
+    parser.add_argument("--max-tokens", type=int, default=768)
+    args = parser.parse_args()
+
+    import torch
+
+    from runbookai_amd.parallel.dist import barrier, destroy, init_distributed
+
+    rank, world = init_distributed()
+    has_gpu = torch.cuda.is_available()
+    device = f"cuda:{rank % torch.cuda.device_count()}" if has_gpu else "cpu"
+    model_name = args.model or ("llama3-8b" if has_gpu else "tiny")
+
+    from runbookai_amd.engine.client import LocalEngineClient
+    from runbookai_amd.engine.engine import LLMEngine
+    from runbookai_amd.evals.benchmark import load_fixtures
+    from runbookai_amd.evals.scoring import score_investigation_result
+    from runbookai_amd.agent.orchestrator import InvestigationOrchestrator
+    from runbookai_amd.providers.simulation import SimScenario, set_scenario
+    from runbookai_amd.tools.registry import ToolRegistry
+
+    fixtures = load_fixtures(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                          "examples", "evals",
+                                          "investigation-fixtures.sample.json"))
+    cases = fixtures["cases"]
+
+    kv_blocks = 2048 if model_name != "tiny" else 512
+    engine = LLMEngine(model=model_name, device=device, tp=None,
+                       kv_blocks=kv_blocks, background=True)
+    # one shared scenario for concurrent runs (scenario registry is global)
+    set_scenario(SimScenario.redis_exhaustion())
+
+    def run_investigation(i: int) -> dict:
+        case = cases[i % len(cases)]
+        client = LocalEngineClient(engine, max_tokens=args.max_tokens)
+        registry = ToolRegistry()
+        orch = InvestigationOrchestrator(
+            llm=client, tool_executor=registry,
+            max_iterations=int(case.get("execute", {}).get("maxIterations", 6)),
+        )
+        result = orch.investigate(case["query"], incident_id=case.get("incidentId"))
+        score = score_investigation_result(result.to_dict(), case.get("expected", {}))
+        return {"score": score["overall"],
+                "passed": score["overall"] >= fixtures.get("passThreshold", 0.7),
+                "success": result.success}
+
+    def run_batch(n: int) -> list[dict]:
+        with ThreadPoolExecutor(max_workers=args.concurrency) as pool:
+            return list(pool.map(run_investigation, range(n)))
+
+    # warmup (untimed)
+    if args.warmup > 0:
+        run_batch(args.warmup)
+
+    # timed region: barrier + device sync on both sides
+    barrier()
+    if has_gpu:
+        torch.cuda.synchronize()
+    t0 = time.time()
+    results = run_batch(args.steps)
+    if has_gpu:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.time() - t0
+
+    # MAX over ranks
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if dist.get_backend() == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    stats = engine.throughput_stats()
+    ms_per_step = elapsed * 1000.0 / args.steps
+    inv_per_hour = world * args.steps / elapsed * 3600.0
+    pass_rate = sum(1 for r in results if r["passed"]) / max(1, len(results))
+
+    if rank == 0:
+        line = {
+            "metric": "investigations_per_hour",
+            "value": round(inv_per_hour, 3),
+            "unit": "investigations/hour",
+            "n_gpus": world if has_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic (simulated incident set, random-init weights, "
+                    "grammar-constrained decoding)",
+            "config": {
+                "model": model_name,
+                "global_batch": args.concurrency * world,
+                "seq_len": engine.cfg.max_seq_len,
+                "parallelism": f"dp{world}",
+                "pass_rate": pass_rate,
+                "decode_tok_per_s": round(stats.get("decode_tok_per_s", 0.0), 1),
+                "prefill_tok_per_s": round(stats.get("prefill_tok_per_s", 0.0), 1),
+                "llm_calls_total": stats.get("requests", 0),
+            },
+        }
+        print(json.dumps(line), flush=True)
+    engine.shutdown()
+    destroy()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,331 @@
+"""LLM engine: continuous batching over the paged-KV Llama model.
+
+Implements SURVEY.md §2.11 component 3: iteration-level scheduling with
+prefill/decode interleaving and per-request sampling state. Concurrent
+investigations (the eval harness runs up to 32 at once) submit requests
+from their own threads; a single engine thread batches whatever is
+pending each iteration, so decode batches grow/shrink as agent loops
+issue and await LLM calls — replacing "the provider's server does
+batching" (reference src/model/llm.ts).
+
+Grammar-constrained decoding: requests carrying a JSON schema get a
+JsonFsm; every step the sampler masks logits to the FSM's allowed bytes
+(plus engine-level masking to the active byte vocab), so outputs are
+schema-valid by construction.
+"""
+from __future__ import annotations
+
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+import torch
+
+from .. import ops
+from ..agent.llm_parser import PROMPT_SCHEMAS
+from .json_fsm import NUMBER_CLOSE_SENTINEL, JsonFsm
+from .kv_cache import PagedKvCache
+from .llama import CONFIGS, LlamaModel
+from .tokenizer import ACTIVE_VOCAB, ByteTokenizer, SpecialTokens
+
+MASK_REGION = 512  # byte vocab + specials, padded; legal ids live below this
+
+
+@dataclass
+class Request:
+    rid: int
+    prompt_ids: list[int]
+    max_new_tokens: int = 512
+    temperature: float = 0.0
+    schema: Optional[dict[str, Any]] = None
+    fsm: Optional[JsonFsm] = None
+    out_ids: list[int] = field(default_factory=list)
+    state: str = "waiting"           # waiting | running | done
+    done_event: threading.Event = field(default_factory=threading.Event)
+    prompt_len: int = 0
+    pos: int = 0                     # next position to write
+    error: str = ""
+    submitted_at: float = field(default_factory=time.time)
+    first_token_at: float = 0.0
+    finished_at: float = 0.0
+
+
+class LLMEngine:
+    def __init__(
+        self,
+        model: str = "tiny",
+        device: Optional[str] = None,
+        tp: Optional[int] = None,
+        max_prefill_tokens: int = 8192,
+        max_batch: int = 64,
+        kv_blocks: Optional[int] = None,
+        seed: int = 1234,
+        background: bool = True,
+    ) -> None:
+        cfg = CONFIGS[model]
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = device
+        self.cfg = cfg
+        self.model = LlamaModel(cfg, device=device, tp=tp, seed=seed, kv_blocks=kv_blocks)
+        self.tokenizer = ByteTokenizer(cfg.vocab_size)
+        self.max_prefill_tokens = max_prefill_tokens
+        self.max_batch = max_batch
+        self._rid = 0
+        self._lock = threading.Condition()
+        self.waiting: list[Request] = []
+        self.running: list[Request] = []
+        self.stats = {"requests": 0, "prefill_tokens": 0, "decode_tokens": 0,
+                      "steps": 0, "prefill_time": 0.0, "decode_time": 0.0}
+        self._stop = False
+        self._thread: Optional[threading.Thread] = None
+        if background:
+            self._thread = threading.Thread(target=self._loop, daemon=True,
+                                            name="llm-engine")
+            self._thread.start()
+
+    # -- submission ---------------------------------------------------------------
+
+    def submit(self, prompt_ids: list[int], max_new_tokens: int = 512,
+               temperature: float = 0.0, schema: Optional[dict[str, Any]] = None) -> Request:
+        with self._lock:
+            self._rid += 1
+            # clamp to the model's context window (RoPE table bound)
+            room = self.cfg.max_seq_len - len(prompt_ids) - 1
+            if room <= 0:
+                prompt_ids = prompt_ids[-(self.cfg.max_seq_len // 2):]
+                room = self.cfg.max_seq_len - len(prompt_ids) - 1
+            req = Request(
+                rid=self._rid,
+                prompt_ids=list(prompt_ids),
+                max_new_tokens=max(1, min(max_new_tokens, room)),
+                temperature=temperature,
+                schema=schema,
+                fsm=JsonFsm(schema) if schema else None,
+            )
+            req.prompt_len = len(req.prompt_ids)
+            self.waiting.append(req)
+            self.stats["requests"] += 1
+            self._lock.notify_all()
+            return req
+
+    def generate(self, prompt_ids: list[int], max_new_tokens: int = 512,
+                 temperature: float = 0.0, schema: Optional[dict[str, Any]] = None,
+                 timeout_s: float = 600.0) -> Request:
+        req = self.submit(prompt_ids, max_new_tokens, temperature, schema)
+        if self._thread is None:
+            self.run_until_idle()
+        else:
+            req.done_event.wait(timeout=timeout_s)
+        if not req.done_event.is_set():
+            raise TimeoutError(f"generation timed out after {timeout_s}s")
+        if req.error:
+            raise RuntimeError(req.error)
+        return req
+
+    # -- engine loop ----------------------------------------------------------------
+
+    def _loop(self) -> None:
+        while not self._stop:
+            with self._lock:
+                while not self.waiting and not self.running and not self._stop:
+                    self._lock.wait(timeout=1.0)
+                if self._stop:
+                    return
+            try:
+                self.step()
+            except Exception as e:  # noqa: BLE001 — fail requests, not the thread
+                with self._lock:
+                    for req in self.running + self.waiting:
+                        req.error = f"{type(e).__name__}: {e}"
+                        req.state = "done"
+                        req.done_event.set()
+                    self.running.clear()
+                    self.waiting.clear()
+
+    def run_until_idle(self, max_steps: int = 100_000) -> None:
+        for _ in range(max_steps):
+            with self._lock:
+                if not self.waiting and not self.running:
+                    return
+            self.step()
+
+    def shutdown(self) -> None:
+        self._stop = True
+        with self._lock:
+            self._lock.notify_all()
+        if self._thread:
+            self._thread.join(timeout=5.0)
+
+    # -- scheduling ------------------------------------------------------------------
+
+    def step(self) -> None:
+        """One engine iteration: admit a prefill batch if any request is
+        waiting (and fits), else run one decode step over all running."""
+        with self._lock:
+            prefill_batch = self._admit_locked()
+        if prefill_batch:
+            self._run_prefill(prefill_batch)
+        else:
+            with self._lock:
+                decode_batch = list(self.running)
+            if decode_batch:
+                self._run_decode(decode_batch)
+        self.stats["steps"] += 1
+
+    def _admit_locked(self) -> list[Request]:
+        batch: list[Request] = []
+        tokens = 0
+        kv = self.model.kv
+        while self.waiting and len(self.running) + len(batch) < self.max_batch:
+            req = self.waiting[0]
+            need = req.prompt_len + req.max_new_tokens
+            if batch and tokens + req.prompt_len > self.max_prefill_tokens:
+                break
+            if not kv.can_allocate(need):
+                break
+            kv.allocate(req.rid, need)
+            self.waiting.pop(0)
+            batch.append(req)
+            tokens += req.prompt_len
+        return batch
+
+    # -- execution --------------------------------------------------------------------
+
+    def _run_prefill(self, batch: list[Request]) -> None:
+        t0 = time.time()
+        kv = self.model.kv
+        token_ids: list[int] = []
+        positions: list[int] = []
+        starts = [0]
+        slots: list[torch.Tensor] = []
+        for req in batch:
+            token_ids.extend(req.prompt_ids)
+            positions.extend(range(req.prompt_len))
+            starts.append(starts[-1] + req.prompt_len)
+            slots.append(kv.slot_mapping(req.rid, 0, req.prompt_len))
+            kv.set_len(req.rid, req.prompt_len)
+            req.pos = req.prompt_len
+        logits = self.model.prefill(
+            torch.tensor(token_ids, dtype=torch.int64),
+            torch.tensor(positions, dtype=torch.int32),
+            torch.tensor(starts, dtype=torch.int32),
+            torch.cat(slots),
+        )
+        self.stats["prefill_tokens"] += len(token_ids)
+        self._sample_and_advance(batch, logits)
+        with self._lock:
+            for req in batch:
+                if req.state != "done":
+                    req.state = "running"
+                    self.running.append(req)
+        self.stats["prefill_time"] += time.time() - t0
+
+    def _run_decode(self, batch: list[Request]) -> None:
+        t0 = time.time()
+        kv = self.model.kv
+        input_ids = []
+        positions = []
+        slot_list = []
+        for req in batch:
+            tok = req.out_ids[-1] if req.out_ids else (req.prompt_ids[-1] if req.prompt_ids else 0)
+            input_ids.append(tok)
+            positions.append(req.pos)
+            kv.extend(req.rid, req.pos + 1)
+            slot_list.append(kv.slot_mapping(req.rid, req.pos, 1))
+            kv.set_len(req.rid, req.pos + 1)
+        bt, lens = kv.batch_tables([r.rid for r in batch], self.device)
+        logits = self.model.decode(
+            torch.tensor(input_ids, dtype=torch.int64),
+            torch.tensor(positions, dtype=torch.int32),
+            bt, lens, torch.cat(slot_list),
+        )
+        for req in batch:
+            req.pos += 1
+        self.stats["decode_tokens"] += len(batch)
+        self._sample_and_advance(batch, logits)
+        with self._lock:
+            self.running = [r for r in self.running if r.state != "done"]
+        self.stats["decode_time"] += time.time() - t0
+
+    # -- sampling ---------------------------------------------------------------------
+
+    def _sample_and_advance(self, batch: list[Request], logits: torch.Tensor) -> None:
+        """Masked sampling per request + FSM/stop bookkeeping.
+
+        All requests are masked to the active byte vocab (ids < MASK_REGION);
+        schema'd requests are further masked to their FSM's allowed bytes.
+        """
+        region = logits[:, :MASK_REGION]
+        mask = torch.zeros((len(batch), MASK_REGION), dtype=torch.bool)
+        for i, req in enumerate(batch):
+            if req.fsm is not None:
+                allowed = req.fsm.allowed_bytes()
+                if not allowed:   # FSM complete -> force EOT
+                    mask[i, SpecialTokens.EOT] = True
+                else:
+                    mask[i, allowed] = True
+            else:
+                mask[i, :ACTIVE_VOCAB] = True
+        mask_d = mask.to(region.device)
+        greedy = all(r.temperature <= 0.0 for r in batch)
+        if greedy:
+            chosen = ops.masked_greedy(region, mask_d)
+        else:
+            temp = max(r.temperature for r in batch)
+            chosen = ops.masked_sample(region, mask_d, temperature=temp)
+        chosen = chosen.cpu().tolist()
+        now = time.time()
+        for req, tok in zip(batch, chosen):
+            if req.first_token_at == 0.0:
+                req.first_token_at = now
+            self._advance_request(req, int(tok))
+
+    def _advance_request(self, req: Request, tok: int) -> None:
+        finished = False
+        if req.fsm is not None:
+            if req.fsm.done or tok == SpecialTokens.EOT:
+                finished = True
+            else:
+                req.fsm.advance(tok)
+                if tok != NUMBER_CLOSE_SENTINEL:
+                    req.out_ids.append(tok)
+                if req.fsm.done:
+                    finished = True
+        else:
+            if tok in (SpecialTokens.EOS, SpecialTokens.EOT):
+                finished = True
+            else:
+                req.out_ids.append(tok)
+        if len(req.out_ids) >= req.max_new_tokens:
+            finished = True
+        if finished:
+            req.state = "done"
+            req.finished_at = time.time()
+            self.model.kv.free(req.rid)
+            req.done_event.set()
+        else:
+            # the chosen token still needs its KV stored next decode step
+            pass
+
+    # -- metrics ----------------------------------------------------------------------
+
+    def throughput_stats(self) -> dict[str, Any]:
+        s = dict(self.stats)
+        if s["decode_time"] > 0:
+            s["decode_tok_per_s"] = s["decode_tokens"] / s["decode_time"]
+        if s["prefill_time"] > 0:
+            s["prefill_tok_per_s"] = s["prefill_tokens"] / s["prefill_time"]
+        return s
+
+
+_engines: dict[str, LLMEngine] = {}
+
+
+def get_engine(model: str = "tiny", **kwargs: Any) -> LLMEngine:
+    """Process-wide engine cache (one engine per model)."""
+    key = f"{model}:{kwargs.get('tp')}:{kwargs.get('device')}"
+    if key not in _engines:
+        _engines[key] = LLMEngine(model=model, **kwargs)
+    return _engines[key]

@@ -1,0 +1,320 @@
+"""JSON-schema grammar FSM for byte-level constrained decoding.
+
+Since the engine's tokens ARE bytes (tokenizer.py), a byte-level automaton
+enforces a JSON schema EXACTLY during decoding: at every step the sampler
+masks logits to `allowed_bytes()`, and structural bytes (braces, keys,
+colons, commas) are forced — so even a random-init model emits
+schema-valid JSON. This implements the constrained-sampling plan of
+SURVEY.md §7 ("we own the decoder, so enforce schema at the logits
+level"), replacing the reference's hope that hosted frontier models
+return parseable JSON (mitigations at llm-parser.ts:215-229).
+
+Supported schema subset (everything llm_parser.py's PROMPT_SCHEMAS use):
+object {properties, required}, array {items, minItems, maxItems},
+string {maxLength}, integer/number {minimum, maximum}, boolean, enum of
+strings.
+
+Design: a stack of frames; each frame knows its allowed next bytes and
+consumes one byte at a time. Keys and punctuation are emitted as FORCED
+single-byte choices, so the model only "chooses" inside value positions.
+"""
+from __future__ import annotations
+
+from typing import Any, Optional
+
+# printable ASCII minus '"' and '\\' (escapes are disallowed to keep the
+# automaton single-byte exact; values never need them)
+_STRING_BYTES = [b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C)]
+_DIGITS = list(range(0x30, 0x3A))
+EOS_CHOICE = -1  # sentinel: the FSM is done; sampler should emit EOS/EOT
+
+
+class JsonFsm:
+    def __init__(self, schema: dict[str, Any], max_total_bytes: int = 8192) -> None:
+        self.schema = schema
+        self.max_total_bytes = max_total_bytes
+        self.emitted = 0
+        self._pending: list[int] = []           # forced bytes queued for emission
+        self._stack: list[dict[str, Any]] = []  # frames
+        self._done = False
+        self._push_value(schema)
+
+    # -- public API -------------------------------------------------------------
+
+    @property
+    def done(self) -> bool:
+        return self._done and not self._pending
+
+    def allowed_bytes(self) -> list[int]:
+        """Byte ids the next token may take; [] when generation must stop."""
+        if self._pending:
+            return [self._pending[0]]
+        if self._done:
+            return []
+        frame = self._stack[-1]
+        return self._frame_allowed(frame)
+
+    def advance(self, byte: int) -> None:
+        """Consume the byte the sampler chose (must be in allowed_bytes())."""
+        self.emitted += 1
+        if self._pending:
+            expected = self._pending.pop(0)
+            if byte != expected:
+                raise ValueError(f"FSM desync: forced {expected}, got {byte}")
+            if not self._pending and not self._stack:
+                self._done = True
+            return
+        if self._done:
+            raise ValueError("FSM already done")
+        self._frame_advance(self._stack[-1], byte)
+
+    # -- frame plumbing -----------------------------------------------------------
+
+    def _force(self, text: str) -> None:
+        self._pending.extend(text.encode("utf-8"))
+
+    def _pop_frame(self) -> None:
+        self._stack.pop()
+        if not self._stack and not self._pending:
+            self._done = True
+
+    def _push_value(self, schema: dict[str, Any]) -> None:
+        """Push a frame for a value of the given schema; may queue forced bytes."""
+        if "enum" in schema:
+            options = [str(o) for o in schema["enum"]]
+            self._stack.append({"kind": "enum", "options": options, "progress": ""})
+            self._force('"')
+            return
+        t = schema.get("type", "string")
+        if t == "object":
+            props = list(schema.get("properties", {}).items())
+            required = set(schema.get("required", [k for k, _ in props]))
+            # fixed order; only required properties are emitted (optional ones
+            # would need a model choice on the KEY — structure stays forced)
+            emit = [(k, v) for k, v in props if k in required] or props[:1]
+            self._stack.append({"kind": "object", "props": emit, "idx": 0})
+            self._force("{")
+            self._object_next_key()
+        elif t == "array":
+            self._stack.append({
+                "kind": "array",
+                "items": schema.get("items", {"type": "string"}),
+                "min": int(schema.get("minItems", 0)),
+                "max": int(schema.get("maxItems", 8)),
+                "count": 0,
+                "state": "start",
+            })
+            self._force("[")
+        elif t == "string":
+            self._stack.append({"kind": "string",
+                                "max": int(schema.get("maxLength", 200)),
+                                "len": 0, "open": False})
+            self._force('"')
+        elif t in ("integer", "number"):
+            self._stack.append({"kind": "number", "float": t == "number",
+                                "len": 0, "max_len": 10, "has_dot": False,
+                                "minimum": schema.get("minimum"),
+                                "maximum": schema.get("maximum")})
+        elif t == "boolean":
+            self._stack.append({"kind": "enum_raw", "options": ["true", "false"],
+                                "progress": ""})
+        elif t == "null":
+            self._force("null")
+        else:  # fallback: treat as string
+            self._push_value({"type": "string"})
+
+    def _object_next_key(self) -> None:
+        frame = self._stack[-1]
+        assert frame["kind"] == "object"
+        if frame["idx"] >= len(frame["props"]):
+            self._force("}")
+            self._pop_frame()
+            return
+        key, subschema = frame["props"][frame["idx"]]
+        prefix = ", " if frame["idx"] > 0 else ""
+        frame["idx"] += 1
+        self._force(f'{prefix}"{key}": ')
+        # value frame goes on top; when it pops, control returns to the object
+        self._push_value(subschema)
+
+    # -- per-kind allowed/advance --------------------------------------------------
+
+    def _frame_allowed(self, frame: dict[str, Any]) -> list[int]:
+        kind = frame["kind"]
+        if kind == "object":
+            # object frames only act through forced bytes
+            return []
+        if kind == "string":
+            # budget guard: force close when at max
+            if frame["len"] >= frame["max"] or self.emitted >= self.max_total_bytes:
+                return [0x22]  # '"'
+            if frame["len"] == 0:
+                return _STRING_BYTES  # require at least 1 char before closing
+            return _STRING_BYTES + [0x22]
+        if kind == "number":
+            if frame["len"] == 0:
+                return _DIGITS
+            if frame.get("dot_pending"):
+                return _DIGITS  # '.' must be followed by a digit
+            if frame["len"] >= frame["max_len"]:
+                return [0x00]  # sentinel handled in advance: close number
+            if frame.get("leading_zero") and not frame["has_dot"]:
+                # JSON forbids further digits after a leading 0
+                allowed = [0x00]
+                if frame["float"]:
+                    allowed.append(0x2E)
+                return allowed
+            allowed = list(_DIGITS)
+            if frame["float"] and not frame["has_dot"]:
+                allowed.append(0x2E)  # '.'
+            allowed.append(0x00)
+            return allowed
+        if kind in ("enum", "enum_raw"):
+            progress = frame["progress"]
+            nexts = {ord(o[len(progress)]) for o in frame["options"]
+                     if o.startswith(progress) and len(o) > len(progress)}
+            return sorted(nexts)
+        if kind == "array":
+            if frame["state"] == "start":
+                # model chooses: emit first item or close (if min allows)
+                choices = [0x7B if _opens_with_brace(frame["items"]) else 0x22]
+                choices = self._array_item_start_bytes(frame)
+                if frame["min"] == 0:
+                    choices = choices + [0x5D]  # ']'
+                return sorted(set(choices))
+            if frame["state"] == "between":
+                choices = []
+                if frame["count"] < frame["max"]:
+                    choices.append(0x2C)  # ','
+                if frame["count"] >= frame["min"]:
+                    choices.append(0x5D)  # ']'
+                return choices or [0x5D]
+        return []
+
+    def _array_item_start_bytes(self, frame: dict[str, Any]) -> list[int]:
+        items = frame["items"]
+        if "enum" in items or items.get("type", "string") == "string":
+            return [0x22]
+        t = items.get("type")
+        if t == "object":
+            return [0x7B]
+        if t == "array":
+            return [0x5B]
+        if t in ("integer", "number"):
+            return _DIGITS
+        if t == "boolean":
+            return [ord("t"), ord("f")]
+        return [0x22]
+
+    def _frame_advance(self, frame: dict[str, Any], byte: int) -> None:
+        kind = frame["kind"]
+        if kind == "string":
+            if byte == 0x22:
+                self._pop_frame()
+                self._resume_parent()
+            else:
+                frame["len"] += 1
+            return
+        if kind == "number":
+            if byte == 0x00:
+                # close-number sentinel: pop and let parent continue; the
+                # sampler maps this to NOT emitting a byte (see sampler)
+                self._pop_frame()
+                self._resume_parent()
+                return
+            if byte == 0x2E:
+                frame["has_dot"] = True
+                frame["dot_pending"] = True
+            elif frame.get("dot_pending"):
+                frame["dot_pending"] = False
+            if frame["len"] == 0 and byte == 0x30:
+                frame["leading_zero"] = True
+            frame["len"] += 1
+            return
+        if kind in ("enum", "enum_raw"):
+            frame["progress"] += chr(byte)
+            exact = frame["progress"] in frame["options"]
+            extendable = any(o.startswith(frame["progress"]) and len(o) > len(frame["progress"])
+                             for o in frame["options"])
+            if exact and not extendable:
+                if kind == "enum":
+                    self._force('"')
+                self._pop_frame()
+                self._resume_parent()
+            elif exact and extendable:
+                # prefix of a longer option (e.g. "low"/"lowest"): prefer exact
+                if kind == "enum":
+                    self._force('"')
+                self._pop_frame()
+                self._resume_parent()
+            return
+        if kind == "array":
+            if frame["state"] == "start":
+                if byte == 0x5D:
+                    self._pop_frame()
+                    self._resume_parent()
+                    return
+                frame["state"] = "between"
+                frame["count"] = 1
+                self._start_item(frame, byte)
+                return
+            if frame["state"] == "between":
+                if byte == 0x5D:
+                    self._pop_frame()
+                    self._resume_parent()
+                elif byte == 0x2C:
+                    frame["count"] += 1
+                    self._force(" ")
+                    self._push_value(frame["items"])  # forced prefix lands after the space
+                return
+        raise ValueError(f"frame {kind} cannot advance on byte {byte}")
+
+    def _start_item(self, frame: dict[str, Any], first_byte: int) -> None:
+        """The model chose to start an array item; push the item frame and
+        replay the first byte into it."""
+        items = frame["items"]
+        self._push_value(items)
+        # _push_value may have queued forced bytes (e.g. '"' or '{'); the
+        # first byte the model emitted IS that forced byte — consume it.
+        if self._pending and self._pending[0] == first_byte:
+            self._pending.pop(0)
+            if not self._pending and frame.get("_noop"):
+                pass
+        else:
+            # number/boolean items: no forced prefix; replay into the new frame
+            self._frame_advance(self._stack[-1], first_byte)
+
+    def _resume_parent(self) -> None:
+        """After a value frame pops, hand control back to its parent."""
+        if not self._stack:
+            return
+        parent = self._stack[-1]
+        if parent["kind"] == "object":
+            self._object_next_key()
+        # array parents continue from their own "between" state naturally
+
+
+NUMBER_CLOSE_SENTINEL = 0x00  # exposed for the sampler
+
+
+def generate_minimal(schema: dict[str, Any], chooser=None, max_bytes: int = 8192) -> str:
+    """Drive the FSM with a chooser(allowed)->byte (default: first allowed).
+    Used in tests and as a CPU fallback 'model'."""
+    fsm = JsonFsm(schema, max_total_bytes=max_bytes)
+    out = bytearray()
+    chooser = chooser or (lambda allowed: allowed[0])
+    steps = 0
+    while not fsm.done and steps < max_bytes * 2:
+        steps += 1
+        allowed = fsm.allowed_bytes()
+        if not allowed:
+            break
+        byte = chooser(allowed)
+        fsm.advance(byte)
+        if byte != NUMBER_CLOSE_SENTINEL:
+            out.append(byte)
+    return out.decode("utf-8", errors="replace")
+
+
+def _opens_with_brace(schema: dict[str, Any]) -> bool:
+    return schema.get("type") == "object"

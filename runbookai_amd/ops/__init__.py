@@ -1,0 +1,144 @@
+"""Custom op dispatch: gfx950 HIP kernels on GPU, torch reference on CPU.
+
+Policy (round contract): on a CUDA/ROCm device the HIP extension is
+REQUIRED — ops raise immediately if it is missing rather than silently
+falling back to eager PyTorch; on CPU the fp32 reference implementations
+(ops/reference.py) run, which are also the numerics baselines the GPU
+kernels are tested against.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from . import reference
+
+_ext = None
+_ext_error: Optional[str] = None
+
+
+def _get_ext():
+    global _ext, _ext_error
+    if _ext is not None:
+        return _ext
+    if _ext_error is not None:
+        raise RuntimeError(_ext_error)
+    from .build import build, load_prebuilt
+
+    try:
+        _ext = load_prebuilt() or build()
+    except Exception as e:  # noqa: BLE001
+        _ext_error = (
+            "runbookai_hip_ops extension unavailable on a GPU host — the HIP "
+            f"path is mandatory on GPU (no silent eager fallback). Build error: {e}"
+        )
+        raise RuntimeError(_ext_error) from e
+    return _ext
+
+
+def extension_loaded() -> bool:
+    return _ext is not None
+
+
+def _on_gpu(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+# ---------------------------------------------------------------- public ops
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if _on_gpu(x):
+        return _get_ext().rmsnorm(x.contiguous(), weight.contiguous(), eps)
+    return reference.rmsnorm(x, weight, eps)
+
+
+def rmsnorm_residual(x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor,
+                     eps: float = 1e-5):
+    if _on_gpu(x):
+        out, res = _get_ext().rmsnorm_residual(x.contiguous(), residual.contiguous(),
+                                               weight.contiguous(), eps)
+        return out, res
+    return reference.rmsnorm_residual(x, residual, weight, eps)
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(gate):
+        return _get_ext().silu_mul(gate.contiguous(), up.contiguous())
+    return reference.silu_mul(gate, up)
+
+
+def apply_rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+               positions: torch.Tensor):
+    if _on_gpu(q):
+        q = q.contiguous()
+        k = k.contiguous()
+        _get_ext().rope_inplace(q, k, cos.contiguous(), sin.contiguous(),
+                                positions.to(torch.int32).contiguous())
+        return q, k
+    return reference.apply_rope(q, k, cos, sin, positions)
+
+
+def prefill_attention(q, k, v, seq_starts, causal: bool = True,
+                      scale: Optional[float] = None, batch_idx=None):
+    scale = scale or 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        if batch_idx is None:
+            batch_idx = _batch_idx_from_starts(seq_starts, q.shape[0])
+        return _get_ext().prefill_attn(q.contiguous(), k.contiguous(), v.contiguous(),
+                                       batch_idx.to(torch.int32).contiguous(),
+                                       seq_starts.to(torch.int32).contiguous(),
+                                       scale, causal)
+    return reference.prefill_attention(q, k, v, seq_starts, causal, scale)
+
+
+def paged_decode_attention(q, k_cache, v_cache, block_tables, seq_lens,
+                           scale: Optional[float] = None):
+    scale = scale or 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        return _get_ext().paged_decode(q.contiguous(), k_cache, v_cache,
+                                       block_tables.to(torch.int32).contiguous(),
+                                       seq_lens.to(torch.int32).contiguous(), scale)
+    return reference.paged_decode_attention(q, k_cache, v_cache, block_tables,
+                                            seq_lens, scale)
+
+
+def store_kv(k, v, k_cache, v_cache, slot_mapping):
+    if _on_gpu(k):
+        _get_ext().store_kv(k.contiguous(), v.contiguous(), k_cache, v_cache,
+                            slot_mapping.to(torch.int32).contiguous())
+        return
+    reference.store_kv(k, v, k_cache, v_cache, slot_mapping)
+
+
+def topk_cosine(matrix: torch.Tensor, query: torch.Tensor, k: int):
+    if _on_gpu(matrix):
+        scores = _get_ext().cosine_scores(matrix.contiguous(), query.contiguous())
+        vals, idx = torch.topk(scores, min(k, matrix.shape[0]))
+        return vals, idx
+    return reference.topk_cosine(matrix, query, k)
+
+
+def masked_greedy(logits: torch.Tensor, allowed_mask: Optional[torch.Tensor] = None):
+    """Greedy token selection under a validity mask. logits [B, V]."""
+    if _on_gpu(logits):
+        mask_u8 = allowed_mask.to(torch.uint8).contiguous() if allowed_mask is not None else None
+        return _get_ext().masked_argmax(logits.to(torch.bfloat16).contiguous(), mask_u8).long()
+    return reference.masked_sample(logits, allowed_mask, temperature=0.0)
+
+
+def masked_sample(logits, allowed_mask=None, temperature: float = 0.0,
+                  top_p: float = 1.0, generator=None):
+    if temperature <= 0.0:
+        return masked_greedy(logits, allowed_mask)
+    # stochastic path: fp32 torch math on-device (vocab-sized, not hot)
+    return reference.masked_sample(logits, allowed_mask, temperature, top_p, generator)
+
+
+def _batch_idx_from_starts(seq_starts: torch.Tensor, T: int) -> torch.Tensor:
+    starts = seq_starts.tolist()
+    idx = torch.empty(T, dtype=torch.int32)
+    for b in range(len(starts) - 1):
+        idx[starts[b]:starts[b + 1]] = b
+    return idx.to(seq_starts.device)

@@ -1,0 +1,175 @@
+// Torch extension bindings for the gfx950 kernels.
+// Kernel launchers are extern "C" in the .hip files; streams are passed as
+// opaque pointers (hipStream_t is a pointer type) so this TU needs no HIP
+// headers and hipify leaves it alone apart from the stream query.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+extern "C" {
+void launch_rmsnorm(const void*, const void*, void*, int, int, float, void*);
+void launch_rmsnorm_residual(const void*, const void*, void*, const void*, void*,
+                             int, int, float, void*);
+void launch_silu_mul(const void*, const void*, void*, long, void*);
+void launch_rope(void*, void*, const void*, const void*, const void*,
+                 int, int, int, int, void*);
+void launch_paged_decode(const void*, const void*, const void*, const void*,
+                         const void*, void*, int, int, int, int, int, int, float, void*);
+void launch_prefill(const void*, const void*, const void*, const void*, const void*,
+                    void*, int, int, int, int, float, int, void*);
+void launch_store_kv(const void*, const void*, void*, void*, const void*,
+                     int, int, int, int, void*);
+void launch_cosine_scores(const void*, const void*, void*, long, int, void*);
+void launch_masked_argmax(const void*, const void*, void*, int, int, void*);
+}
+
+namespace {
+
+void* current_stream() {
+    return (void*)at::cuda::getCurrentCUDAStream().stream();
+}
+
+#define CHECK_IN(t, dt)                                                        \
+    TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                          \
+    TORCH_CHECK((t).is_contiguous(), #t " must be contiguous");                \
+    TORCH_CHECK((t).scalar_type() == dt, #t " must be ", dt)
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+    CHECK_IN(x, torch::kBFloat16);
+    CHECK_IN(w, torch::kBFloat16);
+    auto sizes = x.sizes().vec();
+    int H = (int)sizes.back();
+    long T = x.numel() / H;
+    TORCH_CHECK(H % 8 == 0, "H must be divisible by 8");
+    auto out = torch::empty_like(x);
+    launch_rmsnorm(x.data_ptr(), w.data_ptr(), out.data_ptr(), (int)T, H, (float)eps,
+                   current_stream());
+    return out;
+}
+
+std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res,
+                                            torch::Tensor w, double eps) {
+    CHECK_IN(x, torch::kBFloat16);
+    CHECK_IN(res, torch::kBFloat16);
+    CHECK_IN(w, torch::kBFloat16);
+    int H = (int)x.size(-1);
+    long T = x.numel() / H;
+    auto out = torch::empty_like(x);
+    auto res_out = torch::empty_like(x);
+    launch_rmsnorm_residual(x.data_ptr(), res.data_ptr(), res_out.data_ptr(),
+                            w.data_ptr(), out.data_ptr(), (int)T, H, (float)eps,
+                            current_stream());
+    return {out, res_out};
+}
+
+torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
+    CHECK_IN(gate, torch::kBFloat16);
+    CHECK_IN(up, torch::kBFloat16);
+    TORCH_CHECK(gate.numel() == up.numel());
+    TORCH_CHECK(gate.numel() % 8 == 0);
+    auto out = torch::empty_like(gate);
+    launch_silu_mul(gate.data_ptr(), up.data_ptr(), out.data_ptr(), gate.numel(),
+                    current_stream());
+    return out;
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos,
+                  torch::Tensor sin, torch::Tensor positions) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(k, torch::kBFloat16);
+    CHECK_IN(cos, torch::kFloat32);
+    CHECK_IN(sin, torch::kFloat32);
+    CHECK_IN(positions, torch::kInt32);
+    int T = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+    int Hk = (int)k.size(1);
+    TORCH_CHECK(k.size(0) == T && k.size(2) == D);
+    launch_rope(q.data_ptr(), k.data_ptr(), cos.data_ptr(), sin.data_ptr(),
+                positions.data_ptr(), T, Hq, Hk, D, current_stream());
+}
+
+torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                           torch::Tensor bt, torch::Tensor lens, double scale) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(kc, torch::kBFloat16);
+    CHECK_IN(vc, torch::kBFloat16);
+    CHECK_IN(bt, torch::kInt32);
+    CHECK_IN(lens, torch::kInt32);
+    int B = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+    int Hk = (int)kc.size(1), BS = (int)kc.size(2);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    auto out = torch::empty_like(q);
+    launch_paged_decode(q.data_ptr(), kc.data_ptr(), vc.data_ptr(), bt.data_ptr(),
+                        lens.data_ptr(), out.data_ptr(), B, Hq, Hk, D, BS,
+                        (int)bt.size(1), (float)scale, current_stream());
+    return out;
+}
+
+torch::Tensor prefill_attn(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                           torch::Tensor batch_idx, torch::Tensor seq_starts,
+                           double scale, bool causal) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(k, torch::kBFloat16);
+    CHECK_IN(v, torch::kBFloat16);
+    CHECK_IN(batch_idx, torch::kInt32);
+    CHECK_IN(seq_starts, torch::kInt32);
+    int T = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+    int Hk = (int)k.size(1);
+    TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
+    auto out = torch::empty_like(q);
+    launch_prefill(q.data_ptr(), k.data_ptr(), v.data_ptr(), batch_idx.data_ptr(),
+                   seq_starts.data_ptr(), out.data_ptr(), T, Hq, Hk, D,
+                   (float)scale, causal ? 1 : 0, current_stream());
+    return out;
+}
+
+void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor kc, torch::Tensor vc,
+              torch::Tensor slots) {
+    CHECK_IN(k, torch::kBFloat16);
+    CHECK_IN(v, torch::kBFloat16);
+    CHECK_IN(kc, torch::kBFloat16);
+    CHECK_IN(vc, torch::kBFloat16);
+    CHECK_IN(slots, torch::kInt32);
+    int T = (int)k.size(0), Hk = (int)k.size(1), D = (int)k.size(2);
+    int BS = (int)kc.size(2);
+    launch_store_kv(k.data_ptr(), v.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                    slots.data_ptr(), T, Hk, D, BS, current_stream());
+}
+
+torch::Tensor cosine_scores(torch::Tensor matrix, torch::Tensor query) {
+    CHECK_IN(matrix, torch::kFloat16);
+    CHECK_IN(query, torch::kFloat16);
+    long N = matrix.size(0);
+    int D = (int)matrix.size(1);
+    TORCH_CHECK(D % 8 == 0);
+    auto scores = torch::empty({N}, matrix.options().dtype(torch::kFloat32));
+    launch_cosine_scores(matrix.data_ptr(), query.data_ptr(), scores.data_ptr(), N, D,
+                         current_stream());
+    return scores;
+}
+
+torch::Tensor masked_argmax(torch::Tensor logits, c10::optional<torch::Tensor> mask) {
+    CHECK_IN(logits, torch::kBFloat16);
+    int B = (int)logits.size(0), V = (int)logits.size(1);
+    const void* mptr = nullptr;
+    if (mask.has_value()) {
+        CHECK_IN(mask.value(), torch::kUInt8);
+        mptr = mask->data_ptr();
+    }
+    auto out = torch::empty({B}, logits.options().dtype(torch::kInt32));
+    launch_masked_argmax(logits.data_ptr(), mptr, out.data_ptr(), B, V,
+                         current_stream());
+    return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("rmsnorm", &rmsnorm, "RMSNorm bf16 (gfx950)");
+    m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual-add + RMSNorm");
+    m.def("silu_mul", &silu_mul, "SwiGLU activation");
+    m.def("rope_inplace", &rope_inplace, "RoPE in place on q,k");
+    m.def("paged_decode", &paged_decode, "paged-KV decode attention");
+    m.def("prefill_attn", &prefill_attn, "varlen prefill attention");
+    m.def("store_kv", &store_kv, "scatter K/V into paged cache");
+    m.def("cosine_scores", &cosine_scores, "brute-force cosine scores");
+    m.def("masked_argmax", &masked_argmax, "greedy sampling under validity mask");
+}

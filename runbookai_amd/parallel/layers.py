@@ -1,0 +1,65 @@
+"""Tensor-parallel linear layers (Megatron-style column/row split).
+
+Column-parallel: weight sharded on the OUTPUT dim; no communication on
+forward (each rank computes its head/neuron shard). Row-parallel: weight
+sharded on the INPUT dim; forward ends in ONE RCCL all-reduce over xGMI.
+Per transformer layer that is exactly 2 all-reduces (after attention
+o-proj and MLP down-proj) — the xGMI-aware plan of SURVEY.md §2.11 item 2.
+
+GEMMs go through torch.matmul (hipBLASLt on ROCm) — plain library GEMMs
+per the MI355X design rules; the fused hot ops are the hand-written HIP
+kernels in runbookai_amd/ops.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from .dist import all_reduce, get_rank, get_world_size
+
+
+def _init_weight(out_f: int, in_f: int, dtype, device, gen: Optional[torch.Generator],
+                 std: float = 0.02) -> torch.Tensor:
+    w = torch.empty(out_f, in_f, dtype=torch.float32, device=device)
+    w.normal_(0.0, std, generator=gen)
+    return w.to(dtype)
+
+
+class ColumnParallelLinear:
+    """y_shard = x @ W_shard^T ; W sharded on the output dimension."""
+
+    def __init__(self, in_features: int, out_features: int, tp: Optional[int] = None,
+                 dtype=torch.bfloat16, device="cpu", gen: Optional[torch.Generator] = None):
+        self.tp = tp or get_world_size()
+        assert out_features % self.tp == 0, (out_features, self.tp)
+        self.out_per_rank = out_features // self.tp
+        self.weight = _init_weight(self.out_per_rank, in_features, dtype, device, gen)
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        return x @ self.weight.t()
+
+
+class RowParallelLinear:
+    """y = all_reduce(x_shard @ W_shard^T); W sharded on the input dimension."""
+
+    def __init__(self, in_features: int, out_features: int, tp: Optional[int] = None,
+                 dtype=torch.bfloat16, device="cpu", gen: Optional[torch.Generator] = None):
+        self.tp = tp or get_world_size()
+        assert in_features % self.tp == 0
+        self.in_per_rank = in_features // self.tp
+        self.weight = _init_weight(out_features, self.in_per_rank, dtype, device, gen)
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        y = x @ self.weight.t()
+        return all_reduce(y)
+
+
+class ReplicatedLinear:
+    def __init__(self, in_features: int, out_features: int, dtype=torch.bfloat16,
+                 device="cpu", gen: Optional[torch.Generator] = None):
+        self.weight = _init_weight(out_features, in_features, dtype, device, gen)
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        return x @ self.weight.t()

@@ -1,0 +1,176 @@
+"""Engine tests on CPU with the tiny model: paged KV equivalence,
+continuous batching, grammar-constrained generation, full orchestrator
+integration through the local engine."""
+import json
+import threading
+
+import pytest
+import torch
+
+from runbookai_amd.engine.engine import LLMEngine
+from runbookai_amd.engine.kv_cache import PagedKvCache
+from runbookai_amd.engine.llama import CONFIGS, LlamaModel
+from runbookai_amd.engine.tokenizer import ByteTokenizer, SpecialTokens
+from runbookai_amd.agent.llm_parser import (
+    PROMPT_SCHEMAS,
+    fill_prompt,
+    parse_triage_response,
+)
+
+
+class TestTokenizer:
+    def test_roundtrip(self):
+        tok = ByteTokenizer()
+        text = "redis pool exhausted — ünïcode ✓"
+        assert tok.decode(tok.encode(text)) == text
+
+    def test_chat_template(self):
+        tok = ByteTokenizer()
+        ids = tok.encode_chat("sys", "user msg")
+        assert ids[0] == tok.bos_id
+        assert SpecialTokens.EOT in ids
+        assert "user msg" in tok.decode(ids)
+
+
+class TestKvCache:
+    def test_alloc_free(self):
+        kv = PagedKvCache(2, 2, 64, num_blocks=8, block_size=16)
+        kv.allocate(1, 40)  # 3 blocks
+        assert kv.free_blocks == 5
+        kv.free(1)
+        assert kv.free_blocks == 8
+
+    def test_slot_mapping(self):
+        kv = PagedKvCache(1, 2, 64, num_blocks=8, block_size=16)
+        kv.allocate(1, 40)
+        slots = kv.slot_mapping(1, 14, 4)  # crosses a block boundary
+        table = kv.block_tables[1]
+        assert slots[0] == table[0] * 16 + 14
+        assert slots[2] == table[1] * 16 + 0
+
+    def test_exhaustion(self):
+        kv = PagedKvCache(1, 2, 64, num_blocks=2, block_size=16)
+        kv.allocate(1, 32)
+        with pytest.raises(RuntimeError):
+            kv.allocate(2, 16)
+
+
+class TestModelNumerics:
+    """Paged decode must equal a fresh full prefill on the same tokens."""
+
+    def test_decode_matches_prefill(self):
+        torch.manual_seed(0)
+        model_a = LlamaModel(CONFIGS["tiny"], device="cpu", seed=7)
+        model_b = LlamaModel(CONFIGS["tiny"], device="cpu", seed=7)
+        ids = list(range(10, 22))  # 12 tokens
+
+        # path A: prefill all 12 tokens at once
+        kv = model_a.kv
+        kv.allocate(1, len(ids))
+        slots = kv.slot_mapping(1, 0, len(ids))
+        logits_a = model_a.prefill(
+            torch.tensor(ids), torch.arange(len(ids), dtype=torch.int32),
+            torch.tensor([0, len(ids)], dtype=torch.int32), slots)
+
+        # path B: prefill 11, decode the 12th through the paged path
+        kvb = model_b.kv
+        kvb.allocate(1, len(ids))
+        slots_b = kvb.slot_mapping(1, 0, len(ids) - 1)
+        model_b.prefill(
+            torch.tensor(ids[:-1]), torch.arange(len(ids) - 1, dtype=torch.int32),
+            torch.tensor([0, len(ids) - 1], dtype=torch.int32), slots_b)
+        kvb.set_len(1, len(ids))
+        bt, lens = kvb.batch_tables([1], "cpu")
+        logits_b = model_b.decode(
+            torch.tensor([ids[-1]]), torch.tensor([len(ids) - 1], dtype=torch.int32),
+            bt, lens, kvb.slot_mapping(1, len(ids) - 1, 1))
+
+        diff = (logits_a[0].float() - logits_b[0].float()).abs().max().item()
+        assert diff < 0.05, f"paged decode diverged from prefill: {diff}"
+
+    def test_identical_seeds_identical_weights(self):
+        m1 = LlamaModel(CONFIGS["tiny"], seed=3)
+        m2 = LlamaModel(CONFIGS["tiny"], seed=3)
+        assert torch.equal(m1.layers[0].qkv.weight, m2.layers[0].qkv.weight)
+
+
+@pytest.fixture(scope="module")
+def engine():
+    eng = LLMEngine(model="tiny", device="cpu", background=False)
+    yield eng
+    eng.shutdown()
+
+
+class TestEngine:
+    def test_constrained_generation_valid_json(self, engine):
+        tok = engine.tokenizer
+        schema = PROMPT_SCHEMAS["triage"]
+        ids = tok.encode_chat("sys", "triage this incident")
+        req = engine.generate(ids, max_new_tokens=2048, schema=schema)
+        text = tok.decode(req.out_ids)
+        data = json.loads(text)  # MUST be valid JSON even with random weights
+        assert "summary" in data and "severity" in data
+        parsed = parse_triage_response(text)
+        assert parsed["severity"] in ("low", "medium", "high", "critical")
+
+    def test_unconstrained_stops(self, engine):
+        ids = engine.tokenizer.encode_chat("sys", "hello")
+        req = engine.generate(ids, max_new_tokens=16)
+        assert req.state == "done"
+        assert len(req.out_ids) <= 16
+
+    def test_kv_released(self, engine):
+        free_before = engine.model.kv.free_blocks
+        ids = engine.tokenizer.encode_chat("sys", "short")
+        engine.generate(ids, max_new_tokens=4)
+        assert engine.model.kv.free_blocks == free_before
+
+    def test_batched_requests_share_steps(self):
+        eng = LLMEngine(model="tiny", device="cpu", background=True)
+        try:
+            tok = eng.tokenizer
+            schema = PROMPT_SCHEMAS["generateConclusion"]
+            reqs = []
+            for i in range(6):
+                ids = tok.encode_chat("sys", f"case {i}")
+                reqs.append(eng.submit(ids, max_new_tokens=1024, schema=schema))
+            for r in reqs:
+                assert r.done_event.wait(timeout=120), "request did not finish"
+            for r in reqs:
+                data = json.loads(tok.decode(r.out_ids))
+                assert "rootCause" in data
+            # continuous batching actually batched: fewer steps than serial tokens
+            total_tokens = sum(len(r.out_ids) for r in reqs)
+            assert eng.stats["steps"] < total_tokens
+        finally:
+            eng.shutdown()
+
+
+class TestLocalClientOrchestration:
+    """The flagship integration: a full structured investigation through the
+    local engine with random-init weights — completes with schema-valid
+    output at every phase (BASELINE config 3 shape, tiny model on CPU)."""
+
+    def test_investigation_end_to_end(self):
+        from runbookai_amd.engine.client import LocalEngineClient
+        from runbookai_amd.agent.orchestrator import InvestigationOrchestrator
+        from runbookai_amd.providers.simulation import SimScenario, set_scenario
+        from runbookai_amd.tools.registry import ToolRegistry
+
+        set_scenario(SimScenario.redis_exhaustion())
+        eng = LLMEngine(model="tiny", device="cpu", background=False)
+        try:
+            client = LocalEngineClient(eng, max_tokens=2048)
+            registry = ToolRegistry()
+            orch = InvestigationOrchestrator(llm=client, tool_executor=registry,
+                                             max_iterations=4)
+            result = orch.investigate("Investigate PD-EXAMPLE-001",
+                                      incident_id="PD-EXAMPLE-001")
+            assert result.success, result.error
+            assert result.root_cause  # schema guarantees a non-empty rootCause
+            assert result.confidence in ("low", "medium", "high")
+            assert "complete" in result.phases_visited
+            assert orch.stats["llm_calls"] >= 4
+        finally:
+            eng.shutdown()
+            set_scenario(None)

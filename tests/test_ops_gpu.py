@@ -1,0 +1,209 @@
+"""GPU kernel numerics: every gfx950 HIP kernel vs the plain-PyTorch fp32
+reference (runbookai_amd/ops/reference.py) on random data. All @gpu."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from runbookai_amd import ops
+from runbookai_amd.ops import reference as ref
+
+DEV = "cuda:0"
+
+
+def bf(x):
+    return x.to(torch.bfloat16)
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+
+
+class TestRmsnorm:
+    @pytest.mark.parametrize("shape", [(4, 256), (33, 4096), (128, 8192)])
+    def test_vs_reference(self, shape):
+        x = bf(torch.randn(shape)).to(DEV)
+        w = bf(torch.randn(shape[-1]) * 0.1 + 1.0).to(DEV)
+        out = ops.rmsnorm(x, w)
+        expected = ref.rmsnorm(x.cpu(), w.cpu())
+        diff = (out.cpu().float() - expected.float()).abs().max().item()
+        assert diff < 2e-2, diff
+
+    def test_fused_residual(self):
+        x = bf(torch.randn(16, 4096)).to(DEV)
+        r = bf(torch.randn(16, 4096)).to(DEV)
+        w = bf(torch.ones(4096)).to(DEV)
+        out, res = ops.rmsnorm_residual(x, r, w)
+        e_out, e_res = ref.rmsnorm_residual(x.cpu(), r.cpu(), w.cpu())
+        assert (out.cpu().float() - e_out.float()).abs().max().item() < 2e-2
+        assert (res.cpu().float() - e_res.float()).abs().max().item() < 2e-2
+
+
+class TestSiluMul:
+    def test_vs_reference(self):
+        g = bf(torch.randn(1000, 512)).to(DEV)
+        u = bf(torch.randn(1000, 512)).to(DEV)
+        out = ops.silu_mul(g, u)
+        expected = ref.silu_mul(g.cpu(), u.cpu())
+        assert (out.cpu().float() - expected.float()).abs().max().item() < 2e-2
+
+
+class TestRope:
+    @pytest.mark.parametrize("D", [64, 128])
+    def test_vs_reference(self, D):
+        T, Hq, Hk = 37, 8, 2
+        cos, sin = ref.rope_cos_sin(256, D)
+        q = bf(torch.randn(T, Hq, D))
+        k = bf(torch.randn(T, Hk, D))
+        pos = torch.randint(0, 256, (T,), dtype=torch.int32)
+        eq, ek = ref.apply_rope(q, k, cos, sin, pos)
+        gq, gk = ops.apply_rope(q.to(DEV), k.to(DEV), cos.to(DEV), sin.to(DEV),
+                                pos.to(DEV))
+        assert (gq.cpu().float() - eq.float()).abs().max().item() < 2e-2
+        assert (gk.cpu().float() - ek.float()).abs().max().item() < 2e-2
+
+
+class TestStoreKv:
+    def test_scatter(self):
+        T, Hk, D, BS, NB = 21, 4, 128, 16, 8
+        k = bf(torch.randn(T, Hk, D))
+        v = bf(torch.randn(T, Hk, D))
+        slots = torch.randperm(NB * BS)[:T].to(torch.int32)
+        kc = torch.zeros(NB, Hk, BS, D, dtype=torch.bfloat16)
+        vc = torch.zeros_like(kc)
+        ref.store_kv(k, v, kc, vc, slots)
+        gkc = torch.zeros_like(kc).to(DEV)
+        gvc = torch.zeros_like(kc).to(DEV)
+        ops.store_kv(k.to(DEV), v.to(DEV), gkc, gvc, slots.to(DEV))
+        assert torch.equal(gkc.cpu(), kc)
+        assert torch.equal(gvc.cpu(), vc)
+
+
+class TestPrefillAttention:
+    @pytest.mark.parametrize("D,Hq,Hk", [(128, 8, 2), (64, 4, 4)])
+    def test_varlen_causal_gqa(self, D, Hq, Hk):
+        lens = [17, 64, 5]
+        T = sum(lens)
+        starts = torch.tensor([0, *torch.tensor(lens).cumsum(0).tolist()],
+                              dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.5)
+        k = bf(torch.randn(T, Hk, D) * 0.5)
+        v = bf(torch.randn(T, Hk, D) * 0.5)
+        expected = ref.prefill_attention(q, k, v, starts, causal=True)
+        out = ops.prefill_attention(q.to(DEV), k.to(DEV), v.to(DEV), starts.to(DEV),
+                                    causal=True)
+        diff = (out.cpu().float() - expected.float()).abs().max().item()
+        assert diff < 3e-2, diff
+
+    def test_non_causal(self):
+        T, Hq, Hk, D = 48, 4, 4, 64
+        starts = torch.tensor([0, T], dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.5)
+        k = bf(torch.randn(T, Hk, D) * 0.5)
+        v = bf(torch.randn(T, Hk, D) * 0.5)
+        expected = ref.prefill_attention(q, k, v, starts, causal=False)
+        out = ops.prefill_attention(q.to(DEV), k.to(DEV), v.to(DEV), starts.to(DEV),
+                                    causal=False)
+        assert (out.cpu().float() - expected.float()).abs().max().item() < 3e-2
+
+    def test_softmax_spike(self):
+        """One dominant key: probability mass must follow it (online-softmax
+        rescale correctness per guide §5.4 rule 26)."""
+        T, Hq, Hk, D = 33, 2, 2, 128
+        starts = torch.tensor([0, T], dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.1)
+        k = bf(torch.randn(T, Hk, D) * 0.1)
+        v = bf(torch.randn(T, Hk, D) * 0.5)
+        # spike key 7 against query 30
+        k[7] = (q[30, :Hk] * 20.0).to(torch.bfloat16)
+        expected = ref.prefill_attention(q, k, v, starts, causal=True)
+        out = ops.prefill_attention(q.to(DEV), k.to(DEV), v.to(DEV), starts.to(DEV))
+        assert (out.cpu().float() - expected.float()).abs().max().item() < 3e-2
+
+
+class TestPagedDecode:
+    @pytest.mark.parametrize("D,Hq,Hk", [(128, 32, 8), (64, 4, 2)])
+    def test_vs_reference(self, D, Hq, Hk):
+        B, BS, NB = 4, 16, 64
+        lens = torch.tensor([3, 17, 150, 64], dtype=torch.int32)
+        max_blocks = int((lens.max().item() + BS - 1) // BS)
+        bt = torch.full((B, max_blocks), -1, dtype=torch.int32)
+        used = iter(torch.randperm(NB).tolist())
+        for b in range(B):
+            nb = (int(lens[b]) + BS - 1) // BS
+            bt[b, :nb] = torch.tensor([next(used) for _ in range(nb)], dtype=torch.int32)
+        kc = bf(torch.randn(NB, Hk, BS, D) * 0.5)
+        vc = bf(torch.randn(NB, Hk, BS, D) * 0.5)
+        q = bf(torch.randn(B, Hq, D) * 0.5)
+        expected = ref.paged_decode_attention(q, kc, vc, bt, lens)
+        out = ops.paged_decode_attention(q.to(DEV), kc.to(DEV), vc.to(DEV),
+                                         bt.to(DEV), lens.to(DEV))
+        diff = (out.cpu().float() - expected.float()).abs().max().item()
+        assert diff < 3e-2, diff
+
+
+class TestRetrievalSampling:
+    def test_topk_cosine(self):
+        N, D, K = 5000, 384, 8
+        m = torch.randn(N, D)
+        m = (m / m.norm(dim=1, keepdim=True)).half().to(DEV)
+        q = torch.randn(D)
+        q = (q / q.norm()).half().to(DEV)
+        vals, idx = ops.topk_cosine(m, q, K)
+        e_vals, e_idx = ref.topk_cosine(m.cpu().float(), q.cpu().float(), K)
+        assert set(idx.cpu().tolist()) == set(e_idx.tolist())
+        assert (vals.cpu() - e_vals).abs().max().item() < 1e-2
+
+    def test_masked_argmax(self):
+        B, V = 8, 128256
+        logits = bf(torch.randn(B, V)).to(DEV)
+        mask = torch.zeros(B, V, dtype=torch.bool)
+        mask[:, :300] = True
+        out = ops.masked_greedy(logits, mask.to(DEV))
+        expected = ref.masked_sample(logits.cpu(), mask, temperature=0.0)
+        assert torch.equal(out.cpu(), expected)
+
+    def test_argmax_unmasked(self):
+        logits = bf(torch.randn(4, 1000)).to(DEV)
+        out = ops.masked_greedy(logits, None)
+        expected = logits.cpu().float().argmax(-1)
+        assert torch.equal(out.cpu(), expected)
+
+
+class TestEngineGpu:
+    def test_tiny_engine_constrained_on_gpu(self):
+        import json
+
+        from runbookai_amd.agent.llm_parser import PROMPT_SCHEMAS
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device=DEV, background=False)
+        try:
+            ids = eng.tokenizer.encode_chat("sys", "evaluate evidence")
+            req = eng.generate(ids, max_new_tokens=2048,
+                               schema=PROMPT_SCHEMAS["evaluateEvidence"])
+            data = json.loads(eng.tokenizer.decode(req.out_ids))
+            assert data["action"] in ("branch", "prune", "confirm", "continue")
+            assert ops.extension_loaded()
+        finally:
+            eng.shutdown()
+
+    def test_gpu_matches_cpu_tiny_prefill(self):
+        """Same seed tiny model: GPU logits ≈ CPU logits (bf16 tolerance)."""
+        from runbookai_amd.engine.llama import CONFIGS, LlamaModel
+
+        ids = list(range(50, 80))
+        out = {}
+        for dev in ("cpu", DEV):
+            m = LlamaModel(CONFIGS["tiny"], device=dev, seed=11)
+            m.kv.allocate(1, len(ids))
+            logits = m.prefill(
+                torch.tensor(ids), torch.arange(len(ids), dtype=torch.int32),
+                torch.tensor([0, len(ids)], dtype=torch.int32),
+                m.kv.slot_mapping(1, 0, len(ids)))
+            out[dev] = logits.float().cpu()
+        diff = (out["cpu"] - out[DEV]).abs().max().item()
+        assert diff < 0.1, f"GPU/CPU logits diverged: {diff}"
