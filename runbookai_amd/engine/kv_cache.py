@@ -35,7 +35,9 @@ class PagedKvCache:
         shape = (num_blocks, num_kv_heads, block_size, head_dim)
         self.k = [torch.zeros(shape, dtype=dtype, device=device) for _ in range(num_layers)]
         self.v = [torch.zeros(shape, dtype=dtype, device=device) for _ in range(num_layers)]
-        self._free: list[int] = list(range(num_blocks - 1, -1, -1))
+        # last block is reserved as scratch for hipGraph padding rows
+        self.scratch_block = num_blocks - 1
+        self._free: list[int] = list(range(num_blocks - 2, -1, -1))
         self.block_tables: dict[int, list[int]] = {}   # seq_id -> block ids
         self.seq_lens: dict[int, int] = {}
 

@@ -114,6 +114,11 @@ class LlamaModel:
         self.kv = PagedKvCache(cfg.num_layers, self.layers[0].kv_heads_per_rank,
                                cfg.head_dim, kv_blocks, kv_block_size, dtype, device)
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        import os as _os
+
+        self.use_graphs = (str(device).startswith("cuda")
+                           and _os.environ.get("RUNBOOKAI_NO_GRAPHS", "0") != "1")
+        self._graphs: dict[int, tuple] = {}
 
     # -- setup -------------------------------------------------------------------
 
@@ -215,22 +220,95 @@ class LlamaModel:
         h_last = ops.rmsnorm(h[last], self.final_norm_w, self.cfg.rms_eps)
         return self.lm_head(h_last)
 
-    def decode(self, token_ids: torch.Tensor, positions: torch.Tensor,
-               block_tables: torch.Tensor, seq_lens: torch.Tensor,
-               slot_mapping: torch.Tensor) -> torch.Tensor:
-        """One-token step for B sequences. Returns logits [B, vocab]."""
-        device = self.device
-        token_ids = token_ids.to(device)
-        positions = positions.to(torch.int32).to(device)
-        block_tables = block_tables.to(device)
-        seq_lens = seq_lens.to(device)
-        slots = slot_mapping.to(torch.int32).to(device)
-        h = self.embed.weight[token_ids.long()]
+    def _decode_impl(self, token_ids, positions, block_tables, seq_lens, slots):
+        """Device-tensor decode body (hipGraph-capturable)."""
+        h = self.embed.weight[token_ids]
         for i, layer in enumerate(self.layers):
             h = self._layer_forward_decode(layer, h, positions, block_tables,
                                            seq_lens, slots, i)
         h = ops.rmsnorm(h, self.final_norm_w, self.cfg.rms_eps)
         return self.lm_head(h)
+
+    def decode(self, token_ids: torch.Tensor, positions: torch.Tensor,
+               block_tables: torch.Tensor, seq_lens: torch.Tensor,
+               slot_mapping: torch.Tensor) -> torch.Tensor:
+        """One-token step for B sequences. Returns logits [B, vocab].
+
+        On GPU, the whole ~10*L-kernel decode body replays as ONE hipGraph
+        per padded batch size (decode is launch-bound otherwise); padding
+        rows attend into a reserved scratch KV block.
+        """
+        device = self.device
+        token_ids = token_ids.long()
+        positions = positions.to(torch.int32)
+        seq_lens = seq_lens.to(torch.int32)
+        slots = slot_mapping.to(torch.int32)
+        if self.use_graphs and device != "cpu":
+            return self._decode_with_graph(token_ids, positions, block_tables,
+                                           seq_lens, slots)
+        return self._decode_impl(token_ids.to(device), positions.to(device),
+                                 block_tables.to(device), seq_lens.to(device),
+                                 slots.to(device))
+
+    # -- hipGraph decode ----------------------------------------------------------
+
+    GRAPH_SIZES = (1, 2, 4, 8, 16, 32, 64)
+
+    def _graph_max_blocks(self) -> int:
+        return (self.cfg.max_seq_len + self.kv.block_size - 1) // self.kv.block_size
+
+    def _ensure_graph(self, bpad: int):
+        if bpad in self._graphs:
+            return self._graphs[bpad]
+        dev = self.device
+        maxb = self._graph_max_blocks()
+        static = {
+            "ids": torch.zeros(bpad, dtype=torch.long, device=dev),
+            "pos": torch.zeros(bpad, dtype=torch.int32, device=dev),
+            "bt": torch.zeros((bpad, maxb), dtype=torch.int32, device=dev),
+            "lens": torch.ones(bpad, dtype=torch.int32, device=dev),
+            "slots": torch.full((bpad,), self.kv.scratch_block * self.kv.block_size,
+                                dtype=torch.int32, device=dev),
+        }
+        static["bt"].fill_(self.kv.scratch_block)
+        # warmup on a side stream (required before capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._decode_impl(static["ids"], static["pos"], static["bt"],
+                                  static["lens"], static["slots"])
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static["out"] = self._decode_impl(static["ids"], static["pos"], static["bt"],
+                                              static["lens"], static["slots"])
+        self._graphs[bpad] = (graph, static)
+        return self._graphs[bpad]
+
+    def _decode_with_graph(self, token_ids, positions, block_tables, seq_lens, slots):
+        B = token_ids.shape[0]
+        bpad = next((s for s in self.GRAPH_SIZES if s >= B), None)
+        if bpad is None:
+            dev = self.device
+            return self._decode_impl(token_ids.to(dev), positions.to(dev),
+                                     block_tables.to(dev), seq_lens.to(dev),
+                                     slots.to(dev))
+        graph, static = self._ensure_graph(bpad)
+        scratch_slot = self.kv.scratch_block * self.kv.block_size
+        static["ids"][:B].copy_(token_ids, non_blocking=True)
+        static["ids"][B:].zero_()
+        static["pos"][:B].copy_(positions, non_blocking=True)
+        static["pos"][B:].zero_()
+        static["bt"].fill_(self.kv.scratch_block)
+        nb = block_tables.shape[1]
+        static["bt"][:B, :nb].copy_(block_tables, non_blocking=True)
+        static["lens"][:B].copy_(seq_lens, non_blocking=True)
+        static["lens"][B:].fill_(1)
+        static["slots"][:B].copy_(slots, non_blocking=True)
+        static["slots"][B:].fill_(scratch_slot)
+        graph.replay()
+        return static["out"][:B]
 
     @staticmethod
     def _batch_idx(seq_starts: torch.Tensor, T: int) -> torch.Tensor:

@@ -34,14 +34,15 @@ class TestTokenizer:
 
 class TestKvCache:
     def test_alloc_free(self):
-        kv = PagedKvCache(2, 2, 64, num_blocks=8, block_size=16)
+        kv = PagedKvCache(2, 2, 64, num_blocks=9, block_size=16)
+        assert kv.free_blocks == 8  # one block reserved as graph scratch
         kv.allocate(1, 40)  # 3 blocks
         assert kv.free_blocks == 5
         kv.free(1)
         assert kv.free_blocks == 8
 
     def test_slot_mapping(self):
-        kv = PagedKvCache(1, 2, 64, num_blocks=8, block_size=16)
+        kv = PagedKvCache(1, 2, 64, num_blocks=9, block_size=16)
         kv.allocate(1, 40)
         slots = kv.slot_mapping(1, 14, 4)  # crosses a block boundary
         table = kv.block_tables[1]
@@ -49,7 +50,7 @@ class TestKvCache:
         assert slots[2] == table[1] * 16 + 0
 
     def test_exhaustion(self):
-        kv = PagedKvCache(1, 2, 64, num_blocks=2, block_size=16)
+        kv = PagedKvCache(1, 2, 64, num_blocks=3, block_size=16)
         kv.allocate(1, 32)
         with pytest.raises(RuntimeError):
             kv.allocate(2, 16)
