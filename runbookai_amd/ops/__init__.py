@@ -80,17 +80,43 @@ def apply_rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.T
     return reference.apply_rope(q, k, cos, sin, positions)
 
 
+USE_FLASH_PREFILL = True
+
+
 def prefill_attention(q, k, v, seq_starts, causal: bool = True,
                       scale: Optional[float] = None, batch_idx=None):
     scale = scale or 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
+        starts_i32 = seq_starts.to(torch.int32)
+        if USE_FLASH_PREFILL and q.shape[-1] == 128:
+            tb, tq = _build_qtiles(starts_i32)
+            return _get_ext().flash_prefill(q.contiguous(), k.contiguous(),
+                                            v.contiguous(), tb.to(q.device),
+                                            tq.to(q.device),
+                                            starts_i32.contiguous().to(q.device),
+                                            scale, causal)
         if batch_idx is None:
             batch_idx = _batch_idx_from_starts(seq_starts, q.shape[0])
         return _get_ext().prefill_attn(q.contiguous(), k.contiguous(), v.contiguous(),
                                        batch_idx.to(torch.int32).contiguous(),
-                                       seq_starts.to(torch.int32).contiguous(),
+                                       starts_i32.contiguous(),
                                        scale, causal)
     return reference.prefill_attention(q, k, v, seq_starts, causal, scale)
+
+
+def _build_qtiles(seq_starts: torch.Tensor, qtile: int = 64):
+    """Per-64-row q-tile (segment id, global q start) arrays for the flash
+    prefill grid; tiles never span segment boundaries."""
+    starts = seq_starts.cpu().tolist()
+    tb: list[int] = []
+    tq: list[int] = []
+    for b in range(len(starts) - 1):
+        s, e = starts[b], starts[b + 1]
+        for q0 in range(s, e, qtile):
+            tb.append(b)
+            tq.append(q0)
+    return (torch.tensor(tb, dtype=torch.int32),
+            torch.tensor(tq, dtype=torch.int32))
 
 
 def paged_decode_attention(q, k_cache, v_cache, block_tables, seq_lens,

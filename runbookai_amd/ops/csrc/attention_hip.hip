@@ -123,6 +123,134 @@ extern "C" void launch_paged_decode(const void* q, const void* kc, const void* v
                        Hq, Hk, D, BS, max_blocks, scale);
 }
 
+// ---------------------------------------------------- split-K decode (flash)
+// Long contexts leave B*Hq workgroups streaming KV one token per wave —
+// latency-bound (measured ~150 GB/s effective). Split each sequence's KV
+// range across NSPLIT workgroups (fp32 partials), then merge: parallelism
+// becomes B*Hq*NSPLIT workgroups and the chip reaches its HBM stream rate.
+__global__ void paged_decode_partial_kernel(
+    const ushort_t* __restrict__ q, const ushort_t* __restrict__ k_cache,
+    const ushort_t* __restrict__ v_cache, const int* __restrict__ block_tables,
+    const int* __restrict__ seq_lens,
+    float* __restrict__ part_m,     // [B, Hq, NSPLIT]
+    float* __restrict__ part_l,     // [B, Hq, NSPLIT]
+    float* __restrict__ part_acc,   // [B, Hq, NSPLIT, D]
+    int Hq, int Hk, int D, int BS, int max_blocks, int nsplit, float scale) {
+    const int b = blockIdx.x;
+    const int h = blockIdx.y;
+    const int split = blockIdx.z;
+    const int hk = h / (Hq / Hk);
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    const int nw = blockDim.x / WAVE;
+    const int L = seq_lens[b];
+    const int halfD = D / 2;
+    const bool active = lane < halfD;
+    const long pidx = ((long)b * Hq + h) * nsplit + split;
+
+    const int chunk = (L + nsplit - 1) / nsplit;
+    const int t_begin = split * chunk;
+    const int t_end = min(L, t_begin + chunk);
+
+    float q0 = 0.f, q1 = 0.f;
+    if (active) {
+        ushort2_t qp = *reinterpret_cast<const ushort2_t*>(
+            q + ((long)b * Hq + h) * D + 2 * lane);
+        q0 = bf2f(qp[0]) * scale;
+        q1 = bf2f(qp[1]) * scale;
+    }
+
+    OnlineAcc acc;
+    online_init(acc);
+    const int* bt = block_tables + (long)b * max_blocks;
+    for (int t = t_begin + wid; t < t_end; t += nw) {
+        const int blk = bt[t / BS];
+        const long base = (((long)blk * Hk + hk) * BS + (t % BS)) * D;
+        float s = 0.f, v0 = 0.f, v1 = 0.f;
+        if (active) {
+            ushort2_t kp = *reinterpret_cast<const ushort2_t*>(k_cache + base + 2 * lane);
+            s = q0 * bf2f(kp[0]) + q1 * bf2f(kp[1]);
+            ushort2_t vp = *reinterpret_cast<const ushort2_t*>(v_cache + base + 2 * lane);
+            v0 = bf2f(vp[0]);
+            v1 = bf2f(vp[1]);
+        }
+        s = wave_sum(s);
+        online_update(acc, s, v0, v1);
+    }
+
+    // cross-wave merge, then write fp32 partial
+    __shared__ float lm[8], ll[8];
+    __shared__ float la[8][128];
+    if (lane == 0) { lm[wid] = acc.m; ll[wid] = acc.l; }
+    if (active) { la[wid][2 * lane] = acc.a0; la[wid][2 * lane + 1] = acc.a1; }
+    __syncthreads();
+    if (wid == 0) {
+        float m_all = -1e30f;
+        for (int w = 0; w < nw; ++w) m_all = fmaxf(m_all, lm[w]);
+        float l_all = 0.f, o0 = 0.f, o1 = 0.f;
+        for (int w = 0; w < nw; ++w) {
+            float c = __expf(lm[w] - m_all);
+            l_all += ll[w] * c;
+            if (active) {
+                o0 += la[w][2 * lane] * c;
+                o1 += la[w][2 * lane + 1] * c;
+            }
+        }
+        if (lane == 0) { part_m[pidx] = m_all; part_l[pidx] = l_all; }
+        if (active) {
+            part_acc[pidx * D + 2 * lane] = o0;
+            part_acc[pidx * D + 2 * lane + 1] = o1;
+        }
+    }
+}
+
+__global__ void paged_decode_merge_kernel(
+    const float* __restrict__ part_m, const float* __restrict__ part_l,
+    const float* __restrict__ part_acc, ushort_t* __restrict__ out,
+    int Hq, int D, int nsplit) {
+    const int b = blockIdx.x;
+    const int h = blockIdx.y;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int halfD = D / 2;
+    const bool active = lane < halfD;
+    const long base = ((long)b * Hq + h) * nsplit;
+    float m_all = -1e30f;
+    for (int s = 0; s < nsplit; ++s) m_all = fmaxf(m_all, part_m[base + s]);
+    float l_all = 0.f, o0 = 0.f, o1 = 0.f;
+    for (int s = 0; s < nsplit; ++s) {
+        const float c = __expf(part_m[base + s] - m_all);
+        l_all += part_l[base + s] * c;
+        if (active) {
+            o0 += part_acc[(base + s) * D + 2 * lane] * c;
+            o1 += part_acc[(base + s) * D + 2 * lane + 1] * c;
+        }
+    }
+    if (active) {
+        const float inv = (l_all > 0.f) ? 1.f / l_all : 0.f;
+        ushort2_t o;
+        o[0] = f2bf(o0 * inv);
+        o[1] = f2bf(o1 * inv);
+        *reinterpret_cast<ushort2_t*>(out + ((long)b * Hq + h) * D + 2 * lane) = o;
+    }
+}
+
+extern "C" void launch_paged_decode_splitk(
+    const void* q, const void* kc, const void* vc, const void* bt, const void* lens,
+    void* part_m, void* part_l, void* part_acc, void* out,
+    int B, int Hq, int Hk, int D, int BS, int max_blocks, int nsplit, float scale,
+    hipStream_t stream) {
+    dim3 grid(B, Hq, nsplit), block(256);
+    hipLaunchKernelGGL(paged_decode_partial_kernel, grid, block, 0, stream,
+                       (const ushort_t*)q, (const ushort_t*)kc, (const ushort_t*)vc,
+                       (const int*)bt, (const int*)lens,
+                       (float*)part_m, (float*)part_l, (float*)part_acc,
+                       Hq, Hk, D, BS, max_blocks, nsplit, scale);
+    dim3 grid2(B, Hq), block2(WAVE);
+    hipLaunchKernelGGL(paged_decode_merge_kernel, grid2, block2, 0, stream,
+                       (const float*)part_m, (const float*)part_l,
+                       (const float*)part_acc, (ushort_t*)out, Hq, D, nsplit);
+}
+
 // ------------------------------------------------------------ prefill v1
 // Packed varlen: q:[T,Hq,D] k,v:[T,Hk,D]; batch_idx:[T]; seq_starts:[B+1].
 // One 4-wave workgroup per (token, head); waves split the KV range.

@@ -14,10 +14,17 @@ void launch_rope(void*, void*, const void*, const void*, const void*,
                  int, int, int, int, void*);
 void launch_paged_decode(const void*, const void*, const void*, const void*,
                          const void*, void*, int, int, int, int, int, int, float, void*);
+void launch_paged_decode_splitk(const void*, const void*, const void*, const void*,
+                                const void*, void*, void*, void*, void*,
+                                int, int, int, int, int, int, int, float, void*);
 void launch_prefill(const void*, const void*, const void*, const void*, const void*,
                     void*, int, int, int, int, float, int, void*);
 void launch_store_kv(const void*, const void*, void*, void*, const void*,
                      int, int, int, int, void*);
+void launch_flash_prefill(const void*, const void*, const void*, const void*,
+                          const void*, const void*, void*, int, int, int, float,
+                          int, void*);
+void launch_mfma_probe(const void*, const void*, void*, void*);
 void launch_cosine_scores(const void*, const void*, void*, long, int, void*);
 void launch_masked_argmax(const void*, const void*, void*, int, int, void*);
 }
@@ -97,9 +104,25 @@ torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
     int Hk = (int)kc.size(1), BS = (int)kc.size(2);
     TORCH_CHECK(D == 64 || D == 128, "head dim must be 64 or 128");
     auto out = torch::empty_like(q);
-    launch_paged_decode(q.data_ptr(), kc.data_ptr(), vc.data_ptr(), bt.data_ptr(),
-                        lens.data_ptr(), out.data_ptr(), B, Hq, Hk, D, BS,
-                        (int)bt.size(1), (float)scale, current_stream());
+    // split-K degree: enough workgroups to fill 256 CUs several times over
+    int nsplit = 1;
+    long base_wgs = (long)B * Hq;
+    while (nsplit < 16 && base_wgs * nsplit < 2048) nsplit *= 2;
+    if (nsplit <= 1) {
+        launch_paged_decode(q.data_ptr(), kc.data_ptr(), vc.data_ptr(), bt.data_ptr(),
+                            lens.data_ptr(), out.data_ptr(), B, Hq, Hk, D, BS,
+                            (int)bt.size(1), (float)scale, current_stream());
+        return out;
+    }
+    auto fopt = q.options().dtype(torch::kFloat32);
+    auto part_m = torch::empty({B, Hq, nsplit}, fopt);
+    auto part_l = torch::empty({B, Hq, nsplit}, fopt);
+    auto part_acc = torch::empty({B, Hq, nsplit, D}, fopt);
+    launch_paged_decode_splitk(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                               bt.data_ptr(), lens.data_ptr(), part_m.data_ptr(),
+                               part_l.data_ptr(), part_acc.data_ptr(), out.data_ptr(),
+                               B, Hq, Hk, D, BS, (int)bt.size(1), nsplit, (float)scale,
+                               current_stream());
     return out;
 }
 
@@ -119,6 +142,34 @@ torch::Tensor prefill_attn(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                    seq_starts.data_ptr(), out.data_ptr(), T, Hq, Hk, D,
                    (float)scale, causal ? 1 : 0, current_stream());
     return out;
+}
+
+torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                            torch::Tensor tile_batch, torch::Tensor tile_qstart,
+                            torch::Tensor seq_starts, double scale, bool causal) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(k, torch::kBFloat16);
+    CHECK_IN(v, torch::kBFloat16);
+    CHECK_IN(tile_batch, torch::kInt32);
+    CHECK_IN(tile_qstart, torch::kInt32);
+    CHECK_IN(seq_starts, torch::kInt32);
+    int Hq = (int)q.size(1), Hk = (int)k.size(1), D = (int)q.size(2);
+    TORCH_CHECK(D == 128, "flash prefill supports head dim 128");
+    auto out = torch::empty_like(q);
+    launch_flash_prefill(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                         tile_batch.data_ptr(), tile_qstart.data_ptr(),
+                         seq_starts.data_ptr(), out.data_ptr(),
+                         (int)tile_batch.size(0), Hq, Hk, (float)scale,
+                         causal ? 1 : 0, current_stream());
+    return out;
+}
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+    CHECK_IN(A, torch::kBFloat16);
+    CHECK_IN(B, torch::kBFloat16);
+    auto C = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+    launch_mfma_probe(A.data_ptr(), B.data_ptr(), C.data_ptr(), current_stream());
+    return C;
 }
 
 void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor kc, torch::Tensor vc,
@@ -169,6 +220,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rope_inplace", &rope_inplace, "RoPE in place on q,k");
     m.def("paged_decode", &paged_decode, "paged-KV decode attention");
     m.def("prefill_attn", &prefill_attn, "varlen prefill attention");
+    m.def("flash_prefill", &flash_prefill, "MFMA flash prefill attention (D=128)");
+    m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
     m.def("store_kv", &store_kv, "scatter K/V into paged cache");
     m.def("cosine_scores", &cosine_scores, "brute-force cosine scores");
     m.def("masked_argmax", &masked_argmax, "greedy sampling under validity mask");

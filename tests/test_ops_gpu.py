@@ -124,6 +124,64 @@ class TestPrefillAttention:
         assert (out.cpu().float() - expected.float()).abs().max().item() < 3e-2
 
 
+class TestMfmaLayout:
+    def test_probe_vs_matmul(self):
+        """A=I with ASYMMETRIC B catches transposes (guide §3)."""
+        from runbookai_amd.ops import _get_ext
+
+        ext = _get_ext()
+        A = torch.zeros(16, 32)
+        for i in range(16):
+            A[i, i] = 1.0
+        B = torch.arange(32 * 16, dtype=torch.float32).reshape(32, 16) / 100.0
+        C = ext.mfma_probe(bf(A).to(DEV), bf(B).to(DEV)).cpu()
+        expected = (bf(A).float() @ bf(B).float())
+        assert (C - expected).abs().max().item() < 1e-2, \
+            f"MFMA fragment layout mismatch: {(C - expected).abs().max().item()}"
+
+    def test_probe_random(self):
+        from runbookai_amd.ops import _get_ext
+
+        ext = _get_ext()
+        A = torch.randn(16, 32)
+        B = torch.randn(32, 16)
+        C = ext.mfma_probe(bf(A).to(DEV), bf(B).to(DEV)).cpu()
+        expected = bf(A).float() @ bf(B).float()
+        assert (C - expected).abs().max().item() < 0.1
+
+
+class TestFlashPrefill:
+    @pytest.mark.parametrize("lens", [[64], [17, 130, 5], [200, 64]])
+    @pytest.mark.parametrize("causal", [True, False])
+    def test_vs_reference(self, lens, causal):
+        Hq, Hk, D = 8, 2, 128
+        T = sum(lens)
+        starts = torch.tensor([0, *torch.tensor(lens).cumsum(0).tolist()],
+                              dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.5)
+        k = bf(torch.randn(T, Hk, D) * 0.5)
+        v = bf(torch.randn(T, Hk, D) * 0.5)
+        expected = ref.prefill_attention(q, k, v, starts, causal=causal)
+        out = ops.prefill_attention(q.to(DEV), k.to(DEV), v.to(DEV), starts.to(DEV),
+                                    causal=causal)
+        diff = (out.cpu().float() - expected.float()).abs().max().item()
+        assert diff < 4e-2, diff
+
+    def test_spike_rescale(self):
+        """Forced online-softmax rescale at a chosen tile (guide rule 26)."""
+        Hq, Hk, D = 4, 4, 128
+        T = 200
+        starts = torch.tensor([0, T], dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.1)
+        k = bf(torch.randn(T, Hk, D) * 0.1)
+        v = bf(torch.randn(T, Hk, D) * 0.5)
+        k[150] = (q[190] * 25.0).to(torch.bfloat16)  # max jumps at tile 4
+        expected = ref.prefill_attention(q, k, v, starts, causal=True)
+        out = ops.prefill_attention(q.to(DEV), k.to(DEV), v.to(DEV), starts.to(DEV),
+                                    causal=True)
+        assert (out.cpu().float() - expected.float()).abs().max().item() < 4e-2
+
+
 class TestPagedDecode:
     @pytest.mark.parametrize("D,Hq,Hk", [(128, 32, 8), (64, 4, 2)])
     def test_vs_reference(self, D, Hq, Hk):
