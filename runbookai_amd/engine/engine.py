@@ -41,6 +41,7 @@ class Request:
     schema: Optional[dict[str, Any]] = None
     fsm: Optional[JsonFsm] = None
     out_ids: list[int] = field(default_factory=list)
+    pending_input: list[int] = field(default_factory=list)  # tokens not yet in KV
     state: str = "waiting"           # waiting | running | done
     done_event: threading.Event = field(default_factory=threading.Event)
     prompt_len: int = 0
@@ -105,6 +106,10 @@ class LLMEngine:
                 fsm=JsonFsm(schema) if schema else None,
             )
             req.prompt_len = len(req.prompt_ids)
+            # grammar-forced prefix (e.g. '{"summary": "') enters WITH the
+            # prompt prefill — zero decode steps spent on forced structure
+            prefix = self._drain_forced(req)
+            req.pending_input = req.prompt_ids + prefix
             self.waiting.append(req)
             self.stats["requests"] += 1
             self._lock.notify_all()
@@ -162,16 +167,21 @@ class LLMEngine:
 
     def step(self) -> None:
         """One engine iteration: admit a prefill batch if any request is
-        waiting (and fits), else run one decode step over all running."""
+        waiting (and fits); else flush multi-token chunks (grammar-forced
+        runs) through the chunked-prefill path; else one decode step over
+        all running sequences."""
         with self._lock:
             prefill_batch = self._admit_locked()
         if prefill_batch:
             self._run_prefill(prefill_batch)
         else:
             with self._lock:
-                decode_batch = list(self.running)
-            if decode_batch:
-                self._run_decode(decode_batch)
+                multis = [r for r in self.running if len(r.pending_input) > 1]
+                ones = [r for r in self.running if len(r.pending_input) == 1]
+            if multis:
+                self._run_chunk(multis)
+            elif ones:
+                self._run_decode(ones)
         self.stats["steps"] += 1
 
     def _admit_locked(self) -> list[Request]:
@@ -180,15 +190,15 @@ class LLMEngine:
         kv = self.model.kv
         while self.waiting and len(self.running) + len(batch) < self.max_batch:
             req = self.waiting[0]
-            need = req.prompt_len + req.max_new_tokens
-            if batch and tokens + req.prompt_len > self.max_prefill_tokens:
+            need = len(req.pending_input) + req.max_new_tokens
+            if batch and tokens + len(req.pending_input) > self.max_prefill_tokens:
                 break
             if not kv.can_allocate(need):
                 break
             kv.allocate(req.rid, need)
             self.waiting.pop(0)
             batch.append(req)
-            tokens += req.prompt_len
+            tokens += len(req.pending_input)
         return batch
 
     # -- execution --------------------------------------------------------------------
@@ -201,12 +211,14 @@ class LLMEngine:
         starts = [0]
         slots: list[torch.Tensor] = []
         for req in batch:
-            token_ids.extend(req.prompt_ids)
-            positions.extend(range(req.prompt_len))
-            starts.append(starts[-1] + req.prompt_len)
-            slots.append(kv.slot_mapping(req.rid, 0, req.prompt_len))
-            kv.set_len(req.rid, req.prompt_len)
-            req.pos = req.prompt_len
+            n = len(req.pending_input)
+            token_ids.extend(req.pending_input)
+            positions.extend(range(n))
+            starts.append(starts[-1] + n)
+            slots.append(kv.slot_mapping(req.rid, 0, n))
+            kv.set_len(req.rid, n)
+            req.pos = n
+            req.pending_input = []
         logits = self.model.prefill(
             torch.tensor(token_ids, dtype=torch.int64),
             torch.tensor(positions, dtype=torch.int32),
@@ -222,6 +234,41 @@ class LLMEngine:
                     self.running.append(req)
         self.stats["prefill_time"] += time.time() - t0
 
+    def _run_chunk(self, batch: list[Request]) -> None:
+        """Multi-token append (grammar-forced runs) via chunked prefill."""
+        t0 = time.time()
+        kv = self.model.kv
+        token_ids: list[int] = []
+        positions: list[int] = []
+        starts = [0]
+        slots: list[torch.Tensor] = []
+        hist: list[int] = []
+        for req in batch:
+            n = len(req.pending_input)
+            token_ids.extend(req.pending_input)
+            positions.extend(range(req.pos, req.pos + n))
+            starts.append(starts[-1] + n)
+            kv.extend(req.rid, req.pos + n)
+            slots.append(kv.slot_mapping(req.rid, req.pos, n))
+            hist.append(req.pos)
+            kv.set_len(req.rid, req.pos + n)
+            req.pos += n
+            req.pending_input = []
+        bt, _lens = kv.batch_tables([r.rid for r in batch], self.device)
+        logits = self.model.chunk_step(
+            torch.tensor(token_ids, dtype=torch.int64),
+            torch.tensor(positions, dtype=torch.int32),
+            torch.tensor(starts, dtype=torch.int32),
+            bt,
+            torch.tensor(hist, dtype=torch.int32),
+            torch.cat(slots),
+        )
+        self.stats["chunk_tokens"] = self.stats.get("chunk_tokens", 0) + len(token_ids)
+        self._sample_and_advance(batch, logits)
+        with self._lock:
+            self.running = [r for r in self.running if r.state != "done"]
+        self.stats["decode_time"] += time.time() - t0
+
     def _run_decode(self, batch: list[Request]) -> None:
         t0 = time.time()
         kv = self.model.kv
@@ -229,20 +276,19 @@ class LLMEngine:
         positions = []
         slot_list = []
         for req in batch:
-            tok = req.out_ids[-1] if req.out_ids else (req.prompt_ids[-1] if req.prompt_ids else 0)
-            input_ids.append(tok)
+            input_ids.append(req.pending_input[0])
             positions.append(req.pos)
             kv.extend(req.rid, req.pos + 1)
             slot_list.append(kv.slot_mapping(req.rid, req.pos, 1))
             kv.set_len(req.rid, req.pos + 1)
+            req.pos += 1
+            req.pending_input = []
         bt, lens = kv.batch_tables([r.rid for r in batch], self.device)
         logits = self.model.decode(
             torch.tensor(input_ids, dtype=torch.int64),
             torch.tensor(positions, dtype=torch.int32),
             bt, lens, torch.cat(slot_list),
         )
-        for req in batch:
-            req.pos += 1
         self.stats["decode_tokens"] += len(batch)
         self._sample_and_advance(batch, logits)
         with self._lock:
@@ -277,10 +323,41 @@ class LLMEngine:
             chosen = ops.masked_sample(region, mask_d, temperature=temp)
         chosen = chosen.cpu().tolist()
         now = time.time()
-        for req, tok in zip(batch, chosen):
+        region_cpu = None
+        for i, req in enumerate(batch):
             if req.first_token_at == 0.0:
                 req.first_token_at = now
-            self._advance_request(req, int(tok))
+            tok = int(chosen[i])
+            # zero-width sentinel (number close): advance FSM and re-choose
+            # from the SAME logits row until a real byte appears
+            while (req.fsm is not None and tok == NUMBER_CLOSE_SENTINEL
+                   and not req.fsm.done):
+                req.fsm.advance(NUMBER_CLOSE_SENTINEL)
+                allowed = req.fsm.allowed_bytes()
+                if not allowed:
+                    tok = SpecialTokens.EOT
+                    break
+                if region_cpu is None:
+                    region_cpu = region.float().cpu()
+                row = region_cpu[i]
+                tok = max(allowed, key=lambda b: float(row[b]))
+            self._advance_request(req, tok)
+
+    def _drain_forced(self, req: Request) -> list[int]:
+        """Consume grammar-FORCED bytes (single-choice FSM states) without
+        spending decode steps; they return as a chunk-prefill run."""
+        run: list[int] = []
+        fsm = req.fsm
+        while fsm is not None and not fsm.done:
+            allowed = fsm.allowed_bytes()
+            if len(allowed) != 1:
+                break
+            b = allowed[0]
+            fsm.advance(b)
+            if b != NUMBER_CLOSE_SENTINEL:
+                req.out_ids.append(b)
+                run.append(b)
+        return run
 
     def _advance_request(self, req: Request, tok: int) -> None:
         finished = False
@@ -291,6 +368,10 @@ class LLMEngine:
                 req.fsm.advance(tok)
                 if tok != NUMBER_CLOSE_SENTINEL:
                     req.out_ids.append(tok)
+                    req.pending_input = [tok]
+                else:
+                    req.pending_input = []
+                req.pending_input += self._drain_forced(req)
                 if req.fsm.done:
                     finished = True
         else:
@@ -298,16 +379,19 @@ class LLMEngine:
                 finished = True
             else:
                 req.out_ids.append(tok)
+                req.pending_input = [tok]
         if len(req.out_ids) >= req.max_new_tokens:
+            finished = True
+        if not finished and not req.pending_input:
+            # nothing left to feed the model (pure forced tail / FSM drained):
+            # the request is complete — a running request with no pending
+            # input would starve the scheduler
             finished = True
         if finished:
             req.state = "done"
             req.finished_at = time.time()
             self.model.kv.free(req.rid)
             req.done_event.set()
-        else:
-            # the chosen token still needs its KV stored next decode step
-            pass
 
     # -- metrics ----------------------------------------------------------------------
 

@@ -229,6 +229,74 @@ class LlamaModel:
         h = ops.rmsnorm(h, self.final_norm_w, self.cfg.rms_eps)
         return self.lm_head(h)
 
+    def _layer_forward_chunk(self, layer: LlamaLayer, h: torch.Tensor,
+                             positions, seq_starts, block_tables, hist_lens,
+                             slots, layer_idx: int):
+        T = h.shape[0]
+        normed = ops.rmsnorm(h, layer.input_norm_w, self.cfg.rms_eps)
+        qkv = layer.qkv(normed)
+        q, k, v = layer._split_qkv(qkv, T)
+        q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, positions)
+        ops.store_kv(k, v, self.kv.k[layer_idx], self.kv.v[layer_idx], slots)
+        attn = ops.chunked_prefill_attention(q, self.kv.k[layer_idx],
+                                             self.kv.v[layer_idx], block_tables,
+                                             hist_lens, seq_starts, self.scale)
+        h = h + layer.o_proj(attn.reshape(T, -1))
+        normed = ops.rmsnorm(h, layer.post_norm_w, self.cfg.rms_eps)
+        gu = layer.gate_up(normed)
+        gate, up = gu.chunk(2, dim=-1)
+        h = h + layer.down(ops.silu_mul(gate.contiguous(), up.contiguous()))
+        return h
+
+    def chunk_step(self, token_ids: torch.Tensor, positions: torch.Tensor,
+                   seq_starts: torch.Tensor, block_tables: torch.Tensor,
+                   hist_lens: torch.Tensor, slot_mapping: torch.Tensor) -> torch.Tensor:
+        """Multi-token append for running sequences: each chunk attends over
+        its paged history + itself. Returns logits at each chunk's LAST
+        token: [B, vocab]. GPU path requires head_dim 128 (Llama); other
+        dims fall back to sequential decode steps."""
+        device = self.device
+        if (str(device).startswith("cuda") and self.cfg.head_dim != 128):
+            return self._chunk_by_decode(token_ids, positions, seq_starts,
+                                         block_tables, hist_lens, slot_mapping)
+        token_ids = token_ids.long().to(device)
+        positions = positions.to(torch.int32).to(device)
+        seq_starts_d = seq_starts.to(torch.int32).to(device)
+        bt = block_tables.to(device)
+        hist = hist_lens.to(torch.int32).to(device)
+        slots = slot_mapping.to(torch.int32).to(device)
+        h = self.embed.weight[token_ids]
+        for i, layer in enumerate(self.layers):
+            h = self._layer_forward_chunk(layer, h, positions, seq_starts_d, bt,
+                                          hist, slots, i)
+        last = (seq_starts[1:] - 1).long().to(device)
+        h_last = ops.rmsnorm(h[last], self.final_norm_w, self.cfg.rms_eps)
+        return self.lm_head(h_last)
+
+    def _chunk_by_decode(self, token_ids, positions, seq_starts, block_tables,
+                         hist_lens, slot_mapping):
+        """Fallback: feed chunk tokens one decode step at a time (non-128
+        head dims on GPU); only the last step's logits are returned."""
+        starts = seq_starts.tolist()
+        B = len(starts) - 1
+        max_len = max(starts[b + 1] - starts[b] for b in range(B))
+        logits = None
+        lens_cur = hist_lens.clone()
+        for j in range(max_len):
+            idx = []
+            for b in range(B):
+                n = starts[b + 1] - starts[b]
+                idx.append(starts[b] + min(j, n - 1))  # clamp: re-run last token
+            idx_t = torch.tensor(idx, dtype=torch.long)
+            lens_step = torch.minimum(hist_lens + (j + 1),
+                                      hist_lens + torch.tensor(
+                                          [starts[b + 1] - starts[b] for b in range(B)],
+                                          dtype=hist_lens.dtype))
+            logits = self.decode(token_ids[idx_t], positions[idx_t],
+                                 block_tables, lens_step.to(torch.int32),
+                                 slot_mapping[idx_t])
+        return logits
+
     def decode(self, token_ids: torch.Tensor, positions: torch.Tensor,
                block_tables: torch.Tensor, seq_lens: torch.Tensor,
                slot_mapping: torch.Tensor) -> torch.Tensor:

@@ -119,6 +119,22 @@ def _build_qtiles(seq_starts: torch.Tensor, qtile: int = 64):
             torch.tensor(tq, dtype=torch.int32))
 
 
+def chunked_prefill_attention(q, k_cache, v_cache, block_tables, hist_lens,
+                              seq_starts, scale: Optional[float] = None):
+    """Packed new-token chunks attending over paged history + themselves."""
+    scale = scale or 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        tb, tq = _build_qtiles(seq_starts.to(torch.int32))
+        return _get_ext().flash_prefill_paged(
+            q.contiguous(), k_cache, v_cache,
+            block_tables.to(torch.int32).contiguous(),
+            tb.to(q.device), tq.to(q.device),
+            seq_starts.to(torch.int32).contiguous().to(q.device),
+            hist_lens.to(torch.int32).contiguous().to(q.device), scale)
+    return reference.chunked_prefill_attention(q, k_cache, v_cache, block_tables,
+                                               hist_lens, seq_starts, scale)
+
+
 def paged_decode_attention(q, k_cache, v_cache, block_tables, seq_lens,
                            scale: Optional[float] = None):
     scale = scale or 1.0 / math.sqrt(q.shape[-1])

@@ -24,6 +24,9 @@ void launch_store_kv(const void*, const void*, void*, void*, const void*,
 void launch_flash_prefill(const void*, const void*, const void*, const void*,
                           const void*, const void*, void*, int, int, int, float,
                           int, void*);
+void launch_flash_prefill_paged(const void*, const void*, const void*, const void*,
+                                const void*, const void*, const void*, const void*,
+                                void*, int, int, int, int, int, float, void*);
 void launch_mfma_probe(const void*, const void*, void*, void*);
 void launch_cosine_scores(const void*, const void*, void*, long, int, void*);
 void launch_masked_argmax(const void*, const void*, void*, int, int, void*);
@@ -164,6 +167,31 @@ torch::Tensor flash_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
     return out;
 }
 
+torch::Tensor flash_prefill_paged(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                                  torch::Tensor bt, torch::Tensor tile_batch,
+                                  torch::Tensor tile_qstart, torch::Tensor seq_starts,
+                                  torch::Tensor hist_lens, double scale) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(kc, torch::kBFloat16);
+    CHECK_IN(vc, torch::kBFloat16);
+    CHECK_IN(bt, torch::kInt32);
+    CHECK_IN(tile_batch, torch::kInt32);
+    CHECK_IN(tile_qstart, torch::kInt32);
+    CHECK_IN(seq_starts, torch::kInt32);
+    CHECK_IN(hist_lens, torch::kInt32);
+    int Hq = (int)q.size(1), D = (int)q.size(2);
+    int Hk = (int)kc.size(1), BS = (int)kc.size(2);
+    TORCH_CHECK(D == 128, "paged flash prefill supports head dim 128");
+    auto out = torch::empty_like(q);
+    launch_flash_prefill_paged(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                               bt.data_ptr(), tile_batch.data_ptr(),
+                               tile_qstart.data_ptr(), seq_starts.data_ptr(),
+                               hist_lens.data_ptr(), out.data_ptr(),
+                               (int)tile_batch.size(0), Hq, Hk, BS,
+                               (int)bt.size(1), (float)scale, current_stream());
+    return out;
+}
+
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
     CHECK_IN(A, torch::kBFloat16);
     CHECK_IN(B, torch::kBFloat16);
@@ -221,6 +249,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("paged_decode", &paged_decode, "paged-KV decode attention");
     m.def("prefill_attn", &prefill_attn, "varlen prefill attention");
     m.def("flash_prefill", &flash_prefill, "MFMA flash prefill attention (D=128)");
+    m.def("flash_prefill_paged", &flash_prefill_paged,
+          "MFMA chunked prefill over the paged KV cache (D=128)");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
     m.def("store_kv", &store_kv, "scatter K/V into paged cache");
     m.def("cosine_scores", &cosine_scores, "brute-force cosine scores");

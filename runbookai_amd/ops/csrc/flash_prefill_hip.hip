@@ -223,6 +223,187 @@ __global__ void flash_prefill_kernel(
     }
 }
 
+// ------------------------------------------------- paged chunked prefill
+// Same structure, but K/V come from the PAGED CACHE via block tables, so a
+// chunk of new tokens can attend over its sequence's full history — this
+// powers forced-byte injection (grammar-forced JSON structure enters as
+// 50k-tok/s chunks instead of one decode step per byte) and chunked
+// prefill generally. hist_lens[b] = tokens already in the cache BEFORE
+// this chunk; new K/V must be store_kv'd before calling.
+__launch_bounds__(256, 2)
+__global__ void flash_prefill_paged_kernel(
+    const ushort_t* __restrict__ q, const ushort_t* __restrict__ k_cache,
+    const ushort_t* __restrict__ v_cache, const int* __restrict__ block_tables,
+    const int* __restrict__ tile_batch, const int* __restrict__ tile_qstart,
+    const int* __restrict__ seq_starts, const int* __restrict__ hist_lens,
+    ushort_t* __restrict__ out, int Hq, int Hk, int BS, int max_blocks,
+    float scale) {
+    const int tile = blockIdx.x;
+    const int h = blockIdx.y;
+    const int hk = h / (Hq / Hk);
+    const int b = tile_batch[tile];
+    const int q0g = tile_qstart[tile];
+    const int seg_start = seq_starts[b];
+    const int seg_end = seq_starts[b + 1];
+    const int hist = hist_lens[b];
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    const int* bt = block_tables + (long)b * max_blocks;
+
+    __shared__ ushort_t k_lds[KTILE][DHEAD];
+    __shared__ ushort_t v_lds[KTILE][DHEAD];
+    __shared__ ushort_t p_lds[4][16][KTILE];
+
+    const int my_qrow = q0g + wid * 16 + (lane & 15);
+    const bool row_valid = my_qrow < seg_end;
+    bf16x8_t qfrag[4];
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+        if (row_valid) {
+            const long base = ((long)my_qrow * Hq + h) * DHEAD + s * 32 + (lane >> 4) * 8;
+            qfrag[s] = *reinterpret_cast<const bf16x8_t*>(q + base);
+        } else {
+            qfrag[s] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+    }
+
+    float m_run[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+    float l_run[4] = {0.f, 0.f, 0.f, 0.f};
+    f32x4_t o_acc[8];
+#pragma unroll
+    for (int d = 0; d < 8; ++d) o_acc[d] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+
+    // full-sequence kv extent for this tile (causal): history + local q hi
+    const int q_hi_local = min(q0g + QTILE, seg_end) - 1 - seg_start;
+    const int kv_end = hist + q_hi_local + 1;     // exclusive, full-seq index
+    const int n_tiles = (kv_end + KTILE - 1) / KTILE;
+
+    for (int kt = 0; kt < n_tiles; ++kt) {
+        const int kv0 = kt * KTILE;               // full-seq token of col 0
+        __syncthreads();
+        for (int idx = threadIdx.x; idx < KTILE * DHEAD / 8; idx += 256) {
+            const int row = (idx * 8) / DHEAD;
+            const int col = (idx * 8) % DHEAD;
+            const int tok = kv0 + row;
+            bf16x8_t kv8 = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+            bf16x8_t vv8 = kv8;
+            if (tok < kv_end) {
+                const int blk = bt[tok / BS];
+                const long base = (((long)blk * Hk + hk) * BS + (tok % BS)) * DHEAD + col;
+                kv8 = *reinterpret_cast<const bf16x8_t*>(k_cache + base);
+                vv8 = *reinterpret_cast<const bf16x8_t*>(v_cache + base);
+            }
+            *reinterpret_cast<bf16x8_t*>(&k_lds[row][col]) = kv8;
+            *reinterpret_cast<bf16x8_t*>(&v_lds[row][col]) = vv8;
+        }
+        __syncthreads();
+
+        f32x4_t sc[2];
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+            sc[kc] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int s = 0; s < 4; ++s) {
+                const int kcol = kc * 16 + (lane & 15);
+                const int dbase = s * 32 + (lane >> 4) * 8;
+                bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(&k_lds[kcol][dbase]);
+                sc[kc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[s], bfrag, sc[kc], 0, 0, 0);
+            }
+        }
+
+        float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int qrow = q0g + wid * 16 + (lane >> 4) * 4 + r;
+                const int q_full = hist + (qrow - seg_start);
+                const int ktok = kv0 + kc * 16 + (lane & 15);
+                float sv = sc[kc][r] * scale;
+                const bool masked = (qrow >= seg_end) || (ktok > q_full);
+                sv = masked ? -1e30f : sv;
+                sc[kc][r] = sv;
+                tile_max[r] = fmaxf(tile_max[r], sv);
+            }
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                tile_max[r] = fmaxf(tile_max[r], __shfl_xor(tile_max[r], off, WAVE));
+        }
+        float alpha[4], row_sum[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            const float m_new = fmaxf(m_run[r], tile_max[r]);
+            alpha[r] = __expf(m_run[r] - m_new);
+            m_run[r] = m_new;
+            row_sum[r] = 0.f;
+        }
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const float p = (sc[kc][r] <= -1e29f)
+                                    ? 0.f
+                                    : __expf(sc[kc][r] - m_run[r]);
+                row_sum[r] += p;
+                p_lds[wid][(lane >> 4) * 4 + r][kc * 16 + (lane & 15)] = f2bf(p);
+            }
+        }
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+                row_sum[r] += __shfl_xor(row_sum[r], off, WAVE);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            l_run[r] = l_run[r] * alpha[r] + row_sum[r];
+#pragma unroll
+            for (int d = 0; d < 8; ++d) o_acc[d][r] *= alpha[r];
+        }
+        __syncthreads();
+
+        bf16x8_t pfrag = *reinterpret_cast<const bf16x8_t*>(
+            &p_lds[wid][lane & 15][(lane >> 4) * 8]);
+#pragma unroll
+        for (int d = 0; d < 8; ++d) {
+            bf16x8_t vfrag;
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                vfrag[j] = (short)v_lds[(lane >> 4) * 8 + j][d * 16 + (lane & 15)];
+            o_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag, o_acc[d], 0, 0, 0);
+        }
+    }
+
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+        const int qrow = q0g + wid * 16 + (lane >> 4) * 4 + r;
+        if (qrow >= seg_end) continue;
+        const float inv = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+#pragma unroll
+        for (int d = 0; d < 8; ++d) {
+            out[((long)qrow * Hq + h) * DHEAD + d * 16 + (lane & 15)] =
+                f2bf(o_acc[d][r] * inv);
+        }
+    }
+}
+
+extern "C" void launch_flash_prefill_paged(
+    const void* q, const void* kc, const void* vc, const void* bt,
+    const void* tile_batch, const void* tile_qstart, const void* seq_starts,
+    const void* hist_lens, void* out, int n_tiles, int Hq, int Hk, int BS,
+    int max_blocks, float scale, hipStream_t stream) {
+    dim3 grid(n_tiles, Hq), block(256);
+    hipLaunchKernelGGL(flash_prefill_paged_kernel, grid, block, 0, stream,
+                       (const ushort_t*)q, (const ushort_t*)kc, (const ushort_t*)vc,
+                       (const int*)bt, (const int*)tile_batch,
+                       (const int*)tile_qstart, (const int*)seq_starts,
+                       (const int*)hist_lens, (ushort_t*)out, Hq, Hk, BS,
+                       max_blocks, scale);
+}
+
 extern "C" void launch_flash_prefill(const void* q, const void* k, const void* v,
                                      const void* tile_batch, const void* tile_qstart,
                                      const void* seq_starts, void* out,

@@ -125,6 +125,45 @@ def paged_decode_attention(
     return out
 
 
+def chunked_prefill_attention(
+    q: torch.Tensor,              # [Tnew, Hq, D] packed new tokens
+    k_cache: torch.Tensor,        # [NB, Hk, BS, D]
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,   # [B, MB]
+    hist_lens: torch.Tensor,      # [B] tokens in cache BEFORE the chunk
+    seq_starts: torch.Tensor,     # [B+1] packed starts of the new chunks
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """New tokens attend causally over history + themselves (K/V already in
+    the cache). Numerics reference for flash_prefill_paged."""
+    Tnew, Hq, D = q.shape
+    Hk = k_cache.shape[1]
+    bs = k_cache.shape[2]
+    group = Hq // Hk
+    scale = scale or 1.0 / math.sqrt(D)
+    out = torch.empty_like(q)
+    for b in range(seq_starts.numel() - 1):
+        s, e = int(seq_starts[b]), int(seq_starts[b + 1])
+        n_new = e - s
+        hist = int(hist_lens[b])
+        L = hist + n_new
+        nblocks = (L + bs - 1) // bs
+        blocks = block_tables[b, :nblocks].long()
+        k = k_cache[blocks].float().permute(1, 0, 2, 3).reshape(Hk, nblocks * bs, D)[:, :L]
+        v = v_cache[blocks].float().permute(1, 0, 2, 3).reshape(Hk, nblocks * bs, D)[:, :L]
+        k_g = k.repeat_interleave(group, dim=0)
+        v_g = v.repeat_interleave(group, dim=0)
+        qs = q[s:e].float()  # [n_new, Hq, D]
+        scores = torch.einsum("qhd,hld->hql", qs, k_g) * scale
+        kv_pos = torch.arange(L, device=q.device)
+        q_pos = hist + torch.arange(n_new, device=q.device)
+        mask = kv_pos.unsqueeze(0) > q_pos.unsqueeze(1)  # [n_new, L]
+        scores.masked_fill_(mask.unsqueeze(0), float("-inf"))
+        probs = scores.softmax(-1)
+        out[s:e] = torch.einsum("hql,hld->qhd", probs, v_g).to(q.dtype)
+    return out
+
+
 def store_kv(
     k: torch.Tensor,              # [T, Hk, D] new keys
     v: torch.Tensor,
