@@ -168,39 +168,27 @@ class LlamaModel:
 
     # -- forward -----------------------------------------------------------------
 
-    def _layer_forward_prefill(self, layer: LlamaLayer, h: torch.Tensor,
-                               positions, seq_starts, batch_idx, slots, layer_idx: int):
+    def _transformer_body(self, h: torch.Tensor, positions, slots, attn_fn) -> torch.Tensor:
+        """Shared fused body: every residual add is fused into the following
+        RMSNorm (rmsnorm_residual), SwiGLU reads the fused [gate|up] rows.
+        attn_fn(q, layer_idx) -> attention output [T, heads_per_rank, D].
+        Returns the FINAL-normed hidden states [T, H]."""
         T = h.shape[0]
-        normed = ops.rmsnorm(h, layer.input_norm_w, self.cfg.rms_eps)
-        qkv = layer.qkv(normed)
-        q, k, v = layer._split_qkv(qkv, T)
-        q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, positions)
-        ops.store_kv(k, v, self.kv.k[layer_idx], self.kv.v[layer_idx], slots)
-        attn = ops.prefill_attention(q, k, v, seq_starts, causal=True, scale=self.scale,
-                                     batch_idx=batch_idx)
-        h = h + layer.o_proj(attn.reshape(T, -1))
-        normed = ops.rmsnorm(h, layer.post_norm_w, self.cfg.rms_eps)
-        gu = layer.gate_up(normed)
-        gate, up = gu.chunk(2, dim=-1)
-        h = h + layer.down(ops.silu_mul(gate.contiguous(), up.contiguous()))
-        return h
-
-    def _layer_forward_decode(self, layer: LlamaLayer, h: torch.Tensor,
-                              positions, block_tables, seq_lens, slots, layer_idx: int):
-        B = h.shape[0]
-        normed = ops.rmsnorm(h, layer.input_norm_w, self.cfg.rms_eps)
-        qkv = layer.qkv(normed)
-        q, k, v = layer._split_qkv(qkv, B)
-        q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, positions)
-        ops.store_kv(k, v, self.kv.k[layer_idx], self.kv.v[layer_idx], slots)
-        attn = ops.paged_decode_attention(q, self.kv.k[layer_idx], self.kv.v[layer_idx],
-                                          block_tables, seq_lens, self.scale)
-        h = h + layer.o_proj(attn.reshape(B, -1))
-        normed = ops.rmsnorm(h, layer.post_norm_w, self.cfg.rms_eps)
-        gu = layer.gate_up(normed)
-        gate, up = gu.chunk(2, dim=-1)
-        h = h + layer.down(ops.silu_mul(gate.contiguous(), up.contiguous()))
-        return h
+        normed = ops.rmsnorm(h, self.layers[0].input_norm_w, self.cfg.rms_eps)
+        for i, layer in enumerate(self.layers):
+            qkv = layer.qkv(normed)
+            q, k, v = layer._split_qkv(qkv, T)
+            q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, positions)
+            ops.store_kv(k, v, self.kv.k[i], self.kv.v[i], slots)
+            attn = attn_fn(q, k, v, i)
+            attn_out = layer.o_proj(attn.reshape(T, -1))
+            normed, h = ops.rmsnorm_residual(attn_out, h, layer.post_norm_w,
+                                             self.cfg.rms_eps)
+            mlp_out = layer.down(ops.silu_mul_fused(layer.gate_up(normed)))
+            next_w = (self.layers[i + 1].input_norm_w if i + 1 < len(self.layers)
+                      else self.final_norm_w)
+            normed, h = ops.rmsnorm_residual(mlp_out, h, next_w, self.cfg.rms_eps)
+        return normed  # final-normed states
 
     def prefill(self, token_ids: torch.Tensor, positions: torch.Tensor,
                 seq_starts: torch.Tensor, slot_mapping: torch.Tensor) -> torch.Tensor:
@@ -213,40 +201,25 @@ class LlamaModel:
         slots = slot_mapping.to(torch.int32).to(device)
         batch_idx = self._batch_idx(seq_starts, token_ids.shape[0]).to(device)
         h = self.embed.weight[token_ids.long()]
-        for i, layer in enumerate(self.layers):
-            h = self._layer_forward_prefill(layer, h, positions, seq_starts_d,
-                                            batch_idx, slots, i)
+
+        def attn_fn(q, k, v, i):
+            return ops.prefill_attention(q, k, v, seq_starts_d, causal=True,
+                                         scale=self.scale, batch_idx=batch_idx)
+
+        normed = self._transformer_body(h, positions, slots, attn_fn)
         last = (seq_starts[1:] - 1).long().to(device)
-        h_last = ops.rmsnorm(h[last], self.final_norm_w, self.cfg.rms_eps)
-        return self.lm_head(h_last)
+        return self.lm_head(normed[last])
 
     def _decode_impl(self, token_ids, positions, block_tables, seq_lens, slots):
         """Device-tensor decode body (hipGraph-capturable)."""
         h = self.embed.weight[token_ids]
-        for i, layer in enumerate(self.layers):
-            h = self._layer_forward_decode(layer, h, positions, block_tables,
-                                           seq_lens, slots, i)
-        h = ops.rmsnorm(h, self.final_norm_w, self.cfg.rms_eps)
-        return self.lm_head(h)
 
-    def _layer_forward_chunk(self, layer: LlamaLayer, h: torch.Tensor,
-                             positions, seq_starts, block_tables, hist_lens,
-                             slots, layer_idx: int):
-        T = h.shape[0]
-        normed = ops.rmsnorm(h, layer.input_norm_w, self.cfg.rms_eps)
-        qkv = layer.qkv(normed)
-        q, k, v = layer._split_qkv(qkv, T)
-        q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, positions)
-        ops.store_kv(k, v, self.kv.k[layer_idx], self.kv.v[layer_idx], slots)
-        attn = ops.chunked_prefill_attention(q, self.kv.k[layer_idx],
-                                             self.kv.v[layer_idx], block_tables,
-                                             hist_lens, seq_starts, self.scale)
-        h = h + layer.o_proj(attn.reshape(T, -1))
-        normed = ops.rmsnorm(h, layer.post_norm_w, self.cfg.rms_eps)
-        gu = layer.gate_up(normed)
-        gate, up = gu.chunk(2, dim=-1)
-        h = h + layer.down(ops.silu_mul(gate.contiguous(), up.contiguous()))
-        return h
+        def attn_fn(q, k, v, i):
+            return ops.paged_decode_attention(q, self.kv.k[i], self.kv.v[i],
+                                              block_tables, seq_lens, self.scale)
+
+        normed = self._transformer_body(h, positions, slots, attn_fn)
+        return self.lm_head(normed)
 
     def chunk_step(self, token_ids: torch.Tensor, positions: torch.Tensor,
                    seq_starts: torch.Tensor, block_tables: torch.Tensor,
@@ -266,12 +239,14 @@ class LlamaModel:
         hist = hist_lens.to(torch.int32).to(device)
         slots = slot_mapping.to(torch.int32).to(device)
         h = self.embed.weight[token_ids]
-        for i, layer in enumerate(self.layers):
-            h = self._layer_forward_chunk(layer, h, positions, seq_starts_d, bt,
-                                          hist, slots, i)
+
+        def attn_fn(q, k, v, i):
+            return ops.chunked_prefill_attention(q, self.kv.k[i], self.kv.v[i],
+                                                 bt, hist, seq_starts_d, self.scale)
+
+        normed = self._transformer_body(h, positions, slots, attn_fn)
         last = (seq_starts[1:] - 1).long().to(device)
-        h_last = ops.rmsnorm(h[last], self.final_norm_w, self.cfg.rms_eps)
-        return self.lm_head(h_last)
+        return self.lm_head(normed[last])
 
     def _chunk_by_decode(self, token_ids, positions, seq_starts, block_tables,
                          hist_lens, slot_mapping):

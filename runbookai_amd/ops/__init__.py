@@ -69,6 +69,15 @@ def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return reference.silu_mul(gate, up)
 
 
+def silu_mul_fused(gu: torch.Tensor) -> torch.Tensor:
+    """gu: [T, 2I] rows laid out as [gate | up]. Returns silu(gate)*up [T, I]
+    without materializing the strided halves."""
+    if _on_gpu(gu):
+        return _get_ext().silu_mul_fused(gu.contiguous())
+    gate, up = gu.chunk(2, dim=-1)
+    return reference.silu_mul(gate.contiguous(), up.contiguous())
+
+
 def apply_rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
                positions: torch.Tensor):
     if _on_gpu(q):

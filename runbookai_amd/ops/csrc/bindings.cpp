@@ -10,6 +10,7 @@ void launch_rmsnorm(const void*, const void*, void*, int, int, float, void*);
 void launch_rmsnorm_residual(const void*, const void*, void*, const void*, void*,
                              int, int, float, void*);
 void launch_silu_mul(const void*, const void*, void*, long, void*);
+void launch_silu_mul_fused(const void*, void*, long, int, void*);
 void launch_rope(void*, void*, const void*, const void*, const void*,
                  int, int, int, int, void*);
 void launch_paged_decode(const void*, const void*, const void*, const void*,
@@ -79,6 +80,17 @@ torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
     auto out = torch::empty_like(gate);
     launch_silu_mul(gate.data_ptr(), up.data_ptr(), out.data_ptr(), gate.numel(),
                     current_stream());
+    return out;
+}
+
+torch::Tensor silu_mul_fused(torch::Tensor gu) {
+    CHECK_IN(gu, torch::kBFloat16);
+    long T = gu.size(0);
+    int I2 = (int)gu.size(1);
+    TORCH_CHECK(I2 % 16 == 0, "fused gate_up width must be divisible by 16");
+    int I = I2 / 2;
+    auto out = torch::empty({T, (long)I}, gu.options());
+    launch_silu_mul_fused(gu.data_ptr(), out.data_ptr(), T, I, current_stream());
     return out;
 }
 
@@ -245,6 +257,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("rmsnorm", &rmsnorm, "RMSNorm bf16 (gfx950)");
     m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual-add + RMSNorm");
     m.def("silu_mul", &silu_mul, "SwiGLU activation");
+    m.def("silu_mul_fused", &silu_mul_fused, "SwiGLU from fused [gate|up] rows");
     m.def("rope_inplace", &rope_inplace, "RoPE in place on q,k");
     m.def("paged_decode", &paged_decode, "paged-KV decode attention");
     m.def("prefill_attn", &prefill_attn, "varlen prefill attention");

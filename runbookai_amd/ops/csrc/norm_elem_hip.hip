@@ -126,6 +126,36 @@ extern "C" void launch_silu_mul(const void* gate, const void* up, void* out,
                        (const ushort_t*)gate, (const ushort_t*)up, (ushort_t*)out, n8);
 }
 
+// gu: [T, 2*I] (gate | up concatenated per row) -> out [T, I].
+// Avoids the two .contiguous() copies a strided chunk() would cost.
+__global__ void silu_mul_fused_kernel(const ushort_t* __restrict__ gu,
+                                      ushort_t* __restrict__ out, long T, int I) {
+    const int i8 = I / 8;
+    for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < T * i8;
+         idx += (long)gridDim.x * blockDim.x) {
+        const long t = idx / i8;
+        const int c = (int)(idx % i8) * 8;
+        ushort8_t g = *reinterpret_cast<const ushort8_t*>(gu + t * 2 * I + c);
+        ushort8_t u = *reinterpret_cast<const ushort8_t*>(gu + t * 2 * I + I + c);
+        ushort8_t o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float gf = bf2f(g[j]);
+            float s = gf / (1.f + __expf(-gf));
+            o[j] = f2bf(s * bf2f(u[j]));
+        }
+        *reinterpret_cast<ushort8_t*>(out + t * I + c) = o;
+    }
+}
+
+extern "C" void launch_silu_mul_fused(const void* gu, void* out, long T, int I,
+                                      hipStream_t stream) {
+    long work = T * (I / 8);
+    int blocks = (int)min((work + 255) / 256, (long)2048);
+    hipLaunchKernelGGL(silu_mul_fused_kernel, dim3(blocks), dim3(256), 0, stream,
+                       (const ushort_t*)gu, (ushort_t*)out, T, I);
+}
+
 // ---------------------------------------------------------------- rope
 // q: [T, Hq, D], k: [T, Hk, D] bf16 (modified in place); cos/sin: [S, D/2]
 // f32; positions: [T] int32. Pair-interleaved rotation: elements (2i,2i+1)

@@ -50,6 +50,13 @@ class TestSiluMul:
         expected = ref.silu_mul(g.cpu(), u.cpu())
         assert (out.cpu().float() - expected.float()).abs().max().item() < 2e-2
 
+    def test_fused_rows(self):
+        gu = bf(torch.randn(64, 2 * 14336)).to(DEV)
+        out = ops.silu_mul_fused(gu)
+        g, u = gu.cpu().chunk(2, dim=-1)
+        expected = ref.silu_mul(g.contiguous(), u.contiguous())
+        assert (out.cpu().float() - expected.float()).abs().max().item() < 2e-2
+
 
 class TestRope:
     @pytest.mark.parametrize("D", [64, 128])
