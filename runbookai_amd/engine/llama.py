@@ -65,13 +65,15 @@ class LlamaLayer:
         self.kv_heads_per_rank = max(1, cfg.num_kv_heads // tp)
         self.input_norm_w = torch.ones(H, dtype=dtype, device=device)
         self.post_norm_w = torch.ones(H, dtype=dtype, device=device)
-        # fused QKV projection (column-parallel over heads)
+        # Parallel layers are built FULL-SIZE with a shared seed (so every
+        # rank generates identical weights); LlamaModel._shard_and_move then
+        # keeps this rank's shard. tp=1 here avoids double-sharding.
         qkv_out = (cfg.num_heads + 2 * cfg.num_kv_heads) * cfg.head_dim
-        self.qkv = ColumnParallelLinear(H, qkv_out, tp, dtype, device, gen)
-        self.o_proj = RowParallelLinear(cfg.q_size, H, tp, dtype, device, gen)
+        self.qkv = ColumnParallelLinear(H, qkv_out, 1, dtype, device, gen)
+        self.o_proj = RowParallelLinear(cfg.q_size, H, 1, dtype, device, gen)
         # fused gate+up (column-parallel)
-        self.gate_up = ColumnParallelLinear(H, 2 * cfg.intermediate_size, tp, dtype, device, gen)
-        self.down = RowParallelLinear(cfg.intermediate_size, H, tp, dtype, device, gen)
+        self.gate_up = ColumnParallelLinear(H, 2 * cfg.intermediate_size, 1, dtype, device, gen)
+        self.down = RowParallelLinear(cfg.intermediate_size, H, 1, dtype, device, gen)
 
     def _split_qkv(self, qkv: torch.Tensor, T: int):
         cfg = self.cfg

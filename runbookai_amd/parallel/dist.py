@@ -46,9 +46,17 @@ def get_world_size() -> int:
 
 
 def all_reduce(t: torch.Tensor) -> torch.Tensor:
-    """Sum all-reduce across the TP group (in place; returns t)."""
+    """Sum all-reduce across the TP group (in place; returns t).
+
+    gloo (CPU test backend) lacks bf16 — round-trip through fp32 there;
+    RCCL reduces bf16 natively."""
     if dist.is_initialized() and dist.get_world_size() > 1:
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        if dist.get_backend() == "gloo" and t.dtype == torch.bfloat16:
+            f = t.float()
+            dist.all_reduce(f, op=dist.ReduceOp.SUM)
+            t.copy_(f.to(t.dtype))
+        else:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
     return t
 
 
