@@ -57,23 +57,45 @@ CONFIGS: dict[str, LlamaConfig] = {
 
 
 class LlamaLayer:
-    def __init__(self, cfg: LlamaConfig, tp: int, dtype, device, gen) -> None:
+    def __init__(self, cfg: LlamaConfig, tp: int, rank: int, dtype, device,
+                 gen_factory) -> None:
+        """Builds this rank's TP SHARD directly: each full tensor is
+        generated on-device with a per-tensor seed (identical across ranks),
+        sliced, and the full copy freed — no host-side fp32 staging, so 70B
+        initializes in seconds within one GPU's memory headroom."""
         self.cfg = cfg
         self.tp = tp
         H = cfg.hidden_size
-        self.heads_per_rank = cfg.num_heads // tp
-        self.kv_heads_per_rank = max(1, cfg.num_kv_heads // tp)
+        d = cfg.head_dim
+        hq, hk = cfg.num_heads, cfg.num_kv_heads
+        self.heads_per_rank = hq // tp
+        self.kv_heads_per_rank = max(1, hk // tp)
+        hq_r, hk_r = self.heads_per_rank, self.kv_heads_per_rank
         self.input_norm_w = torch.ones(H, dtype=dtype, device=device)
         self.post_norm_w = torch.ones(H, dtype=dtype, device=device)
-        # Parallel layers are built FULL-SIZE with a shared seed (so every
-        # rank generates identical weights); LlamaModel._shard_and_move then
-        # keeps this rank's shard. tp=1 here avoids double-sharding.
-        qkv_out = (cfg.num_heads + 2 * cfg.num_kv_heads) * cfg.head_dim
-        self.qkv = ColumnParallelLinear(H, qkv_out, 1, dtype, device, gen)
-        self.o_proj = RowParallelLinear(cfg.q_size, H, 1, dtype, device, gen)
-        # fused gate+up (column-parallel)
-        self.gate_up = ColumnParallelLinear(H, 2 * cfg.intermediate_size, 1, dtype, device, gen)
-        self.down = RowParallelLinear(cfg.intermediate_size, H, 1, dtype, device, gen)
+
+        qkv_out = (hq + 2 * hk) * d
+        self.qkv = ColumnParallelLinear(H, qkv_out, 1, dtype, device, gen_factory())
+        w = self.qkv.weight
+        q_w = w[: hq * d].view(hq, d, -1)[rank * hq_r:(rank + 1) * hq_r].reshape(hq_r * d, -1)
+        k_w = w[hq * d:(hq + hk) * d].view(hk, d, -1)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, -1)
+        v_w = w[(hq + hk) * d:].view(hk, d, -1)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, -1)
+        self.qkv.weight = torch.cat([q_w, k_w, v_w], 0).contiguous()
+
+        self.o_proj = RowParallelLinear(cfg.q_size, H, 1, dtype, device, gen_factory())
+        ow = self.o_proj.weight.view(-1, hq, d)[:, rank * hq_r:(rank + 1) * hq_r]
+        self.o_proj.weight = ow.reshape(H, hq_r * d).contiguous()
+
+        inter = cfg.intermediate_size
+        ipr = inter // tp
+        self.gate_up = ColumnParallelLinear(H, 2 * inter, 1, dtype, device, gen_factory())
+        gw = self.gate_up.weight
+        self.gate_up.weight = torch.cat(
+            [gw[:inter][rank * ipr:(rank + 1) * ipr],
+             gw[inter:][rank * ipr:(rank + 1) * ipr]], 0).contiguous()
+
+        self.down = RowParallelLinear(inter, H, 1, dtype, device, gen_factory())
+        self.down.weight = self.down.weight[:, rank * ipr:(rank + 1) * ipr].contiguous()
 
     def _split_qkv(self, qkv: torch.Tensor, T: int):
         cfg = self.cfg
@@ -98,16 +120,28 @@ class LlamaModel:
         self.dtype = dtype
         self.tp = tp or get_world_size()
         assert cfg.num_heads % self.tp == 0
-        gen = torch.Generator(device="cpu")
-        gen.manual_seed(seed)
-        # weights are generated on CPU for determinism across TP ranks, then moved
-        cpu = "cpu"
-        self.embed = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size, dtype, cpu, gen)
+        from ..parallel.dist import get_rank
+
+        rank = get_rank() % self.tp
+        # per-tensor seeded generators ON DEVICE: identical full tensors on
+        # every rank (before sharding), no host fp32 staging
+        gen_dev = device if str(device).startswith("cuda") else "cpu"
+        self._gen_counter = 0
+
+        def gen_factory() -> torch.Generator:
+            self._gen_counter += 1
+            g = torch.Generator(device=gen_dev)
+            g.manual_seed(seed * 100_003 + self._gen_counter)
+            return g
+
+        self.embed = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size, dtype, device,
+                                      gen_factory())
         # (embedding lookup uses embed.weight as the table)
-        self.layers = [LlamaLayer(cfg, self.tp, dtype, cpu, gen) for _ in range(cfg.num_layers)]
-        self.final_norm_w = torch.ones(cfg.hidden_size, dtype=dtype)
-        self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size, dtype, cpu, gen)
-        self._shard_and_move(device)
+        self.layers = [LlamaLayer(cfg, self.tp, rank, dtype, device, gen_factory)
+                       for _ in range(cfg.num_layers)]
+        self.final_norm_w = torch.ones(cfg.hidden_size, dtype=dtype, device=device)
+        self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size, dtype, device,
+                                        gen_factory())
         cos, sin = self._rope_tables()
         self.rope_cos = cos.to(device)
         self.rope_sin = sin.to(device)
@@ -128,45 +162,6 @@ class LlamaModel:
         from ..ops.reference import rope_cos_sin
 
         return rope_cos_sin(self.cfg.max_seq_len, self.cfg.head_dim, self.cfg.rope_theta)
-
-    def _shard_and_move(self, device: str) -> None:
-        """Take this rank's TP shard of every parallel weight and move to device.
-
-        Weights were initialized FULL-SIZE on CPU with a shared seed, so all
-        ranks agree; each rank keeps rows/cols for its shard — numerics match
-        TP=1 exactly (modulo all-reduce summation order).
-        """
-        from ..parallel.dist import get_rank
-
-        rank = get_rank() % self.tp
-        cfg = self.cfg
-        d = cfg.head_dim
-        for layer in self.layers:
-            hq, hk = cfg.num_heads, cfg.num_kv_heads
-            hq_r, hk_r = layer.heads_per_rank, layer.kv_heads_per_rank
-            w = layer.qkv.weight  # [(hq+2hk)*d, H] full
-            q_w = w[: hq * d].view(hq, d, -1)[rank * hq_r:(rank + 1) * hq_r].reshape(hq_r * d, -1)
-            k_w = w[hq * d:(hq + hk) * d].view(hk, d, -1)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, -1)
-            v_w = w[(hq + hk) * d:].view(hk, d, -1)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, -1)
-            layer.qkv.weight = torch.cat([q_w, k_w, v_w], 0).to(device).contiguous()
-            # o_proj: input dim sharded by head
-            ow = layer.o_proj.weight  # [H, hq*d]
-            ow = ow.view(-1, hq, d)[:, rank * hq_r:(rank + 1) * hq_r].reshape(ow.shape[0], hq_r * d)
-            layer.o_proj.weight = ow.to(device).contiguous()
-            # gate_up: [2*I, H] — shard each half separately
-            gw = layer.gate_up.weight
-            inter = cfg.intermediate_size
-            ipr = inter // self.tp
-            gate = gw[:inter][rank * ipr:(rank + 1) * ipr]
-            up = gw[inter:][rank * ipr:(rank + 1) * ipr]
-            layer.gate_up.weight = torch.cat([gate, up], 0).to(device).contiguous()
-            dw = layer.down.weight  # [H, I]
-            layer.down.weight = dw[:, rank * ipr:(rank + 1) * ipr].to(device).contiguous()
-            layer.input_norm_w = layer.input_norm_w.to(device)
-            layer.post_norm_w = layer.post_norm_w.to(device)
-        self.embed.weight = self.embed.weight.to(device)
-        self.lm_head.weight = self.lm_head.weight.to(device)
-        self.final_norm_w = self.final_norm_w.to(device)
 
     # -- forward -----------------------------------------------------------------
 

@@ -22,7 +22,10 @@ from .dist import all_reduce, get_rank, get_world_size
 
 def _init_weight(out_f: int, in_f: int, dtype, device, gen: Optional[torch.Generator],
                  std: float = 0.02) -> torch.Tensor:
-    w = torch.empty(out_f, in_f, dtype=torch.float32, device=device)
+    """Random init on the generator's device (caller controls placement and
+    per-tensor seeding for rank determinism)."""
+    target = gen.device if gen is not None else device
+    w = torch.empty(out_f, in_f, dtype=torch.float32, device=target)
     w.normal_(0.0, std, generator=gen)
     return w.to(dtype)
 
