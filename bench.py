@@ -67,7 +67,10 @@ def main() -> None:
     cases = fixtures["cases"]
 
     kv_blocks = 2048 if model_name != "tiny" else 512
-    engine = LLMEngine(model=model_name, device=device, tp=None,
+    # weak-scaling benchmark = DATA parallel replicas (tp=1 per rank);
+    # --tp > 1 shards one model across all ranks instead (70B config)
+    engine = LLMEngine(model=model_name, device=device,
+                       tp=(args.tp if args.tp > 1 else 1),
                        kv_blocks=kv_blocks, background=True)
     # one shared scenario for concurrent runs (scenario registry is global)
     set_scenario(SimScenario.redis_exhaustion())

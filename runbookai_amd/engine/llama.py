@@ -85,6 +85,7 @@ class LlamaLayer:
         self.o_proj = RowParallelLinear(cfg.q_size, H, 1, dtype, device, gen_factory())
         ow = self.o_proj.weight.view(-1, hq, d)[:, rank * hq_r:(rank + 1) * hq_r]
         self.o_proj.weight = ow.reshape(H, hq_r * d).contiguous()
+        self.o_proj.tp = tp  # reduce across the TP group only when tp > 1
 
         inter = cfg.intermediate_size
         ipr = inter // tp
@@ -96,6 +97,7 @@ class LlamaLayer:
 
         self.down = RowParallelLinear(inter, H, 1, dtype, device, gen_factory())
         self.down.weight = self.down.weight[:, rank * ipr:(rank + 1) * ipr].contiguous()
+        self.down.tp = tp
 
     def _split_qkv(self, qkv: torch.Tensor, T: int):
         cfg = self.cfg
@@ -120,6 +122,8 @@ class LlamaModel:
         self.dtype = dtype
         self.tp = tp or get_world_size()
         assert cfg.num_heads % self.tp == 0
+        assert self.tp in (1, get_world_size()), \
+            "tp must equal world size (TP) or 1 (DP replicas); mixed DPxTP is not wired yet"
         from ..parallel.dist import get_rank
 
         rank = get_rank() % self.tp

@@ -45,7 +45,11 @@ class ColumnParallelLinear:
 
 
 class RowParallelLinear:
-    """y = all_reduce(x_shard @ W_shard^T); W sharded on the input dimension."""
+    """y = all_reduce(x_shard @ W_shard^T); W sharded on the input dimension.
+
+    The reduction fires only when `tp > 1` — under pure data parallelism
+    (tp=1, world>1) each replica's forward is independent. Set .tp after
+    external sharding (engine/llama.py) to enable the reduce."""
 
     def __init__(self, in_features: int, out_features: int, tp: Optional[int] = None,
                  dtype=torch.bfloat16, device="cpu", gen: Optional[torch.Generator] = None):
@@ -56,7 +60,7 @@ class RowParallelLinear:
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         y = x @ self.weight.t()
-        return all_reduce(y)
+        return all_reduce(y) if self.tp > 1 else y
 
 
 class ReplicatedLinear:
