@@ -271,8 +271,14 @@ def status(ctx: click.Context) -> None:
 @cli.command()
 @click.option("--template", type=click.Choice(["ecs-rds", "serverless", "enterprise"]),
               default="ecs-rds")
-def init(template: str) -> None:
-    """Initialize .runbook/ with a config template."""
+@click.option("--interactive", is_flag=True, help="Run the setup wizard")
+def init(template: str, interactive: bool) -> None:
+    """Initialize .runbook/ with a config template (or the wizard)."""
+    if interactive:
+        from .config.wizard import run_wizard
+
+        run_wizard()
+        return
     from .config.onboarding import quick_setup
 
     paths = quick_setup(template)
@@ -397,6 +403,34 @@ def eval_cmd(ctx: click.Context, fixtures: str, offline: bool, concurrency: int,
             json.dump(report, f, indent=1)
         _echo(f"{DIM}report → {report_path}{RESET}")
     sys.exit(0 if report["failed"] == 0 else 1)
+
+
+@cli.command("eval-all")
+@click.option("--offline", is_flag=True)
+@click.option("--concurrency", default=1)
+@click.option("--provider", default=None)
+@click.option("--out", "out_path", default=".runbook/evals/summary.json")
+@click.pass_context
+def eval_all_cmd(ctx: click.Context, offline: bool, concurrency: int,
+                 provider: Optional[str], out_path: str) -> None:
+    """Run every fixture suite (sample + converted datasets) and aggregate."""
+    from .evals.run_all import run_all
+
+    config: Config = ctx.obj["config"]
+    llm_cfg = config.llm.model_dump(by_alias=False)
+    if provider:
+        llm_cfg["provider"] = provider
+
+    summary = run_all(
+        llm_factory=None if offline else (lambda: create_llm_client(llm_cfg)),
+        offline=offline, concurrency=concurrency, out_path=out_path)
+    for name, s in summary["suites"].items():
+        if "skipped" in s:
+            _echo(f"{DIM}{name}: skipped ({s['skipped']}){RESET}")
+        else:
+            _echo(f"{name}: {s['passed']}/{s['total']} pass "
+                  f"(avg {s['averageOverallScore']:.2f}, {s['wallMs']} ms)")
+    _echo(f"\noverall pass rate {summary['overallPassRate']:.0%} → {out_path}")
 
 
 # -- checkpoint ------------------------------------------------------------------
