@@ -29,6 +29,8 @@ void launch_flash_prefill_paged(const void*, const void*, const void*, const voi
                                 const void*, const void*, const void*, const void*,
                                 void*, int, int, int, int, int, float, void*);
 void launch_mfma_probe(const void*, const void*, void*, void*);
+void launch_skinny_gemm(const void*, const void*, void*, void*, int, int, long,
+                        int, void*);
 void launch_cosine_scores(const void*, const void*, void*, long, int, void*);
 void launch_masked_argmax(const void*, const void*, void*, int, int, void*);
 }
@@ -225,6 +227,31 @@ void store_kv(torch::Tensor k, torch::Tensor v, torch::Tensor kc, torch::Tensor 
                     slots.data_ptr(), T, Hk, D, BS, current_stream());
 }
 
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
+    CHECK_IN(x, torch::kBFloat16);
+    CHECK_IN(w, torch::kBFloat16);
+    int M = (int)x.size(0);
+    long K = x.size(1);
+    int N = (int)w.size(0);
+    TORCH_CHECK(M <= 64, "skinny_gemm supports M <= 64");
+    TORCH_CHECK(w.size(1) == K);
+    TORCH_CHECK(N % 64 == 0 && K % 32 == 0, "need N%64==0, K%32==0");
+    auto out = torch::empty({M, (long)N}, x.options());
+    // split-K so the grid fills the chip several blocks per CU
+    int splitk = 1;
+    while (splitk < 16 && (long)(N / 64) * splitk < 1024 && (K / 32) % (splitk * 2) == 0)
+        splitk *= 2;
+    torch::Tensor partial;
+    void* pptr = nullptr;
+    if (splitk > 1) {
+        partial = torch::empty({splitk, M, (long)N}, x.options().dtype(torch::kFloat32));
+        pptr = partial.data_ptr();
+    }
+    launch_skinny_gemm(x.data_ptr(), w.data_ptr(), pptr, out.data_ptr(), M, N, K,
+                       splitk, current_stream());
+    return out;
+}
+
 torch::Tensor cosine_scores(torch::Tensor matrix, torch::Tensor query) {
     CHECK_IN(matrix, torch::kFloat16);
     CHECK_IN(query, torch::kFloat16);
@@ -267,5 +294,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
     m.def("store_kv", &store_kv, "scatter K/V into paged cache");
     m.def("cosine_scores", &cosine_scores, "brute-force cosine scores");
+    m.def("skinny_gemm", &skinny_gemm, "MFMA skinny GEMM x[M<=64,K] @ W[N,K]^T");
     m.def("masked_argmax", &masked_argmax, "greedy sampling under validity mask");
 }

@@ -30,6 +30,12 @@ def _init_weight(out_f: int, in_f: int, dtype, device, gen: Optional[torch.Gener
     return w.to(dtype)
 
 
+def _linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    from .. import ops
+
+    return ops.linear(x, weight)
+
+
 class ColumnParallelLinear:
     """y_shard = x @ W_shard^T ; W sharded on the output dimension."""
 
@@ -41,7 +47,7 @@ class ColumnParallelLinear:
         self.weight = _init_weight(self.out_per_rank, in_features, dtype, device, gen)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
-        return x @ self.weight.t()
+        return _linear(x, self.weight)
 
 
 class RowParallelLinear:
@@ -59,7 +65,7 @@ class RowParallelLinear:
         self.weight = _init_weight(out_features, self.in_per_rank, dtype, device, gen)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
-        y = x @ self.weight.t()
+        y = _linear(x, self.weight)
         return all_reduce(y) if self.tp > 1 else y
 
 
@@ -69,4 +75,4 @@ class ReplicatedLinear:
         self.weight = _init_weight(out_features, in_features, dtype, device, gen)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
-        return x @ self.weight.t()
+        return _linear(x, self.weight)

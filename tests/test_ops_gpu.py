@@ -243,6 +243,35 @@ class TestPagedDecode:
         assert diff < 3e-2, diff
 
 
+class TestSkinnyGemm:
+    @pytest.mark.parametrize("M,N,K", [
+        (1, 4096, 4096), (8, 6144, 4096), (13, 4096, 14336),
+        (32, 28672, 4096), (64, 128256, 4096),
+    ])
+    def test_vs_matmul(self, M, N, K):
+        from runbookai_amd.ops import _get_ext
+
+        torch.manual_seed(3)
+        x = bf(torch.randn(M, K) * 0.3).to(DEV)
+        w = bf(torch.randn(N, K) * 0.3).to(DEV)
+        out = _get_ext().skinny_gemm(x, w)
+        expected = (x.float() @ w.float().t())
+        diff = (out.float() - expected).abs().max().item()
+        # bf16 inputs, fp32 accum: tolerance scales with sqrt(K)
+        tol = 0.02 * (K ** 0.5) * 0.3 * 0.3
+        assert diff < max(0.15, tol), f"diff {diff} (tol {tol})"
+
+    def test_linear_dispatch_uses_kernel(self):
+        x = bf(torch.randn(4, 4096)).to(DEV)
+        w = bf(torch.randn(4096, 4096)).to(DEV)
+        out = ops.linear(x, w)
+        expected = x.float() @ w.float().t()
+        assert (out.float() - expected).abs().max().item() < 1.0
+        # shapes the kernel can't take fall back to matmul
+        w_odd = bf(torch.randn(100, 4096)).to(DEV)
+        assert ops.linear(x, w_odd).shape == (4, 100)
+
+
 class TestRetrievalSampling:
     def test_topk_cosine(self):
         N, D, K = 5000, 384, 8
