@@ -170,7 +170,7 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """x [T, K] @ weight[N, K]^T with the decode-shaped MFMA kernel when it
     applies (GPU, T<=64, N%64==0, K%32==0), else hipBLASLt via matmul."""
     if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 64
-            and weight.shape[0] % 64 == 0 and weight.shape[1] % 32 == 0):
+            and weight.shape[0] % 64 == 0 and weight.shape[1] % 64 == 0):
         return _get_ext().skinny_gemm(x.contiguous(), weight)
     return x @ weight.t()
 

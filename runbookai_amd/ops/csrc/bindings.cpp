@@ -235,11 +235,11 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
     int N = (int)w.size(0);
     TORCH_CHECK(M <= 64, "skinny_gemm supports M <= 64");
     TORCH_CHECK(w.size(1) == K);
-    TORCH_CHECK(N % 64 == 0 && K % 32 == 0, "need N%64==0, K%32==0");
+    TORCH_CHECK(N % 64 == 0 && K % 64 == 0, "need N%64==0, K%64==0");
     auto out = torch::empty({M, (long)N}, x.options());
     // split-K so the grid fills the chip several blocks per CU
     int splitk = 1;
-    while (splitk < 16 && (long)(N / 64) * splitk < 1024 && (K / 32) % (splitk * 2) == 0)
+    while (splitk < 16 && (long)(N / 64) * splitk < 1024 && (K / 64) % (splitk * 2) == 0)
         splitk *= 2;
     torch::Tensor partial;
     void* pptr = nullptr;

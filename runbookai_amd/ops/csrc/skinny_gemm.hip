@@ -2,30 +2,36 @@
 // M <= 64 (decode batch), bf16 in / bf16 out, fp32 accumulate.
 //
 // Decode GEMMs are pure weight streaming (every W element used once, x is
-// L2-resident); the guide's "M = 256 sampling/decode projection GEMM"
-// recipe applies: W tiles staged through LDS with coalesced cooperative
-// loads, MFMA 16x16x32 bf16 for the dots, split-K so the grid fills 256
-// CUs even at small N. hipBLASLt runs these shapes at ~3.5-5 TB/s; this
-// kernel targets the ~6.3 TB/s HBM stream ceiling.
-//
-// Layout notes: W is [N, K] row-major (torch Linear convention), so the
-// MFMA B fragment (lane l: B[k=(l>>4)*8+j][n=l&15] = W[n][k]) is 8
-// CONTIGUOUS bf16 per lane from an LDS-staged W tile. The A fragment
-// (x[m=l&15][k-slice]) reads global directly — x is tiny and cache-hot.
+// L2-resident). Structure per the guide's decode-projection recipe:
+// - W tiles [64 n][64 k] staged to LDS by global_load_lds (lane-linear
+//   dest), double-buffered: stage tile t+1 while computing tile t; the
+//   __syncthreads() drain is the simple 2-phase pattern.
+// - XOR swizzle on BOTH the glds source address and the LDS read
+//   (rule 21): a linear [64][128 B] image would put the 16 rows of each
+//   MFMA B-fragment group on 2 bank slots (8-way conflict); byte ^=
+//   ((row & 7) << 4) spreads them across 8 slots.
+// - MFMA 16x16x32 bf16; A fragments (x) read global directly (L2-hot).
+// - split-K across workgroups with an fp32 partial slab + merge kernel so
+//   small-N projections still fill 256 CUs.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 #define BN 64          // output columns per workgroup
-#define BK 32          // k per MFMA tile
+#define BKG 64         // k per staged tile (2 MFMA k-steps)
+#define TILE_BYTES (BN * BKG * 2)   // 8 KB
 #define MAX_MTILES 4   // up to 64 rows (4 x 16)
+
+DEVINL unsigned swz(unsigned row, unsigned colb) {
+    return row * (BKG * 2) + (colb ^ ((row & 7u) << 4));
+}
 
 __launch_bounds__(256, 2)
 __global__ void skinny_gemm_kernel(
     const ushort_t* __restrict__ x,   // [M, K]
     const ushort_t* __restrict__ w,   // [N, K]
-    float* __restrict__ partial,      // [SPLITK, M, N] fp32 (or null if SPLITK==1)
+    float* __restrict__ partial,      // [SPLITK, M, N] fp32 (null if SPLITK==1)
     ushort_t* __restrict__ out,       // [M, N] bf16 (used when SPLITK==1)
     int M, int N, long K, int splitk, int m_tiles) {
     const int n_base = blockIdx.x * BN;
@@ -33,43 +39,80 @@ __global__ void skinny_gemm_kernel(
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x / WAVE;
 
-    // this split's k range (multiple of BK)
-    const long k_per_split = ((K / BK + splitk - 1) / splitk) * BK;
+    const long k_per_split = ((K / BKG + splitk - 1) / splitk) * BKG;
     const long k_begin = split * k_per_split;
     const long k_end = min(K, k_begin + k_per_split);
+    if (k_begin >= k_end) {
+        // empty split: zero its slab slice
+#pragma unroll
+        for (int t = 0; t < MAX_MTILES; ++t) {
+            if (t >= m_tiles) break;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = t * 16 + (lane >> 4) * 4 + r;
+                if (m >= M) continue;
+                const int n = n_base + wid * 16 + (lane & 15);
+                if (splitk > 1) partial[((long)split * M + m) * N + n] = 0.f;
+            }
+        }
+        return;
+    }
 
-    __shared__ ushort_t w_lds[BN][BK];
+    __shared__ ushort_t w_lds[2][BN * BKG];
+
+    // glds staging: thread t covers LDS bytes [pass*4096 + t*16); the
+    // SOURCE address carries the inverse swizzle (dest is lane-linear).
+    const unsigned tid = threadIdx.x;
+
+    auto stage = [&](int buf, long k0) {
+#pragma unroll
+        for (int pass = 0; pass < 2; ++pass) {
+            const unsigned p = pass * 4096u + tid * 16u;      // linear LDS byte
+            const unsigned row = p >> 7;                      // 128 B per row
+            const unsigned colb = (p & 127u) ^ ((row & 7u) << 4);
+            const ushort_t* src = w + ((long)(n_base + row) * K + k0) + colb / 2;
+            // wave-uniform LDS base for this (wave, pass) segment
+            ushort_t* lds_base = &w_lds[buf][0] + (pass * 4096u + wid * 1024u) / 2;
+            __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)src,
+                                             (__attribute__((address_space(3))) void*)lds_base,
+                                             16, 0, 0);
+        }
+    };
 
     f32x4_t acc[MAX_MTILES];
 #pragma unroll
     for (int t = 0; t < MAX_MTILES; ++t) acc[t] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
-    const int a_row_raw = lane & 15;          // m within tile
+    const int a_row_raw = lane & 15;
     const int a_kslice = (lane >> 4) * 8;
-    const int b_col = wid * 16 + (lane & 15); // n within BN block
+    const unsigned b_row = wid * 16 + (lane & 15);  // n within BN block
 
-    for (long k0 = k_begin; k0 < k_end; k0 += BK) {
-        // cooperative W tile load: 256 threads x 16B = 4 KB = [64][32] bf16.
-        // thread t loads row t/4, 8 elements at (t%4)*8 — 64 B chunks/row.
-        __syncthreads();
-        {
-            const int row = threadIdx.x >> 2;
-            const int col = (threadIdx.x & 3) * 8;
-            *reinterpret_cast<bf16x8_t*>(&w_lds[row][col]) =
-                *reinterpret_cast<const bf16x8_t*>(w + (long)(n_base + row) * K + k0 + col);
-        }
-        __syncthreads();
-        // B fragment: 8 contiguous bf16 of W_lds[b_col]
-        bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(&w_lds[b_col][a_kslice]);
+    stage(0, k_begin);
+    __syncthreads();   // drains the prologue glds
+
+    int cur = 0;
+    for (long k0 = k_begin; k0 < k_end; k0 += BKG) {
+        const long k_next = k0 + BKG;
+        if (k_next < k_end) stage(cur ^ 1, k_next);
 #pragma unroll
-        for (int t = 0; t < MAX_MTILES; ++t) {
-            if (t >= m_tiles) break;
-            const int m = t * 16 + a_row_raw;
-            const int m_clamped = m < M ? m : 0;  // pad rows recompute row 0
-            bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-                x + (long)m_clamped * K + k0 + a_kslice);
-            acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[t], 0, 0, 0);
+        for (int ks = 0; ks < 2; ++ks) {
+            // B fragment: 16 contiguous bytes at swizzled offset
+            const unsigned colb = (unsigned)(ks * 64 + (lane >> 4) * 16);
+            bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
+                &w_lds[cur][swz(b_row, colb) / 2]);
+#pragma unroll
+            for (int t = 0; t < MAX_MTILES; ++t) {
+                if (t >= m_tiles) break;
+                const int m = t * 16 + a_row_raw;
+                const int m_clamped = m < M ? m : 0;
+                bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
+                    x + (long)m_clamped * K + k0 + ks * 32 + a_kslice);
+                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[t],
+                                                                 0, 0, 0);
+            }
         }
+        __syncthreads();   // next tile's glds has landed; LDS reads done
+        cur ^= 1;
     }
 
     // epilogue: C layout row=(l>>4)*4+r, col=l&15

@@ -98,6 +98,32 @@ def main() -> None:
         prefill[f"T{T}_tok_s"] = round(tps, 1)
     report["prefill"] = prefill
 
+    # skinny GEMM vs hipBLASLt on the decode projection shapes
+    from runbookai_amd.ops import _get_ext
+
+    ext = _get_ext()
+    gemm = {}
+    for (name, N, K) in (("qkv", 6144, 4096), ("o", 4096, 4096),
+                         ("gate_up", 28672, 4096), ("down", 4096, 14336),
+                         ("lm_head", 128256, 4096)):
+        for M in (1, 8, 32):
+            x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+            w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+            for label, fn in (("skinny", lambda: ext.skinny_gemm(x, w)),
+                              ("blaslt", lambda: x @ w.t())):
+                for _ in range(3):
+                    fn()
+                torch.cuda.synchronize()
+                t0 = time.time()
+                for _ in range(20):
+                    fn()
+                torch.cuda.synchronize()
+                us = (time.time() - t0) / 20 * 1e6
+                tbps = (N * K * 2) / (us / 1e6) / 1e12
+                gemm[f"{name}_M{M}_{label}"] = f"{us:.1f}us {tbps:.2f}TB/s"
+            del x, w
+    report["gemm_shapes"] = gemm
+
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/perf_probe.json", "w") as f:
         json.dump(report, f, indent=1)
