@@ -167,9 +167,15 @@ USE_SKINNY_GEMM = True
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
-    """x [T, K] @ weight[N, K]^T with the decode-shaped MFMA kernel when it
-    applies (GPU, T<=64, N%64==0, K%32==0), else hipBLASLt via matmul."""
-    if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 64
+    """x [T, K] @ weight[N, K]^T.
+
+    Measured dispatch (profiles/PERF_LOG.md): the hand MFMA skinny kernel
+    beats hipBLASLt on small-M x small-N decode projections (qkv/o shapes:
+    1.5-2.5 -> 2.7-3.0 TB/s); hipBLASLt already streams big-N shapes
+    (gate_up/down/lm_head) at 5.5-6.5 TB/s, so those stay on the library.
+    """
+    if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 16
+            and weight.shape[0] <= 8192
             and weight.shape[0] % 64 == 0 and weight.shape[1] % 64 == 0):
         return _get_ext().skinny_gemm(x.contiguous(), weight)
     return x @ weight.t()
