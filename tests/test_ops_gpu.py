@@ -247,6 +247,9 @@ class TestSkinnyGemm:
     @pytest.mark.parametrize("M,N,K", [
         (1, 4096, 4096), (8, 6144, 4096), (13, 4096, 14336),
         (32, 28672, 4096), (64, 128256, 4096),
+        # tiny-model shapes (small K exercises short split-K ranges)
+        (30, 512, 256), (30, 256, 256), (30, 1024, 256), (30, 256, 512),
+        (30, 4096, 256),
     ])
     def test_vs_matmul(self, M, N, K):
         from runbookai_amd.ops import _get_ext
@@ -332,7 +335,13 @@ class TestEngineGpu:
                 torch.tensor([0, len(ids)], dtype=torch.int32),
                 m.kv.slot_mapping(1, 0, len(ids)))
             out[dev] = logits.float().cpu()
-        diff = (out["cpu"] - out[DEV]).abs().max().item()
-        # bf16 rounding + split-K summation-order differences amplify over
-        # layers; per-op numerics are covered by the kernel-vs-reference tests
-        assert diff < 0.3, f"GPU/CPU logits diverged: {diff}"
+        # bf16 rounding + summation-order differences amplify over layers;
+        # per-op numerics are covered by the kernel-vs-reference tests.
+        # Check relative closeness + ranking agreement, not bitwise max-abs.
+        a, b = out["cpu"][0], out[DEV][0]
+        rel = (a - b).norm().item() / a.norm().item()
+        assert rel < 0.05, f"GPU/CPU logits rel-diff {rel}"
+        top_cpu = a.topk(20).indices.tolist()
+        top_gpu = b.topk(20).indices.tolist()
+        overlap = len(set(top_cpu) & set(top_gpu))
+        assert overlap >= 15, f"top-20 overlap only {overlap}"
