@@ -116,6 +116,7 @@ class LlamaModel:
         seed: int = 1234,
         kv_blocks: Optional[int] = None,
         kv_block_size: int = 16,
+        init_device: Optional[str] = None,
     ) -> None:
         self.cfg = cfg
         self.device = device
@@ -128,8 +129,10 @@ class LlamaModel:
 
         rank = get_rank() % self.tp
         # per-tensor seeded generators ON DEVICE: identical full tensors on
-        # every rank (before sharding), no host fp32 staging
-        gen_dev = device if str(device).startswith("cuda") else "cpu"
+        # every rank (before sharding), no host fp32 staging. init_device
+        # overrides placement (e.g. "cpu" for cross-device parity tests —
+        # CPU and CUDA RNG streams differ for the same seed).
+        gen_dev = init_device or (device if str(device).startswith("cuda") else "cpu")
         self._gen_counter = 0
 
         def gen_factory() -> torch.Generator:
@@ -146,6 +149,15 @@ class LlamaModel:
         self.final_norm_w = torch.ones(cfg.hidden_size, dtype=dtype, device=device)
         self.lm_head = ReplicatedLinear(cfg.hidden_size, cfg.vocab_size, dtype, device,
                                         gen_factory())
+        if str(gen_dev) != str(device):
+            # init happened off-device (parity mode): move weights over
+            self.embed.weight = self.embed.weight.to(device)
+            self.lm_head.weight = self.lm_head.weight.to(device)
+            for layer in self.layers:
+                layer.qkv.weight = layer.qkv.weight.to(device)
+                layer.o_proj.weight = layer.o_proj.weight.to(device)
+                layer.gate_up.weight = layer.gate_up.weight.to(device)
+                layer.down.weight = layer.down.weight.to(device)
         cos, sin = self._rope_tables()
         self.rope_cos = cos.to(device)
         self.rope_sin = sin.to(device)

@@ -328,7 +328,9 @@ class TestEngineGpu:
         ids = list(range(50, 80))
         out = {}
         for dev in ("cpu", DEV):
-            m = LlamaModel(CONFIGS["tiny"], device=dev, seed=11)
+            # init_device="cpu" so both models share identical weights
+            # (CPU vs CUDA RNG streams differ for the same seed)
+            m = LlamaModel(CONFIGS["tiny"], device=dev, seed=11, init_device="cpu")
             m.kv.allocate(1, len(ids))
             logits = m.prefill(
                 torch.tensor(ids), torch.arange(len(ids), dtype=torch.int32),
