@@ -26,8 +26,18 @@ class LocalEngineClient:
 
     @classmethod
     def from_config(cls, cfg: dict[str, Any]) -> "LocalEngineClient":
+        import sys
+
+        import torch
+
+        model = cfg.get("model", "llama3-8b")
+        if not torch.cuda.is_available() and model != "tiny":
+            print(f"[runbook] no GPU visible: substituting the 'tiny' engine for "
+                  f"'{model}' (CPU cannot serve an 8B/70B policy interactively)",
+                  file=sys.stderr)
+            model = "tiny"
         engine = get_engine(
-            model=cfg.get("model", "llama3-8b"),
+            model=model,
             tp=cfg.get("tensor_parallel") or cfg.get("tensorParallel"),
             device=cfg.get("device"),
         )
