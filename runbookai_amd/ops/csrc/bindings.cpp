@@ -237,9 +237,11 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
     TORCH_CHECK(w.size(1) == K);
     TORCH_CHECK(N % 64 == 0 && K % 64 == 0, "need N%64==0, K%64==0");
     auto out = torch::empty({M, (long)N}, x.options());
-    // split-K so the grid fills the chip several blocks per CU
+    // split-K so the grid fills the chip, but keep >= 8 tiles per split so
+    // the glds pipeline reaches steady state
     int splitk = 1;
-    while (splitk < 16 && (long)(N / 64) * splitk < 1024 && (K / 64) % (splitk * 2) == 0)
+    while (splitk < 16 && (long)(N / 64) * splitk < 1024
+           && (K / 64) % (splitk * 2) == 0 && (K / 64) / (splitk * 2) >= 8)
         splitk *= 2;
     torch::Tensor partial;
     void* pptr = nullptr;
