@@ -120,6 +120,31 @@ class ApprovalManager:
         return self._record(operation, resource, risk, False, "policy",
                             "no approval channel available; denied by default")
 
+    @classmethod
+    def with_slack_pending_store(cls, policy: Optional[ApprovalPolicy] = None,
+                                 pending_dir: str = ".runbook/pending",
+                                 timeout_s: float = 300.0,
+                                 notify_channel: str = "#incidents") -> "ApprovalManager":
+        """Approval via the Slack button webhook: creates a pending-approval
+        file, posts the request to Slack, and blocks until the webhook
+        resolves it (reference approval.ts Slack path + webhooks flow)."""
+        from ..tools.incident.slack import post_update
+        from ..webhooks.slack_webhook import PendingApprovalStore
+
+        store = PendingApprovalStore(pending_dir)
+
+        def slack_approver(request: dict[str, Any]) -> Optional[bool]:
+            approval_id = store.create(request)
+            post_update(
+                notify_channel,
+                f":rotating_light: Approval needed [{request.get('risk', '?')}]: "
+                f"{request.get('operation', '?')} on {request.get('resource', '?')}\n"
+                f"approve/deny id `{approval_id}` via the approval webhook",
+            )
+            return store.wait_for(approval_id, timeout_s=timeout_s)
+
+        return cls(policy=policy, slack_approver=slack_approver)
+
     def _record(self, operation: str, resource: str, risk: str, approved: bool,
                 approver: str, reason: str) -> ApprovalRecord:
         rec = ApprovalRecord(operation=operation, resource=resource, risk=risk,
