@@ -5,6 +5,8 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
+#include <mutex>
+
 extern "C" {
 void launch_rmsnorm(const void*, const void*, void*, int, int, float, void*);
 void launch_rmsnorm_residual(const void*, const void*, void*, const void*, void*,
@@ -29,8 +31,8 @@ void launch_flash_prefill_paged(const void*, const void*, const void*, const voi
                                 const void*, const void*, const void*, const void*,
                                 void*, int, int, int, int, int, float, void*);
 void launch_mfma_probe(const void*, const void*, void*, void*);
-void launch_skinny_gemm(const void*, const void*, void*, void*, int, int, long,
-                        int, void*);
+void launch_skinny_gemm(const void*, const void*, void*, void*, void*, int, int,
+                        long, int, void*);
 void launch_cosine_scores(const void*, const void*, void*, long, int, void*);
 void launch_masked_argmax(const void*, const void*, void*, int, int, void*);
 }
@@ -245,12 +247,28 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
         splitk *= 2;
     torch::Tensor partial;
     void* pptr = nullptr;
+    void* cptr = nullptr;
     if (splitk > 1) {
         partial = torch::empty({splitk, M, (long)N}, x.options().dtype(torch::kFloat32));
         pptr = partial.data_ptr();
+        // Persistent arrival counters for the in-launch split-K combine:
+        // zero-initialized ONCE; each tile's last arriver resets its slot, so
+        // the buffer is 0 again before every subsequent launch (stream order).
+        static torch::Tensor cnt;
+        static std::mutex cnt_mu;
+        {
+            std::lock_guard<std::mutex> g(cnt_mu);
+            long need = N / 64;
+            if (!cnt.defined() || cnt.numel() < need
+                || cnt.device() != x.device()) {
+                cnt = torch::zeros({std::max(need, (long)2048)},
+                                   x.options().dtype(torch::kInt32));
+            }
+        }
+        cptr = cnt.data_ptr();
     }
-    launch_skinny_gemm(x.data_ptr(), w.data_ptr(), pptr, out.data_ptr(), M, N, K,
-                       splitk, current_stream());
+    launch_skinny_gemm(x.data_ptr(), w.data_ptr(), pptr, out.data_ptr(), cptr,
+                       M, N, K, splitk, current_stream());
     return out;
 }
 

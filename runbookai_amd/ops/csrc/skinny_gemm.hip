@@ -27,6 +27,8 @@ DEVINL unsigned swz(unsigned row, unsigned colb) {
     return row * (BKG * 2) + (colb ^ ((row & 7u) << 4));
 }
 
+typedef __attribute__((address_space(1))) unsigned gu32_t;
+
 template <int NT>
 __launch_bounds__(256, 2)
 __global__ void skinny_gemm_kernel(
@@ -34,6 +36,8 @@ __global__ void skinny_gemm_kernel(
     const ushort_t* __restrict__ w,   // [N, K]
     float* __restrict__ partial,      // [SPLITK, M, N] fp32 (null if SPLITK==1)
     ushort_t* __restrict__ out,       // [M, N] bf16 (used when SPLITK==1)
+    unsigned* __restrict__ cnt,       // [N/BN] arrival counters (pre-zeroed;
+                                      // last arriver resets its slot)
     int M, int N, long K, int splitk) {
     const int n_base = blockIdx.x * BN;
     const int split = blockIdx.y;
@@ -41,21 +45,11 @@ __global__ void skinny_gemm_kernel(
     const int wid = threadIdx.x / WAVE;
 
     const long k_per_split = ((K / BKG + splitk - 1) / splitk) * BKG;
-    const long k_begin = split * k_per_split;
+    const long k_begin = min(K, (long)split * k_per_split);
     const long k_end = min(K, k_begin + k_per_split);
-    if (k_begin >= k_end) {
-#pragma unroll
-        for (int t = 0; t < NT; ++t) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int m = t * 16 + (lane >> 4) * 4 + r;
-                if (m >= M) continue;
-                const int n = n_base + wid * 16 + (lane & 15);
-                if (splitk > 1) partial[((long)split * M + m) * N + n] = 0.f;
-            }
-        }
-        return;
-    }
+    const long n_tiles = (k_end - k_begin) / BKG;
+    // NOTE: empty splits (n_tiles == 0) still run the epilogue + combine
+    // protocol below — every block MUST draw its arrival ticket.
 
     __shared__ ushort_t w_lds[2][BN * BKG];
 
@@ -102,11 +96,12 @@ __global__ void skinny_gemm_kernel(
         xp[t] = x + (long)m_clamped * K + k_begin + a_kslice;
     }
 
-    stage(0, 0);
+    if (n_tiles > 0) {
+        stage(0, 0);
+    }
     __syncthreads();   // drains the prologue glds
 
     int cur = 0;
-    const long n_tiles = (k_end - k_begin) / BKG;
     for (long ti = 0; ti < n_tiles; ++ti) {
         if (ti + 1 < n_tiles) stage(cur ^ 1, ti + 1);
         const long xoff = ti * BKG;
@@ -141,6 +136,39 @@ __global__ void skinny_gemm_kernel(
             }
         }
     }
+    if (splitk == 1) return;
+
+    // ---- in-launch split-K combine (guide Guideline 16, counter form) ----
+    // Publish the fp32 slab with an agent-scope release; the LAST arriving
+    // slice block for this tile reduces all slabs and writes bf16. No block
+    // ever spins — non-last blocks exit after their ticket.
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");   // every wave drains its stores
+    __syncthreads();
+    unsigned* flag = reinterpret_cast<unsigned*>(&w_lds[0][0]);  // reuse the one LDS array
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // keep the post-wbl2 wait (pitfall 12)
+        const unsigned ticket = __hip_atomic_fetch_add(
+            (gu32_t*)&cnt[blockIdx.x], 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        flag[0] = (ticket == (unsigned)splitk - 1) ? 1u : 0u;
+    }
+    __syncthreads();
+    if (flag[0] == 0u) return;   // not the last arriver
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        // reset the counter for the next launch (stream order guarantees
+        // no concurrent user of this slot)
+        __hip_atomic_store((gu32_t*)&cnt[blockIdx.x], 0u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < M * BN; i += 256) {
+        const int m = i / BN;
+        const int n = n_base + (i % BN);
+        float s = 0.f;
+        for (int sp = 0; sp < splitk; ++sp) s += partial[((long)sp * M + m) * N + n];
+        out[(long)m * N + n] = f2bf(s);
+    }
 }
 
 __global__ void skinny_gemm_merge_kernel(const float* __restrict__ partial,
@@ -155,36 +183,19 @@ __global__ void skinny_gemm_merge_kernel(const float* __restrict__ partial,
 }
 
 extern "C" void launch_skinny_gemm(const void* x, const void* w, void* partial,
-                                   void* out, int M, int N, long K, int splitk,
-                                   hipStream_t stream) {
+                                   void* out, void* cnt, int M, int N, long K,
+                                   int splitk, hipStream_t stream) {
     const int m_tiles = (M + 15) / 16;
     dim3 grid(N / BN, splitk), block(256);
+#define LAUNCH_NT(NT)                                                            \
+    hipLaunchKernelGGL(skinny_gemm_kernel<NT>, grid, block, 0, stream,           \
+                       (const ushort_t*)x, (const ushort_t*)w, (float*)partial,  \
+                       (ushort_t*)out, (unsigned*)cnt, M, N, K, splitk)
     switch (m_tiles) {
-    case 1:
-        hipLaunchKernelGGL(skinny_gemm_kernel<1>, grid, block, 0, stream,
-                           (const ushort_t*)x, (const ushort_t*)w, (float*)partial,
-                           (ushort_t*)out, M, N, K, splitk);
-        break;
-    case 2:
-        hipLaunchKernelGGL(skinny_gemm_kernel<2>, grid, block, 0, stream,
-                           (const ushort_t*)x, (const ushort_t*)w, (float*)partial,
-                           (ushort_t*)out, M, N, K, splitk);
-        break;
-    case 3:
-        hipLaunchKernelGGL(skinny_gemm_kernel<3>, grid, block, 0, stream,
-                           (const ushort_t*)x, (const ushort_t*)w, (float*)partial,
-                           (ushort_t*)out, M, N, K, splitk);
-        break;
-    default:
-        hipLaunchKernelGGL(skinny_gemm_kernel<4>, grid, block, 0, stream,
-                           (const ushort_t*)x, (const ushort_t*)w, (float*)partial,
-                           (ushort_t*)out, M, N, K, splitk);
-        break;
+    case 1: LAUNCH_NT(1); break;
+    case 2: LAUNCH_NT(2); break;
+    case 3: LAUNCH_NT(3); break;
+    default: LAUNCH_NT(4); break;
     }
-    if (splitk > 1) {
-        long mn = (long)M * N;
-        int blocks = (int)min((mn + 255) / 256, (long)2048);
-        hipLaunchKernelGGL(skinny_gemm_merge_kernel, dim3(blocks), dim3(256), 0, stream,
-                           (const float*)partial, (ushort_t*)out, mn, mn, splitk);
-    }
+#undef LAUNCH_NT
 }

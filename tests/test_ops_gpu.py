@@ -264,6 +264,26 @@ class TestSkinnyGemm:
         tol = 0.02 * (K ** 0.5) * 0.3 * 0.3
         assert diff < max(0.15, tol), f"diff {diff} (tol {tol})"
 
+    def test_splitk_combine_race_screen(self):
+        """The in-launch split-K combine is placement/timing-sensitive code:
+        screen it over many runs and interleaved shapes (guide two-lane
+        discipline for sync-structure edits)."""
+        from runbookai_amd.ops import _get_ext
+
+        ext = _get_ext()
+        torch.manual_seed(9)
+        shapes = [(1, 4096, 4096), (8, 6144, 4096), (16, 4096, 14336)]
+        tensors = []
+        for M, N, K in shapes:
+            x = bf(torch.randn(M, K) * 0.3).to(DEV)
+            w = bf(torch.randn(N, K) * 0.3).to(DEV)
+            tensors.append((x, w, x.float() @ w.float().t()))
+        for round_i in range(15):
+            for x, w, expected in tensors:
+                out = ext.skinny_gemm(x, w)
+                diff = (out.float() - expected).abs().max().item()
+                assert diff < 0.6, f"round {round_i}: diff {diff}"
+
     def test_linear_dispatch_uses_kernel(self):
         x = bf(torch.randn(4, 4096)).to(DEV)
         w = bf(torch.randn(4096, 4096)).to(DEV)
