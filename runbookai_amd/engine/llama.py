@@ -181,18 +181,26 @@ class LlamaModel:
 
     # -- forward -----------------------------------------------------------------
 
-    def _transformer_body(self, h: torch.Tensor, positions, slots, attn_fn) -> torch.Tensor:
+    def _transformer_body(self, h: torch.Tensor, positions, slots, attn_fn,
+                          fused_kv: bool = False) -> torch.Tensor:
         """Shared fused body: every residual add is fused into the following
         RMSNorm (rmsnorm_residual), SwiGLU reads the fused [gate|up] rows.
-        attn_fn(q, layer_idx) -> attention output [T, heads_per_rank, D].
-        Returns the FINAL-normed hidden states [T, H]."""
+        attn_fn(q, k, v, layer_idx) -> attention output [T, heads, D].
+        fused_kv: RoPE + KV scatter as ONE kernel — only valid when attn_fn
+        reads K/V from the paged cache (decode/chunk), since the packed k
+        stays unrotated. Returns the FINAL-normed hidden states [T, H]."""
         T = h.shape[0]
         normed = ops.rmsnorm(h, self.layers[0].input_norm_w, self.cfg.rms_eps)
         for i, layer in enumerate(self.layers):
             qkv = layer.qkv(normed)
             q, k, v = layer._split_qkv(qkv, T)
-            q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, positions)
-            ops.store_kv(k, v, self.kv.k[i], self.kv.v[i], slots)
+            if fused_kv:
+                q = ops.rope_store_kv_fused(q, k, v, self.rope_cos, self.rope_sin,
+                                            positions, self.kv.k[i], self.kv.v[i],
+                                            slots)
+            else:
+                q, k = ops.apply_rope(q, k, self.rope_cos, self.rope_sin, positions)
+                ops.store_kv(k, v, self.kv.k[i], self.kv.v[i], slots)
             attn = attn_fn(q, k, v, i)
             attn_out = layer.o_proj(attn.reshape(T, -1))
             normed, h = ops.rmsnorm_residual(attn_out, h, layer.post_norm_w,
@@ -231,7 +239,7 @@ class LlamaModel:
             return ops.paged_decode_attention(q, self.kv.k[i], self.kv.v[i],
                                               block_tables, seq_lens, self.scale)
 
-        normed = self._transformer_body(h, positions, slots, attn_fn)
+        normed = self._transformer_body(h, positions, slots, attn_fn, fused_kv=True)
         return self.lm_head(normed)
 
     def chunk_step(self, token_ids: torch.Tensor, positions: torch.Tensor,
@@ -257,7 +265,7 @@ class LlamaModel:
             return ops.chunked_prefill_attention(q, self.kv.k[i], self.kv.v[i],
                                                  bt, hist, seq_starts_d, self.scale)
 
-        normed = self._transformer_body(h, positions, slots, attn_fn)
+        normed = self._transformer_body(h, positions, slots, attn_fn, fused_kv=True)
         last = (seq_starts[1:] - 1).long().to(device)
         return self.lm_head(normed[last])
 

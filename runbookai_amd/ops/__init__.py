@@ -92,6 +92,21 @@ def apply_rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.T
 USE_FLASH_PREFILL = True
 
 
+def rope_store_kv_fused(q, k, v, cos, sin, positions, k_cache, v_cache, slot_mapping):
+    """Fused RoPE + paged-KV scatter: returns the rotated q; the rotated k
+    and raw v land in the cache (packed k/v are NOT rotated — callers on
+    this path must read K from the cache)."""
+    if _on_gpu(q):
+        q = q.contiguous()
+        _get_ext().rope_store_kv(q, k.contiguous(), v.contiguous(), k_cache, v_cache,
+                                 cos, sin, positions.to(torch.int32).contiguous(),
+                                 slot_mapping.to(torch.int32).contiguous())
+        return q
+    q, k = reference.apply_rope(q, k, cos, sin, positions)
+    reference.store_kv(k, v, k_cache, v_cache, slot_mapping)
+    return q
+
+
 def prefill_attention(q, k, v, seq_starts, causal: bool = True,
                       scale: Optional[float] = None, batch_idx=None):
     scale = scale or 1.0 / math.sqrt(q.shape[-1])

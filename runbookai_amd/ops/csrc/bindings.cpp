@@ -15,6 +15,9 @@ void launch_silu_mul(const void*, const void*, void*, long, void*);
 void launch_silu_mul_fused(const void*, void*, long, int, void*);
 void launch_rope(void*, void*, const void*, const void*, const void*,
                  int, int, int, int, void*);
+void launch_rope_store_kv(void*, const void*, const void*, void*, void*,
+                          const void*, const void*, const void*, const void*,
+                          int, int, int, int, int, void*);
 void launch_paged_decode(const void*, const void*, const void*, const void*,
                          const void*, void*, int, int, int, int, int, int, float, void*);
 void launch_paged_decode_splitk(const void*, const void*, const void*, const void*,
@@ -110,6 +113,26 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos,
     TORCH_CHECK(k.size(0) == T && k.size(2) == D);
     launch_rope(q.data_ptr(), k.data_ptr(), cos.data_ptr(), sin.data_ptr(),
                 positions.data_ptr(), T, Hq, Hk, D, current_stream());
+}
+
+void rope_store_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                   torch::Tensor kc, torch::Tensor vc, torch::Tensor cos,
+                   torch::Tensor sin, torch::Tensor positions, torch::Tensor slots) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(k, torch::kBFloat16);
+    CHECK_IN(v, torch::kBFloat16);
+    CHECK_IN(kc, torch::kBFloat16);
+    CHECK_IN(vc, torch::kBFloat16);
+    CHECK_IN(cos, torch::kFloat32);
+    CHECK_IN(sin, torch::kFloat32);
+    CHECK_IN(positions, torch::kInt32);
+    CHECK_IN(slots, torch::kInt32);
+    int T = (int)q.size(0), Hq = (int)q.size(1), D = (int)q.size(2);
+    int Hk = (int)k.size(1), BS = (int)kc.size(2);
+    launch_rope_store_kv(q.data_ptr(), k.data_ptr(), v.data_ptr(), kc.data_ptr(),
+                         vc.data_ptr(), cos.data_ptr(), sin.data_ptr(),
+                         positions.data_ptr(), slots.data_ptr(), T, Hq, Hk, D, BS,
+                         current_stream());
 }
 
 torch::Tensor paged_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
@@ -306,6 +329,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("silu_mul", &silu_mul, "SwiGLU activation");
     m.def("silu_mul_fused", &silu_mul_fused, "SwiGLU from fused [gate|up] rows");
     m.def("rope_inplace", &rope_inplace, "RoPE in place on q,k");
+    m.def("rope_store_kv", &rope_store_kv, "fused RoPE + paged KV scatter");
     m.def("paged_decode", &paged_decode, "paged-KV decode attention");
     m.def("prefill_attn", &prefill_attn, "varlen prefill attention");
     m.def("flash_prefill", &flash_prefill, "MFMA flash prefill attention (D=128)");

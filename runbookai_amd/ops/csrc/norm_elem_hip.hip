@@ -188,6 +188,86 @@ __global__ void rope_kernel(ushort_t* __restrict__ q, ushort_t* __restrict__ k,
     }
 }
 
+// Fused RoPE + paged-KV scatter: rotates q (in place) and k, writing the
+// rotated k and raw v straight into the paged cache — one kernel where
+// rope_kernel + store_kv_kernel were two launches per layer.
+// Row space: t*(Hq+2*Hk) rows; q rows rotate in place, k rows rotate into
+// the cache, v rows copy into the cache.
+__global__ void rope_store_kv_kernel(ushort_t* __restrict__ q,
+                                     const ushort_t* __restrict__ k,
+                                     const ushort_t* __restrict__ v,
+                                     ushort_t* __restrict__ k_cache,
+                                     ushort_t* __restrict__ v_cache,
+                                     const float* __restrict__ cost,
+                                     const float* __restrict__ sint,
+                                     const int* __restrict__ positions,
+                                     const int* __restrict__ slots,
+                                     int T, int Hq, int Hk, int D, int BS) {
+    const int halfD = D / 2;
+    const int rows_per_tok = Hq + 2 * Hk;
+    const int waves_per_block = blockDim.x / WAVE;
+    const long wave_global = (long)blockIdx.x * waves_per_block + threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    if (wave_global >= (long)T * rows_per_tok) return;
+    const int t = (int)(wave_global / rows_per_tok);
+    const int h = (int)(wave_global % rows_per_tok);
+    const int pos = positions[t];
+    const int slot = slots[t];
+    const int blk = slot / BS, off = slot % BS;
+
+    if (h < Hq) {                       // q row: rotate in place
+        ushort_t* row = q + ((long)t * Hq + h) * D;
+        for (int i = lane; i < halfD; i += WAVE) {
+            ushort2_t pair = *reinterpret_cast<ushort2_t*>(row + 2 * i);
+            const float c = cost[(long)pos * halfD + i];
+            const float s = sint[(long)pos * halfD + i];
+            const float x0 = bf2f(pair[0]), x1 = bf2f(pair[1]);
+            ushort2_t o;
+            o[0] = f2bf(x0 * c - x1 * s);
+            o[1] = f2bf(x0 * s + x1 * c);
+            *reinterpret_cast<ushort2_t*>(row + 2 * i) = o;
+        }
+    } else if (h < Hq + Hk) {           // k row: rotate into the cache
+        const int hk = h - Hq;
+        const ushort_t* row = k + ((long)t * Hk + hk) * D;
+        ushort_t* dst = k_cache + (((long)blk * Hk + hk) * BS + off) * D;
+        for (int i = lane; i < halfD; i += WAVE) {
+            ushort2_t pair = *reinterpret_cast<const ushort2_t*>(row + 2 * i);
+            const float c = cost[(long)pos * halfD + i];
+            const float s = sint[(long)pos * halfD + i];
+            const float x0 = bf2f(pair[0]), x1 = bf2f(pair[1]);
+            ushort2_t o;
+            o[0] = f2bf(x0 * c - x1 * s);
+            o[1] = f2bf(x0 * s + x1 * c);
+            *reinterpret_cast<ushort2_t*>(dst + 2 * i) = o;
+        }
+    } else {                            // v row: straight copy into the cache
+        const int hk = h - Hq - Hk;
+        const ushort_t* row = v + ((long)t * Hk + hk) * D;
+        ushort_t* dst = v_cache + (((long)blk * Hk + hk) * BS + off) * D;
+        for (int i = lane * 4; i < D; i += WAVE * 4) {
+            *reinterpret_cast<ushort4_t*>(dst + i) =
+                *reinterpret_cast<const ushort4_t*>(row + i);
+        }
+    }
+}
+
+extern "C" void launch_rope_store_kv(void* q, const void* k, const void* v,
+                                     void* kc, void* vc, const void* cost,
+                                     const void* sint, const void* positions,
+                                     const void* slots, int T, int Hq, int Hk,
+                                     int D, int BS, hipStream_t stream) {
+    const int waves_per_block = 4;
+    long rows = (long)T * (Hq + 2 * Hk);
+    long blocks = (rows + waves_per_block - 1) / waves_per_block;
+    hipLaunchKernelGGL(rope_store_kv_kernel, dim3((unsigned)blocks),
+                       dim3(waves_per_block * WAVE), 0, stream,
+                       (ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
+                       (ushort_t*)kc, (ushort_t*)vc, (const float*)cost,
+                       (const float*)sint, (const int*)positions, (const int*)slots,
+                       T, Hq, Hk, D, BS);
+}
+
 extern "C" void launch_rope(void* q, void* k, const void* cost, const void* sint,
                             const void* positions, int T, int Hq, int Hk, int D,
                             hipStream_t stream) {
