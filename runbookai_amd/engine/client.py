@@ -17,6 +17,32 @@ from ..agent.types import ChatResponse, ToolCall, new_id
 from .engine import LLMEngine, get_engine
 
 
+def _sanitize_args_schema(params: dict[str, Any], max_props: int = 3) -> dict[str, Any]:
+    """Reduce a tool's JSON-schema parameters to the FSM-supported subset:
+    flat string/number/integer/boolean/enum properties with bounded lengths."""
+    props_in = params.get("properties", {}) if isinstance(params, dict) else {}
+    props_out: dict[str, Any] = {}
+    preferred = list(params.get("required", [])) + [k for k in props_in
+                                                    if k not in params.get("required", [])]
+    for key in preferred:
+        if len(props_out) >= max_props:
+            break
+        spec = props_in.get(key)
+        if not isinstance(spec, dict):
+            continue
+        if "enum" in spec:
+            props_out[key] = {"enum": [str(v) for v in spec["enum"]][:8]}
+        elif spec.get("type") == "string":
+            props_out[key] = {"type": "string", "maxLength": 60}
+        elif spec.get("type") in ("integer", "number", "boolean"):
+            props_out[key] = {"type": spec["type"]}
+        # objects/arrays are skipped — tools treat missing args as defaults
+    if not props_out:
+        props_out = {"query": {"type": "string", "maxLength": 60}}
+    return {"type": "object", "properties": props_out,
+            "required": list(props_out.keys())}
+
+
 class LocalEngineClient:
     def __init__(self, engine: LLMEngine, max_tokens: int = 1024,
                  temperature: float = 0.0) -> None:
@@ -60,17 +86,95 @@ class LocalEngineClient:
 
     # -- chat (free-form with optional tool calls) --------------------------------
 
+    #: fallback schema bounding free-text answers (random-init greedy decode
+    #: would otherwise never emit EOT and run to max_tokens)
+    ANSWER_SCHEMA: dict[str, Any] = {
+        "type": "object",
+        "properties": {"answer": {"type": "string", "maxLength": 600}},
+        "required": ["answer"],
+    }
+
     def chat(self, system: str, user: str,
              tools: Optional[list[dict[str, Any]]] = None) -> ChatResponse:
         kind, body = split_schema_tag(user)
         schema = PROMPT_SCHEMAS.get(kind) if kind else None
         tok = self.engine.tokenizer
+        if schema is None and tools:
+            return self._chat_with_tools(system, body, tools)
         ids = tok.encode_chat(system, body)
+        if schema is not None:
+            req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
+                                       temperature=self.temperature, schema=schema)
+            return ChatResponse(content=tok.decode(req.out_ids))
+        # free text: bound it with the answer schema and unwrap
         req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
-                                   temperature=self.temperature, schema=schema)
+                                   temperature=self.temperature,
+                                   schema=self.ANSWER_SCHEMA)
         text = tok.decode(req.out_ids)
-        tool_calls = self._parse_tool_calls(text, tools)
-        return ChatResponse(content=text, tool_calls=tool_calls)
+        try:
+            from ..agent.llm_parser import parse_json
+
+            text = str(parse_json(text).get("answer", text))
+        except Exception:  # noqa: BLE001
+            pass
+        return ChatResponse(content=text)
+
+    # -- grammar-constrained tool calling ------------------------------------------
+
+    def _chat_with_tools(self, system: str, user: str,
+                         tools: list[dict[str, Any]]) -> ChatResponse:
+        """Two-stage constrained decode: (1) pick an action + tool from an
+        enum grammar, (2) fill that tool's argument schema. The free-form
+        agent loop therefore executes REAL tool calls from any checkpoint —
+        schema discipline at the logits level instead of parse-and-hope."""
+        from ..agent.llm_parser import parse_json
+
+        tok = self.engine.tokenizer
+        names = [t["name"] for t in tools][:32]
+        decision_schema = {
+            "type": "object",
+            "properties": {
+                "thinking": {"type": "string", "maxLength": 160},
+                "action": {"enum": ["tool", "final"]},
+                "tool": {"enum": names},
+            },
+            "required": ["thinking", "action", "tool"],
+        }
+        ids = tok.encode_chat(system, user + "\n\nDecide: call a tool or give the final answer.")
+        req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
+                                   temperature=self.temperature, schema=decision_schema)
+        try:
+            decision = parse_json(tok.decode(req.out_ids))
+        except Exception:  # noqa: BLE001
+            decision = {"action": "final", "thinking": ""}
+        thinking = str(decision.get("thinking", ""))
+        if decision.get("action") == "tool" and decision.get("tool") in names:
+            name = decision["tool"]
+            spec = next(t for t in tools if t["name"] == name)
+            args_schema = _sanitize_args_schema(spec.get("parameters", {}))
+            ids2 = tok.encode_chat(
+                system, f"{user}\n\nProvide arguments for the tool `{name}`.")
+            req2 = self.engine.generate(ids2, max_new_tokens=self.max_tokens,
+                                        temperature=self.temperature, schema=args_schema)
+            try:
+                args = parse_json(tok.decode(req2.out_ids))
+                if not isinstance(args, dict):
+                    args = {}
+            except Exception:  # noqa: BLE001
+                args = {}
+            call = ToolCall(id=new_id("call-"), name=name, arguments=args)
+            return ChatResponse(content="", tool_calls=[call], thinking=thinking)
+        # final answer
+        ids3 = tok.encode_chat(system, user + "\n\nGive the final answer.")
+        req3 = self.engine.generate(ids3, max_new_tokens=self.max_tokens,
+                                    temperature=self.temperature,
+                                    schema=self.ANSWER_SCHEMA)
+        text = tok.decode(req3.out_ids)
+        try:
+            text = str(parse_json(text).get("answer", text))
+        except Exception:  # noqa: BLE001
+            pass
+        return ChatResponse(content=text, thinking=thinking)
 
     def chat_stream(self, system: str, user: str,
                     tools: Optional[list[dict[str, Any]]] = None) -> Iterator[str]:
@@ -96,7 +200,7 @@ class LocalEngineClient:
             yield tok.decode(req.out_ids[emitted:])
 
     @staticmethod
-    def _parse_tool_calls(text: str, tools: Optional[list[dict[str, Any]]]) -> list[ToolCall]:
+    def _parse_tool_calls_legacy(text: str, tools: Optional[list[dict[str, Any]]]) -> list[ToolCall]:
         """Extract {"toolCalls": [...]} patterns from model text."""
         if not tools or '"toolCalls"' not in text:
             return []
