@@ -341,6 +341,33 @@ def knowledge_validate() -> None:
     _echo(f"{GREEN}knowledge base valid{RESET}: {json.dumps(stats)}")
 
 
+@knowledge.group("auth")
+def knowledge_auth() -> None:
+    """Knowledge-source authentication."""
+
+
+@knowledge_auth.command("google")
+@click.option("--client-id", envvar="GOOGLE_CLIENT_ID", default="")
+@click.option("--client-secret", envvar="GOOGLE_CLIENT_SECRET", default="")
+def knowledge_auth_google(client_id: str, client_secret: str) -> None:
+    """Run the Google Drive OAuth loopback flow (requires network egress)."""
+    from .knowledge.sources.google_auth import TokenStore, run_auth_flow
+
+    store = TokenStore()
+    if store.valid():
+        _echo(f"{GREEN}already authenticated{RESET} (token at {store.path})")
+        return
+    if not client_id:
+        _echo(f"{RED}set GOOGLE_CLIENT_ID / GOOGLE_CLIENT_SECRET first{RESET}")
+        sys.exit(1)
+    try:
+        run_auth_flow(client_id, client_secret)
+        _echo(f"{GREEN}authenticated{RESET}")
+    except RuntimeError as e:
+        _echo(f"{YELLOW}{e}{RESET}")
+        sys.exit(1)
+
+
 @knowledge.command("stats")
 def knowledge_stats() -> None:
     from .knowledge.retriever.default import create_retriever
