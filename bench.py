@@ -68,10 +68,20 @@ def main() -> None:
 
     kv_blocks = 2048 if model_name != "tiny" else 512
     # weak-scaling benchmark = DATA parallel replicas (tp=1 per rank);
-    # --tp > 1 shards one model across all ranks instead (70B config)
-    engine = LLMEngine(model=model_name, device=device,
-                       tp=(args.tp if args.tp > 1 else 1),
-                       kv_blocks=kv_blocks, background=True)
+    # --tp > 1 shards ONE model across all ranks instead (70B config):
+    # rank 0 schedules + samples, other ranks follow broadcast steps
+    tp = args.tp if args.tp > 1 else 1
+    if tp > 1:
+        assert world == tp, f"--tp {tp} needs torchrun with {tp} ranks"
+    engine = LLMEngine(model=model_name, device=device, tp=tp,
+                       kv_blocks=kv_blocks, background=(tp == 1 or rank == 0))
+    if tp > 1 and rank != 0:
+        from runbookai_amd.parallel.tp_serving import run_follower_loop
+
+        run_follower_loop(engine.model)   # returns on rank 0's stop payload
+        barrier()
+        destroy()
+        return
     # one shared scenario for concurrent runs (scenario registry is global)
     set_scenario(SimScenario.redis_exhaustion())
 
@@ -97,19 +107,32 @@ def main() -> None:
     if args.warmup > 0:
         run_batch(args.warmup)
 
-    # timed region: barrier + device sync on both sides
-    barrier()
+    # timed region: barrier + device sync on both sides. Under TP the
+    # follower ranks are inside their broadcast loop (they execute every
+    # timed step with rank 0), so rank 0 times alone and the barrier moves
+    # to after the stop payload.
+    tp_mode = tp > 1
+    if not tp_mode:
+        barrier()
     if has_gpu:
         torch.cuda.synchronize()
     t0 = time.time()
     results = run_batch(args.steps)
     if has_gpu:
         torch.cuda.synchronize()
-    barrier()
+    if not tp_mode:
+        barrier()
     elapsed = time.time() - t0
+    if tp_mode:
+        from runbookai_amd.parallel.tp_serving import broadcast_stop
 
-    # MAX over ranks
-    if world > 1:
+        engine.shutdown()
+        broadcast_stop()
+        barrier()
+
+    # MAX over ranks (DP mode only; under TP rank 0 is the sole timer and
+    # follower ranks have already exited past the final barrier)
+    if world > 1 and not tp_mode:
         import torch.distributed as dist
 
         t = torch.tensor([elapsed], dtype=torch.float64,
@@ -119,7 +142,8 @@ def main() -> None:
 
     stats = engine.throughput_stats()
     ms_per_step = elapsed * 1000.0 / args.steps
-    inv_per_hour = world * args.steps / elapsed * 3600.0
+    total_investigations = args.steps if tp_mode else world * args.steps
+    inv_per_hour = total_investigations / elapsed * 3600.0
     pass_rate = sum(1 for r in results if r["passed"]) / max(1, len(results))
 
     if rank == 0:
@@ -141,7 +165,7 @@ def main() -> None:
                 "model": model_name,
                 "global_batch": args.concurrency * world,
                 "seq_len": engine.cfg.max_seq_len,
-                "parallelism": f"dp{world}",
+                "parallelism": f"tp{world}" if tp_mode else f"dp{world}",
                 "pass_rate": pass_rate,
                 "decode_tok_per_s": round(stats.get("decode_tok_per_s", 0.0), 1),
                 "prefill_tok_per_s": round(stats.get("prefill_tok_per_s", 0.0), 1),

@@ -70,6 +70,10 @@ class LLMEngine:
         self.device = device
         self.cfg = cfg
         self.model = LlamaModel(cfg, device=device, tp=tp, seed=seed, kv_blocks=kv_blocks)
+        if self.model.tp > 1:
+            # every rank must replay identical step sequences; keep eager
+            # (RCCL collectives inside hipGraph capture are not validated)
+            self.model.use_graphs = False
         self.tokenizer = ByteTokenizer(cfg.vocab_size)
         self.max_prefill_tokens = max_prefill_tokens
         self.max_batch = max_batch
@@ -203,6 +207,15 @@ class LLMEngine:
 
     # -- execution --------------------------------------------------------------------
 
+    def _tp_dispatch(self, payload: dict) -> torch.Tensor:
+        """Execute one model step; under TP, rank 0 broadcasts the payload so
+        every rank runs the same collectives in the same order."""
+        from ..parallel.tp_serving import broadcast_step, execute_step, tp_active
+
+        if tp_active(self.model.tp):
+            broadcast_step(payload)
+        return execute_step(self.model, payload)
+
     def _run_prefill(self, batch: list[Request]) -> None:
         t0 = time.time()
         kv = self.model.kv
@@ -219,12 +232,13 @@ class LLMEngine:
             kv.set_len(req.rid, n)
             req.pos = n
             req.pending_input = []
-        logits = self.model.prefill(
-            torch.tensor(token_ids, dtype=torch.int64),
-            torch.tensor(positions, dtype=torch.int32),
-            torch.tensor(starts, dtype=torch.int32),
-            torch.cat(slots),
-        )
+        logits = self._tp_dispatch({
+            "op": "prefill",
+            "token_ids": token_ids,
+            "positions": positions,
+            "seq_starts": starts,
+            "slots": torch.cat(slots).tolist(),
+        })
         self.stats["prefill_tokens"] += len(token_ids)
         self._sample_and_advance(batch, logits)
         with self._lock:
@@ -254,15 +268,17 @@ class LLMEngine:
             kv.set_len(req.rid, req.pos + n)
             req.pos += n
             req.pending_input = []
-        bt, _lens = kv.batch_tables([r.rid for r in batch], self.device)
-        logits = self.model.chunk_step(
-            torch.tensor(token_ids, dtype=torch.int64),
-            torch.tensor(positions, dtype=torch.int32),
-            torch.tensor(starts, dtype=torch.int32),
-            bt,
-            torch.tensor(hist, dtype=torch.int32),
-            torch.cat(slots),
-        )
+        bt, _lens = kv.batch_tables([r.rid for r in batch], "cpu")
+        logits = self._tp_dispatch({
+            "op": "chunk",
+            "token_ids": token_ids,
+            "positions": positions,
+            "seq_starts": starts,
+            "block_tables": bt.flatten().tolist(),
+            "bt_shape": list(bt.shape),
+            "hist_lens": hist,
+            "slots": torch.cat(slots).tolist(),
+        })
         self.stats["chunk_tokens"] = self.stats.get("chunk_tokens", 0) + len(token_ids)
         self._sample_and_advance(batch, logits)
         with self._lock:
@@ -283,12 +299,16 @@ class LLMEngine:
             kv.set_len(req.rid, req.pos + 1)
             req.pos += 1
             req.pending_input = []
-        bt, lens = kv.batch_tables([r.rid for r in batch], self.device)
-        logits = self.model.decode(
-            torch.tensor(input_ids, dtype=torch.int64),
-            torch.tensor(positions, dtype=torch.int32),
-            bt, lens, torch.cat(slot_list),
-        )
+        bt, lens = kv.batch_tables([r.rid for r in batch], "cpu")
+        logits = self._tp_dispatch({
+            "op": "decode",
+            "token_ids": input_ids,
+            "positions": positions,
+            "block_tables": bt.flatten().tolist(),
+            "bt_shape": list(bt.shape),
+            "seq_lens": lens.tolist(),
+            "slots": torch.cat(slot_list).tolist(),
+        })
         self.stats["decode_tokens"] += len(batch)
         self._sample_and_advance(batch, logits)
         with self._lock:

@@ -68,6 +68,55 @@ def test_tp2_matches_tp1():
     assert diff < 0.05, f"TP=2 diverged from TP=1 by {diff}"
 
 
+def _tp_serving_worker(rank: int, world: int, port: int, q) -> None:
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["RANK"] = str(rank)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from runbookai_amd.agent.llm_parser import PROMPT_SCHEMAS
+        from runbookai_amd.engine.engine import LLMEngine
+        from runbookai_amd.parallel.tp_serving import broadcast_stop, run_follower_loop
+
+        eng = LLMEngine(model="tiny", device="cpu", tp=world, background=False, seed=42)
+        if rank == 0:
+            req = eng.generate(eng.tokenizer.encode_chat("sys", "triage this"),
+                               max_new_tokens=2048, schema=PROMPT_SCHEMAS["triage"],
+                               timeout_s=200)
+            text = eng.tokenizer.decode(req.out_ids)
+            broadcast_stop()
+            q.put(("ok", text))
+        else:
+            steps = run_follower_loop(eng.model)
+            assert steps > 0
+    except Exception as e:  # noqa: BLE001
+        if rank == 0:
+            q.put(("err", repr(e)))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp_serving_constrained_generation():
+    """TP=2 serving over gloo: rank 0 schedules + samples, rank 1 follows
+    broadcast steps; output is schema-valid JSON (BASELINE config 5 path)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_serving_worker, args=(r, 2, 29641, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    data = json.loads(payload)
+    assert data["severity"] in ("low", "medium", "high", "critical")
+
+
 def test_column_row_parallel_shapes():
     from runbookai_amd.parallel.layers import ColumnParallelLinear, RowParallelLinear
 
