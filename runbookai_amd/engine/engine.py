@@ -27,9 +27,23 @@ from ..agent.llm_parser import PROMPT_SCHEMAS
 from .json_fsm import NUMBER_CLOSE_SENTINEL, JsonFsm
 from .kv_cache import PagedKvCache
 from .llama import CONFIGS, LlamaModel
-from .tokenizer import ACTIVE_VOCAB, ByteTokenizer, SpecialTokens
+from .tokenizer import (
+    ACTIVE_VOCAB,
+    MAX_WORD_LEN,
+    WORD_BASE,
+    WORD_STRINGS,
+    ByteTokenizer,
+    SpecialTokens,
+)
 
-MASK_REGION = 512  # byte vocab + specials, padded; legal ids live below this
+MASK_REGION = 1024  # bytes + specials + word tokens, padded
+assert ACTIVE_VOCAB <= MASK_REGION, "word vocab outgrew the sampler mask region"
+
+# word ids admissible at a given string capacity (word length <= cap)
+_WORD_IDS_BY_CAP: dict[int, list[int]] = {}
+for _cap in range(3, MAX_WORD_LEN + 1):
+    _WORD_IDS_BY_CAP[_cap] = [WORD_BASE + _i for _i, _w in enumerate(WORD_STRINGS)
+                              if len(_w) <= _cap]
 
 
 @dataclass
@@ -332,6 +346,11 @@ class LLMEngine:
                     mask[i, SpecialTokens.EOT] = True
                 else:
                     mask[i, allowed] = True
+                    # inside free string content, admit multi-byte word
+                    # tokens (one decode step per word)
+                    cap = req.fsm.string_capacity()
+                    if cap >= 3:
+                        mask[i, _WORD_IDS_BY_CAP[min(cap, MAX_WORD_LEN)]] = True
             else:
                 mask[i, :ACTIVE_VOCAB] = True
         mask_d = mask.to(region.device)
@@ -384,6 +403,13 @@ class LLMEngine:
         if req.fsm is not None:
             if req.fsm.done or tok == SpecialTokens.EOT:
                 finished = True
+            elif tok >= WORD_BASE:
+                # word token: advance the byte automaton through the word
+                for b in WORD_STRINGS[tok - WORD_BASE].encode("ascii"):
+                    req.fsm.advance(b)
+                req.out_ids.append(tok)
+                req.pending_input = [tok]
+                req.pending_input += self._drain_forced(req)
             else:
                 req.fsm.advance(tok)
                 if tok != NUMBER_CLOSE_SENTINEL:

@@ -54,6 +54,20 @@ class JsonFsm:
         frame = self._stack[-1]
         return self._frame_allowed(frame)
 
+    def string_capacity(self) -> int:
+        """Remaining free-content bytes when the FSM sits INSIDE a string
+        value (>=1 char already emitted, no forced bytes pending); 0
+        otherwise. The sampler uses this to admit multi-byte word tokens —
+        words are plain printable ASCII, so any word of length <= capacity
+        keeps the automaton exact (advanced byte-by-byte)."""
+        if self._pending or self._done or not self._stack:
+            return 0
+        frame = self._stack[-1]
+        if frame.get("kind") != "string" or frame["len"] == 0:
+            return 0
+        return max(0, min(frame["max"] - frame["len"],
+                          self.max_total_bytes - self.emitted))
+
     def advance(self, byte: int) -> None:
         """Consume the byte the sampler chose (must be in allowed_bytes())."""
         self.emitted += 1

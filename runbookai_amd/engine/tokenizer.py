@@ -39,8 +39,59 @@ class SpecialTokens:
     }
 
 
-#: number of ids that encode() can actually produce
-ACTIVE_VOCAB = BYTE_VOCAB + len(SpecialTokens.ALL)
+#: Multi-byte WORD tokens (ids from WORD_BASE): common English + SRE/ops
+#: vocabulary with leading spaces. Inside grammar-unconstrained string
+#: content the sampler may emit these (one decode step per word instead of
+#: per byte); structural JSON stays byte-exact. All characters are
+#: printable ASCII excluding '"' and '\\' so any word is legal inside a
+#: JSON string.
+_WORD_SRC = (
+    "the of and to in is for on with as at by from that this it an be are was "
+    "not has have had will can should could would may might must service "
+    "services error errors timeout timeouts latency connection connections "
+    "pool pools exhausted exhaustion redis database cache memory cpu disk "
+    "network gateway upstream downstream deploy deployment deployments config "
+    "configuration rollback restart scale scaling replica replicas pod pods "
+    "node nodes cluster clusters alarm alarms alert alerts metric metrics log "
+    "logs trace traces request requests response responses failure failures "
+    "failed failing spike spiked increased decreased degraded unavailable "
+    "unhealthy healthy saturated queue queues backlog throttled limit limits "
+    "rate capacity incident incidents root cause evidence hypothesis confirm "
+    "confirmed investigate investigation check checked because after before "
+    "during since between high low medium critical severity impact affected "
+    "checkout cart payment user auth api server client process thread load "
+    "traffic volume percent seconds minutes hours time start started stop "
+    "stopped running pending active issue issues problem problems fix fixed "
+    "mitigate mitigation resolve resolved monitor monitoring observed shows "
+    "indicates suggests correlates caused causing leading due likely possible "
+    "probable verify verified restarted increase decrease raise lower apply "
+    "applied revert reverted version release instance instances container "
+    "containers kubernetes docker aws cloud region zone endpoint endpoints "
+    "certificate certificates expired quota storage bucket shard shards index "
+    "identity provider callback webhook worker workers job jobs task tasks "
+    "event events message messages topic partition consumer producer lag "
+    "retry retries backoff circuit breaker health probe liveness readiness "
+    "oom kill killed leak leaking socket sockets port ports dns tls ssl http "
+    "https grpc tcp udp ip host hosts slow fast normal elevated dropped "
+    "refused reset closed open opened count total average peak p99 p95 max "
+    "min sum last first new old current previous next more less than all "
+    "some none other same different multiple single several"
+)
+
+_words: list[str] = []
+_seen: set[str] = set()
+for _w in _WORD_SRC.split():
+    for _cand in (" " + _w, " " + _w.capitalize()):
+        if _cand not in _seen:
+            _seen.add(_cand)
+            _words.append(_cand)
+WORD_STRINGS: tuple[str, ...] = tuple(_words)
+WORD_BASE = 262
+WORD_TOKENS: dict[str, int] = {w: WORD_BASE + i for i, w in enumerate(WORD_STRINGS)}
+MAX_WORD_LEN = max(len(w) for w in WORD_STRINGS)
+
+#: number of ids that encode()/the sampler can actually produce
+ACTIVE_VOCAB = WORD_BASE + len(WORD_STRINGS)
 
 
 class ByteTokenizer:
@@ -55,8 +106,33 @@ class ByteTokenizer:
         self.eot_id = SpecialTokens.EOT
 
     def encode(self, text: str, bos: bool = False, eot: bool = False) -> list[int]:
+        """Greedy longest-match over word tokens, bytes elsewhere —
+        lossless either way (word tokens decode to their exact strings)."""
         ids: list[int] = [self.bos_id] if bos else []
-        ids.extend(text.encode("utf-8"))
+        data = text.encode("utf-8")
+        i = 0
+        n = len(data)
+        while i < n:
+            if data[i] == 0x20:  # words all start with a space
+                matched = 0
+                for L in range(min(MAX_WORD_LEN, n - i), 2, -1):
+                    chunk = data[i:i + L]
+                    try:
+                        tok = WORD_TOKENS.get(chunk.decode("ascii"))
+                    except UnicodeDecodeError:
+                        tok = None
+                    if tok is not None:
+                        # word boundary: next byte must not extend the word
+                        nxt = data[i + L] if i + L < n else 0x20
+                        if not (0x61 <= nxt <= 0x7A or 0x41 <= nxt <= 0x5A):
+                            ids.append(tok)
+                            matched = L
+                            break
+                if matched:
+                    i += matched
+                    continue
+            ids.append(data[i])
+            i += 1
         if eot:
             ids.append(self.eot_id)
         return ids
@@ -66,7 +142,9 @@ class ByteTokenizer:
         for i in ids:
             if 0 <= i < BYTE_VOCAB:
                 out.append(i)
-            # specials and out-of-range ids render as nothing
+            elif i >= WORD_BASE and i - WORD_BASE < len(WORD_STRINGS):
+                out.extend(WORD_STRINGS[i - WORD_BASE].encode("ascii"))
+            # specials render as nothing
         return out.decode("utf-8", errors="replace")
 
     def encode_chat(self, system: str, user: str,
@@ -79,11 +157,11 @@ class ByteTokenizer:
             return [S.START_HEADER, *role.encode("utf-8"), S.END_HEADER, 10]
 
         if system:
-            ids += header("system") + list(system.encode("utf-8")) + [S.EOT]
-        ids += header("user") + list(user.encode("utf-8")) + [S.EOT]
+            ids += header("system") + self.encode(system) + [S.EOT]
+        ids += header("user") + self.encode(user) + [S.EOT]
         ids += header("assistant")
         if assistant_prefix:
-            ids += list(assistant_prefix.encode("utf-8"))
+            ids += self.encode(assistant_prefix)
         return ids
 
     def stop_ids(self) -> set[int]:
