@@ -1,0 +1,310 @@
+"""Context-engineering + support component tests (parity with the
+reference's per-component suites: scratchpad, tool-summarizer,
+context-compactor, investigation-memory, conversation-memory, log
+analyzer, tool cache, confidence, citations, safety, parallel executor)."""
+import time
+
+import pytest
+
+from runbookai_amd.agent.citation_context import CitationContext
+from runbookai_amd.agent.confidence import (
+    aggregate_confidence,
+    calculate_confidence,
+    classify_evidence,
+    level_at_least,
+    to_level,
+)
+from runbookai_amd.agent.context_compactor import ContextCompactor, create_compactor
+from runbookai_amd.agent.investigation_memory import InvestigationMemory
+from runbookai_amd.agent.log_analyzer import LogAnalyzer
+from runbookai_amd.agent.parallel_executor import (
+    ParallelToolExecutor,
+    analyze_tool_dependencies,
+)
+from runbookai_amd.agent.safety import SafetyManager, classify_aws_operation
+from runbookai_amd.agent.scratchpad import Scratchpad, jaccard
+from runbookai_amd.agent.tool_cache import ToolCache
+from runbookai_amd.agent.tool_summarizer import ToolSummarizer
+from runbookai_amd.agent.types import Tool, ToolCall
+
+
+class TestScratchpad:
+    def test_tiers_and_drilldown(self):
+        pad = Scratchpad("s1")
+        rec = pad.append_tool_result("aws_query", {"service": "ecs"}, "2 items",
+                                     {"items": [1, 2]})
+        assert pad.get_result_by_id(rec.result_id).full_result == {"items": [1, 2]}
+        ctx = pad.build_tiered_context()
+        assert rec.result_id in ctx and '"items"' in ctx
+
+    def test_full_result_cap(self):
+        pad = Scratchpad("s2")
+        pad.append_tool_result("t", {}, "big", {"blob": "x" * 10_000})
+        assert "truncated" in pad.build_tiered_context()
+
+    def test_soft_limits_warn_not_block(self):
+        pad = Scratchpad("s3")
+        for _ in range(5):
+            pad.append_tool_result("search_knowledge", {"q": "x"}, "s", {})
+        warning = pad.check_tool_limit("search_knowledge", {"search_knowledge": 5})
+        assert warning and "5" in warning
+        assert pad.check_tool_limit("aws_query", {"aws_query": 10}) is None
+
+    def test_jaccard_retry_loop(self):
+        pad = Scratchpad("s4")
+        for _ in range(3):
+            pad.append_tool_result("logs", {"filter": "redis pool exhausted"}, "s", {})
+        assert pad.detect_retry_loop("logs", {"filter": "redis pool exhausted"})
+        assert not pad.detect_retry_loop("logs", {"filter": "completely different thing"})
+        assert jaccard("redis pool exhausted", "redis pool exhausted now") > 0.6
+
+    def test_compaction_plan_archives(self):
+        pad = Scratchpad("s5")
+        ids = [pad.append_tool_result("t", {"i": i}, f"s{i}", {"i": i}).result_id
+               for i in range(4)]
+        from runbookai_amd.agent.scratchpad import CompactionPlan
+
+        cleared = pad.apply_compaction_plan(CompactionPlan(
+            keep_full=[ids[3]], keep_compact=[ids[2]], clear=ids[:2]))
+        assert cleared == 2
+        assert pad.get_result_by_id(ids[0]) is not None  # archived, still retrievable
+        assert "cleared" in pad.build_tiered_context()
+
+    def test_jsonl_resume(self, tmp_path):
+        pad = Scratchpad("sess-resume", str(tmp_path))
+        pad.append_tool_result("aws_query", {"service": "ecs"}, "summary-1", {"x": 1})
+        pad.append("thinking", text="thought about redis")
+        pad2 = Scratchpad.load("sess-resume", str(tmp_path))
+        assert pad2.tool_counts == {"aws_query": 1}
+        assert pad2.tool_uses[0].summary == "summary-1"
+        assert any(e.kind == "thinking" for e in pad2.entries)
+
+
+class TestToolSummarizer:
+    def test_alarm_summary(self):
+        s = ToolSummarizer().summarize("cloudwatch_alarms", {}, {
+            "alarms": [{"name": "a1", "state": "ALARM", "reason": "hot"},
+                       {"name": "a2", "state": "OK"}]})
+        assert "1/2" in s.summary
+        assert s.has_errors and s.health_status == "alarming"
+
+    def test_error_summary(self):
+        s = ToolSummarizer().summarize("datadog", {}, None, error="boom")
+        assert s.has_errors and "boom" in s.summary
+
+    def test_generic_and_services(self):
+        s = ToolSummarizer().summarize("aws_query", {"service": "ecs"}, {
+            "items": [{"name": "x", "service": "checkout-api"}]})
+        assert s.item_count == 1
+        assert "checkout-api" in s.services
+
+
+class TestCompactor:
+    def _pad(self, n=20):
+        pad = Scratchpad("c1")
+        for i in range(n):
+            pad.append_tool_result("cloudwatch_logs", {"i": i},
+                                   "redis errors" if i % 3 == 0 else "quiet",
+                                   {"i": i}, has_errors=i % 3 == 0)
+        return pad
+
+    def test_plan_respects_caps(self):
+        pad = self._pad(40)
+        plan = ContextCompactor().compact(pad, query="redis errors")
+        assert len(plan.keep_full) <= 10
+        assert len(plan.keep_compact) <= 15
+        assert len(plan.keep_full) + len(plan.keep_compact) + len(plan.clear) == 40
+
+    def test_error_results_score_higher(self):
+        pad = self._pad(20)
+        c = ContextCompactor()
+        plan = c.compact(pad, query="redis errors")
+        kept = set(plan.keep_full)
+        err_ids = {r.result_id for r in pad.tool_uses if r.has_errors}
+        assert kept & err_ids, "error-bearing results should be kept full"
+
+    def test_budgeted_variant(self):
+        pad = self._pad(30)
+        c = ContextCompactor()
+        plan = c.compact(pad, token_budget=5000)
+        assert c.estimated_tokens(plan) <= 6000
+
+    def test_presets(self):
+        assert create_compactor("research").config.max_full == 15
+        assert create_compactor("unknown").config.max_full == 10
+
+
+class TestInvestigationMemory:
+    def test_extract_from_thinking(self):
+        mem = InvestigationMemory("m1")
+        mem.extract_from_thinking(
+            "I found that the redis pool is exhausted. Errors from checkout-api "
+            "suggest saturation. Next I should check the cart-service deploy.")
+        kinds = {n.kind for n in mem.notes}
+        assert "finding" in kinds and "question" in kinds
+        assert "checkout-api" in mem.discovered_services
+
+    def test_drain_new_discoveries(self):
+        mem = InvestigationMemory("m2")
+        mem.track_service("cart-service")
+        mem.track_symptom("latency spike")
+        s, y = mem.drain_new_discoveries()
+        assert s == ["cart-service"] and y == ["latency spike"]
+        assert mem.drain_new_discoveries() == ([], [])
+
+    def test_persistence(self, tmp_path):
+        mem = InvestigationMemory("m3", str(tmp_path))
+        mem.add_note("finding", "pool exhausted")
+        mem.save()
+        mem2 = InvestigationMemory("m3", str(tmp_path))
+        mem2.init()
+        assert mem2.notes[0].text == "pool exhausted"
+
+
+class TestLogAnalyzer:
+    LINES = [
+        "2026-02-10T09:12:03Z ERROR checkout-api redis: connection pool exhausted",
+        "2026-02-10T09:12:09Z ERROR cart-service dial tcp: i/o timeout",
+        "2026-02-10T09:12:11Z WARN cart-service retrying after 429 TooManyRequests",
+        "2026-02-10T09:13:00Z INFO checkout-api served request in 40ms",
+        "java.lang.OutOfMemoryError: Java heap space",
+    ]
+
+    def test_patterns_ranked_by_severity(self):
+        hits = LogAnalyzer().analyze_patterns(self.LINES)
+        names = [h.name for h in hits]
+        assert "oom" in names and "connection_pool" in names and "throttle" in names
+        sev = [h.severity for h in hits]
+        assert sev.index("critical") == 0 if "critical" in sev else True
+
+    def test_level_filter(self):
+        out = LogAnalyzer().filter_by_level(self.LINES, "ERROR")
+        assert len(out) == 2
+
+    def test_analyze_produces_hypotheses(self):
+        result = LogAnalyzer().analyze(self.LINES)
+        assert result["suggestedHypotheses"]
+        assert "checkout-api" in result["services"]
+
+
+class TestToolCache:
+    def test_ttl_expiry(self):
+        c = ToolCache()
+        c.put("datadog", {"q": 1}, {"v": 1})
+        assert c.get("datadog", {"q": 1}) == {"v": 1}
+        # force-expire
+        from runbookai_amd.agent.tool_cache import call_signature
+
+        c._store[call_signature("datadog", {"q": 1})].expires_at = time.time() - 1
+        assert c.get("datadog", {"q": 1}) is None
+
+    def test_non_cacheable(self):
+        c = ToolCache()
+        c.put("aws_mutate", {"op": "delete"}, {"done": True})
+        assert c.get("aws_mutate", {"op": "delete"}) is None
+
+    def test_invalidate_by_name(self):
+        c = ToolCache()
+        c.put("datadog", {"q": 1}, 1)
+        c.put("prometheus", {"q": 1}, 2)
+        assert c.invalidate(tool_name="datadog") == 1
+        assert c.get("prometheus", {"q": 1}) == 2
+
+    def test_lru_eviction(self):
+        c = ToolCache(max_entries=2)
+        for i in range(3):
+            c.put("datadog", {"q": i}, i)
+        assert c.stats()["evictions"] == 1
+
+
+class TestConfidenceCitations:
+    def test_levels(self):
+        assert to_level(0.9) == "high" and to_level(0.5) == "medium" and to_level(0.1) == "low"
+        assert level_at_least("high", "medium") and not level_at_least("low", "medium")
+
+    def test_calculate_bounds(self):
+        assert 0.05 <= calculate_confidence(10, 0) <= 0.95
+        assert calculate_confidence(0, 5) < calculate_confidence(5, 0)
+
+    def test_classify(self):
+        assert classify_evidence("connection pool exhausted in logs") == "supporting"
+        assert classify_evidence("no alarms, all healthy") == "contradicting"
+
+    def test_aggregate_weighted_to_max(self):
+        assert aggregate_confidence([0.9, 0.1]) > 0.6
+
+    def test_citations_dedupe_and_cap(self):
+        c = CitationContext(max_citations=2)
+        assert c.ref("Doc A") == "[1]"
+        assert c.ref("Doc A") == "[1]"
+        assert c.ref("Doc B") == "[2]"
+        assert c.ref("Doc C") == ""  # over cap
+        assert "## Sources" in c.format_markdown()
+
+
+class TestSafety:
+    def test_classification(self):
+        assert classify_aws_operation("describe-instances") == "none"
+        assert classify_aws_operation("update-service") == "medium"
+        assert classify_aws_operation("terminate-instances") == "critical"
+
+    def test_budget_and_cooldown(self):
+        s = SafetyManager(max_mutations_per_session=1, critical_cooldown_s=60)
+        ok, _ = s.can_proceed("update-service", "x")
+        assert ok
+        s.record_mutation("delete-cluster", "db")
+        ok, reason = s.can_proceed("update-service", "y")
+        assert not ok and "budget" in reason
+
+    def test_critical_cooldown(self):
+        s = SafetyManager(max_mutations_per_session=10, critical_cooldown_s=60)
+        s.record_mutation("terminate-instances", "i-1")
+        ok, reason = s.can_proceed("delete-table", "t")
+        assert not ok and "cooldown" in reason
+
+
+class TestParallelExecutor:
+    def test_executes_and_reports_durations(self):
+        def slow(**_):
+            time.sleep(0.01)
+            return {"ok": True}
+
+        tools = {"a": Tool(name="a", description="", parameters={}, execute=slow)}
+        calls = [ToolCall(id=str(i), name="a", arguments={}) for i in range(4)]
+        results = ParallelToolExecutor(max_concurrent=4).execute_all(tools, calls)
+        assert all(r.ok for r in results)
+        assert all(r.duration_ms >= 0 for r in results)
+
+    def test_errors_captured(self):
+        def boom(**_):
+            raise RuntimeError("nope")
+
+        tools = {"b": Tool(name="b", description="", parameters={}, execute=boom)}
+        [r] = ParallelToolExecutor().execute_all(tools, [ToolCall(id="1", name="b",
+                                                                  arguments={})])
+        assert not r.ok and "nope" in r.error
+
+    def test_dependency_batching(self):
+        calls = [ToolCall(id="1", name="aws_query", arguments={}),
+                 ToolCall(id="2", name="datadog", arguments={}),
+                 ToolCall(id="3", name="get_full_result", arguments={})]
+        batches = analyze_tool_dependencies(calls)
+        assert len(batches[0]) == 2
+        assert batches[1][0].name == "get_full_result"
+
+
+class TestHypothesisEngineRoundtrip:
+    def test_json_roundtrip_and_scoring(self):
+        from runbookai_amd.agent.hypothesis import HypothesisEngine
+
+        eng = HypothesisEngine()
+        root = eng.add("db issue", priority=1)
+        eng.branch(root.id, ["conn exhaustion", "slow query"])
+        eng.add_evidence(root.children[0] if isinstance(root.children[0], str) else "",
+                         "conn errors seen", True, "logs")
+        blob = eng.to_json()
+        eng2 = HypothesisEngine.from_json(blob)
+        assert len(eng2.hypotheses) == 3
+        assert eng2.to_markdown().count("⑂") == 1
+        tree = eng2.to_tree_data()
+        assert len(tree) == 1 and len(tree[0]["children"]) == 2
