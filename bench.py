@@ -66,6 +66,14 @@ def main() -> None:
                                           "investigation-fixtures.sample.json"))
     cases = fixtures["cases"]
 
+    # knowledge base: runbooks synced through the (GPU when available)
+    # embedder + vector store, searched during triage/remediation
+    from runbookai_amd.knowledge.indexer.embedder import create_embedder
+    from runbookai_amd.knowledge.retriever.default import create_retriever
+
+    retriever = create_retriever(in_memory=True, embedder=create_embedder())
+    retriever.sync()
+
     kv_blocks = 2048 if model_name != "tiny" else 512
     # weak-scaling benchmark = DATA parallel replicas (tp=1 per rank);
     # --tp > 1 shards ONE model across all ranks instead (70B config):
@@ -88,9 +96,9 @@ def main() -> None:
     def run_investigation(i: int) -> dict:
         case = cases[i % len(cases)]
         client = LocalEngineClient(engine, max_tokens=args.max_tokens)
-        registry = ToolRegistry()
+        registry = ToolRegistry(knowledge_retriever=retriever)
         orch = InvestigationOrchestrator(
-            llm=client, tool_executor=registry,
+            llm=client, tool_executor=registry, knowledge_retriever=retriever,
             max_iterations=int(case.get("execute", {}).get("maxIterations", 6)),
         )
         result = orch.investigate(case["query"], incident_id=case.get("incidentId"))
