@@ -147,6 +147,38 @@ class TestEngine:
             eng.shutdown()
 
 
+class TestEnginePressure:
+    def test_kv_exhaustion_queues_and_completes(self):
+        """More concurrent requests than the KV pool holds: admission defers
+        until blocks free up; every request still completes (failure-
+        recovery parity: graceful queuing, not crashes)."""
+        from runbookai_amd.agent.llm_parser import PROMPT_SCHEMAS
+
+        # tiny pool: ~16 blocks of 16 tokens (minus graph scratch)
+        eng = LLMEngine(model="tiny", device="cpu", background=True, kv_blocks=40)
+        try:
+            tok = eng.tokenizer
+            reqs = []
+            for i in range(6):
+                ids = tok.encode_chat("sys", f"case {i}")
+                reqs.append(eng.submit(ids, max_new_tokens=96,
+                                       schema=PROMPT_SCHEMAS["generateConclusion"]))
+            for r in reqs:
+                assert r.done_event.wait(timeout=180), "request starved under KV pressure"
+            assert all(r.error == "" for r in reqs)
+        finally:
+            eng.shutdown()
+
+    def test_oversized_prompt_clamped(self):
+        eng = LLMEngine(model="tiny", device="cpu", background=False)
+        try:
+            ids = [65] * (eng.cfg.max_seq_len + 500)
+            req = eng.generate(ids, max_new_tokens=4)
+            assert req.state == "done"
+        finally:
+            eng.shutdown()
+
+
 class TestConstrainedToolCalling:
     def test_free_form_agent_executes_tools_on_random_weights(self):
         """The two-stage tool-calling grammar: even a random-init model
