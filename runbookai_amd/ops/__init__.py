@@ -216,7 +216,12 @@ def masked_sample(logits, allowed_mask=None, temperature: float = 0.0,
                   top_p: float = 1.0, generator=None):
     if temperature <= 0.0:
         return masked_greedy(logits, allowed_mask)
-    # stochastic path: fp32 torch math on-device (vocab-sized, not hot)
+    if _on_gpu(logits) and logits.shape[1] <= 2048:
+        u = torch.rand(logits.shape[0], device=logits.device, generator=generator)
+        mask_u8 = (allowed_mask.to(torch.uint8).contiguous()
+                   if allowed_mask is not None else None)
+        return _get_ext().masked_topp(logits.to(torch.bfloat16).contiguous(), mask_u8,
+                                      u.float(), temperature, top_p).long()
     return reference.masked_sample(logits, allowed_mask, temperature, top_p, generator)
 
 

@@ -38,6 +38,8 @@ void launch_skinny_gemm(const void*, const void*, void*, void*, void*, int, int,
                         long, int, void*);
 void launch_cosine_scores(const void*, const void*, void*, long, int, void*);
 void launch_masked_argmax(const void*, const void*, void*, int, int, void*);
+void launch_masked_topp(const void*, const void*, const void*, void*, int, int,
+                        float, float, void*);
 }
 
 namespace {
@@ -321,6 +323,23 @@ torch::Tensor masked_argmax(torch::Tensor logits, c10::optional<torch::Tensor> m
     return out;
 }
 
+torch::Tensor masked_topp(torch::Tensor logits, c10::optional<torch::Tensor> mask,
+                          torch::Tensor uniform, double temperature, double top_p) {
+    CHECK_IN(logits, torch::kBFloat16);
+    CHECK_IN(uniform, torch::kFloat32);
+    int B = (int)logits.size(0), V = (int)logits.size(1);
+    TORCH_CHECK(V <= 2048, "masked_topp supports V <= 2048 (grammar region)");
+    const void* mptr = nullptr;
+    if (mask.has_value()) {
+        CHECK_IN(mask.value(), torch::kUInt8);
+        mptr = mask->data_ptr();
+    }
+    auto out = torch::empty({B}, logits.options().dtype(torch::kInt32));
+    launch_masked_topp(logits.data_ptr(), mptr, uniform.data_ptr(), out.data_ptr(),
+                       B, V, (float)temperature, (float)top_p, current_stream());
+    return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -340,4 +359,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("cosine_scores", &cosine_scores, "brute-force cosine scores");
     m.def("skinny_gemm", &skinny_gemm, "MFMA skinny GEMM x[M<=64,K] @ W[N,K]^T");
     m.def("masked_argmax", &masked_argmax, "greedy sampling under validity mask");
+    m.def("masked_topp", &masked_topp,
+          "fused temperature softmax + nucleus sampling under validity mask");
 }

@@ -316,6 +316,45 @@ class TestRetrievalSampling:
         expected = ref.masked_sample(logits.cpu(), mask, temperature=0.0)
         assert torch.equal(out.cpu(), expected)
 
+    def test_masked_topp_respects_mask_and_distribution(self):
+        torch.manual_seed(12)
+        B, V = 64, 1024
+        logits = bf(torch.randn(B, V)).to(DEV)
+        mask = torch.zeros(B, V, dtype=torch.bool)
+        mask[:, 100:200] = True
+        out = ops.masked_sample(logits.float(), mask.to(DEV), temperature=1.0, top_p=0.9)
+        assert ((out >= 100) & (out < 200)).all(), "samples escaped the mask"
+
+    def test_masked_topp_peaked_distribution(self):
+        """A strongly peaked row must (almost) always return its mode."""
+        B, V = 32, 512
+        logits = torch.full((B, V), -5.0)
+        logits[:, 7] = 10.0
+        out = ops.masked_sample(bf(logits).float().to(DEV), None,
+                                temperature=1.0, top_p=0.9)
+        assert (out.cpu() == 7).all()
+
+    def test_masked_topp_statistics(self):
+        """Two-outcome distribution: empirical frequency tracks the softmax
+        ratio (temperature sampling actually samples)."""
+        V = 256
+        logits = torch.full((1, V), -20.0)
+        logits[0, 3] = 1.0
+        logits[0, 9] = 0.0   # p(3)/p(9) = e
+        counts = {3: 0, 9: 0}
+        g = torch.Generator(device=DEV)
+        g.manual_seed(5)
+        row = bf(logits).float().to(DEV)
+        batch = row.expand(256, V).contiguous()
+        for _ in range(4):
+            out = ops.masked_sample(batch, None, temperature=1.0, top_p=1.0,
+                                    generator=g)
+            for t in out.cpu().tolist():
+                assert t in (3, 9)
+                counts[t] += 1
+        ratio = counts[3] / max(1, counts[9])
+        assert 1.8 < ratio < 4.2, f"empirical ratio {ratio} vs e~2.72"
+
     def test_argmax_unmasked(self):
         logits = bf(torch.randn(4, 1000)).to(DEV)
         out = ops.masked_greedy(logits, None)
