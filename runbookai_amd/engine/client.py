@@ -198,22 +198,3 @@ class LocalEngineClient:
                     time.sleep(0.005)
         if emitted < len(req.out_ids):
             yield tok.decode(req.out_ids[emitted:])
-
-    @staticmethod
-    def _parse_tool_calls_legacy(text: str, tools: Optional[list[dict[str, Any]]]) -> list[ToolCall]:
-        """Extract {"toolCalls": [...]} patterns from model text."""
-        if not tools or '"toolCalls"' not in text:
-            return []
-        try:
-            from ..agent.llm_parser import parse_json
-
-            data = parse_json(text)
-            calls = []
-            known = {t["name"] for t in tools}
-            for c in data.get("toolCalls", []) if isinstance(data, dict) else []:
-                if isinstance(c, dict) and c.get("name") in known:
-                    calls.append(ToolCall(id=new_id("call-"), name=c["name"],
-                                          arguments=c.get("arguments", {}) or {}))
-            return calls
-        except Exception:  # noqa: BLE001
-            return []
