@@ -154,6 +154,13 @@ def main() -> None:
     inv_per_hour = total_investigations / elapsed * 3600.0
     pass_rate = sum(1 for r in results if r["passed"]) / max(1, len(results))
 
+    # accuracy tier (hermetic): the reference-scorer offline gate over the
+    # fixtures' mockResults — semantic accuracy needs trained weights, so
+    # the random-init pass_rate below is reported next to this gate
+    from runbookai_amd.evals.benchmark import run_benchmark
+
+    offline_pass_rate = run_benchmark(fixtures, offline=True)["passRate"]
+
     if rank == 0:
         line = {
             "metric": "investigations_per_hour",
@@ -175,6 +182,7 @@ def main() -> None:
                 "seq_len": engine.cfg.max_seq_len,
                 "parallelism": f"tp{world}" if tp_mode else f"dp{world}",
                 "pass_rate": pass_rate,
+                "offline_gate_pass_rate": offline_pass_rate,
                 "decode_tok_per_s": round(stats.get("decode_tok_per_s", 0.0), 1),
                 "prefill_tok_per_s": round(stats.get("prefill_tok_per_s", 0.0), 1),
                 "llm_calls_total": stats.get("requests", 0),

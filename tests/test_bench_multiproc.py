@@ -1,0 +1,41 @@
+"""bench.py under torchrun (gloo, tiny model): DP replicas and TP serving —
+the exact launch mode the driver uses for the scaling run."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_bench(extra, port):
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", str(port), os.path.join(REPO, "bench.py"),
+           "--gpus", "2", "--steps", "1", "--warmup", "0",
+           "--concurrency", "1", "--model", "tiny", *extra]
+    proc = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
+                          cwd=REPO)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    lines = [l for l in proc.stdout.splitlines() if l.startswith("{")]
+    assert lines, proc.stdout[-2000:]
+    return json.loads(lines[-1])
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(500)
+def test_bench_dp2():
+    out = _run_bench([], 29721)
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["value"] > 0
+    assert out["config"]["offline_gate_pass_rate"] == 1.0
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(500)
+def test_bench_tp2():
+    out = _run_bench(["--tp", "2"], 29722)
+    assert out["config"]["parallelism"] == "tp2"
+    assert out["value"] > 0
