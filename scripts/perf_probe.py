@@ -98,6 +98,32 @@ def main() -> None:
         prefill[f"T{T}_tok_s"] = round(tps, 1)
     report["prefill"] = prefill
 
+    # attention kernels in isolation (Llama-3-8B shapes)
+    from runbookai_amd import ops as rops
+
+    attn = {}
+    B_, Hq_, Hk_, D_, S_ = 16, 32, 8, 128, 2048
+    T_ = B_ * S_
+    q = torch.randn(T_, Hq_, D_, dtype=torch.bfloat16, device="cuda") * 0.5
+    k = torch.randn(T_, Hk_, D_, dtype=torch.bfloat16, device="cuda") * 0.5
+    v = torch.randn(T_, Hk_, D_, dtype=torch.bfloat16, device="cuda") * 0.5
+    starts = torch.arange(0, T_ + 1, S_, dtype=torch.int32).to("cuda")
+    for _ in range(2):
+        rops.prefill_attention(q, k, v, starts, causal=True)
+    torch.cuda.synchronize()
+    t0 = time.time()
+    iters_a = 5
+    for _ in range(iters_a):
+        rops.prefill_attention(q, k, v, starts, causal=True)
+    torch.cuda.synchronize()
+    dt = (time.time() - t0) / iters_a
+    # causal flops: 2 matmuls * 2*S^2/2*D per (head, seq)
+    flops = 2 * 2 * (S_ * S_ / 2) * D_ * Hq_ * B_
+    attn["flash_prefill_B16_H32_S2048_ms"] = round(dt * 1000, 2)
+    attn["flash_prefill_TF"] = round(flops / dt / 1e12, 1)
+    del q, k, v
+    report["attention"] = attn
+
     # skinny GEMM vs hipBLASLt on the decode projection shapes
     from runbookai_amd.ops import _get_ext
 
