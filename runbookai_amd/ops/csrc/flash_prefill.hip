@@ -26,9 +26,10 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 #define QTILE 64   // q rows per workgroup (16 per wave)
-#define KTILE 32   // kv tokens per LDS tile
+#define KTILE 64   // kv tokens per LDS tile
 #define DHEAD 128  // head dim (Llama-3)
-#define VPAD 40    // padded row length (elements) for transposed V / P tiles
+#define VPAD 72    // padded row length (elements) for transposed V / P tiles
+#define NKC (KTILE / 16)   // 16-col score sub-tiles per kv tile
 
 // ------------------------------------------------------------------ probe
 __global__ void mfma_probe_kernel(const ushort_t* __restrict__ A,
@@ -147,9 +148,9 @@ __global__ void flash_prefill_kernel(
         __syncthreads();
 
         // ---- QK^T ----
-        f32x4_t sc[2];
+        f32x4_t sc[NKC];
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < NKC; ++kc) {
             sc[kc] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int s = 0; s < 4; ++s) {
@@ -164,7 +165,7 @@ __global__ void flash_prefill_kernel(
         // ---- mask + online softmax ----
         float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < NKC; ++kc) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int qrow = q0g + wid * 16 + (lane >> 4) * 4 + r;
@@ -193,7 +194,7 @@ __global__ void flash_prefill_kernel(
             row_sum[r] = 0.f;
         }
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < NKC; ++kc) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 // masked elements contribute EXACTLY zero (whole-row-masked
@@ -219,15 +220,22 @@ __global__ void flash_prefill_kernel(
         }
         __syncthreads();
 
-        // ---- PV: O[16 q][128 d] += P[16x32] * V[32x128] ----
-        bf16x8_t pfrag = *reinterpret_cast<const bf16x8_t*>(
-            &p_lds[wid][lane & 15][(lane >> 4) * 8]);
+        // ---- PV: O[16 q][128 d] += P[16xKTILE] * V[KTILEx128] ----
+        bf16x8_t pfrag[NKC / 2];
+#pragma unroll
+        for (int ks = 0; ks < NKC / 2; ++ks)
+            pfrag[ks] = *reinterpret_cast<const bf16x8_t*>(
+                &p_lds[wid][lane & 15][ks * 32 + (lane >> 4) * 8]);
 #pragma unroll
         for (int d = 0; d < 8; ++d) {
-            // B fragment = V^T rows: one contiguous 16-B read per MFMA
-            bf16x8_t vfrag = *reinterpret_cast<const bf16x8_t*>(
-                &v_lds[d * 16 + (lane & 15)][(lane >> 4) * 8]);
-            o_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag, o_acc[d], 0, 0, 0);
+#pragma unroll
+            for (int ks = 0; ks < NKC / 2; ++ks) {
+                // B fragment = V^T rows: one contiguous 16-B read per MFMA
+                bf16x8_t vfrag = *reinterpret_cast<const bf16x8_t*>(
+                    &v_lds[d * 16 + (lane & 15)][ks * 32 + (lane >> 4) * 8]);
+                o_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag[ks], vfrag,
+                                                                   o_acc[d], 0, 0, 0);
+            }
         }
     }
 
