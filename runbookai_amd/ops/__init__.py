@@ -133,8 +133,8 @@ def _qtile_rows(hq: int, hk: int) -> int:
     (shared K/V tile) and 16-row blocks — group>=4 -> 16 rows, group 2 ->
     32, group 1 -> 64."""
     group = max(1, hq // hk)
-    heads_per_wg = min(group, 8)
-    return 16 * (8 // heads_per_wg)
+    heads_per_wg = min(group, 4)
+    return 16 * (4 // heads_per_wg)
 
 
 def _build_qtiles(seq_starts: torch.Tensor, qtile: int = 64):

@@ -61,7 +61,7 @@ DEVINL unsigned kswz(unsigned row, unsigned colb) {
 
 // ------------------------------------------------------------- flash kernel
 template <bool PAGED, bool CAUSAL>
-__launch_bounds__(512, 1)
+__launch_bounds__(256, 2)
 __global__ void flash_prefill_kernel(
     const ushort_t* __restrict__ q,
     const ushort_t* __restrict__ k,        // packed [T,Hk,D]  (non-paged)
@@ -80,8 +80,8 @@ __global__ void flash_prefill_kernel(
     const int tile = blockIdx.x;
     const int hk = blockIdx.y;
     const int group = Hq / Hk;
-    const int heads_per_wg = group < 8 ? group : 8;
-    const int row_blocks = 8 / heads_per_wg;     // 16-row blocks per wave set
+    const int heads_per_wg = group < 4 ? group : 4;
+    const int row_blocks = 4 / heads_per_wg;     // 16-row blocks per wave set
     const int b = tile_batch[tile];
     const int q0g = tile_qstart[tile];
     const int seg_start = seq_starts[b];
@@ -89,13 +89,13 @@ __global__ void flash_prefill_kernel(
     const int hist = PAGED ? hist_lens[b] : 0;
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x / WAVE;
-    const int h = hk * group + blockIdx.z * 8 + (wid % heads_per_wg);
+    const int h = hk * group + blockIdx.z * 4 + (wid % heads_per_wg);
     const int wrow0 = q0g + 16 * (wid / heads_per_wg);   // this wave's 16 q rows
     const int* bt = PAGED ? block_tables + (long)b * max_blocks : nullptr;
 
     __shared__ ushort_t k_lds[KTILE * DHEAD];        // swizzled rows
     __shared__ ushort_t v_lds[DHEAD][VPAD];          // transposed, padded
-    __shared__ ushort_t p_lds[8][16][VPAD];          // per-wave P staging
+    __shared__ ushort_t p_lds[4][16][VPAD];          // per-wave P staging
 
     const int my_qrow = wrow0 + (lane & 15);
     const bool row_valid = my_qrow < seg_end;
@@ -130,13 +130,13 @@ __global__ void flash_prefill_kernel(
     // latency hides under the QK/softmax/PV compute phase — the sync-staged
     // version stalls a full memory latency per tile at 2 waves/SIMD.
     // Each thread owns NCHUNK 16-B pieces of K and V.
-    constexpr int NCHUNK = KTILE * DHEAD / 8 / 512;
+    constexpr int NCHUNK = KTILE * DHEAD / 8 / 256;
     bf16x8_t skA[NCHUNK], svA[NCHUNK], skB[NCHUNK], svB[NCHUNK];
 
     auto load_tile = [&](int kt, bf16x8_t (&sk)[NCHUNK], bf16x8_t (&sv)[NCHUNK]) {
 #pragma unroll
         for (int c = 0; c < NCHUNK; ++c) {
-            const int idx = threadIdx.x + c * 512;
+            const int idx = threadIdx.x + c * 256;
             const int row = (idx * 8) / DHEAD;
             const int col = (idx * 8) % DHEAD;
             const int tok = kt * KTILE + row;
@@ -158,7 +158,7 @@ __global__ void flash_prefill_kernel(
     auto write_tile = [&](bf16x8_t (&sk)[NCHUNK], bf16x8_t (&sv)[NCHUNK]) {
 #pragma unroll
         for (int c = 0; c < NCHUNK; ++c) {
-            const int idx = threadIdx.x + c * 512;
+            const int idx = threadIdx.x + c * 256;
             const int row = (idx * 8) / DHEAD;
             const int col = (idx * 8) % DHEAD;
             *reinterpret_cast<bf16x8_t*>(
@@ -295,8 +295,8 @@ extern "C" void launch_flash_prefill(const void* q, const void* k, const void* v
                                      int n_tiles, int Hq, int Hk, float scale,
                                      int causal, hipStream_t stream) {
     const int group = Hq / Hk;
-    const int head_slices = (group + 7) / 8;
-    dim3 grid(n_tiles, Hk, head_slices), block(512);
+    const int head_slices = (group + 3) / 4;
+    dim3 grid(n_tiles, Hk, head_slices), block(256);
     if (causal) {
         hipLaunchKernelGGL((flash_prefill_kernel<false, true>), grid, block, 0, stream,
                            (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
@@ -318,8 +318,8 @@ extern "C" void launch_flash_prefill_paged(
     const void* hist_lens, void* out, int n_tiles, int Hq, int Hk, int BS,
     int max_blocks, float scale, hipStream_t stream) {
     const int group = Hq / Hk;
-    const int head_slices = (group + 7) / 8;
-    dim3 grid(n_tiles, Hk, head_slices), block(512);
+    const int head_slices = (group + 3) / 4;
+    dim3 grid(n_tiles, Hk, head_slices), block(256);
     hipLaunchKernelGGL((flash_prefill_kernel<true, true>), grid, block, 0, stream,
                        (const ushort_t*)q, nullptr, nullptr,
                        (const ushort_t*)kc, (const ushort_t*)vc,
