@@ -113,7 +113,7 @@ def prefill_attention(q, k, v, seq_starts, causal: bool = True,
     if _on_gpu(q):
         starts_i32 = seq_starts.to(torch.int32)
         if USE_FLASH_PREFILL and q.shape[-1] == 128:
-            tb, tq = _build_qtiles(starts_i32, qtile=_qtile_rows(q.shape[1], k.shape[1]))
+            tb, tq = _build_qtiles(starts_i32)
             return _get_ext().flash_prefill(q.contiguous(), k.contiguous(),
                                             v.contiguous(), tb.to(q.device),
                                             tq.to(q.device),
@@ -128,17 +128,8 @@ def prefill_attention(q, k, v, seq_starts, causal: bool = True,
     return reference.prefill_attention(q, k, v, seq_starts, causal, scale)
 
 
-def _qtile_rows(hq: int, hk: int) -> int:
-    """Q rows per flash workgroup: the 4 waves split between GQA q-heads
-    (shared K/V tile) and 16-row blocks — group>=4 -> 16 rows, group 2 ->
-    32, group 1 -> 64."""
-    group = max(1, hq // hk)
-    heads_per_wg = min(group, 4)
-    return 16 * (4 // heads_per_wg)
-
-
 def _build_qtiles(seq_starts: torch.Tensor, qtile: int = 64):
-    """Per-q-tile (segment id, global q start) arrays for the flash
+    """Per-64-row q-tile (segment id, global q start) arrays for the flash
     prefill grid; tiles never span segment boundaries."""
     starts = seq_starts.cpu().tolist()
     tb: list[int] = []
@@ -157,8 +148,7 @@ def chunked_prefill_attention(q, k_cache, v_cache, block_tables, hist_lens,
     """Packed new-token chunks attending over paged history + themselves."""
     scale = scale or 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
-        tb, tq = _build_qtiles(seq_starts.to(torch.int32),
-                               qtile=_qtile_rows(q.shape[1], k_cache.shape[1]))
+        tb, tq = _build_qtiles(seq_starts.to(torch.int32))
         return _get_ext().flash_prefill_paged(
             q.contiguous(), k_cache, v_cache,
             block_tables.to(torch.int32).contiguous(),

@@ -73,15 +73,9 @@ __global__ void flash_prefill_kernel(
     const int* __restrict__ tile_batch, const int* __restrict__ tile_qstart,
     const int* __restrict__ seq_starts, ushort_t* __restrict__ out,
     int Hq, int Hk, int BS, int max_blocks, float scale) {
-    // GQA-shared mapping: one workgroup per (q-tile, KV head, head-slice);
-    // the 4 waves cover the kv head's q-heads x 16-row blocks, so the K/V
-    // tile in LDS is read by the WHOLE GQA group (4x less HBM traffic than
-    // a workgroup-per-q-head layout at group 4).
     const int tile = blockIdx.x;
-    const int hk = blockIdx.y;
-    const int group = Hq / Hk;
-    const int heads_per_wg = group < 4 ? group : 4;
-    const int row_blocks = 4 / heads_per_wg;     // 16-row blocks per wave set
+    const int h = blockIdx.y;
+    const int hk = h / (Hq / Hk);
     const int b = tile_batch[tile];
     const int q0g = tile_qstart[tile];
     const int seg_start = seq_starts[b];
@@ -89,15 +83,13 @@ __global__ void flash_prefill_kernel(
     const int hist = PAGED ? hist_lens[b] : 0;
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x / WAVE;
-    const int h = hk * group + blockIdx.z * 4 + (wid % heads_per_wg);
-    const int wrow0 = q0g + 16 * (wid / heads_per_wg);   // this wave's 16 q rows
     const int* bt = PAGED ? block_tables + (long)b * max_blocks : nullptr;
 
     __shared__ ushort_t k_lds[KTILE * DHEAD];        // swizzled rows
     __shared__ ushort_t v_lds[DHEAD][VPAD];          // transposed, padded
     __shared__ ushort_t p_lds[4][16][VPAD];          // per-wave P staging
 
-    const int my_qrow = wrow0 + (lane & 15);
+    const int my_qrow = q0g + wid * 16 + (lane & 15);
     const bool row_valid = my_qrow < seg_end;
     bf16x8_t qfrag[4];
 #pragma unroll
@@ -119,7 +111,7 @@ __global__ void flash_prefill_kernel(
     // kv extent in LOCAL kv coordinates (0-based within this sequence's
     // attendable space: history + packed segment for PAGED, segment only
     // otherwise)
-    const int q_hi_local = min(q0g + 16 * row_blocks, seg_end) - 1 - seg_start;
+    const int q_hi_local = min(q0g + QTILE, seg_end) - 1 - seg_start;
     const int seg_len = seg_end - seg_start;
     const int kv_len = CAUSAL ? (hist + q_hi_local + 1)
                               : (hist + seg_len);
@@ -204,7 +196,7 @@ __global__ void flash_prefill_kernel(
         for (int kc = 0; kc < NKC; ++kc) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int qrow = wrow0 + (lane >> 4) * 4 + r;
+                const int qrow = q0g + wid * 16 + (lane >> 4) * 4 + r;
                 const int q_full = hist + (qrow - seg_start);
                 const int ktok = kv0 + kc * 16 + (lane & 15);
                 float sv = sc[kc][r] * scale;
@@ -278,7 +270,7 @@ __global__ void flash_prefill_kernel(
     // ---- epilogue: O / l, store bf16 ----
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-        const int qrow = wrow0 + (lane >> 4) * 4 + r;
+        const int qrow = q0g + wid * 16 + (lane >> 4) * 4 + r;
         if (qrow >= seg_end) continue;
         const float inv = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
 #pragma unroll
@@ -294,9 +286,7 @@ extern "C" void launch_flash_prefill(const void* q, const void* k, const void* v
                                      const void* seq_starts, void* out,
                                      int n_tiles, int Hq, int Hk, float scale,
                                      int causal, hipStream_t stream) {
-    const int group = Hq / Hk;
-    const int head_slices = (group + 3) / 4;
-    dim3 grid(n_tiles, Hk, head_slices), block(256);
+    dim3 grid(n_tiles, Hq), block(256);
     if (causal) {
         hipLaunchKernelGGL((flash_prefill_kernel<false, true>), grid, block, 0, stream,
                            (const ushort_t*)q, (const ushort_t*)k, (const ushort_t*)v,
@@ -317,9 +307,7 @@ extern "C" void launch_flash_prefill_paged(
     const void* tile_batch, const void* tile_qstart, const void* seq_starts,
     const void* hist_lens, void* out, int n_tiles, int Hq, int Hk, int BS,
     int max_blocks, float scale, hipStream_t stream) {
-    const int group = Hq / Hk;
-    const int head_slices = (group + 3) / 4;
-    dim3 grid(n_tiles, Hk, head_slices), block(256);
+    dim3 grid(n_tiles, Hq), block(256);
     hipLaunchKernelGGL((flash_prefill_kernel<true, true>), grid, block, 0, stream,
                        (const ushort_t*)q, nullptr, nullptr,
                        (const ushort_t*)kc, (const ushort_t*)vc,
