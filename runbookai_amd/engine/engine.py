@@ -241,13 +241,17 @@ class LLMEngine:
 
     # -- execution --------------------------------------------------------------------
 
-    def _tp_dispatch(self, payload: dict) -> torch.Tensor:
-        """Execute one model step; under TP, rank 0 broadcasts the payload so
-        every rank runs the same collectives in the same order."""
-        from ..parallel.tp_serving import broadcast_step, execute_step, tp_active
+    def _tp_active(self) -> bool:
+        from ..parallel.tp_serving import tp_active
 
-        if tp_active(self.model.tp):
-            broadcast_step(payload)
+        return tp_active(self.model.tp)
+
+    def _tp_dispatch(self, payload: dict) -> torch.Tensor:
+        """Broadcast + execute one model step (TP > 1 only): every rank runs
+        the same collectives in the same order."""
+        from ..parallel.tp_serving import broadcast_step, execute_step
+
+        broadcast_step(payload)
         return execute_step(self.model, payload)
 
     def _run_prefill(self, batch: list[Request]) -> None:
@@ -266,13 +270,17 @@ class LLMEngine:
             kv.set_len(req.rid, n)
             req.pos = n
             req.pending_input = []
-        logits = self._tp_dispatch({
-            "op": "prefill",
-            "token_ids": token_ids,
-            "positions": positions,
-            "seq_starts": starts,
-            "slots": torch.cat(slots).tolist(),
-        })
+        slots_t = torch.cat(slots)
+        if self._tp_active():
+            logits = self._tp_dispatch({
+                "op": "prefill", "token_ids": token_ids, "positions": positions,
+                "seq_starts": starts, "slots": slots_t.tolist(),
+            })
+        else:
+            logits = self.model.prefill(
+                torch.tensor(token_ids, dtype=torch.int64),
+                torch.tensor(positions, dtype=torch.int32),
+                torch.tensor(starts, dtype=torch.int32), slots_t)
         self.stats["prefill_tokens"] += len(token_ids)
         self._sample_and_advance(batch, logits)
         with self._lock:
@@ -303,16 +311,20 @@ class LLMEngine:
             req.pos += n
             req.pending_input = []
         bt, _lens = kv.batch_tables([r.rid for r in batch], "cpu")
-        logits = self._tp_dispatch({
-            "op": "chunk",
-            "token_ids": token_ids,
-            "positions": positions,
-            "seq_starts": starts,
-            "block_tables": bt.flatten().tolist(),
-            "bt_shape": list(bt.shape),
-            "hist_lens": hist,
-            "slots": torch.cat(slots).tolist(),
-        })
+        slots_t = torch.cat(slots)
+        if self._tp_active():
+            logits = self._tp_dispatch({
+                "op": "chunk", "token_ids": token_ids, "positions": positions,
+                "seq_starts": starts, "block_tables": bt.flatten().tolist(),
+                "bt_shape": list(bt.shape), "hist_lens": hist,
+                "slots": slots_t.tolist(),
+            })
+        else:
+            logits = self.model.chunk_step(
+                torch.tensor(token_ids, dtype=torch.int64),
+                torch.tensor(positions, dtype=torch.int32),
+                torch.tensor(starts, dtype=torch.int32), bt,
+                torch.tensor(hist, dtype=torch.int32), slots_t)
         self.stats["chunk_tokens"] = self.stats.get("chunk_tokens", 0) + len(token_ids)
         self._sample_and_advance(batch, logits)
         with self._lock:
@@ -334,15 +346,18 @@ class LLMEngine:
             req.pos += 1
             req.pending_input = []
         bt, lens = kv.batch_tables([r.rid for r in batch], "cpu")
-        logits = self._tp_dispatch({
-            "op": "decode",
-            "token_ids": input_ids,
-            "positions": positions,
-            "block_tables": bt.flatten().tolist(),
-            "bt_shape": list(bt.shape),
-            "seq_lens": lens.tolist(),
-            "slots": torch.cat(slot_list).tolist(),
-        })
+        slots_t = torch.cat(slot_list)
+        if self._tp_active():
+            logits = self._tp_dispatch({
+                "op": "decode", "token_ids": input_ids, "positions": positions,
+                "block_tables": bt.flatten().tolist(), "bt_shape": list(bt.shape),
+                "seq_lens": lens.tolist(), "slots": slots_t.tolist(),
+            })
+        else:
+            logits = self.model.decode(
+                torch.tensor(input_ids, dtype=torch.int64),
+                torch.tensor(positions, dtype=torch.int32),
+                bt, lens, slots_t)
         self.stats["decode_tokens"] += len(batch)
         self._sample_and_advance(batch, logits)
         with self._lock:
