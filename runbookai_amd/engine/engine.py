@@ -76,6 +76,9 @@ class Request:
     fsm: Optional[JsonFsm] = None
     out_ids: list[int] = field(default_factory=list)
     pending_input: list[int] = field(default_factory=list)  # tokens not yet in KV
+    prefill_tokens: list[int] = field(default_factory=list)  # prompt + forced prefix
+    cached_len: int = 0              # prefix tokens served from the KV prefix pool
+    registered: bool = False         # prompt blocks published to the prefix pool
     state: str = "waiting"           # waiting | running | done
     done_event: threading.Event = field(default_factory=threading.Event)
     prompt_len: int = 0
@@ -97,6 +100,7 @@ class LLMEngine:
         kv_blocks: Optional[int] = None,
         seed: int = 1234,
         background: bool = True,
+        prefix_cache: bool = True,
     ) -> None:
         cfg = CONFIGS[model]
         if device is None:
@@ -111,12 +115,14 @@ class LLMEngine:
         self.tokenizer = ByteTokenizer(cfg.vocab_size)
         self.max_prefill_tokens = max_prefill_tokens
         self.max_batch = max_batch
+        self.prefix_cache = prefix_cache
         self._rid = 0
         self._lock = threading.Condition()
         self.waiting: list[Request] = []
         self.running: list[Request] = []
         self.stats = {"requests": 0, "prefill_tokens": 0, "decode_tokens": 0,
-                      "steps": 0, "prefill_time": 0.0, "decode_time": 0.0}
+                      "steps": 0, "prefill_time": 0.0, "decode_time": 0.0,
+                      "cached_prefix_tokens": 0}
         self._stop = False
         self._thread: Optional[threading.Thread] = None
         if background:
@@ -148,6 +154,7 @@ class LLMEngine:
             # prompt prefill — zero decode steps spent on forced structure
             prefix = self._drain_forced(req)
             req.pending_input = req.prompt_ids + prefix
+            req.prefill_tokens = list(req.pending_input)
             self.waiting.append(req)
             self.stats["requests"] += 1
             self._lock.notify_all()
@@ -233,10 +240,26 @@ class LLMEngine:
                 break
             if not kv.can_allocate(need):
                 break
-            kv.allocate(req.rid, need)
+            cached = 0
+            if self.prefix_cache:
+                cached = kv.allocate_with_prefix(req.rid, req.pending_input, need)
+            else:
+                kv.allocate(req.rid, need)
             self.waiting.pop(0)
-            batch.append(req)
-            tokens += len(req.pending_input)
+            if cached:
+                # prefix served from the pool: skip straight to running with
+                # only the suffix pending — it flows through the chunked-
+                # prefill path (history attention over the shared blocks)
+                kv.set_len(req.rid, cached)
+                req.pos = cached
+                req.cached_len = cached
+                req.pending_input = req.pending_input[cached:]
+                req.state = "running"
+                self.running.append(req)
+                self.stats["cached_prefix_tokens"] += cached
+            else:
+                batch.append(req)
+                tokens += len(req.pending_input)
         return batch
 
     # -- execution --------------------------------------------------------------------
@@ -282,6 +305,7 @@ class LLMEngine:
                 torch.tensor(positions, dtype=torch.int32),
                 torch.tensor(starts, dtype=torch.int32), slots_t)
         self.stats["prefill_tokens"] += len(token_ids)
+        self._maybe_register(batch)
         self._sample_and_advance(batch, logits)
         with self._lock:
             for req in batch:
@@ -326,6 +350,7 @@ class LLMEngine:
                 torch.tensor(starts, dtype=torch.int32), bt,
                 torch.tensor(hist, dtype=torch.int32), slots_t)
         self.stats["chunk_tokens"] = self.stats.get("chunk_tokens", 0) + len(token_ids)
+        self._maybe_register(batch)
         self._sample_and_advance(batch, logits)
         with self._lock:
             self.running = [r for r in self.running if r.state != "done"]
@@ -359,10 +384,23 @@ class LLMEngine:
                 torch.tensor(positions, dtype=torch.int32),
                 bt, lens, slots_t)
         self.stats["decode_tokens"] += len(batch)
+        self._maybe_register(batch)
         self._sample_and_advance(batch, logits)
         with self._lock:
             self.running = [r for r in self.running if r.state != "done"]
         self.stats["decode_time"] += time.time() - t0
+
+    def _maybe_register(self, batch: list[Request]) -> None:
+        """Once a request's full prompt (prompt + forced prefix) is resident
+        in KV, publish its whole blocks to the prefix pool — BEFORE sampling,
+        which may finish the request and free its blocks."""
+        if not self.prefix_cache:
+            return
+        kv = self.model.kv
+        for req in batch:
+            if not req.registered and req.pos >= len(req.prefill_tokens):
+                kv.register_prefix(req.rid, req.prefill_tokens)
+                req.registered = True
 
     # -- sampling ---------------------------------------------------------------------
 
