@@ -42,6 +42,8 @@ def main() -> None:
     parser.add_argument("--model", default=None, help="tiny | llama3-8b | llama3-70b")
     parser.add_argument("--tp", type=int, default=1)
     parser.add_argument("--max-tokens", type=int, default=768)
+    parser.add_argument("--no-prefix-cache", action="store_true",
+                        help="disable shared-prompt KV reuse (A/B)")
     args = parser.parse_args()
 
     import torch
@@ -82,7 +84,8 @@ def main() -> None:
     if tp > 1:
         assert world == tp, f"--tp {tp} needs torchrun with {tp} ranks"
     engine = LLMEngine(model=model_name, device=device, tp=tp,
-                       kv_blocks=kv_blocks, background=(tp == 1 or rank == 0))
+                       kv_blocks=kv_blocks, background=(tp == 1 or rank == 0),
+                       prefix_cache=not args.no_prefix_cache)
     if tp > 1 and rank != 0:
         from runbookai_amd.parallel.tp_serving import run_follower_loop
 
