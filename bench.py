@@ -108,7 +108,8 @@ def main() -> None:
         score = score_investigation_result(result.to_dict(), case.get("expected", {}))
         return {"score": score["overall"],
                 "passed": score["overall"] >= fixtures.get("passThreshold", 0.7),
-                "success": result.success}
+                "success": result.success, "error": result.error,
+                "phases": len(result.phases_visited)}
 
     def run_batch(n: int) -> list[dict]:
         with ThreadPoolExecutor(max_workers=args.concurrency) as pool:
@@ -152,6 +153,14 @@ def main() -> None:
         elapsed = float(t.item())
 
     stats = engine.throughput_stats()
+    # surface investigation/engine errors on stderr — a run full of failed
+    # investigations is NOT a valid fast run
+    errors = sorted({r["error"] for r in results if r.get("error")})
+    if errors or stats.get("step_errors"):
+        print(f"[bench] {len([r for r in results if r.get('error')])} of "
+              f"{len(results)} investigations errored; engine step_errors="
+              f"{stats.get('step_errors', 0)} last={stats.get('last_error', '')!r} "
+              f"samples={errors[:3]}", file=sys.stderr)
     ms_per_step = elapsed * 1000.0 / args.steps
     total_investigations = args.steps if tp_mode else world * args.steps
     inv_per_hour = total_investigations / elapsed * 3600.0

@@ -124,6 +124,7 @@ class LLMEngine:
                       "steps": 0, "prefill_time": 0.0, "decode_time": 0.0,
                       "cached_prefix_tokens": 0}
         self._stop = False
+        self._step_batch: list[Request] = []   # requests in the executing step
         self._thread: Optional[threading.Thread] = None
         if background:
             self._thread = threading.Thread(target=self._loop, daemon=True,
@@ -186,20 +187,40 @@ class LLMEngine:
             try:
                 self.step()
             except Exception as e:  # noqa: BLE001 — fail requests, not the thread
-                with self._lock:
-                    for req in self.running + self.waiting:
-                        req.error = f"{type(e).__name__}: {e}"
-                        req.state = "done"
-                        req.done_event.set()
-                    self.running.clear()
-                    self.waiting.clear()
+                self._fail_step(e)
 
     def run_until_idle(self, max_steps: int = 100_000) -> None:
         for _ in range(max_steps):
             with self._lock:
                 if not self.waiting and not self.running:
                     return
-            self.step()
+            try:
+                self.step()
+            except Exception as e:  # noqa: BLE001
+                self._fail_step(e)
+
+    def _fail_step(self, e: Exception) -> None:
+        """Fail ONLY the requests touched by the failing step (their KV/FSM
+        state is suspect); other in-flight investigations keep running. The
+        step batch is in neither `waiting` nor `running` while it executes,
+        so it must be failed explicitly or its callers block until timeout."""
+        msg = f"{type(e).__name__}: {e}"
+        self.stats["step_errors"] = self.stats.get("step_errors", 0) + 1
+        self.stats["last_error"] = msg
+        if self.device.startswith("cuda"):
+            try:
+                torch.cuda.synchronize()
+            except Exception:  # noqa: BLE001 — device itself is gone
+                pass
+        with self._lock:
+            batch = self._step_batch
+            self._step_batch = []
+            for req in batch:
+                req.error = msg
+                req.state = "done"
+                self.model.kv.free(req.rid)
+                req.done_event.set()
+            self.running = [r for r in self.running if r not in batch]
 
     def shutdown(self) -> None:
         self._stop = True
@@ -218,15 +239,19 @@ class LLMEngine:
         with self._lock:
             prefill_batch = self._admit_locked()
         if prefill_batch:
+            self._step_batch = prefill_batch
             self._run_prefill(prefill_batch)
         else:
             with self._lock:
                 multis = [r for r in self.running if len(r.pending_input) > 1]
                 ones = [r for r in self.running if len(r.pending_input) == 1]
             if multis:
+                self._step_batch = multis
                 self._run_chunk(multis)
             elif ones:
+                self._step_batch = ones
                 self._run_decode(ones)
+        self._step_batch = []
         self.stats["steps"] += 1
 
     def _admit_locked(self) -> list[Request]:

@@ -186,11 +186,13 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
 
     Measured dispatch (profiles/PERF_LOG.md): the hand MFMA skinny kernel
     beats hipBLASLt on small-M x small-N decode projections (qkv/o shapes:
-    1.5-2.5 -> 2.7-3.0 TB/s); hipBLASLt already streams big-N shapes
-    (gate_up/down/lm_head) at 5.5-6.5 TB/s, so those stay on the library.
+    1.5-2.5 -> 2.7-3.0 TB/s) and matches it on gate_up (6.4 vs 6.5 TB/s)
+    — which also sidesteps a hipBLASLt HIPBLAS_STATUS_INTERNAL_ERROR seen
+    at m=28672, n=1..2 on some boxes; lm_head stays on the library
+    (5.5-6.5 TB/s on the huge-N shape).
     """
     if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 16
-            and weight.shape[0] <= 8192
+            and weight.shape[0] <= 28672
             and weight.shape[0] % 64 == 0 and weight.shape[1] % 64 == 0):
         return _get_ext().skinny_gemm(x.contiguous(), weight)
     return x @ weight.t()

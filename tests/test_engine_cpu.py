@@ -169,6 +169,33 @@ class TestEnginePressure:
         finally:
             eng.shutdown()
 
+    def test_step_failure_is_isolated(self):
+        """A model-step exception fails only the requests in that step —
+        other in-flight requests keep running and later submissions work
+        (one flaky library call must not take down 32 investigations)."""
+        eng = LLMEngine(model="tiny", device="cpu", background=True, kv_blocks=128)
+        try:
+            orig_prefill = eng.model.prefill
+            boom = {"armed": True}
+
+            def flaky(*a, **kw):
+                if boom.pop("armed", False):
+                    raise RuntimeError("transient library error")
+                return orig_prefill(*a, **kw)
+
+            eng.model.prefill = flaky
+            bad = eng.submit([65] * 40, max_new_tokens=8)
+            assert bad.done_event.wait(timeout=60)
+            assert "transient library error" in bad.error
+            assert eng.stats["step_errors"] == 1
+            # engine still serves fresh requests afterwards
+            good = eng.submit([66] * 40, max_new_tokens=8)
+            assert good.done_event.wait(timeout=60)
+            assert good.error == ""
+            assert good.out_ids
+        finally:
+            eng.shutdown()
+
     def test_oversized_prompt_clamped(self):
         eng = LLMEngine(model="tiny", device="cpu", background=False)
         try:

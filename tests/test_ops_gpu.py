@@ -246,6 +246,7 @@ class TestPagedDecode:
 class TestSkinnyGemm:
     @pytest.mark.parametrize("M,N,K", [
         (1, 4096, 4096), (8, 6144, 4096), (13, 4096, 14336),
+        (1, 28672, 4096), (2, 28672, 4096),   # gate_up decode (dispatch hot path)
         (32, 28672, 4096), (64, 128256, 4096),
         # tiny-model shapes (small K exercises short split-K ranges)
         (30, 512, 256), (30, 256, 256), (30, 1024, 256), (30, 256, 512),
