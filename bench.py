@@ -37,8 +37,9 @@ def main() -> None:
     parser.add_argument("--steps", type=int, default=2,
                         help="timed investigations per rank")
     parser.add_argument("--warmup", type=int, default=1)
-    parser.add_argument("--concurrency", type=int, default=8,
-                        help="investigations in flight per rank (continuous batching)")
+    parser.add_argument("--concurrency", type=int, default=32,
+                        help="investigations in flight per rank (continuous "
+                             "batching; BASELINE config 4 names 32)")
     parser.add_argument("--model", default=None, help="tiny | llama3-8b | llama3-70b")
     parser.add_argument("--tp", type=int, default=1)
     parser.add_argument("--max-tokens", type=int, default=768)
