@@ -57,7 +57,8 @@ class LocalEngineClient:
         import torch
 
         model = cfg.get("model", "llama3-8b")
-        if not torch.cuda.is_available() and model != "tiny":
+        if (not torch.cuda.is_available() and model != "tiny"
+                and not (cfg.get("checkpoint") or cfg.get("checkpoint_path"))):
             print(f"[runbook] no GPU visible: substituting the 'tiny' engine for "
                   f"'{model}' (CPU cannot serve an 8B/70B policy interactively)",
                   file=sys.stderr)
@@ -66,23 +67,43 @@ class LocalEngineClient:
             model=model,
             tp=cfg.get("tensor_parallel") or cfg.get("tensorParallel"),
             device=cfg.get("device"),
+            checkpoint=cfg.get("checkpoint") or cfg.get("checkpoint_path"),
         )
         return cls(engine, max_tokens=int(cfg.get("max_tokens", 1024)),
                    temperature=float(cfg.get("temperature", 0.0)))
+
+    # -- generation core -----------------------------------------------------------
+
+    def _tok(self):
+        """Checkpoint BPE tokenizer when the engine serves trained weights;
+        the byte tokenizer otherwise."""
+        return self.engine.hf_tokenizer or self.engine.tokenizer
+
+    def _gen(self, system: str, body: str,
+             schema: Optional[dict[str, Any]]) -> str:
+        """One generation. Byte-tokenizer engines enforce the schema at the
+        logits level (FSM masks); checkpoint engines get the schema as a
+        prompt instruction instead (the reference's parse-and-hope contract,
+        reference src/model/llm.ts prompt assembly) and the llm_parser
+        fallbacks absorb deviations."""
+        tok = self._tok()
+        if schema is not None and self.engine.hf_tokenizer is not None:
+            body = (body + "\n\nRespond with ONLY a JSON object matching this "
+                    "schema:\n" + json.dumps(schema))
+            schema = None
+        ids = tok.encode_chat(system, body)
+        req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
+                                   temperature=self.temperature, schema=schema)
+        return tok.decode(req.out_ids)
 
     # -- complete (JSON-disciplined; used by the orchestrator) --------------------
 
     def complete(self, prompt: str) -> str:
         kind, body = split_schema_tag(prompt)
         schema = PROMPT_SCHEMAS.get(kind) if kind else None
-        tok = self.engine.tokenizer
-        ids = tok.encode_chat(
+        return self._gen(
             "You are Runbook, an SRE agent. Respond with ONLY the requested JSON.",
-            body,
-        )
-        req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
-                                   temperature=self.temperature, schema=schema)
-        return tok.decode(req.out_ids)
+            body, schema)
 
     # -- chat (free-form with optional tool calls) --------------------------------
 
@@ -98,19 +119,16 @@ class LocalEngineClient:
              tools: Optional[list[dict[str, Any]]] = None) -> ChatResponse:
         kind, body = split_schema_tag(user)
         schema = PROMPT_SCHEMAS.get(kind) if kind else None
-        tok = self.engine.tokenizer
         if schema is None and tools:
             return self._chat_with_tools(system, body, tools)
-        ids = tok.encode_chat(system, body)
         if schema is not None:
-            req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
-                                       temperature=self.temperature, schema=schema)
-            return ChatResponse(content=tok.decode(req.out_ids))
-        # free text: bound it with the answer schema and unwrap
-        req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
-                                   temperature=self.temperature,
-                                   schema=self.ANSWER_SCHEMA)
-        text = tok.decode(req.out_ids)
+            return ChatResponse(content=self._gen(system, body, schema))
+        if self.engine.hf_tokenizer is not None:
+            # trained weights emit EOT on their own — no bounding schema
+            return ChatResponse(content=self._gen(system, body, None))
+        # free text on random-init weights: bound it with the answer schema
+        # (greedy decode would otherwise never emit EOT) and unwrap
+        text = self._gen(system, body, self.ANSWER_SCHEMA)
         try:
             from ..agent.llm_parser import parse_json
 
@@ -129,7 +147,6 @@ class LocalEngineClient:
         schema discipline at the logits level instead of parse-and-hope."""
         from ..agent.llm_parser import parse_json
 
-        tok = self.engine.tokenizer
         names = [t["name"] for t in tools][:32]
         decision_schema = {
             "type": "object",
@@ -140,11 +157,10 @@ class LocalEngineClient:
             },
             "required": ["thinking", "action", "tool"],
         }
-        ids = tok.encode_chat(system, user + "\n\nDecide: call a tool or give the final answer.")
-        req = self.engine.generate(ids, max_new_tokens=self.max_tokens,
-                                   temperature=self.temperature, schema=decision_schema)
+        out = self._gen(system, user + "\n\nDecide: call a tool or give the final answer.",
+                        decision_schema)
         try:
-            decision = parse_json(tok.decode(req.out_ids))
+            decision = parse_json(out)
         except Exception:  # noqa: BLE001
             decision = {"action": "final", "thinking": ""}
         thinking = str(decision.get("thinking", ""))
@@ -152,12 +168,11 @@ class LocalEngineClient:
             name = decision["tool"]
             spec = next(t for t in tools if t["name"] == name)
             args_schema = _sanitize_args_schema(spec.get("parameters", {}))
-            ids2 = tok.encode_chat(
-                system, f"{user}\n\nProvide arguments for the tool `{name}`.")
-            req2 = self.engine.generate(ids2, max_new_tokens=self.max_tokens,
-                                        temperature=self.temperature, schema=args_schema)
+            out2 = self._gen(system,
+                             f"{user}\n\nProvide arguments for the tool `{name}`.",
+                             args_schema)
             try:
-                args = parse_json(tok.decode(req2.out_ids))
+                args = parse_json(out2)
                 if not isinstance(args, dict):
                     args = {}
             except Exception:  # noqa: BLE001
@@ -165,11 +180,9 @@ class LocalEngineClient:
             call = ToolCall(id=new_id("call-"), name=name, arguments=args)
             return ChatResponse(content="", tool_calls=[call], thinking=thinking)
         # final answer
-        ids3 = tok.encode_chat(system, user + "\n\nGive the final answer.")
-        req3 = self.engine.generate(ids3, max_new_tokens=self.max_tokens,
-                                    temperature=self.temperature,
-                                    schema=self.ANSWER_SCHEMA)
-        text = tok.decode(req3.out_ids)
+        text = self._gen(system, user + "\n\nGive the final answer.",
+                         None if self.engine.hf_tokenizer is not None
+                         else self.ANSWER_SCHEMA)
         try:
             text = str(parse_json(text).get("answer", text))
         except Exception:  # noqa: BLE001
@@ -181,7 +194,7 @@ class LocalEngineClient:
         """Real streaming: yields text chunks as tokens are sampled."""
         import time
 
-        tok = self.engine.tokenizer
+        tok = self._tok()
         ids = tok.encode_chat(system, user)
         req = self.engine.submit(ids, max_new_tokens=self.max_tokens,
                                  temperature=self.temperature)

@@ -101,13 +101,33 @@ class LLMEngine:
         seed: int = 1234,
         background: bool = True,
         prefix_cache: bool = True,
+        checkpoint: Optional[str] = None,
     ) -> None:
-        cfg = CONFIGS[model]
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = device
+        self.hf_tokenizer = None
+        if checkpoint is not None:
+            # trained weights: config comes from the checkpoint's config.json;
+            # a tokenizer.json beside it switches the engine to full-vocab
+            # free decoding (byte-FSM grammar masks only apply to the byte
+            # tokenizer — see engine/bpe_tokenizer.py docstring)
+            import os as _os
+
+            from .checkpoint import load_model
+            self.model = load_model(checkpoint, device=device, tp=tp,
+                                    kv_blocks=kv_blocks, seed=seed)
+            cfg = self.model.cfg
+            tok_file = _os.path.join(checkpoint, "tokenizer.json")
+            if _os.path.exists(tok_file):
+                from .bpe_tokenizer import BpeTokenizer
+
+                self.hf_tokenizer = BpeTokenizer.from_file(tok_file)
+        else:
+            cfg = CONFIGS[model]
+            self.model = LlamaModel(cfg, device=device, tp=tp, seed=seed,
+                                    kv_blocks=kv_blocks)
         self.cfg = cfg
-        self.model = LlamaModel(cfg, device=device, tp=tp, seed=seed, kv_blocks=kv_blocks)
         if self.model.tp > 1:
             # every rank must replay identical step sequences; keep eager
             # (RCCL collectives inside hipGraph capture are not validated)
@@ -135,6 +155,11 @@ class LLMEngine:
 
     def submit(self, prompt_ids: list[int], max_new_tokens: int = 512,
                temperature: float = 0.0, schema: Optional[dict[str, Any]] = None) -> Request:
+        if schema is not None and self.hf_tokenizer is not None:
+            raise ValueError(
+                "grammar-constrained decoding runs on the byte tokenizer; with a "
+                "BPE checkpoint tokenizer, put the schema in the prompt instead "
+                "(LocalEngineClient does this automatically)")
         with self._lock:
             self._rid += 1
             # clamp to the model's context window (RoPE table bound)
@@ -434,7 +459,24 @@ class LLMEngine:
 
         All requests are masked to the active byte vocab (ids < MASK_REGION);
         schema'd requests are further masked to their FSM's allowed bytes.
+        With a checkpoint BPE tokenizer the whole vocabulary is live and
+        decoding is unmasked (schemas are rejected at submit).
         """
+        if self.hf_tokenizer is not None:
+            greedy = all(r.temperature <= 0.0 for r in batch)
+            if greedy:
+                chosen_t = logits.argmax(-1)
+            else:
+                temp = max(r.temperature for r in batch)
+                probs = torch.softmax(logits.float() / temp, dim=-1)
+                chosen_t = torch.multinomial(probs, 1).squeeze(-1)
+            now = time.time()
+            chosen_l = chosen_t.cpu().tolist()
+            for i, req in enumerate(batch):
+                if req.first_token_at == 0.0:
+                    req.first_token_at = now
+                self._advance_request(req, int(chosen_l[i]))
+            return
         region = logits[:, :MASK_REGION]
         mask = torch.zeros((len(batch), MASK_REGION), dtype=torch.bool)
         for i, req in enumerate(batch):
@@ -517,7 +559,11 @@ class LLMEngine:
                 if req.fsm.done:
                     finished = True
         else:
-            if tok in (SpecialTokens.EOS, SpecialTokens.EOT):
+            if self.hf_tokenizer is not None:
+                stops = {self.hf_tokenizer.eot_id}
+            else:
+                stops = {SpecialTokens.EOS, SpecialTokens.EOT}
+            if tok in stops:
                 finished = True
             else:
                 req.out_ids.append(tok)
@@ -551,7 +597,7 @@ _engines: dict[str, LLMEngine] = {}
 
 def get_engine(model: str = "tiny", **kwargs: Any) -> LLMEngine:
     """Process-wide engine cache (one engine per model)."""
-    key = f"{model}:{kwargs.get('tp')}:{kwargs.get('device')}"
+    key = f"{model}:{kwargs.get('tp')}:{kwargs.get('device')}:{kwargs.get('checkpoint')}"
     if key not in _engines:
         _engines[key] = LLMEngine(model=model, **kwargs)
     return _engines[key]

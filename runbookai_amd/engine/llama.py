@@ -385,12 +385,14 @@ class LlamaModel:
     # -- weights -----------------------------------------------------------------
 
     def load_safetensors(self, path: str) -> None:
-        """Hook for loading real Llama-3 checkpoints (safetensors shards).
-        Not exercised in this offline environment (no checkpoints on disk),
-        but keeps the deployment path explicit."""
-        raise NotImplementedError(
-            "no checkpoints available offline; weights are random-init by contract"
-        )
+        """Load a real Llama checkpoint (HF safetensors layout) over this
+        model's weights, sharded for (tp, rank) — see engine/checkpoint.py.
+        The benchmarks stay on random-init weights by contract (no
+        checkpoints ship in this offline image); round-trip correctness is
+        covered by tests/test_checkpoint.py."""
+        from .checkpoint import load_hf_checkpoint
+
+        load_hf_checkpoint(self, path)
 
 
 def param_count(cfg: LlamaConfig) -> int:

@@ -1,0 +1,277 @@
+"""Trained-weight loading: HF safetensors round-trip + BPE tokenizer.
+
+The loader (engine/checkpoint.py) is validated by exporting a random-init
+tiny model to the HF on-disk format and loading it back: logits must
+match exactly. The BPE tokenizer (engine/bpe_tokenizer.py) is validated
+against the `tokenizers` library on a ByteLevelBPE trained in-test —
+same ids on the same text, no network needed.
+"""
+from __future__ import annotations
+
+import json
+import os
+
+import pytest
+import torch
+
+from runbookai_amd.engine.checkpoint import (
+    config_from_hf,
+    export_hf_checkpoint,
+    load_model,
+)
+from runbookai_amd.engine.llama import CONFIGS, LlamaModel
+
+
+def tiny_model(seed: int = 7) -> LlamaModel:
+    return LlamaModel(CONFIGS["tiny"], device="cpu", seed=seed, kv_blocks=64)
+
+
+class TestCheckpointRoundTrip:
+    def test_export_then_load_matches_logits(self, tmp_path):
+        src = tiny_model(seed=7)
+        export_hf_checkpoint(src, str(tmp_path))
+        assert os.path.exists(tmp_path / "model.safetensors")
+        assert os.path.exists(tmp_path / "config.json")
+
+        dst = load_model(str(tmp_path), device="cpu", kv_blocks=64)
+        assert dst.cfg.hidden_size == src.cfg.hidden_size
+
+        T = 12
+        ids = torch.arange(T, dtype=torch.int64) + 3
+        pos = torch.arange(T, dtype=torch.int32)
+        starts = torch.tensor([0, T], dtype=torch.int32)
+        for m in (src, dst):
+            m.kv.allocate(1, T)
+        out_src = src.prefill(ids, pos, starts, src.kv.slot_mapping(1, 0, T))
+        out_dst = dst.prefill(ids, pos, starts, dst.kv.slot_mapping(1, 0, T))
+        assert torch.equal(out_src, out_dst)
+
+    def test_load_differs_from_fresh_random(self, tmp_path):
+        src = tiny_model(seed=7)
+        export_hf_checkpoint(src, str(tmp_path))
+        fresh = tiny_model(seed=8)
+        loaded = load_model(str(tmp_path), device="cpu", kv_blocks=64)
+        assert not torch.equal(fresh.layers[0].qkv.weight,
+                               loaded.layers[0].qkv.weight)
+        assert torch.equal(src.layers[0].qkv.weight,
+                           loaded.layers[0].qkv.weight)
+
+    def test_config_from_hf_reads_fields(self, tmp_path):
+        export_hf_checkpoint(tiny_model(), str(tmp_path))
+        cfg = config_from_hf(str(tmp_path))
+        tiny = CONFIGS["tiny"]
+        assert (cfg.hidden_size, cfg.intermediate_size, cfg.num_layers,
+                cfg.num_heads, cfg.num_kv_heads, cfg.head_dim,
+                cfg.vocab_size) == (
+            tiny.hidden_size, tiny.intermediate_size, tiny.num_layers,
+            tiny.num_heads, tiny.num_kv_heads, tiny.head_dim, tiny.vocab_size)
+
+    def test_tied_embeddings_fallback(self, tmp_path):
+        """Checkpoints without lm_head.weight tie it to the embedding."""
+        src = tiny_model()
+        export_hf_checkpoint(src, str(tmp_path))
+        from safetensors import safe_open
+        from safetensors.torch import save_file
+
+        fp = str(tmp_path / "model.safetensors")
+        with safe_open(fp, framework="pt") as f:
+            state = {k: f.get_tensor(k) for k in f.keys()
+                     if k != "lm_head.weight"}
+        save_file(state, fp)
+        loaded = load_model(str(tmp_path), device="cpu", kv_blocks=64)
+        assert torch.equal(loaded.lm_head.weight, loaded.embed.weight)
+
+    def test_sharded_index_layout(self, tmp_path):
+        """Multi-shard checkpoints with model.safetensors.index.json load
+        identically to the single-file layout."""
+        src = tiny_model(seed=11)
+        export_hf_checkpoint(src, str(tmp_path))
+        from safetensors import safe_open
+        from safetensors.torch import save_file
+
+        fp = str(tmp_path / "model.safetensors")
+        with safe_open(fp, framework="pt") as f:
+            state = {k: f.get_tensor(k) for k in f.keys()}
+        os.remove(fp)
+        keys = sorted(state)
+        half = len(keys) // 2
+        shards = {"model-00001-of-00002.safetensors": keys[:half],
+                  "model-00002-of-00002.safetensors": keys[half:]}
+        weight_map = {}
+        for fn, ks in shards.items():
+            save_file({k: state[k].contiguous() for k in ks}, str(tmp_path / fn))
+            weight_map.update({k: fn for k in ks})
+        with open(tmp_path / "model.safetensors.index.json", "w") as f:
+            json.dump({"weight_map": weight_map}, f)
+        loaded = load_model(str(tmp_path), device="cpu", kv_blocks=64)
+        assert torch.equal(src.layers[1].down.weight, loaded.layers[1].down.weight)
+
+    def test_tp_shard_slices_match_manual(self, tmp_path, monkeypatch):
+        """Rank-1-of-2 shard of every projection == manual slice of the
+        full tensors (the exact layout LlamaLayer builds)."""
+        src = tiny_model(seed=13)
+        export_hf_checkpoint(src, str(tmp_path))
+        import runbookai_amd.engine.llama as llama_mod
+        import runbookai_amd.parallel.dist as dist_mod
+        from runbookai_amd.engine import checkpoint as ckpt_mod
+
+        # llama.py binds get_world_size at import; patch both references
+        monkeypatch.setattr(llama_mod, "get_world_size", lambda: 2)
+        monkeypatch.setattr(dist_mod, "get_world_size", lambda: 2)
+        monkeypatch.setattr(dist_mod, "get_rank", lambda: 1)
+        sharded = LlamaModel(CONFIGS["tiny"], device="cpu", tp=2, kv_blocks=64)
+        ckpt_mod.load_hf_checkpoint(sharded, str(tmp_path))
+
+        cfg = src.cfg
+        d, hq, hk = cfg.head_dim, cfg.num_heads, cfg.num_kv_heads
+        hq_r, hk_r = hq // 2, hk // 2
+        H, inter = cfg.hidden_size, cfg.intermediate_size
+        full = src.layers[0].qkv.weight
+        q_full, k_full, v_full = full.split([hq * d, hk * d, hk * d], 0)
+        expect_q = q_full.view(hq, d, H)[hq_r:].reshape(hq_r * d, H)
+        expect_k = k_full.view(hk, d, H)[hk_r:].reshape(hk_r * d, H)
+        got = sharded.layers[0].qkv.weight
+        assert torch.equal(got[:hq_r * d], expect_q)
+        assert torch.equal(got[hq_r * d:hq_r * d + hk_r * d], expect_k)
+        # row-parallel down: rank 1 takes the second half of the columns
+        assert torch.equal(sharded.layers[0].down.weight,
+                           src.layers[0].down.weight[:, inter // 2:])
+        # column-parallel gate_up: [gate second half | up second half]
+        ipr = inter // 2
+        gu = src.layers[0].gate_up.weight
+        assert torch.equal(sharded.layers[0].gate_up.weight[:ipr], gu[ipr:inter])
+        assert torch.equal(sharded.layers[0].gate_up.weight[ipr:], gu[inter + ipr:])
+
+
+class TestCheckpointEngine:
+    @pytest.fixture()
+    def ckpt_dir(self, tmp_path):
+        export_hf_checkpoint(tiny_model(seed=21), str(tmp_path))
+        tokenizers = pytest.importorskip("tokenizers")
+        tok = tokenizers.ByteLevelBPETokenizer()
+        tok.train_from_iterator(["redis pool exhausted on checkout"] * 30,
+                                vocab_size=400, min_frequency=1,
+                                special_tokens=["<|eot_id|>"])
+        tok.save(str(tmp_path / "tokenizer.json"))
+        return str(tmp_path)
+
+    def test_engine_serves_checkpoint_free_decode(self, ckpt_dir):
+        from runbookai_amd.engine.client import LocalEngineClient
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(device="cpu", background=False, checkpoint=ckpt_dir,
+                        kv_blocks=64)
+        try:
+            assert eng.hf_tokenizer is not None
+            assert eng.cfg.hidden_size == CONFIGS["tiny"].hidden_size
+            # schemas are a byte-tokenizer feature: reject loudly
+            with pytest.raises(ValueError):
+                eng.submit([1, 2, 3], schema={"type": "object", "properties": {}})
+            client = LocalEngineClient(eng, max_tokens=8)
+            resp = client.chat("You are Runbook.", "status of checkout?")
+            assert isinstance(resp.content, str)
+            # schema'd complete routes the schema into the prompt (no FSM)
+            out = client.complete("what failed?")
+            assert isinstance(out, str)
+            assert eng.stats["requests"] >= 2
+        finally:
+            eng.shutdown()
+
+    def test_checkpoint_without_tokenizer_keeps_byte_mode(self, tmp_path):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        export_hf_checkpoint(tiny_model(seed=22), str(tmp_path))
+        eng = LLMEngine(device="cpu", background=False, checkpoint=str(tmp_path),
+                        kv_blocks=64)
+        try:
+            assert eng.hf_tokenizer is None
+            req = eng.generate(eng.tokenizer.encode("hi"), max_new_tokens=4,
+                               schema={"type": "object", "properties": {
+                                   "a": {"type": "number"}}, "required": ["a"]})
+            assert req.error == ""
+        finally:
+            eng.shutdown()
+
+
+class TestBpeTokenizer:
+    @pytest.fixture(scope="class")
+    def trained(self, tmp_path_factory):
+        tokenizers = pytest.importorskip("tokenizers")
+        corpus = [
+            "Redis connection pool exhausted on checkout-api",
+            "error rate spiked to 40% after deploy 2024-06-01",
+            "kubectl get pods -n prod | grep CrashLoopBackOff",
+            "The quick brown fox jumps over the lazy dog's tail, twice!",
+            "investigate high latency   between   services\n\nnow",
+        ] * 20
+        tok = tokenizers.ByteLevelBPETokenizer()
+        tok.train_from_iterator(corpus, vocab_size=600, min_frequency=2,
+                                special_tokens=["<|eot_id|>"])
+        path = tmp_path_factory.mktemp("bpe") / "tokenizer.json"
+        tok.save(str(path))
+        return tokenizers, str(path)
+
+    def test_parity_with_tokenizers_lib(self, trained):
+        tokenizers, path = trained
+        from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer
+
+        ours = BpeTokenizer.from_file(path)
+        theirs = tokenizers.Tokenizer.from_file(path)
+        samples = [
+            "Redis connection pool exhausted",
+            "error rate spiked to 40%",
+            "kubectl get pods -n prod",
+            "dog's tail, twice!",
+            "high latency   between   services",
+            "unseen wörds with ümlaute und 数字123",
+        ]
+        for text in samples:
+            assert ours.encode(text) == theirs.encode(text).ids, text
+
+    def test_decode_round_trip(self, trained):
+        _, path = trained
+        from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer
+
+        tok = BpeTokenizer.from_file(path)
+        for text in ["Redis pool exhausted!", "a\n\nb  c", "ünïcode 漢字"]:
+            assert tok.decode(tok.encode(text)) == text
+
+    def test_llama3_style_file_with_split_pattern(self, tmp_path):
+        """A tokenizer.json carrying its own Split regex (Llama-3 layout)
+        is honored; specials map through encode_chat."""
+        from runbookai_amd.engine.bpe_tokenizer import (
+            LLAMA3_PATTERN,
+            BpeTokenizer,
+            bytes_to_unicode,
+        )
+
+        b2u = bytes_to_unicode()
+        base = sorted({b2u[b] for b in range(256)})
+        vocab = {ch: i for i, ch in enumerate(base)}
+        he = b2u[ord("h")] + b2u[ord("e")]
+        vocab[he] = len(vocab)
+        specials_start = len(vocab)
+        data = {
+            "model": {"type": "BPE", "vocab": vocab,
+                      "merges": [f"{b2u[ord('h')]} {b2u[ord('e')]}"]},
+            "added_tokens": [
+                {"id": specials_start + i, "content": name}
+                for i, name in enumerate(
+                    ["<|begin_of_text|>", "<|start_header_id|>",
+                     "<|end_header_id|>", "<|eot_id|>"])],
+            "pre_tokenizer": {
+                "type": "Sequence",
+                "pretokenizers": [
+                    {"type": "Split", "pattern": {"Regex": LLAMA3_PATTERN},
+                     "behavior": "Isolated"}]},
+        }
+        path = tmp_path / "tokenizer.json"
+        path.write_text(json.dumps(data))
+        tok = BpeTokenizer.from_file(path)
+        ids = tok.encode("hello")
+        assert ids[0] == vocab[he]   # the merge applied
+        chat = tok.encode_chat("sys", "user text")
+        assert chat[0] == tok.special_tokens["<|begin_of_text|>"]
+        assert tok.special_tokens["<|eot_id|>"] in chat
+        assert tok.eot_id == tok.special_tokens["<|eot_id|>"]
+        assert tok._pat.pattern == LLAMA3_PATTERN
