@@ -27,6 +27,9 @@ class LlmConfig(BaseModel):
     temperature: float = 0.0
     tensor_parallel: int = Field(default=1, alias="tensorParallel")
     dtype: str = "bf16"
+    # HF-format checkpoint dir (safetensors [+ tokenizer.json]); empty =
+    # seeded random-init weights (the offline default)
+    checkpoint: str = Field(default="", alias="checkpointPath")
 
     model_config = {"populate_by_name": True, "extra": "allow", "protected_namespaces": ()}
 
