@@ -199,6 +199,12 @@ def main() -> None:
                 "decode_tok_per_s": round(stats.get("decode_tok_per_s", 0.0), 1),
                 "prefill_tok_per_s": round(stats.get("prefill_tok_per_s", 0.0), 1),
                 "llm_calls_total": stats.get("requests", 0),
+                "engine_prefill_s": round(stats.get("prefill_time", 0.0), 1),
+                "engine_decode_s": round(stats.get("decode_time", 0.0), 1),
+                "engine_steps": stats.get("steps", 0),
+                "decode_tokens": stats.get("decode_tokens", 0),
+                "chunk_tokens": stats.get("chunk_tokens", 0),
+                "cached_prefix_tokens": stats.get("cached_prefix_tokens", 0),
             },
         }
         print(json.dumps(line), flush=True)
