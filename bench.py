@@ -205,6 +205,12 @@ def main() -> None:
                 "decode_tokens": stats.get("decode_tokens", 0),
                 "chunk_tokens": stats.get("chunk_tokens", 0),
                 "cached_prefix_tokens": stats.get("cached_prefix_tokens", 0),
+                "t_decode_pre": round(stats.get("decode_pre_time", 0.0), 1),
+                "t_decode_model": round(stats.get("decode_model_time", 0.0), 1),
+                "t_chunk_pre": round(stats.get("chunk_pre_time", 0.0), 1),
+                "t_chunk_model": round(stats.get("chunk_model_time", 0.0), 1),
+                "t_sample": round(stats.get("sample_time", 0.0), 1),
+                "chunk_steps": stats.get("chunk_steps", 0),
             },
         }
         print(json.dumps(line), flush=True)

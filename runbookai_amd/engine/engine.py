@@ -386,6 +386,8 @@ class LLMEngine:
             req.pending_input = []
         bt, _lens = kv.batch_tables([r.rid for r in batch], "cpu")
         slots_t = torch.cat(slots)
+        self.stats["chunk_pre_time"] = (self.stats.get("chunk_pre_time", 0.0)
+                                        + time.time() - t0)
         if self._tp_active():
             logits = self._tp_dispatch({
                 "op": "chunk", "token_ids": token_ids, "positions": positions,
@@ -400,8 +402,14 @@ class LLMEngine:
                 torch.tensor(starts, dtype=torch.int32), bt,
                 torch.tensor(hist, dtype=torch.int32), slots_t)
         self.stats["chunk_tokens"] = self.stats.get("chunk_tokens", 0) + len(token_ids)
+        t1 = time.time()
+        self.stats["chunk_model_time"] = (self.stats.get("chunk_model_time", 0.0)
+                                          + t1 - t0)
+        self.stats["chunk_steps"] = self.stats.get("chunk_steps", 0) + 1
         self._maybe_register(batch)
         self._sample_and_advance(batch, logits)
+        self.stats["sample_time"] = (self.stats.get("sample_time", 0.0)
+                                     + time.time() - t1)
         with self._lock:
             self.running = [r for r in self.running if r.state != "done"]
         self.stats["decode_time"] += time.time() - t0
@@ -422,6 +430,8 @@ class LLMEngine:
             req.pending_input = []
         bt, lens = kv.batch_tables([r.rid for r in batch], "cpu")
         slots_t = torch.cat(slot_list)
+        self.stats["decode_pre_time"] = (self.stats.get("decode_pre_time", 0.0)
+                                         + time.time() - t0)
         if self._tp_active():
             logits = self._tp_dispatch({
                 "op": "decode", "token_ids": input_ids, "positions": positions,
@@ -434,8 +444,13 @@ class LLMEngine:
                 torch.tensor(positions, dtype=torch.int32),
                 bt, lens, slots_t)
         self.stats["decode_tokens"] += len(batch)
+        t1 = time.time()
+        self.stats["decode_model_time"] = (self.stats.get("decode_model_time", 0.0)
+                                           + t1 - t0)
         self._maybe_register(batch)
         self._sample_and_advance(batch, logits)
+        self.stats["sample_time"] = (self.stats.get("sample_time", 0.0)
+                                     + time.time() - t1)
         with self._lock:
             self.running = [r for r in self.running if r.state != "done"]
         self.stats["decode_time"] += time.time() - t0
