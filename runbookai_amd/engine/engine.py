@@ -132,6 +132,11 @@ class LLMEngine:
             # every rank must replay identical step sequences; keep eager
             # (RCCL collectives inside hipGraph capture are not validated)
             self.model.use_graphs = False
+        elif device.startswith("cuda"):
+            # capture every decode graph NOW, before request threads exist:
+            # a capture racing concurrent GPU work (knowledge search on the
+            # default stream) dies with hipErrorStreamCaptureUnsupported
+            self.model.capture_decode_graphs()
         self.tokenizer = ByteTokenizer(cfg.vocab_size)
         self.max_prefill_tokens = max_prefill_tokens
         self.max_batch = max_batch

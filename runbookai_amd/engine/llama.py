@@ -344,21 +344,47 @@ class LlamaModel:
                                   static["lens"], static["slots"])
         torch.cuda.current_stream().wait_stream(s)
         graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
+        # thread_local capture mode: OTHER threads (knowledge search /
+        # embedder on the default stream) may launch GPU work while this
+        # thread captures — global mode turns those into
+        # hipErrorStreamCaptureUnsupported and kills the batch
+        with torch.cuda.graph(graph, capture_error_mode="thread_local"):
             static["out"] = self._decode_impl(static["ids"], static["pos"], static["bt"],
                                               static["lens"], static["slots"])
         self._graphs[bpad] = (graph, static)
         return self._graphs[bpad]
 
+    def capture_decode_graphs(self, max_size: Optional[int] = None) -> int:
+        """Pre-capture the decode graphs (serving engines call this at init,
+        BEFORE request threads exist, so no capture happens mid-traffic)."""
+        if self.device == "cpu" or not self.use_graphs:
+            return 0
+        n = 0
+        for bpad in self.GRAPH_SIZES:
+            if max_size is not None and bpad > max_size:
+                break
+            self._ensure_graph(bpad)
+            n += 1
+        torch.cuda.synchronize()
+        return n
+
     def _decode_with_graph(self, token_ids, positions, block_tables, seq_lens, slots):
         B = token_ids.shape[0]
         bpad = next((s for s in self.GRAPH_SIZES if s >= B), None)
+        if bpad is not None and bpad not in self._graphs:
+            try:
+                self._ensure_graph(bpad)
+            except RuntimeError:
+                # capture raced with another thread's GPU work: run this
+                # batch eagerly and let a later quiet moment capture it
+                torch.cuda.synchronize()
+                bpad = None
         if bpad is None:
             dev = self.device
             return self._decode_impl(token_ids.to(dev), positions.to(dev),
                                      block_tables.to(dev), seq_lens.to(dev),
                                      slots.to(dev))
-        graph, static = self._ensure_graph(bpad)
+        graph, static = self._graphs[bpad]
         scratch_slot = self.kv.scratch_block * self.kv.block_size
         static["ids"][:B].copy_(token_ids, non_blocking=True)
         static["ids"][B:].zero_()
