@@ -185,14 +185,14 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """x [T, K] @ weight[N, K]^T.
 
     Measured dispatch (profiles/PERF_LOG.md): the hand MFMA skinny kernel
-    beats hipBLASLt on small-M x small-N decode projections (qkv/o shapes:
-    1.5-2.5 -> 2.7-3.0 TB/s) and matches it on gate_up (6.4 vs 6.5 TB/s)
-    — which also sidesteps a hipBLASLt HIPBLAS_STATUS_INTERNAL_ERROR seen
-    at m=28672, n=1..2 on some boxes; lm_head stays on the library
-    (5.5-6.5 TB/s on the huge-N shape).
+    beats hipBLASLt on the small-M qkv/o decode projections (1.8-2.6 ->
+    2.8-3.2 TB/s); hipBLASLt keeps gate_up (N=28672), down (K=14336) and
+    lm_head — a one-call probe that looked like parity on gate_up did not
+    reproduce across boxes (82 vs 36 us; see the dispatch-correction entry
+    in profiles/PERF_LOG.md), so the bound is both N<=8192 AND K<=8192.
     """
     if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 16
-            and weight.shape[0] <= 28672
+            and weight.shape[0] <= 8192 and weight.shape[1] <= 8192
             and weight.shape[0] % 64 == 0 and weight.shape[1] % 64 == 0):
         return _get_ext().skinny_gemm(x.contiguous(), weight)
     return x @ weight.t()
