@@ -261,11 +261,21 @@ class LLMEngine:
 
     # -- scheduling ------------------------------------------------------------------
 
+    #: pending runs longer than this go through the chunked-prefill path;
+    #: shorter runs (grammar-forced bytes, word-token tails) ride the
+    #: graphed decode batch one token per iteration — an eager chunk call
+    #: costs ~13 ms of CPU launch time (32 layers, no hipGraph) while the
+    #: whole decode batch advances alongside for free (measured:
+    #: profiles/PERF_LOG.md, scheduling section)
+    CHUNK_THRESHOLD = 8
+
     def step(self) -> None:
         """One engine iteration: admit a prefill batch if any request is
-        waiting (and fits); else flush multi-token chunks (grammar-forced
-        runs) through the chunked-prefill path; else one decode step over
-        all running sequences."""
+        waiting (and fits); else, if any request has a LONG pending run
+        (admission suffix after a prefix-cache hit, long forced run), flush
+        all pending runs through the chunked-prefill path; else one decode
+        step over all running sequences, each feeding the next pending
+        token (sampling only where the queue drained)."""
         with self._lock:
             prefill_batch = self._admit_locked()
         if prefill_batch:
@@ -273,14 +283,15 @@ class LLMEngine:
             self._run_prefill(prefill_batch)
         else:
             with self._lock:
-                multis = [r for r in self.running if len(r.pending_input) > 1]
-                ones = [r for r in self.running if len(r.pending_input) == 1]
-            if multis:
-                self._step_batch = multis
-                self._run_chunk(multis)
-            elif ones:
-                self._step_batch = ones
-                self._run_decode(ones)
+                pending = [r for r in self.running if r.pending_input]
+                long_run = any(len(r.pending_input) > self.CHUNK_THRESHOLD
+                               for r in pending)
+            if long_run:
+                self._step_batch = pending
+                self._run_chunk(pending)
+            elif pending:
+                self._step_batch = pending
+                self._run_decode(pending)
         self._step_batch = []
         self.stats["steps"] += 1
 
@@ -426,13 +437,15 @@ class LLMEngine:
         positions = []
         slot_list = []
         for req in batch:
+            # feed ONE token from each request's pending queue; requests
+            # mid-run (forced bytes still queued) skip sampling this step
             input_ids.append(req.pending_input[0])
             positions.append(req.pos)
             kv.extend(req.rid, req.pos + 1)
             slot_list.append(kv.slot_mapping(req.rid, req.pos, 1))
             kv.set_len(req.rid, req.pos + 1)
             req.pos += 1
-            req.pending_input = []
+            req.pending_input = req.pending_input[1:]
         bt, lens = kv.batch_tables([r.rid for r in batch], "cpu")
         slots_t = torch.cat(slot_list)
         self.stats["decode_pre_time"] = (self.stats.get("decode_pre_time", 0.0)
@@ -453,7 +466,12 @@ class LLMEngine:
         self.stats["decode_model_time"] = (self.stats.get("decode_model_time", 0.0)
                                            + t1 - t0)
         self._maybe_register(batch)
-        self._sample_and_advance(batch, logits)
+        drained = [i for i, r in enumerate(batch) if not r.pending_input]
+        if len(drained) == len(batch):
+            self._sample_and_advance(batch, logits)
+        elif drained:
+            idx = torch.tensor(drained, device=logits.device)
+            self._sample_and_advance([batch[i] for i in drained], logits[idx])
         self.stats["sample_time"] = (self.stats.get("sample_time", 0.0)
                                      + time.time() - t1)
         with self._lock:
