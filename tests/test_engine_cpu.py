@@ -169,6 +169,31 @@ class TestEnginePressure:
         finally:
             eng.shutdown()
 
+    def test_short_run_decode_matches_chunk_path(self, monkeypatch):
+        """Forced runs <=CHUNK_THRESHOLD ride the decode batch one token
+        per iteration (mid-run requests skip sampling); forcing EVERY run
+        through the chunk path instead must yield identical outputs."""
+        from runbookai_amd.agent.llm_parser import PROMPT_SCHEMAS
+
+        def run(threshold):
+            monkeypatch.setattr(LLMEngine, "CHUNK_THRESHOLD", threshold)
+            eng = LLMEngine(model="tiny", device="cpu", background=False,
+                            kv_blocks=256, prefix_cache=False)
+            try:
+                tok = eng.tokenizer
+                reqs = [eng.submit(tok.encode_chat("sys", f"case {i}"),
+                                   max_new_tokens=64,
+                                   schema=PROMPT_SCHEMAS["generateConclusion"])
+                        for i in range(3)]
+                eng.run_until_idle()
+                assert all(r.error == "" for r in reqs)
+                return [r.out_ids for r in reqs]
+            finally:
+                eng.shutdown()
+
+        # threshold 0 = every run chunks (old behavior); 8 = decode rides
+        assert run(0) == run(8)
+
     def test_step_failure_is_isolated(self):
         """A model-step exception fails only the requests in that step —
         other in-flight requests keep running and later submissions work
