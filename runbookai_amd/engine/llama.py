@@ -171,6 +171,8 @@ class LlamaModel:
         self.use_graphs = (str(device).startswith("cuda")
                            and _os.environ.get("RUNBOOKAI_NO_GRAPHS", "0") != "1")
         self._graphs: dict[int, tuple] = {}
+        self._chunk_graphs: dict[tuple, tuple] = {}
+        self._chunk_pool = None   # shared mempool across chunk graphs
 
     # -- setup -------------------------------------------------------------------
 
@@ -248,25 +250,57 @@ class LlamaModel:
         """Multi-token append for running sequences: each chunk attends over
         its paged history + itself. Returns logits at each chunk's LAST
         token: [B, vocab]. GPU path requires head_dim 128 (Llama); other
-        dims fall back to sequential decode steps."""
+        dims fall back to sequential decode steps. On GPU the step replays
+        as a hipGraph bucketed by (padded tokens, padded segments) — the
+        eager path costs ~13 ms of CPU launch time per call."""
         device = self.device
-        if (str(device).startswith("cuda") and self.cfg.head_dim != 128):
+        if not str(device).startswith("cuda"):
+            token_ids = token_ids.long()
+            seq_starts_d = seq_starts.to(torch.int32)
+            hist = hist_lens.to(torch.int32)
+            slots = slot_mapping.to(torch.int32)
+            h = self.embed.weight[token_ids]
+
+            def attn_fn(q, k, v, i):
+                return ops.chunked_prefill_attention(q, self.kv.k[i], self.kv.v[i],
+                                                     block_tables, hist,
+                                                     seq_starts_d, self.scale)
+
+            normed = self._transformer_body(h, positions.to(torch.int32), slots,
+                                            attn_fn, fused_kv=True)
+            return self.lm_head(normed[(seq_starts[1:] - 1).long()])
+        if self.cfg.head_dim != 128:
             return self._chunk_by_decode(token_ids, positions, seq_starts,
                                          block_tables, hist_lens, slot_mapping)
-        token_ids = token_ids.long().to(device)
-        positions = positions.to(torch.int32).to(device)
-        seq_starts_d = seq_starts.to(torch.int32).to(device)
-        bt = block_tables.to(device)
-        hist = hist_lens.to(torch.int32).to(device)
-        slots = slot_mapping.to(torch.int32).to(device)
+        if self.use_graphs:
+            out = self._chunk_with_graph(token_ids, positions, seq_starts,
+                                         block_tables, hist_lens, slot_mapping)
+            if out is not None:
+                return out
+        # eager flash path: tiles built ONCE per call (not per layer)
+        tb, tq = ops._build_qtiles(seq_starts.to(torch.int32))
+        return self._chunk_impl(
+            token_ids.long().to(device),
+            positions.to(torch.int32).to(device),
+            seq_starts.to(torch.int32).to(device),
+            block_tables.to(device),
+            hist_lens.to(torch.int32).to(device),
+            slot_mapping.to(torch.int32).to(device),
+            tb.to(device), tq.to(device),
+            (seq_starts[1:] - 1).long().to(device))
+
+    def _chunk_impl(self, token_ids, positions, seq_starts_d, bt, hist, slots,
+                    tb, tq, last):
+        """Device-tensor chunk body (hipGraph-capturable: all inputs are
+        device tensors, attention takes prebuilt tiles)."""
         h = self.embed.weight[token_ids]
 
         def attn_fn(q, k, v, i):
-            return ops.chunked_prefill_attention(q, self.kv.k[i], self.kv.v[i],
-                                                 bt, hist, seq_starts_d, self.scale)
+            return ops.chunked_prefill_attention_tiles(
+                q, self.kv.k[i], self.kv.v[i], bt, tb, tq, seq_starts_d, hist,
+                self.scale)
 
         normed = self._transformer_body(h, positions, slots, attn_fn, fused_kv=True)
-        last = (seq_starts[1:] - 1).long().to(device)
         return self.lm_head(normed[last])
 
     def _chunk_by_decode(self, token_ids, positions, seq_starts, block_tables,
@@ -365,6 +399,12 @@ class LlamaModel:
                 break
             self._ensure_graph(bpad)
             n += 1
+        if self.cfg.head_dim == 128:
+            # common chunk buckets (admission suffixes land here); rarer
+            # (T, B) combinations capture lazily with an eager fallback
+            for tpad in self.CHUNK_T_BUCKETS:
+                self._ensure_chunk_graph(tpad, self.CHUNK_B_BUCKETS[0])
+                n += 1
         torch.cuda.synchronize()
         return n
 
@@ -397,6 +437,124 @@ class LlamaModel:
         static["lens"][B:].fill_(1)
         static["slots"][:B].copy_(slots, non_blocking=True)
         static["slots"][B:].fill_(scratch_slot)
+        graph.replay()
+        return static["out"][:B]
+
+    # -- hipGraph chunk steps -----------------------------------------------------
+    #
+    # Chunk shapes vary per call, so graphs are bucketed by (T_pad, B_pad):
+    # real tokens/segments fill the static buffers, rows [T_real, T_pad)
+    # form one extra padding segment (hist 0, scratch slots — keeps every
+    # attention output row finite), leftover segment slots are empty
+    # (start == end) and leftover tile slots point at an empty segment, so
+    # the kernel's row/kv guards skip them. ~13 ms of eager CPU launch
+    # time becomes one graph replay.
+
+    CHUNK_T_BUCKETS = (64, 128, 256, 512, 1024, 2048)
+    CHUNK_B_BUCKETS = (4, 8, 16, 32)
+
+    def _ensure_chunk_graph(self, tpad: int, bpad: int):
+        key = (tpad, bpad)
+        if key in self._chunk_graphs:
+            return self._chunk_graphs[key]
+        dev = self.device
+        maxb = self._graph_max_blocks()
+        ntiles = tpad // 64 + bpad
+        scratch_slot = self.kv.scratch_block * self.kv.block_size
+        starts = torch.full((bpad + 1,), tpad, dtype=torch.int32, device=dev)
+        starts[0] = 0   # warmup layout: segment 0 covers all rows, rest empty
+        tb = torch.full((ntiles,), bpad - 1, dtype=torch.int32, device=dev)
+        tq = torch.full((ntiles,), tpad, dtype=torch.int32, device=dev)
+        for i in range(tpad // 64):
+            tb[i] = 0
+            tq[i] = i * 64
+        static = {
+            "ids": torch.zeros(tpad, dtype=torch.long, device=dev),
+            "pos": torch.zeros(tpad, dtype=torch.int32, device=dev),
+            "starts": starts,
+            "bt": torch.full((bpad, maxb), self.kv.scratch_block,
+                             dtype=torch.int32, device=dev),
+            "hist": torch.zeros(bpad, dtype=torch.int32, device=dev),
+            "slots": torch.full((tpad,), scratch_slot, dtype=torch.int32, device=dev),
+            "tb": tb,
+            "tq": tq,
+            "last": torch.zeros(bpad, dtype=torch.long, device=dev),
+        }
+
+        def run():
+            return self._chunk_impl(static["ids"], static["pos"], static["starts"],
+                                    static["bt"], static["hist"], static["slots"],
+                                    static["tb"], static["tq"], static["last"])
+
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                run()
+        torch.cuda.current_stream().wait_stream(s)
+        if self._chunk_pool is None:
+            self._chunk_pool = torch.cuda.graph_pool_handle()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph, pool=self._chunk_pool,
+                              capture_error_mode="thread_local"):
+            static["out"] = run()
+        self._chunk_graphs[key] = (graph, static)
+        return self._chunk_graphs[key]
+
+    def _chunk_with_graph(self, token_ids, positions, seq_starts, block_tables,
+                          hist_lens, slot_mapping):
+        """Replay a bucketed chunk graph; returns None when the call does
+        not fit any bucket (caller falls back to the eager flash path)."""
+        T = int(token_ids.shape[0])
+        B = int(seq_starts.shape[0]) - 1
+        tpad = next((t for t in self.CHUNK_T_BUCKETS if t >= T), None)
+        bpad = next((b for b in self.CHUNK_B_BUCKETS if b >= B + 1), None)
+        if tpad is None or bpad is None:
+            return None
+        key = (tpad, bpad)
+        if key not in self._chunk_graphs:
+            try:
+                self._ensure_chunk_graph(tpad, bpad)
+            except RuntimeError:
+                torch.cuda.synchronize()
+                return None
+        graph, static = self._chunk_graphs[key]
+        scratch_slot = self.kv.scratch_block * self.kv.block_size
+        # token rows
+        static["ids"][:T].copy_(token_ids.long(), non_blocking=True)
+        static["ids"][T:].zero_()
+        static["pos"][:T].copy_(positions.to(torch.int32), non_blocking=True)
+        static["pos"][T:].zero_()
+        static["slots"][:T].copy_(slot_mapping.to(torch.int32), non_blocking=True)
+        static["slots"][T:].fill_(scratch_slot)
+        # segments: B real, one padding segment [T, tpad), rest empty
+        starts_host = torch.full((bpad + 1,), tpad, dtype=torch.int32)
+        starts_host[:B + 1] = seq_starts.to(torch.int32)   # [..., T]
+        static["starts"].copy_(starts_host, non_blocking=True)
+        static["hist"][:B].copy_(hist_lens.to(torch.int32), non_blocking=True)
+        static["hist"][B:].zero_()
+        static["bt"].fill_(self.kv.scratch_block)
+        nb = block_tables.shape[1]
+        static["bt"][:B, :nb].copy_(block_tables, non_blocking=True)
+        last_host = torch.zeros(bpad, dtype=torch.long)
+        last_host[:B] = (seq_starts[1:] - 1).long()
+        static["last"].copy_(last_host, non_blocking=True)
+        # tiles: real segments, then the padding segment, then inert fillers
+        starts_l = seq_starts.tolist()
+        tb_host = torch.full_like(static["tb"], bpad - 1, device="cpu")
+        tq_host = torch.full_like(static["tq"], tpad, device="cpu")
+        n = 0
+        for b in range(B):
+            for q0 in range(starts_l[b], starts_l[b + 1], 64):
+                tb_host[n] = b
+                tq_host[n] = q0
+                n += 1
+        for q0 in range(T, tpad, 64):
+            tb_host[n] = B                       # padding segment
+            tq_host[n] = q0
+            n += 1
+        static["tb"].copy_(tb_host, non_blocking=True)
+        static["tq"].copy_(tq_host, non_blocking=True)
         graph.replay()
         return static["out"][:B]
 

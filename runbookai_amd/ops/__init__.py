@@ -159,6 +159,16 @@ def chunked_prefill_attention(q, k_cache, v_cache, block_tables, hist_lens,
                                                hist_lens, seq_starts, scale)
 
 
+def chunked_prefill_attention_tiles(q, k_cache, v_cache, block_tables, tb, tq,
+                                    seq_starts_dev, hist_lens_dev, scale: float):
+    """Device-tile variant of chunked_prefill_attention: every argument is
+    already a device tensor (tiles included), so the call is hipGraph-
+    capturable — no host-side tile building or H2D copies."""
+    return _get_ext().flash_prefill_paged(q, k_cache, v_cache, block_tables,
+                                          tb, tq, seq_starts_dev, hist_lens_dev,
+                                          scale)
+
+
 def paged_decode_attention(q, k_cache, v_cache, block_tables, seq_lens,
                            scale: Optional[float] = None):
     scale = scale or 1.0 / math.sqrt(q.shape[-1])

@@ -381,6 +381,40 @@ class TestEngineGpu:
         finally:
             eng.shutdown()
 
+    def test_chunk_graph_matches_eager(self):
+        """Bucketed hipGraph chunk replay == eager flash chunk path on the
+        8B model (same kernels, static-buffer padding must not leak into
+        real rows)."""
+        from runbookai_amd.engine.llama import CONFIGS, LlamaModel
+
+        m = LlamaModel(CONFIGS["llama3-8b"], device=DEV, kv_blocks=256)
+        hist_lens = [40, 100, 7]
+        new_lens = [5, 90, 17]          # T=112 (unaligned), B=3 -> bucket (128, 4)
+        for s, (h, n) in enumerate(zip(hist_lens, new_lens), start=1):
+            m.kv.allocate(s, h + n + 8)
+            ids = torch.randint(0, 255, (h,))
+            m.prefill(ids, torch.arange(h, dtype=torch.int32),
+                      torch.tensor([0, h], dtype=torch.int32),
+                      m.kv.slot_mapping(s, 0, h))
+            m.kv.set_len(s, h)
+        token_ids = torch.randint(0, 255, (sum(new_lens),))
+        positions = torch.cat([torch.arange(h, h + n, dtype=torch.int32)
+                               for h, n in zip(hist_lens, new_lens)])
+        starts = torch.tensor([0, 5, 95, 112], dtype=torch.int32)
+        bt, _ = m.kv.batch_tables([1, 2, 3], "cpu")
+        hist = torch.tensor(hist_lens, dtype=torch.int32)
+        slots = torch.cat([m.kv.slot_mapping(s, h, n)
+                           for s, (h, n) in enumerate(zip(hist_lens, new_lens), 1)])
+        m.use_graphs = True
+        out_graph = m.chunk_step(token_ids, positions, starts, bt, hist, slots)
+        assert (128, 4) in m._chunk_graphs, "graph bucket was not used"
+        m.use_graphs = False
+        out_eager = m.chunk_step(token_ids, positions, starts, bt, hist, slots)
+        a, b = out_graph.float().cpu(), out_eager.float().cpu()
+        rel = (a - b).norm().item() / max(a.norm().item(), 1e-6)
+        assert rel < 1e-2, f"graph vs eager chunk rel-diff {rel}"
+        assert torch.equal(a.argmax(-1), b.argmax(-1))
+
     def test_gpu_matches_cpu_tiny_prefill(self):
         """Same seed tiny model: GPU logits ≈ CPU logits (bf16 tolerance)."""
         from runbookai_amd.engine.llama import CONFIGS, LlamaModel
