@@ -170,6 +170,7 @@ class LlamaModel:
 
         self.use_graphs = (str(device).startswith("cuda")
                            and _os.environ.get("RUNBOOKAI_NO_GRAPHS", "0") != "1")
+        self.use_chunk_graphs = _os.environ.get("RUNBOOKAI_CHUNK_GRAPHS", "1") != "0"
         self._graphs: dict[int, tuple] = {}
         self._chunk_graphs: dict[tuple, tuple] = {}
         self._chunk_pool = None   # shared mempool across chunk graphs
@@ -272,7 +273,7 @@ class LlamaModel:
         if self.cfg.head_dim != 128:
             return self._chunk_by_decode(token_ids, positions, seq_starts,
                                          block_tables, hist_lens, slot_mapping)
-        if self.use_graphs:
+        if self.use_graphs and self.use_chunk_graphs:
             out = self._chunk_with_graph(token_ids, positions, seq_starts,
                                          block_tables, hist_lens, slot_mapping)
             if out is not None:
@@ -399,7 +400,7 @@ class LlamaModel:
                 break
             self._ensure_graph(bpad)
             n += 1
-        if self.cfg.head_dim == 128:
+        if self.cfg.head_dim == 128 and self.use_chunk_graphs:
             # common chunk buckets (admission suffixes land here); rarer
             # (T, B) combinations capture lazily with an eager fallback
             for tpad in self.CHUNK_T_BUCKETS:
@@ -450,7 +451,11 @@ class LlamaModel:
     # the kernel's row/kv guards skip them. ~13 ms of eager CPU launch
     # time becomes one graph replay.
 
-    CHUNK_T_BUCKETS = (64, 128, 256, 512, 1024, 2048)
+    # Graphs only where CPU launch time dominates GPU compute: a T=512
+    # chunk is ~10 ms of GPU work vs ~13 ms eager launch cost, so beyond
+    # 512 the eager path overlaps fine and padding waste would dominate.
+    # Finer buckets bound padding waste to ~25%.
+    CHUNK_T_BUCKETS = (64, 128, 192, 256, 384, 512)
     CHUNK_B_BUCKETS = (4, 8, 16, 32)
 
     def _ensure_chunk_graph(self, tpad: int, bpad: int):
