@@ -226,9 +226,20 @@ class LlamaModel:
         batch_idx = self._batch_idx(seq_starts, token_ids.shape[0]).to(device)
         h = self.embed.weight[token_ids.long()]
 
-        def attn_fn(q, k, v, i):
-            return ops.prefill_attention(q, k, v, seq_starts_d, causal=True,
-                                         scale=self.scale, batch_idx=batch_idx)
+        use_flash = (str(device).startswith("cuda") and ops.USE_FLASH_PREFILL
+                     and self.cfg.head_dim == 128)
+        if use_flash:
+            # tiles built ONCE per call, not per layer
+            tb, tq = ops._build_qtiles(seq_starts.to(torch.int32))
+            tb, tq = tb.to(device), tq.to(device)
+
+            def attn_fn(q, k, v, i):
+                return ops.prefill_attention_tiles(q, k, v, tb, tq, seq_starts_d,
+                                                   self.scale, causal=True)
+        else:
+            def attn_fn(q, k, v, i):
+                return ops.prefill_attention(q, k, v, seq_starts_d, causal=True,
+                                             scale=self.scale, batch_idx=batch_idx)
 
         normed = self._transformer_body(h, positions, slots, attn_fn)
         last = (seq_starts[1:] - 1).long().to(device)

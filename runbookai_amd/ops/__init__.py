@@ -159,6 +159,14 @@ def chunked_prefill_attention(q, k_cache, v_cache, block_tables, hist_lens,
                                                hist_lens, seq_starts, scale)
 
 
+def prefill_attention_tiles(q, k, v, tb, tq, seq_starts_dev, scale: float,
+                            causal: bool = True):
+    """Device-tile variant of flash prefill (D=128): tiles prebuilt once
+    per model call instead of once per layer."""
+    return _get_ext().flash_prefill(q.contiguous(), k.contiguous(), v.contiguous(),
+                                    tb, tq, seq_starts_dev, scale, causal)
+
+
 def chunked_prefill_attention_tiles(q, k_cache, v_cache, block_tables, tb, tq,
                                     seq_starts_dev, hist_lens_dev, scale: float):
     """Device-tile variant of chunked_prefill_attention: every argument is
