@@ -143,6 +143,9 @@ class JsonFsm:
         if frame["idx"] >= len(frame["props"]):
             self._force("}")
             self._pop_frame()
+            # chain: this object may itself complete its parent (nested
+            # object values) — found by schema fuzzing, tests/test_json_fsm_fuzz.py
+            self._resume_parent()
             return
         key, subschema = frame["props"][frame["idx"]]
         prefix = ", " if frame["idx"] > 0 else ""
