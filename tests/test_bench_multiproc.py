@@ -10,10 +10,21 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def _run_bench(extra, port):
+def _free_port() -> int:
+    """OS-assigned port: fixed ports collide with a rendezvous left over
+    from a previous suite run shutting down."""
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _run_bench(extra, port=None):
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", str(port), os.path.join(REPO, "bench.py"),
+           "--master-port", str(port or _free_port()),
+           os.path.join(REPO, "bench.py"),
            "--gpus", "2", "--steps", "1", "--warmup", "0",
            "--concurrency", "1", "--model", "tiny", *extra]
     proc = subprocess.run(cmd, capture_output=True, text=True, timeout=420,
@@ -27,7 +38,7 @@ def _run_bench(extra, port):
 @pytest.mark.slow
 @pytest.mark.timeout(500)
 def test_bench_dp2():
-    out = _run_bench([], 29721)
+    out = _run_bench([])
     assert out["config"]["parallelism"] == "dp2"
     assert out["value"] > 0
     assert out["config"]["offline_gate_pass_rate"] == 1.0
@@ -36,6 +47,6 @@ def test_bench_dp2():
 @pytest.mark.slow
 @pytest.mark.timeout(500)
 def test_bench_tp2():
-    out = _run_bench(["--tp", "2"], 29722)
+    out = _run_bench(["--tp", "2"])
     assert out["config"]["parallelism"] == "tp2"
     assert out["value"] > 0
