@@ -186,57 +186,99 @@ class InvestigationOrchestrator:
             self._generate_hypotheses(query)
             if m.phase == Phase.HYPOTHESIZE:
                 m.transition(Phase.INVESTIGATE)
+            return self._finish(query, start)
+        except Exception as e:  # noqa: BLE001 — surface failure as a result
+            return self._failure(start, e)
 
-            # Investigation loop (reference L651-653, cycle L901)
-            while m.can_continue() and m.phase in (Phase.INVESTIGATE, Phase.EVALUATE):
-                m.next_iteration()
-                confirmed = self._run_investigation_cycle(query)
-                if confirmed or not m.active_hypotheses():
-                    if m.can_transition(Phase.CONCLUDE):
-                        m.transition(Phase.CONCLUDE)
-                    break
-            if m.phase not in (Phase.CONCLUDE, Phase.COMPLETE, Phase.FAILED):
+    def resume_from_checkpoint(self, query: str, checkpoint: Any,
+                               incident_id: Optional[str] = None) -> InvestigationResult:
+        """Continue an investigation from a saved checkpoint: the state
+        machine is rehydrated (phase, hypothesis tree, services, symptoms)
+        and the pipeline picks up at the restored phase — skipping triage
+        and hypothesis generation when their results are already in the
+        snapshot. Goes beyond the reference, which stores and lists
+        checkpoints but never resumes them (src/session/checkpoint.ts)."""
+        from ..session.checkpoint import machine_from_checkpoint
+
+        self.machine = machine_from_checkpoint(
+            checkpoint, max_hypotheses=self.machine.max_hypotheses,
+            max_depth=self.machine.max_depth,
+            max_iterations=self.machine.max_iterations)
+        m = self.machine
+        m.on("phase_change", lambda d: self._on_phase(d))
+        self._emit("resumed", investigation_id=m.investigation_id,
+                   phase=m.phase.value, hypotheses=len(m.hypotheses))
+        start = time.time()
+        try:
+            if m.phase == Phase.TRIAGE:
+                self._run_triage(query, incident_id)
+                if m.phase == Phase.TRIAGE:
+                    m.transition(Phase.HYPOTHESIZE)
+            if m.phase == Phase.HYPOTHESIZE:
+                if not m.hypotheses:
+                    self._generate_hypotheses(query)
+                if m.phase == Phase.HYPOTHESIZE:
+                    m.transition(Phase.INVESTIGATE)
+            return self._finish(query, start)
+        except Exception as e:  # noqa: BLE001
+            return self._failure(start, e)
+
+    def _finish(self, query: str, start: float) -> InvestigationResult:
+        """Investigation loop + conclusion + remediation + result assembly
+        (shared by investigate() and resume_from_checkpoint())."""
+        m = self.machine
+        # Investigation loop (reference L651-653, cycle L901)
+        while m.can_continue() and m.phase in (Phase.INVESTIGATE, Phase.EVALUATE):
+            m.next_iteration()
+            confirmed = self._run_investigation_cycle(query)
+            if confirmed or not m.active_hypotheses():
                 if m.can_transition(Phase.CONCLUDE):
                     m.transition(Phase.CONCLUDE)
+                break
+        if m.phase not in (Phase.CONCLUDE, Phase.COMPLETE, Phase.FAILED):
+            if m.can_transition(Phase.CONCLUDE):
+                m.transition(Phase.CONCLUDE)
 
-            self._run_conclusion(query)
-            self._run_remediation(query)
-            if m.phase == Phase.REMEDIATE:
-                m.transition(Phase.COMPLETE)
-            elif m.phase == Phase.CONCLUDE:
-                m.transition(Phase.COMPLETE)
+        self._run_conclusion(query)
+        self._run_remediation(query)
+        if m.phase == Phase.REMEDIATE:
+            m.transition(Phase.COMPLETE)
+        elif m.phase == Phase.CONCLUDE:
+            m.transition(Phase.COMPLETE)
 
-            c = m.conclusion
-            return InvestigationResult(
-                investigation_id=m.investigation_id,
-                root_cause=c.root_cause if c else "inconclusive",
-                confidence=c.confidence if c else "low",
-                summary=m.get_summary(),
-                affected_services=m.affected_services,
-                remediation_plan=_plan_dict(m.remediation_plan),
-                duration_ms=int((time.time() - start) * 1000),
-                phases_visited=self.phases_visited,
-                hypotheses=[h.to_dict() for h in m.hypotheses.values()],
-                evidence=[e.description for h in m.hypotheses.values() for e in h.evidence],
-            )
-        except Exception as e:  # noqa: BLE001 — surface failure as a result
-            try:
-                m.fail(str(e))
-            except Exception:  # noqa: BLE001
-                pass
-            return InvestigationResult(
-                investigation_id=m.investigation_id,
-                root_cause="",
-                confidence="low",
-                summary=m.get_summary(),
-                affected_services=m.affected_services,
-                remediation_plan=None,
-                duration_ms=int((time.time() - start) * 1000),
-                phases_visited=self.phases_visited,
-                hypotheses=[h.to_dict() for h in m.hypotheses.values()],
-                success=False,
-                error=f"{type(e).__name__}: {e}",
-            )
+        c = m.conclusion
+        return InvestigationResult(
+            investigation_id=m.investigation_id,
+            root_cause=c.root_cause if c else "inconclusive",
+            confidence=c.confidence if c else "low",
+            summary=m.get_summary(),
+            affected_services=m.affected_services,
+            remediation_plan=_plan_dict(m.remediation_plan),
+            duration_ms=int((time.time() - start) * 1000),
+            phases_visited=self.phases_visited,
+            hypotheses=[h.to_dict() for h in m.hypotheses.values()],
+            evidence=[e.description for h in m.hypotheses.values() for e in h.evidence],
+        )
+
+    def _failure(self, start: float, e: Exception) -> InvestigationResult:
+        m = self.machine
+        try:
+            m.fail(str(e))
+        except Exception:  # noqa: BLE001
+            pass
+        return InvestigationResult(
+            investigation_id=m.investigation_id,
+            root_cause="",
+            confidence="low",
+            summary=m.get_summary(),
+            affected_services=m.affected_services,
+            remediation_plan=None,
+            duration_ms=int((time.time() - start) * 1000),
+            phases_visited=self.phases_visited,
+            hypotheses=[h.to_dict() for h in m.hypotheses.values()],
+            success=False,
+            error=f"{type(e).__name__}: {e}",
+        )
 
     # -- triage (reference L723-872) ------------------------------------------
 

@@ -497,6 +497,48 @@ def checkpoint_show(investigation_id: str, checkpoint_id: Optional[str]) -> None
     _echo(cp.format())
 
 
+@checkpoint.command("resume")
+@click.argument("investigation_id")
+@click.argument("checkpoint_id", required=False)
+@click.option("--query", default="", help="updated incident description (defaults to the triage summary)")
+@click.option("--scenario", default=None, help="simulated incident scenario")
+@click.pass_context
+def checkpoint_resume(ctx: click.Context, investigation_id: str,
+                      checkpoint_id: Optional[str], query: str,
+                      scenario: Optional[str]) -> None:
+    """Rehydrate the saved state machine and continue the investigation
+    from its checkpointed phase (beyond the reference, which only stores
+    checkpoints)."""
+    from .session.checkpoint import CheckpointStore, checkpoint_from_machine
+
+    store = CheckpointStore()
+    cp = (store.load(investigation_id, checkpoint_id) if checkpoint_id
+          else store.load_latest(investigation_id))
+    if cp is None:
+        _echo(f"{RED}no checkpoint found for {investigation_id}{RESET}")
+        sys.exit(1)
+    config = ctx.obj["config"]
+    rt = _build_runtime(config, scenario=scenario)
+    from .agent.orchestrator import InvestigationOrchestrator
+
+    orch = InvestigationOrchestrator(
+        llm=rt["llm"], tool_executor=rt["registry"],
+        available_tools=set(t.name for t in rt["tools"]),
+        knowledge_retriever=rt["retriever"],
+        max_iterations=config.agent.max_iterations * 2,
+    )
+    _echo(f"resuming {investigation_id} from checkpoint {cp.checkpoint_id} "
+          f"(phase: {cp.phase}, {len(cp.hypotheses)} hypotheses)")
+    result = orch.resume_from_checkpoint(
+        query or f"resume: {'; '.join(cp.symptoms) or investigation_id}", cp)
+    _echo(result.to_dict().get("summary", ""))
+    new_cp = checkpoint_from_machine(orch.machine, label="resumed-final")
+    store.save(new_cp)
+    _echo(f"{DIM}checkpoint saved: {new_cp.checkpoint_id}{RESET}")
+    if not result.success:
+        sys.exit(1)
+
+
 @checkpoint.command("delete")
 @click.argument("investigation_id")
 @click.argument("checkpoint_id", required=False)

@@ -168,6 +168,36 @@ class CheckpointStore:
         return removed
 
 
+def machine_from_checkpoint(cp: InvestigationCheckpoint,
+                            max_hypotheses: int = 10, max_depth: int = 4,
+                            max_iterations: int = 20) -> Any:
+    """Rehydrate an InvestigationStateMachine from a checkpoint — the
+    inverse of checkpoint_from_machine, used by
+    `InvestigationOrchestrator.resume_from_checkpoint` / `runbook
+    checkpoint resume` (the reference stores checkpoints but never
+    auto-resumes them: session/checkpoint.ts L164-407 is storage only)."""
+    from ..agent.state_machine import InvestigationStateMachine, Phase
+    from ..agent.types import Hypothesis, now_ms
+
+    m = InvestigationStateMachine(
+        investigation_id=cp.investigation_id, max_hypotheses=max_hypotheses,
+        max_depth=max_depth, max_iterations=max_iterations)
+    try:
+        m.phase = Phase(cp.phase) if cp.phase else Phase.TRIAGE
+    except ValueError:
+        m.phase = Phase.TRIAGE
+    if m.phase in (Phase.IDLE, Phase.COMPLETE, Phase.FAILED):
+        # nothing mid-flight to continue; restart the pipeline on the
+        # restored evidence instead of replaying a terminal state
+        m.phase = Phase.TRIAGE
+    m.hypotheses = {h["id"]: Hypothesis.from_dict(h)
+                    for h in cp.hypotheses if h.get("id")}
+    m.affected_services = list(cp.services)
+    m.symptoms = list(cp.symptoms)
+    m.started_at = now_ms()
+    return m
+
+
 def checkpoint_from_machine(machine: Any, label: str = "") -> InvestigationCheckpoint:
     """Snapshot an InvestigationStateMachine."""
     return InvestigationCheckpoint(
