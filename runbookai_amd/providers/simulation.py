@@ -143,6 +143,108 @@ class SimScenario:
         return s
 
     @classmethod
+    def kafka_disk_pressure(cls) -> "SimScenario":
+        """Broker disk filling up → producers time out; the cause is a
+        retention misconfiguration, NOT a deploy (exercises non-deploy
+        causality: the deploy history is clean)."""
+        s = cls(name="kafka-disk-pressure")
+        s.incident = {
+            "id": "PD-EXAMPLE-003",
+            "title": "order events delayed; kafka producers timing out",
+            "status": "triggered", "urgency": "high", "service": "order-events",
+            "createdAt": "2026-02-12T03:40:00Z",
+        }
+        s.services = [
+            {"name": "order-events", "status": "degraded", "type": "ecs"},
+            {"name": "kafka-broker-2", "status": "unhealthy", "type": "msk"},
+            {"name": "analytics-ingest", "status": "degraded", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "kafka-broker-2-disk-used", "state": "ALARM",
+             "reason": "disk used > 95% on /var/kafka (volume kafka-data-2)",
+             "service": "kafka-broker-2"},
+            {"name": "order-events-producer-errors", "state": "ALARM",
+             "reason": "producer error rate > 2% for 20 minutes", "service": "order-events"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-12T03:41:12Z", "service": "kafka-broker-2", "level": "ERROR",
+             "message": "No space left on device: /var/kafka/order-events-7/00000000.log"},
+            {"timestamp": "2026-02-12T03:41:40Z", "service": "order-events", "level": "ERROR",
+             "message": "KafkaTimeoutError: batch for order-events-7 expired after 30000ms"},
+            {"timestamp": "2026-02-12T03:42:05Z", "service": "kafka-broker-2", "level": "WARN",
+             "message": "log retention check skipped: retention.ms=-1 on topic order-events (unbounded)"},
+            {"timestamp": "2026-02-12T03:44:20Z", "service": "analytics-ingest", "level": "WARN",
+             "message": "consumer lag growing on order-events: 1.2M messages behind"},
+        ]
+        s.metrics = {
+            "kafka-broker-2.disk.used_pct": [71, 78, 84, 89, 93, 96, 98, 99],
+            "order-events.producer.error_rate": [0.0, 0.0, 0.1, 0.4, 1.8, 3.5, 5.2, 6.0],
+            "analytics-ingest.consumer.lag": [1200, 1500, 9000, 120000, 480000, 1200000],
+        }
+        s.deployments = []   # clean deploy history — the cause is config drift
+        s.pods = [
+            {"name": "order-events-5d4f", "namespace": "prod", "status": "Running",
+             "restarts": 0, "cpu": "310m", "memory": "800Mi"},
+        ]
+        s.monitors = [
+            {"name": "kafka disk", "status": "Alert",
+             "query": "max:kafka.disk.used_pct{broker:2} > 90"},
+        ]
+        s.resources = {
+            "msk": [{"id": "kafka-prod", "brokers": 3, "status": "active",
+                     "storagePerBrokerGiB": 512}],
+            "ebs": [{"id": "kafka-data-2", "sizeGiB": 512, "usedPct": 99,
+                     "attachedTo": "kafka-broker-2"}],
+        }
+        return s
+
+    @classmethod
+    def cert_expiry(cls) -> "SimScenario":
+        """Expired TLS certificate on an internal endpoint → handshake
+        failures cascading into upstream 503s; no alarm fires on the cert
+        itself (exercises log-driven causality over alarm-driven)."""
+        s = cls(name="tls-cert-expiry")
+        s.incident = {
+            "id": "PD-EXAMPLE-004",
+            "title": "payments failing: upstream TLS errors to auth-service",
+            "status": "triggered", "urgency": "high", "service": "payment-service",
+            "createdAt": "2026-02-13T00:02:00Z",
+        }
+        s.services = [
+            {"name": "payment-service", "status": "degraded", "type": "ecs"},
+            {"name": "auth-service", "status": "degraded", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "payment-success-rate", "state": "ALARM",
+             "reason": "success rate < 90% for 10 minutes", "service": "payment-service"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-13T00:01:02Z", "service": "payment-service", "level": "ERROR",
+             "message": "x509: certificate has expired or is not yet valid: auth-internal.prod "
+                        "notAfter=2026-02-13T00:00:00Z"},
+            {"timestamp": "2026-02-13T00:01:30Z", "service": "payment-service", "level": "ERROR",
+             "message": "TLS handshake error dialing auth-internal.prod:8443: bad certificate"},
+            {"timestamp": "2026-02-13T00:02:10Z", "service": "auth-service", "level": "WARN",
+             "message": "cert-rotation job auth-internal last succeeded 92 days ago (expected every 60d)"},
+            {"timestamp": "2026-02-13T00:03:00Z", "service": "payment-service", "level": "ERROR",
+             "message": "POST /charge returning 503: auth unavailable"},
+        ]
+        s.metrics = {
+            "payment-service.success_rate": [99.9, 99.8, 99.9, 41.0, 8.2, 3.1],
+            "auth-service.tls.handshake_errors": [0, 0, 0, 180, 1450, 2600],
+        }
+        s.deployments = []
+        s.monitors = [
+            {"name": "payment success", "status": "Alert",
+             "query": "avg:payment.success_rate{*} < 90"},
+        ]
+        s.resources = {
+            "acm": [{"id": "auth-internal.prod", "status": "EXPIRED",
+                     "notAfter": "2026-02-13T00:00:00Z", "renewalEligibility": "ELIGIBLE"}],
+        }
+        return s
+
+    @classmethod
     def from_fixture(cls, case: dict[str, Any]) -> "SimScenario":
         """Generate a scenario from an eval fixture case: the telemetry
         reflects the expected root cause so a competent agent can find it."""
@@ -189,6 +291,8 @@ class SimScenario:
 _SCENARIOS = {
     "redis-conn-exhaustion": SimScenario.redis_exhaustion,
     "api-gateway-5xx": SimScenario.gateway_5xx,
+    "kafka-disk-pressure": SimScenario.kafka_disk_pressure,
+    "tls-cert-expiry": SimScenario.cert_expiry,
 }
 
 _current: Optional[SimScenario] = None
