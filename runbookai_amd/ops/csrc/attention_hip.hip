@@ -163,7 +163,32 @@ __global__ void paged_decode_partial_kernel(
     OnlineAcc acc;
     online_init(acc);
     const int* bt = block_tables + (long)b * max_blocks;
-    for (int t = t_begin + wid; t < t_end; t += nw) {
+    // 2-token unroll: both K/V pair loads issue before either reduction, so
+    // two HBM fetches are in flight per online-update chain (the serial
+    // m/l dependency otherwise leaves one 256-B fetch per latency).
+    int t = t_begin + wid;
+    for (; t + nw < t_end; t += 2 * nw) {
+        const int ta = t, tb = t + nw;
+        const long base_a = (((long)bt[ta / BS] * Hk + hk) * BS + (ta % BS)) * D;
+        const long base_b = (((long)bt[tb / BS] * Hk + hk) * BS + (tb % BS)) * D;
+        float sa = 0.f, va0 = 0.f, va1 = 0.f;
+        float sb = 0.f, vb0 = 0.f, vb1 = 0.f;
+        if (active) {
+            ushort2_t ka = *reinterpret_cast<const ushort2_t*>(k_cache + base_a + 2 * lane);
+            ushort2_t kb = *reinterpret_cast<const ushort2_t*>(k_cache + base_b + 2 * lane);
+            ushort2_t va = *reinterpret_cast<const ushort2_t*>(v_cache + base_a + 2 * lane);
+            ushort2_t vb = *reinterpret_cast<const ushort2_t*>(v_cache + base_b + 2 * lane);
+            sa = q0 * bf2f(ka[0]) + q1 * bf2f(ka[1]);
+            sb = q0 * bf2f(kb[0]) + q1 * bf2f(kb[1]);
+            va0 = bf2f(va[0]); va1 = bf2f(va[1]);
+            vb0 = bf2f(vb[0]); vb1 = bf2f(vb[1]);
+        }
+        sa = wave_sum(sa);
+        sb = wave_sum(sb);
+        online_update(acc, sa, va0, va1);
+        online_update(acc, sb, vb0, vb1);
+    }
+    for (; t < t_end; t += nw) {
         const int blk = bt[t / BS];
         const long base = (((long)blk * Hk + hk) * BS + (t % BS)) * D;
         float s = 0.f, v0 = 0.f, v1 = 0.f;

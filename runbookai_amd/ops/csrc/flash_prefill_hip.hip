@@ -27,9 +27,10 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 #define QTILE 64   // q rows per workgroup (16 per wave)
-#define KTILE 32   // kv tokens per LDS tile
+#define KTILE 64   // kv tokens per LDS tile
 #define DHEAD 128  // head dim (Llama-3)
-#define VPAD 40    // padded row length (elements) for transposed V / P tiles
+#define VPAD 72    // padded row length (elements) for transposed V / P tiles
+#define NKC (KTILE / 16)   // 16-col score sub-tiles per kv tile
 
 // ------------------------------------------------------------------ probe
 __global__ void mfma_probe_kernel(const ushort_t* __restrict__ A,
@@ -117,18 +118,23 @@ __global__ void flash_prefill_kernel(
                               : (hist + seg_len);
     const int n_tiles = (kv_len + KTILE - 1) / KTILE;
 
-    for (int kt = 0; kt < n_tiles; ++kt) {
-        const int kv0 = kt * KTILE;               // local token of tile col 0
-        __syncthreads();
-        // cooperative K/V tile load. K: bf16x8 chunks into swizzled rows.
-        // V: each 16-B chunk of a token row is TRANSPOSED into v_lds
-        // (8 scalar writes), giving b128 fragment reads in the PV loop.
-        for (int idx = threadIdx.x; idx < KTILE * DHEAD / 8; idx += 256) {
-            const int row = (idx * 8) / DHEAD;    // token within tile
-            const int col = (idx * 8) % DHEAD;    // d
-            const int tok = kv0 + row;
-            bf16x8_t kv8 = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-            bf16x8_t vv8 = kv8;
+    // Register-staged double buffering (guide T14): the NEXT tile's global
+    // loads are issued right after the current tile's LDS write, so HBM
+    // latency hides under the QK/softmax/PV compute phase — the sync-staged
+    // version stalls a full memory latency per tile at 2 waves/SIMD.
+    // Each thread owns NCHUNK 16-B pieces of K and V.
+    constexpr int NCHUNK = KTILE * DHEAD / 8 / 256;
+    bf16x8_t skA[NCHUNK], svA[NCHUNK], skB[NCHUNK], svB[NCHUNK];
+
+    auto load_tile = [&](int kt, bf16x8_t (&sk)[NCHUNK], bf16x8_t (&sv)[NCHUNK]) {
+#pragma unroll
+        for (int c = 0; c < NCHUNK; ++c) {
+            const int idx = threadIdx.x + c * 256;
+            const int row = (idx * 8) / DHEAD;
+            const int col = (idx * 8) % DHEAD;
+            const int tok = kt * KTILE + row;
+            sk[c] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+            sv[c] = sk[c];
             if (tok < kv_len) {
                 long base;
                 if (PAGED) {
@@ -137,20 +143,43 @@ __global__ void flash_prefill_kernel(
                 } else {
                     base = ((long)(seg_start + tok) * Hk + hk) * DHEAD + col;
                 }
-                kv8 = *reinterpret_cast<const bf16x8_t*>((PAGED ? k_cache : k) + base);
-                vv8 = *reinterpret_cast<const bf16x8_t*>((PAGED ? v_cache : v) + base);
+                sk[c] = *reinterpret_cast<const bf16x8_t*>((PAGED ? k_cache : k) + base);
+                sv[c] = *reinterpret_cast<const bf16x8_t*>((PAGED ? v_cache : v) + base);
             }
-            *reinterpret_cast<bf16x8_t*>(
-                reinterpret_cast<char*>(k_lds) + kswz((unsigned)row, (unsigned)col * 2)) = kv8;
-#pragma unroll
-            for (int j = 0; j < 8; ++j) v_lds[col + j][row] = vv8[j];
         }
+    };
+    auto write_tile = [&](bf16x8_t (&sk)[NCHUNK], bf16x8_t (&sv)[NCHUNK]) {
+#pragma unroll
+        for (int c = 0; c < NCHUNK; ++c) {
+            const int idx = threadIdx.x + c * 256;
+            const int row = (idx * 8) / DHEAD;
+            const int col = (idx * 8) % DHEAD;
+            *reinterpret_cast<bf16x8_t*>(
+                reinterpret_cast<char*>(k_lds) + kswz((unsigned)row, (unsigned)col * 2)) = sk[c];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) v_lds[col + j][row] = sv[c][j];
+        }
+    };
+
+    load_tile(0, skA, svA);
+    int parity = 0;
+    for (int kt = 0; kt < n_tiles; ++kt) {
+        const int kv0 = kt * KTILE;               // local token of tile col 0
+        __syncthreads();   // previous tile's LDS readers are done
+        if (parity == 0) write_tile(skA, svA); else write_tile(skB, svB);
+        if (kt + 1 < n_tiles) {
+            // issue next tile's loads NOW; first use is next iteration's
+            // write phase, so the waits land after this tile's compute
+            if (parity == 0) load_tile(kt + 1, skB, svB);
+            else load_tile(kt + 1, skA, svA);
+        }
+        parity ^= 1;
         __syncthreads();
 
         // ---- QK^T ----
-        f32x4_t sc[2];
+        f32x4_t sc[NKC];
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < NKC; ++kc) {
             sc[kc] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int s = 0; s < 4; ++s) {
@@ -165,7 +194,7 @@ __global__ void flash_prefill_kernel(
         // ---- mask + online softmax ----
         float tile_max[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < NKC; ++kc) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 const int qrow = q0g + wid * 16 + (lane >> 4) * 4 + r;
@@ -194,7 +223,7 @@ __global__ void flash_prefill_kernel(
             row_sum[r] = 0.f;
         }
 #pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
+        for (int kc = 0; kc < NKC; ++kc) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 // masked elements contribute EXACTLY zero (whole-row-masked
@@ -220,15 +249,22 @@ __global__ void flash_prefill_kernel(
         }
         __syncthreads();
 
-        // ---- PV: O[16 q][128 d] += P[16x32] * V[32x128] ----
-        bf16x8_t pfrag = *reinterpret_cast<const bf16x8_t*>(
-            &p_lds[wid][lane & 15][(lane >> 4) * 8]);
+        // ---- PV: O[16 q][128 d] += P[16xKTILE] * V[KTILEx128] ----
+        bf16x8_t pfrag[NKC / 2];
+#pragma unroll
+        for (int ks = 0; ks < NKC / 2; ++ks)
+            pfrag[ks] = *reinterpret_cast<const bf16x8_t*>(
+                &p_lds[wid][lane & 15][ks * 32 + (lane >> 4) * 8]);
 #pragma unroll
         for (int d = 0; d < 8; ++d) {
-            // B fragment = V^T rows: one contiguous 16-B read per MFMA
-            bf16x8_t vfrag = *reinterpret_cast<const bf16x8_t*>(
-                &v_lds[d * 16 + (lane & 15)][(lane >> 4) * 8]);
-            o_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfrag, o_acc[d], 0, 0, 0);
+#pragma unroll
+            for (int ks = 0; ks < NKC / 2; ++ks) {
+                // B fragment = V^T rows: one contiguous 16-B read per MFMA
+                bf16x8_t vfrag = *reinterpret_cast<const bf16x8_t*>(
+                    &v_lds[d * 16 + (lane & 15)][ks * 32 + (lane >> 4) * 8]);
+                o_acc[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag[ks], vfrag,
+                                                                   o_acc[d], 0, 0, 0);
+            }
         }
     }
 

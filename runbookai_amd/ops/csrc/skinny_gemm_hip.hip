@@ -21,8 +21,6 @@ typedef __attribute__((ext_vector_type(4))) float f32x4_t;
 
 #define BN 64          // output columns per workgroup
 #define BKG 64         // k per staged tile (2 MFMA k-steps)
-#define TILE_BYTES (BN * BKG * 2)   // 8 KB
-#define MAX_MTILES 4   // up to 64 rows (4 x 16)
 
 DEVINL unsigned swz(unsigned row, unsigned colb) {
     return row * (BKG * 2) + (colb ^ ((row & 7u) << 4));
@@ -169,17 +167,6 @@ __global__ void skinny_gemm_kernel(
         float s = 0.f;
         for (int sp = 0; sp < splitk; ++sp) s += partial[((long)sp * M + m) * N + n];
         out[(long)m * N + n] = f2bf(s);
-    }
-}
-
-__global__ void skinny_gemm_merge_kernel(const float* __restrict__ partial,
-                                         ushort_t* __restrict__ out,
-                                         long MN, long stride, int splitk) {
-    for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < MN;
-         i += (long)gridDim.x * blockDim.x) {
-        float s = 0.f;
-        for (int sp = 0; sp < splitk; ++sp) s += partial[sp * stride + i];
-        out[i] = f2bf(s);
     }
 }
 
