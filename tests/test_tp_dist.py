@@ -137,3 +137,66 @@ def test_70b_config_sharding_math():
 
     # ~70B parameters at the declared shapes
     assert 60e9 < param_count(cfg) < 80e9
+
+
+def _tp4_worker(rank: int, world: int, port: int, ids: list, q) -> None:
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["RANK"] = str(rank)
+    import torch.distributed as dist
+
+    from runbookai_amd.engine.llama import LlamaConfig
+
+    cfg = LlamaConfig(name="tiny4", hidden_size=256, intermediate_size=512,
+                      num_layers=2, num_heads=8, num_kv_heads=4, head_dim=32,
+                      vocab_size=4096, max_seq_len=512)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        model = LlamaModel(cfg, device="cpu", tp=world, seed=43, kv_blocks=32)
+        model.kv.allocate(1, len(ids))
+        logits = model.prefill(
+            torch.tensor(ids), torch.arange(len(ids), dtype=torch.int32),
+            torch.tensor([0, len(ids)], dtype=torch.int32),
+            model.kv.slot_mapping(1, 0, len(ids)))
+        if rank == 0:
+            q.put(("ok", logits[0].float().numpy().tobytes(), list(logits.shape)))
+    except Exception as e:  # noqa: BLE001
+        if rank == 0:
+            q.put(("err", repr(e), None))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp4_matches_tp1():
+    """TP=4 (the 70B half-node degree) on a GQA config where kv-heads
+    still divide: same logits as the unsharded model."""
+    from runbookai_amd.engine.llama import LlamaConfig
+
+    cfg = LlamaConfig(name="tiny4", hidden_size=256, intermediate_size=512,
+                      num_layers=2, num_heads=8, num_kv_heads=4, head_dim=32,
+                      vocab_size=4096, max_seq_len=512)
+    ids = list(range(40, 70))
+    ref_model = LlamaModel(cfg, device="cpu", tp=1, seed=43, kv_blocks=32)
+    ref_model.kv.allocate(1, len(ids))
+    ref = ref_model.prefill(
+        torch.tensor(ids), torch.arange(len(ids), dtype=torch.int32),
+        torch.tensor([0, len(ids)], dtype=torch.int32),
+        ref_model.kv.slot_mapping(1, 0, len(ids)))
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp4_worker, args=(r, 4, 29651, ids, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    status, payload, shape = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    import numpy as np
+
+    tp4 = torch.from_numpy(np.frombuffer(payload, dtype=np.float32).copy()).view(shape[1])
+    diff = (ref[0].float() - tp4).abs().max().item()
+    assert diff < 0.05, f"TP=4 diverged from TP=1 by {diff}"
