@@ -115,12 +115,15 @@ def load_hf_checkpoint(model: LlamaModel, path: str,
     model.embed.weight = place(take("model.embed_tokens.weight"))
     for i, layer in enumerate(model.layers):
         p = f"model.layers.{i}."
+        from .llama import kv_shard_range
+
         q = take(p + "self_attn.q_proj.weight")   # [hq*d, H]
         k = take(p + "self_attn.k_proj.weight")   # [hk*d, H]
         v = take(p + "self_attn.v_proj.weight")
         q_r = q.view(hq, d, H)[rank * hq_r:(rank + 1) * hq_r].reshape(hq_r * d, H)
-        k_r = k.view(hk, d, H)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, H)
-        v_r = v.view(hk, d, H)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, H)
+        ks, ke = kv_shard_range(hk, tp, rank)
+        k_r = k.view(hk, d, H)[ks:ke].reshape(hk_r * d, H)
+        v_r = v.view(hk, d, H)[ks:ke].reshape(hk_r * d, H)
         layer.qkv.weight = place(torch.cat([q_r, k_r, v_r], 0))
         del q, k, v, q_r, k_r, v_r
 

@@ -56,6 +56,19 @@ CONFIGS: dict[str, LlamaConfig] = {
 }
 
 
+def kv_shard_range(hk: int, tp: int, rank: int) -> tuple[int, int]:
+    """This rank's K/V head slice. When tp > num_kv_heads the heads are
+    REPLICATED (tp/hk ranks share each kv head — their q-head groups all
+    attend to it); otherwise heads shard contiguously."""
+    if hk >= tp:
+        assert hk % tp == 0, f"kv heads {hk} must divide by tp {tp}"
+        hk_r = hk // tp
+        return rank * hk_r, (rank + 1) * hk_r
+    assert tp % hk == 0, f"tp {tp} must be a multiple of kv heads {hk}"
+    idx = rank // (tp // hk)
+    return idx, idx + 1
+
+
 class LlamaLayer:
     def __init__(self, cfg: LlamaConfig, tp: int, rank: int, dtype, device,
                  gen_factory) -> None:
@@ -78,8 +91,9 @@ class LlamaLayer:
         self.qkv = ColumnParallelLinear(H, qkv_out, 1, dtype, device, gen_factory())
         w = self.qkv.weight
         q_w = w[: hq * d].view(hq, d, -1)[rank * hq_r:(rank + 1) * hq_r].reshape(hq_r * d, -1)
-        k_w = w[hq * d:(hq + hk) * d].view(hk, d, -1)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, -1)
-        v_w = w[(hq + hk) * d:].view(hk, d, -1)[rank * hk_r:(rank + 1) * hk_r].reshape(hk_r * d, -1)
+        ks, ke = kv_shard_range(hk, tp, rank)
+        k_w = w[hq * d:(hq + hk) * d].view(hk, d, -1)[ks:ke].reshape(hk_r * d, -1)
+        v_w = w[(hq + hk) * d:].view(hk, d, -1)[ks:ke].reshape(hk_r * d, -1)
         self.qkv.weight = torch.cat([q_w, k_w, v_w], 0).contiguous()
 
         self.o_proj = RowParallelLinear(cfg.q_size, H, 1, dtype, device, gen_factory())

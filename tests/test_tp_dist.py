@@ -200,3 +200,58 @@ def test_tp4_matches_tp1():
     tp4 = torch.from_numpy(np.frombuffer(payload, dtype=np.float32).copy()).view(shape[1])
     diff = (ref[0].float() - tp4).abs().max().item()
     assert diff < 0.05, f"TP=4 diverged from TP=1 by {diff}"
+
+
+def _tp4_repl_worker(rank: int, world: int, port: int, ids: list, q) -> None:
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["RANK"] = str(rank)
+    import torch.distributed as dist
+
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        # tiny has 4 q-heads / 2 kv-heads: tp=4 REPLICATES each kv head
+        # across two ranks (kv_shard_range replication path)
+        model = LlamaModel(CONFIGS["tiny"], device="cpu", tp=world, seed=44)
+        model.kv.allocate(1, len(ids))
+        logits = model.prefill(
+            torch.tensor(ids), torch.arange(len(ids), dtype=torch.int32),
+            torch.tensor([0, len(ids)], dtype=torch.int32),
+            model.kv.slot_mapping(1, 0, len(ids)))
+        if rank == 0:
+            q.put(("ok", logits[0].float().numpy().tobytes(), list(logits.shape)))
+    except Exception as e:  # noqa: BLE001
+        if rank == 0:
+            q.put(("err", repr(e), None))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp4_kv_replication_matches_tp1():
+    """tp > num_kv_heads: kv heads replicate across rank groups and the
+    logits still match the unsharded model."""
+    ids = list(range(10, 38))
+    ref_model = LlamaModel(CONFIGS["tiny"], device="cpu", tp=1, seed=44)
+    ref_model.kv.allocate(1, len(ids))
+    ref = ref_model.prefill(
+        torch.tensor(ids), torch.arange(len(ids), dtype=torch.int32),
+        torch.tensor([0, len(ids)], dtype=torch.int32),
+        ref_model.kv.slot_mapping(1, 0, len(ids)))
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp4_repl_worker, args=(r, 4, 29661, ids, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    status, payload, shape = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", payload
+    import numpy as np
+
+    tp4 = torch.from_numpy(np.frombuffer(payload, dtype=np.float32).copy()).view(shape[1])
+    diff = (ref[0].float() - tp4).abs().max().item()
+    assert diff < 0.05, f"TP=4 (kv replication) diverged from TP=1 by {diff}"
