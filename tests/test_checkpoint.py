@@ -66,6 +66,23 @@ class TestCheckpointRoundTrip:
             tiny.hidden_size, tiny.intermediate_size, tiny.num_layers,
             tiny.num_heads, tiny.num_kv_heads, tiny.head_dim, tiny.vocab_size)
 
+    def test_fp32_checkpoint_converts_to_model_dtype(self, tmp_path):
+        """HF checkpoints ship in fp32/fp16/bf16; the loader converts to
+        the model's compute dtype on placement."""
+        src = tiny_model(seed=31)
+        export_hf_checkpoint(src, str(tmp_path))
+        from safetensors import safe_open
+        from safetensors.torch import save_file
+
+        fp = str(tmp_path / "model.safetensors")
+        with safe_open(fp, framework="pt") as f:
+            state = {k: f.get_tensor(k).float() for k in f.keys()}
+        save_file(state, fp)
+        loaded = load_model(str(tmp_path), device="cpu", kv_blocks=64)
+        assert loaded.layers[0].qkv.weight.dtype == torch.bfloat16
+        assert torch.equal(loaded.layers[0].qkv.weight,
+                           src.layers[0].qkv.weight)   # bf16->fp32->bf16 exact
+
     def test_tied_embeddings_fallback(self, tmp_path):
         """Checkpoints without lm_head.weight tie it to the embedding."""
         src = tiny_model()
