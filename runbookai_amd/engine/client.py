@@ -87,7 +87,10 @@ class LocalEngineClient:
         reference src/model/llm.ts prompt assembly) and the llm_parser
         fallbacks absorb deviations."""
         tok = self._tok()
-        if schema is not None and self.engine.hf_tokenizer is not None:
+        if (schema is not None and self.engine.hf_tokenizer is not None
+                and not self.engine.supports_bpe_grammar):
+            # vocab too large for the Python token-trie masker: fall back
+            # to schema-in-prompt + tolerant parsing
             body = (body + "\n\nRespond with ONLY a JSON object matching this "
                     "schema:\n" + json.dumps(schema))
             schema = None

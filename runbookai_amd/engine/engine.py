@@ -156,15 +156,38 @@ class LLMEngine:
                                             name="llm-engine")
             self._thread.start()
 
+    # -- BPE grammar (trained checkpoints) ----------------------------------------
+
+    _BPE_MASKER_UNSET = object()
+
+    @property
+    def bpe_masker(self):
+        """Token-trie x FSM masker for the checkpoint tokenizer (None when
+        absent or over the size gate) — engine/grammar_bpe.py."""
+        if getattr(self, "_bpe_masker", self._BPE_MASKER_UNSET) is self._BPE_MASKER_UNSET:
+            if self.hf_tokenizer is None:
+                self._bpe_masker = None
+            else:
+                from .grammar_bpe import build_masker
+
+                self._bpe_masker = build_masker(self.hf_tokenizer)
+        return self._bpe_masker
+
+    @property
+    def supports_bpe_grammar(self) -> bool:
+        return self.bpe_masker is not None
+
     # -- submission ---------------------------------------------------------------
 
     def submit(self, prompt_ids: list[int], max_new_tokens: int = 512,
                temperature: float = 0.0, schema: Optional[dict[str, Any]] = None) -> Request:
-        if schema is not None and self.hf_tokenizer is not None:
+        if schema is not None and self.hf_tokenizer is not None \
+                and not self.supports_bpe_grammar:
             raise ValueError(
-                "grammar-constrained decoding runs on the byte tokenizer; with a "
-                "BPE checkpoint tokenizer, put the schema in the prompt instead "
-                "(LocalEngineClient does this automatically)")
+                "grammar-constrained decoding over this BPE vocab exceeds the "
+                "token-trie masker's size gate (grammar_bpe.MAX_VOCAB); put the "
+                "schema in the prompt instead (LocalEngineClient does this "
+                "automatically)")
         with self._lock:
             self._rid += 1
             # clamp to the model's context window (RoPE table bound)
@@ -304,6 +327,16 @@ class LLMEngine:
             need = len(req.pending_input) + req.max_new_tokens
             if batch and tokens + len(req.pending_input) > self.max_prefill_tokens:
                 break
+            if need > (kv.num_blocks - 1) * kv.block_size:
+                # can NEVER fit, even with the whole pool free: fail loudly
+                # instead of starving the queue forever
+                self.waiting.pop(0)
+                req.error = (f"request needs {need} KV tokens but the pool holds "
+                             f"{(kv.num_blocks - 1) * kv.block_size}; lower "
+                             f"max_new_tokens or raise kv_blocks")
+                req.state = "done"
+                req.done_event.set()
+                continue
             if not kv.can_allocate(need):
                 break
             cached = 0
@@ -501,6 +534,22 @@ class LLMEngine:
         decoding is unmasked (schemas are rejected at submit).
         """
         if self.hf_tokenizer is not None:
+            masker = self.bpe_masker
+            if any(r.fsm is not None for r in batch):
+                # token-trie grammar masks over the full checkpoint vocab
+                mask = torch.zeros((len(batch), logits.shape[1]), dtype=torch.bool)
+                for i, r in enumerate(batch):
+                    if r.fsm is None:
+                        mask[i] = True
+                        continue
+                    allowed = masker.allowed_tokens(r.fsm)
+                    if allowed:
+                        mask[i, allowed] = True
+                    elif masker.eot_id is not None:
+                        mask[i, masker.eot_id] = True
+                    else:
+                        mask[i] = True
+                logits = logits.masked_fill(~mask.to(logits.device), float("-inf"))
             greedy = all(r.temperature <= 0.0 for r in batch)
             if greedy:
                 chosen_t = logits.argmax(-1)
@@ -562,6 +611,8 @@ class LLMEngine:
         """Consume grammar-FORCED bytes (single-choice FSM states) without
         spending decode steps; they return as a chunk-prefill run."""
         run: list[int] = []
+        if self.hf_tokenizer is not None:
+            return run   # byte-forcing only applies to the byte tokenizer
         fsm = req.fsm
         while fsm is not None and not fsm.done:
             allowed = fsm.allowed_bytes()
@@ -576,6 +627,27 @@ class LLMEngine:
 
     def _advance_request(self, req: Request, tok: int) -> None:
         finished = False
+        if self.hf_tokenizer is not None and req.fsm is not None:
+            # BPE grammar path: one TOKEN advances the byte FSM through its
+            # whole expansion (trie admission guarantees legality)
+            masker = self.bpe_masker
+            token_bytes = masker.token_bytes.get(tok)
+            if req.fsm.done or tok == masker.eot_id or token_bytes is None:
+                finished = True
+            else:
+                masker.advance_token(req.fsm, token_bytes)
+                req.out_ids.append(tok)
+                req.pending_input = [tok]
+                if req.fsm.done:
+                    finished = True
+            if len(req.out_ids) >= req.max_new_tokens:
+                finished = True
+            if finished:
+                req.state = "done"
+                req.finished_at = time.time()
+                self.model.kv.free(req.rid)
+                req.done_event.set()
+            return
         if req.fsm is not None:
             if req.fsm.done or tok == SpecialTokens.EOT:
                 finished = True

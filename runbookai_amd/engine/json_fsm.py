@@ -41,6 +41,20 @@ class JsonFsm:
 
     # -- public API -------------------------------------------------------------
 
+    def clone(self) -> "JsonFsm":
+        """Cheap state snapshot (frames are flat dicts whose nested lists —
+        props, options — are never mutated, so shallow frame copies
+        suffice). Used by the BPE token-trie masker (engine/grammar_bpe.py)
+        to explore candidate token expansions."""
+        c = JsonFsm.__new__(JsonFsm)
+        c.schema = self.schema
+        c.max_total_bytes = self.max_total_bytes
+        c.emitted = self.emitted
+        c._pending = list(self._pending)
+        c._stack = [dict(f) for f in self._stack]
+        c._done = self._done
+        return c
+
     @property
     def done(self) -> bool:
         return self._done and not self._pending

@@ -160,18 +160,21 @@ class TestCheckpointRoundTrip:
         assert torch.equal(sharded.layers[0].gate_up.weight[ipr:], gu[inter + ipr:])
 
 
-class TestCheckpointEngine:
-    @pytest.fixture()
-    def ckpt_dir(self, tmp_path):
-        export_hf_checkpoint(tiny_model(seed=21), str(tmp_path))
-        tokenizers = pytest.importorskip("tokenizers")
-        tok = tokenizers.ByteLevelBPETokenizer()
-        tok.train_from_iterator(["redis pool exhausted on checkout"] * 30,
-                                vocab_size=400, min_frequency=1,
-                                special_tokens=["<|eot_id|>"])
-        tok.save(str(tmp_path / "tokenizer.json"))
-        return str(tmp_path)
+@pytest.fixture()
+def ckpt_dir(tmp_path):
+    """Tiny checkpoint + small trained BPE tokenizer.json (shared by the
+    checkpoint-engine and BPE-grammar suites)."""
+    export_hf_checkpoint(tiny_model(seed=21), str(tmp_path))
+    tokenizers = pytest.importorskip("tokenizers")
+    tok = tokenizers.ByteLevelBPETokenizer()
+    tok.train_from_iterator(["redis pool exhausted on checkout"] * 30,
+                            vocab_size=400, min_frequency=1,
+                            special_tokens=["<|eot_id|>"])
+    tok.save(str(tmp_path / "tokenizer.json"))
+    return str(tmp_path)
 
+
+class TestCheckpointEngine:
     def test_engine_serves_checkpoint_free_decode(self, ckpt_dir):
         from runbookai_amd.engine.client import LocalEngineClient
         from runbookai_amd.engine.engine import LLMEngine
@@ -181,9 +184,8 @@ class TestCheckpointEngine:
         try:
             assert eng.hf_tokenizer is not None
             assert eng.cfg.hidden_size == CONFIGS["tiny"].hidden_size
-            # schemas are a byte-tokenizer feature: reject loudly
-            with pytest.raises(ValueError):
-                eng.submit([1, 2, 3], schema={"type": "object", "properties": {}})
+            # small vocab: the token-trie grammar masker is active
+            assert eng.supports_bpe_grammar
             client = LocalEngineClient(eng, max_tokens=8)
             resp = client.chat("You are Runbook.", "status of checkout?")
             assert isinstance(resp.content, str)
@@ -206,6 +208,114 @@ class TestCheckpointEngine:
                                schema={"type": "object", "properties": {
                                    "a": {"type": "number"}}, "required": ["a"]})
             assert req.error == ""
+        finally:
+            eng.shutdown()
+
+
+SCHEMA_NESTED = {
+    "type": "object",
+    "properties": {
+        "summary": {"type": "string", "maxLength": 40},
+        "confidence": {"type": "number"},
+        "nested": {"type": "object",
+                   "properties": {"sev": {"enum": ["low", "high"]}},
+                   "required": ["sev"]},
+        "tags": {"type": "array", "items": {"type": "string", "maxLength": 8},
+                 "minItems": 1, "maxItems": 3},
+    },
+    "required": ["summary", "confidence", "nested", "tags"],
+}
+
+
+class TestBpeGrammar:
+    """Grammar-constrained decoding over a BPE vocab (token-trie x FSM,
+    engine/grammar_bpe.py): schema-valid JSON from a trained-format
+    checkpoint with random weights."""
+
+    def test_masker_random_walks_always_schema_valid(self, tmp_path):
+        import random
+
+        from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer
+        from runbookai_amd.engine.grammar_bpe import build_masker
+        from runbookai_amd.engine.json_fsm import JsonFsm
+
+        tokenizers = pytest.importorskip("tokenizers")
+        tk = tokenizers.ByteLevelBPETokenizer()
+        tk.train_from_iterator(["redis pool exhausted on checkout",
+                                "confidence high medium low 0.9"] * 30,
+                               vocab_size=500, min_frequency=1,
+                               special_tokens=["<|eot_id|>"])
+        tk.save(str(tmp_path / "tokenizer.json"))
+        bt = BpeTokenizer.from_file(str(tmp_path / "tokenizer.json"))
+        masker = build_masker(bt)
+        assert masker is not None
+        for seed in range(4):
+            rng = random.Random(seed)
+            fsm = JsonFsm(SCHEMA_NESTED)
+            out_ids = []
+            for _ in range(3000):
+                allowed = masker.allowed_tokens(fsm)
+                if not allowed:
+                    break
+                tid = rng.choice(sorted(allowed))
+                out_ids.append(tid)
+                masker.advance_token(fsm, masker.token_bytes[tid])
+            else:
+                raise AssertionError("walk did not terminate")
+            data = json.loads(bt.decode(out_ids))
+            assert len(data["summary"]) <= 40
+            assert isinstance(data["confidence"], (int, float))
+            assert data["nested"]["sev"] in ("low", "high")
+            assert 1 <= len(data["tags"]) <= 3
+
+    def test_engine_constrained_decoding_on_checkpoint(self, ckpt_dir):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(device="cpu", background=False, checkpoint=ckpt_dir,
+                        kv_blocks=128)
+        try:
+            assert eng.supports_bpe_grammar
+            ids = eng.hf_tokenizer.encode_chat("You are Runbook.",
+                                               "status of checkout?")
+            req = eng.generate(ids, max_new_tokens=256, schema=SCHEMA_NESTED)
+            assert req.error == ""
+            text = eng.hf_tokenizer.decode(req.out_ids)
+            data = json.loads(text)   # random weights, still schema-valid
+            assert set(SCHEMA_NESTED["required"]) <= set(data)
+        finally:
+            eng.shutdown()
+
+    def test_client_passes_schema_through(self, ckpt_dir):
+        from runbookai_amd.engine.client import LocalEngineClient
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(device="cpu", background=False, checkpoint=ckpt_dir,
+                        kv_blocks=384)
+        try:
+            client = LocalEngineClient(eng, max_tokens=2048)
+            out = client.complete("what failed?")   # free text (no tag)
+            # triage default path: complete() without a tag is free text;
+            # force a schema'd call through chat():
+            from runbookai_amd.agent.llm_parser import SCHEMA_TAG_PREFIX
+
+            resp = client.chat("sys",
+                               f"{SCHEMA_TAG_PREFIX}triage\x00what failed?")
+            data = json.loads(resp.content)
+            assert "summary" in data and "severity" in data
+        finally:
+            eng.shutdown()
+
+    def test_oversized_vocab_falls_back(self, monkeypatch, ckpt_dir):
+        from runbookai_amd.engine import grammar_bpe
+        from runbookai_amd.engine.engine import LLMEngine
+
+        monkeypatch.setattr(grammar_bpe, "MAX_VOCAB", 10)
+        eng = LLMEngine(device="cpu", background=False, checkpoint=ckpt_dir,
+                        kv_blocks=128)
+        try:
+            assert not eng.supports_bpe_grammar
+            with pytest.raises(ValueError):
+                eng.submit([1, 2], schema={"type": "object", "properties": {}})
         finally:
             eng.shutdown()
 
