@@ -305,6 +305,36 @@ class TestBpeGrammar:
         finally:
             eng.shutdown()
 
+    @pytest.mark.timeout(600)
+    def test_full_investigation_on_bpe_checkpoint(self, ckpt_dir):
+        """The whole product loop — orchestrator, real tool registry,
+        simulated incident — served by a trained-format checkpoint with
+        BPE-grammar constrained decoding. Every phase parses."""
+        from runbookai_amd.agent.orchestrator import InvestigationOrchestrator
+        from runbookai_amd.engine.client import LocalEngineClient
+        from runbookai_amd.engine.engine import LLMEngine
+        from runbookai_amd.providers.simulation import SimScenario, set_scenario
+        from runbookai_amd.tools.registry import ToolRegistry
+
+        set_scenario(SimScenario.redis_exhaustion())
+        eng = LLMEngine(device="cpu", background=True, checkpoint=ckpt_dir,
+                        kv_blocks=512)
+        try:
+            client = LocalEngineClient(eng, max_tokens=1024)
+            orch = InvestigationOrchestrator(llm=client,
+                                             tool_executor=ToolRegistry(),
+                                             max_iterations=2,
+                                             queries_per_hypothesis=1)
+            result = orch.investigate("checkout latency spike",
+                                      incident_id="PD-EXAMPLE-001")
+            d = result.to_dict()
+            assert d["phasesVisited"]
+            assert "conclude" in d["phasesVisited"] or d["error"]
+            assert eng.stats["requests"] >= 3
+        finally:
+            eng.shutdown()
+            set_scenario(None)
+
     def test_oversized_vocab_falls_back(self, monkeypatch, ckpt_dir):
         from runbookai_amd.engine import grammar_bpe
         from runbookai_amd.engine.engine import LLMEngine
