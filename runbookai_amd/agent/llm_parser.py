@@ -81,6 +81,14 @@ def _normalize_json_text(text: str) -> str:
 
 
 def parse_json(text: str) -> Any:
+    # clean output parses as-is — salvage heuristics (fence stripping,
+    # brace scanning) only run on failure, so valid JSON whose STRING
+    # content contains ``` or braces is never mangled (schema fuzzing,
+    # tests/test_llm_parser_fuzz.py)
+    try:
+        return json.loads(text.strip())
+    except (json.JSONDecodeError, AttributeError):
+        pass
     raw = extract_json(text)
     try:
         return json.loads(raw)

@@ -10,10 +10,15 @@ step). Byte-level BPE vocabs contain all single-byte tokens, so the
 allowed set is never empty until the grammar completes — progress is
 guaranteed.
 
-This masker is pure Python and scales with reachable trie nodes per
-step; it is gated to vocabs ≤ RUNBOOKAI_BPE_GRAMMAR_MAX_VOCAB (default
-16384). A native trie walker for the full 128k Llama-3 vocab is roadmap
-work (docs/ROADMAP.md item 7).
+Scaling: the only state with wide byte fan-out is STRING CONTENT (~95
+legal bytes); there the admissible set is precomputed — every
+all-printable quote-free token of length ≤ remaining capacity is legal
+and leaves the FSM in-string, so the mask is a length-bucket lookup plus
+a walk over the small subtrie of quote-bearing tokens (they may close
+the string and continue structure). Structural/enum/number states
+branch over a handful of bytes, so the full-trie walk stays cheap.
+Together this handles the full 128k Llama-3 vocab in Python; the gate
+RUNBOOKAI_BPE_GRAMMAR_MAX_VOCAB (default 200000) is a safety valve.
 """
 from __future__ import annotations
 
@@ -21,9 +26,9 @@ import os
 from typing import Optional
 
 from .bpe_tokenizer import BpeTokenizer, unicode_to_bytes
-from .json_fsm import NUMBER_CLOSE_SENTINEL, JsonFsm
+from .json_fsm import _STRING_BYTES, NUMBER_CLOSE_SENTINEL, JsonFsm
 
-MAX_VOCAB = int(os.environ.get("RUNBOOKAI_BPE_GRAMMAR_MAX_VOCAB", "16384"))
+MAX_VOCAB = int(os.environ.get("RUNBOOKAI_BPE_GRAMMAR_MAX_VOCAB", "200000"))
 
 
 class _TrieNode:
@@ -51,7 +56,10 @@ class GrammarTokenMasker:
     def __init__(self, tokenizer: BpeTokenizer) -> None:
         self.vocab_size = max(tokenizer.vocab.values(), default=0) + 1
         self.root = _TrieNode()
+        self.string_root = _TrieNode()   # quote-bearing / non-printable tokens
         self.token_bytes: dict[int, bytes] = {}
+        safe = set(_STRING_BYTES)
+        by_len: dict[int, list[int]] = {}
         special_ids = set(tokenizer.special_tokens.values())
         for tok, tid in tokenizer.vocab.items():
             if tid in special_ids:
@@ -67,6 +75,24 @@ class GrammarTokenMasker:
                     nxt = node.children.setdefault(b, _TrieNode())
                 node = nxt
             node.ends.append(tid)
+            if all(b in safe for b in bs):
+                by_len.setdefault(len(bs), []).append(tid)
+            else:
+                node = self.string_root
+                for b in bs:
+                    nxt = node.children.get(b)
+                    if nxt is None:
+                        nxt = node.children.setdefault(b, _TrieNode())
+                    node = nxt
+                node.ends.append(tid)
+        # string fast path: prefix lists — safe_upto[L] = every quote-free
+        # printable token of length <= L
+        self.max_safe_len = max(by_len, default=0)
+        self.safe_upto: list[list[int]] = [[]]
+        acc: list[int] = []
+        for length in range(1, self.max_safe_len + 1):
+            acc = acc + by_len.get(length, [])
+            self.safe_upto.append(acc)
         self.eot_id = tokenizer.eot_id
 
     # -- per-step mask ------------------------------------------------------------
@@ -78,6 +104,16 @@ class GrammarTokenMasker:
         if fsm.done:
             return []
         out: list[int] = []
+        st = fsm.string_state()
+        if st is not None:
+            chars, cap = st
+            # fast path: quote-free printable tokens of length <= capacity
+            # stay inside the string — no per-token FSM walk needed
+            out.extend(self.safe_upto[min(cap, self.max_safe_len)])
+            # quote-bearing tokens may close the string and continue into
+            # structure: walk just that subtrie (closing needs >=1 char)
+            self._walk(self.string_root, fsm, out)
+            return out
         self._walk(self.root, fsm, out)
         return out
 

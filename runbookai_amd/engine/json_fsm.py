@@ -68,6 +68,19 @@ class JsonFsm:
         frame = self._stack[-1]
         return self._frame_allowed(frame)
 
+    def string_state(self) -> Optional[tuple[int, int]]:
+        """(chars_emitted, remaining_capacity) when the FSM sits inside a
+        string VALUE with no forced bytes pending; None otherwise. The BPE
+        masker's fast path keys on this (grammar_bpe.py)."""
+        if self._pending or self._done or not self._stack:
+            return None
+        frame = self._stack[-1]
+        if frame.get("kind") != "string":
+            return None
+        cap = max(0, min(frame["max"] - frame["len"],
+                         self.max_total_bytes - self.emitted))
+        return frame["len"], cap
+
     def string_capacity(self) -> int:
         """Remaining free-content bytes when the FSM sits INSIDE a string
         value (>=1 char already emitted, no forced bytes pending); 0

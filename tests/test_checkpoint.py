@@ -268,6 +268,75 @@ class TestBpeGrammar:
             assert data["nested"]["sev"] in ("low", "high")
             assert 1 <= len(data["tags"]) <= 3
 
+    def test_string_fast_path_equals_brute_force(self, tmp_path):
+        """The precomputed in-string token set + quote-subtrie walk must
+        equal the exhaustive full-trie walk at EVERY step of random
+        decodes (the fast path is what makes 128k vocabs tractable)."""
+        import random
+
+        from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer
+        from runbookai_amd.engine.grammar_bpe import build_masker
+        from runbookai_amd.engine.json_fsm import JsonFsm
+
+        tokenizers = pytest.importorskip("tokenizers")
+        tk = tokenizers.ByteLevelBPETokenizer()
+        tk.train_from_iterator(
+            ['say "high" or "low", confidence 0.9!', "redis pool exhausted"] * 30,
+            vocab_size=600, min_frequency=1, special_tokens=["<|eot_id|>"])
+        tk.save(str(tmp_path / "tokenizer.json"))
+        masker = build_masker(BpeTokenizer.from_file(str(tmp_path / "tokenizer.json")))
+
+        def brute(fsm):
+            out = []
+            masker._walk(masker.root, fsm, out)
+            return sorted(out)
+
+        for seed in range(3):
+            rng = random.Random(seed)
+            fsm = JsonFsm(SCHEMA_NESTED)
+            for _ in range(2000):
+                fast = sorted(masker.allowed_tokens(fsm))
+                assert fast == brute(fsm.clone())
+                if not fast:
+                    break
+                tid = rng.choice(fast)
+                masker.advance_token(fsm, masker.token_bytes[tid])
+            else:
+                raise AssertionError("walk did not terminate")
+
+    def test_large_vocab_step_is_fast(self):
+        """Synthetic ~33k vocab: a string-state masking step must use the
+        fast path (well under the time a full-trie walk would take)."""
+        import itertools
+        import string
+        import time
+
+        from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer, bytes_to_unicode
+        from runbookai_amd.engine.grammar_bpe import GrammarTokenMasker
+        from runbookai_amd.engine.json_fsm import JsonFsm
+
+        b2u = bytes_to_unicode()
+        vocab = {b2u[b]: b for b in range(256)}
+        alpha = string.ascii_lowercase
+        for i, (a, b, c) in enumerate(itertools.islice(
+                itertools.product(alpha, alpha, alpha), 17000)):
+            vocab[f"{b2u[ord(a)]}{b2u[ord(b)]}{b2u[ord(c)]}"] = 256 + i
+        for i, (a, b) in enumerate(itertools.product(alpha + '"', alpha)):
+            vocab[f"{b2u[ord(a)]}{b2u[ord(b)]}"] = 256 + 17000 + i
+        class FakeTok:
+            special_tokens = {"<|eot_id|>": len(vocab)}
+            eot_id = len(vocab)
+        FakeTok.vocab = vocab
+        masker = GrammarTokenMasker(FakeTok)
+        fsm = JsonFsm({"type": "object", "properties": {
+            "s": {"type": "string", "maxLength": 60}}, "required": ["s"]})
+        masker.advance_token(fsm, b'{"s": "ab')
+        t0 = time.time()
+        allowed = masker.allowed_tokens(fsm)
+        dt = time.time() - t0
+        assert len(allowed) > 15000          # the safe bucket is in play
+        assert dt < 0.2, f"string-state step took {dt*1000:.0f} ms"
+
     def test_engine_constrained_decoding_on_checkpoint(self, ckpt_dir):
         from runbookai_amd.engine.engine import LLMEngine
 
