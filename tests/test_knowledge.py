@@ -258,3 +258,22 @@ class TestServiceGraph:
 
     def test_stats(self):
         assert self._graph().stats() == {"nodes": 5, "edges": 4}
+
+
+class TestFtsAdversarialQueries:
+    def test_search_never_raises_on_hostile_input(self):
+        """User queries flow into FTS5 MATCH: operators, quotes, column
+        filters and injection attempts must all degrade to sanitized term
+        search, never a sqlite error."""
+        store = KnowledgeStore(":memory:")
+        for doc in load_from_filesystem(EXAMPLES):
+            store.upsert_document(doc)
+        hostile = ['"', "'", "AND OR NOT", "(((", "*",
+                   "a;DROP TABLE chunks;--", "漢字 テスト", "", "  ", "-",
+                   "a-b c_d", "near(a,b)", '"unclosed', "NOT redis",
+                   "^start", "col:value", "redis OR ' OR 1=1 --"]
+        for q in hostile:
+            hits = store.search(q, limit=3)
+            assert isinstance(hits, list), q
+        # and the injection attempt did not damage the table
+        assert store.search("redis connection pool")
