@@ -79,6 +79,7 @@ class Request:
     prefill_tokens: list[int] = field(default_factory=list)  # prompt + forced prefix
     cached_len: int = 0              # prefix tokens served from the KV prefix pool
     registered: bool = False         # prompt blocks published to the prefix pool
+    dedup_deferred: bool = False     # held back one round behind a same-prefix twin
     state: str = "waiting"           # waiting | running | done
     done_event: threading.Event = field(default_factory=threading.Event)
     prompt_len: int = 0
@@ -322,15 +323,17 @@ class LLMEngine:
         batch: list[Request] = []
         tokens = 0
         kv = self.model.kv
-        while self.waiting and len(self.running) + len(batch) < self.max_batch:
-            req = self.waiting[0]
+        admitted_first: set = set()   # first full block of each admission this call
+        i = 0
+        while i < len(self.waiting) and len(self.running) + len(batch) < self.max_batch:
+            req = self.waiting[i]
             need = len(req.pending_input) + req.max_new_tokens
             if batch and tokens + len(req.pending_input) > self.max_prefill_tokens:
                 break
             if need > (kv.num_blocks - 1) * kv.block_size:
                 # can NEVER fit, even with the whole pool free: fail loudly
                 # instead of starving the queue forever
-                self.waiting.pop(0)
+                self.waiting.pop(i)
                 req.error = (f"request needs {need} KV tokens but the pool holds "
                              f"{(kv.num_blocks - 1) * kv.block_size}; lower "
                              f"max_new_tokens or raise kv_blocks")
@@ -339,12 +342,26 @@ class LLMEngine:
                 continue
             if not kv.can_allocate(need):
                 break
+            fb = (tuple(req.pending_input[:kv.block_size])
+                  if len(req.pending_input) >= kv.block_size else None)
+            if (self.prefix_cache and fb is not None and fb in admitted_first
+                    and not req.dedup_deferred):
+                # same first block as a request admitted THIS call: its
+                # prefix isn't registered yet, so admitting now would
+                # re-prefill it — hold back ONE round (a cold 32-way wave
+                # of identical system prompts prefills once, not 32×);
+                # the flag guarantees admission next round regardless
+                req.dedup_deferred = True
+                i += 1
+                continue
             cached = 0
             if self.prefix_cache:
                 cached = kv.allocate_with_prefix(req.rid, req.pending_input, need)
             else:
                 kv.allocate(req.rid, need)
-            self.waiting.pop(0)
+            self.waiting.pop(i)
+            if fb is not None:
+                admitted_first.add(fb)
             if cached:
                 # prefix served from the pool: skip straight to running with
                 # only the suffix pending — it flows through the chunked-

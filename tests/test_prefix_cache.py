@@ -150,6 +150,51 @@ class TestEngineParity:
         for r in reqs:
             assert r.error == ""
 
+    def test_cold_wave_dedup_prefills_shared_prefix_once(self):
+        """Simultaneous identical prompts (a cold 32-way wave): the first
+        admits, twins defer ONE round and then hit the registered prefix —
+        the shared blocks are computed once, outputs stay identical."""
+        eng = LLMEngine(model="tiny", device="cpu", background=False,
+                        prefix_cache=True, kv_blocks=256)
+        prompt = list(range(2, 120))   # 7 full blocks
+        reqs = [eng.submit(prompt, max_new_tokens=8) for _ in range(4)]
+        eng.run_until_idle()
+        assert all(r.error == "" for r in reqs)
+        assert len({tuple(r.out_ids) for r in reqs}) == 1
+        # 3 of 4 served the 112-token prefix from the pool
+        assert eng.stats["cached_prefix_tokens"] == 3 * 112
+        assert any(r.dedup_deferred for r in reqs)
+
+    def test_distinct_prompts_not_deferred(self):
+        eng = LLMEngine(model="tiny", device="cpu", background=False,
+                        prefix_cache=True, kv_blocks=256)
+        reqs = [eng.submit([10 + i] * 60, max_new_tokens=4) for i in range(4)]
+        eng.run_until_idle()
+        assert all(r.error == "" for r in reqs)
+        assert not any(r.dedup_deferred for r in reqs)
+        assert eng.stats["cached_prefix_tokens"] == 0
+
+    def test_deferred_twin_completes_when_first_fails(self):
+        """The defer-once flag guarantees admission even if the twin's
+        prefill step dies."""
+        eng = LLMEngine(model="tiny", device="cpu", background=False,
+                        prefix_cache=True, kv_blocks=256)
+        boom = {"armed": True}
+        orig = eng.model.prefill
+
+        def flaky(*a, **kw):
+            if boom.pop("armed", False):
+                raise RuntimeError("chaos")
+            return orig(*a, **kw)
+
+        eng.model.prefill = flaky
+        prompt = list(range(3, 90))
+        r1 = eng.submit(prompt, max_new_tokens=6)
+        r2 = eng.submit(prompt, max_new_tokens=6)
+        eng.run_until_idle()
+        assert "chaos" in r1.error
+        assert r2.error == "" and r2.out_ids
+
     def test_free_text_requests_cache_too(self):
         eng = LLMEngine(model="tiny", device="cpu", background=False,
                         prefix_cache=True, kv_blocks=256)
