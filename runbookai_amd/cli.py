@@ -549,6 +549,40 @@ def checkpoint_delete(investigation_id: str, checkpoint_id: Optional[str]) -> No
     _echo(f"deleted {removed} checkpoint(s)")
 
 
+# -- metrics ---------------------------------------------------------------------
+
+@cli.group()
+def metrics() -> None:
+    """Prometheus metrics for the serving engine."""
+
+
+@metrics.command("serve")
+@click.option("--port", type=int, default=9464)
+@click.option("--addr", default="127.0.0.1")
+@click.pass_context
+def metrics_serve(ctx: click.Context, port: int, addr: str) -> None:
+    """Expose engine counters + KV pool occupancy on /metrics (the
+    reference ships no metrics backend — SURVEY §5)."""
+    import time as _time
+
+    from .engine.metrics import serve_metrics
+    from .model.client import create_llm_client
+
+    config = ctx.obj["config"]
+    client = create_llm_client(config.llm.model_dump(by_alias=False))
+    engine = getattr(client, "engine", None)
+    if engine is None:
+        _echo(f"{RED}metrics need the local engine (llm.provider=local){RESET}")
+        sys.exit(1)
+    serve_metrics(engine, port=port, addr=addr)
+    _echo(f"metrics on http://{addr}:{port}/metrics (ctrl-c to stop)")
+    try:
+        while True:
+            _time.sleep(3600)
+    except KeyboardInterrupt:
+        pass
+
+
 # -- mcp -------------------------------------------------------------------------
 
 @cli.group()
