@@ -282,6 +282,17 @@ class LLMEngine:
             self._lock.notify_all()
         if self._thread:
             self._thread.join(timeout=5.0)
+        # fail anything still in flight so callers unblock immediately
+        # instead of sitting in done_event.wait until their timeout
+        with self._lock:
+            for req in self.running + self.waiting + self._step_batch:
+                if req.state != "done":
+                    req.error = req.error or "engine shut down"
+                    req.state = "done"
+                    req.done_event.set()
+            self.running.clear()
+            self.waiting.clear()
+            self._step_batch = []
 
     # -- scheduling ------------------------------------------------------------------
 

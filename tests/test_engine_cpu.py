@@ -353,3 +353,14 @@ class TestLocalClientOrchestration:
         finally:
             eng.shutdown()
             set_scenario(None)
+
+
+class TestShutdown:
+    def test_shutdown_unblocks_in_flight_requests(self):
+        eng = LLMEngine(model="tiny", device="cpu", background=True, kv_blocks=128)
+        reqs = [eng.submit(eng.tokenizer.encode_chat("sys", f"q{i}"),
+                           max_new_tokens=512) for i in range(4)]
+        eng.shutdown()
+        for r in reqs:
+            assert r.done_event.wait(timeout=10)
+            assert r.state == "done"
