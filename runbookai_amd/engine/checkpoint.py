@@ -48,6 +48,22 @@ def config_from_hf(path: str, name: str = "hf") -> LlamaConfig:
     )
 
 
+def unpermute_rope_rows(w: torch.Tensor, n_heads: int, head_dim: int) -> torch.Tensor:
+    """HF-format Llama q_proj/k_proj rows are permuted per head for the
+    rotate-half RoPE convention ([even dims | odd dims]); this engine's
+    RoPE rotates interleaved pairs (ops/reference.py apply_rope). Inverse
+    of the HF convert-script permute: HF row i of a head maps to
+    interleaved row 2i (i < d/2) or 2(i-d/2)+1."""
+    d, H = head_dim, w.shape[-1]
+    return w.view(n_heads, 2, d // 2, H).transpose(1, 2).reshape(n_heads * d, H)
+
+
+def permute_rope_rows(w: torch.Tensor, n_heads: int, head_dim: int) -> torch.Tensor:
+    """Interleaved-pair rows -> HF rotate-half row order (export side)."""
+    d, H = head_dim, w.shape[-1]
+    return w.view(n_heads, d // 2, 2, H).transpose(1, 2).reshape(n_heads * d, H)
+
+
 class _ShardReader:
     """name -> tensor across one or many .safetensors files, opened lazily."""
 
@@ -117,8 +133,10 @@ def load_hf_checkpoint(model: LlamaModel, path: str,
         p = f"model.layers.{i}."
         from .llama import kv_shard_range
 
-        q = take(p + "self_attn.q_proj.weight")   # [hq*d, H]
-        k = take(p + "self_attn.k_proj.weight")   # [hk*d, H]
+        # un-permute q/k from HF rotate-half row order to this engine's
+        # interleaved-pair RoPE convention (see unpermute_rope_rows)
+        q = unpermute_rope_rows(take(p + "self_attn.q_proj.weight"), hq, d)
+        k = unpermute_rope_rows(take(p + "self_attn.k_proj.weight"), hk, d)
         v = take(p + "self_attn.v_proj.weight")
         q_r = q.view(hq, d, H)[rank * hq_r:(rank + 1) * hq_r].reshape(hq_r * d, H)
         ks, ke = kv_shard_range(hk, tp, rank)
@@ -160,9 +178,18 @@ def load_model(path: str, device: str = "cpu", tp: Optional[int] = None,
     architectures reuse their tuned CONFIGS entry (graph sizes etc.)."""
     cfg = config_from_hf(path, name=name or os.path.basename(os.path.normpath(path)))
     for known in CONFIGS.values():
+        # adopt a tuned CONFIGS entry only on a FULL architectural match —
+        # matching on hidden/layers/heads alone would silently swap in the
+        # wrong kv-heads/intermediate/vocab/rope for lookalike models
+        # (e.g. Llama-2-7B vs llama3-8b, both 4096/32/32)
         if (known.hidden_size == cfg.hidden_size
                 and known.num_layers == cfg.num_layers
-                and known.num_heads == cfg.num_heads):
+                and known.num_heads == cfg.num_heads
+                and known.num_kv_heads == cfg.num_kv_heads
+                and known.intermediate_size == cfg.intermediate_size
+                and known.vocab_size == cfg.vocab_size
+                and known.head_dim == cfg.head_dim
+                and known.rope_theta == cfg.rope_theta):
             cfg = known
             break
     model = LlamaModel(cfg, device=device, tp=tp, kv_blocks=kv_blocks, **kwargs)
@@ -190,8 +217,11 @@ def export_hf_checkpoint(model: LlamaModel, path: str) -> None:
     for i, layer in enumerate(model.layers):
         p = f"model.layers.{i}."
         w = layer.qkv.weight
-        state[p + "self_attn.q_proj.weight"] = w[:hq * d]
-        state[p + "self_attn.k_proj.weight"] = w[hq * d:(hq + hk) * d]
+        # permute q/k back to HF rotate-half row order so exports are true
+        # HF checkpoints (and load()'s un-permute round-trips exactly)
+        state[p + "self_attn.q_proj.weight"] = permute_rope_rows(w[:hq * d], hq, d)
+        state[p + "self_attn.k_proj.weight"] = permute_rope_rows(
+            w[hq * d:(hq + hk) * d], hk, d)
         state[p + "self_attn.v_proj.weight"] = w[(hq + hk) * d:]
         state[p + "self_attn.o_proj.weight"] = layer.o_proj.weight
         state[p + "mlp.gate_proj.weight"] = layer.gate_up.weight[:inter]

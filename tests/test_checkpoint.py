@@ -46,6 +46,76 @@ class TestCheckpointRoundTrip:
         out_dst = dst.prefill(ids, pos, starts, dst.kv.slot_mapping(1, 0, T))
         assert torch.equal(out_src, out_dst)
 
+    def test_hf_rotate_half_convention(self, tmp_path):
+        """A checkpoint authored in HF convention (q/k rows permuted for
+        rotate-half RoPE) must produce the same attention scores through
+        this engine's interleaved-pair RoPE as an HF-reference forward.
+
+        Reference: HF apply_rotary_pos_emb rotates [x1|x2] halves with
+        cos/sin repeated twice; our apply_rope rotates interleaved pairs.
+        The invariant is q.k dot products (attention scores), which are
+        permutation-covariant: equal iff the loader un-permutes rows."""
+        from runbookai_amd.engine.checkpoint import unpermute_rope_rows
+        from runbookai_amd.ops.reference import apply_rope, rope_cos_sin
+
+        torch.manual_seed(0)
+        H, d, heads, T = 32, 16, 2, 5
+        w_hf_q = torch.randn(heads * d, H)
+        w_hf_k = torch.randn(heads * d, H)
+        x = torch.randn(T, H)
+        pos = torch.arange(T, dtype=torch.int64)
+
+        # --- HF reference: rotate-half on the HF-ordered projections ---
+        def hf_rope(v):  # v: [T, heads, d]
+            inv = 1.0 / (10000.0 ** (torch.arange(0, d, 2).float() / d))
+            freqs = torch.outer(pos.float(), inv)           # [T, d/2]
+            emb = torch.cat([freqs, freqs], dim=-1)          # [T, d]
+            cos, sin = emb.cos()[:, None, :], emb.sin()[:, None, :]
+            x1, x2 = v[..., : d // 2], v[..., d // 2:]
+            return v * cos + torch.cat([-x2, x1], dim=-1) * sin
+
+        q_hf = hf_rope((x @ w_hf_q.T).view(T, heads, d))
+        k_hf = hf_rope((x @ w_hf_k.T).view(T, heads, d))
+        scores_hf = torch.einsum("qhd,khd->hqk", q_hf, k_hf)
+
+        # --- this engine: un-permuted rows + interleaved-pair RoPE ---
+        w_q = unpermute_rope_rows(w_hf_q, heads, d)
+        w_k = unpermute_rope_rows(w_hf_k, heads, d)
+        cos, sin = rope_cos_sin(T, d, theta=10000.0)
+        q_e, k_e = apply_rope((x @ w_q.T).view(T, heads, d),
+                              (x @ w_k.T).view(T, heads, d), cos, sin,
+                              pos.to(torch.int32))
+        scores_e = torch.einsum("qhd,khd->hqk", q_e.float(), k_e.float())
+        assert torch.allclose(scores_hf, scores_e, atol=1e-4), \
+            (scores_hf - scores_e).abs().max()
+
+        # and WITHOUT the un-permute the scores must differ (guards against
+        # the test passing vacuously)
+        q_bad, k_bad = apply_rope((x @ w_hf_q.T).view(T, heads, d),
+                                  (x @ w_hf_k.T).view(T, heads, d), cos, sin,
+                                  pos.to(torch.int32))
+        scores_bad = torch.einsum("qhd,khd->hqk", q_bad.float(), k_bad.float())
+        assert not torch.allclose(scores_hf, scores_bad, atol=1e-3)
+
+    def test_config_entry_not_adopted_on_partial_match(self, tmp_path):
+        """A lookalike architecture (same hidden/layers/heads, different
+        kv-heads/intermediate/vocab) must NOT silently adopt a tuned
+        CONFIGS entry — the checkpoint's own config wins."""
+        src = tiny_model(seed=7)
+        export_hf_checkpoint(src, str(tmp_path))
+        cfg_path = tmp_path / "config.json"
+        cfg = json.loads(cfg_path.read_text())
+        cfg["num_key_value_heads"] = 4    # tiny has 2
+        cfg_path.write_text(json.dumps(cfg))
+        # weights no longer match the config; just check config resolution
+        loaded_cfg = config_from_hf(str(tmp_path))
+        from runbookai_amd.engine.checkpoint import CONFIGS as _C
+        assert loaded_cfg.num_kv_heads == 4
+        for known in _C.values():
+            assert not (known.hidden_size == loaded_cfg.hidden_size
+                        and known.num_kv_heads == loaded_cfg.num_kv_heads
+                        and known.num_layers == loaded_cfg.num_layers)
+
     def test_load_differs_from_fresh_random(self, tmp_path):
         src = tiny_model(seed=7)
         export_hf_checkpoint(src, str(tmp_path))
