@@ -75,8 +75,26 @@ class PendingApprovalStore:
 
 
 class ApprovalWebhook:
-    def __init__(self, store: Optional[PendingApprovalStore] = None) -> None:
+    """Approval-button interaction server.
+
+    With a `signing_secret`, every POST must carry a valid Slack v0
+    signature (X-Slack-Signature over `v0:<timestamp>:<body>`, constant-
+    time compare, stale timestamps rejected) BEFORE any approval
+    resolves — an unsigned request must never flip a high/critical
+    remediation gate (reference src/webhooks/slack-webhook.ts verifies
+    the same way). Without a secret (local dev), requests pass."""
+
+    def __init__(self, store: Optional[PendingApprovalStore] = None,
+                 signing_secret: str = "") -> None:
         self.store = store or PendingApprovalStore()
+        self.signing_secret = signing_secret
+
+    def verify(self, timestamp: str, body: bytes, signature: str) -> bool:
+        if not self.signing_secret:
+            return True
+        from ..slack.gateway import verify_signature
+
+        return verify_signature(self.signing_secret, timestamp, body, signature)
 
     def handle_interaction(self, payload: dict[str, Any]) -> dict[str, Any]:
         actions = payload.get("actions", [])
@@ -97,7 +115,14 @@ class ApprovalWebhook:
         class Handler(BaseHTTPRequestHandler):
             def do_POST(self) -> None:  # noqa: N802
                 length = int(self.headers.get("Content-Length", 0))
-                body = self.rfile.read(length).decode("utf-8", "replace")
+                raw = self.rfile.read(length)
+                if not webhook.verify(
+                        self.headers.get("X-Slack-Request-Timestamp", ""),
+                        raw, self.headers.get("X-Slack-Signature", "")):
+                    self.send_response(401)
+                    self.end_headers()
+                    return
+                body = raw.decode("utf-8", "replace")
                 # Slack sends interactions as form-encoded payload=<json>
                 try:
                     if body.startswith("payload="):

@@ -51,3 +51,32 @@ def test_slack_approval_timeout_denies(tmp_path):
         assert not rec.approved
     finally:
         set_scenario(None)
+
+
+def test_webhook_rejects_unsigned_request(tmp_path):
+    """With a signing secret configured, an unsigned/forged POST must not
+    resolve an approval (advisor finding: the approval gate was forgeable)."""
+    import hashlib
+    import hmac as hmac_mod
+    import time as time_mod
+
+    store = PendingApprovalStore(str(tmp_path / "pending"))
+    approval_id = store.create({"operation": "scale-down", "resource": "prod"})
+    webhook = ApprovalWebhook(store, signing_secret="s3cret")
+
+    body = b'{"actions":[{"value":"approve:' + approval_id.encode() + b'"}]}'
+    ts = str(int(time_mod.time()))
+
+    # forged: bad signature
+    assert not webhook.verify(ts, body, "v0=deadbeef")
+    # stale timestamp: valid HMAC but older than the replay window
+    old_ts = str(int(time_mod.time()) - 3600)
+    stale_sig = "v0=" + hmac_mod.new(
+        b"s3cret", f"v0:{old_ts}:".encode() + body, hashlib.sha256).hexdigest()
+    assert not webhook.verify(old_ts, body, stale_sig)
+    # properly signed: passes
+    good_sig = "v0=" + hmac_mod.new(
+        b"s3cret", f"v0:{ts}:".encode() + body, hashlib.sha256).hexdigest()
+    assert webhook.verify(ts, body, good_sig)
+    # and without a secret (local dev) verification is a no-op
+    assert ApprovalWebhook(store).verify(ts, body, "")
