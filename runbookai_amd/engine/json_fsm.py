@@ -154,6 +154,7 @@ class JsonFsm:
         elif t in ("integer", "number"):
             self._stack.append({"kind": "number", "float": t == "number",
                                 "len": 0, "max_len": 10, "has_dot": False,
+                                "lit": "",
                                 "minimum": schema.get("minimum"),
                                 "maximum": schema.get("maximum")})
         elif t == "boolean":
@@ -197,9 +198,9 @@ class JsonFsm:
             return _STRING_BYTES + [0x22]
         if kind == "number":
             if frame["len"] == 0:
-                return _DIGITS
+                return self._number_filter(frame, _DIGITS)
             if frame.get("dot_pending"):
-                return _DIGITS  # '.' must be followed by a digit
+                return self._number_filter(frame, _DIGITS)  # '.' needs a digit
             if frame["len"] >= frame["max_len"]:
                 return [0x00]  # sentinel handled in advance: close number
             if frame.get("leading_zero") and not frame["has_dot"]:
@@ -207,12 +208,12 @@ class JsonFsm:
                 allowed = [0x00]
                 if frame["float"]:
                     allowed.append(0x2E)
-                return allowed
+                return self._number_filter(frame, allowed)
             allowed = list(_DIGITS)
             if frame["float"] and not frame["has_dot"]:
                 allowed.append(0x2E)  # '.'
             allowed.append(0x00)
-            return allowed
+            return self._number_filter(frame, allowed)
         if kind in ("enum", "enum_raw"):
             progress = frame["progress"]
             nexts = {ord(o[len(progress)]) for o in frame["options"]
@@ -234,6 +235,75 @@ class JsonFsm:
                     choices.append(0x5D)  # ']'
                 return choices or [0x5D]
         return []
+
+    def _number_filter(self, frame: dict[str, Any], allowed: list[int]) -> list[int]:
+        """Enforce schema minimum/maximum at the digit level: a byte stays
+        allowed only if SOME completion of the literal lies within bounds.
+        Appending digits (to the integer part or the fraction) only ever
+        increases the value, so completions of literal s span
+        [float(s), float(s + '9'*room)] — a byte survives iff that interval
+        intersects [minimum, maximum]. The close sentinel survives iff the
+        literal itself is within bounds. A schema whose bounds are
+        unsatisfiable within max_len keeps the unfiltered set (the parser's
+        downstream clamp still applies) rather than deadlocking the FSM."""
+        lo_b = frame.get("minimum")
+        hi_b = frame.get("maximum")
+        if lo_b is None and hi_b is None:
+            return allowed
+        lo_b = float("-inf") if lo_b is None else float(lo_b)
+        hi_b = float("inf") if hi_b is None else float(hi_b)
+        lit = frame["lit"]
+        room_total = frame["max_len"]
+        is_f = frame["float"]
+
+        def viable(s: str) -> bool:
+            """Can s complete (stop now or extend with digits) in bounds?
+            Fractional extensions of 'X.Y' span the contiguous interval
+            [v, v+ulp); INTEGER extensions of 'p' are the disjoint union of
+            [p*10^k, (p+1)*10^k) per appended digit count k — each k is
+            checked separately (a contiguous-interval shortcut would let
+            prefix '5' claim it reaches [100, 200])."""
+            room = room_total - len(s)
+            if "." in s:
+                if s.endswith("."):
+                    if room <= 0:
+                        return False
+                    return (float(s + "0") <= hi_b
+                            and float(s + "9" * room) >= lo_b)
+                v = float(s)
+                if lo_b <= v <= hi_b:
+                    return True
+                return room > 0 and v <= hi_b and float(s + "9" * room) >= lo_b
+            p = int(s)
+            if lo_b <= p <= hi_b:
+                return True
+            if is_f and room >= 2 and p <= hi_b and p + 1 > lo_b:
+                return True   # a fraction 'p.x' lands in [p, p+1)
+            if s == "0" or room <= 0:
+                return False  # JSON forbids digits after a leading zero
+            ten_k = 1
+            for _ in range(room):
+                ten_k *= 10
+                lo_k = p * ten_k
+                if lo_k > hi_b:
+                    return False  # grows monotonically with k
+                hi_k = lo_k + ten_k - (0 if is_f else 1)
+                if hi_k >= lo_b:
+                    return True
+            return False
+
+        out = []
+        for b in allowed:
+            if b == 0x00:
+                if lit and not lit.endswith(".") and lo_b <= float(lit) <= hi_b:
+                    out.append(b)
+            elif b == 0x2E:
+                if viable(lit + "."):
+                    out.append(b)
+            else:  # digit
+                if viable(lit + chr(b)):
+                    out.append(b)
+        return out or allowed
 
     def _array_item_start_bytes(self, frame: dict[str, Any]) -> list[int]:
         items = frame["items"]
@@ -273,6 +343,7 @@ class JsonFsm:
                 frame["dot_pending"] = False
             if frame["len"] == 0 and byte == 0x30:
                 frame["leading_zero"] = True
+            frame["lit"] += chr(byte)
             frame["len"] += 1
             return
         if kind in ("enum", "enum_raw"):

@@ -48,6 +48,36 @@ class TestBasics:
         out = drive({"type": "number"}, chooser)
         assert isinstance(json.loads(out), float)
 
+    @pytest.mark.parametrize("schema,lo,hi", [
+        ({"type": "number", "minimum": 0, "maximum": 1}, 0.0, 1.0),
+        ({"type": "integer", "minimum": 1, "maximum": 5}, 1, 5),
+        ({"type": "integer", "minimum": 100, "maximum": 200}, 100, 200),
+        ({"type": "number", "minimum": 0.5, "maximum": 2.5}, 0.5, 2.5),
+        ({"type": "integer", "maximum": 9}, float("-inf"), 9),
+        ({"type": "integer", "minimum": 42}, 42, float("inf")),
+    ])
+    def test_number_bounds_enforced_by_construction(self, schema, lo, hi):
+        """Digit-level min/max enforcement: ANY choice sequence the FSM
+        allows must decode inside the bounds (advisor finding: bounds were
+        stored but never enforced)."""
+        for seed in range(40):
+            out = drive(schema, random_chooser(seed))
+            val = json.loads(out)
+            assert lo <= val <= hi, (schema, out)
+
+    def test_number_bounds_prefix_trap(self):
+        """Prefix '5' can never reach [100, 200]: the first digit must be
+        restricted to 1 or 2 (a contiguous-interval check would miss it)."""
+        fsm = JsonFsm({"type": "integer", "minimum": 100, "maximum": 200})
+        allowed = {chr(b) for b in fsm.allowed_bytes()}
+        assert allowed == {"1", "2"}
+
+    def test_unsatisfiable_bounds_fall_back_unfiltered(self):
+        # min > max: keep the FSM alive (downstream clamping still applies)
+        out = drive({"type": "integer", "minimum": 9, "maximum": 1},
+                    lambda a: a[0])
+        json.loads(out)   # still valid JSON
+
     def test_boolean(self):
         out = drive({"type": "boolean"}, lambda a: a[0])
         assert json.loads(out) in (True, False)
