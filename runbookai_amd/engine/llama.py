@@ -49,6 +49,11 @@ CONFIGS: dict[str, LlamaConfig] = {
     "tiny": LlamaConfig(name="tiny", hidden_size=256, intermediate_size=512,
                         num_layers=2, num_heads=4, num_kv_heads=2, head_dim=64,
                         vocab_size=4096, max_seq_len=4096),
+    # 8-head test model: shards at tp=8 (kv replication path) on CPU gloo —
+    # the launch-readiness tier for the driver's 8-GPU scaling run
+    "tiny8": LlamaConfig(name="tiny8", hidden_size=256, intermediate_size=512,
+                         num_layers=2, num_heads=8, num_kv_heads=2, head_dim=32,
+                         vocab_size=4096, max_seq_len=4096),
     "llama3-8b": LlamaConfig(name="llama3-8b", hidden_size=4096, intermediate_size=14336,
                              num_layers=32, num_heads=32, num_kv_heads=8, head_dim=128),
     "llama3-70b": LlamaConfig(name="llama3-70b", hidden_size=8192, intermediate_size=28672,
