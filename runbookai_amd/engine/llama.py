@@ -249,7 +249,8 @@ class LlamaModel:
                      and self.cfg.head_dim == 128)
         if use_flash:
             # tiles built ONCE per call, not per layer
-            tb, tq = ops._build_qtiles(seq_starts.to(torch.int32))
+            tb, tq = ops._build_qtiles(seq_starts.to(torch.int32),
+                                       ops.PREFILL_QTILE)
             tb, tq = tb.to(device), tq.to(device)
 
             def attn_fn(q, k, v, i):
@@ -309,7 +310,7 @@ class LlamaModel:
             if out is not None:
                 return out
         # eager flash path: tiles built ONCE per call (not per layer)
-        tb, tq = ops._build_qtiles(seq_starts.to(torch.int32))
+        tb, tq = ops._build_qtiles(seq_starts.to(torch.int32), ops.CHUNK_QTILE)
         return self._chunk_impl(
             token_ids.long().to(device),
             positions.to(torch.int32).to(device),
@@ -494,15 +495,16 @@ class LlamaModel:
             return self._chunk_graphs[key]
         dev = self.device
         maxb = self._graph_max_blocks()
-        ntiles = tpad // 64 + bpad
+        qt = ops.CHUNK_QTILE
+        ntiles = tpad // qt + bpad
         scratch_slot = self.kv.scratch_block * self.kv.block_size
         starts = torch.full((bpad + 1,), tpad, dtype=torch.int32, device=dev)
         starts[0] = 0   # warmup layout: segment 0 covers all rows, rest empty
         tb = torch.full((ntiles,), bpad - 1, dtype=torch.int32, device=dev)
         tq = torch.full((ntiles,), tpad, dtype=torch.int32, device=dev)
-        for i in range(tpad // 64):
+        for i in range(tpad // qt):
             tb[i] = 0
-            tq[i] = i * 64
+            tq[i] = i * qt
         static = {
             "ids": torch.zeros(tpad, dtype=torch.long, device=dev),
             "pos": torch.zeros(tpad, dtype=torch.int32, device=dev),
@@ -578,13 +580,14 @@ class LlamaModel:
         starts_l = seq_starts.tolist()
         tb_host = torch.full_like(static["tb"], bpad - 1, device="cpu")
         tq_host = torch.full_like(static["tq"], tpad, device="cpu")
+        qt = ops.CHUNK_QTILE
         n = 0
         for b in range(B):
-            for q0 in range(starts_l[b], starts_l[b + 1], 64):
+            for q0 in range(starts_l[b], starts_l[b + 1], qt):
                 tb_host[n] = b
                 tq_host[n] = q0
                 n += 1
-        for q0 in range(T, tpad, 64):
+        for q0 in range(T, tpad, qt):
             tb_host[n] = B                       # padding segment
             tq_host[n] = q0
             n += 1

@@ -91,6 +91,15 @@ def apply_rope(q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.T
 
 USE_FLASH_PREFILL = True
 
+# v2 flash prefill (in-register softmax, 32x32 MFMA, swapped QK^T): 256-row
+# tiles (8-wave WGs) for packed prefill, 128-row (4-wave) for the paged
+# chunk path. RUNBOOKAI_FLASH_V2=0 falls back to the v1 16x16 kernel (A/B).
+import os as _os
+
+USE_FLASH_V2 = _os.environ.get("RUNBOOKAI_FLASH_V2", "1") != "0"
+PREFILL_QTILE = 256 if USE_FLASH_V2 else 64
+CHUNK_QTILE = 128 if USE_FLASH_V2 else 64
+
 
 def rope_store_kv_fused(q, k, v, cos, sin, positions, k_cache, v_cache, slot_mapping):
     """Fused RoPE + paged-KV scatter: returns the rotated q; the rotated k
@@ -113,12 +122,14 @@ def prefill_attention(q, k, v, seq_starts, causal: bool = True,
     if _on_gpu(q):
         starts_i32 = seq_starts.to(torch.int32)
         if USE_FLASH_PREFILL and q.shape[-1] == 128:
-            tb, tq = _build_qtiles(starts_i32)
-            return _get_ext().flash_prefill(q.contiguous(), k.contiguous(),
-                                            v.contiguous(), tb.to(q.device),
-                                            tq.to(q.device),
-                                            starts_i32.contiguous().to(q.device),
-                                            scale, causal)
+            tb, tq = _build_qtiles(starts_i32, PREFILL_QTILE)
+            fn = (_get_ext().flash_prefill2 if USE_FLASH_V2
+                  else _get_ext().flash_prefill)
+            return fn(q.contiguous(), k.contiguous(),
+                      v.contiguous(), tb.to(q.device),
+                      tq.to(q.device),
+                      starts_i32.contiguous().to(q.device),
+                      scale, causal)
         if batch_idx is None:
             batch_idx = _batch_idx_from_starts(seq_starts, q.shape[0])
         return _get_ext().prefill_attn(q.contiguous(), k.contiguous(), v.contiguous(),
@@ -129,8 +140,9 @@ def prefill_attention(q, k, v, seq_starts, causal: bool = True,
 
 
 def _build_qtiles(seq_starts: torch.Tensor, qtile: int = 64):
-    """Per-64-row q-tile (segment id, global q start) arrays for the flash
-    prefill grid; tiles never span segment boundaries."""
+    """Per-q-tile (segment id, global q start) arrays for the flash
+    prefill grid; tiles never span segment boundaries. qtile must match
+    the kernel the tiles feed (PREFILL_QTILE / CHUNK_QTILE for v2)."""
     starts = seq_starts.cpu().tolist()
     tb: list[int] = []
     tq: list[int] = []
@@ -148,8 +160,10 @@ def chunked_prefill_attention(q, k_cache, v_cache, block_tables, hist_lens,
     """Packed new-token chunks attending over paged history + themselves."""
     scale = scale or 1.0 / math.sqrt(q.shape[-1])
     if _on_gpu(q):
-        tb, tq = _build_qtiles(seq_starts.to(torch.int32))
-        return _get_ext().flash_prefill_paged(
+        tb, tq = _build_qtiles(seq_starts.to(torch.int32), CHUNK_QTILE)
+        fn = (_get_ext().flash_prefill2_paged if USE_FLASH_V2
+              else _get_ext().flash_prefill_paged)
+        return fn(
             q.contiguous(), k_cache, v_cache,
             block_tables.to(torch.int32).contiguous(),
             tb.to(q.device), tq.to(q.device),
@@ -162,19 +176,23 @@ def chunked_prefill_attention(q, k_cache, v_cache, block_tables, hist_lens,
 def prefill_attention_tiles(q, k, v, tb, tq, seq_starts_dev, scale: float,
                             causal: bool = True):
     """Device-tile variant of flash prefill (D=128): tiles prebuilt once
-    per model call instead of once per layer."""
-    return _get_ext().flash_prefill(q.contiguous(), k.contiguous(), v.contiguous(),
-                                    tb, tq, seq_starts_dev, scale, causal)
+    per model call (with PREFILL_QTILE) instead of once per layer."""
+    fn = _get_ext().flash_prefill2 if USE_FLASH_V2 else _get_ext().flash_prefill
+    return fn(q.contiguous(), k.contiguous(), v.contiguous(),
+              tb, tq, seq_starts_dev, scale, causal)
 
 
 def chunked_prefill_attention_tiles(q, k_cache, v_cache, block_tables, tb, tq,
                                     seq_starts_dev, hist_lens_dev, scale: float):
     """Device-tile variant of chunked_prefill_attention: every argument is
     already a device tensor (tiles included), so the call is hipGraph-
-    capturable — no host-side tile building or H2D copies."""
-    return _get_ext().flash_prefill_paged(q, k_cache, v_cache, block_tables,
-                                          tb, tq, seq_starts_dev, hist_lens_dev,
-                                          scale)
+    capturable — no host-side tile building or H2D copies. Tiles must be
+    built with CHUNK_QTILE."""
+    fn = (_get_ext().flash_prefill2_paged if USE_FLASH_V2
+          else _get_ext().flash_prefill_paged)
+    return fn(q, k_cache, v_cache, block_tables,
+              tb, tq, seq_starts_dev, hist_lens_dev,
+              scale)
 
 
 def paged_decode_attention(q, k_cache, v_cache, block_tables, seq_lens,

@@ -19,6 +19,7 @@ SOURCES = [
     os.path.join(CSRC, "norm_elem.hip"),
     os.path.join(CSRC, "attention.hip"),
     os.path.join(CSRC, "flash_prefill.hip"),
+    os.path.join(CSRC, "flash_prefill2.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
     os.path.join(CSRC, "retrieval_sampling.hip"),
 ]

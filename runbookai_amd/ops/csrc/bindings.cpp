@@ -33,6 +33,12 @@ void launch_flash_prefill(const void*, const void*, const void*, const void*,
 void launch_flash_prefill_paged(const void*, const void*, const void*, const void*,
                                 const void*, const void*, const void*, const void*,
                                 void*, int, int, int, int, int, float, void*);
+void launch_flash_prefill2(const void*, const void*, const void*, const void*,
+                           const void*, const void*, void*, int, int, int, float,
+                           int, void*);
+void launch_flash_prefill2_paged(const void*, const void*, const void*, const void*,
+                                 const void*, const void*, const void*, const void*,
+                                 void*, int, int, int, int, int, float, void*);
 void launch_mfma_probe(const void*, const void*, void*, void*);
 void launch_skinny_gemm(const void*, const void*, void*, void*, void*, int, int,
                         long, int, void*);
@@ -233,6 +239,51 @@ torch::Tensor flash_prefill_paged(torch::Tensor q, torch::Tensor kc, torch::Tens
     return out;
 }
 
+torch::Tensor flash_prefill2(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                             torch::Tensor tile_batch, torch::Tensor tile_qstart,
+                             torch::Tensor seq_starts, double scale, bool causal) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(k, torch::kBFloat16);
+    CHECK_IN(v, torch::kBFloat16);
+    CHECK_IN(tile_batch, torch::kInt32);
+    CHECK_IN(tile_qstart, torch::kInt32);
+    CHECK_IN(seq_starts, torch::kInt32);
+    int Hq = (int)q.size(1), Hk = (int)k.size(1), D = (int)q.size(2);
+    TORCH_CHECK(D == 128, "flash prefill v2 supports head dim 128");
+    auto out = torch::empty_like(q);
+    launch_flash_prefill2(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          tile_batch.data_ptr(), tile_qstart.data_ptr(),
+                          seq_starts.data_ptr(), out.data_ptr(),
+                          (int)tile_batch.size(0), Hq, Hk, (float)scale,
+                          causal ? 1 : 0, current_stream());
+    return out;
+}
+
+torch::Tensor flash_prefill2_paged(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                                   torch::Tensor bt, torch::Tensor tile_batch,
+                                   torch::Tensor tile_qstart, torch::Tensor seq_starts,
+                                   torch::Tensor hist_lens, double scale) {
+    CHECK_IN(q, torch::kBFloat16);
+    CHECK_IN(kc, torch::kBFloat16);
+    CHECK_IN(vc, torch::kBFloat16);
+    CHECK_IN(bt, torch::kInt32);
+    CHECK_IN(tile_batch, torch::kInt32);
+    CHECK_IN(tile_qstart, torch::kInt32);
+    CHECK_IN(seq_starts, torch::kInt32);
+    CHECK_IN(hist_lens, torch::kInt32);
+    int Hq = (int)q.size(1), D = (int)q.size(2);
+    int Hk = (int)kc.size(1), BS = (int)kc.size(2);
+    TORCH_CHECK(D == 128, "paged flash prefill v2 supports head dim 128");
+    auto out = torch::empty_like(q);
+    launch_flash_prefill2_paged(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                                bt.data_ptr(), tile_batch.data_ptr(),
+                                tile_qstart.data_ptr(), seq_starts.data_ptr(),
+                                hist_lens.data_ptr(), out.data_ptr(),
+                                (int)tile_batch.size(0), Hq, Hk, BS,
+                                (int)bt.size(1), (float)scale, current_stream());
+    return out;
+}
+
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
     CHECK_IN(A, torch::kBFloat16);
     CHECK_IN(B, torch::kBFloat16);
@@ -352,6 +403,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("paged_decode", &paged_decode, "paged-KV decode attention");
     m.def("prefill_attn", &prefill_attn, "varlen prefill attention");
     m.def("flash_prefill", &flash_prefill, "MFMA flash prefill attention (D=128)");
+    m.def("flash_prefill2", &flash_prefill2,
+          "MFMA flash prefill v2: in-register softmax (D=128)");
+    m.def("flash_prefill2_paged", &flash_prefill2_paged,
+          "MFMA flash prefill v2, paged history (D=128)");
     m.def("flash_prefill_paged", &flash_prefill_paged,
           "MFMA chunked prefill over the paged KV cache (D=128)");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");

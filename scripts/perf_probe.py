@@ -107,20 +107,36 @@ def main() -> None:
     q = torch.randn(T_, Hq_, D_, dtype=torch.bfloat16, device="cuda") * 0.5
     k = torch.randn(T_, Hk_, D_, dtype=torch.bfloat16, device="cuda") * 0.5
     v = torch.randn(T_, Hk_, D_, dtype=torch.bfloat16, device="cuda") * 0.5
-    starts = torch.arange(0, T_ + 1, S_, dtype=torch.int32).to("cuda")
-    for _ in range(2):
-        rops.prefill_attention(q, k, v, starts, causal=True)
-    torch.cuda.synchronize()
-    t0 = time.time()
-    iters_a = 5
-    for _ in range(iters_a):
-        rops.prefill_attention(q, k, v, starts, causal=True)
-    torch.cuda.synchronize()
-    dt = (time.time() - t0) / iters_a
+    starts_h = torch.arange(0, T_ + 1, S_, dtype=torch.int32)
+    starts = starts_h.to("cuda")
     # causal flops: 2 matmuls * 2*S^2/2*D per (head, seq)
     flops = 2 * 2 * (S_ * S_ / 2) * D_ * Hq_ * B_
-    attn["flash_prefill_B16_H32_S2048_ms"] = round(dt * 1000, 2)
-    attn["flash_prefill_TF"] = round(flops / dt / 1e12, 1)
+    scale = 1.0 / (D_ ** 0.5)
+    from runbookai_amd.ops import _get_ext as _ge
+    ext_a = _ge()
+    variants = {
+        "v1_16x16": (ext_a.flash_prefill, 64),
+        "v2_inreg": (ext_a.flash_prefill2, 256),
+    }
+    # within-probe interleaved A/B (guide rule 24): 6 rounds each, report min
+    times = {name: [] for name in variants}
+    tiles = {name: tuple(t.to("cuda") for t in rops._build_qtiles(starts_h, qt))
+             for name, (_, qt) in variants.items()}
+    for name, (fn, _) in variants.items():
+        fn(q, k, v, *tiles[name], starts, scale, True)   # warmup
+    torch.cuda.synchronize()
+    for _ in range(6):
+        for name, (fn, _) in variants.items():
+            torch.cuda.synchronize()
+            t0 = time.time()
+            for _ in range(3):
+                fn(q, k, v, *tiles[name], starts, scale, True)
+            torch.cuda.synchronize()
+            times[name].append((time.time() - t0) / 3)
+    for name, ts in times.items():
+        best = min(ts)
+        attn[f"flash_{name}_ms"] = round(best * 1000, 2)
+        attn[f"flash_{name}_TF"] = round(flops / best / 1e12, 1)
     del q, k, v
     report["attention"] = attn
 

@@ -189,6 +189,66 @@ class TestFlashPrefill:
         assert (out.cpu().float() - expected.float()).abs().max().item() < 4e-2
 
 
+class TestFlashPrefillV2:
+    """v2 kernel (in-register softmax, 32x32 MFMA, swapped QK^T) — explicit
+    v1/v2 parity plus shapes that exercise the 8-wave tile structure."""
+
+    def _tiles(self, starts, qtile):
+        return [t.to(DEV) for t in ops._build_qtiles(starts, qtile)]
+
+    @pytest.mark.parametrize("lens", [[300, 520, 64], [256], [1024]])
+    def test_v1_v2_parity_and_reference(self, lens):
+        from runbookai_amd.ops import _get_ext
+        ext = _get_ext()
+        Hq, Hk, D = 8, 2, 128
+        T = sum(lens)
+        starts = torch.tensor([0, *torch.tensor(lens).cumsum(0).tolist()],
+                              dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.5).to(DEV)
+        k = bf(torch.randn(T, Hk, D) * 0.5).to(DEV)
+        v = bf(torch.randn(T, Hk, D) * 0.5).to(DEV)
+        scale = 1.0 / math.sqrt(D)
+        sd = starts.to(DEV)
+        tb1, tq1 = self._tiles(starts, 64)
+        tb2, tq2 = self._tiles(starts, 256)
+        out1 = ext.flash_prefill(q, k, v, tb1, tq1, sd, scale, True)
+        out2 = ext.flash_prefill2(q, k, v, tb2, tq2, sd, scale, True)
+        expected = ref.prefill_attention(q.cpu(), k.cpu(), v.cpu(), starts,
+                                         causal=True)
+        d1 = (out1.cpu().float() - expected.float()).abs().max().item()
+        d2 = (out2.cpu().float() - expected.float()).abs().max().item()
+        assert d1 < 4e-2, d1
+        assert d2 < 4e-2, d2
+
+    def test_forced_rescale_past_defer_threshold(self):
+        """Spike large enough that (tmax - m)*scale*log2e >> THRESH=11.5:
+        the non-deferred rescale branch must produce reference numerics
+        (guide rule 26: the rare branch needs its own forcing input)."""
+        Hq, Hk, D = 4, 4, 128
+        T = 200
+        starts = torch.tensor([0, T], dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.1)
+        k = bf(torch.randn(T, Hk, D) * 0.1)
+        v = bf(torch.randn(T, Hk, D) * 0.5)
+        k[150] = (q[190] * 400.0).to(torch.bfloat16)  # raw score ~512
+        expected = ref.prefill_attention(q, k, v, starts, causal=True)
+        out = ops.prefill_attention(q.to(DEV), k.to(DEV), v.to(DEV),
+                                    starts.to(DEV), causal=True)
+        assert (out.cpu().float() - expected.float()).abs().max().item() < 4e-2
+
+    def test_gqa_llama_shape_noncausal(self):
+        Hq, Hk, D = 32, 8, 128
+        T = 640
+        starts = torch.tensor([0, T], dtype=torch.int32)
+        q = bf(torch.randn(T, Hq, D) * 0.5).to(DEV)
+        k = bf(torch.randn(T, Hk, D) * 0.5).to(DEV)
+        v = bf(torch.randn(T, Hk, D) * 0.5).to(DEV)
+        expected = ref.prefill_attention(q.cpu(), k.cpu(), v.cpu(), starts,
+                                         causal=False)
+        out = ops.prefill_attention(q, k, v, starts.to(DEV), causal=False)
+        assert (out.cpu().float() - expected.float()).abs().max().item() < 4e-2
+
+
 class TestChunkedPrefill:
     def test_paged_chunk_vs_reference(self):
         """New-token chunks attend over paged history (forced-byte injection
@@ -218,6 +278,40 @@ class TestChunkedPrefill:
         expected = ref.chunked_prefill_attention(q, kc, vc, bt, hist_t, starts)
         out = ops.chunked_prefill_attention(q.to(DEV), kc.to(DEV), vc.to(DEV),
                                             bt.to(DEV), hist_t.to(DEV), starts.to(DEV))
+        diff = (out.cpu().float() - expected.float()).abs().max().item()
+        assert diff < 4e-2, diff
+
+
+class TestChunkedPrefillLarge:
+    def test_large_chunks_paged_v2(self):
+        """Chunks spanning multiple 128-row v2 tiles over long paged
+        history (the GPU bench's dominant attention shape)."""
+        Hq, Hk, D, BS, NB = 8, 2, 128, 16, 96
+        hist = [100, 37]
+        new = [200, 150]
+        B = 2
+        bt = torch.full((B, 32), -1, dtype=torch.int32)
+        used = iter(torch.randperm(NB - 1).tolist())
+        kc = torch.zeros(NB, Hk, BS, D, dtype=torch.bfloat16)
+        vc = torch.zeros_like(kc)
+        torch.manual_seed(11)
+        for b in range(B):
+            total = hist[b] + new[b]
+            nb = (total + BS - 1) // BS
+            bt[b, :nb] = torch.tensor([next(used) for _ in range(nb)],
+                                      dtype=torch.int32)
+            for t in range(total):
+                blk = int(bt[b, t // BS])
+                kc[blk, :, t % BS] = torch.randn(Hk, D).to(torch.bfloat16) * 0.5
+                vc[blk, :, t % BS] = torch.randn(Hk, D).to(torch.bfloat16) * 0.5
+        Tnew = sum(new)
+        starts = torch.tensor([0, new[0], Tnew], dtype=torch.int32)
+        q = bf(torch.randn(Tnew, Hq, D) * 0.5)
+        hist_t = torch.tensor(hist, dtype=torch.int32)
+        expected = ref.chunked_prefill_attention(q, kc, vc, bt, hist_t, starts)
+        out = ops.chunked_prefill_attention(q.to(DEV), kc.to(DEV), vc.to(DEV),
+                                            bt.to(DEV), hist_t.to(DEV),
+                                            starts.to(DEV))
         diff = (out.cpu().float() - expected.float()).abs().max().item()
         assert diff < 4e-2, diff
 
