@@ -49,8 +49,24 @@ DEVINL unsigned kswz2(unsigned row, unsigned colb) {
     return row * (DHEAD2 * 2) + (colb ^ ((row & 7u) << 4));
 }
 
+// Transposed-V k-slot swizzle, 8-element (16-B) block granular: the write
+// pass scatters one k across dcols at a 144-B row stride (288 dwords = 0
+// mod 32 banks -> 16-way write conflict without it); XORing the 8-k block
+// index with dcol bits 3-5 spreads a 16-lane write group over 8 banks
+// (2-way) while b128 reads of 8 consecutive k stay one contiguous 16-B
+// slot (k0 is always 8-aligned).
+DEVINL unsigned vswz2(unsigned dcol, unsigned kk) {
+    return kk ^ (8u * ((dcol >> 3) & 7u));
+}
+
+// ONE v_cvt_pk_bf16_f32 instead of ~18 VALU of hand-rolled RNE bit math
+// (guide T12 recipe: no builtin for cvt_pk on gfx950; the asm statement is
+// register-only, and its consumer permlane32_swap is a BUILTIN, so hipcc
+// inserts the VALU-write->permlane hazard nops itself — T21 note)
 DEVINL unsigned pack_bf16(float lo, float hi) {
-    return (unsigned)f2bf(lo) | ((unsigned)f2bf(hi) << 16);
+    unsigned r;
+    asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+    return r;
 }
 
 // C-fragment row for reg r, lane-half hi
@@ -85,8 +101,11 @@ __global__ void flash_prefill2_kernel(
     const int hi = lane >> 5;      // lane half
     const int* bt = PAGED ? block_tables + (long)b * max_blocks : nullptr;
 
-    __shared__ ushort_t k_lds[KTILE2 * DHEAD2];      // swizzled rows
-    __shared__ ushort_t v_lds[DHEAD2][VPAD2];        // transposed, padded
+    // LDS double-buffered: the staging writes for tile kt+1 overlap OTHER
+    // waves' compute on tile kt (different buffer), so the loop needs only
+    // ONE barrier per KV tile instead of two barrier-fenced phases
+    __shared__ ushort_t k_lds[2][KTILE2 * DHEAD2];   // swizzled rows
+    __shared__ ushort_t v_lds[2][DHEAD2][VPAD2];     // transposed, padded
 
     const int q0w = q0g + wid * 32;          // this wave's q-block start
     const int my_qrow = q0w + col;           // this lane's q-row (global)
@@ -122,12 +141,17 @@ __global__ void flash_prefill2_kernel(
     const int wave_q_hi = min(q0w + 32, seg_end) - 1 - seg_start;
     const int wave_kv_hi = CAUSAL ? (hist + wave_q_hi + 1) : kv_len;
     const int q_full = hist + (my_qrow - seg_start);   // causal bound, local kv
+    // every ktok < wave_kv_lo is attendable by EVERY row of this wave
+    const int wave_kv_lo = CAUSAL ? (hist + (q0w - seg_start) + 1) : kv_len;
+    const bool row_valid_wave = (q0w + 31 < seg_end);
 
-    // register-staged double buffer (T14)
+    // register staging, single set: tile kt+1 sits in registers through
+    // compute(kt), drains into LDS buf[cur^1] after it (T14 write-late),
+    // then tile kt+2's loads re-issue into the same registers
     constexpr int NCHUNK = KTILE2 * DHEAD2 / 8 / NTHREADS;
-    bf16x8_t skA[NCHUNK], svA[NCHUNK], skB[NCHUNK], svB[NCHUNK];
+    bf16x8_t sk[NCHUNK], sv[NCHUNK];
 
-    auto load_tile = [&](int kt, bf16x8_t (&sk)[NCHUNK], bf16x8_t (&sv)[NCHUNK]) {
+    auto load_tile = [&](int kt) {
 #pragma unroll
         for (int c = 0; c < NCHUNK; ++c) {
             const int idx = (int)threadIdx.x + c * NTHREADS;
@@ -149,35 +173,34 @@ __global__ void flash_prefill2_kernel(
             }
         }
     };
-    auto write_tile = [&](bf16x8_t (&sk)[NCHUNK], bf16x8_t (&sv)[NCHUNK]) {
+    auto write_tile = [&](int buf) {
 #pragma unroll
         for (int c = 0; c < NCHUNK; ++c) {
             const int idx = (int)threadIdx.x + c * NTHREADS;
             const int row = (idx * 8) / DHEAD2;
             const int dcol = (idx * 8) % DHEAD2;
             *reinterpret_cast<bf16x8_t*>(
-                reinterpret_cast<char*>(k_lds) + kswz2((unsigned)row, (unsigned)dcol * 2)) = sk[c];
+                reinterpret_cast<char*>(k_lds[buf])
+                + kswz2((unsigned)row, (unsigned)dcol * 2)) = sk[c];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) v_lds[dcol + j][row] = sv[c][j];
+            for (int j = 0; j < 8; ++j)
+                v_lds[buf][dcol + j][vswz2(dcol + j, row)] = sv[c][j];
         }
     };
 
-    load_tile(0, skA, svA);
-    int parity = 0;
+    // prologue: tile 0 into buf 0, tile 1's loads in flight
+    load_tile(0);
+    write_tile(0);
+    if (1 < n_tiles) load_tile(1);
+    __syncthreads();
+    int cur = 0;
     for (int kt = 0; kt < n_tiles; ++kt) {
         const int kv0 = kt * KTILE2;
-        __syncthreads();   // previous tile's LDS readers done
-        if (parity == 0) write_tile(skA, svA); else write_tile(skB, svB);
-        if (kt + 1 < n_tiles) {
-            if (parity == 0) load_tile(kt + 1, skB, svB);
-            else load_tile(kt + 1, skA, svA);
-        }
-        parity ^= 1;
-        __syncthreads();
-        if (kv0 >= wave_kv_hi) continue;   // fully above this wave's causal bound
+        if (kv0 < wave_kv_hi) {
 
         // ---- QK^T swapped: S^T[ktok][qcol] per 32-token sub-tile ----
         f32x16_t sc[2];
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
@@ -186,32 +209,45 @@ __global__ void flash_prefill2_kernel(
             for (int s = 0; s < 8; ++s) {
                 const unsigned krow = sub * 32 + col;   // k-token row
                 bf16x8_t kfrag = *reinterpret_cast<const bf16x8_t*>(
-                    reinterpret_cast<const char*>(k_lds)
+                    reinterpret_cast<const char*>(k_lds[cur])
                     + kswz2(krow, (s * 16 + hi * 8) * 2));
                 sc[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qf[s],
                                                                   sc[sub], 0, 0, 0);
             }
         }
+        __builtin_amdgcn_s_setprio(0);
 
         // ---- mask + in-register online softmax (row = this lane's q-row) ----
+        // interior tiles (every ktok attendable by every row of the wave)
+        // skip the 32-select mask pass entirely — wave-uniform condition
+        const bool interior = row_valid_wave &&
+                              (kv0 + KTILE2 <= wave_kv_lo) &&
+                              (kv0 + KTILE2 <= kv_len);
         float tmax = -1e30f;
+        if (interior) {
 #pragma unroll
-        for (int sub = 0; sub < 2; ++sub) {
+            for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
-            for (int r = 0; r < 16; ++r) {
-                const int ktok = kv0 + sub * 32 + crow(r, hi);
-                const bool masked = !row_valid || (ktok >= kv_len) ||
-                                    (CAUSAL && ktok > q_full);
-                const float sv0 = masked ? -1e30f : sc[sub][r];
-                sc[sub][r] = sv0;
-                tmax = fmaxf(tmax, sv0);
+                for (int r = 0; r < 16; ++r) tmax = fmaxf(tmax, sc[sub][r]);
+        } else {
+#pragma unroll
+            for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+                for (int r = 0; r < 16; ++r) {
+                    const int ktok = kv0 + sub * 32 + crow(r, hi);
+                    const bool masked = !row_valid || (ktok >= kv_len) ||
+                                        (CAUSAL && ktok > q_full);
+                    const float sv0 = masked ? -1e30f : sc[sub][r];
+                    sc[sub][r] = sv0;
+                    tmax = fmaxf(tmax, sv0);
+                }
             }
         }
         tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));   // partner half
 
         const bool defer = __all((tmax - m_run) * c2 <= THRESH);
         const float m_new = defer ? m_run : fmaxf(m_run, tmax);
-        const float alpha = defer ? 1.0f : exp2f((m_run - m_new) * c2);
+        const float alpha = defer ? 1.0f : __builtin_amdgcn_exp2f((m_run - m_new) * c2);
         m_run = m_new;
 
         float row_sum = 0.f;
@@ -221,7 +257,7 @@ __global__ void flash_prefill2_kernel(
             for (int r = 0; r < 16; ++r) {
                 const float p = (sc[sub][r] <= -1e29f)
                                     ? 0.f
-                                    : exp2f((sc[sub][r] - m_new) * c2);
+                                    : __builtin_amdgcn_exp2f((sc[sub][r] - m_new) * c2);
                 sc[sub][r] = p;
                 row_sum += p;
             }
@@ -261,15 +297,28 @@ __global__ void flash_prefill2_kernel(
         }
 
         // ---- PV: O[32q x 128d] += P[32q x 64k] · V[64k x 128d] ----
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
 #pragma unroll
             for (int db = 0; db < 4; ++db) {
+                const unsigned dcol = db * 32 + col;
                 bf16x8_t vfrag = *reinterpret_cast<const bf16x8_t*>(
-                    &v_lds[db * 32 + col][ks * 16 + hi * 8]);
+                    &v_lds[cur][dcol][vswz2(dcol, ks * 16 + hi * 8)]);
                 o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], vfrag,
                                                                     o_acc[db], 0, 0, 0);
             }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        }   // compute (skipped above this wave's causal bound)
+        if (kt + 1 < n_tiles) {
+            // drain tile kt+1's registers into the other buffer (safe: every
+            // wave finished reading it at the PREVIOUS barrier) and re-issue
+            // tile kt+2's loads — they hide under compute(kt+1)
+            write_tile(cur ^ 1);
+            if (kt + 2 < n_tiles) load_tile(kt + 2);
+            __syncthreads();
+            cur ^= 1;
         }
     }
 
