@@ -217,99 +217,95 @@ __global__ void flash_prefill2_kernel(
         }
         __builtin_amdgcn_s_setprio(0);
 
-        // ---- mask + in-register online softmax (row = this lane's q-row) ----
+        // ---- per-SUB online softmax + PV (finer pipelining: sub1's
+        // softmax VALU overlaps the partner wave's PV MFMAs on sub0) ----
         // interior tiles (every ktok attendable by every row of the wave)
         // skip the 32-select mask pass entirely — wave-uniform condition
         const bool interior = row_valid_wave &&
                               (kv0 + KTILE2 <= wave_kv_lo) &&
                               (kv0 + KTILE2 <= kv_len);
-        float tmax = -1e30f;
-        if (interior) {
+
+        auto sm_sub = [&](f32x16_t& s, int sub, bf16x8_t (&pas)[2]) {
+            float tmax = -1e30f;
+            if (interior) {
 #pragma unroll
-            for (int sub = 0; sub < 2; ++sub)
-#pragma unroll
-                for (int r = 0; r < 16; ++r) tmax = fmaxf(tmax, sc[sub][r]);
-        } else {
-#pragma unroll
-            for (int sub = 0; sub < 2; ++sub) {
+                for (int r = 0; r < 16; ++r) tmax = fmaxf(tmax, s[r]);
+            } else {
 #pragma unroll
                 for (int r = 0; r < 16; ++r) {
                     const int ktok = kv0 + sub * 32 + crow(r, hi);
                     const bool masked = !row_valid || (ktok >= kv_len) ||
                                         (CAUSAL && ktok > q_full);
-                    const float sv0 = masked ? -1e30f : sc[sub][r];
-                    sc[sub][r] = sv0;
+                    const float sv0 = masked ? -1e30f : s[r];
+                    s[r] = sv0;
                     tmax = fmaxf(tmax, sv0);
                 }
             }
-        }
-        tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));   // partner half
-
-        const bool defer = __all((tmax - m_run) * c2 <= THRESH);
-        const float m_new = defer ? m_run : fmaxf(m_run, tmax);
-        const float alpha = defer ? 1.0f : __builtin_amdgcn_exp2f((m_run - m_new) * c2);
-        m_run = m_new;
-
-        float row_sum = 0.f;
-#pragma unroll
-        for (int sub = 0; sub < 2; ++sub) {
+            tmax = fmaxf(tmax, __shfl_xor(tmax, 32, WAVE));   // partner half
+            const bool defer = __all((tmax - m_run) * c2 <= THRESH);
+            const float m_new = defer ? m_run : fmaxf(m_run, tmax);
+            const float alpha = defer ? 1.0f
+                                      : __builtin_amdgcn_exp2f((m_run - m_new) * c2);
+            m_run = m_new;
+            float row_sum = 0.f;
 #pragma unroll
             for (int r = 0; r < 16; ++r) {
-                const float p = (sc[sub][r] <= -1e29f)
-                                    ? 0.f
-                                    : __builtin_amdgcn_exp2f((sc[sub][r] - m_new) * c2);
-                sc[sub][r] = p;
-                row_sum += p;
+                const float pv = (s[r] <= -1e29f)
+                                     ? 0.f
+                                     : __builtin_amdgcn_exp2f((s[r] - m_new) * c2);
+                s[r] = pv;
+                row_sum += pv;
             }
-        }
-        row_sum += __shfl_xor(row_sum, 32, WAVE);
-        l_run = l_run * alpha + row_sum;
-
-        if (!defer) {
-            // per-reg alpha: O regs live in PV fragment layout, so reg r
-            // belongs to q-row crow(r, hi) — broadcast from the owning lane
+            row_sum += __shfl_xor(row_sum, 32, WAVE);
+            l_run = l_run * alpha + row_sum;
+            if (!defer) {
+                // per-reg alpha: O regs live in PV fragment layout, so reg
+                // r belongs to q-row crow(r, hi) — broadcast from its lane
 #pragma unroll
-            for (int r = 0; r < 16; ++r) {
-                const float a_r = __shfl(alpha, crow(r, hi), WAVE);
+                for (int r = 0; r < 16; ++r) {
+                    const float a_r = __shfl(alpha, crow(r, hi), WAVE);
 #pragma unroll
-                for (int d = 0; d < 4; ++d) o_acc[d][r] *= a_r;
+                    for (int d = 0; d < 4; ++d) o_acc[d][r] *= a_r;
+                }
             }
-        }
-
-        // ---- P -> bf16 A-fragments via pack + permlane32_swap ----
-        // slice ks covers k = 16*ks..16*ks+15; regs 0..7 of sc[sub] hold
-        // rows 0..15 (+4*hi), regs 8..15 hold rows 16..31 (+4*hi)
-        bf16x8_t pa[4];
+            // P -> bf16 A-fragments via cvt_pk + permlane32_swap: regs 0..7
+            // hold rows 0..15 (+4*hi) = k-slice 0, regs 8..15 rows 16..31
 #pragma unroll
-        for (int ks = 0; ks < 4; ++ks) {
-            const int sub = ks >> 1;
-            const int rb = (ks & 1) * 8;   // reg base within the accumulator
-            unsigned a_lo = pack_bf16(sc[sub][rb + 0], sc[sub][rb + 1]);
-            unsigned a_hi = pack_bf16(sc[sub][rb + 2], sc[sub][rb + 3]);
-            unsigned b_lo = pack_bf16(sc[sub][rb + 4], sc[sub][rb + 5]);
-            unsigned b_hi = pack_bf16(sc[sub][rb + 6], sc[sub][rb + 7]);
-            auto r1 = __builtin_amdgcn_permlane32_swap(a_lo, b_lo, false, false);
-            auto r2 = __builtin_amdgcn_permlane32_swap(a_hi, b_hi, false, false);
-            // dwords [j01, j23, j45, j67] of the 8-element A fragment
-            unsigned fr[4] = {(unsigned)r1[0], (unsigned)r2[0],
-                              (unsigned)r1[1], (unsigned)r2[1]};
-            pa[ks] = *reinterpret_cast<bf16x8_t*>(fr);
-        }
-
-        // ---- PV: O[32q x 128d] += P[32q x 64k] · V[64k x 128d] ----
-        __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-        for (int ks = 0; ks < 4; ++ks) {
-#pragma unroll
-            for (int db = 0; db < 4; ++db) {
-                const unsigned dcol = db * 32 + col;
-                bf16x8_t vfrag = *reinterpret_cast<const bf16x8_t*>(
-                    &v_lds[cur][dcol][vswz2(dcol, ks * 16 + hi * 8)]);
-                o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ks], vfrag,
-                                                                    o_acc[db], 0, 0, 0);
+            for (int half = 0; half < 2; ++half) {
+                const int rb = half * 8;
+                unsigned a_lo = pack_bf16(s[rb + 0], s[rb + 1]);
+                unsigned a_hi = pack_bf16(s[rb + 2], s[rb + 3]);
+                unsigned b_lo = pack_bf16(s[rb + 4], s[rb + 5]);
+                unsigned b_hi = pack_bf16(s[rb + 6], s[rb + 7]);
+                auto r1 = __builtin_amdgcn_permlane32_swap(a_lo, b_lo, false, false);
+                auto r2 = __builtin_amdgcn_permlane32_swap(a_hi, b_hi, false, false);
+                unsigned fr[4] = {(unsigned)r1[0], (unsigned)r2[0],
+                                  (unsigned)r1[1], (unsigned)r2[1]};
+                pas[half] = *reinterpret_cast<bf16x8_t*>(fr);
             }
-        }
-        __builtin_amdgcn_s_setprio(0);
+        };
+        auto pv_sub = [&](int sub, bf16x8_t (&pas)[2]) {
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int half = 0; half < 2; ++half) {
+                const int ks = sub * 2 + half;
+#pragma unroll
+                for (int db = 0; db < 4; ++db) {
+                    const unsigned dcol = db * 32 + col;
+                    bf16x8_t vfrag = *reinterpret_cast<const bf16x8_t*>(
+                        &v_lds[cur][dcol][vswz2(dcol, ks * 16 + hi * 8)]);
+                    o_acc[db] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                        pas[half], vfrag, o_acc[db], 0, 0, 0);
+                }
+            }
+            __builtin_amdgcn_s_setprio(0);
+        };
+
+        bf16x8_t pa0[2], pa1[2];
+        sm_sub(sc[0], 0, pa0);
+        pv_sub(0, pa0);
+        sm_sub(sc[1], 1, pa1);
+        pv_sub(1, pa1);
         }   // compute (skipped above this wave's causal bound)
         if (kt + 1 < n_tiles) {
             // drain tile kt+1's registers into the other buffer (safe: every
