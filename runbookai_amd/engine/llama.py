@@ -189,6 +189,7 @@ class LlamaModel:
 
         self.use_graphs = (str(device).startswith("cuda")
                            and _os.environ.get("RUNBOOKAI_NO_GRAPHS", "0") != "1")
+        self.use_decode_fused = _os.environ.get("RUNBOOKAI_DECODE_FUSED", "1") != "0"
         self.use_chunk_graphs = _os.environ.get("RUNBOOKAI_CHUNK_GRAPHS", "1") != "0"
         self._graphs: dict[int, tuple] = {}
         self._chunk_graphs: dict[tuple, tuple] = {}
@@ -268,12 +269,44 @@ class LlamaModel:
     def _decode_impl(self, token_ids, positions, block_tables, seq_lens, slots):
         """Device-tensor decode body (hipGraph-capturable)."""
         h = self.embed.weight[token_ids]
+        if (self.use_decode_fused and h.shape[0] <= 4 and self.tp == 1
+                and str(self.device).startswith("cuda")):
+            return self._decode_impl_fused(h, positions, block_tables, seq_lens,
+                                           slots)
 
         def attn_fn(q, k, v, i):
             return ops.paged_decode_attention(q, self.kv.k[i], self.kv.v[i],
                                               block_tables, seq_lens, self.scale)
 
         normed = self._transformer_body(h, positions, slots, attn_fn, fused_kv=True)
+        return self.lm_head(normed)
+
+    def _decode_impl_fused(self, h, positions, block_tables, seq_lens, slots):
+        """Small-batch (B <= 4) decode layer on the fused GEMV path: per
+        layer 4 decode_gemv launches (rmsnorm folded into the qkv/gate_up
+        prologues, silu into down's, residual adds into the o/down
+        epilogues) + rope-scatter + paged attention — vs ~10 kernels on
+        the generic body. tp == 1 only: the residual-add-in-epilogue would
+        land before the TP all-reduce."""
+        T = h.shape[0]
+        cfg = self.cfg
+        eps = cfg.rms_eps
+        for i, layer in enumerate(self.layers):
+            qkv = ops.gemv(h, layer.qkv.weight, pre=1,
+                           norm_w=layer.input_norm_w, eps=eps)
+            q, k, v = layer._split_qkv(qkv, T)
+            q = ops.rope_store_kv_fused(q, k, v, self.rope_cos, self.rope_sin,
+                                        positions, self.kv.k[i], self.kv.v[i],
+                                        slots)
+            attn = ops.paged_decode_attention(q, self.kv.k[i], self.kv.v[i],
+                                              block_tables, seq_lens, self.scale)
+            h = ops.gemv(attn.reshape(T, -1), layer.o_proj.weight, res=h)
+            gu = ops.gemv(h, layer.gate_up.weight, pre=1,
+                          norm_w=layer.post_norm_w, eps=eps)
+            h = ops.gemv(gu, layer.down.weight, pre=2, res=h)
+        # lm_head stays on hipBLASLt (it wins at vocab-sized N); final norm
+        # is one small kernel next to a 1 GB weight stream
+        normed = ops.rmsnorm(h, self.final_norm_w, eps)
         return self.lm_head(normed)
 
     def chunk_step(self, token_ids: torch.Tensor, positions: torch.Tensor,

@@ -215,18 +215,43 @@ def store_kv(k, v, k_cache, v_cache, slot_mapping):
 
 
 USE_SKINNY_GEMM = True
+USE_DECODE_GEMV = _os.environ.get("RUNBOOKAI_DECODE_GEMV", "1") != "0"
+
+
+def gemv(x: torch.Tensor, w: torch.Tensor, pre: int = 0,
+         norm_w: Optional[torch.Tensor] = None,
+         res: Optional[torch.Tensor] = None, eps: float = 1e-5) -> torch.Tensor:
+    """Decode GEMV (M <= 4) with fused prologue/epilogue:
+    pre=0: out = x @ w^T; pre=1: out = rmsnorm(x, norm_w) @ w^T;
+    pre=2: x is packed [gate|up] rows, out = (silu(gate)*up) @ w^T.
+    res (optional): out += res (the residual-stream add). One kernel
+    replaces rmsnorm/silu + GEMM + residual-add launches on the decode
+    path (ops/csrc/decode_gemv.hip)."""
+    if _on_gpu(x):
+        return _get_ext().decode_gemv(x.contiguous(), w, norm_w, res, pre, eps)
+    if pre == 1:
+        x = reference.rmsnorm(x, norm_w, eps)
+    elif pre == 2:
+        g, u = x.chunk(2, dim=-1)
+        x = reference.silu_mul(g.contiguous(), u.contiguous())
+    out = x.float() @ w.t().float()
+    if res is not None:
+        out = out + res.float()
+    return out.to(w.dtype)
 
 
 def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """x [T, K] @ weight[N, K]^T.
 
-    Measured dispatch (profiles/PERF_LOG.md): the hand MFMA skinny kernel
-    beats hipBLASLt on the small-M qkv/o decode projections (1.8-2.6 ->
-    2.8-3.2 TB/s); hipBLASLt keeps gate_up (N=28672), down (K=14336) and
-    lm_head — a one-call probe that looked like parity on gate_up did not
-    reproduce across boxes (82 vs 36 us; see the dispatch-correction entry
-    in profiles/PERF_LOG.md), so the bound is both N<=8192 AND K<=8192.
-    """
+    Measured dispatch (profiles/PERF_LOG.md): at M <= 4 the register-
+    streaming GEMV owns every decode projection (the LDS round trip is
+    pure overhead at that M — guide decode-GEMV rule); at M 5-16 the hand
+    MFMA skinny kernel keeps qkv/o (N,K <= 8192) and hipBLASLt keeps
+    gate_up (N=28672), down (K=14336) and lm_head."""
+    if (USE_DECODE_GEMV and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 4
+            and weight.shape[0] <= 32768 and weight.shape[1] % 8 == 0):
+        # N cap: hipBLASLt wins the vocab-sized lm_head (182 vs 223 us at M=1)
+        return _get_ext().decode_gemv(x.contiguous(), weight, None, None, 0, 1e-5)
     if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 16
             and weight.shape[0] <= 8192 and weight.shape[1] <= 8192
             and weight.shape[0] % 64 == 0 and weight.shape[1] % 64 == 0):

@@ -21,6 +21,7 @@ SOURCES = [
     os.path.join(CSRC, "flash_prefill.hip"),
     os.path.join(CSRC, "flash_prefill2.hip"),
     os.path.join(CSRC, "skinny_gemm.hip"),
+    os.path.join(CSRC, "decode_gemv.hip"),
     os.path.join(CSRC, "retrieval_sampling.hip"),
 ]
 

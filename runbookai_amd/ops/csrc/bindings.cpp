@@ -42,6 +42,8 @@ void launch_flash_prefill2_paged(const void*, const void*, const void*, const vo
 void launch_mfma_probe(const void*, const void*, void*, void*);
 void launch_skinny_gemm(const void*, const void*, void*, void*, void*, int, int,
                         long, int, void*);
+void launch_decode_gemv(const void*, const void*, const void*, const void*,
+                        void*, int, int, long, int, int, float, void*);
 void launch_cosine_scores(const void*, const void*, void*, long, int, void*);
 void launch_masked_argmax(const void*, const void*, void*, int, int, void*);
 void launch_masked_topp(const void*, const void*, const void*, void*, int, int,
@@ -348,6 +350,37 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
     return out;
 }
 
+torch::Tensor decode_gemv(torch::Tensor x, torch::Tensor w,
+                          c10::optional<torch::Tensor> norm_w,
+                          c10::optional<torch::Tensor> res,
+                          int64_t pre, double eps) {
+    // x: [M, K] (pre 0/1) or [M, 2K] packed [gate|up] (pre 2); w: [N, K].
+    CHECK_IN(x, torch::kBFloat16);
+    CHECK_IN(w, torch::kBFloat16);
+    long K = w.size(1);
+    int N = (int)w.size(0);
+    int M = (int)x.size(0);
+    TORCH_CHECK(M <= 4, "decode_gemv supports M <= 4");
+    TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+    TORCH_CHECK(x.size(1) == (pre == 2 ? 2 * K : K), "x/w shape mismatch");
+    const void* nw_ptr = nullptr;
+    if (pre == 1) {
+        TORCH_CHECK(norm_w.has_value() && norm_w->numel() == K);
+        nw_ptr = norm_w->data_ptr();
+    }
+    const void* res_ptr = nullptr;
+    if (res.has_value()) {
+        TORCH_CHECK(res->size(0) == M && res->size(1) == N);
+        res_ptr = res->data_ptr();
+    }
+    auto out = torch::empty({M, (long)N}, x.options());
+    launch_decode_gemv(x.data_ptr(), w.data_ptr(), nw_ptr, res_ptr,
+                       out.data_ptr(), M, N, K, (int)pre,
+                       res_ptr != nullptr ? 1 : 0, (float)eps,
+                       current_stream());
+    return out;
+}
+
 torch::Tensor cosine_scores(torch::Tensor matrix, torch::Tensor query) {
     CHECK_IN(matrix, torch::kFloat16);
     CHECK_IN(query, torch::kFloat16);
@@ -403,6 +436,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("paged_decode", &paged_decode, "paged-KV decode attention");
     m.def("prefill_attn", &prefill_attn, "varlen prefill attention");
     m.def("flash_prefill", &flash_prefill, "MFMA flash prefill attention (D=128)");
+    m.def("decode_gemv", &decode_gemv,
+          "decode GEMV M<=4 with fused rmsnorm/silu prologue + residual epilogue",
+          py::arg("x"), py::arg("w"), py::arg("norm_w") = py::none(),
+          py::arg("res") = py::none(), py::arg("pre") = 0,
+          py::arg("eps") = 1e-5);
     m.def("flash_prefill2", &flash_prefill2,
           "MFMA flash prefill v2: in-register softmax (D=128)");
     m.def("flash_prefill2_paged", &flash_prefill2_paged,

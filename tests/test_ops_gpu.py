@@ -535,3 +535,91 @@ class TestEngineGpu:
         top_gpu = b.topk(20).indices.tolist()
         overlap = len(set(top_cpu) & set(top_gpu))
         assert overlap >= 15, f"top-20 overlap only {overlap}"
+
+
+class TestDecodeGemv:
+    """Register-streaming decode GEMV (M<=4) with fused prologues/epilogue
+    vs fp32 reference compositions."""
+
+    @pytest.mark.parametrize("M,N,K", [(1, 512, 256), (2, 6144, 4096),
+                                       (4, 4096, 4096), (1, 4096, 14336),
+                                       (3, 1000, 264)])
+    def test_plain_vs_reference(self, M, N, K):
+        from runbookai_amd.ops import _get_ext
+        ext = _get_ext()
+        x = bf(torch.randn(M, K) * 0.5).to(DEV)
+        w = bf(torch.randn(N, K) * 0.1).to(DEV)
+        out = ext.decode_gemv(x, w, None, None, 0, 1e-5)
+        expected = x.cpu().float() @ w.cpu().float().t()
+        tol = 2e-2 * (K ** 0.5) * 0.05 + 0.2
+        assert (out.cpu().float() - expected).abs().max().item() < tol
+
+    def test_rmsnorm_prologue(self):
+        from runbookai_amd.ops import _get_ext
+        ext = _get_ext()
+        M, N, K = 2, 1024, 4096
+        x = bf(torch.randn(M, K)).to(DEV)
+        w = bf(torch.randn(N, K) * 0.1).to(DEV)
+        nw = bf(torch.randn(K) * 0.1 + 1.0).to(DEV)
+        out = ext.decode_gemv(x, w, nw, None, 1, 1e-5)
+        normed = ref.rmsnorm(x.cpu(), nw.cpu(), 1e-5)
+        expected = normed.float() @ w.cpu().float().t()
+        assert (out.cpu().float() - expected).abs().max().item() < 0.5
+
+    def test_silu_prologue_and_residual(self):
+        from runbookai_amd.ops import _get_ext
+        ext = _get_ext()
+        M, N, K = 1, 512, 2048   # K = intermediate
+        gu = bf(torch.randn(M, 2 * K)).to(DEV)
+        w = bf(torch.randn(N, K) * 0.1).to(DEV)
+        res = bf(torch.randn(M, N)).to(DEV)
+        out = ext.decode_gemv(gu, w, None, res, 2, 1e-5)
+        g, u = gu.cpu().chunk(2, dim=-1)
+        act = ref.silu_mul(g.contiguous(), u.contiguous())
+        expected = act.float() @ w.cpu().float().t() + res.cpu().float()
+        assert (out.cpu().float() - expected).abs().max().item() < 0.6
+
+    def test_splitk_matches_nosplit(self):
+        """K large enough to trigger the in-launch split-K combine; the
+        result must match a single-split run (combine-protocol screen)."""
+        from runbookai_amd.ops import _get_ext
+        ext = _get_ext()
+        M, N, K = 2, 512, 8192   # nblk=2 -> splitk 16
+        x = bf(torch.randn(M, K) * 0.5).to(DEV)
+        w = bf(torch.randn(N, K) * 0.1).to(DEV)
+        outs = [ext.decode_gemv(x, w, None, None, 0, 1e-5) for _ in range(5)]
+        expected = x.cpu().float() @ w.cpu().float().t()
+        for o in outs:   # repeated: counter reset must survive relaunch
+            assert (o.cpu().float() - expected).abs().max().item() < 2.0
+
+    def test_fused_decode_matches_generic(self):
+        """Whole-model parity: the fused decode body (B<=4) vs the generic
+        kernel path on the same weights/KV."""
+        import runbookai_amd.engine.llama as L
+        torch.manual_seed(3)
+        cfg = L.CONFIGS["tiny"]
+        m = L.LlamaModel(cfg, device=DEV, kv_blocks=64, seed=11)
+        m.use_graphs = False
+        kv = m.kv
+        B, ctx = 2, 33
+        for s in range(B):
+            kv.allocate(50 + s, ctx + 8)
+            kv.set_len(50 + s, ctx)
+            # populate history KV with random content
+            kv.k[0][:, :, :, :].normal_(0, 0.3)
+        for li in range(cfg.num_layers):
+            m.kv.k[li].normal_(0, 0.3)
+            m.kv.v[li].normal_(0, 0.3)
+        bt, lens = kv.batch_tables([50 + s for s in range(B)], DEV)
+        ids = torch.randint(0, 255, (B,), dtype=torch.long).to(DEV)
+        pos = torch.full((B,), ctx - 1, dtype=torch.int32).to(DEV)
+        slots = torch.cat([kv.slot_mapping(50 + s, ctx - 1, 1)
+                           for s in range(B)]).to(torch.int32).to(DEV)
+        m.use_decode_fused = True
+        out_fused = m._decode_impl(ids, pos, bt.to(DEV), lens.to(DEV), slots)
+        # the fused run scattered this step's k/v; generic run overwrites
+        # the same slots with identical values, so state matches
+        m.use_decode_fused = False
+        out_generic = m._decode_impl(ids, pos, bt.to(DEV), lens.to(DEV), slots)
+        diff = (out_fused.float() - out_generic.float()).abs().max().item()
+        assert diff < 0.5, diff
