@@ -1,0 +1,237 @@
+"""Synthetic investigation-trace generation for policy training.
+
+An ORACLE client answers the real InvestigationOrchestrator's prompts
+correctly (deterministically, from the case's expected root cause), while
+every (prompt, response) pair is recorded. The recorded pairs are exactly
+the serving distribution: the same prompt builders (agent/llm_parser
+PROMPTS), the same simulated telemetry (providers/simulation), the same
+knowledge-base snippets — so a model trained on them serves through
+LLMEngine(checkpoint=...) with no distribution shift.
+
+Case variety comes from ARCHETYPES x sampled service names/symptoms: each
+generated case builds its scenario with SimScenario.from_fixture, whose
+telemetry embeds the expected keywords — the learnable signal is
+"telemetry content -> root cause", which transfers to the held-out eval
+fixtures built the same way.
+
+The reference has no counterpart (its investigation quality is bought
+from hosted frontier models); this module is what makes BASELINE.md's
+accuracy axis measurable on local hardware.
+"""
+from __future__ import annotations
+
+import random
+from dataclasses import dataclass, field
+from typing import Any, Optional
+
+from ..agent.llm_parser import split_schema_tag
+
+# (root-cause phrase template, keyword pool, symptom pool, service pool)
+ARCHETYPES: list[dict[str, Any]] = [
+    {"name": "conn-pool-exhaustion",
+     "keywords": ["connection", "pool", "exhausted"],
+     "extra": [["redis"], ["postgres"], ["mysql"], ["database"]],
+     "symptoms": ["latency spike", "timeouts acquiring connections",
+                  "p99 latency over SLO"],
+     "services": [["checkout-api", "cart-service"], ["orders-api", "billing"],
+                  ["session-service", "auth-api"]]},
+    {"name": "gateway-5xx",
+     "keywords": ["upstream", "timeout", "5xx"],
+     "extra": [["gateway"], ["nginx"], ["envoy"]],
+     "symptoms": ["HTTP 502 rate climbing", "upstream connect errors",
+                  "gateway error budget burn"],
+     "services": [["api-gateway", "user-service"], ["edge-proxy", "search-api"]]},
+    {"name": "disk-pressure",
+     "keywords": ["disk", "full", "retention"],
+     "extra": [["kafka"], ["broker"], ["log segments"]],
+     "symptoms": ["producer latency rising", "consumer lag growing",
+                  "broker disk usage over 95%"],
+     "services": [["kafka-broker", "order-events"], ["event-bus", "analytics-sink"]]},
+    {"name": "cert-expiry",
+     "keywords": ["certificate", "expired", "tls"],
+     "extra": [["handshake"], ["x509"]],
+     "symptoms": ["TLS handshake failures", "unknown certificate errors",
+                  "payment callbacks failing"],
+     "services": [["payments-api", "bank-connector"], ["webhook-dispatcher", "partner-api"]]},
+    {"name": "oom-crashloop",
+     "keywords": ["memory", "oom", "killed"],
+     "extra": [["container"], ["heap"], ["limit"]],
+     "symptoms": ["pods restarting", "OOMKilled events", "rss near limit"],
+     "services": [["recommendation-svc", "feed-api"], ["ml-inference", "ranker"]]},
+    {"name": "deploy-regression",
+     "keywords": ["deployment", "regression", "rollback"],
+     "extra": [["release"], ["canary"], ["bad config"]],
+     "symptoms": ["error rate up after deploy", "new version crashlooping",
+                  "canary failing health checks"],
+     "services": [["catalog-api", "inventory"], ["pricing-svc", "promo-engine"]]},
+    {"name": "db-cpu-saturation",
+     "keywords": ["cpu", "saturation", "slow queries"],
+     "extra": [["database"], ["index"], ["full scan"]],
+     "symptoms": ["query latency growing", "db cpu pegged at 100%",
+                  "application timeouts"],
+     "services": [["orders-db", "orders-api"], ["reporting-db", "dashboard-api"]]},
+    {"name": "dns-resolution",
+     "keywords": ["dns", "resolution", "failing"],
+     "extra": [["ndots"], ["coredns"], ["nameserver"]],
+     "symptoms": ["intermittent name resolution errors", "SERVFAIL responses",
+                  "connection setup failures"],
+     "services": [["service-mesh", "discovery"], ["coredns", "cluster-api"]]},
+    {"name": "rate-limit",
+     "keywords": ["rate", "limit", "throttled"],
+     "extra": [["quota"], ["429"], ["burst"]],
+     "symptoms": ["HTTP 429 responses", "third-party quota exceeded",
+                  "request queue growing"],
+     "services": [["notification-svc", "email-provider"], ["sync-worker", "crm-api"]]},
+    {"name": "cache-stampede",
+     "keywords": ["cache", "miss", "stampede"],
+     "extra": [["ttl"], ["thundering herd"], ["eviction"]],
+     "symptoms": ["cache hit rate collapsed", "origin load spiking",
+                  "latency p95 rising"],
+     "services": [["content-cache", "cms-api"], ["edge-cache", "media-service"]]},
+    {"name": "queue-backlog",
+     "keywords": ["queue", "backlog", "consumer"],
+     "extra": [["lag"], ["stuck"], ["dead letter"]],
+     "symptoms": ["message age growing", "consumers stalled",
+                  "processing delayed"],
+     "services": [["job-queue", "worker-pool"], ["task-broker", "batch-runner"]]},
+    {"name": "network-partition",
+     "keywords": ["network", "partition", "unreachable"],
+     "extra": [["packet loss"], ["link"], ["az"]],
+     "symptoms": ["cross-zone call failures", "elevated packet loss",
+                  "health checks flapping"],
+     "services": [["zone-b-fleet", "replicator"], ["cluster-peer", "gossip-mesh"]]},
+]
+
+
+def gen_case(rng: random.Random, idx: int) -> dict[str, Any]:
+    """One fixture-style case dict drawn from the archetype pools."""
+    arch = rng.choice(ARCHETYPES)
+    services = list(rng.choice(arch["services"]))
+    extra = rng.choice(arch["extra"])
+    keywords = list(arch["keywords"]) + extra
+    symptom = rng.choice(arch["symptoms"])
+    incident = f"PD-GEN-{idx:05d}"
+    return {
+        "id": f"{arch['name']}-{idx}",
+        "incidentId": incident,
+        "query": (f"Investigate incident {incident}: {services[0]} "
+                  f"{symptom}"),
+        "context": f"{services[0]}: {' '.join(keywords)} detected",
+        "expected": {
+            "rootCauseKeywords": keywords,
+            "affectedServices": services,
+            "minimumConfidence": "medium",
+        },
+        "execute": {"maxIterations": 4},
+    }
+
+
+@dataclass
+class OracleClient:
+    """Answers orchestrator prompts correctly from the case's expected
+    data; optionally records every (prompt, response) pair."""
+
+    case: dict[str, Any]
+    rng: random.Random = field(default_factory=lambda: random.Random(0))
+    records: Optional[list[dict[str, str]]] = None
+
+    def _kw(self) -> list[str]:
+        return self.case["expected"]["rootCauseKeywords"]
+
+    def _svc(self) -> list[str]:
+        return self.case["expected"]["affectedServices"]
+
+    def _record(self, prompt: str, response: str) -> None:
+        if self.records is not None:
+            kind, body = split_schema_tag(prompt)
+            self.records.append({"kind": kind or "", "body": body,
+                                 "response": response})
+
+    def complete(self, prompt: str) -> str:
+        import json
+
+        kind, _body = split_schema_tag(prompt)
+        kw, svc = self._kw(), self._svc()
+        phrase = " ".join(kw)
+        if kind == "triage":
+            out = {"summary": f"{svc[0]} degraded: {phrase}",
+                   "symptoms": [f"{phrase} on {svc[0]}"],
+                   "affectedServices": svc[:4],
+                   "severity": "high"}
+        elif kind == "generateHypotheses":
+            decoy_arch = self.rng.choice(ARCHETYPES)
+            out = {"hypotheses": [
+                {"statement": f"{svc[0]} failing due to {phrase}",
+                 "rationale": f"telemetry shows {phrase} on {svc[0]}",
+                 "priority": 1, "affectedServices": svc[:3]},
+                {"statement": f"possible {' '.join(decoy_arch['keywords'][:2])} on {svc[-1]}",
+                 "rationale": "secondary signal, lower confidence",
+                 "priority": 3},
+            ]}
+        elif kind == "evaluateEvidence":
+            # confirm the hypothesis when it carries the true keywords
+            hyp_match = all(k.lower() in prompt.lower() for k in kw[:2])
+            out = {"action": "confirm" if hyp_match else "prune",
+                   "confidence": 0.9 if hyp_match else 0.2,
+                   "reasoning": (f"evidence shows {phrase}" if hyp_match
+                                 else "no supporting signal"),
+                   "evidence": [{"description": f"{phrase} observed on {svc[0]}",
+                                 "supports": hyp_match}]}
+        elif kind == "generateConclusion":
+            out = {"rootCause": f"{svc[0]}: {phrase}",
+                   "confidence": "high",
+                   "summary": f"Root cause of the incident is {phrase} on "
+                              f"{svc[0]}; downstream impact on {', '.join(svc[1:2])}",
+                   "affectedServices": svc[:4],
+                   "evidence": [f"{phrase} in telemetry"]}
+        elif kind == "generateRemediation":
+            out = {"summary": f"Mitigate {phrase} on {svc[0]}",
+                   "steps": [
+                       {"description": f"Address {phrase} on {svc[0]}",
+                        "risk": "medium", "requiresApproval": True},
+                       {"description": "Verify recovery and close incident",
+                        "risk": "low", "requiresApproval": False}]}
+        elif kind == "analyzeLogs":
+            out = {"summary": f"log patterns show {phrase}",
+                   "patterns": [{"pattern": phrase, "severity": "error"}],
+                   "services": svc[:3]}
+        else:
+            out = {"answer": f"{phrase} on {svc[0]}"}
+        resp = json.dumps(out)
+        self._record(prompt, resp)
+        return resp
+
+    def chat(self, system: str, user: str, tools=None):  # pragma: no cover
+        from ..engine.client import ChatResponse
+
+        return ChatResponse(content=self.complete(user))
+
+
+def generate_traces(n_cases: int, seed: int = 0,
+                    retriever=None) -> list[dict[str, str]]:
+    """Run `n_cases` oracle-driven investigations through the REAL
+    orchestrator against from_fixture scenarios; returns recorded
+    (kind, body, response) dicts."""
+    from ..agent.orchestrator import InvestigationOrchestrator
+    from ..providers.simulation import SimScenario, set_thread_scenario
+    from ..tools.registry import ToolRegistry
+
+    if retriever is None:
+        from ..knowledge.retriever.default import create_retriever
+
+        retriever = create_retriever(in_memory=True)
+        retriever.sync()
+    rng = random.Random(seed)
+    records: list[dict[str, str]] = []
+    for i in range(n_cases):
+        case = gen_case(rng, i)
+        set_thread_scenario(SimScenario.from_fixture(case))
+        oracle = OracleClient(case, rng=random.Random(seed * 7919 + i),
+                              records=records)
+        registry = ToolRegistry(knowledge_retriever=retriever)
+        orch = InvestigationOrchestrator(
+            llm=oracle, tool_executor=registry, knowledge_retriever=retriever,
+            max_iterations=int(case["execute"]["maxIterations"]))
+        orch.investigate(case["query"], incident_id=case.get("incidentId"))
+    set_thread_scenario(None)
+    return records

@@ -1,0 +1,116 @@
+"""Policy-training pipeline: trace generation, trainable model parity with
+the inference engine, and the export->serve round trip (CPU)."""
+import json
+import random
+
+import pytest
+import torch
+
+from runbookai_amd.engine.train import CONFIGS, TrainableLlama, to_inference_model
+from runbookai_amd.evals.trace_gen import OracleClient, gen_case, generate_traces
+
+
+def test_gen_case_fields():
+    rng = random.Random(0)
+    case = gen_case(rng, 3)
+    assert case["expected"]["rootCauseKeywords"]
+    assert case["expected"]["affectedServices"]
+    assert case["incidentId"].startswith("PD-GEN-")
+
+
+def test_oracle_answers_parse_and_score():
+    """Oracle responses must parse through the real llm_parser schemas and
+    score >= threshold with the reference scorer (they are the training
+    targets — a non-scoring oracle would teach the wrong policy)."""
+    from runbookai_amd.agent.llm_parser import (
+        parse_conclusion,
+        parse_hypothesis_generation,
+        parse_triage_response,
+    )
+    from runbookai_amd.evals.scoring import score_investigation_result
+
+    rng = random.Random(1)
+    for i in range(6):
+        case = gen_case(rng, i)
+        oracle = OracleClient(case)
+        tri = parse_triage_response(oracle.complete(
+            "\x00schema:triage\x00investigate"))
+        assert tri["severity"] in ("low", "medium", "high", "critical")
+        hyp = parse_hypothesis_generation(oracle.complete(
+            "\x00schema:generateHypotheses\x00go"))
+        assert hyp and hyp[0]["statement"]
+        conc = parse_conclusion(oracle.complete(
+            "\x00schema:generateConclusion\x00go"))
+        result = {"rootCause": conc["rootCause"], "confidence": conc["confidence"],
+                  "affectedServices": conc.get("affectedServices", []),
+                  "summary": conc["summary"]}
+        score = score_investigation_result(result, case["expected"])
+        assert score["overall"] >= 0.7, (case["id"], score)
+
+
+def test_generate_traces_covers_phases():
+    recs = generate_traces(2, seed=5)
+    kinds = {r["kind"] for r in recs}
+    assert {"triage", "generateHypotheses", "generateConclusion"} <= kinds
+    assert all(r["body"] and r["response"] for r in recs)
+
+
+def test_trainable_matches_inference_model():
+    """TrainableLlama forward == LlamaModel prefill on the same weights
+    (same RoPE/RMSNorm/SwiGLU conventions), so trained weights serve
+    unchanged."""
+    cfg = CONFIGS["tiny"]
+    torch.manual_seed(0)
+    tm = TrainableLlama(cfg)
+    inf = to_inference_model(tm, device="cpu", kv_blocks=16)
+    T = 9
+    ids = torch.arange(T) + 5
+    logits_t = tm(ids.unsqueeze(0))[0, -1]
+    inf.kv.allocate(1, T)
+    logits_i = inf.prefill(ids, torch.arange(T, dtype=torch.int32),
+                           torch.tensor([0, T], dtype=torch.int32),
+                           inf.kv.slot_mapping(1, 0, T))[0]
+    # inference path is bf16; compare top-5 agreement + correlation
+    t5t = set(logits_t.topk(5).indices.tolist())
+    t5i = set(logits_i.float().topk(5).indices.tolist())
+    assert len(t5t & t5i) >= 3, (t5t, t5i)
+    corr = torch.corrcoef(torch.stack(
+        [logits_t.float(), logits_i.float()]))[0, 1]
+    assert corr > 0.98, float(corr)
+
+
+@pytest.mark.slow
+def test_train_export_serve_round_trip(tmp_path):
+    """Two optimizer steps -> export -> LLMEngine(checkpoint) serves with
+    the BPE grammar active and produces schema-valid JSON."""
+    import subprocess
+    import sys as _sys
+
+    out = tmp_path / "ckpt"
+    proc = subprocess.run(
+        [_sys.executable, "scripts/train_policy.py", "--cases", "4",
+         "--steps", "2", "--out", str(out), "--max-len", "768",
+         "--batch-tokens", "4096"],
+        capture_output=True, text=True, timeout=600)
+    assert proc.returncode == 0, proc.stderr[-1500:]
+    from runbookai_amd.agent.llm_parser import parse_json
+    from runbookai_amd.engine.client import LocalEngineClient
+    from runbookai_amd.engine.engine import LLMEngine
+
+    engine = LLMEngine(device="cpu", checkpoint=str(out), kv_blocks=128)
+    assert engine.supports_bpe_grammar
+    # a bounded schema the 2-step model can complete within the budget:
+    # grammar-constrained output must be valid JSON by construction
+    schema = {"type": "object",
+              "properties": {"a": {"type": "string", "maxLength": 16}},
+              "required": ["a"]}
+    tok = engine.hf_tokenizer
+    req = engine.generate(tok.encode_chat("sys", "answer"), max_new_tokens=64,
+                          schema=schema)
+    data = parse_json(tok.decode(req.out_ids))
+    assert isinstance(data, dict) and "a" in data
+    # and the client path produces SOMETHING textual through the chat wrap
+    client = LocalEngineClient(engine, max_tokens=24)
+    text = client.complete("\x00schema:triage\x00What is wrong with checkout?")
+    assert isinstance(text, str)
+    engine.shutdown()
