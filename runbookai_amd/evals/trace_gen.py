@@ -102,6 +102,129 @@ ARCHETYPES: list[dict[str, Any]] = [
      "services": [["zone-b-fleet", "replicator"], ["cluster-peer", "gossip-mesh"]]},
 ]
 
+NATURAL_TELEMETRY: dict[str, dict] = {
+ "conn-pool-exhaustion": {
+  "logs": [
+   "could not acquire connection from pool after 5000ms ({extra0} pool size 50, in use 50)",
+   "{extra0}: max clients reached, rejecting new connections",
+   "timeout waiting for a free connection slot; pool utilization at 100%",
+   "upstream call aborted: connection pool saturated"
+  ],
+  "alarm": "connection wait time p99 over threshold on {svc}",
+  "metric": "{svc}.pool_wait_ms"
+ },
+ "gateway-5xx": {
+  "logs": [
+   "upstream unavailable: {svc1} timed out after 10s (gateway timeout)",
+   "504 gateway timeout routing /v1/requests",
+   "proxy error: no healthy upstream endpoints for {svc1}",
+   "5xx rate over 5% across listener group"
+  ],
+  "alarm": "5xx rate > 5% for 10 minutes on {svc}",
+  "metric": "{svc}.5xx_rate"
+ },
+ "disk-pressure": {
+  "logs": [
+   "disk usage on /var/lib/{extra0} at 97%, segments cannot be rolled",
+   "No space left on device while appending to partition log",
+   "retention enforcement behind: oldest segments not deleted",
+   "producer request timed out; broker flush stalled on full volume"
+  ],
+  "alarm": "disk usage above 95% on {svc}",
+  "metric": "{svc}.disk_used_pct"
+ },
+ "cert-expiry": {
+  "logs": [
+   "x509: certificate has expired or is not yet valid (notAfter in the past)",
+   "TLS handshake error from peer: bad certificate",
+   "ssl verification failed calling {svc1}: certificate expired",
+   "remote rejected connection during handshake"
+  ],
+  "alarm": "TLS handshake failure rate rising on {svc}",
+  "metric": "{svc}.tls_handshake_errors"
+ },
+ "oom-crashloop": {
+  "logs": [
+   "container killed: OOMKilled (memory limit 512Mi exceeded)",
+   "java.lang.OutOfMemoryError: heap space exhausted",
+   "pod restarted 7 times in 10m; last state: OOMKilled",
+   "rss climbed to limit before kill signal"
+  ],
+  "alarm": "container restarts exceed threshold on {svc}",
+  "metric": "{svc}.memory_rss_bytes"
+ },
+ "deploy-regression": {
+  "logs": [
+   "error rate tripled within 2m of rollout v{ver} on {svc}",
+   "new release failing readiness probe; previous version was healthy",
+   "canary analysis failed: comparison against baseline exceeded error budget",
+   "panic in handler introduced by latest deploy"
+  ],
+  "alarm": "error budget burn after deployment on {svc}",
+  "metric": "{svc}.error_rate"
+ },
+ "db-cpu-saturation": {
+  "logs": [
+   "slow query log: SELECT ... full table scan taking 12s",
+   "database CPU at 100%; connection queue building",
+   "query planner falling back to sequential scan (missing index?)",
+   "statement timeout reached for 14% of queries"
+  ],
+  "alarm": "database CPU utilization sustained above 95% on {svc}",
+  "metric": "{svc}.cpu_util"
+ },
+ "dns-resolution": {
+  "logs": [
+   "lookup {svc1} on 10.0.0.2:53: server misbehaving (SERVFAIL)",
+   "dial tcp: lookup failed: no such host (intermittent)",
+   "coredns: upstream nameserver timeout, retrying",
+   "resolution latency spiking; ndots expansion generating extra queries"
+  ],
+  "alarm": "DNS error rate above threshold on {svc}",
+  "metric": "{svc}.dns_errors"
+ },
+ "rate-limit": {
+  "logs": [
+   "HTTP 429 Too Many Requests from provider API (quota exceeded)",
+   "request rejected: rate limit bucket empty, retry after 30s",
+   "burst above contracted quota; requests being throttled",
+   "backoff engaged after repeated 429 responses"
+  ],
+  "alarm": "throttled request rate rising on {svc}",
+  "metric": "{svc}.throttled_requests"
+ },
+ "cache-stampede": {
+  "logs": [
+   "cache hit ratio fell from 98% to 11% after key expiry wave",
+   "origin overwhelmed: concurrent regeneration of the same hot key",
+   "TTL expiry storm: thousands of misses for identical entries",
+   "eviction pressure high; working set exceeds cache capacity"
+  ],
+  "alarm": "cache hit ratio below threshold on {svc}",
+  "metric": "{svc}.cache_hit_ratio"
+ },
+ "queue-backlog": {
+  "logs": [
+   "queue depth growing: 1.2M messages, oldest 42m",
+   "consumer group stalled; no acknowledgements in 10m",
+   "dead letter queue receiving poison messages repeatedly",
+   "processing lag exceeds SLA for downstream jobs"
+  ],
+  "alarm": "message age above threshold on {svc}",
+  "metric": "{svc}.queue_depth"
+ },
+ "network-partition": {
+  "logs": [
+   "peer unreachable: i/o timeout dialing 10.2.0.0/16 endpoints",
+   "packet loss to zone-b above 30% on inter-az link",
+   "gossip membership flapping; suspect marks rising",
+   "replication halted: cannot reach quorum across zones"
+  ],
+  "alarm": "cross-zone connectivity degraded for {svc}",
+  "metric": "{svc}.packet_loss_pct"
+ }
+}
+
 
 def gen_case(rng: random.Random, idx: int) -> dict[str, Any]:
     """One fixture-style case dict drawn from the archetype pools."""
@@ -113,6 +236,7 @@ def gen_case(rng: random.Random, idx: int) -> dict[str, Any]:
     incident = f"PD-GEN-{idx:05d}"
     return {
         "id": f"{arch['name']}-{idx}",
+        "_arch": arch["name"],
         "incidentId": incident,
         "query": (f"Investigate incident {incident}: {services[0]} "
                   f"{symptom}"),
@@ -124,6 +248,60 @@ def gen_case(rng: random.Random, idx: int) -> dict[str, Any]:
         },
         "execute": {"maxIterations": 4},
     }
+
+
+def build_scenario(case: dict[str, Any], rng: random.Random):
+    """Scenario with NATURAL telemetry phrasing (log lines written like the
+    hand-built demo scenarios, not keyword echoes): the policy must learn
+    telemetry -> root-cause extraction, which is what transfers to the
+    held-out eval scenarios."""
+    from ..providers.simulation import SimScenario
+
+    arch_name = case.get("_arch")
+    tmpl = NATURAL_TELEMETRY.get(arch_name)
+    svc = case["expected"]["affectedServices"]
+    kw = case["expected"]["rootCauseKeywords"]
+    s = SimScenario(name=case["id"])
+    s.incident = {
+        "id": case["incidentId"], "title": case["query"],
+        "status": "triggered", "urgency": rng.choice(["high", "critical"]),
+        "service": svc[0], "createdAt": "2026-02-12T10:00:00Z",
+    }
+    noise_svc = rng.choice(["metrics-agent", "log-shipper", "cron-runner",
+                            "fluentd", "health-prober"])
+    s.services = ([{"name": x, "status": "degraded" if i < 2 else "healthy",
+                    "type": "ecs"} for i, x in enumerate(svc)]
+                  + [{"name": noise_svc, "status": "healthy", "type": "ecs"}])
+    fmt = {"svc": svc[0], "svc1": svc[-1], "extra0": kw[-1],
+           "ver": f"{rng.randint(1,9)}.{rng.randint(0,20)}.{rng.randint(0,9)}"}
+    if tmpl:
+        s.alarms = [{"name": f"{svc[0]}-alert", "state": "ALARM",
+                     "reason": tmpl["alarm"].format(**fmt), "service": svc[0]}]
+        msgs = list(tmpl["logs"])
+        rng.shuffle(msgs)
+        for i, msg in enumerate(msgs[:rng.randint(2, 4)]):
+            s.log_events.append({
+                "timestamp": f"2026-02-12T10:{i:02d}:{(i * 13) % 60:02d}Z",
+                "service": rng.choice(svc), "level": "ERROR",
+                "message": msg.format(**fmt)})
+        # noise line that must NOT become the root cause
+        s.log_events.append({
+            "timestamp": "2026-02-12T10:09:00Z", "service": noise_svc,
+            "level": "INFO", "message": "scheduled healthcheck completed"})
+        base = rng.uniform(40, 300)
+        s.metrics = {tmpl["metric"].format(**fmt):
+                     [round(base * (1 + 0.4 * i), 1) for i in range(6)]}
+        s.monitors = [{"name": f"{svc[0]} {arch_name}", "status": "Alert",
+                       "query": f"avg:{tmpl['metric'].format(**fmt)} > 10"}]
+    else:
+        return SimScenario.from_fixture(case)
+    if arch_name == "deploy-regression":
+        s.deployments = [{"service": svc[0], "version": f"v{fmt['ver']}",
+                          "at": "2026-02-12T09:58:00Z",
+                          "change": f"release v{fmt['ver']}"}]
+    s.resources = {"ecs": [{"name": x, "desiredCount": 3, "runningCount": 3,
+                            "taskDefinition": f"{x}:1"} for x in svc]}
+    return s
 
 
 @dataclass
@@ -213,7 +391,7 @@ def generate_traces(n_cases: int, seed: int = 0,
     orchestrator against from_fixture scenarios; returns recorded
     (kind, body, response) dicts."""
     from ..agent.orchestrator import InvestigationOrchestrator
-    from ..providers.simulation import SimScenario, set_thread_scenario
+    from ..providers.simulation import set_thread_scenario
     from ..tools.registry import ToolRegistry
 
     if retriever is None:
@@ -225,7 +403,7 @@ def generate_traces(n_cases: int, seed: int = 0,
     records: list[dict[str, str]] = []
     for i in range(n_cases):
         case = gen_case(rng, i)
-        set_thread_scenario(SimScenario.from_fixture(case))
+        set_thread_scenario(build_scenario(case, rng))
         oracle = OracleClient(case, rng=random.Random(seed * 7919 + i),
                               records=records)
         registry = ToolRegistry(knowledge_retriever=retriever)
