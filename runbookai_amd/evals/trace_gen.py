@@ -42,8 +42,8 @@ ARCHETYPES: list[dict[str, Any]] = [
                   "gateway error budget burn"],
      "services": [["api-gateway", "user-service"], ["edge-proxy", "search-api"]]},
     {"name": "disk-pressure",
-     "keywords": ["disk", "full", "retention"],
-     "extra": [["kafka"], ["broker"], ["log segments"]],
+     "keywords": ["disk", "space", "retention"],
+     "extra": [["kafka"], ["broker"], ["full"]],
      "symptoms": ["producer latency rising", "consumer lag growing",
                   "broker disk usage over 95%"],
      "services": [["kafka-broker", "order-events"], ["event-bus", "analytics-sink"]]},
@@ -94,6 +94,12 @@ ARCHETYPES: list[dict[str, Any]] = [
      "symptoms": ["message age growing", "consumers stalled",
                   "processing delayed"],
      "services": [["job-queue", "worker-pool"], ["task-broker", "batch-runner"]]},
+    {"name": "gateway-deploy",
+     "keywords": ["deploy", "gateway", "timeout", "upstream"],
+     "extra": [["5xx"], ["regression"], ["rollback"]],
+     "symptoms": ["5xx spike minutes after a deploy", "upstream timeouts after release",
+                  "gateway errors following rollout"],
+     "services": [["api-gateway", "user-service"], ["edge-router", "profile-api"]]},
     {"name": "network-partition",
      "keywords": ["network", "partition", "unreachable"],
      "extra": [["packet loss"], ["link"], ["az"]],
@@ -213,6 +219,16 @@ NATURAL_TELEMETRY: dict[str, dict] = {
   "alarm": "message age above threshold on {svc}",
   "metric": "{svc}.queue_depth"
  },
+ "gateway-deploy": {
+  "logs": [
+   "upstream unavailable: {svc1} timed out after 10s (gateway timeout)",
+   "panic: nil pointer dereference in handler v{ver} (deployed minutes ago)",
+   "504 gateway timeout routing /v1/users",
+   "5xx rate jumped from 0.1% to 9% right after the {svc1} rollout"
+  ],
+  "alarm": "5xx rate > 5% for 10 minutes on {svc}",
+  "metric": "{svc}.5xx_rate"
+ },
  "network-partition": {
   "logs": [
    "peer unreachable: i/o timeout dialing 10.2.0.0/16 endpoints",
@@ -226,10 +242,28 @@ NATURAL_TELEMETRY: dict[str, dict] = {
 }
 
 
+_SVC_HEADS = ["auth", "user", "cart", "order", "billing", "search", "media",
+              "sync", "edge", "feed", "profile", "payment", "invoice", "ship",
+              "geo", "notify", "ledger", "catalog", "session", "ingest"]
+_SVC_TAILS = ["api", "svc", "service", "worker", "gateway", "db", "cache",
+              "broker", "proxy", "runner", "engine", "store"]
+
+
+def _svc_name(rng: random.Random) -> str:
+    return f"{rng.choice(_SVC_HEADS)}-{rng.choice(_SVC_TAILS)}"
+
+
 def gen_case(rng: random.Random, idx: int) -> dict[str, Any]:
-    """One fixture-style case dict drawn from the archetype pools."""
+    """One fixture-style case dict drawn from the archetype pools. 70% of
+    cases use freshly combined service names — the policy cannot memorize
+    an archetype->service mapping and must COPY names from the prompt."""
     arch = rng.choice(ARCHETYPES)
-    services = list(rng.choice(arch["services"]))
+    if rng.random() < 0.7:
+        services = [_svc_name(rng) for _ in range(2)]
+        while services[1] == services[0]:
+            services[1] = _svc_name(rng)
+    else:
+        services = list(rng.choice(arch["services"]))
     extra = rng.choice(arch["extra"])
     keywords = list(arch["keywords"]) + extra
     symptom = rng.choice(arch["symptoms"])
@@ -356,7 +390,7 @@ class OracleClient:
                    "evidence": [{"description": f"{phrase} observed on {svc[0]}",
                                  "supports": hyp_match}]}
         elif kind == "generateConclusion":
-            out = {"rootCause": f"{svc[0]}: {phrase}",
+            out = {"rootCause": f"{phrase} on {' and '.join(svc[:2])}",
                    "confidence": "high",
                    "summary": f"Root cause of the incident is {phrase} on "
                               f"{svc[0]}; downstream impact on {', '.join(svc[1:2])}",
