@@ -31,6 +31,12 @@ CONFIGS.setdefault("policy-small", LlamaConfig(
     name="policy-small", hidden_size=512, intermediate_size=1408,
     num_layers=6, num_heads=8, num_kv_heads=4, head_dim=64,
     vocab_size=4096, max_seq_len=4096))
+# deeper variant: conclusion-stage induction copying (carry the confirmed
+# hypothesis phrase through the final prompt) wants more layers
+CONFIGS.setdefault("policy-base", LlamaConfig(
+    name="policy-base", hidden_size=512, intermediate_size=1408,
+    num_layers=8, num_heads=8, num_kv_heads=4, head_dim=64,
+    vocab_size=4096, max_seq_len=4096))
 
 
 class RMSNorm(nn.Module):

@@ -228,7 +228,9 @@ def eval_checkpoint(ckpt: str, concurrency: int = 4) -> dict:
         score = score_investigation_result(result.to_dict(), case.get("expected", {}))
         return {"id": case["id"], "score": round(score["overall"], 3),
                 "passed": score["overall"] >= 0.7,
-                "rootCause": (result.root_cause or "")[:120]}
+                "rootCause": (result.root_cause or "")[:120],
+                "hypotheses": [h.get("statement", "")[:90]
+                               for h in (result.hypotheses or [])[:3]]}
 
     t0 = time.time()
     with ThreadPoolExecutor(max_workers=concurrency) as pool:
