@@ -226,8 +226,10 @@ def main() -> None:
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16",
-            "data": "synthetic (simulated incident set, random-init weights, "
-                    "grammar-constrained decoding)",
+            "data": ("synthetic (simulated incident set, trained policy "
+                     "checkpoint, grammar-constrained decoding)" if args.checkpoint
+                     else "synthetic (simulated incident set, random-init "
+                          "weights, grammar-constrained decoding)"),
             "config": {
                 "model": model_name,
                 "checkpoint": args.checkpoint or "random-init",
