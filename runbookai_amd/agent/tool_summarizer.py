@@ -159,16 +159,191 @@ def _sum_pagerduty(args: dict[str, Any], data: Any) -> CompactToolResult:
     return _generic("pagerduty", args, data)
 
 
+def _sum_prometheus(args: dict[str, Any], data: Any) -> CompactToolResult:
+    """Dedicated prometheus summarizer (reference summarizePrometheus,
+    tool-summarizer.ts:558): instant/range series counts + firing alerts."""
+    action = args.get("action", "instant")
+    if isinstance(data, dict) and "alerts" in data:
+        alerts = data.get("alerts") or []
+        firing = [a for a in alerts if isinstance(a, dict)
+                  and a.get("state") in ("firing", "Alert", "ALERT")]
+        return CompactToolResult(
+            summary=f"prometheus alerts: {len(firing)}/{len(alerts)} firing",
+            highlights=[str(a.get("name", a.get("alertname", "?")))
+                        for a in firing[:3]],
+            item_count=len(alerts), has_errors=bool(firing),
+            health_status="alarming" if firing else "ok")
+    series = (data.get("result") or data.get("series") or []) \
+        if isinstance(data, dict) else (data or [])
+    peak = ""
+    try:
+        vals = [float(v) for s in series if isinstance(s, dict)
+                for v in ([p[1] for p in s.get("values", [])]
+                          or [s.get("value", [0, 0])[1]])]
+        if vals:
+            peak = f", peak {max(vals):g}"
+    except (TypeError, ValueError, IndexError):
+        pass
+    return CompactToolResult(
+        summary=f"prometheus {action} '{args.get('query', '')}': "
+                f"{len(series)} series{peak}",
+        item_count=len(series), has_errors=_find_errors(data))
+
+
+def _sum_incident_list(args: dict[str, Any], data: Any) -> CompactToolResult:
+    """pagerduty_list_incidents / opsgenie_list_* (reference
+    summarizePagerdutyList, tool-summarizer.ts:436): counts by status,
+    top titles."""
+    items = []
+    if isinstance(data, dict):
+        items = (data.get("incidents") or data.get("alerts") or [])
+    elif isinstance(data, list):
+        items = data
+    by_status: dict[str, int] = {}
+    for it in items:
+        if isinstance(it, dict):
+            st = str(it.get("status", it.get("state", "?")))
+            by_status[st] = by_status.get(st, 0) + 1
+    status_str = ", ".join(f"{n} {s}" for s, n in sorted(by_status.items()))
+    open_like = sum(n for s, n in by_status.items()
+                    if s.lower() in ("triggered", "acknowledged", "open"))
+    return CompactToolResult(
+        summary=f"{len(items)} incidents/alerts ({status_str})" if items
+                else "no open incidents",
+        highlights=[str(it.get("title", it.get("message", "?")))[:90]
+                    for it in items[:3] if isinstance(it, dict)],
+        item_count=len(items), has_errors=open_like > 0,
+        services=_extract_services(items))
+
+
+def _sum_incident_action(args: dict[str, Any], data: Any) -> CompactToolResult:
+    """Note/ack/close style incident mutations: one-line receipt."""
+    ok = not _find_errors(data)
+    target = args.get("incident_id") or args.get("alert_id") or args.get("id", "?")
+    return CompactToolResult(
+        summary=f"incident action on {target}: {'ok' if ok else 'FAILED'}",
+        item_count=1, has_errors=not ok)
+
+
+def _sum_slack(args: dict[str, Any], data: Any) -> CompactToolResult:
+    if isinstance(data, dict) and "messages" in data:
+        msgs = data.get("messages") or []
+        return CompactToolResult(
+            summary=f"slack thread: {len(msgs)} messages in "
+                    f"{args.get('channel', '?')}",
+            highlights=[str(m.get('text', ''))[:90] for m in msgs[:3]
+                        if isinstance(m, dict)],
+            item_count=len(msgs))
+    ok = not _find_errors(data)
+    return CompactToolResult(
+        summary=f"slack message to {args.get('channel', '?')}: "
+                f"{'sent' if ok else 'FAILED'}",
+        item_count=1, has_errors=not ok)
+
+
+def _sum_code_fix(args: dict[str, Any], data: Any) -> CompactToolResult:
+    """github_query / gitlab_query fix_candidates: top candidate titles."""
+    cands = []
+    if isinstance(data, dict):
+        cands = (data.get("candidates") or data.get("results")
+                 or data.get("items") or [])
+    return CompactToolResult(
+        summary=f"{len(cands)} code-fix candidates for "
+                f"'{args.get('query', args.get('service', ''))}'",
+        highlights=[str(c.get("title", c.get("path", "?")))[:90]
+                    for c in cands[:3] if isinstance(c, dict)],
+        item_count=len(cands))
+
+
+def _sum_skill(args: dict[str, Any], data: Any) -> CompactToolResult:
+    name = args.get("name", args.get("skill", "?"))
+    action = args.get("action", "execute")
+    steps = []
+    if isinstance(data, dict):
+        steps = data.get("steps") or data.get("results") or []
+    failed = sum(1 for s in steps if isinstance(s, dict)
+                 and (s.get("error") or s.get("status") == "failed"))
+    return CompactToolResult(
+        summary=f"skill {action} '{name}': {len(steps)} steps"
+                + (f", {failed} failed" if failed else ""),
+        item_count=len(steps) or 1, has_errors=failed > 0 or _find_errors(data))
+
+
+def _sum_diagram(args: dict[str, Any], data: Any) -> CompactToolResult:
+    """Diagram/chart tools: the rendering is bulky ASCII — keep it out of
+    the compact context entirely; the agent drills down by result id."""
+    kind = args.get("type", args.get("chart_type", "diagram"))
+    size = len(str(data)) if data is not None else 0
+    return CompactToolResult(
+        summary=f"rendered {kind} ({size} chars; use get_full_result to view)",
+        item_count=1)
+
+
+def _sum_aws_cli(args: dict[str, Any], data: Any) -> CompactToolResult:
+    cmd = str(args.get("command", ""))[:80]
+    lines = str(data).count("\n") + 1 if data else 0
+    return CompactToolResult(
+        summary=f"aws cli `{cmd}`: {lines} output lines",
+        item_count=lines, has_errors=_find_errors(data))
+
+
+def _sum_aws_mutate(args: dict[str, Any], data: Any) -> CompactToolResult:
+    ok = not _find_errors(data)
+    return CompactToolResult(
+        summary=f"aws_mutate {args.get('operation', '?')} on "
+                f"{args.get('resource', '?')}: {'applied' if ok else 'FAILED'}",
+        item_count=1, has_errors=not ok,
+        services=_extract_services(args))
+
+
+def _sum_context(args: dict[str, Any], data: Any) -> CompactToolResult:
+    """get_full_result / list_results drill-down tools: never re-summarize
+    (their whole point is raw access); note the retrieval only."""
+    rid = args.get("result_id", "")
+    n = _count_items(data)
+    return CompactToolResult(
+        summary=(f"retrieved full result {rid}" if rid
+                 else f"listed {n} stored results"),
+        item_count=max(n, 1))
+
+
+# Per-tool summarizers for the FULL tool surface (reference keeps 8 and
+# lets the rest bloat the generic path, tool-summarizer.ts:723-740; every
+# unsummarized tool here costs context budget at the compaction tier)
 SUMMARIZERS: dict[str, Callable[[dict[str, Any], Any], CompactToolResult]] = {
     "aws_query": _sum_aws_query,
+    "aws_mutate": _sum_aws_mutate,
+    "aws_cli": _sum_aws_cli,
     "cloudwatch_alarms": _sum_cloudwatch_alarms,
     "cloudwatch_logs": _sum_cloudwatch_logs,
     "datadog": _sum_datadog,
-    "prometheus": _sum_datadog,
+    "prometheus": _sum_prometheus,
     "search_knowledge": _sum_search_knowledge,
     "kubernetes_query": _sum_kubernetes,
     "pagerduty_get_incident": _sum_pagerduty,
+    "pagerduty_list_incidents": _sum_incident_list,
+    "pagerduty_add_note": _sum_incident_action,
     "opsgenie_get_incident": _sum_pagerduty,
+    "opsgenie_get_alert": _sum_pagerduty,
+    "opsgenie_list_alerts": _sum_incident_list,
+    "opsgenie_list_incidents": _sum_incident_list,
+    "opsgenie_add_note": _sum_incident_action,
+    "opsgenie_acknowledge_alert": _sum_incident_action,
+    "opsgenie_close_alert": _sum_incident_action,
+    "slack_post_update": _sum_slack,
+    "slack_post_root_cause": _sum_slack,
+    "slack_read_thread": _sum_slack,
+    "slack_message": _sum_slack,
+    "github_query": _sum_code_fix,
+    "gitlab_query": _sum_code_fix,
+    "skill": _sum_skill,
+    "generate_flowchart": _sum_diagram,
+    "generate_sequence_diagram": _sum_diagram,
+    "generate_architecture_diagram": _sum_diagram,
+    "visualize_metrics": _sum_diagram,
+    "render_mermaid": _sum_diagram,
+    "get_full_result": _sum_context,
+    "list_results": _sum_context,
 }
 
 
