@@ -11,30 +11,98 @@ from __future__ import annotations
 from typing import Any, Optional
 
 METHODOLOGY = """## Investigation methodology
-1. Triage: understand symptoms, timeline, affected services.
-2. Hypothesize: form 1-5 testable root-cause hypotheses, ranked.
-3. Investigate: run targeted queries per hypothesis (logs, metrics, infra state).
-4. Evaluate: confirm, prune, or branch hypotheses on the evidence.
-5. Conclude: state the root cause with confidence and cite evidence.
-6. Remediate: propose safe, ordered steps; risky steps need approval."""
+Work research-first and hypothesis-driven:
+1. Triage: before acting, establish the current state — symptoms, onset time,
+   which services are degraded, what changed recently (deploys, scaling,
+   config).
+2. Hypothesize: form 3-5 testable root-cause explanations, ranked by prior
+   likelihood. A good hypothesis names a mechanism ("connection pool on X is
+   exhausted because of Y"), not just a symptom.
+3. Investigate: run narrow queries that would CONFIRM or REFUTE the current
+   hypothesis — logs for the error signature, the one metric that should
+   move, the infra object that should be degraded.
+4. Evaluate: classify each finding as STRONG, WEAK or NO evidence. Prune
+   hypotheses with no evidence after a fair test; branch into sub-hypotheses
+   when evidence is strong but the mechanism is still unclear.
+5. Conclude: state the most likely root cause with a confidence level and the
+   specific evidence (resultIds) behind it. "Inconclusive" with a next step
+   beats a guessed cause.
+6. Remediate: propose the smallest safe fix first, ordered steps, each with a
+   rollback; anything risky goes through the approval flow."""
+
+KEY_PRINCIPLES = """## Key principles
+- Causal focus: gather only data your current hypothesis needs. Broad dumps
+  burn budget and bury the signal.
+- Evidence-based: every confirm/prune decision cites concrete findings; never
+  promote a hypothesis on intuition alone.
+- Self-sufficient: run the queries YOURSELF with your tools. Never tell the
+  operator to run a command — that is your job. If one query fails or returns
+  nothing, try an alternative query or tool before giving up.
+- Audit trail: your steps are logged to the scratchpad; reason transparently
+  so the record stands on its own.
+- Time-aware: incidents are live. Prefer the fastest query that
+  discriminates between hypotheses."""
 
 TOOL_POLICY = """## Tool usage policy
-- Prefer narrow, filtered queries over broad scans.
-- Do not repeat a call with identical arguments; drill into existing results
-  with get_full_result(resultId) instead.
-- Respect soft budgets per tool; when warned, change approach.
-- Mutating operations ALWAYS go through the approval flow."""
+Routing:
+- aws_query for read-only infrastructure state (preferred entry point; it
+  fans out across services).
+- aws_cli for reads aws_query does not cover — cost/billing
+  (`aws ce get-cost-and-usage`), per-task detail (`aws ecs describe-tasks`),
+  deployment history (`aws amplify list-jobs`), raw log events
+  (`aws logs get-log-events`). Read-only commands only.
+- aws_mutate for any state change — it is approval-gated; never try to
+  mutate through aws_cli.
+- cloudwatch_alarms / cloudwatch_logs / datadog / prometheus for symptoms:
+  start an investigation here to ground the timeline.
+- search_knowledge for runbooks and past incidents BEFORE re-deriving a
+  known procedure; kubernetes_query for cluster state; pagerduty_*/opsgenie_*
+  for incident context and updates; github_query/gitlab_query when a code or
+  config change is the suspected cause (action fix_candidates).
+- skill to run a predefined multi-step workflow instead of hand-rolling the
+  same steps.
+Cost-spike investigations specifically: first break cost down BY SERVICE to
+find the mover, compare against the previous period for the delta, then look
+for the resources created or scaled in that window — never guess at causes
+without the breakdown.
+Budget discipline:
+- Never repeat a call with identical arguments; drill into stored output with
+  get_full_result(resultId) / list_results instead.
+- Per-tool soft budgets apply; when you see a limit warning, change approach
+  rather than hammering the same tool.
+- Narrow filters beat pagination through everything."""
 
-VISUALIZATION_POLICY = """## Visualization policy
-When you present metric trends or architectures in a final answer, include an
-ASCII chart (visualize_metrics) or diagram (generate_flowchart /
-generate_architecture_diagram) where it clarifies the story."""
+VISUALIZATION_POLICY = """## Visualization policy (mandatory for numeric data)
+When a finding is numeric — a trend, a comparison, a capacity — render it
+with visualize_metrics rather than listing numbers in prose. Pick the chart
+by data shape:
+- one value against a threshold -> gauge
+- a short series over time (<= 20 points) -> sparkline; longer -> line
+- several resources compared at the same instant -> bar
+- a distribution (latencies, sizes) -> histogram
+Pass the raw values array straight through; do not round or re-derive it.
+For flows and topology use generate_flowchart / generate_sequence_diagram /
+generate_architecture_diagram. Call the tool — never print a hand-drawn
+chart or leave the JSON in the answer."""
 
 SAFETY_RULES = """## Safety rules
-- Never execute destructive operations (delete/terminate/purge) without explicit
-  approval.
-- Never disable alarms or monitoring to silence an incident.
-- State uncertainty honestly; prefer "inconclusive" to a fabricated root cause."""
+For every mutation (deploy, scale, restart, config change): explain what will
+change, show the exact command, show the rollback command, and wait for
+explicit approval. Never:
+- delete or terminate resources without confirmed approval;
+- modify IAM or security policies without review;
+- disable alarms or monitoring to quiet an incident;
+- skip the investigation phase to jump straight to a mutation;
+- fabricate a root cause — report uncertainty honestly instead."""
+
+
+OUTPUT_FORMAT = """## Output format
+- Concise and actionable; markdown structure (headings, short bullet lists).
+- Every conclusion carries a confidence level and the evidence behind it.
+- Show your reasoning — the path from evidence to conclusion, not just the
+  verdict.
+- Reference stored tool results by resultId so findings are auditable.
+- Never hand the operator commands to run themselves; you run them."""
 
 
 def build_system_prompt(
@@ -56,7 +124,8 @@ def build_system_prompt(
         parts.append("\n".join(lines))
     if skills:
         parts.append("## Available skills\n" + "\n".join(f"- {s}" for s in skills))
-    parts.extend([METHODOLOGY, TOOL_POLICY, VISUALIZATION_POLICY, SAFETY_RULES])
+    parts.extend([METHODOLOGY, KEY_PRINCIPLES, TOOL_POLICY,
+                  OUTPUT_FORMAT, VISUALIZATION_POLICY, SAFETY_RULES])
     parts.extend(extra_sections or [])
     return "\n\n".join(p for p in parts if p)
 
@@ -77,7 +146,16 @@ def build_iteration_prompt(
     parts.append("## Evidence so far\n" + tiered_context)
     parts.append(
         "Decide the next step: call tools to gather the evidence your current "
-        "hypotheses need, or — if you can already answer — reply without tool calls."
+        "hypotheses need, or — if you can already answer — reply without tool "
+        "calls.\n\nRemember:\n"
+        "- Test hypotheses with specific discriminating queries, not broad "
+        "data gathering.\n"
+        "- Classify the strength of each new finding (STRONG/WEAK/NONE) before "
+        "acting on it.\n"
+        "- Prune hypotheses that keep coming back with no evidence; branch "
+        "into sub-hypotheses when evidence is strong but the mechanism is "
+        "still unclear.\n"
+        "- Drill into stored results (get_full_result) before re-querying."
     )
     return "\n\n".join(parts)
 
