@@ -384,3 +384,78 @@ class TestFullSurfaceSummarizers:
     def test_context_drilldown(self):
         s = self._s("get_full_result", {"result_id": "r42"}, {"data": 1})
         assert "r42" in s.summary
+
+
+class TestInfraContextDepth:
+    """Pre-discovery behavior sized to the reference's cases
+    (infra-context.ts:212-437): per-service health rules, alarm service
+    extraction, health rollup thresholds, key-service identification."""
+
+    class FakeExec:
+        def execute(self, tool, args):
+            if tool == "aws_query" and args.get("service") == "ecs":
+                return {"items": [
+                    {"name": "checkout", "status": "ACTIVE",
+                     "desiredCount": 3, "runningCount": 3},
+                    {"name": "cart", "status": "DRAINING",
+                     "desiredCount": 3, "runningCount": 1},
+                    {"name": "a"}, {"name": "b"}, {"name": "c"},
+                ]}
+            if tool == "aws_query" and args.get("service") == "ec2":
+                return {"items": [{"state": "running"}, {"state": "stopped"}]}
+            if tool == "cloudwatch_alarms":
+                return {"alarms": [
+                    {"name": "checkout-alarm", "state": "ALARM",
+                     "reason": "latency"},
+                    {"name": "noisy", "state": "ALARM",
+                     "dimensions": [{"Name": "ServiceName", "Value": "cart"}]},
+                ]}
+            if tool == "aws_query" and args.get("service") == "codedeploy":
+                return {"items": [{"service": "checkout", "version": "v2"}]}
+            return {"items": []}
+
+    def _mgr(self):
+        from runbookai_amd.agent.infra_context import InfraContextManager
+        m = InfraContextManager(tool_executor=self.FakeExec())
+        m.discover()
+        return m
+
+    def test_per_service_health_counting(self):
+        m = self._mgr()
+        ecs = m.inventory["ecs"]
+        # ACTIVE + 3 status-less count healthy; DRAINING w/ mismatch doesn't
+        assert ecs["count"] == 5 and ecs["unhealthy"] == 1
+        assert m.inventory["ec2"]["unhealthy"] == 1
+
+    def test_alarm_service_extraction(self):
+        m = self._mgr()
+        by_name = {a["name"]: a for a in m.alarms}
+        assert by_name["checkout-alarm"]["service"] == "checkout"  # name pattern
+        assert by_name["noisy"]["service"] == "cart"               # dimension
+
+    def test_health_rollup_thresholds(self):
+        m = self._mgr()
+        h = m.health
+        assert h["overall"] == "degraded"   # warnings + 2 alarms (not >2)
+        assert h["healthy"] == 5 and h["warning"] == 2
+
+    def test_key_services_by_count_and_alarms(self):
+        m = self._mgr()
+        ks = m.key_services()
+        assert "ecs" in ks          # count >= 5
+        assert "ec2" in ks          # unhealthy > 0
+        assert "checkout" in ks     # alarm-owning service
+
+    def test_prompt_overview_sections(self):
+        m = self._mgr()
+        text = m.prompt_overview()
+        assert "Service inventory" in text
+        assert "Active alarms" in text
+        assert "Recent deployments" in text
+        assert "ecs: 5 resource(s) (1 unhealthy)" in text
+
+    def test_cache_staleness(self):
+        m = self._mgr()
+        first = m._discovered_at
+        m.discover()   # within the 5-min window: no re-discovery
+        assert m._discovered_at == first
