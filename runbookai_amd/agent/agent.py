@@ -166,7 +166,10 @@ class Agent:
             if knowledge_md and iteration == 1:
                 user += "\n\n## Retrieved knowledge\n" + knowledge_md
             if self.conversation_memory:
-                conv = self.conversation_memory.get_context_for_prompt()
+                # related-context recall: past investigations / messages
+                # matching THIS query surface alongside the rolling window
+                conv = self.conversation_memory.get_context_for_prompt(
+                    query=query)
                 if conv:
                     user = conv + "\n\n" + user
 

@@ -90,12 +90,62 @@ class ConversationMemory:
         scored.sort(key=lambda t: t[0], reverse=True)
         return [s for _, s in scored[:limit]]
 
+    def get_related_context(self, query: str) -> dict[str, Any]:
+        """Combined recall for a new turn (reference getRelatedContext,
+        conversation-memory.ts:539-552): past investigations whose query,
+        digest OR touched services match, plus the matching recent
+        messages. Matching is token-overlap (jaccard) rather than the
+        reference's substring test, so paraphrases recall too."""
+        q_lower = query.lower()
+        stop = {"on", "the", "a", "an", "is", "are", "was", "to", "of", "in",
+                "and", "or", "with", "for", "why", "what", "did", "do"}
+        q_words = {w for w in q_lower.split() if w not in stop}
+
+        def _score(text: str) -> float:
+            words = {w for w in text.lower().split() if w not in stop}
+            if not words or not q_words:
+                return 0.0
+            return len(words & q_words) / len(words | q_words)
+
+        invs = [s for s in self.investigations
+                if _score(f"{s.query} {s.answer_digest}") > 0.08]
+        invs.sort(key=lambda s: -_score(f"{s.query} {s.answer_digest}"))
+        invs = invs[:3]
+        # service-mention recall: the user naming a service a past
+        # investigation touched should surface it even with no word overlap
+        for s in self.investigations:
+            if s in invs:
+                continue
+            if any(svc and svc.lower() in q_lower for svc in s.services):
+                invs.append(s)
+        return {"investigations": invs[:3],
+                "messages": self.search(query, limit=5)}
+
+    def related_context_section(self, query: str) -> str:
+        """Prompt section for the chat turn; empty when nothing recalls."""
+        rel = self.get_related_context(query)
+        if not rel["investigations"] and not rel["messages"]:
+            return ""
+        lines = ["## Related earlier context"]
+        for s in rel["investigations"]:
+            svcs = f" [{', '.join(s.services[:3])}]" if s.services else ""
+            lines.append(f"- Past investigation{svcs}: {s.query} -> "
+                         f"{s.answer_digest[:160]}")
+        for m in rel["messages"][:3]:
+            lines.append(f"- {m.role} said earlier: {m.content[:140]}")
+        return "\n".join(lines)
+
     # -- prompt context (reference L249-292) ---------------------------------
 
-    def get_context_for_prompt(self, token_budget: int = 2000) -> str:
+    def get_context_for_prompt(self, token_budget: int = 2000,
+                               query: str = "") -> str:
         parts: list[str] = []
         if self.compressed_summary:
             parts.append("## Earlier conversation (compressed)\n" + self.compressed_summary)
+        if query:
+            rel = self.related_context_section(query)
+            if rel:
+                parts.append(rel)
         recent: list[str] = []
         used = estimate_tokens("\n".join(parts))
         for m in reversed(self.messages):

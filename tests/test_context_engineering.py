@@ -459,3 +459,43 @@ class TestInfraContextDepth:
         first = m._discovered_at
         m.discover()   # within the 5-min window: no re-discovery
         assert m._discovered_at == first
+
+
+class TestConversationRecall:
+    """Related-context recall (reference conversation-memory.ts:228-246,
+    539-552): past investigations surface for new queries by word overlap
+    AND by service mention."""
+
+    def _mem(self):
+        from runbookai_amd.agent.conversation_memory import ConversationMemory
+        m = ConversationMemory()
+        m.add_investigation(
+            "Investigate checkout latency spike",
+            "Root cause: redis connection pool exhausted on checkout-api",
+            services=["checkout-api", "redis"])
+        m.add_investigation(
+            "Why did deploys fail yesterday", "CI runner out of disk",
+            services=["ci-runner"])
+        m.add_message("user", "the cart page is slow again")
+        m.add_message("assistant", "checking redis pool metrics")
+        return m
+
+    def test_recall_by_word_overlap(self):
+        rel = self._mem().get_related_context("checkout latency is back")
+        assert any("checkout" in s.query for s in rel["investigations"])
+
+    def test_recall_by_service_mention(self):
+        # no word overlap with the stored query — only the service name
+        rel = self._mem().get_related_context("anything odd with redis?")
+        assert any("redis" in s.services for s in rel["investigations"])
+
+    def test_section_renders_and_empty_when_unrelated(self):
+        m = self._mem()
+        sec = m.related_context_section("checkout latency again")
+        assert "Related earlier context" in sec and "redis connection pool" in sec
+        assert m.related_context_section("weather on mars zzz") == ""
+
+    def test_prompt_context_includes_recall(self):
+        m = self._mem()
+        ctx = m.get_context_for_prompt(query="checkout latency again")
+        assert "Related earlier context" in ctx
