@@ -177,8 +177,71 @@ class ToolRegistry:
                           "authorize", "purge", "register", "deregister", "run-instances")
     _DATE_EXPR = re.compile(r"\b(now|today|yesterday)(?:\s*-\s*(\d+)([dhm]))?\b")
 
+    @staticmethod
+    def preprocess_date_expressions(command: str) -> str:
+        """Resolve shell date substitutions to REAL dates computed now
+        (reference preprocessDateExpressions, registry.ts:1375-1455):
+        GNU `$(date -d 'N days ago' +%Y-%m-%d)`, BSD `$(date -v-30d ...)`,
+        `$(date +%Y-%m-%d)`, yesterday / last month / first day of last
+        month — plus this repo's bare now/today/yesterday[-Nd/h/m] forms.
+        Runs BEFORE the shell-operator guard so the benign `$(date ...)`
+        forms pass while any other `$(...)` is still rejected."""
+        import datetime as _dt
+
+        now = _dt.datetime.now(_dt.timezone.utc)
+
+        def _fmt(d: _dt.datetime) -> str:
+            return d.strftime("%Y-%m-%d")
+
+        def _shift(days: int = 0, months: int = 0) -> _dt.datetime:
+            d = now - _dt.timedelta(days=days)
+            if months:
+                m = d.month - 1 - months
+                d = d.replace(year=d.year + m // 12, month=m % 12 + 1, day=1)
+            return d
+
+        def gnu(m: "re.Match[str]") -> str:
+            n, unit = int(m.group(1)), m.group(2).lower()
+            days = n * (7 if unit.startswith("week") else 1)
+            if unit.startswith("month"):
+                return _fmt(_shift(months=n))
+            return _fmt(_shift(days=days))
+
+        def bsd(m: "re.Match[str]") -> str:
+            n, unit = int(m.group(1)), m.group(2).lower()
+            return _fmt(_shift(days=n * {"d": 1, "w": 7, "m": 30, "y": 365}[unit]))
+
+        cmd = command
+        cmd = re.sub(r"\$\(date\s+-d\s+['\"]?(\d+)\s+(day|days|week|weeks|month|"
+                     r"months)\s+ago['\"]?\s+\+%Y-%m-%d\)", gnu, cmd, flags=re.I)
+        cmd = re.sub(r"\$\(date\s+-v-(\d+)([dwmy])\s+\+%Y-%m-%d\)", bsd, cmd,
+                     flags=re.I)
+        cmd = re.sub(r"\$\(date\s+-d\s+['\"]?first\s+day\s+of\s+last\s+month"
+                     r"['\"]?\s+\+%Y-%m-%d\)", _fmt(_shift(months=1)), cmd, flags=re.I)
+        cmd = re.sub(r"\$\(date\s+-d\s+['\"]?last\s+month['\"]?\s+\+%Y-%m-%d\)",
+                     _fmt(_shift(days=30)), cmd, flags=re.I)
+        cmd = re.sub(r"\$\(date\s+-d\s+['\"]?yesterday['\"]?\s+\+%Y-%m-%d\)",
+                     _fmt(_shift(days=1)), cmd, flags=re.I)
+        cmd = re.sub(r"\$\(date\s+\+%Y-%m-%d\)", _fmt(now), cmd, flags=re.I)
+        # bare relative forms: now-2h, yesterday, today-3d
+        def bare(m: "re.Match[str]") -> str:
+            base = now
+            if m.group(1) == "yesterday":
+                base = base - _dt.timedelta(days=1)
+            if m.group(2):
+                n, unit = int(m.group(2)), m.group(3)
+                base -= _dt.timedelta(days=n if unit == "d" else 0,
+                                      hours=n if unit == "h" else 0,
+                                      minutes=n if unit == "m" else 0)
+            return base.strftime("%Y-%m-%dT%H:%M:%SZ")
+
+        cmd = ToolRegistry._DATE_EXPR.sub(bare, cmd)
+        return cmd
+
     def _aws_cli(self, command: str, **_: Any) -> dict[str, Any]:
-        cmd = command.strip()
+        # resolve the benign $(date ...) substitutions FIRST; any remaining
+        # shell construct is rejected below (reference order, L1620-1635)
+        cmd = self.preprocess_date_expressions(command.strip())
         if self._SHELL_OPERATORS.search(cmd):
             raise ValueError("aws_cli: shell operators are not allowed")
         try:
@@ -187,22 +250,57 @@ class ToolRegistry:
             raise ValueError(f"aws_cli: unparseable command: {e}") from None
         if not parts or parts[0] != "aws":
             raise ValueError("aws_cli: command must start with 'aws'")
-        if len(parts) < 3:
+        # skip global flags before the service token (reference
+        # parseAwsCliServiceAndOperation, registry.ts:1507-1527)
+        i = 1
+        while i < len(parts) and parts[i].startswith("-"):
+            flag = parts[i]
+            nxt = parts[i + 1] if i + 1 < len(parts) else None
+            if not flag.startswith("--no-") and nxt and not nxt.startswith("-"):
+                i += 2
+            else:
+                i += 1
+        if i + 1 >= len(parts):
             raise ValueError("aws_cli: expected 'aws <service> <operation> ...'")
-        service, operation = parts[1], parts[2]
+        service, operation = parts[i], parts[i + 1]
         op_lower = operation.lower()
         if any(op_lower.startswith(k) or k in op_lower for k in self._MUTATION_KEYWORDS):
             raise ValueError(f"aws_cli: operation '{operation}' is not read-only — "
                              "use aws_mutate (approval-gated) instead")
         if not any(op_lower.startswith(v) for v in self._READ_ONLY_VERBS):
             raise ValueError(f"aws_cli: operation '{operation}' is not on the read-only whitelist")
-        # date-expression preprocessing (reference L1375-1455): now-2h etc.
-        resolved = self._DATE_EXPR.sub("2026-02-12T00:00:00Z", cmd)
+        # build the structured invocation the executor runs (simulated
+        # backend): option flags become the operation's parameters
+        options: dict[str, str] = {}
+        j = i + 2
+        while j < len(parts):
+            tok = parts[j]
+            if tok.startswith("--"):
+                key = tok[2:]
+                nxt = parts[j + 1] if j + 1 < len(parts) else None
+                if nxt is not None and not nxt.startswith("--"):
+                    options[key] = nxt
+                    j += 2
+                else:
+                    options[key] = "true"
+                    j += 1
+            else:
+                j += 1
         sdef = get_service(service)
-        if sdef is None:
+        if sdef is None and service not in ("ce", "logs"):
             raise ValueError(f"aws_cli: unknown service '{service}'")
-        result = execute_list_operation(service, operation)
-        out: dict[str, Any] = {"command": resolved, **result}
+        invocation = {"service": service, "operation": operation,
+                      "options": options,
+                      "region": options.get("region", "us-east-1")}
+        result = execute_list_operation(service, operation) if sdef else {"items": []}
+        out: dict[str, Any] = {"command": cmd, "invocation": invocation, **result}
+        if "limit" in options or "max-items" in options:
+            try:
+                lim = int(options.get("limit", options.get("max-items", "50")))
+                if isinstance(out.get("items"), list):
+                    out["items"] = out["items"][:lim]
+            except ValueError:
+                pass
         # auto cost-chart for cost-explorer-ish queries (reference L1637-1668)
         if "cost" in cmd or service == "ce":
             scenario = get_scenario()

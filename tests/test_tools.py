@@ -248,3 +248,62 @@ class TestContextTools:
             assert listing["results"][0]["tool"] == "aws_query"
         finally:
             set_active_scratchpad(None)
+
+
+class TestAwsCliRealism:
+    """aws_cli date-expression preprocessing + structured invocation
+    (reference registry.ts:1375-1455, 1507-1527): real computed dates,
+    CLI flag parsing, whitelist after substitution."""
+
+    def _r(self):
+        from runbookai_amd.tools.registry import ToolRegistry
+        return ToolRegistry()
+
+    def test_gnu_date_expression_resolves_to_real_dates(self):
+        import datetime as dt
+        out = self._r().execute("aws_cli", {"command":
+            "aws ce get-cost-and-usage --time-period "
+            "Start=$(date -d '30 days ago' +%Y-%m-%d),End=$(date +%Y-%m-%d) "
+            "--granularity MONTHLY"})
+        today = dt.datetime.now(dt.timezone.utc).strftime("%Y-%m-%d")
+        start = (dt.datetime.now(dt.timezone.utc)
+                 - dt.timedelta(days=30)).strftime("%Y-%m-%d")
+        assert f"End={today}" in out["command"]
+        assert f"Start={start}" in out["command"]
+
+    def test_bsd_and_relative_forms(self):
+        out = self._r().execute("aws_cli", {"command":
+            "aws logs get-log-events --start-time now-2h"})
+        assert "now-2h" not in out["command"]
+        assert "T" in out["command"] and "Z" in out["command"]
+        out2 = self._r().execute("aws_cli", {"command":
+            "aws ec2 describe-instances --filters created=$(date -v-7d +%Y-%m-%d)"})
+        assert "$(date" not in out2["command"]
+
+    def test_invocation_structure_built(self):
+        out = self._r().execute("aws_cli", {"command":
+            "aws ecs list-tasks --cluster prod --region eu-west-1 --max-items 2"})
+        inv = out["invocation"]
+        assert inv["service"] == "ecs" and inv["operation"] == "list-tasks"
+        assert inv["options"]["cluster"] == "prod"
+        assert inv["region"] == "eu-west-1"
+
+    def test_global_flags_before_service_are_skipped(self):
+        out = self._r().execute("aws_cli", {"command":
+            "aws --profile prod --no-cli-pager ecs list-services"})
+        assert out["invocation"]["service"] == "ecs"
+
+    def test_non_date_shell_construct_still_rejected(self):
+        import pytest as _pytest
+        with _pytest.raises(Exception):
+            self._r().execute("aws_cli", {"command":
+                "aws ecs list-tasks --cluster $(whoami)"})
+        with _pytest.raises(Exception):
+            self._r().execute("aws_cli", {"command":
+                "aws s3 ls; rm -rf /"})
+
+    def test_mutations_still_blocked_after_preprocessing(self):
+        import pytest as _pytest
+        with _pytest.raises(Exception):
+            self._r().execute("aws_cli", {"command":
+                "aws ec2 terminate-instances --instance-ids i-1"})
