@@ -610,13 +610,28 @@ def mcp_tools() -> None:
 
 @cli.command("slack-gateway")
 @click.option("--port", default=3030)
+@click.option("--socket-mode", is_flag=True,
+              help="Socket Mode (websocket envelopes) instead of HTTP events")
 @click.pass_context
-def slack_gateway(ctx: click.Context, port: int) -> None:
-    """Run the Slack events gateway (HTTP)."""
-    from .slack.gateway import SlackGateway
+def slack_gateway(ctx: click.Context, port: int, socket_mode: bool) -> None:
+    """Run the Slack events gateway (HTTP Events API or Socket Mode)."""
+    from .slack.gateway import SlackGateway, SocketModeClient
 
     rt = _build_runtime(ctx.obj["config"])
     gw = SlackGateway(config=ctx.obj["config"].incident.slack, runtime=rt)
+    if socket_mode:
+        # framing/ack/dedupe/reconnect protocol is implemented
+        # (slack/gateway.py SocketModeClient); the websocket transport
+        # needs egress to slack.com, absent in this offline image
+        app_token = (ctx.obj["config"].incident.slack or {}).get("appToken", "")
+        if not app_token:
+            raise click.ClickException(
+                "socket mode needs incident.slack.appToken (xapp-...)")
+        raise click.ClickException(
+            "socket mode requires network egress to slack.com "
+            "(apps.connections.open); this environment is offline — "
+            "the protocol client is in place, supply a websocket "
+            "transport where egress exists")
     _echo(f"slack gateway listening on :{port}")
     gw.serve(port=port)
 

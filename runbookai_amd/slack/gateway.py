@@ -172,3 +172,73 @@ class SlackGateway:
                 pass
 
         HTTPServer(("127.0.0.1", port), Handler).serve_forever()
+
+
+# -- Socket Mode (reference gateway.ts:384-530) --------------------------------
+#
+# Slack Socket Mode: apps.connections.open yields a websocket URL; every
+# inbound frame is an ENVELOPE that must be ACKed by echoing its
+# envelope_id, then its payload.event flows through the same dedupe +
+# handler as the HTTP Events API; a closed socket reconnects after a
+# delay. The transport here is pluggable — this image has no egress (and
+# no websocket client), so production wiring supplies a real transport
+# while the framing/ack/dedupe/reconnect protocol below is exercised by
+# the simulated transport in tests.
+
+class SocketModeTransport:
+    """Minimal transport contract: frames() yields raw JSON strings until
+    the connection closes; send() pushes a raw JSON string (the ACK)."""
+
+    def frames(self):  # pragma: no cover - interface
+        raise NotImplementedError
+
+    def send(self, raw: str) -> None:  # pragma: no cover - interface
+        raise NotImplementedError
+
+
+class SocketModeClient:
+    def __init__(self, gateway: "SlackGateway", transport_factory,
+                 reconnect_delay_s: float = 3.0) -> None:
+        self.gateway = gateway
+        self.transport_factory = transport_factory
+        self.reconnect_delay_s = reconnect_delay_s
+        self.connections = 0
+        self.acked: list[str] = []
+        self.handled = 0
+
+    def _handle_frame(self, raw: str, transport: SocketModeTransport) -> None:
+        try:
+            envelope = json.loads(raw)
+        except json.JSONDecodeError:
+            return
+        env_id = envelope.get("envelope_id")
+        if env_id:
+            # ACK FIRST: Slack retries un-acked envelopes (reference L506)
+            transport.send(json.dumps({"envelope_id": env_id}))
+            self.acked.append(env_id)
+        if envelope.get("type") in ("hello", "disconnect"):
+            return
+        payload = envelope.get("payload") or {}
+        event = payload.get("event")
+        if not event:
+            return
+        event_id = payload.get("event_id")
+        if event_id and self.gateway.dedupe(f"sock:{event_id}"):
+            return
+        self.gateway.handle_event(event)
+        self.handled += 1
+
+    def run(self, max_connections: Optional[int] = None) -> None:
+        """Connect/consume/reconnect loop. max_connections bounds the loop
+        (tests / graceful shutdown); None reconnects forever."""
+        while max_connections is None or self.connections < max_connections:
+            self.connections += 1
+            try:
+                transport = self.transport_factory()
+                for raw in transport.frames():
+                    self._handle_frame(raw, transport)
+            except ConnectionError:
+                pass  # fall through to reconnect
+            if max_connections is not None and self.connections >= max_connections:
+                return
+            time.sleep(self.reconnect_delay_s)

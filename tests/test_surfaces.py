@@ -336,3 +336,91 @@ class TestOperability:
                          spool_path=spool)
         assert n == 1
         assert spool_status(spool)["spooled"] == 0
+
+
+class TestSocketMode:
+    """Socket Mode framing (reference gateway.ts:470-530): envelope ACK
+    before processing, hello/disconnect frames ignored, event dedupe,
+    reconnect loop over transport failures."""
+
+    class FakeTransport:
+        def __init__(self, frame_batches):
+            self.batches = frame_batches
+            self.sent = []
+            self.connects = 0
+
+        def factory(self):
+            import json as _json
+            from runbookai_amd.slack.gateway import SocketModeTransport
+
+            outer = self
+
+            class _T(SocketModeTransport):
+                def __init__(self):
+                    outer.connects += 1
+                    if outer.connects > len(outer.batches):
+                        self.batch = []
+                    else:
+                        self.batch = outer.batches[outer.connects - 1]
+                    if self.batch == "FAIL":
+                        raise ConnectionError("connect refused")
+
+                def frames(self):
+                    yield from self.batch
+
+                def send(self, raw):
+                    outer.sent.append(_json.loads(raw))
+
+            return _T
+
+    def _gateway(self):
+        from runbookai_amd.slack.gateway import SlackGateway
+
+        return SlackGateway(config={}, runtime={})
+
+    def _env(self, env_id, event_id, text="help", ts="1.0"):
+        import json
+
+        return json.dumps({
+            "envelope_id": env_id, "type": "events_api",
+            "payload": {"event_id": event_id,
+                        "event": {"type": "app_mention", "channel": "C1",
+                                  "user": "U1", "text": text, "ts": ts}}})
+
+    def test_ack_and_handle(self):
+        import json
+        from runbookai_amd.slack.gateway import SocketModeClient
+
+        gw = self._gateway()
+        hello = json.dumps({"type": "hello"})
+        ft = self.FakeTransport([[hello, self._env("e1", "Ev1")]])
+        client = SocketModeClient(gw, ft.factory(), reconnect_delay_s=0.0)
+        client.run(max_connections=1)
+        assert ft.sent == [{"envelope_id": "e1"}]   # ACKed
+        assert client.handled == 1
+        assert gw.replies and "commands" in gw.replies[0]["text"].lower()
+
+    def test_dedupe_across_frames(self):
+        from runbookai_amd.slack.gateway import SocketModeClient
+
+        gw = self._gateway()
+        ft = self.FakeTransport([[self._env("e1", "EvX"),
+                                  self._env("e2", "EvX")]])   # same event_id
+        client = SocketModeClient(gw, ft.factory(), reconnect_delay_s=0.0)
+        client.run(max_connections=1)
+        assert len(ft.sent) == 2       # both envelopes ACKed regardless
+        assert client.handled == 1     # but processed once
+
+    def test_reconnects_after_drop_and_failure(self):
+        from runbookai_amd.slack.gateway import SocketModeClient
+
+        gw = self._gateway()
+        ft = self.FakeTransport([
+            [self._env("e1", "Ev1", ts="1.0")],
+            "FAIL",                                  # connection refused
+            [self._env("e2", "Ev2", ts="2.0")],
+        ])
+        client = SocketModeClient(gw, ft.factory(), reconnect_delay_s=0.0)
+        client.run(max_connections=3)
+        assert client.connections == 3
+        assert client.handled == 2
