@@ -251,11 +251,21 @@ class Agent:
             memory.build_final_summary(),
             self.hypothesis_engine.to_markdown(),
         )
-        resp = self.llm.chat(
-            "You are Runbook, an SRE investigation agent writing a final incident answer.",
-            final_prompt,
-        )
-        answer = resp.content or self.last_answer or "Investigation produced no conclusive answer."
+        final_system = ("You are Runbook, an SRE investigation agent writing "
+                        "a final incident answer.")
+        if hasattr(self.llm, "chat_stream"):
+            # REAL token streaming: ANSWER_CHUNK events as tokens sample
+            # (reference's chatStream fakes this by re-chunking a finished
+            # response, llm.ts:152-203; this engine streams live)
+            chunks: list[str] = []
+            for chunk in self.llm.chat_stream(final_system, final_prompt):
+                chunks.append(chunk)
+                yield AgentEvent(EventType.ANSWER_CHUNK, {"text": chunk})
+            answer = "".join(chunks)
+        else:
+            resp = self.llm.chat(final_system, final_prompt)
+            answer = resp.content
+        answer = answer or self.last_answer or "Investigation produced no conclusive answer."
         appendices = []
         tree = self.hypothesis_engine.to_markdown()
         if tree:
@@ -275,7 +285,9 @@ class Agent:
             self.conversation_memory.add_message("user", query)
             self.conversation_memory.add_message("assistant", answer[:800])
             self.conversation_memory.add_investigation(query, answer, memory.discovered_services)
-        yield AgentEvent(EventType.ANSWER_FINAL, {"text": answer})
+        streamed_len = len("".join(chunks)) if hasattr(self.llm, "chat_stream") else 0
+        yield AgentEvent(EventType.ANSWER_FINAL, {"text": answer,
+                                                  "streamedLen": streamed_len})
         yield AgentEvent(EventType.DONE, {"iterations": iteration,
                                           "cacheStats": self.cache.stats()})
         set_active_scratchpad(None)

@@ -106,8 +106,15 @@ def _render_agent_event(event: Any) -> None:
         _echo(f"{DIM}📚 retrieved {event.data.get('count')} knowledge docs{RESET}")
     elif t == EventType.CONTEXT_CLEARED:
         _echo(f"{DIM}🧹 compacted context ({event.data.get('cleared')} results cleared){RESET}")
+    elif t == EventType.ANSWER_CHUNK:
+        # live token stream: print without newline as the engine samples
+        print(event.data.get("text", ""), end="", flush=True)
     elif t == EventType.ANSWER_FINAL:
-        _echo("\n" + event.data.get("text", ""))
+        text = event.data.get("text", "")
+        streamed = event.data.get("streamedLen", 0)
+        # when the body already streamed, print only the appendices
+        _echo(("\n" if streamed else "\n" + "") + text[streamed:]
+              if streamed else "\n" + text)
     elif t == EventType.DONE:
         _echo(f"\n{DIM}done in {event.data.get('iterations')} iterations{RESET}")
 

@@ -146,3 +146,29 @@ class TestFreeFormLoop:
         list(agent.run("first question"))
         assert len(mem.messages) == 2
         assert mem.investigations[0].query == "first question"
+
+
+def test_final_answer_streams_chunks(make_agent=None):
+    """ANSWER_CHUNK events precede ANSWER_FINAL when the client supports
+    chat_stream; the final text equals the streamed body (+ appendices)."""
+    from runbookai_amd.agent.agent import Agent
+    from runbookai_amd.agent.types import AgentConfig, ChatResponse, EventType
+
+    class StreamingMock:
+        def chat(self, system, user, tools=None):
+            return ChatResponse(content="no tools needed")
+
+        def chat_stream(self, system, user, tools=None):
+            yield "Root cause: "
+            yield "redis pool exhausted."
+
+    agent = Agent(llm=StreamingMock(), tools=[], knowledge_retriever=None,
+                  config=AgentConfig(max_iterations=1))
+    events = list(agent.run("why is checkout slow?"))
+    chunks = [e for e in events if e.type == EventType.ANSWER_CHUNK]
+    final = [e for e in events if e.type == EventType.ANSWER_FINAL][0]
+    assert [c.data["text"] for c in chunks] == ["Root cause: ",
+                                                "redis pool exhausted."]
+    body = "".join(c.data["text"] for c in chunks)
+    assert final.data["text"].startswith(body)
+    assert final.data["streamedLen"] == len(body)
