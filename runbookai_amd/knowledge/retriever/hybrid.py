@@ -87,6 +87,33 @@ class HybridRetriever:
         fused = reciprocal_rank_fusion([(FTS_WEIGHT, fts_hits), (VECTOR_WEIGHT, vec_hits)])
         return fused[:limit]
 
+    def search_many(
+        self,
+        queries: list[str],
+        limit: int = 5,
+        doc_type: Optional[str] = None,
+        service: Optional[str] = None,
+    ) -> list[list[SearchHit]]:
+        """Batched hybrid search: one encoder forward covers every query's
+        vector leg (engine-side: concurrent investigations batch their
+        knowledge lookups instead of paying a per-query encode)."""
+        mode = self._effective_mode()
+        if mode == "fts" or self.vector_store is None:
+            return [self.store.search(q, limit=limit, doc_type=doc_type,
+                                      service=service) for q in queries]
+        vec_lists = self.vector_store.search_many(
+            queries, limit=limit * 2, min_score=0.0,
+            service=service, doc_type=doc_type)
+        if mode == "vector":
+            return [v[:limit] for v in vec_lists]
+        out = []
+        for q, vec_hits in zip(queries, vec_lists):
+            fts_hits = self.store.search(q, limit * 2, doc_type, service)
+            fused = reciprocal_rank_fusion(
+                [(FTS_WEIGHT, fts_hits), (VECTOR_WEIGHT, vec_hits)])
+            out.append(fused[:limit])
+        return out
+
     # -- typed views (reference L156-226) --------------------------------------
 
     def search_by_type(self, query: str, doc_type: str, limit: int = 5) -> list[SearchHit]:

@@ -152,6 +152,63 @@ class VectorStore:
                 break
         return hits
 
+    def search_many(
+        self,
+        queries: list[str],
+        limit: int = 5,
+        min_score: float = MIN_SCORE,
+        service: Optional[str] = None,
+        doc_type: Optional[str] = None,
+    ) -> list[list[SearchHit]]:
+        """Batched search: ONE encoder forward for every query (the
+        round-1 knowledge bench showed per-query encoding at batch 1 is
+        the e2e wall: 303 QPS vs 12k search-only QPS at 100k docs), then
+        one batched matmul top-k against the resident corpus."""
+        if self.embedder is None or self._matrix is None or not self._ids:
+            return [[] for _ in queries]
+        qvecs = self.embedder.embed_texts(list(queries))
+        fetch = limit * 4 if (service or doc_type) else limit
+        all_pairs = self._topk_batch(np.asarray(qvecs, dtype=np.float32), fetch)
+        out: list[list[SearchHit]] = []
+        for pairs in all_pairs:
+            hits: list[SearchHit] = []
+            for idx, score in pairs:
+                if score < min_score:
+                    continue
+                cid = self._ids[idx]
+                meta = self._meta[cid]
+                if service and service not in meta["services"]:
+                    continue
+                if doc_type and meta["type"] != doc_type:
+                    continue
+                hits.append(SearchHit(
+                    doc_id=meta["docId"], chunk_id=cid, title=meta["title"],
+                    content=meta["content"], doc_type=meta["type"], score=score,
+                    services=meta["services"]))
+                if len(hits) >= limit:
+                    break
+            out.append(hits)
+        return out
+
+    def _topk_batch(self, qvecs: np.ndarray, k: int) -> list[list[tuple[int, float]]]:
+        """[B, D] queries -> per-query (idx, score) pairs via one batched
+        matmul + topk on the resident corpus."""
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                matrix = self._ensure_gpu_matrix()
+                q = torch.from_numpy(qvecs)
+                q = (q / q.norm(dim=1, keepdim=True).clamp_min(1e-12)).half()
+                scores = q.to(matrix.device) @ matrix.t()       # [B, N]
+                kk = min(k, matrix.shape[0])
+                vals, idx = torch.topk(scores.float(), kk, dim=1)
+                vals_l, idx_l = vals.cpu().tolist(), idx.cpu().tolist()
+                return [list(zip(i, v)) for i, v in zip(idx_l, vals_l)]
+        except ImportError:
+            pass
+        return [self._topk(qvecs[b], k) for b in range(qvecs.shape[0])]
+
     def _topk(self, qvec: np.ndarray, k: int) -> list[tuple[int, float]]:
         """Brute-force cosine top-k. GPU: fused HIP kernel over the fp16
         corpus matrix; CPU: numpy reference."""

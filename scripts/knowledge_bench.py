@@ -88,6 +88,22 @@ def main() -> None:
     torch.cuda.synchronize()
     report["e2e_search_qps_n100k"] = round(iters / (time.time() - t0), 1)
 
+    # batched e2e: one encoder forward per BATCH of queries (the serving
+    # shape — concurrent investigations batch their lookups)
+    for B in (16, 64):
+        batch = [queries[i % len(queries)] for i in range(B)]
+        torch.cuda.synchronize()
+        t0 = time.time()
+        reps = max(1, 512 // B)
+        for _ in range(reps):
+            qv = enc.encode(batch, batch_size=B)
+            qt = torch.from_numpy(qv).half().to("cuda:0")
+            scores = qt @ matrix.t()
+            torch.topk(scores.float(), 5, dim=1)
+        torch.cuda.synchronize()
+        report[f"e2e_search_qps_n100k_batch{B}"] = round(
+            reps * B / (time.time() - t0), 1)
+
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/knowledge_bench.json", "w") as f:
         json.dump(report, f, indent=1)

@@ -277,3 +277,52 @@ class TestFtsAdversarialQueries:
             assert isinstance(hits, list), q
         # and the injection attempt did not damage the table
         assert store.search("redis connection pool")
+
+
+def test_vector_search_many_matches_per_query(tmp_path):
+    """Batched search_many == N x search on the same store (CPU path)."""
+    import numpy as np
+
+    from runbookai_amd.knowledge.store.vector_store import VectorStore
+
+    class FakeEmbedder:
+        dim = 8
+
+        def _vec(self, text):
+            rng = np.random.default_rng(abs(hash(text)) % (2**31))
+            v = rng.standard_normal(8).astype(np.float32)
+            return v / np.linalg.norm(v)
+
+        def embed_text(self, text):
+            return self._vec(text)
+
+        def embed_texts(self, texts):
+            return np.stack([self._vec(t) for t in texts])
+
+        def embed_chunk(self, chunk):
+            return self._vec(chunk["content"])
+
+    store = VectorStore(str(tmp_path / "vec.db"), embedder=FakeEmbedder())
+    chunks = [{"chunkId": f"c{i}", "docId": f"d{i}", "title": f"t{i}",
+               "section": "", "content": f"content about topic {i % 5}",
+               "type": "runbook", "services": []} for i in range(40)]
+    store.add_chunks(chunks)
+    queries = ["content about topic 1", "content about topic 3",
+               "totally unrelated query"]
+    batched = store.search_many(queries, limit=4, min_score=0.0)
+    singles = [store.search(q, limit=4, min_score=0.0) for q in queries]
+    assert len(batched) == 3
+    for b, s in zip(batched, singles):
+        assert [h.chunk_id for h in b] == [h.chunk_id for h in s]
+
+
+def test_hybrid_search_many_fuses_like_single(tmp_path):
+    from runbookai_amd.knowledge.retriever.hybrid import HybridRetriever
+
+    # FTS-only degrade path: batched == per-query
+    class FtsStore:
+        def search(self, query, limit=5, doc_type=None, service=None):
+            return []
+
+    r = HybridRetriever(store=FtsStore(), vector_store=None)
+    assert r.search_many(["a", "b"], limit=3) == [[], []]
