@@ -499,3 +499,55 @@ class TestConversationRecall:
         m = self._mem()
         ctx = m.get_context_for_prompt(query="checkout latency again")
         assert "Related earlier context" in ctx
+
+
+class TestServiceContextDepth:
+    """Blast-radius structure + per-service context bundles (reference
+    service-context.ts:120-205, 262-302)."""
+
+    def _graph(self):
+        from runbookai_amd.knowledge.store.graph_store import ServiceGraph
+        g = ServiceGraph()
+        g.add_node("payments", tier="critical", owner="payments-team",
+                   type="ecs", oncall="pay-oncall", slack="#pay")
+        g.add_node("checkout", type="ecs", tier="standard")
+        g.add_node("cart")
+        g.add_node("redis")
+        # checkout -> redis (critical), cart -> checkout, payments -> checkout
+        g.add_dependency("checkout", "redis", critical=True, criticality="critical")
+        g.add_dependency("cart", "checkout")
+        g.add_dependency("payments", "checkout")
+        return g
+
+    def test_blast_radius_structure(self):
+        from runbookai_amd.agent.service_context import ServiceContextManager
+        m = ServiceContextManager(self._graph())
+        br = m.blast_radius_info("redis")
+        assert "checkout" in br["direct"]
+        assert set(br["transitive"]) >= {"cart", "payments"} - set(br["direct"])
+        assert "payments" in br["criticalAffected"]   # critical tier reached
+        assert br["totalAffected"] == 3
+        assert any("payments" in p for p in br["criticalPaths"])
+
+    def test_service_context_bundle(self):
+        from runbookai_amd.agent.service_context import ServiceContextManager
+
+        class R:
+            def search(self, q, limit=1):
+                class H: title = f"runbook for {q.split()[0]}"
+                return [H()]
+
+        m = ServiceContextManager(self._graph(), retriever=R())
+        ctx = m.service_context("checkout")
+        assert ctx["criticalDependencies"] == ["redis"]
+        assert ctx["escalation"]["service"] == "checkout"
+        assert any("runbook" in r for r in ctx["runbooks"])
+        both = m.contexts_for_services(["checkout", "ghost-svc"])
+        assert set(both) == {"checkout"}
+
+    def test_prompt_section_renders_depth(self):
+        from runbookai_amd.agent.service_context import ServiceContextManager
+        m = ServiceContextManager(self._graph())
+        text = m.prompt_section(["redis", "unknown-svc"])
+        assert "blast radius: 3 services" in text
+        assert "CRITICAL tier affected" in text and "payments" in text
