@@ -79,14 +79,56 @@ class KnowledgeContextManager:
     # -- symptom -> known-issue matcher (reference L590-606) -----------------
 
     def match_known_issues(self, symptoms: list[str]) -> list[dict[str, Any]]:
+        """Bidirectional containment like the reference matcher PLUS
+        per-symptom frontmatter matching: an issue whose declared symptoms
+        contain (or are contained by) a reported symptom matches even when
+        the title/body phrase differs."""
         matches = []
+        sym_lower = [s.lower() for s in symptoms if s]
         for issue in self.known_issues:
+            score = 0
+            declared = issue.get("symptoms") or []
+            for isym in declared:
+                il = str(isym).lower()
+                for s in sym_lower:
+                    if il in s or s in il:
+                        score += 2   # declared-symptom match outranks body text
             text = f"{issue.get('title', '')} {issue.get('content', '')}".lower()
-            score = sum(1 for s in symptoms if s.lower() in text)
+            for s in sym_lower:
+                if s in text:
+                    score += 1
             if score > 0:
                 matches.append({**issue, "_match_score": score})
         matches.sort(key=lambda m: m["_match_score"], reverse=True)
         return matches[:3]
+
+    # -- runbook/service coverage views (reference L556-578) -----------------
+
+    def has_runbook_for_service(self, service: str) -> bool:
+        sl = service.lower()
+        for rb in self.runbook_index:
+            for svc in rb.get("services") or []:
+                if sl in str(svc).lower():
+                    return True
+            if sl in str(rb.get("title", "")).lower():
+                return True
+        return False
+
+    def unqueried_services_with_runbooks(self) -> list[str]:
+        """Services the index has runbooks for but the investigation has
+        not queried yet — the agent can proactively pull these."""
+        covered: set[str] = set()
+        for rb in self.runbook_index:
+            for svc in rb.get("services") or []:
+                covered.add(str(svc))
+        return sorted(covered - self._queried_services)
+
+    def reset(self) -> None:
+        """New investigation: keep the pre-built index, drop per-run state
+        (reference reset(), knowledge-context.ts:606-612)."""
+        self._queried_services.clear()
+        self._queried_symptoms.clear()
+        self.jit_results.clear()
 
     # -- prompt sections (reference L452-557) --------------------------------
 

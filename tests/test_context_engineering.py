@@ -551,3 +551,47 @@ class TestServiceContextDepth:
         text = m.prompt_section(["redis", "unknown-svc"])
         assert "blast radius: 3 services" in text
         assert "CRITICAL tier affected" in text and "payments" in text
+
+
+class TestKnowledgeContextDepth:
+    """Symptom matcher + coverage views (reference
+    knowledge-context.ts:556-612)."""
+
+    def _mgr(self):
+        from runbookai_amd.agent.knowledge_context import KnowledgeContextManager
+        m = KnowledgeContextManager(retriever=None)
+        m.runbook_index = [
+            {"title": "Redis pool exhaustion", "services": ["checkout-api", "redis"]},
+            {"title": "Kafka disk pressure", "services": ["kafka-broker"]},
+        ]
+        m.known_issues = [
+            {"title": "Stale DNS entries", "content": "SERVFAIL responses seen",
+             "symptoms": ["dns resolution failures"]},
+            {"title": "Pool leak v2.1", "content": "connections never returned",
+             "symptoms": ["timeouts acquiring connections", "latency spike"]},
+        ]
+        return m
+
+    def test_declared_symptom_match_outranks_body(self):
+        m = self._mgr()
+        hits = m.match_known_issues(["latency spike on checkout"])
+        assert hits and hits[0]["title"] == "Pool leak v2.1"
+        # bidirectional: reported symptom contained in declared symptom
+        hits2 = m.match_known_issues(["dns resolution"])
+        assert hits2 and hits2[0]["title"] == "Stale DNS entries"
+
+    def test_runbook_coverage_views(self):
+        m = self._mgr()
+        assert m.has_runbook_for_service("redis")
+        assert not m.has_runbook_for_service("ghost-svc")
+        unq = m.unqueried_services_with_runbooks()
+        assert "kafka-broker" in unq
+        m._queried_services.add("kafka-broker")
+        assert "kafka-broker" not in m.unqueried_services_with_runbooks()
+
+    def test_reset_keeps_index(self):
+        m = self._mgr()
+        m.jit_results.append({"title": "x"})
+        m._queried_services.add("a")
+        m.reset()
+        assert m.runbook_index and not m.jit_results and not m._queried_services
