@@ -228,6 +228,18 @@ def gemv(x: torch.Tensor, w: torch.Tensor, pre: int = 0,
     replaces rmsnorm/silu + GEMM + residual-add launches on the decode
     path (ops/csrc/decode_gemv.hip)."""
     if _on_gpu(x):
+        M, K = x.shape[0], w.shape[1]
+        mt = 2 if M <= 2 else 4
+        # the kernel stages MT full x rows in LDS; past the 160 KiB/CU cap
+        # (70B down-proj K=28672 at M 3-4 = 229 KiB) split the batch in
+        # half — two mt=2 launches fit (116 KiB each)
+        if mt * K * 2 + 8192 > 160 * 1024 and M > 2:
+            half = M // 2
+            return torch.cat([
+                gemv(x[:half], w, pre, norm_w,
+                     res[:half] if res is not None else None, eps),
+                gemv(x[half:], w, pre, norm_w,
+                     res[half:] if res is not None else None, eps)], dim=0)
         return _get_ext().decode_gemv(x.contiguous(), w, norm_w, res, pre, eps)
     if pre == 1:
         x = reference.rmsnorm(x, norm_w, eps)
@@ -251,7 +263,7 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     if (USE_DECODE_GEMV and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 4
             and weight.shape[0] <= 32768 and weight.shape[1] % 8 == 0):
         # N cap: hipBLASLt wins the vocab-sized lm_head (182 vs 223 us at M=1)
-        return _get_ext().decode_gemv(x.contiguous(), weight, None, None, 0, 1e-5)
+        return gemv(x, weight)
     if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 16
             and weight.shape[0] <= 8192 and weight.shape[1] <= 8192
             and weight.shape[0] % 64 == 0 and weight.shape[1] % 64 == 0):

@@ -623,3 +623,18 @@ class TestDecodeGemv:
         out_generic = m._decode_impl(ids, pos, bt.to(DEV), lens.to(DEV), slots)
         diff = (out_fused.float() - out_generic.float()).abs().max().item()
         assert diff < 0.5, diff
+
+
+class TestDecodeGemvLdsCap:
+    def test_70b_down_shape_m4(self):
+        """M=4 x K=28672 would want 229 KiB of LDS (> the 160 KiB/CU cap);
+        the wrapper splits the batch — numerics must match the reference."""
+        M, N, K = 4, 1024, 28672
+        gu = bf(torch.randn(M, 2 * K) * 0.3).to(DEV)
+        w = bf(torch.randn(N, K) * 0.05).to(DEV)
+        res = bf(torch.randn(M, N)).to(DEV)
+        out = ops.gemv(gu, w, pre=2, res=res)
+        g, u = gu.cpu().chunk(2, dim=-1)
+        act = ref.silu_mul(g.contiguous(), u.contiguous())
+        expected = act.float() @ w.cpu().float().t() + res.cpu().float()
+        assert (out.cpu().float() - expected).abs().max().item() < 2.0
