@@ -40,106 +40,116 @@ __global__ void decode_gemv_kernel(
     const ushort_t* __restrict__ nw,   // [K] norm weight (PRE 1)
     const ushort_t* __restrict__ res,  // [M, N] residual (RES)
     ushort_t* __restrict__ out,        // [M, N]
-    int M, int N, long K, float eps) {
+    int M, int N, long K, int KC, float eps) {
     constexpr int ROWS = 256 / KS;     // output rows per block
     extern __shared__ __attribute__((aligned(16))) char smem[];
     ushort_t* x_lds = reinterpret_cast<ushort_t*>(smem);
     const int tid = threadIdx.x;
+    float* wred = reinterpret_cast<float*>(smem + (size_t)MT * KC * 2);
 
-    // ---- stage pre(x) rows into LDS (16-B pieces across the block) ----
-    const int pieces = (int)(K / 8);
-#pragma unroll 1
-    for (int i = tid; i < MT * pieces; i += 256) {
-        const int m = i / pieces;
-        const int kk = (i % pieces) * 8;
-        bf16x8_t v{0, 0, 0, 0, 0, 0, 0, 0};
-        if (m < M) {
-            if (PRE == 2) {
-                const bf16x8_t g = *reinterpret_cast<const bf16x8_t*>(
-                    x + (long)m * 2 * K + kk);
-                const bf16x8_t u = *reinterpret_cast<const bf16x8_t*>(
-                    x + (long)m * 2 * K + K + kk);
-#pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    v[j] = (short)f2bf(silu_f(bf2f((ushort_t)g[j]))
-                                       * bf2f((ushort_t)u[j]));
-            } else if (PRE == 1) {
-                const bf16x8_t h = *reinterpret_cast<const bf16x8_t*>(
-                    x + (long)m * K + kk);
-                const bf16x8_t wn = *reinterpret_cast<const bf16x8_t*>(nw + kk);
-#pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    v[j] = (short)f2bf(bf2f((ushort_t)h[j]) * bf2f((ushort_t)wn[j]));
-            } else {
-                v = *reinterpret_cast<const bf16x8_t*>(x + (long)m * K + kk);
-            }
-        }
-        *reinterpret_cast<bf16x8_t*>(&x_lds[(long)m * K + kk]) = v;
-    }
-
-    float rs[MT];
-#pragma unroll
-    for (int m = 0; m < MT; ++m) rs[m] = 1.f;
-    float* wred = reinterpret_cast<float*>(smem + (size_t)MT * K * 2);
-    __syncthreads();   // staged x visible
-
-    // ---- stream W: row = tid % ROWS, k-segment = tid / ROWS ----
     const int rlocal = tid % ROWS;
     const int kseg = tid / ROWS;
     const long nrow = (long)blockIdx.x * ROWS + rlocal;
     const bool nvalid = nrow < N;
-    const long seg_len = ((K / 8 + KS - 1) / KS) * 8;
-    const long k0 = min(K, (long)kseg * seg_len);
-    const int klen = (int)(min(K, k0 + seg_len) - k0);
-    const ushort_t* wp = w + nrow * K + k0;
+    float rs[MT];
+#pragma unroll
+    for (int m = 0; m < MT; ++m) rs[m] = 1.f;
     float acc[MT];
 #pragma unroll
     for (int m = 0; m < MT; ++m) acc[m] = 0.f;
 
-    // register double-buffer: iteration c's dots run with iteration c+1's
-    // loads already in flight (without it the stream stalls a full memory
-    // latency between every 128-B group — measured ~2x off the stream roof)
-    const int full = nvalid ? klen / (GV_UNROLL * 8) : 0;
-    bf16x8_t wrA[GV_UNROLL], wrB[GV_UNROLL];
-    auto wload = [&](bf16x8_t (&wr)[GV_UNROLL], int c) {
+    // ---- chunked over K: stage MT x-rows for [kc0, kc0+KC) into LDS,
+    // stream each thread's W segment of the chunk, repeat. Bounding the
+    // staged window at KC keeps LDS <= 68 KiB at ANY K (full-row staging
+    // at 70B's K=28672 was 229 KiB at MT=4 — over the 160 KiB/CU cap —
+    // and 1 block/CU even where it fit) ----
+    for (long kc0 = 0; kc0 < K; kc0 += KC) {
+        const int clen = (int)min((long)KC, K - kc0);
+        const int cpieces = clen / 8;
+#pragma unroll 1
+        for (int i = tid; i < MT * cpieces; i += 256) {
+            const int m = i / cpieces;
+            const long kk = kc0 + (i % cpieces) * 8;
+            bf16x8_t v{0, 0, 0, 0, 0, 0, 0, 0};
+            if (m < M) {
+                if (PRE == 2) {
+                    const bf16x8_t g = *reinterpret_cast<const bf16x8_t*>(
+                        x + (long)m * 2 * K + kk);
+                    const bf16x8_t u = *reinterpret_cast<const bf16x8_t*>(
+                        x + (long)m * 2 * K + K + kk);
 #pragma unroll
-        for (int u = 0; u < GV_UNROLL; ++u)
-            wr[u] = *reinterpret_cast<const bf16x8_t*>(
-                wp + (long)c * GV_UNROLL * 8 + u * 8);
-    };
-    auto wdot = [&](bf16x8_t (&wr)[GV_UNROLL], int c) {
+                    for (int j = 0; j < 8; ++j)
+                        v[j] = (short)f2bf(silu_f(bf2f((ushort_t)g[j]))
+                                           * bf2f((ushort_t)u[j]));
+                } else if (PRE == 1) {
+                    const bf16x8_t h = *reinterpret_cast<const bf16x8_t*>(
+                        x + (long)m * K + kk);
+                    const bf16x8_t wn = *reinterpret_cast<const bf16x8_t*>(nw + kk);
 #pragma unroll
-        for (int u = 0; u < GV_UNROLL; ++u) {
-            const long kk = k0 + c * GV_UNROLL * 8 + u * 8;
+                    for (int j = 0; j < 8; ++j)
+                        v[j] = (short)f2bf(bf2f((ushort_t)h[j])
+                                           * bf2f((ushort_t)wn[j]));
+                } else {
+                    v = *reinterpret_cast<const bf16x8_t*>(x + (long)m * K + kk);
+                }
+            }
+            *reinterpret_cast<bf16x8_t*>(
+                &x_lds[(long)m * KC + (kk - kc0)]) = v;
+        }
+        __syncthreads();   // chunk staged
+
+        // this thread's segment WITHIN the chunk
+        const long seg_len = ((clen / 8 + KS - 1) / KS) * 8;
+        const long k0 = min((long)clen, (long)kseg * seg_len);
+        const int klen = (int)(min((long)clen, k0 + seg_len) - k0);
+        const ushort_t* wp = w + nrow * K + kc0 + k0;
+
+        // register double-buffer: iteration c's dots run with iteration
+        // c+1's loads already in flight (without it the stream stalls a
+        // full memory latency between every 128-B group)
+        const int full = nvalid ? klen / (GV_UNROLL * 8) : 0;
+        bf16x8_t wrA[GV_UNROLL], wrB[GV_UNROLL];
+        auto wload = [&](bf16x8_t (&wr)[GV_UNROLL], int c) {
 #pragma unroll
-            for (int m = 0; m < MT; ++m) {
-                const bf16x8_t xc = *reinterpret_cast<const bf16x8_t*>(
-                    &x_lds[(long)m * K + kk]);
+            for (int u = 0; u < GV_UNROLL; ++u)
+                wr[u] = *reinterpret_cast<const bf16x8_t*>(
+                    wp + (long)c * GV_UNROLL * 8 + u * 8);
+        };
+        auto wdot = [&](bf16x8_t (&wr)[GV_UNROLL], int c) {
 #pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    acc[m] += bf2f((ushort_t)wr[u][j]) * bf2f((ushort_t)xc[j]);
+            for (int u = 0; u < GV_UNROLL; ++u) {
+                const long kk = k0 + c * GV_UNROLL * 8 + u * 8;
+#pragma unroll
+                for (int m = 0; m < MT; ++m) {
+                    const bf16x8_t xc = *reinterpret_cast<const bf16x8_t*>(
+                        &x_lds[(long)m * KC + kk]);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        acc[m] += bf2f((ushort_t)wr[u][j]) * bf2f((ushort_t)xc[j]);
+                }
+            }
+        };
+        if (full > 0) wload(wrA, 0);
+        for (int c = 0; c < full; c += 2) {
+            if (c + 1 < full) wload(wrB, c + 1);
+            wdot(wrA, c);
+            if (c + 2 < full) wload(wrA, c + 2);
+            if (c + 1 < full) wdot(wrB, c + 1);
+        }
+        if (nvalid) {   // ragged tail, 8-element pieces
+            for (int kk = full * GV_UNROLL * 8; kk < klen; kk += 8) {
+                const bf16x8_t wr = *reinterpret_cast<const bf16x8_t*>(wp + kk);
+#pragma unroll
+                for (int m = 0; m < MT; ++m) {
+                    const bf16x8_t xc = *reinterpret_cast<const bf16x8_t*>(
+                        &x_lds[(long)m * KC + k0 + kk]);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j)
+                        acc[m] += bf2f((ushort_t)wr[j]) * bf2f((ushort_t)xc[j]);
+                }
             }
         }
-    };
-    if (full > 0) wload(wrA, 0);
-    for (int c = 0; c < full; c += 2) {
-        if (c + 1 < full) wload(wrB, c + 1);
-        wdot(wrA, c);
-        if (c + 2 < full) wload(wrA, c + 2);
-        if (c + 1 < full) wdot(wrB, c + 1);
-    }
-    if (nvalid) {   // ragged tail, 8-element pieces
-        for (int kk = full * GV_UNROLL * 8; kk < klen; kk += 8) {
-            const bf16x8_t wr = *reinterpret_cast<const bf16x8_t*>(wp + kk);
-#pragma unroll
-            for (int m = 0; m < MT; ++m) {
-                const bf16x8_t xc = *reinterpret_cast<const bf16x8_t*>(
-                    &x_lds[(long)m * K + k0 + kk]);
-#pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    acc[m] += bf2f((ushort_t)wr[j]) * bf2f((ushort_t)xc[j]);
-            }
-        }
+        __syncthreads();   // chunk consumed before the next overwrites
     }
 
     // ---- PRE 1: row 1/rms, computed AFTER the W stream (the raw-row
@@ -219,24 +229,25 @@ extern "C" void launch_decode_gemv(const void* x, const void* w, const void* nw,
                                    const void* res, void* out, int M, int N,
                                    long K, int pre, int with_res, float eps,
                                    hipStream_t stream) {
+    const int KC = (int)((K < 8192) ? K : 8192);   // staged k-window
     // pick the in-block k-split so the grid reaches >= 224 blocks (the
     // decode projections are per-CU-stream-rate bound: blocks ~ CUs)
     int ks = 1;
-    while (ks < 16 && (long)(N + (256 / (ks * 2)) - 1) / (256 / (ks * 2)) <= 288
+    while (ks < 16 && (long)(N + (256 / (ks * 2)) - 1) / (256 / (ks * 2)) <= 520
            && K / (ks * 2) >= 64 && (N + 255) / 256 * ks < 224)
         ks *= 2;
     const int rows = 256 / ks;
     const int nblk = (N + rows - 1) / rows;
     const int mt = (M <= 2) ? 2 : 4;
-    // LDS: MT full x rows + reduce scratch max(KS*ROWS*MT, 4*MT) floats
-    const size_t lds = (size_t)mt * K * 2
+    // LDS: MT staged x-chunk rows + reduce scratch (<= 68 KiB at KC=8192)
+    const size_t lds = (size_t)mt * KC * 2
                        + (size_t)256 * mt * sizeof(float);
     dim3 grid(nblk), block(256);
 #define GV_L(MT_, KS_, PRE_, RES_)                                                     \
     hipLaunchKernelGGL((decode_gemv_kernel<MT_, KS_, PRE_, RES_>), grid, block, lds,   \
                        stream, (const ushort_t*)x, (const ushort_t*)w,                 \
                        (const ushort_t*)nw, (const ushort_t*)res,                      \
-                       (ushort_t*)out, M, N, K, eps)
+                       (ushort_t*)out, M, N, K, KC, eps)
 #define GV_KS(MT_, PRE_, RES_)                                   \
     do {                                                         \
         if (ks == 1) GV_L(MT_, 1, PRE_, RES_);                   \
