@@ -117,6 +117,50 @@ class GrammarTokenMasker:
         self._walk(self.root, fsm, out)
         return out
 
+    def mask_row(self, fsm: JsonFsm, vocab_size: int):
+        """Boolean mask row over the vocab for this FSM state. For the
+        dominant string-content states the quote-free-token base row is a
+        PRECOMPUTED tensor per remaining-capacity bucket (the Python list
+        `mask[i, allowed] = True` with thousands of safe ids per request
+        per step was ~14 ms/step at c8 in the checkpoint bench); only the
+        small quote-bearing subtrie still walks per step."""
+        import torch
+
+        cache = getattr(self, "_safe_rows", None)
+        if cache is None or self._safe_rows_v != vocab_size:
+            cache = {}
+            for cap in range(0, self.max_safe_len + 1):
+                row = torch.zeros(vocab_size, dtype=torch.bool)
+                if self.safe_upto[cap]:
+                    row[self.safe_upto[cap]] = True
+                cache[cap] = row
+            self._safe_rows = cache
+            self._safe_rows_v = vocab_size
+        if fsm.done:
+            row = torch.zeros(vocab_size, dtype=torch.bool)
+            if self.eot_id is not None:
+                row[self.eot_id] = True
+            return row
+        st = fsm.string_state()
+        if st is not None:
+            _chars, cap = st
+            base = cache[min(max(cap, 0), self.max_safe_len)]
+            extra: list[int] = []
+            self._walk(self.string_root, fsm, extra)
+            if not extra:
+                return base
+            row = base.clone()
+            row[extra] = True
+            return row
+        ids: list[int] = []
+        self._walk(self.root, fsm, ids)
+        row = torch.zeros(vocab_size, dtype=torch.bool)
+        if ids:
+            row[ids] = True
+        elif self.eot_id is not None:
+            row[self.eot_id] = True
+        return row
+
     def _walk(self, node: _TrieNode, fsm: JsonFsm, out: list[int]) -> None:
         out.extend(node.ends)
         if fsm.done:

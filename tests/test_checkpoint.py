@@ -571,3 +571,48 @@ class TestBpeTokenizer:
         assert tok.special_tokens["<|eot_id|>"] in chat
         assert tok.eot_id == tok.special_tokens["<|eot_id|>"]
         assert tok._pat.pattern == LLAMA3_PATTERN
+
+
+def test_mask_row_matches_allowed_tokens(tmp_path):
+    """mask_row (cached-row fast path) must equal the set semantics of
+    allowed_tokens at every step of a grammar-constrained walk."""
+    import random
+
+    import tokenizers
+    import torch
+
+    from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer
+    from runbookai_amd.engine.grammar_bpe import build_masker
+    from runbookai_amd.engine.json_fsm import JsonFsm
+
+    corpus = ["redis connection pool exhausted", "error rate spiked 40%",
+              'json {"summary": "x", "severity": "high"}'] * 30
+    tok = tokenizers.ByteLevelBPETokenizer()
+    tok.train_from_iterator(corpus, vocab_size=400, min_frequency=1,
+                            special_tokens=["<|eot_id|>"])
+    path = str(tmp_path / "tokenizer.json")
+    tok.save(path)
+    masker = build_masker(BpeTokenizer.from_file(path))
+    schema = {"type": "object", "properties": {
+        "summary": {"type": "string", "maxLength": 40},
+        "severity": {"enum": ["low", "high"]},
+        "count": {"type": "integer", "minimum": 1, "maximum": 9},
+    }, "required": ["summary", "severity", "count"]}
+    rng = random.Random(7)
+    V = 512
+    fsm = JsonFsm(schema)
+    for _ in range(60):
+        if fsm.done:
+            break
+        allowed = masker.allowed_tokens(fsm)
+        row = masker.mask_row(fsm, V)
+        expect = torch.zeros(V, dtype=torch.bool)
+        if allowed:
+            expect[allowed] = True
+        elif masker.eot_id is not None:
+            expect[masker.eot_id] = True
+        assert torch.equal(row, expect)
+        if not allowed:
+            break
+        tid = rng.choice(allowed)
+        masker.advance_token(fsm, masker.token_bytes[tid])
