@@ -183,8 +183,11 @@ def run_training(args):
     return args.out
 
 
-def eval_checkpoint(ckpt: str, concurrency: int = 4) -> dict:
-    """Held-out eval: the REAL fixture cases through the served checkpoint."""
+def eval_checkpoint(ckpt: str, concurrency: int = 4,
+                    include_generated: bool = False) -> dict:
+    """Held-out eval: the REAL fixture cases through the served checkpoint.
+    include_generated adds the converter-generated suites (rcaeval /
+    rootly / tracerca — reported separately as out-of-distribution)."""
     from concurrent.futures import ThreadPoolExecutor
 
     from runbookai_amd.agent.orchestrator import InvestigationOrchestrator
@@ -209,12 +212,21 @@ def eval_checkpoint(ckpt: str, concurrency: int = 4) -> dict:
         os.path.abspath(__file__))), "examples", "evals")
     cases = []
     seen = set()
-    for fn in ("investigation-fixtures.sample.json",
-               "investigation-fixtures.extended.json"):
+    files = ["investigation-fixtures.sample.json",
+             "investigation-fixtures.extended.json"]
+    gen_files = ["rcaeval-fixtures.generated.json",
+                 "rootly-logs-fixtures.generated.json",
+                 "tracerca-fixtures.generated.json"]
+    gen_ids = set()
+    if include_generated:
+        files += gen_files
+    for fn in files:
         for c in load_fixtures(os.path.join(fix_dir, fn))["cases"]:
             if c["id"] not in seen:
                 seen.add(c["id"])
                 cases.append(c)
+                if fn in gen_files:
+                    gen_ids.add(c["id"])
 
     def run(case):
         scen = (_SCENARIOS[case["id"]]() if case["id"] in _SCENARIOS
@@ -236,15 +248,23 @@ def eval_checkpoint(ckpt: str, concurrency: int = 4) -> dict:
     with ThreadPoolExecutor(max_workers=concurrency) as pool:
         results = list(pool.map(run, cases))
     elapsed = time.time() - t0
+    core = [r for r in results if r["id"] not in gen_ids]
+    gen = [r for r in results if r["id"] in gen_ids]
     report = {
         "checkpoint": ckpt,
         "cases": results,
-        "pass_rate": sum(r["passed"] for r in results) / len(results),
-        "mean_score": round(sum(r["score"] for r in results) / len(results), 3),
+        "pass_rate": sum(r["passed"] for r in core) / max(1, len(core)),
+        "mean_score": round(sum(r["score"] for r in core) / max(1, len(core)), 3),
         "wall_s": round(elapsed, 1),
         "grammar_constrained": engine.supports_bpe_grammar,
         "device": dev,
     }
+    if gen:
+        report["generated_suites"] = {
+            "pass_rate": sum(r["passed"] for r in gen) / len(gen),
+            "mean_score": round(sum(r["score"] for r in gen) / len(gen), 3),
+            "n": len(gen),
+        }
     engine.shutdown()
     return report
 
@@ -262,12 +282,15 @@ def main():
     p.add_argument("--out", default="gpurun_out/policy_ckpt")
     p.add_argument("--eval", action="store_true")
     p.add_argument("--eval-only", action="store_true")
+    p.add_argument("--eval-generated", action="store_true",
+                   help="also evaluate the converter-generated suites")
     args = p.parse_args()
 
     if not args.eval_only:
         run_training(args)
     if args.eval or args.eval_only:
-        report = eval_checkpoint(args.out)
+        report = eval_checkpoint(args.out,
+                                 include_generated=args.eval_generated)
         print(json.dumps(report, indent=1))
         with open(os.path.join(args.out, "eval_report.json"), "w") as f:
             json.dump(report, f, indent=1)
