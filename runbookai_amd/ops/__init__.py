@@ -264,10 +264,10 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
             and weight.shape[0] <= 32768 and weight.shape[1] % 8 == 0):
         # N cap: hipBLASLt wins the vocab-sized lm_head (182 vs 223 us at M=1)
         return gemv(x, weight)
-    if (USE_SKINNY_GEMM and _on_gpu(x) and x.dim() == 2 and x.shape[0] <= 16
-            and weight.shape[0] <= 8192 and weight.shape[1] <= 8192
-            and weight.shape[0] % 64 == 0 and weight.shape[1] % 64 == 0):
-        return _get_ext().skinny_gemm(x.contiguous(), weight)
+    # M 5-16: hipBLASLt wins every decode projection once weights are COLD
+    # (cycled-weight probe: qkv M8 19.7 vs skinny 27.4 us; round-1's
+    # same-weight loops were L3-flattered — profiles/PERF_LOG.md). The MFMA
+    # skinny kernel stays available as ext.skinny_gemm (parity + tests).
     return x @ weight.t()
 
 
