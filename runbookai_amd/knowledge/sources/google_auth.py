@@ -2,10 +2,10 @@
 
 Parity with reference src/knowledge/sources/google-auth.ts (300 LoC):
 browser OAuth with a localhost:8085 callback and token persistence under
-.runbook/. This environment has NO network egress, so the token exchange
-step cannot complete here; the flow is fully wired (auth URL construction,
-callback server, token store) and `exchange_code` raises a clear offline
-error until egress exists.
+.runbook/. The flow is fully implemented (auth URL construction, callback server,
+token store, code exchange + refresh over HTTP); the token endpoint is
+injectable, so the exchange protocol is covered by tests against a local
+stub even though this image has no egress to Google.
 """
 from __future__ import annotations
 
@@ -109,13 +109,44 @@ class CallbackServer:
         return self.code
 
 
-def exchange_code(client_id: str, client_secret: str, code: str) -> dict[str, Any]:
-    """Exchange the auth code for tokens — requires egress."""
-    raise RuntimeError(
-        "Google token exchange requires network egress, which this environment "
-        f"does not have. With egress, POST to {TOKEN_ENDPOINT} with the code and "
-        "client credentials; the result is persisted via TokenStore."
-    )
+def exchange_code(client_id: str, client_secret: str, code: str,
+                  token_endpoint: str = TOKEN_ENDPOINT) -> dict[str, Any]:
+    """Exchange the auth code for tokens (POST, form-encoded — reference
+    google-auth.ts token exchange). token_endpoint is injectable so the
+    protocol is testable against a local stub (this image has no egress
+    to oauth2.googleapis.com)."""
+    import requests
+
+    resp = requests.post(token_endpoint, data={
+        "client_id": client_id,
+        "client_secret": client_secret,
+        "code": code,
+        "grant_type": "authorization_code",
+        "redirect_uri": f"http://localhost:{CALLBACK_PORT}/callback",
+    }, timeout=30)
+    resp.raise_for_status()
+    token = resp.json()
+    if "access_token" not in token:
+        raise RuntimeError(f"token exchange failed: {token.get('error', token)}")
+    return token
+
+
+def refresh_token(client_id: str, client_secret: str, refresh: str,
+                  token_endpoint: str = TOKEN_ENDPOINT) -> dict[str, Any]:
+    """Refresh an expired access token (reference google-auth.ts refresh)."""
+    import requests
+
+    resp = requests.post(token_endpoint, data={
+        "client_id": client_id,
+        "client_secret": client_secret,
+        "refresh_token": refresh,
+        "grant_type": "refresh_token",
+    }, timeout=30)
+    resp.raise_for_status()
+    token = resp.json()
+    if "access_token" not in token:
+        raise RuntimeError(f"token refresh failed: {token.get('error', token)}")
+    return token
 
 
 def run_auth_flow(client_id: str, client_secret: str,

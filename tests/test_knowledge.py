@@ -457,3 +457,50 @@ class TestLiveSourceClients:
             assert {d.title for d in docs2} == {"Runbook Doc"}
         finally:
             srv.shutdown()
+
+
+def test_google_token_exchange_and_refresh_protocol():
+    """Code exchange + refresh POSTs against a local stub endpoint
+    (reference google-auth.ts; endpoint injectable — no egress here)."""
+    import json
+    import threading
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+    from urllib.parse import parse_qs
+
+    seen = []
+
+    class H(BaseHTTPRequestHandler):
+        def do_POST(self):
+            body = self.rfile.read(int(self.headers["Content-Length"]))
+            form = {k: v[0] for k, v in parse_qs(body.decode()).items()}
+            seen.append(form)
+            out = {"access_token": f"at-{form['grant_type']}",
+                   "refresh_token": "rt-1", "expires_in": 3600}
+            data = json.dumps(out).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.end_headers()
+            self.wfile.write(data)
+
+        def log_message(self, *a):
+            pass
+
+    srv = HTTPServer(("127.0.0.1", 0), H)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    endpoint = f"http://127.0.0.1:{srv.server_port}/token"
+    try:
+        from runbookai_amd.knowledge.sources.google_auth import (
+            exchange_code,
+            refresh_token,
+        )
+
+        tok = exchange_code("cid", "sec", "code123", token_endpoint=endpoint)
+        assert tok["access_token"] == "at-authorization_code"
+        assert seen[0]["code"] == "code123"
+        assert seen[0]["redirect_uri"].endswith("/callback")
+        tok2 = refresh_token("cid", "sec", tok["refresh_token"],
+                             token_endpoint=endpoint)
+        assert tok2["access_token"] == "at-refresh_token"
+        assert seen[1]["refresh_token"] == "rt-1"
+    finally:
+        srv.shutdown()
