@@ -248,6 +248,199 @@ class SimScenario:
         return s
 
     @classmethod
+    def oom_crashloop(cls) -> "SimScenario":
+        """Recommendation service OOM-killed in a loop after a feature
+        flag doubled the embedding cache; exercises k8s pod-state +
+        memory-metric causality."""
+        s = cls(name="oom-crashloop")
+        s.incident = {
+            "id": "PD-EXAMPLE-005",
+            "title": "recommendation-svc unavailable: pods restarting repeatedly",
+            "status": "triggered", "urgency": "high",
+            "service": "recommendation-svc", "createdAt": "2026-02-14T08:10:00Z",
+        }
+        s.services = [
+            {"name": "recommendation-svc", "status": "unhealthy", "type": "eks"},
+            {"name": "feed-api", "status": "degraded", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "recommendation-svc-restarts", "state": "ALARM",
+             "reason": "container restarts > 5 in 10 minutes",
+             "service": "recommendation-svc"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-14T08:08:12Z", "service": "recommendation-svc",
+             "level": "ERROR",
+             "message": "Last State: Terminated, Reason: OOMKilled, exit code 137 "
+                        "(memory limit 2Gi)"},
+            {"timestamp": "2026-02-14T08:08:40Z", "service": "recommendation-svc",
+             "level": "WARN",
+             "message": "embedding cache resident size 1.9GiB after flag "
+                        "rec_cache_v2=on (was 0.8GiB)"},
+            {"timestamp": "2026-02-14T08:09:30Z", "service": "feed-api",
+             "level": "ERROR",
+             "message": "upstream recommendation-svc: connection refused (pod restarting)"},
+        ]
+        s.metrics = {
+            "recommendation-svc.memory_rss_gib": [0.8, 0.9, 1.4, 1.9, 2.0, 2.0],
+            "recommendation-svc.restart_count": [0, 0, 1, 3, 6, 9],
+        }
+        s.pods = [
+            {"name": "recommendation-svc-7d9f-1", "phase": "CrashLoopBackOff",
+             "restarts": 9, "reason": "OOMKilled"},
+            {"name": "recommendation-svc-7d9f-2", "phase": "CrashLoopBackOff",
+             "restarts": 8, "reason": "OOMKilled"},
+        ]
+        s.monitors = [{"name": "rec-svc availability", "status": "Alert",
+                       "query": "avg:rec.availability{*} < 95"}]
+        s.resources = {"eks": [{"name": "recommendation-svc", "desiredCount": 4,
+                                "runningCount": 0, "taskDefinition": "rec:44"}]}
+        return s
+
+    @classmethod
+    def dns_resolution(cls) -> "SimScenario":
+        """Intermittent SERVFAIL from cluster DNS after a nameserver
+        config push; exercises infrastructure-layer causality below the
+        application services."""
+        s = cls(name="dns-resolution")
+        s.incident = {
+            "id": "PD-EXAMPLE-006",
+            "title": "intermittent connection failures across services: name "
+                     "resolution errors",
+            "status": "triggered", "urgency": "high",
+            "service": "service-mesh", "createdAt": "2026-02-15T13:30:00Z",
+        }
+        s.services = [
+            {"name": "service-mesh", "status": "degraded", "type": "eks"},
+            {"name": "coredns", "status": "degraded", "type": "eks"},
+            {"name": "orders-api", "status": "degraded", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "mesh-5xx-rate", "state": "ALARM",
+             "reason": "upstream resolution failures above 2%",
+             "service": "service-mesh"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-15T13:28:05Z", "service": "orders-api",
+             "level": "ERROR",
+             "message": "dial tcp: lookup payments-internal.prod.svc on "
+                        "10.100.0.10:53: server misbehaving (SERVFAIL)"},
+            {"timestamp": "2026-02-15T13:28:40Z", "service": "coredns",
+             "level": "ERROR",
+             "message": "plugin/forward: upstream 10.0.9.9:53 timeout "
+                        "(configured 13:25 by dns-config push #812)"},
+            {"timestamp": "2026-02-15T13:29:15Z", "service": "coredns",
+             "level": "WARN",
+             "message": "dns resolution latency p99 840ms; ndots:5 expansion "
+                        "multiplying query volume"},
+        ]
+        s.metrics = {
+            "coredns.servfail_rate": [0.0, 0.0, 0.1, 2.4, 4.8, 5.1],
+            "coredns.forward_latency_ms": [2, 2, 3, 410, 790, 840],
+        }
+        s.deployments = [
+            {"service": "coredns", "version": "config-812",
+             "at": "2026-02-15T13:25:00Z",
+             "change": "dns-config push #812: forward nameserver changed"},
+        ]
+        s.monitors = [{"name": "cluster dns health", "status": "Alert",
+                       "query": "avg:coredns.servfail{*} > 1"}]
+        s.resources = {"eks": [{"name": "coredns", "desiredCount": 2,
+                                "runningCount": 2, "taskDefinition": "coredns:9"}]}
+        return s
+
+    @classmethod
+    def queue_backlog(cls) -> "SimScenario":
+        """Job queue backlog after the consumer group stalled on a poison
+        message; exercises lag/age metrics + dead-letter causality."""
+        s = cls(name="queue-backlog")
+        s.incident = {
+            "id": "PD-EXAMPLE-007",
+            "title": "order processing delayed: job queue backlog growing",
+            "status": "triggered", "urgency": "high",
+            "service": "job-queue", "createdAt": "2026-02-16T19:05:00Z",
+        }
+        s.services = [
+            {"name": "job-queue", "status": "degraded", "type": "sqs"},
+            {"name": "worker-pool", "status": "degraded", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "job-queue-age", "state": "ALARM",
+             "reason": "oldest message age > 30 minutes", "service": "job-queue"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-16T18:40:20Z", "service": "worker-pool",
+             "level": "ERROR",
+             "message": "consumer stalled: message 9f31 failed deserialization 12 "
+                        "times, not acked (poison message)"},
+            {"timestamp": "2026-02-16T18:55:00Z", "service": "job-queue",
+             "level": "WARN",
+             "message": "queue depth 1.2M and rising; consumer group lag growing"},
+            {"timestamp": "2026-02-16T19:01:10Z", "service": "worker-pool",
+             "level": "ERROR",
+             "message": "backlog processing ETA exceeds SLA; dead letter queue "
+                        "not configured for orders topic"},
+        ]
+        s.metrics = {
+            "job-queue.depth": [12000, 15000, 180000, 560000, 910000, 1200000],
+            "job-queue.oldest_age_min": [1, 1, 8, 19, 33, 42],
+        }
+        s.monitors = [{"name": "queue backlog", "status": "Alert",
+                       "query": "max:queue.age{*} > 30m"}]
+        s.resources = {"sqs": [{"name": "orders-jobs", "depth": 1200000,
+                                "dlq": "absent"}]}
+        return s
+
+    @classmethod
+    def db_cpu_saturation(cls) -> "SimScenario":
+        """Reporting database CPU pinned by a full-scan query after an
+        index was dropped in a migration; exercises slow-query causality."""
+        s = cls(name="db-cpu-saturation")
+        s.incident = {
+            "id": "PD-EXAMPLE-008",
+            "title": "dashboard timeouts: reporting database CPU saturated",
+            "status": "triggered", "urgency": "high",
+            "service": "reporting-db", "createdAt": "2026-02-17T09:45:00Z",
+        }
+        s.services = [
+            {"name": "reporting-db", "status": "degraded", "type": "rds"},
+            {"name": "dashboard-api", "status": "degraded", "type": "ecs"},
+        ]
+        s.alarms = [
+            {"name": "reporting-db-cpu", "state": "ALARM",
+             "reason": "CPUUtilization > 95% for 15 minutes",
+             "service": "reporting-db"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-02-17T09:40:00Z", "service": "reporting-db",
+             "level": "WARN",
+             "message": "slow query (11.8s): SELECT ... FROM order_events WHERE "
+                        "tenant_id = $1 — Seq Scan on order_events "
+                        "(idx_order_events_tenant missing)"},
+            {"timestamp": "2026-02-17T09:41:30Z", "service": "reporting-db",
+             "level": "WARN",
+             "message": "cpu saturation: 97% utilization, 48 active backends "
+                        "waiting on CPU"},
+            {"timestamp": "2026-02-17T09:43:00Z", "service": "dashboard-api",
+             "level": "ERROR",
+             "message": "statement timeout after 10s querying reporting-db"},
+        ]
+        s.metrics = {
+            "reporting-db.cpu_util": [22, 24, 61, 93, 97, 98],
+            "dashboard-api.timeout_rate": [0, 0, 2, 11, 29, 35],
+        }
+        s.deployments = [
+            {"service": "reporting-db", "version": "migration-0412",
+             "at": "2026-02-17T09:30:00Z",
+             "change": "migration 0412: rebuilt order_events (index drop)"},
+        ]
+        s.monitors = [{"name": "reporting db cpu", "status": "Alert",
+                       "query": "avg:rds.cpu{id:reporting-db} > 90"}]
+        s.resources = {"rds": [{"name": "reporting-db", "status": "available",
+                                "class": "db.r6g.4xlarge"}]}
+        return s
+
+    @classmethod
     def from_fixture(cls, case: dict[str, Any]) -> "SimScenario":
         """Generate a scenario from an eval fixture case: the telemetry
         reflects the expected root cause so a competent agent can find it."""
@@ -296,6 +489,10 @@ _SCENARIOS = {
     "api-gateway-5xx": SimScenario.gateway_5xx,
     "kafka-disk-pressure": SimScenario.kafka_disk_pressure,
     "tls-cert-expiry": SimScenario.cert_expiry,
+    "oom-crashloop": SimScenario.oom_crashloop,
+    "dns-resolution": SimScenario.dns_resolution,
+    "queue-backlog": SimScenario.queue_backlog,
+    "db-cpu-saturation": SimScenario.db_cpu_saturation,
 }
 
 _current: Optional[SimScenario] = None

@@ -42,6 +42,10 @@ def test_scenario_loads_and_tools_see_it(name):
     ("api-gateway-5xx", "panic"),
     ("kafka-disk-pressure", "No space left on device"),
     ("tls-cert-expiry", "certificate has expired"),
+    ("oom-crashloop", "OOMKilled"),
+    ("dns-resolution", "SERVFAIL"),
+    ("queue-backlog", "poison message"),
+    ("db-cpu-saturation", "Seq Scan"),
 ])
 def test_causal_needle_reachable_through_logs(name, needle):
     set_scenario(load_scenario(name))
