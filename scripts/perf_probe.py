@@ -151,8 +151,12 @@ def main() -> None:
         for M in (1, 8, 32):
             x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
             w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
-            for label, fn in (("skinny", lambda: ext.skinny_gemm(x, w)),
-                              ("blaslt", lambda: x @ w.t())):
+            variants = [("skinny", lambda: ext.skinny_gemm(x, w)),
+                        ("blaslt", lambda: x @ w.t())]
+            if M <= 4:
+                variants.append(("gemv", lambda: ext.decode_gemv(
+                    x, w, None, None, 0, 1e-5)))
+            for label, fn in variants:
                 for _ in range(3):
                     fn()
                 torch.cuda.synchronize()
