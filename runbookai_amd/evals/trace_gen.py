@@ -171,10 +171,10 @@ NATURAL_TELEMETRY: dict[str, dict] = {
  },
  "db-cpu-saturation": {
   "logs": [
-   "slow query log: SELECT ... full table scan taking 12s",
-   "database CPU at 100%; connection queue building",
-   "query planner falling back to sequential scan (missing index?)",
-   "statement timeout reached for 14% of queries"
+   "slow query log (11.2s): SELECT ... Seq Scan on large table (index missing)",
+   "database CPU pegged at 100%; active backends waiting on CPU",
+   "query planner falling back to sequential scan after index drop",
+   "statement timeout reached for 14% of queries; cpu saturation sustained"
   ],
   "alarm": "database CPU utilization sustained above 95% on {svc}",
   "metric": "{svc}.cpu_util"
