@@ -47,15 +47,47 @@ class CallableAdapter(BaseAdapter):
         return bool(self.fn(claim.to_dict()))
 
 
-class UnavailableAdapter(BaseAdapter):
-    """Represents http/sourcegraph/entireio adapters with no egress:
-    dispatch always fails so claims spool locally and replay later."""
+class HttpAdapter(BaseAdapter):
+    """REST provider (reference adapters/http.ts): POSTs the claim to
+    `/v1/ingest/change-session/{stage}` with the api key header; a
+    non-2xx / unreachable provider returns False so the ingestion layer
+    spools locally and replays later (the reference's degradation path).
+    sourcegraph / entireio are preconfigured variants of this transport
+    (reference adapters/sourcegraph.ts, entireio.ts: the 34-line shims
+    add only an id + `x-runbook-adapter` header)."""
 
-    def __init__(self, name: str) -> None:
+    name = "http"
+
+    def __init__(self, base_url: str, api_key: str = "",
+                 headers: Optional[dict[str, str]] = None,
+                 timeout_s: float = 10.0, name: str = "http") -> None:
+        self.base_url = base_url.rstrip("/")
+        self.api_key = api_key
+        self.headers = dict(headers or {})
+        self.timeout_s = timeout_s
         self.name = name
 
     def dispatch(self, claim: AgentChangeClaim) -> bool:
-        return False
+        import requests
+
+        stage = getattr(claim, "stage", "") or "checkpoint"
+        headers = {"Content-Type": "application/json", **self.headers}
+        if self.api_key:
+            headers["Authorization"] = f"Bearer {self.api_key}"
+        try:
+            resp = requests.post(
+                f"{self.base_url}/v1/ingest/change-session/{stage}",
+                json={"stage": stage, "claim": claim.to_dict()},
+                headers=headers, timeout=self.timeout_s)
+        except requests.RequestException:
+            return False
+        if not resp.ok:
+            return False
+        try:
+            payload = resp.json()
+        except ValueError:
+            return True   # 2xx without a JSON ack still counts as accepted
+        return bool(payload.get("accepted", payload.get("ok", True)))
 
 
 def create_adapter(config: dict[str, Any]) -> BaseAdapter:
@@ -65,5 +97,17 @@ def create_adapter(config: dict[str, Any]) -> BaseAdapter:
     if kind == "custom" and callable(config.get("fn")):
         return CallableAdapter(config["fn"])
     if kind in ("http", "sourcegraph", "entireio", "webhook"):
-        return UnavailableAdapter(kind)
+        base = config.get("baseUrl", "")
+        if not base:
+            # no endpoint configured: fail dispatch -> local spool+replay
+            a = HttpAdapter("http://unconfigured.invalid", name=kind,
+                            timeout_s=0.2)
+            return a
+        headers = dict(config.get("headers") or {})
+        if kind in ("sourcegraph", "entireio"):
+            headers.setdefault("x-runbook-adapter", kind)
+        return HttpAdapter(base, api_key=config.get("apiKey", ""),
+                           headers=headers,
+                           timeout_s=float(config.get("timeoutS", 10.0)),
+                           name=kind)
     raise ValueError(f"unknown operability-context adapter kind '{kind}'")

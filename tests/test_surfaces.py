@@ -322,10 +322,12 @@ class TestOperability:
             replay_spool,
             spool_status,
         )
-        from runbookai_amd.providers.operability_context.factory import UnavailableAdapter
+        from runbookai_amd.providers.operability_context.factory import HttpAdapter
 
         spool = str(tmp_path / "spool" / "claims.jsonl")
-        out = ingest_claim("start", adapter=UnavailableAdapter("http"), spool_path=spool,
+        # unreachable endpoint: dispatch fails fast -> local spool
+        unreachable = HttpAdapter("http://127.0.0.1:9", timeout_s=0.2)
+        out = ingest_claim("start", adapter=unreachable, spool_path=spool,
                            summary="did things")
         assert out["spooled"]
         assert spool_status(spool)["spooled"] == 1
@@ -424,3 +426,61 @@ class TestSocketMode:
         client.run(max_connections=3)
         assert client.connections == 3
         assert client.handled == 2
+
+
+class TestHttpOperabilityAdapter:
+    """REST operability provider transport (reference adapters/http.ts:
+    stage ingest endpoint, bearer auth, ack parsing, spool degradation)."""
+
+    def test_dispatch_posts_stage_and_parses_ack(self, tmp_path):
+        import json
+        import threading
+        from http.server import BaseHTTPRequestHandler, HTTPServer
+
+        seen = []
+
+        class H(BaseHTTPRequestHandler):
+            def do_POST(self):
+                body = json.loads(self.rfile.read(
+                    int(self.headers["Content-Length"])))
+                seen.append((self.path, self.headers.get("Authorization"),
+                             self.headers.get("x-runbook-adapter"), body))
+                accepted = body["claim"].get("summary") != "reject-me"
+                data = json.dumps({"accepted": accepted}).encode()
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.end_headers()
+                self.wfile.write(data)
+
+            def log_message(self, *a):
+                pass
+
+        srv = HTTPServer(("127.0.0.1", 0), H)
+        threading.Thread(target=srv.serve_forever, daemon=True).start()
+        base = f"http://127.0.0.1:{srv.server_port}"
+        try:
+            from runbookai_amd.providers.operability_context.factory import (
+                create_adapter,
+            )
+            from runbookai_amd.providers.operability_context.types import (
+                AgentChangeClaim,
+            )
+
+            adapter = create_adapter({"kind": "sourcegraph", "baseUrl": base,
+                                      "apiKey": "k1"})
+            claim = AgentChangeClaim(claim_id="c1", agent="claude",
+                                     summary="scaled svc",
+                                     metadata={"stage": "checkpoint"})
+            claim.stage = "checkpoint"
+            assert adapter.dispatch(claim)
+            path, auth, marker, body = seen[0]
+            assert path == "/v1/ingest/change-session/checkpoint"
+            assert auth == "Bearer k1"
+            assert marker == "sourcegraph"
+            assert body["claim"]["summary"] == "scaled svc"
+            # provider-side rejection -> False -> caller spools
+            bad = AgentChangeClaim(claim_id="c2", agent="claude",
+                                    summary="reject-me")
+            assert not adapter.dispatch(bad)
+        finally:
+            srv.shutdown()
