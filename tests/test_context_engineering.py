@@ -178,7 +178,7 @@ class TestLogAnalyzer:
         assert sev.index("critical") == 0 if "critical" in sev else True
 
     def test_level_filter(self):
-        out = LogAnalyzer().filter_by_level(self.LINES, "ERROR")
+        out = LogAnalyzer().filter_by_level(self.LINES, "ERROR", keep_unleveled=False)
         assert len(out) == 2
 
     def test_analyze_produces_hypotheses(self):
