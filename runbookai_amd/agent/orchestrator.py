@@ -333,9 +333,12 @@ class InvestigationOrchestrator:
         except ParseError:
             triage = {"summary": f"Triage for: {query}", "symptoms": [], "affectedServices": [],
                       "severity": "medium", "timeline": ""}
-        m.triage_summary = triage["summary"]
-        m.symptoms = triage["symptoms"]
-        m.affected_services = triage["affectedServices"]
+        if m.phase == Phase.TRIAGE:
+            m.set_triage_result(triage["summary"], triage["symptoms"],
+                                triage["affectedServices"],
+                                severity=triage.get("severity", ""))
+        else:  # resumed past triage: keep prior triage, just refresh summary
+            m.triage_summary = m.triage_summary or triage["summary"]
         self._emit("triage_done", **triage)
 
     # -- hypothesis generation (reference L877-896) ----------------------------

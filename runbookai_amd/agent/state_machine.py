@@ -76,6 +76,7 @@ class RemediationStep:
     risk: str = "low"
     requires_approval: bool = False
     matching_skill: Optional[str] = None
+    status: str = "pending"  # pending | in_progress | completed | failed | skipped
 
 
 @dataclass
@@ -92,27 +93,38 @@ class InvestigationStateMachine:
     def __init__(
         self,
         investigation_id: Optional[str] = None,
+        incident_id: Optional[str] = None,
         max_hypotheses: int = 10,
         max_depth: int = 4,
         max_iterations: int = 20,
     ) -> None:
         self.investigation_id = investigation_id or new_id("inv-")
+        self.incident_id = incident_id
         self.phase: Phase = Phase.IDLE
         self.max_hypotheses = max_hypotheses
         self.max_depth = max_depth
         self.max_iterations = max_iterations
         self.iteration = 0
         self.hypotheses: dict[str, Hypothesis] = {}
+        self._hyp_seq = 0
         self.query_results: list[QueryRecord] = []
+        self.evaluations: list[dict[str, Any]] = []
+        self.errors: list[dict[str, Any]] = []
+        self.phase_history: list[dict[str, Any]] = []
         self.conclusion: Optional[Conclusion] = None
         self.remediation_plan: Optional[RemediationPlan] = None
         self.triage_summary: str = ""
+        self.triage_severity: str = ""
         self.symptoms: list[str] = []
         self.affected_services: list[str] = []
         self.started_at: int = 0
         self.completed_at: int = 0
         self._listeners: dict[str, list[Callable[[dict[str, Any]], None]]] = {}
         self.failure_reason: str = ""
+
+    @property
+    def is_complete(self) -> bool:
+        return self.phase in (Phase.COMPLETE, Phase.FAILED)
 
     # -- events ------------------------------------------------------------
 
@@ -139,6 +151,7 @@ class InvestigationStateMachine:
             raise IllegalTransition(f"illegal transition {self.phase.value} -> {to.value}")
         prev = self.phase
         self.phase = to
+        self.phase_history.append({"from": prev.value, "to": to.value, "at": now_ms()})
         if to in (Phase.COMPLETE, Phase.FAILED):
             self.completed_at = now_ms()
         self._emit("phase_change", {"from": prev.value, "to": to.value})
@@ -158,6 +171,33 @@ class InvestigationStateMachine:
         self.iteration += 1
         return self.iteration
 
+    # -- triage (reference setTriageResult L318-327) -------------------------
+
+    def set_triage_result(
+        self,
+        summary: str,
+        symptoms: Optional[list[str]] = None,
+        affected_services: Optional[list[str]] = None,
+        severity: str = "",
+    ) -> None:
+        """Record the triage outcome. Only legal while in the triage phase
+        (reference throws on wrong-phase triage writes)."""
+        if self.phase != Phase.TRIAGE:
+            raise IllegalTransition(f"cannot set triage result in phase {self.phase.value}")
+        self.triage_summary = summary
+        self.triage_severity = severity
+        self.symptoms = list(symptoms or [])
+        self.affected_services = list(affected_services or [])
+        self._emit("triage_set", {"summary": summary, "severity": severity})
+
+    # -- errors (reference recordError L646-659) -----------------------------
+
+    def record_error(self, message: str, context: str = "") -> None:
+        entry = {"message": message, "context": context,
+                 "phase": self.phase.value, "at": now_ms()}
+        self.errors.append(entry)
+        self._emit("error", entry)
+
     # -- hypothesis tree (reference L329-426) --------------------------------
 
     def add_hypothesis(
@@ -176,8 +216,11 @@ class InvestigationStateMachine:
                 return None
             if self.depth_of(parent_id) + 1 >= self.max_depth:
                 return None
+        self._hyp_seq += 1
+        while f"hyp-{self._hyp_seq}" in self.hypotheses:  # rehydrated machines
+            self._hyp_seq += 1
         h = Hypothesis(
-            id=new_id("hyp-"),
+            id=f"hyp-{self._hyp_seq}",
             statement=statement,
             rationale=rationale,
             priority=max(1, min(5, int(priority))),
@@ -189,6 +232,9 @@ class InvestigationStateMachine:
             self.hypotheses[parent_id].children.append(h.id)
         self._emit("hypothesis_added", {"hypothesis": h.to_dict()})
         return h
+
+    def get_hypothesis(self, hypothesis_id: str) -> Optional[Hypothesis]:
+        return self.hypotheses.get(hypothesis_id)
 
     def depth_of(self, hypothesis_id: str) -> int:
         depth = 0
@@ -246,6 +292,11 @@ class InvestigationStateMachine:
         h = self.hypotheses.get(hypothesis_id)
         if h is None:
             return []
+        self.evaluations.append({
+            "hypothesisId": hypothesis_id, "action": action,
+            "confidence": float(confidence), "reasoning": reasoning,
+            "iteration": self.iteration, "at": now_ms(),
+        })
         h.confidence = max(0.0, min(1.0, float(confidence)))
         for ev in evidence or []:
             h.evidence.append(
@@ -283,13 +334,37 @@ class InvestigationStateMachine:
 
     # -- conclusion / remediation (reference L499-544) -----------------------
 
-    def set_conclusion(self, conclusion: Conclusion) -> None:
+    def set_conclusion(self, conclusion: Conclusion,
+                       confirmed_hypothesis_id: Optional[str] = None) -> None:
         self.conclusion = conclusion
         merged = set(self.affected_services) | set(conclusion.affected_services)
         self.affected_services = sorted(merged)
+        if confirmed_hypothesis_id:
+            h = self.hypotheses.get(confirmed_hypothesis_id)
+            if h is not None:
+                h.status = HypothesisStatus.CONFIRMED
+        self._emit("conclusion_reached", {
+            "rootCause": conclusion.root_cause,
+            "confidence": conclusion.confidence,
+        })
 
     def set_remediation_plan(self, plan: RemediationPlan) -> None:
         self.remediation_plan = plan
+
+    def update_step_status(self, step_index: int, status: str) -> None:
+        """Mark a remediation step's progress; emits step_completed when a
+        step reaches completed (reference updateStepStatus L546-563)."""
+        if self.remediation_plan is None:
+            raise IllegalTransition("no remediation plan set")
+        steps = self.remediation_plan.steps
+        if not 0 <= step_index < len(steps):
+            raise IndexError(f"step index {step_index} out of range")
+        steps[step_index].status = status
+        if status == "completed":
+            self._emit("step_completed", {
+                "stepIndex": step_index,
+                "description": steps[step_index].description,
+            })
 
     # -- summary (reference getSummary L566-643) -----------------------------
 
@@ -306,9 +381,15 @@ class InvestigationStateMachine:
         if self.hypotheses:
             lines.append("")
             lines.append("## Hypotheses")
-            for h in self.hypotheses.values():
-                if h.parent_id is None:
-                    lines.extend(self._hypothesis_lines(h, 0))
+            # Proven hypotheses lead; rejected (pruned) ones trail, so the
+            # reader sees the confirmed narrative first (reference L600-617).
+            rank = {HypothesisStatus.CONFIRMED: 0, HypothesisStatus.ACTIVE: 1,
+                    HypothesisStatus.INVESTIGATING: 1, HypothesisStatus.BRANCHED: 2,
+                    HypothesisStatus.PRUNED: 3}
+            roots = [h for h in self.hypotheses.values() if h.parent_id is None]
+            roots.sort(key=lambda h: (rank.get(h.status, 2), -h.confidence))
+            for h in roots:
+                lines.extend(self._hypothesis_lines(h, 0))
         if self.conclusion:
             lines.append("")
             lines.append("## Conclusion")
@@ -349,11 +430,36 @@ class InvestigationStateMachine:
     def to_dict(self) -> dict[str, Any]:
         return {
             "investigationId": self.investigation_id,
+            "incidentId": self.incident_id,
             "phase": self.phase.value,
+            "phaseHistory": list(self.phase_history),
             "iteration": self.iteration,
             "hypotheses": {k: h.to_dict() for k, h in self.hypotheses.items()},
+            "evaluations": list(self.evaluations),
+            "errors": list(self.errors),
             "triageSummary": self.triage_summary,
+            "triageSeverity": self.triage_severity,
             "symptoms": list(self.symptoms),
             "affectedServices": list(self.affected_services),
+            "conclusion": {
+                "rootCause": self.conclusion.root_cause,
+                "confidence": self.conclusion.confidence,
+                "summary": self.conclusion.summary,
+                "affectedServices": list(self.conclusion.affected_services),
+                "evidence": list(self.conclusion.evidence),
+                "contributingFactors": list(self.conclusion.contributing_factors),
+            } if self.conclusion else None,
+            "remediationPlan": {
+                "summary": self.remediation_plan.summary,
+                "rollback": self.remediation_plan.rollback,
+                "steps": [
+                    {"description": s.description, "risk": s.risk,
+                     "requiresApproval": s.requires_approval, "status": s.status,
+                     "tool": s.tool, "command": s.command}
+                    for s in self.remediation_plan.steps
+                ],
+            } if self.remediation_plan else None,
+            "startedAt": self.started_at,
+            "completedAt": self.completed_at,
             "failureReason": self.failure_reason,
         }
