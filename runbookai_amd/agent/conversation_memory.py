@@ -13,17 +13,20 @@ from typing import Any, Optional
 
 from ..utils.tokens import estimate_tokens
 from .scratchpad import jaccard
-from .types import now_ms
+from .types import new_id, now_ms
 
 
 @dataclass
 class Message:
-    role: str  # user | assistant
+    role: str  # user | assistant | system
     content: str
     timestamp: int = field(default_factory=now_ms)
+    id: str = field(default_factory=lambda: new_id("msg-"))
+    metadata: dict[str, Any] = field(default_factory=dict)
 
     def to_dict(self) -> dict[str, Any]:
-        return {"role": self.role, "content": self.content, "timestamp": self.timestamp}
+        return {"id": self.id, "role": self.role, "content": self.content,
+                "timestamp": self.timestamp, "metadata": self.metadata}
 
 
 @dataclass
@@ -39,21 +42,114 @@ class InvestigationSummary:
 
 
 class ConversationMemory:
-    def __init__(self, summarize_after_messages: int = 16, llm: Any = None) -> None:
+    def __init__(self, summarize_after_messages: int = 16, llm: Any = None,
+                 max_messages: int = 200, max_investigations: int = 50) -> None:
         self.summarize_after = summarize_after_messages
+        self.max_messages = max_messages
+        self.max_investigations = max_investigations
         self.llm = llm
         self.messages: list[Message] = []
         self.compressed_summary: str = ""
         self.investigations: list[InvestigationSummary] = []
 
-    def add_message(self, role: str, content: str) -> None:
-        self.messages.append(Message(role=role, content=content))
+    def add_message(self, role: str, content: str,
+                    metadata: Optional[dict[str, Any]] = None) -> Message:
+        msg = Message(role=role, content=content, metadata=dict(metadata or {}))
+        self.messages.append(msg)
         self.maybe_compress()
+        if len(self.messages) > self.max_messages:
+            self.messages = self.messages[-self.max_messages:]
+        return msg
 
     def add_investigation(self, query: str, answer: str, services: Optional[list[str]] = None) -> None:
         self.investigations.append(
             InvestigationSummary(query=query, answer_digest=answer[:400], services=list(services or []))
         )
+        if len(self.investigations) > self.max_investigations:
+            self.investigations = self.investigations[-self.max_investigations:]
+
+    # -- accessors (reference L61-123) ---------------------------------------
+
+    def get_messages(self) -> list[Message]:
+        return list(self.messages)
+
+    def recent_messages(self, n: int = 10) -> list[Message]:
+        return self.messages[-n:]
+
+    def last_message(self) -> Optional[Message]:
+        return self.messages[-1] if self.messages else None
+
+    def last_user_message(self) -> Optional[Message]:
+        for m in reversed(self.messages):
+            if m.role == "user":
+                return m
+        return None
+
+    def messages_since(self, message_id: str) -> list[Message]:
+        """Messages after the one with the given ID; empty if unknown."""
+        for i, m in enumerate(self.messages):
+            if m.id == message_id:
+                return self.messages[i + 1:]
+        return []
+
+    def get_investigations(self) -> list[InvestigationSummary]:
+        return list(self.investigations)
+
+    def recent_investigations(self, n: int = 5) -> list[InvestigationSummary]:
+        return self.investigations[-n:]
+
+    def search_investigations(self, query: str, limit: int = 3) -> list[InvestigationSummary]:
+        q = query.lower()
+        hits = [s for s in self.investigations
+                if q in s.query.lower() or q in s.answer_digest.lower()
+                or any(q in svc.lower() for svc in s.services)]
+        return hits[:limit]
+
+    # -- reference resolution (reference getReference L336-377) --------------
+
+    def get_reference(self, topic: str) -> Optional[str]:
+        """Resolve a topic mention against past investigations first, then
+        assistant messages — used for 'that incident from earlier' turns."""
+        t = topic.lower()
+        for s in reversed(self.investigations):
+            if t in s.query.lower() or t in s.answer_digest.lower():
+                return f"Earlier investigation '{s.query}': {s.answer_digest}"
+        for m in reversed(self.messages):
+            if m.role == "assistant" and t in m.content.lower():
+                return m.content
+        return None
+
+    # -- management (reference L395-436) -------------------------------------
+
+    def clear(self) -> None:
+        self.messages = []
+        self.investigations = []
+        self.compressed_summary = ""
+
+    def clear_messages(self) -> None:
+        self.messages = []
+        self.compressed_summary = ""
+
+    def stats(self) -> dict[str, Any]:
+        text = "\n".join(m.content for m in self.messages)
+        return {
+            "messageCount": len(self.messages),
+            "investigationCount": len(self.investigations),
+            "estimatedTokens": estimate_tokens(text + self.compressed_summary),
+            "compressed": bool(self.compressed_summary),
+        }
+
+    # -- conversation summary (reference createSummary L293-334) -------------
+
+    def summarize(self) -> dict[str, str]:
+        convo = "; ".join(
+            f"{m.role}: {m.content[:80]}" for m in self.messages[-8:])
+        invs = "; ".join(
+            f"{s.query} -> {s.answer_digest[:100]}" for s in self.investigations[-5:])
+        return {
+            "conversationSummary": (self.compressed_summary + " " + convo).strip(),
+            "investigationsSummary": invs,
+        }
 
     # -- compression (reference L422) ----------------------------------------
 
@@ -141,7 +237,10 @@ class ConversationMemory:
                                query: str = "") -> str:
         parts: list[str] = []
         if self.compressed_summary:
-            parts.append("## Earlier conversation (compressed)\n" + self.compressed_summary)
+            # the compressed digest may not eat more than half the budget
+            cap = max(200, token_budget * 4 // 2)  # ≈4 chars/token
+            parts.append("## Earlier conversation (compressed)\n"
+                         + self.compressed_summary[-cap:])
         if query:
             rel = self.related_context_section(query)
             if rel:
@@ -172,7 +271,10 @@ class ConversationMemory:
     def from_json(cls, raw: str) -> "ConversationMemory":
         data = json.loads(raw)
         mem = cls(summarize_after_messages=data.get("summarizeAfter", 16))
-        mem.messages = [Message(role=m["role"], content=m["content"], timestamp=m.get("timestamp", 0))
+        mem.messages = [Message(role=m["role"], content=m["content"],
+                                timestamp=m.get("timestamp", 0),
+                                id=m.get("id", new_id("msg-")),
+                                metadata=m.get("metadata", {}))
                         for m in data.get("messages", [])]
         mem.compressed_summary = data.get("compressedSummary", "")
         mem.investigations = [
