@@ -207,12 +207,19 @@ def parse_triage_response(text: str) -> dict[str, Any]:
     severity = str(d.get("severity", "medium")).lower()
     if severity not in ("low", "medium", "high", "critical"):
         severity = "medium"
+    initial = []
+    for h in d.get("initialHypotheses", []) if isinstance(d.get("initialHypotheses"), list) else []:
+        try:
+            initial.append(validate_hypothesis(h))
+        except ParseError:
+            continue
     return {
         "summary": str(d.get("summary", "")),
         "symptoms": _as_str_list(d.get("symptoms")),
         "affectedServices": _as_str_list(d.get("affectedServices")),
         "severity": severity,
         "timeline": str(d.get("timeline", "")),
+        "initialHypotheses": initial,
     }
 
 

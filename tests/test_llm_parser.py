@@ -159,3 +159,75 @@ class TestPromptFill:
         kind, body = split_schema_tag(p)
         assert kind is None
         assert body == p
+
+
+class TestNormalizationEdges:
+    """Reference llm-parser.test.ts:104-156, 178-230, 253-320, 522-560."""
+
+    def test_priority_clamped_into_range(self):
+        out = parse_hypothesis_generation(
+            '{"hypotheses": [{"statement": "a", "rationale": "r", "priority": 99},'
+            '{"statement": "b", "rationale": "r", "priority": -3},'
+            '{"statement": "c", "rationale": "r", "priority": "bogus"}]}')
+        assert [h["priority"] for h in out] == [5, 1, 3]
+
+    def test_confidence_clamped(self):
+        from runbookai_amd.agent.llm_parser import parse_evidence_evaluation
+
+        hi = parse_evidence_evaluation('{"action": "confirm", "confidence": 4.2, "reasoning": ""}')
+        lo = parse_evidence_evaluation('{"action": "prune", "confidence": -1, "reasoning": ""}')
+        assert hi["confidence"] == 1.0 and lo["confidence"] == 0.0
+
+    def test_evaluation_with_sub_hypotheses(self):
+        from runbookai_amd.agent.llm_parser import parse_evidence_evaluation
+
+        out = parse_evidence_evaluation(
+            '{"action": "branch", "confidence": 0.5, "reasoning": "split",'
+            '"subHypotheses": [{"statement": "s1", "rationale": "r", "priority": 2},'
+            '{"rationale": "no statement"}]}')
+        assert len(out["subHypotheses"]) == 1
+        assert out["subHypotheses"][0]["statement"] == "s1"
+
+    def test_triage_initial_hypotheses(self):
+        from runbookai_amd.agent.llm_parser import parse_triage_response
+
+        out = parse_triage_response(
+            '{"summary": "s", "severity": "high",'
+            '"initialHypotheses": [{"statement": "bad deploy", "rationale": "recent"}]}')
+        assert out["initialHypotheses"][0]["statement"] == "bad deploy"
+
+    def test_conclusion_string_fields_coerced_to_lists(self):
+        from runbookai_amd.agent.llm_parser import parse_conclusion
+
+        out = parse_conclusion(
+            '{"rootCause": "rc", "confidence": "high", "summary": "s",'
+            '"evidence": "a single string", "contributingFactors": "also one"}')
+        assert out["evidence"] == ["a single string"]
+        assert out["contributingFactors"] == ["also one"]
+
+    def test_nested_json_extracted(self):
+        from runbookai_amd.agent.llm_parser import parse_json
+
+        out = parse_json('Before text {"a": {"b": {"c": [1, 2, {"d": 3}]}}} after')
+        assert out["a"]["b"]["c"][2]["d"] == 3
+
+    def test_fill_multiple_occurrences(self):
+        from runbookai_amd.agent.llm_parser import PROMPTS, fill_prompt
+
+        PROMPTS["_twice"] = "service {svc} then {svc} again"
+        try:
+            out = fill_prompt("_twice", tag_schema=False, svc="api")
+            assert out.count("api") == 2
+        finally:
+            del PROMPTS["_twice"]
+
+    def test_all_templates_have_placeholders(self):
+        import re
+
+        from runbookai_amd.agent.llm_parser import PROMPTS
+
+        for kind in ("triage", "generateHypotheses", "evaluateEvidence",
+                     "generateConclusion", "generateRemediation", "analyzeLogs"):
+            assert kind in PROMPTS
+            # every template has at least one single-brace placeholder
+            assert re.search(r"(?<!\{)\{[a-zA-Z_]+\}(?!\})", PROMPTS[kind]), kind
