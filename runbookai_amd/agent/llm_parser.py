@@ -474,6 +474,7 @@ PROMPTS: dict[str, str] = {
         "Root cause: {rootCause}\n"
         "Affected services: {services}\n"
         "Relevant runbooks:\n{runbooks}\n"
+        "Available skills (pre-approved automation):\n{skills}\n"
         "Code-fix candidates:\n{codeFixes}\n\n"
         "Plan safe, ordered remediation steps. Mark risky steps requiresApproval.\n"
         "Respond with ONLY a JSON object:\n"

@@ -138,6 +138,10 @@ class InvestigationOrchestrator:
     def on(self, cb: Callable[[InvestigationEvent], None]) -> None:
         self._listeners.append(cb)
 
+    def off(self, cb: Callable[[InvestigationEvent], None]) -> None:
+        if cb in self._listeners:
+            self._listeners.remove(cb)
+
     def _emit(self, type_: str, **data: Any) -> None:
         ev = InvestigationEvent(type=type_, data=data)
         for cb in self._listeners:
@@ -501,6 +505,16 @@ class InvestigationOrchestrator:
 
     def _fetch_knowledge_digest(self, query: str, limit: int = 3) -> str:
         if self.retriever is None:
+            # no in-process retriever: the knowledge tool (if wired) serves
+            # the same purpose (reference always goes through search_knowledge)
+            if self._tool_available("search_knowledge"):
+                result, error = self._execute("search_knowledge",
+                                              {"query": query, "limit": limit})
+                if not error and isinstance(result, dict):
+                    lines = [f"- {h.get('title', '?')}: {str(h.get('content', ''))[:200]}"
+                             for h in result.get("results", []) if isinstance(h, dict)]
+                    if lines:
+                        return "\n".join(lines)
             return "(no knowledge base configured)"
         try:
             hits = self.retriever.search(query, limit=limit)
@@ -530,6 +544,34 @@ class InvestigationOrchestrator:
                 return "\n".join(f"- {i.get('title', '?')} ({i.get('url', '')})" for i in items[:5])
         return "(no code-fix candidates found)"
 
+    def analyze_logs_for_hypothesis(self, lines: list[str],
+                                    known_services: Optional[list[str]] = None) -> dict[str, Any]:
+        """Pattern-analyze raw log lines into hypothesis seeds (reference
+        analyzeLogsForHypothesis, L691-718): dictionary analysis always runs;
+        the orchestrator's LLM merges a narrative summary when available."""
+        from .log_analyzer import LogAnalyzer
+
+        return LogAnalyzer(llm=self.llm).analyze(lines, known_services=known_services)
+
+    def _fetch_available_skills(self) -> str:
+        """List pre-approved automation skills so the planner can map steps
+        onto them (reference remediation context, L1049-1093)."""
+        if not self._tool_available("skill"):
+            return "(no skill runner configured)"
+        result, error = self._execute("skill", {"action": "list"})
+        if error or not isinstance(result, dict):
+            return "(skill listing unavailable)"
+        skills = result.get("skills", [])
+        if not skills:
+            return "(no skills registered)"
+        lines = []
+        for s in skills[:10]:
+            if isinstance(s, dict):
+                lines.append(f"- {s.get('name', '?')}: {s.get('description', '')}")
+            else:
+                lines.append(f"- {s}")
+        return "\n".join(lines)
+
     def _run_remediation(self, query: str) -> None:
         m = self.machine
         if m.conclusion is None or m.phase != Phase.CONCLUDE:
@@ -538,6 +580,7 @@ class InvestigationOrchestrator:
             return
         m.transition(Phase.REMEDIATE)
         runbooks = self._fetch_knowledge_digest(m.conclusion.root_cause or query)
+        skills = self._fetch_available_skills()
         code_fixes = self._fetch_code_fix_candidates()
         try:
             p = parse_remediation_plan(
@@ -547,6 +590,7 @@ class InvestigationOrchestrator:
                         rootCause=m.conclusion.root_cause,
                         services=", ".join(m.conclusion.affected_services),
                         runbooks=runbooks,
+                        skills=skills,
                         codeFixes=code_fixes,
                     )
                 )

@@ -206,3 +206,68 @@ def test_event_stream():
     assert "phase" in events
     assert "hypothesis" in events
     assert "conclusion" in events
+
+
+def test_off_removes_event_handler():
+    llm = scripted_llm()
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=MockToolExecutor())
+    seen = []
+    cb = seen.append
+    orch.on(cb)
+    orch.off(cb)
+    orch.investigate("checkout latency")
+    assert seen == []
+    orch.off(cb)  # double-off is a no-op
+
+
+def test_analyze_logs_for_hypothesis():
+    orch = InvestigationOrchestrator(llm=MockLLMClient(), tool_executor=MockToolExecutor())
+    logs = [
+        "2024-01-15T10:00:00Z ERROR connection timed out",
+        "2024-01-15T10:01:00Z ERROR database connection pool exhausted",
+    ]
+    analysis = orch.analyze_logs_for_hypothesis(logs)
+    assert analysis["totalLines"] == 2
+    assert analysis["patterns"]
+    assert analysis["suggestedHypotheses"]
+
+
+def test_remediation_prompt_includes_skills_and_runbooks():
+    llm = scripted_llm()
+    tools = MockToolExecutor(overrides={
+        "skill": {"skills": [{"name": "restart-service", "description": "rolling restart"}]},
+    })
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=tools)
+    orch.investigate("checkout latency redis")
+    remediation_prompts = [c["prompt"] for c in llm.calls
+                           if c["kind"] == "complete" and "planning remediation" in c["prompt"]]
+    assert remediation_prompts
+    assert "restart-service" in remediation_prompts[-1]
+    assert "Redis connection exhaustion runbook" in remediation_prompts[-1]
+
+
+def test_incident_id_passed_to_provider():
+    llm = scripted_llm()
+    tools = MockToolExecutor()
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=tools)
+    orch.investigate("checkout latency", incident_id="PD-42")
+    pd_calls = [params for tool, params in tools.calls if tool == "pagerduty_get_incident"]
+    assert pd_calls and pd_calls[0]["incidentId"] == "PD-42"
+
+
+def test_max_iterations_bounds_loop():
+    llm = MockLLMClient()
+    llm.on(r"triaging", json.dumps({"summary": "s", "symptoms": [], "affectedServices": [],
+                                    "severity": "low", "timeline": ""}))
+    llm.on(r"generating root-cause hypotheses", json.dumps({
+        "hypotheses": [{"statement": "h", "rationale": "r", "priority": 1}]}))
+    # evaluation never resolves: always continue
+    llm.on(r"evaluating evidence", json.dumps({
+        "action": "continue", "confidence": 0.5, "reasoning": "inconclusive"}))
+    llm.on(r"writing the conclusion", json.dumps({
+        "rootCause": "undetermined", "confidence": "low", "summary": "s"}))
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=MockToolExecutor(),
+                                     max_iterations=3)
+    result = orch.investigate("mystery")
+    assert result.success
+    assert orch.machine.iteration <= 3
