@@ -25,12 +25,43 @@ DANGEROUS_PATTERNS = [
 ]
 
 _SERVICE_RE = re.compile(r"\b([a-z][a-z0-9]*(?:-[a-z0-9]+)+)\b")
+_SERVICE_KV_RE = re.compile(r"\bservice[=:]\s*\"?([\w-]+)\"?", re.I)
 _SYMPTOM_WORDS = ("latency", "timeout", "error", "5xx", "oom", "crash", "exhausted",
                   "deadlock", "throttl", "unavailable", "degraded")
+_HTTP_CODE_RE = re.compile(r"\b([45]\d\d)\b")
+_PERF_RE = re.compile(r"\bslow(?:ly|ness)?\b|high (?:latency|cpu|memory)|spik(?:e|ing)|p9[59]", re.I)
 
 
-def handle_session_start(payload: dict[str, Any], retriever: Any = None) -> dict[str, Any]:
-    """Knowledge-stats banner (reference @244-282)."""
+def extract_services(prompt: str) -> list[str]:
+    """Dashed service names plus service=NAME mentions (reference @326-370)."""
+    found = [m.group(1) for m in _SERVICE_KV_RE.finditer(prompt)]
+    found += _SERVICE_RE.findall(prompt)
+    return list(dict.fromkeys(s for s in found if len(s) >= 4))[:4]
+
+
+def extract_symptoms(prompt: str) -> list[str]:
+    """Symptom keywords, HTTP error codes and performance phrasing
+    (reference @372-404)."""
+    low = prompt.lower()
+    symptoms = [w for w in _SYMPTOM_WORDS if w in low]
+    for code in _HTTP_CODE_RE.findall(prompt):
+        tag = f"http-{code[0]}xx"
+        if tag not in symptoms:
+            symptoms.append(tag)
+    if _PERF_RE.search(prompt) and "performance" not in symptoms:
+        symptoms.append("performance")
+    return symptoms[:6]
+
+
+def handle_session_start(payload: dict[str, Any], retriever: Any = None,
+                         store: Any = None) -> dict[str, Any]:
+    """Session-state seed + knowledge-stats banner (reference @244-282)."""
+    if store is not None:
+        try:
+            store.append_event(payload.get("session_id", "unknown"),
+                               {"kind": "session_start", "promptCount": 0})
+        except Exception:  # noqa: BLE001
+            pass
     stats: dict[str, Any] = {}
     if retriever is not None:
         try:
@@ -45,11 +76,22 @@ def handle_session_start(payload: dict[str, Any], retriever: Any = None) -> dict
     return {"continue": True, "systemMessage": banner}
 
 
-def handle_user_prompt_submit(payload: dict[str, Any], retriever: Any = None) -> dict[str, Any]:
+#: session_id -> prompts seen this process (reference tracks in state file)
+_PROMPT_COUNTS: dict[str, int] = {}
+
+
+def handle_user_prompt_submit(payload: dict[str, Any], retriever: Any = None,
+                              inject_context: bool = True) -> dict[str, Any]:
     """Context injection (reference @288-374)."""
     prompt = str(payload.get("prompt", payload.get("user_prompt", "")))
-    services = list(dict.fromkeys(_SERVICE_RE.findall(prompt)))[:4]
-    symptoms = [w for w in _SYMPTOM_WORDS if w in prompt.lower()][:4]
+    if not prompt.strip():
+        return {"continue": True}
+    sid = str(payload.get("session_id", "unknown"))
+    _PROMPT_COUNTS[sid] = _PROMPT_COUNTS.get(sid, 0) + 1
+    if not inject_context:
+        return {"continue": True}
+    services = extract_services(prompt)
+    symptoms = extract_symptoms(prompt)[:4]
     if retriever is None or not (services or symptoms):
         return {"continue": True}
     query = " ".join(services + symptoms)
@@ -97,13 +139,18 @@ def handle_stop(payload: dict[str, Any], store: Any = None) -> dict[str, Any]:
     return {"continue": True}
 
 
-def dispatch(payload: dict[str, Any], retriever: Any = None, store: Any = None) -> dict[str, Any]:
+def prompt_count(session_id: str) -> int:
+    return _PROMPT_COUNTS.get(session_id, 0)
+
+
+def dispatch(payload: dict[str, Any], retriever: Any = None, store: Any = None,
+             inject_context: bool = True) -> dict[str, Any]:
     """Stdin dispatcher (reference @455-508)."""
     event = payload.get("hook_event_name", payload.get("event", ""))
     if event == "SessionStart":
-        return handle_session_start(payload, retriever)
+        return handle_session_start(payload, retriever, store)
     if event == "UserPromptSubmit":
-        return handle_user_prompt_submit(payload, retriever)
+        return handle_user_prompt_submit(payload, retriever, inject_context=inject_context)
     if event == "PreToolUse":
         return handle_pre_tool_use(payload)
     if event == "PostToolUse":
