@@ -33,6 +33,7 @@ class InvestigationCheckpoint:
     root_cause: str = ""
     created_at: float = field(default_factory=time.time)
     label: str = ""
+    session_id: str = ""
 
     def to_dict(self) -> dict[str, Any]:
         return {
@@ -48,6 +49,7 @@ class InvestigationCheckpoint:
             "rootCause": self.root_cause,
             "createdAt": self.created_at,
             "label": self.label,
+            "sessionId": self.session_id,
         }
 
     @classmethod
@@ -65,6 +67,7 @@ class InvestigationCheckpoint:
             root_cause=d.get("rootCause", ""),
             created_at=d.get("createdAt", 0.0),
             label=d.get("label", ""),
+            session_id=d.get("sessionId", ""),
         )
 
     def format(self) -> str:
@@ -86,6 +89,41 @@ class InvestigationCheckpoint:
         if self.evidence:
             lines.append(f"  evidence items: {len(self.evidence)}")
         return "\n".join(lines)
+
+    def format_markdown(self) -> str:
+        """Markdown rendering for chat surfaces (reference L467-517)."""
+        ts = time.strftime("%Y-%m-%d %H:%M:%S", time.localtime(self.created_at))
+        lines = [f"## Checkpoint `{self.checkpoint_id}`",
+                 f"**Investigation:** {self.investigation_id}  ·  "
+                 f"**Phase:** {self.phase or '?'}  ·  **Created:** {ts}"]
+        if self.label:
+            lines.append(f"**Label:** {self.label}")
+        if self.root_cause:
+            lines.append(f"**Root cause:** {self.root_cause}")
+        if self.services:
+            lines.append("**Services:** " + ", ".join(self.services[:8]))
+        if self.hypotheses:
+            lines.append("")
+            lines.append(f"### Hypotheses ({len(self.hypotheses)})")
+            for h in self.hypotheses[:8]:
+                lines.append(f"- [{h.get('status', '?')}] {h.get('statement', '')[:120]}")
+        if self.evidence:
+            lines.append("")
+            lines.append(f"_{len(self.evidence)} evidence items captured._")
+        return "\n".join(lines)
+
+
+def format_checkpoint_list_markdown(checkpoints: list["InvestigationCheckpoint"]) -> str:
+    """Markdown table of checkpoints (reference L544-589)."""
+    if not checkpoints:
+        return "_No checkpoints found._"
+    lines = ["| Checkpoint | Phase | Hypotheses | Label | Created |",
+             "|---|---|---|---|---|"]
+    for cp in checkpoints:
+        ts = time.strftime("%Y-%m-%d %H:%M", time.localtime(cp.created_at))
+        lines.append(f"| `{cp.checkpoint_id}` | {cp.phase or '?'} | "
+                     f"{len(cp.hypotheses)} | {cp.label or ''} | {ts} |")
+    return "\n".join(lines)
 
 
 class CheckpointStore:
@@ -116,7 +154,26 @@ class CheckpointStore:
         with open(self._path(checkpoint.investigation_id, checkpoint.checkpoint_id),
                   "w", encoding="utf-8") as f:
             json.dump(checkpoint.to_dict(), f, indent=1)
+        self._write_latest(checkpoint.investigation_id, checkpoint.checkpoint_id)
         return checkpoint.checkpoint_id
+
+    # latest.json pointer (reference L162-178, L373-423)
+    def _latest_path(self, investigation_id: str) -> str:
+        return os.path.join(self._inv_dir(investigation_id), "latest.json")
+
+    def _write_latest(self, investigation_id: str, checkpoint_id: str) -> None:
+        with open(self._latest_path(investigation_id), "w", encoding="utf-8") as f:
+            json.dump({"checkpointId": checkpoint_id}, f)
+
+    def latest_id(self, investigation_id: str) -> Optional[str]:
+        path = self._latest_path(investigation_id)
+        if not os.path.exists(path):
+            return None
+        try:
+            with open(path, encoding="utf-8") as f:
+                return json.load(f).get("checkpointId")
+        except (json.JSONDecodeError, OSError):
+            return None
 
     def load(self, investigation_id: str, checkpoint_id: str) -> Optional[InvestigationCheckpoint]:
         path = self._path(investigation_id, checkpoint_id)
@@ -126,6 +183,11 @@ class CheckpointStore:
             return InvestigationCheckpoint.from_dict(json.load(f))
 
     def load_latest(self, investigation_id: str) -> Optional[InvestigationCheckpoint]:
+        lid = self.latest_id(investigation_id)
+        if lid:
+            cp = self.load(investigation_id, lid)
+            if cp is not None:
+                return cp
         cps = self.list(investigation_id)
         return cps[-1] if cps else None
 
@@ -135,7 +197,7 @@ class CheckpointStore:
             return []
         out = []
         for fn in os.listdir(inv_dir):
-            if not fn.endswith(".json"):
+            if not fn.endswith(".json") or fn == "latest.json":
                 continue
             try:
                 with open(os.path.join(inv_dir, fn), encoding="utf-8") as f:
@@ -151,19 +213,51 @@ class CheckpointStore:
         return sorted(d for d in os.listdir(self.base_dir)
                       if os.path.isdir(os.path.join(self.base_dir, d)))
 
+    def investigations_summary(self) -> list[dict[str, Any]]:
+        """Per-investigation rollup with checkpoint count and the latest
+        checkpoint's phase/hypothesis count (reference L314-349)."""
+        out = []
+        for inv in self.list_investigations():
+            cps = self.list(inv)
+            if not cps:
+                continue
+            latest = self.load_latest(inv) or cps[-1]
+            out.append({
+                "investigationId": inv,
+                "checkpointCount": len(cps),
+                "latest": {
+                    "checkpointId": latest.checkpoint_id,
+                    "phase": latest.phase,
+                    "createdAt": latest.created_at,
+                    "hypothesisCount": len(latest.hypotheses),
+                    "rootCause": latest.root_cause,
+                },
+            })
+        return out
+
     def delete(self, investigation_id: str, checkpoint_id: Optional[str] = None) -> int:
         if checkpoint_id is not None:
             path = self._path(investigation_id, checkpoint_id)
-            if os.path.exists(path):
-                os.remove(path)
-                return 1
-            return 0
+            if not os.path.exists(path):
+                return 0
+            os.remove(path)
+            if self.latest_id(investigation_id) == checkpoint_id:
+                remaining = self.list(investigation_id)
+                if remaining:
+                    self._write_latest(investigation_id, remaining[-1].checkpoint_id)
+                else:
+                    try:
+                        os.remove(self._latest_path(investigation_id))
+                    except OSError:
+                        pass
+            return 1
         removed = 0
         inv_dir = self._inv_dir(investigation_id)
         if os.path.isdir(inv_dir):
             for fn in os.listdir(inv_dir):
                 os.remove(os.path.join(inv_dir, fn))
-                removed += 1
+                if fn != "latest.json":  # the pointer is not a checkpoint
+                    removed += 1
             os.rmdir(inv_dir)
         return removed
 

@@ -101,3 +101,131 @@ class TestResume:
         assert result.success
         # the resumed event fired with the restored context
         assert result.investigation_id == "inv-resume-1"
+
+
+class TestStoreParity:
+    """Reference session/__tests__/checkpoint.test.ts:23-458."""
+
+    def _cp(self, inv="inv-a", **kw):
+        from runbookai_amd.session.checkpoint import CheckpointStore, InvestigationCheckpoint
+
+        return InvestigationCheckpoint(
+            checkpoint_id=CheckpointStore.new_id(), investigation_id=inv, **kw)
+
+    def test_new_id_is_12_hex_and_unique(self):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        ids = {CheckpointStore.new_id() for _ in range(20)}
+        assert len(ids) == 20
+        assert all(len(i) == 12 and all(c in "0123456789abcdef" for c in i) for i in ids)
+
+    def test_session_id_round_trip(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore, InvestigationCheckpoint
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        cp = self._cp(session_id="sess-7")
+        store.save(cp)
+        assert store.load(cp.investigation_id, cp.checkpoint_id).session_id == "sess-7"
+
+    def test_latest_pointer_updated_on_save(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        a, b = self._cp(), self._cp()
+        store.save(a)
+        assert store.latest_id("inv-a") == a.checkpoint_id
+        store.save(b)
+        assert store.latest_id("inv-a") == b.checkpoint_id
+        assert store.load_latest("inv-a").checkpoint_id == b.checkpoint_id
+
+    def test_load_nonexistent_returns_none(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        assert store.load("inv-a", "nope") is None
+        assert store.load_latest("inv-a") is None
+        assert store.list("inv-a") == []
+
+    def test_delete_latest_repoints(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        a, b = self._cp(), self._cp()
+        a.created_at -= 5
+        store.save(a)
+        store.save(b)
+        assert store.delete("inv-a", b.checkpoint_id) == 1
+        assert store.latest_id("inv-a") == a.checkpoint_id
+
+    def test_delete_only_checkpoint_removes_pointer(self, tmp_path):
+        import os
+
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        a = self._cp()
+        store.save(a)
+        store.delete("inv-a", a.checkpoint_id)
+        assert store.latest_id("inv-a") is None
+        assert not os.path.exists(str(tmp_path / "inv-a" / "latest.json"))
+
+    def test_delete_nonexistent_returns_zero(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        assert store.delete("inv-a", "nope") == 0
+
+    def test_delete_all_for_investigation(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        for _ in range(3):
+            store.save(self._cp())
+        assert store.delete("inv-a") == 3
+        assert store.list("inv-a") == []
+
+    def test_latest_json_not_listed_as_checkpoint(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        store.save(self._cp())
+        assert len(store.list("inv-a")) == 1
+
+    def test_investigations_summary(self, tmp_path):
+        from runbookai_amd.session.checkpoint import CheckpointStore
+
+        store = CheckpointStore(base_dir=str(tmp_path))
+        store.save(self._cp("inv-a", hypotheses=[{"id": "hyp-1", "statement": "x"}],
+                            phase="investigate"))
+        store.save(self._cp("inv-a"))
+        store.save(self._cp("inv-b", root_cause="bad deploy"))
+        summary = {s["investigationId"]: s for s in store.investigations_summary()}
+        assert summary["inv-a"]["checkpointCount"] == 2
+        assert summary["inv-b"]["latest"]["rootCause"] == "bad deploy"
+        assert "hypothesisCount" in summary["inv-a"]["latest"]
+
+    def test_format_markdown(self):
+        cp = self._cp(phase="evaluate", label="mid-flight",
+                      root_cause="redis pool exhaustion",
+                      services=["checkout-api"],
+                      hypotheses=[{"status": "confirmed", "statement": "pool too small"}])
+        md = cp.format_markdown()
+        assert "## Checkpoint" in md
+        assert "redis pool exhaustion" in md
+        assert "pool too small" in md
+        assert "checkout-api" in md
+
+    def test_format_list_markdown(self):
+        from runbookai_amd.session.checkpoint import format_checkpoint_list_markdown
+
+        md = format_checkpoint_list_markdown([
+            self._cp(phase="triage", label="start"),
+            self._cp(phase="conclude"),
+        ])
+        assert md.startswith("| Checkpoint |")
+        assert "triage" in md and "conclude" in md
+
+    def test_format_empty_list(self):
+        from runbookai_amd.session.checkpoint import format_checkpoint_list_markdown
+
+        assert "No checkpoints" in format_checkpoint_list_markdown([])
