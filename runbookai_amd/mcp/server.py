@@ -25,15 +25,23 @@ class MCPServer:
     # -- tool surface -----------------------------------------------------------
 
     def tool_specs(self) -> list[dict[str, Any]]:
-        q = {"type": "object", "properties": {"query": {"type": "string"},
-                                              "limit": {"type": "integer"}},
+        q = {"type": "object",
+             "properties": {"query": {"type": "string"},
+                            "limit": {"type": "integer"},
+                            "services": {"type": "array", "items": {"type": "string"}}},
              "required": ["query"]}
-        svc = {"type": "object", "properties": {"service": {"type": "string"}}}
+        issues = {"type": "object",
+                  "properties": {"query": {"type": "string"},
+                                 "limit": {"type": "integer"},
+                                 "symptoms": {"type": "array", "items": {"type": "string"}}}}
+        svc = {"type": "object", "properties": {"service": {"type": "string"},
+                                                "type": {"type": "string"}}}
         return [
             {"name": "search_runbooks", "description": "Search operational runbooks.",
              "inputSchema": q},
-            {"name": "get_known_issues", "description": "Known issues matching symptoms.",
-             "inputSchema": q},
+            {"name": "get_known_issues",
+             "description": "Known issues matching symptoms or a query.",
+             "inputSchema": issues},
             {"name": "search_postmortems", "description": "Find similar past incidents.",
              "inputSchema": q},
             {"name": "get_knowledge_stats", "description": "Knowledge base statistics.",
@@ -42,22 +50,39 @@ class MCPServer:
              "inputSchema": svc},
         ]
 
+    @staticmethod
+    def _filter_by_services(results: list[dict[str, Any]],
+                            services: list[str]) -> list[dict[str, Any]]:
+        if not services:
+            return results
+        wanted = {s.lower() for s in services}
+        return [r for r in results
+                if wanted & {str(s).lower() for s in r.get("services", [])}]
+
     def call_tool(self, name: str, args: dict[str, Any]) -> Any:
         query = str(args.get("query", ""))
         limit = int(args.get("limit", 5))
+        services = [str(s) for s in args.get("services", []) or []]
         if name == "search_runbooks":
-            return {"results": self.retriever.search(query, limit=limit, doc_type="runbook")}
+            hits = self.retriever.search(query, limit=limit, doc_type="runbook")
+            return {"results": self._filter_by_services(hits, services)}
         if name == "get_known_issues":
-            return {"results": self.retriever.search(query, limit=limit, doc_type="known_issue")}
+            symptoms = [str(s) for s in args.get("symptoms", []) or []]
+            q = " ".join([query] + symptoms).strip()
+            return {"results": self.retriever.search(q, limit=limit, doc_type="known_issue")}
         if name == "search_postmortems":
-            return {"results": self.retriever.search(query, limit=limit, doc_type="postmortem")}
+            hits = self.retriever.search(query, limit=limit, doc_type="postmortem")
+            return {"results": self._filter_by_services(hits, services)}
         if name == "get_knowledge_stats":
             return self.retriever.stats()
         if name == "list_services":
-            services: set[str] = set()
+            type_filter = str(args.get("type", "")) or None
+            svcs: set[str] = set()
             for doc in self.retriever.store.list_documents():
-                services.update(doc.get("services", []))
-            return {"services": sorted(services)}
+                if type_filter and doc.get("type") != type_filter:
+                    continue
+                svcs.update(doc.get("services", []))
+            return {"services": sorted(svcs)}
         raise ValueError(f"unknown tool '{name}'")
 
     # -- JSON-RPC ----------------------------------------------------------------
