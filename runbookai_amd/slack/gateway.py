@@ -30,6 +30,13 @@ def parse_command(text: str) -> dict[str, Any]:
     return {"command": "ask", "args": clean, "raw": clean}
 
 
+def sign_request(signing_secret: str, timestamp: str, body: bytes) -> str:
+    """Produce a Slack v0 signature (the inverse of verify_signature;
+    used by tests and by outgoing signed requests)."""
+    base = f"v0:{timestamp}:{body.decode('utf-8', 'replace')}".encode()
+    return "v0=" + hmac.new(signing_secret.encode(), base, hashlib.sha256).hexdigest()
+
+
 def verify_signature(signing_secret: str, timestamp: str, body: bytes,
                      signature: str, max_age_s: float = 300.0) -> bool:
     """Slack v0 signature verification (reference L190-258)."""

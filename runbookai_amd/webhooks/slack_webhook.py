@@ -55,12 +55,34 @@ class PendingApprovalStore:
             return []
         out = []
         for fn in os.listdir(self.directory):
-            if fn.endswith(".json"):
+            if not fn.endswith(".json"):
+                continue
+            try:
                 with open(os.path.join(self.directory, fn), encoding="utf-8") as f:
                     data = json.load(f)
-                if data.get("status") == "pending":
-                    out.append(data)
+            except (json.JSONDecodeError, OSError):  # malformed files are skipped
+                continue
+            if isinstance(data, dict) and data.get("status") == "pending":
+                out.append(data)
         return out
+
+    def cleanup_old(self, max_age_s: float = 24 * 3600.0) -> int:
+        """Remove approval files older than max_age_s (reference
+        cleanupOldFiles, slack-webhook.ts:322-349). max_age_s=0 clears
+        everything; a missing directory removes nothing."""
+        if not os.path.isdir(self.directory):
+            return 0
+        cutoff = time.time() - max_age_s
+        removed = 0
+        for fn in os.listdir(self.directory):
+            path = os.path.join(self.directory, fn)
+            try:
+                if os.path.getmtime(path) <= cutoff:
+                    os.remove(path)
+                    removed += 1
+            except OSError:
+                continue
+        return removed
 
     def wait_for(self, approval_id: str, timeout_s: float = 300.0,
                  poll_s: float = 1.0) -> Optional[bool]:
@@ -72,6 +94,24 @@ class PendingApprovalStore:
                 return data["status"] == "approved"
             time.sleep(poll_s)
         return None
+
+
+def get_webhook_config_from_env() -> Optional[dict[str, Any]]:
+    """Webhook config from the environment (reference
+    getWebhookConfigFromEnv, slack-webhook.ts:40-76): None unless
+    SLACK_SIGNING_SECRET is set; port defaults to 3000."""
+    secret = os.environ.get("SLACK_SIGNING_SECRET", "")
+    if not secret:
+        return None
+    try:
+        port = int(os.environ.get("SLACK_WEBHOOK_PORT", "3000"))
+    except ValueError:
+        port = 3000
+    return {
+        "signingSecret": secret,
+        "port": port,
+        "pendingDir": os.environ.get("RUNBOOK_PENDING_DIR", ".runbook/pending"),
+    }
 
 
 class ApprovalWebhook:
@@ -96,13 +136,25 @@ class ApprovalWebhook:
 
         return verify_signature(self.signing_secret, timestamp, body, signature)
 
+    @staticmethod
+    def parse_action(action: dict[str, Any]) -> tuple[str, str]:
+        """(verb, approval_id) from a block action: value "approve:<id>"
+        or action_id "approve_<id>" / "deny_<id>" (reference L266-292)."""
+        value = str(action.get("value", ""))
+        if ":" in value:
+            verb, _, approval_id = value.partition(":")
+            return verb, approval_id
+        action_id = str(action.get("action_id", ""))
+        for verb in ("approve", "deny", "reject"):
+            if action_id.startswith(verb + "_"):
+                return ("deny" if verb == "reject" else verb), action_id[len(verb) + 1:]
+        return "", ""
+
     def handle_interaction(self, payload: dict[str, Any]) -> dict[str, Any]:
         actions = payload.get("actions", [])
         if not actions:
             return {"ok": False, "error": "no actions"}
-        action = actions[0]
-        value = action.get("value", "")  # "approve:<id>" / "deny:<id>"
-        verb, _, approval_id = value.partition(":")
+        verb, approval_id = self.parse_action(actions[0])
         user = payload.get("user", {}).get("username", "")
         resolved = self.store.resolve(approval_id, verb == "approve", approver=user)
         if resolved is None:

@@ -80,3 +80,148 @@ def test_webhook_rejects_unsigned_request(tmp_path):
     assert webhook.verify(ts, body, good_sig)
     # and without a secret (local dev) verification is a no-op
     assert ApprovalWebhook(store).verify(ts, body, "")
+
+
+class TestWebhookEnvConfig:
+    """Reference slack-webhook.test.ts:40-76."""
+
+    def test_none_without_secret(self, monkeypatch):
+        from runbookai_amd.webhooks.slack_webhook import get_webhook_config_from_env
+
+        monkeypatch.delenv("SLACK_SIGNING_SECRET", raising=False)
+        assert get_webhook_config_from_env() is None
+
+    def test_defaults_with_secret(self, monkeypatch):
+        from runbookai_amd.webhooks.slack_webhook import get_webhook_config_from_env
+
+        monkeypatch.setenv("SLACK_SIGNING_SECRET", "test-secret")
+        monkeypatch.delenv("SLACK_WEBHOOK_PORT", raising=False)
+        cfg = get_webhook_config_from_env()
+        assert cfg["signingSecret"] == "test-secret" and cfg["port"] == 3000
+
+    def test_custom_port_and_dir(self, monkeypatch):
+        from runbookai_amd.webhooks.slack_webhook import get_webhook_config_from_env
+
+        monkeypatch.setenv("SLACK_SIGNING_SECRET", "s")
+        monkeypatch.setenv("SLACK_WEBHOOK_PORT", "8080")
+        monkeypatch.setenv("RUNBOOK_PENDING_DIR", "/custom/path")
+        cfg = get_webhook_config_from_env()
+        assert cfg["port"] == 8080 and cfg["pendingDir"] == "/custom/path"
+
+    def test_bad_port_falls_back(self, monkeypatch):
+        from runbookai_amd.webhooks.slack_webhook import get_webhook_config_from_env
+
+        monkeypatch.setenv("SLACK_SIGNING_SECRET", "s")
+        monkeypatch.setenv("SLACK_WEBHOOK_PORT", "not-a-port")
+        assert get_webhook_config_from_env()["port"] == 3000
+
+
+class TestPendingStoreParity:
+    """Reference slack-webhook.test.ts:79-186."""
+
+    def test_empty_for_missing_and_empty_dir(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import PendingApprovalStore
+
+        assert PendingApprovalStore(str(tmp_path / "nope")).list_pending() == []
+        assert PendingApprovalStore(str(tmp_path)).list_pending() == []
+
+    def test_lists_pending_skips_resolved(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import PendingApprovalStore
+
+        store = PendingApprovalStore(str(tmp_path))
+        a = store.create({"op": "restart"})
+        b = store.create({"op": "scale"})
+        store.resolve(b, approved=True, approver="alice")
+        pending = store.list_pending()
+        assert [p["id"] for p in pending] == [a]
+
+    def test_malformed_json_skipped(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import PendingApprovalStore
+
+        store = PendingApprovalStore(str(tmp_path))
+        store.create({"op": "x"})
+        (tmp_path / "broken.json").write_text("{nope")
+        assert len(store.list_pending()) == 1
+
+    def test_cleanup_missing_dir(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import PendingApprovalStore
+
+        assert PendingApprovalStore(str(tmp_path / "nope")).cleanup_old(0) == 0
+
+    def test_cleanup_keeps_recent(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import PendingApprovalStore
+
+        store = PendingApprovalStore(str(tmp_path))
+        store.create({"op": "x"})
+        assert store.cleanup_old(max_age_s=3600) == 0
+        assert len(store.list_pending()) == 1
+
+    def test_cleanup_age_zero_removes_all(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import PendingApprovalStore
+
+        store = PendingApprovalStore(str(tmp_path))
+        store.create({"op": "x"})
+        store.create({"op": "y"})
+        assert store.cleanup_old(max_age_s=0) == 2
+        assert store.list_pending() == []
+
+    def test_resolution_record_format(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import PendingApprovalStore
+
+        store = PendingApprovalStore(str(tmp_path))
+        aid = store.create({"op": "restart"})
+        store.resolve(aid, approved=False, approver="bob")
+        data = store.get(aid)
+        assert data["status"] == "denied" and data["approver"] == "bob"
+        assert data["resolvedAt"] > 0
+
+
+class TestActionParsing:
+    """Reference slack-webhook.test.ts:266-292."""
+
+    def test_value_format(self):
+        from runbookai_amd.webhooks.slack_webhook import ApprovalWebhook
+
+        assert ApprovalWebhook.parse_action({"value": "approve:abc123"}) == ("approve", "abc123")
+        assert ApprovalWebhook.parse_action({"value": "deny:abc123"}) == ("deny", "abc123")
+
+    def test_action_id_format(self):
+        from runbookai_amd.webhooks.slack_webhook import ApprovalWebhook
+
+        assert ApprovalWebhook.parse_action({"action_id": "approve_mut_123"}) == ("approve", "mut_123")
+        assert ApprovalWebhook.parse_action({"action_id": "reject_mut_123"}) == ("deny", "mut_123")
+
+    def test_unparseable(self):
+        from runbookai_amd.webhooks.slack_webhook import ApprovalWebhook
+
+        assert ApprovalWebhook.parse_action({"action_id": "other_thing"}) == ("", "")
+
+    def test_block_actions_payload(self, tmp_path):
+        from runbookai_amd.webhooks.slack_webhook import ApprovalWebhook, PendingApprovalStore
+
+        store = PendingApprovalStore(str(tmp_path))
+        aid = store.create({"op": "restart"})
+        wh = ApprovalWebhook(store=store)
+        out = wh.handle_interaction({
+            "type": "block_actions",
+            "user": {"username": "carol"},
+            "actions": [{"action_id": f"approve_{aid}", "value": f"approve:{aid}"}],
+        })
+        assert out["ok"] and out["status"] == "approved"
+        assert store.get(aid)["approver"] == "carol"
+
+
+class TestSignatures:
+    """Reference slack-webhook.test.ts:189-218."""
+
+    def test_signature_format_and_uniqueness(self):
+        import hashlib
+        import hmac as hmac_mod
+
+        from runbookai_amd.slack.gateway import sign_request
+
+        sig1 = sign_request("secret", "123", b"body-a")
+        sig2 = sign_request("secret", "123", b"body-b")
+        sig3 = sign_request("other", "123", b"body-a")
+        assert sig1.startswith("v0=") and len(sig1) == 3 + 64
+        assert sig1 != sig2 and sig1 != sig3
