@@ -30,6 +30,32 @@ def parse_command(text: str) -> dict[str, Any]:
     return {"command": "ask", "args": clean, "raw": clean}
 
 
+_INCIDENT_RE = re.compile(r"\b((?:PD|OG|INC)-\w+)\b", re.I)
+
+
+def build_slack_request(parsed: dict[str, Any], event: dict[str, Any]) -> dict[str, Any]:
+    """Structured request from a parsed command + the raw Slack event
+    (reference buildSlackRequest, gateway.ts:108-152): derives the agent
+    query, captures an incident ID mention, and threads the reply under
+    thread_ts when the mention already lives in a thread."""
+    command = parsed["command"]
+    args = parsed.get("args", "")
+    query = args
+    if command == "deploy":
+        query = f"Deploy {args}".strip()
+    elif command == "investigate" and not args:
+        query = parsed.get("raw", "")
+    m = _INCIDENT_RE.search(args)
+    return {
+        "command": command,
+        "query": query,
+        "incidentId": m.group(1) if m else None,
+        "threadTs": event.get("thread_ts") or event.get("ts", ""),
+        "channel": event.get("channel", ""),
+        "user": event.get("user", ""),
+    }
+
+
 def sign_request(signing_secret: str, timestamp: str, body: bytes) -> str:
     """Produce a Slack v0 signature (the inverse of verify_signature;
     used by tests and by outgoing signed requests)."""
@@ -88,14 +114,18 @@ class SlackGateway:
         user = event.get("user", "")
         if not self.allowed(channel, user):
             return {"ok": False, "error": "channel or user not allowed"}
+        if self.config.get("requireThread") and not event.get("thread_ts"):
+            return {"ok": False, "error": "mentions must be in a thread"}
         event_id = event.get("event_ts", event.get("ts", str(time.time())))
         if self.dedupe(event_id):
             return {"ok": True, "deduped": True}
         parsed = parse_command(event.get("text", ""))
+        request = build_slack_request(parsed, event)
         reply = self._run_command(parsed)
-        record = {"channel": channel, "threadTs": event.get("ts", ""), "text": reply}
+        record = {"channel": channel, "threadTs": request["threadTs"], "text": reply}
         self.replies.append(record)
-        return {"ok": True, "command": parsed["command"], "reply": reply}
+        return {"ok": True, "command": parsed["command"], "reply": reply,
+                "request": request}
 
     def _run_command(self, parsed: dict[str, Any]) -> str:
         cmd, args = parsed["command"], parsed["args"]

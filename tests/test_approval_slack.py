@@ -225,3 +225,46 @@ class TestSignatures:
         sig3 = sign_request("other", "123", b"body-a")
         assert sig1.startswith("v0=") and len(sig1) == 3 + 64
         assert sig1 != sig2 and sig1 != sig3
+
+
+class TestBuildSlackRequest:
+    """Reference slack/__tests__/gateway.test.ts:35-113."""
+
+    def test_deploy_request_query(self):
+        from runbookai_amd.slack.gateway import build_slack_request, parse_command
+
+        parsed = parse_command("<@Ubot> deploy checkout-api to production")
+        req = build_slack_request(parsed, {"channel": "C123", "user": "U1", "ts": "123.4"})
+        assert req["command"] == "deploy"
+        assert "Deploy checkout-api to production" in req["query"]
+        assert req["threadTs"] == "123.4"
+
+    def test_incident_id_captured(self):
+        from runbookai_amd.slack.gateway import build_slack_request, parse_command
+
+        parsed = parse_command("<@Ubot> investigate PD-777 redis latency spike")
+        req = build_slack_request(parsed, {"ts": "123.4", "thread_ts": "100.2"})
+        assert req["incidentId"] == "PD-777"
+        assert req["threadTs"] == "100.2"  # reply stays in the thread
+
+    def test_no_incident_id(self):
+        from runbookai_amd.slack.gateway import build_slack_request, parse_command
+
+        req = build_slack_request(parse_command("investigate redis is down"), {"ts": "1"})
+        assert req["incidentId"] is None
+
+    def test_thread_required_rejection(self):
+        from runbookai_amd.slack.gateway import SlackGateway
+
+        gw = SlackGateway(config={"requireThread": True})
+        out = gw.handle_event({"channel": "C1", "user": "U1", "ts": "1.0",
+                               "text": "<@UBOT> status"})
+        assert not out["ok"] and "thread" in out["error"]
+
+    def test_threaded_mention_accepted(self):
+        from runbookai_amd.slack.gateway import SlackGateway
+
+        gw = SlackGateway(config={"requireThread": True})
+        out = gw.handle_event({"channel": "C1", "user": "U1", "ts": "2.0",
+                               "thread_ts": "1.0", "text": "<@UBOT> status"})
+        assert out["ok"] and out["request"]["threadTs"] == "1.0"
