@@ -172,3 +172,55 @@ def test_final_answer_streams_chunks(make_agent=None):
     body = "".join(c.data["text"] for c in chunks)
     assert final.data["text"].startswith(body)
     assert final.data["streamedLen"] == len(body)
+
+
+class TestCitations:
+    """Reference agent/__tests__/agent-citations.test.ts (3 cases)."""
+
+    def _dup_retriever(self):
+        class R(FakeRetriever):
+            def retrieve(self, ctx):
+                from runbookai_amd.agent.types import RetrievedKnowledge
+
+                rk = RetrievedKnowledge()
+                # same doc retrieved twice: must cite once
+                for _ in range(2):
+                    rk.runbooks.append({"title": "Redis runbook", "type": "runbook",
+                                        "content": "raise pool size", "path": "rb.md"})
+                return rk
+        return R()
+
+    def test_knowledge_fast_path_dedupes_references(self):
+        llm = MockLLMClient()
+        llm.on(r"runbook knowledge only", "Follow the Redis runbook steps [1].")
+        agent = Agent(llm=llm, tools=[], knowledge_retriever=self._dup_retriever(),
+                      config=AgentConfig(max_iterations=2))
+        final = [e for e in agent.run("how do I restart redis safely?")
+                 if e.type == EventType.ANSWER_FINAL][-1]
+        text = final.data["text"]
+        assert "## Sources" in text
+        assert text.count("Redis runbook") >= 1
+        # deduplicated: exactly one [1] source row, no [2]
+        assert "[2]" not in text.split("## Sources")[1]
+
+    def test_final_synthesis_includes_references(self):
+        log = []
+        llm = MockLLMClient()
+        llm.on(r"toolCalls", tool_call_response(("cloudwatch_alarms", {})))
+        llm.add(tool_call_response(("cloudwatch_alarms", {})))
+        llm.add(json.dumps({"content": "Redis pool exhausted."}))
+        agent = Agent(llm=llm, tools=make_tools(log), knowledge_retriever=FakeRetriever(),
+                      config=AgentConfig(max_iterations=2))
+        final = [e for e in agent.run("why is checkout slow?")
+                 if e.type == EventType.ANSWER_FINAL][-1]
+        assert "## Sources" in final.data["text"]
+
+    def test_no_references_without_knowledge(self):
+        log = []
+        llm = MockLLMClient()
+        llm.add(json.dumps({"content": "All healthy."}))
+        agent = Agent(llm=llm, tools=make_tools(log), knowledge_retriever=None,
+                      config=AgentConfig(max_iterations=2))
+        final = [e for e in agent.run("anything wrong?")
+                 if e.type == EventType.ANSWER_FINAL][-1]
+        assert "## Sources" not in final.data["text"]
