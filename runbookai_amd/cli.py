@@ -668,8 +668,9 @@ def claude() -> None:
 def claude_enable(scope: str) -> None:
     from .integrations.claude_hooks import install_hooks
 
-    path = install_hooks(scope)
-    _echo(f"{GREEN}claude hooks installed{RESET} → {path}")
+    result = install_hooks(scope)
+    _echo(f"{GREEN}claude hooks installed{RESET} ({result['addedHooks']} added) "
+          f"→ {result['settingsPath']}")
 
 
 @claude.command("status")

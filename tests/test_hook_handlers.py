@@ -205,3 +205,63 @@ class TestStdinProtocol:
 
     def test_invalid_json(self, monkeypatch):
         assert json.loads(self._run(monkeypatch, "{nope"))["continue"] is True
+
+
+class TestClaudeHookInstall:
+    """Reference integrations/__tests__/claude-hooks.test.ts:51-184."""
+
+    def test_installs_seven_events_with_matchers(self, tmp_path):
+        from runbookai_amd.integrations.claude_hooks import (
+            HOOK_COMMAND,
+            install_hooks,
+        )
+
+        result = install_hooks("project", cwd=str(tmp_path))
+        assert result["addedHooks"] == 7
+        assert result["eventsUpdated"] == [
+            "SessionStart", "UserPromptSubmit", "PreToolUse", "PostToolUse",
+            "Stop", "SubagentStop", "PreCompact"]
+        settings = json.load(open(result["settingsPath"]))
+        pre = settings["hooks"]["PreToolUse"][0]
+        assert pre["matcher"] == ".*"
+        assert pre["hooks"][0]["command"] == HOOK_COMMAND
+        assert settings["hooks"]["SessionStart"][0]["matcher"] == ""
+
+    def test_idempotent(self, tmp_path):
+        from runbookai_amd.integrations.claude_hooks import install_hooks
+
+        first = install_hooks("project", cwd=str(tmp_path))
+        second = install_hooks("project", cwd=str(tmp_path))
+        assert first["addedHooks"] == 7 and second["addedHooks"] == 0
+        settings = json.load(open(first["settingsPath"]))
+        total = sum(len(e["hooks"]) for entries in settings["hooks"].values()
+                    for e in entries)
+        assert total == 7
+
+    def test_uninstall_preserves_user_hooks(self, tmp_path):
+        import os
+
+        from runbookai_amd.integrations.claude_hooks import (
+            install_hooks,
+            uninstall_hooks,
+        )
+
+        claude_dir = tmp_path / ".claude"
+        claude_dir.mkdir()
+        user_hook = {"matcher": "", "hooks": [{"type": "command", "command": "my-own-hook"}]}
+        (claude_dir / "settings.json").write_text(json.dumps(
+            {"hooks": {"Stop": [user_hook]}}))
+        install_hooks("project", cwd=str(tmp_path))
+        removed = uninstall_hooks("project", cwd=str(tmp_path))
+        assert removed == 7
+        settings = json.load(open(claude_dir / "settings.json"))
+        assert settings["hooks"]["Stop"] == [user_hook]
+        assert "PreToolUse" not in settings["hooks"]
+
+    def test_status_reports_events(self, tmp_path):
+        from runbookai_amd.integrations.claude_hooks import hooks_status, install_hooks
+
+        install_hooks("project", cwd=str(tmp_path))
+        status = hooks_status(cwd=str(tmp_path))
+        assert status["project"]["enabled"]
+        assert len(status["project"]["installedEvents"]) == 7

@@ -187,8 +187,8 @@ class TestHooks:
         monkeypatch.chdir(tmp_path)
         from runbookai_amd.integrations import claude_hooks
 
-        path = claude_hooks.install_hooks("project")
-        assert os.path.exists(path)
+        result = claude_hooks.install_hooks("project")
+        assert os.path.exists(result["settingsPath"])
         status = claude_hooks.hooks_status()
         assert status["project"]["enabled"]
         claude_hooks.uninstall_hooks("project")
