@@ -756,6 +756,21 @@ def deploy(ctx: click.Context, service: str, version: str) -> None:
     _echo(json.dumps(out, indent=1, default=str))
 
 
+@cli.command()
+@click.option("--model", default="tiny", help="engine model config or checkpoint name")
+@click.option("--checkpoint", default=None, help="path to a trained checkpoint dir")
+@click.option("--host", default="127.0.0.1")
+@click.option("--port", default=8000, type=int)
+@click.option("--tp", default=None, type=int, help="tensor-parallel degree")
+def serve(model: str, checkpoint: str, host: str, port: int, tp: int) -> None:
+    """OpenAI-compatible serving endpoint over the local engine
+    (/v1/chat/completions, /v1/completions, /metrics)."""
+    from .engine.server import serve as _serve
+
+    _echo(f"{GREEN}serving{RESET} {checkpoint or model} on http://{host}:{port}/v1")
+    _serve(model=model, host=host, port=port, checkpoint=checkpoint, tp=tp)
+
+
 def main() -> None:
     cli(obj={})
 

@@ -1,0 +1,147 @@
+"""OpenAI-compatible serving endpoint tests (engine/server.py): protocol
+shapes, grammar-constrained response_format, SSE streaming, metrics,
+error handling. Runs on CPU with the tiny model."""
+from __future__ import annotations
+
+import json
+
+import pytest
+
+from runbookai_amd.engine.engine import LLMEngine
+from runbookai_amd.engine.server import ServingAdapter, create_app
+
+
+@pytest.fixture(scope="module")
+def engine():
+    eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=256)
+    yield eng
+    eng.shutdown()
+
+
+@pytest.fixture(scope="module")
+def adapter(engine):
+    return ServingAdapter(engine, model_name="tiny")
+
+
+@pytest.fixture(scope="module")
+def client(engine):
+    from fastapi.testclient import TestClient
+
+    return TestClient(create_app(engine=engine, model_name="tiny"))
+
+
+class TestAdapter:
+    def test_models(self, adapter):
+        out = adapter.models()
+        assert out["object"] == "list" and out["data"][0]["id"] == "tiny"
+
+    def test_completion_shape(self, adapter):
+        out = adapter.completion({"prompt": "hello", "max_tokens": 8})
+        assert out["object"] == "text_completion"
+        assert out["choices"][0]["index"] == 0
+        assert isinstance(out["choices"][0]["text"], str)
+        u = out["usage"]
+        assert u["total_tokens"] == u["prompt_tokens"] + u["completion_tokens"]
+        assert u["completion_tokens"] <= 8
+
+    def test_chat_completion_shape(self, adapter):
+        out = adapter.chat_completion({
+            "messages": [{"role": "system", "content": "be brief"},
+                         {"role": "user", "content": "hi"}],
+            "max_tokens": 8})
+        msg = out["choices"][0]["message"]
+        assert msg["role"] == "assistant" and isinstance(msg["content"], str)
+        assert out["object"] == "chat.completion"
+
+    def test_multi_turn_folding(self, adapter, engine):
+        ids = adapter._encode_chat([
+            {"role": "system", "content": "sys"},
+            {"role": "user", "content": "first"},
+            {"role": "assistant", "content": "reply"},
+            {"role": "user", "content": "second"}])
+        text = engine.tokenizer.decode(ids)
+        assert "first" in text and "reply" in text and "second" in text
+
+    def test_json_object_response_format(self, adapter):
+        out = adapter.chat_completion({
+            "messages": [{"role": "user", "content": "give me json"}],
+            "response_format": {"type": "json_object"},
+            "max_tokens": 64})
+        parsed = json.loads(out["choices"][0]["message"]["content"])
+        assert isinstance(parsed, dict)
+
+    def test_json_schema_response_format(self, adapter):
+        schema = {"type": "object",
+                  "properties": {"status": {"type": "string"}},
+                  "required": ["status"]}
+        out = adapter.chat_completion({
+            "messages": [{"role": "user", "content": "status?"}],
+            "response_format": {"type": "json_schema",
+                                "json_schema": {"name": "s", "schema": schema}},
+            "max_tokens": 64})
+        parsed = json.loads(out["choices"][0]["message"]["content"])
+        assert "status" in parsed and isinstance(parsed["status"], str)
+
+    def test_empty_messages_rejected(self, adapter):
+        with pytest.raises(ValueError):
+            adapter.chat_completion({"messages": []})
+
+
+class TestHttp:
+    def test_models_endpoint(self, client):
+        r = client.get("/v1/models")
+        assert r.status_code == 200 and r.json()["data"][0]["id"] == "tiny"
+
+    def test_chat_endpoint(self, client):
+        r = client.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "hello"}], "max_tokens": 8})
+        assert r.status_code == 200
+        assert r.json()["choices"][0]["message"]["role"] == "assistant"
+
+    def test_completions_endpoint(self, client):
+        r = client.post("/v1/completions", json={"prompt": "abc", "max_tokens": 4})
+        assert r.status_code == 200
+        assert r.json()["usage"]["completion_tokens"] <= 4
+
+    def test_bad_json_400(self, client):
+        r = client.post("/v1/chat/completions",
+                        content=b"{nope", headers={"content-type": "application/json"})
+        assert r.status_code == 400
+        assert r.json()["error"]["type"] == "invalid_request_error"
+
+    def test_missing_messages_400(self, client):
+        r = client.post("/v1/chat/completions", json={"messages": []})
+        assert r.status_code == 400
+
+    def test_streaming_chat(self, client):
+        with client.stream("POST", "/v1/chat/completions", json={
+                "messages": [{"role": "user", "content": "hi"}],
+                "max_tokens": 6, "stream": True}) as r:
+            assert r.status_code == 200
+            assert "text/event-stream" in r.headers["content-type"]
+            body = "".join(r.iter_text())
+        frames = [l[6:] for l in body.splitlines() if l.startswith("data: ")]
+        assert frames[-1] == "[DONE]"
+        chunks = [json.loads(f) for f in frames[:-1]]
+        assert chunks[0]["choices"][0]["delta"].get("role") == "assistant"
+        assert chunks[-1]["choices"][0]["finish_reason"] == "stop"
+        assert chunks[-1]["usage"]["completion_tokens"] <= 6
+        # the concatenated deltas equal a non-streamed generation's shape
+        text = "".join(c["choices"][0]["delta"].get("content", "") for c in chunks)
+        assert isinstance(text, str)
+
+    def test_streaming_completions(self, client):
+        with client.stream("POST", "/v1/completions", json={
+                "prompt": "xyz", "max_tokens": 4, "stream": True}) as r:
+            body = "".join(r.iter_text())
+        assert body.rstrip().endswith("data: [DONE]")
+
+    def test_metrics_endpoint(self, client):
+        r = client.get("/metrics")
+        assert r.status_code == 200 and "runbook" in r.text.lower() or "engine" in r.text
+
+    def test_healthz(self, client):
+        r = client.get("/healthz")
+        out = r.json()
+        assert out["status"] == "ok" and out["model"] == "tiny"
+        assert "waiting" in out and "running" in out
