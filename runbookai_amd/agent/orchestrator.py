@@ -130,6 +130,8 @@ class InvestigationOrchestrator:
         self.queries_per_hypothesis = queries_per_hypothesis
         self._listeners: list[Callable[[InvestigationEvent], None]] = []
         self.phases_visited: list[str] = []
+        self.inferred_log_group: str = ""
+        self.inferred_lambda: str = ""
         self.stats = {"llm_calls": 0, "tool_calls": 0, "evaluations": 0, "hypotheses": 0}
         self.machine.on("phase_change", lambda d: self._on_phase(d))
 
@@ -420,6 +422,55 @@ class InvestigationOrchestrator:
                    confidence=evaluation["confidence"], branched=len(created))
         return evaluation["action"] == "confirm"
 
+    # -- CloudWatch/Lambda hint inference (reference L233-362) -----------------
+
+    def _update_cloudwatch_hints(self, tool: str, params: dict[str, Any],
+                                 result: Any) -> None:
+        """Learn the serverless context from passing traffic: an explicit
+        log_group param, a Lambda FunctionName in alarm dimensions, or an
+        aws_query lambda listing each pin the log group later
+        cloudwatch_logs queries should target."""
+        log_group = str(params.get("log_group", "") or "").strip()
+        if log_group:
+            self.inferred_log_group = log_group
+            if log_group.startswith("/aws/lambda/"):
+                self.inferred_lambda = log_group[len("/aws/lambda/"):]
+            return
+        if not isinstance(result, dict):
+            return
+        if tool in ("cloudwatch_alarms", "cloudwatch_logs"):
+            for alarm in result.get("alarms", []) or []:
+                if not isinstance(alarm, dict):
+                    continue
+                dims = alarm.get("dimensions", {}) or {}
+                fn = dims.get("FunctionName") if isinstance(dims, dict) else None
+                if fn:
+                    self.inferred_lambda = str(fn)
+                    self.inferred_log_group = f"/aws/lambda/{fn}"
+                    return
+        elif tool == "aws_query":
+            lam = (result.get("results", {}) or {}).get("lambda") \
+                if isinstance(result.get("results"), dict) else None
+            items = lam.get("items", []) if isinstance(lam, dict) else []
+            for item in items or []:
+                name = item.get("FunctionName") or item.get("name") \
+                    if isinstance(item, dict) else None
+                if name:
+                    self.inferred_lambda = str(name)
+                    self.inferred_log_group = f"/aws/lambda/{name}"
+                    return
+
+    def _apply_hints(self, query: CausalQuery) -> CausalQuery:
+        """Target log queries at the inferred Lambda log group when the
+        canned query didn't name one."""
+        if (query.tool == "cloudwatch_logs" and self.inferred_log_group
+                and not query.params.get("log_group")):
+            return CausalQuery(tool=query.tool,
+                               params={**query.params,
+                                       "log_group": self.inferred_log_group},
+                               purpose=query.purpose, priority=query.priority)
+        return query
+
     def _execute_queries_for_hypothesis(self, h: Hypothesis) -> list[dict[str, Any]]:
         """Reference executeQueriesForHypothesis (L937-1000)."""
         m = self.machine
@@ -436,7 +487,9 @@ class InvestigationOrchestrator:
             adapted = self.adapt_query_to_environment(q)
             if adapted is None:
                 continue
+            adapted = self._apply_hints(adapted)
             result, error = self._execute(adapted.tool, adapted.params)
+            self._update_cloudwatch_hints(adapted.tool, adapted.params, result)
             m.record_query_result(h.id, adapted.tool, adapted.params, result, error)
             results.append({"tool": adapted.tool, "purpose": adapted.purpose,
                             "result": result, "error": error})

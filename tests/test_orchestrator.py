@@ -271,3 +271,35 @@ def test_max_iterations_bounds_loop():
     result = orch.investigate("mystery")
     assert result.success
     assert orch.machine.iteration <= 3
+
+
+def test_cloudwatch_hint_inference_targets_lambda_log_group():
+    """Reference updateCloudWatchHints (L233-362): a Lambda FunctionName in
+    alarm dimensions pins later cloudwatch_logs queries to its log group."""
+    llm = scripted_llm()
+    tools = MockToolExecutor(overrides={
+        "cloudwatch_alarms": {"alarms": [{
+            "name": "lambda-errors", "state": "ALARM",
+            "dimensions": {"FunctionName": "checkout-worker"},
+            "reason": "Errors > 5"}]},
+    })
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=tools)
+    orch.investigate("lambda checkout-worker erroring")
+    assert orch.inferred_lambda == "checkout-worker"
+    assert orch.inferred_log_group == "/aws/lambda/checkout-worker"
+    log_calls = [p for t, p in tools.calls if t == "cloudwatch_logs"]
+    assert any(p.get("log_group") == "/aws/lambda/checkout-worker" for p in log_calls)
+
+
+def test_explicit_log_group_param_remembered():
+    orch = InvestigationOrchestrator(llm=MockLLMClient(), tool_executor=MockToolExecutor())
+    orch._update_cloudwatch_hints("cloudwatch_logs",
+                                  {"log_group": "/aws/lambda/billing-fn"}, {})
+    assert orch.inferred_lambda == "billing-fn"
+
+
+def test_aws_query_lambda_listing_sets_hint():
+    orch = InvestigationOrchestrator(llm=MockLLMClient(), tool_executor=MockToolExecutor())
+    orch._update_cloudwatch_hints("aws_query", {}, {
+        "results": {"lambda": {"items": [{"FunctionName": "img-resize"}]}}})
+    assert orch.inferred_log_group == "/aws/lambda/img-resize"
