@@ -476,16 +476,23 @@ def checkpoint() -> None:
 
 @checkpoint.command("list")
 @click.argument("investigation_id", required=False)
-def checkpoint_list(investigation_id: Optional[str]) -> None:
-    from .session.checkpoint import CheckpointStore
+@click.option("--markdown", is_flag=True, help="render as a markdown table")
+def checkpoint_list(investigation_id: Optional[str], markdown: bool) -> None:
+    from .session.checkpoint import CheckpointStore, format_checkpoint_list_markdown
 
     store = CheckpointStore()
     if investigation_id is None:
-        for inv in store.list_investigations():
-            cps = store.list(inv)
-            _echo(f"{inv}: {len(cps)} checkpoints")
+        for s in store.investigations_summary():
+            latest = s["latest"]
+            _echo(f"{s['investigationId']}: {s['checkpointCount']} checkpoints "
+                  f"(latest {latest['checkpointId']} · {latest['phase']} · "
+                  f"{latest['hypothesisCount']} hypotheses)")
         return
-    for cp in store.list(investigation_id):
+    cps = store.list(investigation_id)
+    if markdown:
+        _echo(format_checkpoint_list_markdown(cps))
+        return
+    for cp in cps:
         _echo(cp.format())
 
 

@@ -145,3 +145,45 @@ class TestHttp:
         out = r.json()
         assert out["status"] == "ok" and out["model"] == "tiny"
         assert "waiting" in out and "running" in out
+
+
+class TestConcurrentServing:
+    """Background engine + parallel HTTP clients: requests batch together
+    in the continuous-batching loop and all complete."""
+
+    def test_parallel_requests_background_engine(self):
+        from concurrent.futures import ThreadPoolExecutor
+
+        from fastapi.testclient import TestClient
+
+        eng = LLMEngine(model="tiny", device="cpu", background=True, kv_blocks=256)
+        try:
+            client = TestClient(create_app(engine=eng, model_name="tiny"))
+
+            def one(i):
+                r = client.post("/v1/chat/completions", json={
+                    "messages": [{"role": "user", "content": f"request {i}"}],
+                    "max_tokens": 6})
+                return r.status_code, r.json()
+
+            with ThreadPoolExecutor(max_workers=6) as pool:
+                results = list(pool.map(one, range(6)))
+            assert all(code == 200 for code, _ in results)
+            assert all(out["choices"][0]["message"]["role"] == "assistant"
+                       for _, out in results)
+            assert eng.stats["requests"] >= 6
+        finally:
+            eng.shutdown()
+
+    def test_streaming_with_background_engine(self):
+        from fastapi.testclient import TestClient
+
+        eng = LLMEngine(model="tiny", device="cpu", background=True, kv_blocks=256)
+        try:
+            client = TestClient(create_app(engine=eng, model_name="tiny"))
+            with client.stream("POST", "/v1/completions", json={
+                    "prompt": "abc", "max_tokens": 5, "stream": True}) as r:
+                body = "".join(r.iter_text())
+            assert body.rstrip().endswith("data: [DONE]")
+        finally:
+            eng.shutdown()
