@@ -484,3 +484,54 @@ class TestHttpOperabilityAdapter:
             assert not adapter.dispatch(bad)
         finally:
             srv.shutdown()
+
+
+class TestClaudeSessionIngestion:
+    """Reference learning/__tests__/claude-session-ingestion.test.ts (3 cases)."""
+
+    EVENTS = [
+        {"kind": "session_start", "promptCount": 0},
+        {"kind": "tool_use", "tool_name": "Bash",
+         "tool_input": {"command": "kubectl get pods"}},
+        {"kind": "tool_use", "tool_name": "Bash",
+         "tool_input": {"command": "aws ecs describe-services"}},
+        {"kind": "stop"},
+    ]
+
+    def test_events_converted(self):
+        from runbookai_amd.learning.claude_session_ingestion import events_to_learning_events
+
+        out = events_to_learning_events(self.EVENTS)
+        kinds = [e["type"] for e in out]
+        assert kinds == ["tool", "tool", "session_end"]
+        assert out[0]["input"]["command"] == "kubectl get pods"
+
+    def test_synthesized_metadata(self):
+        from runbookai_amd.learning.claude_session_ingestion import (
+            events_to_learning_events,
+            synthesize_result,
+        )
+
+        le = events_to_learning_events(self.EVENTS)
+        result = synthesize_result("sess-42", le)
+        assert result["investigationId"] == "claude-sess-42"
+        assert "Bash" in result["summary"]
+        assert "kubectl get pods" in result["evidence"]
+        assert result["success"]
+
+    def test_learning_loop_from_session(self, tmp_path):
+        from runbookai_amd.integrations.session_store import SessionStore
+        from runbookai_amd.learning.claude_session_ingestion import ingest_session
+        from runbookai_amd.model.client import MockLLMClient
+
+        store = SessionStore(directory=str(tmp_path / "sessions"))
+        for e in self.EVENTS:
+            store.append_event("sess-42", e)
+        llm = MockLLMClient([json.dumps({
+            "postmortem": {"title": "PM: session review", "summary": "s",
+                           "rootCause": "manual ops session", "timeline": [],
+                           "impact": "", "actionItems": []},
+            "knowledgeSuggestions": []})])
+        out = ingest_session(store, "sess-42", llm,
+                             runbook_dir=str(tmp_path / ".runbook"))
+        assert os.path.exists(out["postmortemPath"])
