@@ -81,6 +81,7 @@ class Request:
     registered: bool = False         # prompt blocks published to the prefix pool
     dedup_deferred: bool = False     # held back one round behind a same-prefix twin
     state: str = "waiting"           # waiting | running | done
+    cancelled: bool = False          # swept at the next step boundary
     done_event: threading.Event = field(default_factory=threading.Event)
     prompt_len: int = 0
     pos: int = 0                     # next position to write
@@ -278,6 +279,29 @@ class LLMEngine:
                 req.done_event.set()
             self.running = [r for r in self.running if r not in batch]
 
+    def cancel(self, req: Request) -> bool:
+        """Cancel a request (e.g. the HTTP client disconnected). Waiting
+        requests finalize immediately; running ones are swept at the next
+        step boundary, where their KV blocks are freed. Returns False if
+        the request had already finished."""
+        with self._lock:
+            if req.state == "done":
+                return False
+            req.cancelled = True
+            if req in self.waiting:
+                self.waiting.remove(req)
+                self._finalize_cancel(req)
+                return True
+            self._lock.notify_all()
+        return True
+
+    def _finalize_cancel(self, req: Request) -> None:
+        """Caller holds the lock (or the request is step-private)."""
+        req.state = "done"
+        req.finished_at = time.time()
+        self.model.kv.free(req.rid)
+        req.done_event.set()
+
     def shutdown(self) -> None:
         self._stop = True
         with self._lock:
@@ -314,6 +338,10 @@ class LLMEngine:
         step over all running sequences, each feeding the next pending
         token (sampling only where the queue drained)."""
         with self._lock:
+            # sweep cancellations at the step boundary, never mid-batch
+            for r in [r for r in self.running if r.cancelled and r.state != "done"]:
+                self.running.remove(r)
+                self._finalize_cancel(r)
             prefill_batch = self._admit_locked()
         if prefill_batch:
             self._step_batch = prefill_batch

@@ -174,20 +174,25 @@ class ServingAdapter:
         tok = self._tok()
         emitted = ""
         background = self.engine._thread is not None
-        while True:
-            if not background:
-                # step the engine inline until this request finishes
-                self.engine.run_until_idle()
-            done = req.done_event.is_set() or req.state == "done"
-            text = tok.decode(req.out_ids)
-            if len(text) > len(emitted):
-                yield text[len(emitted):]
-                emitted = text
-            if done:
-                if req.error:
-                    raise RuntimeError(req.error)
-                return
-            time.sleep(poll_s)
+        try:
+            while True:
+                if not background:
+                    # step the engine inline until this request finishes
+                    self.engine.run_until_idle()
+                done = req.done_event.is_set() or req.state == "done"
+                text = tok.decode(req.out_ids)
+                if len(text) > len(emitted):
+                    yield text[len(emitted):]
+                    emitted = text
+                if done:
+                    if req.error:
+                        raise RuntimeError(req.error)
+                    return
+                time.sleep(poll_s)
+        except GeneratorExit:
+            # client disconnected mid-stream: stop generating, free the KV
+            self.engine.cancel(req)
+            raise
 
     def completion_stream(self, body: dict[str, Any]) -> Iterator[str]:
         prompt = body.get("prompt", "")

@@ -364,3 +364,63 @@ class TestShutdown:
         for r in reqs:
             assert r.done_event.wait(timeout=10)
             assert r.state == "done"
+
+
+class TestCancellation:
+    def test_cancel_waiting_request(self):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            req = eng.submit([1, 2, 3], max_new_tokens=64)
+            assert eng.cancel(req)
+            assert req.state == "done" and req.done_event.is_set()
+            assert req not in eng.waiting
+            # pool untouched; a fresh request still runs fine
+            done = eng.generate([4, 5, 6], max_new_tokens=4)
+            assert done.state == "done" and not done.error
+        finally:
+            eng.shutdown()
+
+    def test_cancel_running_request_frees_kv(self):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            free0 = eng.model.kv.free_blocks
+            req = eng.submit([1, 2, 3, 4], max_new_tokens=512)
+            eng.step()            # admit + prefill: now running with KV
+            assert req in eng.running
+            assert eng.model.kv.free_blocks < free0
+            assert eng.cancel(req)
+            eng.step()            # sweep happens at the step boundary
+            assert req.state == "done"
+            assert req not in eng.running
+            assert eng.model.kv.free_blocks == free0
+        finally:
+            eng.shutdown()
+
+    def test_cancel_finished_request_returns_false(self):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            req = eng.generate([1, 2], max_new_tokens=2)
+            assert not eng.cancel(req)
+        finally:
+            eng.shutdown()
+
+    def test_cancel_with_background_thread(self):
+        import time as _t
+
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=True, kv_blocks=128)
+        try:
+            req = eng.submit(list(range(16)), max_new_tokens=4096)
+            _t.sleep(0.05)        # let the loop admit it
+            eng.cancel(req)
+            assert req.done_event.wait(timeout=10.0)
+            assert req.state == "done"
+        finally:
+            eng.shutdown()
