@@ -150,6 +150,9 @@ class LLMEngine:
         self.stats = {"requests": 0, "prefill_tokens": 0, "decode_tokens": 0,
                       "steps": 0, "prefill_time": 0.0, "decode_time": 0.0,
                       "cached_prefix_tokens": 0}
+        from collections import deque as _deque
+        #: (ttft_s, e2e_s, out_tokens) of recently finished requests
+        self._latency_ring: "_deque" = _deque(maxlen=512)
         self._stop = False
         from collections import deque
         self._ev_queue: "deque" = deque()      # (phase, ev0, ev1, ev2) pending fold
@@ -759,6 +762,7 @@ class LLMEngine:
             if finished:
                 req.state = "done"
                 req.finished_at = time.time()
+                self._record_latency(req)
                 self.model.kv.free(req.rid)
                 req.done_event.set()
             return
@@ -802,10 +806,39 @@ class LLMEngine:
         if finished:
             req.state = "done"
             req.finished_at = time.time()
+            self._record_latency(req)
             self.model.kv.free(req.rid)
             req.done_event.set()
 
+    def _record_latency(self, req: Request) -> None:
+        if req.finished_at and req.submitted_at:
+            ttft = (req.first_token_at - req.submitted_at) \
+                if req.first_token_at else 0.0
+            self._latency_ring.append(
+                (ttft, req.finished_at - req.submitted_at, len(req.out_ids)))
+
     # -- metrics ----------------------------------------------------------------------
+
+    def latency_stats(self) -> dict[str, Any]:
+        """TTFT / end-to-end percentiles over the last ≤512 finished
+        requests (serving SLO view; exposed via /healthz and /metrics)."""
+        samples = list(self._latency_ring)
+        if not samples:
+            return {"samples": 0}
+
+        def pct(vals: list[float], q: float) -> float:
+            vals = sorted(vals)
+            return vals[min(len(vals) - 1, int(q * len(vals)))]
+
+        ttft = [s_[0] for s_ in samples if s_[0] > 0]
+        e2e = [s_[1] for s_ in samples]
+        out = {"samples": len(samples),
+               "e2e_p50_s": round(pct(e2e, 0.50), 4),
+               "e2e_p95_s": round(pct(e2e, 0.95), 4)}
+        if ttft:
+            out["ttft_p50_s"] = round(pct(ttft, 0.50), 4)
+            out["ttft_p95_s"] = round(pct(ttft, 0.95), 4)
+        return out
 
     def throughput_stats(self) -> dict[str, Any]:
         s = dict(self.stats)

@@ -52,6 +52,19 @@ class EngineCollector:
             g.add_metric([], float(len(eng.waiting) + len(eng.running)))
         yield g
 
+        lat = eng.latency_stats() if hasattr(eng, "latency_stats") else {}
+        if lat.get("samples"):
+            lg = GaugeMetricFamily("runbook_engine_latency_seconds",
+                                   "request latency percentiles (recent window)",
+                                   labels=["kind", "quantile"])
+            for key, (kind, q) in (("ttft_p50_s", ("ttft", "0.5")),
+                                   ("ttft_p95_s", ("ttft", "0.95")),
+                                   ("e2e_p50_s", ("e2e", "0.5")),
+                                   ("e2e_p95_s", ("e2e", "0.95"))):
+                if key in lat:
+                    lg.add_metric([kind, q], float(lat[key]))
+            yield lg
+
         kv = eng.model.kv
         pool = GaugeMetricFamily("runbook_engine_kv_blocks", "KV pool blocks",
                                  labels=["state"])

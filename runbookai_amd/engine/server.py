@@ -246,7 +246,8 @@ class ServingAdapter:
             waiting, running = len(self.engine.waiting), len(self.engine.running)
         return {"status": "ok", "model": self.model_name,
                 "waiting": waiting, "running": running,
-                "device": self.engine.device}
+                "device": self.engine.device,
+                "latency": self.engine.latency_stats()}
 
 
 def create_app(engine: Optional[LLMEngine] = None, model: str = "tiny",

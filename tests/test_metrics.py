@@ -47,3 +47,42 @@ def test_cli_help_lists_metrics():
     r = CliRunner().invoke(cli, ["metrics", "serve", "--help"])
     assert r.exit_code == 0
     assert "/metrics" in r.output or "Prometheus" in r.output
+
+
+class TestLatencyStats:
+    def test_percentiles_after_requests(self):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            for _ in range(4):
+                eng.generate([1, 2, 3], max_new_tokens=4)
+            lat = eng.latency_stats()
+            assert lat["samples"] == 4
+            assert lat["e2e_p50_s"] > 0
+            assert lat["e2e_p95_s"] >= lat["e2e_p50_s"]
+            assert "ttft_p50_s" in lat
+        finally:
+            eng.shutdown()
+
+    def test_empty_ring(self):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            assert eng.latency_stats() == {"samples": 0}
+        finally:
+            eng.shutdown()
+
+    def test_latency_in_scrape(self):
+        from runbookai_amd.engine.engine import LLMEngine
+        from runbookai_amd.engine.metrics import render_metrics
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            eng.generate([1, 2], max_new_tokens=2)
+            text = render_metrics(eng).decode()
+            assert "runbook_engine_latency_seconds" in text
+            assert 'kind="e2e"' in text
+        finally:
+            eng.shutdown()
