@@ -90,6 +90,22 @@ class HttpAdapter(BaseAdapter):
         return bool(payload.get("accepted", payload.get("ok", True)))
 
 
+#: plugin adapter constructors by kind (reference
+#: providers/operability-context/registry.ts): a registered kind wins over
+#: nothing but never shadows the built-ins below.
+_ADAPTER_REGISTRY: dict[str, Callable[[dict[str, Any]], BaseAdapter]] = {}
+
+
+def register_adapter_kind(kind: str,
+                          factory: Callable[[dict[str, Any]], BaseAdapter]) -> None:
+    _ADAPTER_REGISTRY[kind] = factory
+
+
+def registered_adapter_kinds() -> list[str]:
+    return sorted(set(_ADAPTER_REGISTRY)
+                  | {"file", "custom", "http", "sourcegraph", "entireio", "webhook"})
+
+
 def create_adapter(config: dict[str, Any]) -> BaseAdapter:
     kind = config.get("kind", "file")
     if kind == "file":
@@ -110,4 +126,6 @@ def create_adapter(config: dict[str, Any]) -> BaseAdapter:
                            headers=headers,
                            timeout_s=float(config.get("timeoutS", 10.0)),
                            name=kind)
+    if kind in _ADAPTER_REGISTRY:
+        return _ADAPTER_REGISTRY[kind](config)
     raise ValueError(f"unknown operability-context adapter kind '{kind}'")

@@ -535,3 +535,39 @@ class TestClaudeSessionIngestion:
         out = ingest_session(store, "sess-42", llm,
                              runbook_dir=str(tmp_path / ".runbook"))
         assert os.path.exists(out["postmortemPath"])
+
+
+class TestAdapterRegistry:
+    """Reference providers/operability-context/registry.ts: plugin adapter
+    kinds resolvable by name."""
+
+    def test_register_and_create(self):
+        from runbookai_amd.providers.operability_context.factory import (
+            BaseAdapter,
+            create_adapter,
+            register_adapter_kind,
+            registered_adapter_kinds,
+        )
+
+        calls = []
+
+        class MyAdapter(BaseAdapter):
+            def dispatch(self, claim):
+                calls.append(claim)
+                return True
+
+        register_adapter_kind("my-backend", lambda cfg: MyAdapter())
+        assert "my-backend" in registered_adapter_kinds()
+        adapter = create_adapter({"kind": "my-backend"})
+        from runbookai_amd.providers.operability_context.types import AgentChangeClaim
+
+        assert adapter.dispatch(AgentChangeClaim(claim_id="c1", agent="a"))
+        assert calls
+
+    def test_unknown_kind_still_raises(self):
+        import pytest as _pytest
+
+        from runbookai_amd.providers.operability_context.factory import create_adapter
+
+        with _pytest.raises(ValueError):
+            create_adapter({"kind": "nope-backend"})
