@@ -424,3 +424,36 @@ class TestCancellation:
             assert req.state == "done"
         finally:
             eng.shutdown()
+
+
+class TestCancelFuzz:
+    def test_random_submit_cancel_no_kv_leak(self):
+        """Random interleaving of submit/cancel/step never leaks KV blocks:
+        after draining, the pool is back to its starting size."""
+        import random
+
+        from runbookai_amd.engine.engine import LLMEngine
+
+        rng = random.Random(7)
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            free0 = eng.model.kv.free_blocks
+            live = []
+            for i in range(60):
+                op = rng.random()
+                if op < 0.45:
+                    live.append(eng.submit(
+                        [rng.randrange(200) for _ in range(rng.randrange(1, 40))],
+                        max_new_tokens=rng.randrange(1, 24)))
+                elif op < 0.7 and live:
+                    eng.cancel(live.pop(rng.randrange(len(live))))
+                else:
+                    eng.step()
+            eng.run_until_idle()
+            for r in live:
+                assert r.done_event.is_set()
+            # retired prefix blocks live in the reusable pool; free_blocks
+            # counts them, so the pool must be whole again
+            assert eng.model.kv.free_blocks == free0
+        finally:
+            eng.shutdown()
