@@ -251,9 +251,12 @@ class ServingAdapter:
 
 
 def create_app(engine: Optional[LLMEngine] = None, model: str = "tiny",
-               model_name: Optional[str] = None, **engine_kwargs: Any):
+               model_name: Optional[str] = None, api_key: str = "",
+               **engine_kwargs: Any):
     """FastAPI app over a ServingAdapter. Engine is built lazily from
-    `model`/`engine_kwargs` when not passed in."""
+    `model`/`engine_kwargs` when not passed in. With `api_key` set, the
+    /v1 endpoints require `Authorization: Bearer <key>` (OpenAI wire
+    convention); /healthz and /metrics stay open for probes/scrapes."""
     from fastapi import FastAPI
     from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
 
@@ -268,12 +271,28 @@ def create_app(engine: Optional[LLMEngine] = None, model: str = "tiny",
                             content={"error": {"message": message,
                                                "type": "invalid_request_error"}})
 
+    def _auth_fail(request: HttpRequest) -> Optional[JSONResponse]:
+        if not api_key:
+            return None
+        import hmac as _hmac
+
+        header = request.headers.get("authorization", "")
+        token = header[7:] if header.startswith("Bearer ") else ""
+        if _hmac.compare_digest(token, api_key):
+            return None
+        return JSONResponse(status_code=401,
+                            content={"error": {"message": "invalid API key",
+                                               "type": "authentication_error"}})
+
     @app.get("/v1/models")
-    def models():
-        return adapter.models()
+    def models(request: HttpRequest):
+        return _auth_fail(request) or adapter.models()
 
     @app.post("/v1/completions")
     async def completions(request: HttpRequest):
+        denied = _auth_fail(request)
+        if denied is not None:
+            return denied
         try:
             body = await request.json()
         except Exception:  # noqa: BLE001
@@ -290,6 +309,9 @@ def create_app(engine: Optional[LLMEngine] = None, model: str = "tiny",
 
     @app.post("/v1/chat/completions")
     async def chat_completions(request: HttpRequest):
+        denied = _auth_fail(request)
+        if denied is not None:
+            return denied
         try:
             body = await request.json()
         except Exception:  # noqa: BLE001

@@ -187,3 +187,38 @@ class TestConcurrentServing:
             assert body.rstrip().endswith("data: [DONE]")
         finally:
             eng.shutdown()
+
+
+class TestApiKeyAuth:
+    @pytest.fixture(scope="class")
+    def auth_client(self):
+        from fastapi.testclient import TestClient
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            yield TestClient(create_app(engine=eng, model_name="tiny",
+                                        api_key="sk-test-123"))
+        finally:
+            eng.shutdown()
+
+    def test_missing_key_401(self, auth_client):
+        r = auth_client.post("/v1/chat/completions", json={
+            "messages": [{"role": "user", "content": "hi"}]})
+        assert r.status_code == 401
+        assert r.json()["error"]["type"] == "authentication_error"
+
+    def test_wrong_key_401(self, auth_client):
+        r = auth_client.get("/v1/models",
+                            headers={"Authorization": "Bearer sk-wrong"})
+        assert r.status_code == 401
+
+    def test_right_key_passes(self, auth_client):
+        r = auth_client.post(
+            "/v1/chat/completions",
+            headers={"Authorization": "Bearer sk-test-123"},
+            json={"messages": [{"role": "user", "content": "hi"}], "max_tokens": 4})
+        assert r.status_code == 200
+
+    def test_health_and_metrics_open(self, auth_client):
+        assert auth_client.get("/healthz").status_code == 200
+        assert auth_client.get("/metrics").status_code == 200
