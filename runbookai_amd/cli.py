@@ -769,13 +769,17 @@ def deploy(ctx: click.Context, service: str, version: str) -> None:
 @click.option("--host", default="127.0.0.1")
 @click.option("--port", default=8000, type=int)
 @click.option("--tp", default=None, type=int, help="tensor-parallel degree")
-def serve(model: str, checkpoint: str, host: str, port: int, tp: int) -> None:
+@click.option("--api-key", default="", envvar="RUNBOOK_API_KEY",
+              help="require Authorization: Bearer <key> on /v1 endpoints")
+def serve(model: str, checkpoint: str, host: str, port: int, tp: int,
+          api_key: str) -> None:
     """OpenAI-compatible serving endpoint over the local engine
     (/v1/chat/completions, /v1/completions, /metrics)."""
     from .engine.server import serve as _serve
 
     _echo(f"{GREEN}serving{RESET} {checkpoint or model} on http://{host}:{port}/v1")
-    _serve(model=model, host=host, port=port, checkpoint=checkpoint, tp=tp)
+    _serve(model=model, host=host, port=port, checkpoint=checkpoint, tp=tp,
+           api_key=api_key)
 
 
 def main() -> None:
