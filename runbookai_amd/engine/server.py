@@ -145,10 +145,60 @@ class ServingAdapter:
             "usage": self._usage(req),
         }
 
+    def _chat_with_tools(self, body: dict[str, Any],
+                         messages: list[dict[str, Any]]) -> dict[str, Any]:
+        """OpenAI tool calling via the engine's two-stage constrained
+        decode (engine/client.py _chat_with_tools): stage 1 picks
+        tool-vs-final from an enum grammar, stage 2 fills that tool's
+        argument schema — real tool calls from any checkpoint."""
+        from .client import LocalEngineClient
+
+        tools_in = body.get("tools") or []
+        mapped = []
+        for t in tools_in:
+            fn = t.get("function", t) if isinstance(t, dict) else {}
+            if fn.get("name"):
+                mapped.append({"name": str(fn["name"]),
+                               "description": str(fn.get("description", "")),
+                               "parameters": fn.get("parameters", {}) or {}})
+        system_parts = [str(m.get("content", "")) for m in messages
+                        if m.get("role") == "system"]
+        convo = [m for m in messages if m.get("role") != "system"]
+        user_body = "\n".join(f"{m.get('role', 'user')}: {m.get('content', '')}"
+                              for m in convo)
+        client = LocalEngineClient(
+            self.engine,
+            max_tokens=max(1, min(int(body.get("max_tokens") or 256),
+                                  self.max_tokens_cap)),
+            temperature=float(body.get("temperature") or 0.0))
+        resp = client.chat("\n".join(system_parts), user_body, tools=mapped)
+        message: dict[str, Any] = {"role": "assistant",
+                                   "content": resp.content or None}
+        finish = "stop"
+        if resp.tool_calls:
+            message["tool_calls"] = [
+                {"id": c.id, "type": "function",
+                 "function": {"name": c.name,
+                              "arguments": json.dumps(c.arguments)}}
+                for c in resp.tool_calls]
+            finish = "tool_calls"
+        return {
+            "id": _gen_id("chatcmpl"),
+            "object": "chat.completion",
+            "created": _now(),
+            "model": self.model_name,
+            "choices": [{"index": 0, "message": message,
+                         "finish_reason": finish}],
+            "usage": {"prompt_tokens": 0, "completion_tokens": 0,
+                      "total_tokens": 0},
+        }
+
     def chat_completion(self, body: dict[str, Any]) -> dict[str, Any]:
         messages = body.get("messages") or []
         if not isinstance(messages, list) or not messages:
             raise ValueError("messages must be a non-empty array")
+        if body.get("tools"):
+            return self._chat_with_tools(body, messages)
         schema = self._resolve_schema(body.get("response_format"))
         ids = self._encode_chat(messages)
         req = self._submit(ids, body, schema)
