@@ -291,6 +291,36 @@ class ServingAdapter:
         }) + "\n\n"
         yield "data: [DONE]\n\n"
 
+    # -- embeddings ---------------------------------------------------------
+
+    @property
+    def embedder(self):
+        """Lazy local embedder (GPU BGE-class encoder when a device is
+        visible, deterministic hash embedder on CPU)."""
+        if not hasattr(self, "_embedder"):
+            from ..knowledge.indexer.embedder import create_embedder
+
+            self._embedder = create_embedder({"backend": "auto"})
+        return self._embedder
+
+    def embeddings(self, body: dict[str, Any]) -> dict[str, Any]:
+        inputs = body.get("input")
+        if isinstance(inputs, str):
+            inputs = [inputs]
+        if not isinstance(inputs, list) or not inputs or \
+                not all(isinstance(t, str) for t in inputs):
+            raise ValueError("input must be a string or array of strings")
+        vecs = self.embedder.embed_texts(inputs)
+        return {
+            "object": "list",
+            "model": body.get("model") or f"{self.model_name}-embed",
+            "data": [{"object": "embedding", "index": i,
+                      "embedding": [float(x) for x in vec]}
+                     for i, vec in enumerate(vecs)],
+            "usage": {"prompt_tokens": sum(len(t) // 4 for t in inputs),
+                      "total_tokens": sum(len(t) // 4 for t in inputs)},
+        }
+
     def health(self) -> dict[str, Any]:
         with self.engine._lock:
             waiting, running = len(self.engine.waiting), len(self.engine.running)
@@ -371,6 +401,22 @@ def create_app(engine: Optional[LLMEngine] = None, model: str = "tiny",
                 return StreamingResponse(adapter.chat_completion_stream(body),
                                          media_type="text/event-stream")
             return adapter.chat_completion(body)
+        except (ValueError, TypeError) as e:
+            return _error(400, str(e))
+        except Exception as e:  # noqa: BLE001
+            return _error(500, f"{type(e).__name__}: {e}")
+
+    @app.post("/v1/embeddings")
+    async def embeddings(request: HttpRequest):
+        denied = _auth_fail(request)
+        if denied is not None:
+            return denied
+        try:
+            body = await request.json()
+        except Exception:  # noqa: BLE001
+            return _error(400, "invalid JSON body")
+        try:
+            return adapter.embeddings(body)
         except (ValueError, TypeError) as e:
             return _error(400, str(e))
         except Exception as e:  # noqa: BLE001

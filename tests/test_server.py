@@ -262,3 +262,27 @@ class TestToolCalling:
         assert r.status_code == 200
         choice = r.json()["choices"][0]
         assert choice["finish_reason"] in ("tool_calls", "stop")
+
+
+class TestEmbeddings:
+    def test_single_string(self, client):
+        r = client.post("/v1/embeddings", json={"input": "redis pool exhausted"})
+        assert r.status_code == 200
+        out = r.json()
+        assert out["object"] == "list" and len(out["data"]) == 1
+        vec = out["data"][0]["embedding"]
+        assert len(vec) > 0 and all(isinstance(x, float) for x in vec[:4])
+
+    def test_batch_and_order(self, client):
+        r = client.post("/v1/embeddings", json={"input": ["a", "b", "c"]})
+        data = r.json()["data"]
+        assert [d["index"] for d in data] == [0, 1, 2]
+
+    def test_deterministic(self, client):
+        v1 = client.post("/v1/embeddings", json={"input": "same text"}).json()
+        v2 = client.post("/v1/embeddings", json={"input": "same text"}).json()
+        assert v1["data"][0]["embedding"] == v2["data"][0]["embedding"]
+
+    def test_bad_input_400(self, client):
+        assert client.post("/v1/embeddings", json={"input": 42}).status_code == 400
+        assert client.post("/v1/embeddings", json={}).status_code == 400
