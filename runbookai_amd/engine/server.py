@@ -45,6 +45,37 @@ def _gen_id(prefix: str) -> str:
     return f"{prefix}-{uuid.uuid4().hex[:24]}"
 
 
+def _stop_list(body: dict[str, Any]) -> list[str]:
+    stop = body.get("stop")
+    if isinstance(stop, str):
+        return [stop] if stop else []
+    if isinstance(stop, list):
+        return [str(s) for s in stop if s][:4]
+    return []
+
+
+def _truncate_at_stop(text: str, stops: list[str]) -> tuple[str, bool]:
+    """(text up to the earliest stop sequence, whether one was hit)."""
+    cut = -1
+    for s in stops:
+        i = text.find(s)
+        if i >= 0 and (cut < 0 or i < cut):
+            cut = i
+    return (text[:cut], True) if cut >= 0 else (text, False)
+
+
+def _safe_emit_len(text: str, stops: list[str]) -> int:
+    """Length of text safe to emit now: hold back any suffix that is a
+    proper prefix of a stop sequence (it may complete next chunk)."""
+    hold = 0
+    for s in stops:
+        for k in range(min(len(s) - 1, len(text)), 0, -1):
+            if text.endswith(s[:k]):
+                hold = max(hold, k)
+                break
+    return len(text) - hold
+
+
 class ServingAdapter:
     """Protocol logic, framework-free (unit-testable without HTTP)."""
 
@@ -133,15 +164,16 @@ class ServingAdapter:
         ids = tok.encode(str(prompt))
         req = self._submit(ids, body, None)
         self._await(req)
+        text, stopped = _truncate_at_stop(tok.decode(req.out_ids), _stop_list(body))
+        finish = "stop" if stopped or len(req.out_ids) < req.max_new_tokens \
+            else "length"
         return {
             "id": _gen_id("cmpl"),
             "object": "text_completion",
             "created": _now(),
             "model": self.model_name,
-            "choices": [{"index": 0, "text": tok.decode(req.out_ids),
-                         "finish_reason": "length"
-                         if len(req.out_ids) >= req.max_new_tokens else "stop",
-                         "logprobs": None}],
+            "choices": [{"index": 0, "text": text,
+                         "finish_reason": finish, "logprobs": None}],
             "usage": self._usage(req),
         }
 
@@ -203,7 +235,10 @@ class ServingAdapter:
         ids = self._encode_chat(messages)
         req = self._submit(ids, body, schema)
         self._await(req)
-        text = self._tok().decode(req.out_ids)
+        text, stopped = _truncate_at_stop(self._tok().decode(req.out_ids),
+                                          _stop_list(body))
+        finish = "stop" if stopped or len(req.out_ids) < req.max_new_tokens \
+            else "length"
         return {
             "id": _gen_id("chatcmpl"),
             "object": "chat.completion",
@@ -211,18 +246,21 @@ class ServingAdapter:
             "model": self.model_name,
             "choices": [{"index": 0,
                          "message": {"role": "assistant", "content": text},
-                         "finish_reason": "length"
-                         if len(req.out_ids) >= req.max_new_tokens else "stop"}],
+                         "finish_reason": finish}],
             "usage": self._usage(req),
         }
 
     # -- streaming ----------------------------------------------------------
 
-    def _stream_text(self, req: Request, poll_s: float = 0.01) -> Iterator[str]:
+    def _stream_text(self, req: Request, poll_s: float = 0.01,
+                     stops: Optional[list[str]] = None) -> Iterator[str]:
         """Yields text deltas as tokens land. Decodes the full output
-        prefix each poll so multi-byte UTF-8 never splits mid-character."""
+        prefix each poll so multi-byte UTF-8 never splits mid-character.
+        With stop sequences, any suffix that could begin one is held back
+        until it resolves; on a hit the request is cancelled."""
         tok = self._tok()
-        emitted = ""
+        stops = stops or []
+        emitted = 0
         background = self.engine._thread is not None
         try:
             while True:
@@ -231,9 +269,19 @@ class ServingAdapter:
                     self.engine.run_until_idle()
                 done = req.done_event.is_set() or req.state == "done"
                 text = tok.decode(req.out_ids)
-                if len(text) > len(emitted):
-                    yield text[len(emitted):]
-                    emitted = text
+                if stops:
+                    cut, hit = _truncate_at_stop(text, stops)
+                    if hit:
+                        if len(cut) > emitted:
+                            yield cut[emitted:]
+                        self.engine.cancel(req)
+                        return
+                    safe = _safe_emit_len(text, stops) if not done else len(text)
+                else:
+                    safe = len(text)
+                if safe > emitted:
+                    yield text[emitted:safe]
+                    emitted = safe
                 if done:
                     if req.error:
                         raise RuntimeError(req.error)
@@ -251,7 +299,7 @@ class ServingAdapter:
         ids = self._tok().encode(str(prompt))
         req = self._submit(ids, body, None)
         cid = _gen_id("cmpl")
-        for delta in self._stream_text(req):
+        for delta in self._stream_text(req, stops=_stop_list(body)):
             yield "data: " + json.dumps({
                 "id": cid, "object": "text_completion", "created": _now(),
                 "model": self.model_name,
@@ -273,7 +321,7 @@ class ServingAdapter:
         req = self._submit(self._encode_chat(messages), body, schema)
         cid = _gen_id("chatcmpl")
         first = True
-        for delta in self._stream_text(req):
+        for delta in self._stream_text(req, stops=_stop_list(body)):
             payload: dict[str, Any] = {"content": delta}
             if first:
                 payload["role"] = "assistant"

@@ -286,3 +286,42 @@ class TestEmbeddings:
     def test_bad_input_400(self, client):
         assert client.post("/v1/embeddings", json={"input": 42}).status_code == 400
         assert client.post("/v1/embeddings", json={}).status_code == 400
+
+
+class TestStopSequences:
+    def test_helpers(self):
+        from runbookai_amd.engine.server import _safe_emit_len, _truncate_at_stop
+
+        assert _truncate_at_stop("hello STOP world", ["STOP"]) == ("hello ", True)
+        assert _truncate_at_stop("hello world", ["STOP"]) == ("hello world", False)
+        # earliest of several stops wins
+        assert _truncate_at_stop("a END b STOP", ["STOP", "END"]) == ("a ", True)
+        # suffix that may grow into a stop is held back
+        assert _safe_emit_len("abc ST", ["STOP"]) == 4
+        assert _safe_emit_len("abc", ["STOP"]) == 3
+
+    def test_nonstream_truncates(self, adapter):
+        out = adapter.completion({"prompt": "q", "max_tokens": 16, "stop": []})
+        full = out["choices"][0]["text"]
+        if len(full) >= 2:
+            stop_char = full[1]
+            out2 = adapter.completion({"prompt": "q", "max_tokens": 16,
+                                       "stop": stop_char})
+            assert stop_char not in out2["choices"][0]["text"]
+            assert out2["choices"][0]["finish_reason"] == "stop"
+
+    def test_stream_respects_stop(self, client):
+        # find what the model emits, then stream with its 2nd char as stop
+        r = client.post("/v1/completions", json={"prompt": "q", "max_tokens": 12})
+        full = r.json()["choices"][0]["text"]
+        if len(full) < 3:
+            pytest.skip("model emitted too little to split")
+        stop_char = full[1]
+        with client.stream("POST", "/v1/completions", json={
+                "prompt": "q", "max_tokens": 12, "stream": True,
+                "stop": stop_char}) as resp:
+            body = "".join(resp.iter_text())
+        frames = [l[6:] for l in body.splitlines() if l.startswith("data: ")]
+        text = "".join(json.loads(f)["choices"][0]["text"]
+                       for f in frames[:-1] if f != "[DONE]")
+        assert stop_char not in text
