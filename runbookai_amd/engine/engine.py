@@ -66,12 +66,29 @@ def _string_mask_row(has_quote: bool, cap: int) -> torch.Tensor:
     return row
 
 
+def _nucleus_mask(logits: torch.Tensor, top_ps: list) -> torch.Tensor:
+    """Drop tokens outside each row's top-p nucleus (smallest set whose
+    probability mass reaches top_p). Rows with top_p >= 1 pass through."""
+    if all(p >= 1.0 for p in top_ps):
+        return logits
+    probs = torch.softmax(logits.float(), dim=-1)
+    sp, si = probs.sort(dim=-1, descending=True)
+    cum = sp.cumsum(-1)
+    tp = torch.tensor([max(0.01, min(1.0, float(p))) for p in top_ps],
+                      device=logits.device, dtype=cum.dtype).unsqueeze(1)
+    # a sorted token is dropped when the mass BEFORE it already reaches p
+    drop_sorted = (cum - sp) >= tp
+    drop = torch.zeros_like(drop_sorted).scatter(1, si, drop_sorted)
+    return logits.masked_fill(drop, float("-inf"))
+
+
 @dataclass
 class Request:
     rid: int
     prompt_ids: list[int]
     max_new_tokens: int = 512
     temperature: float = 0.0
+    top_p: float = 1.0               # nucleus sampling (1.0 = disabled)
     schema: Optional[dict[str, Any]] = None
     fsm: Optional[JsonFsm] = None
     out_ids: list[int] = field(default_factory=list)
@@ -187,7 +204,8 @@ class LLMEngine:
     # -- submission ---------------------------------------------------------------
 
     def submit(self, prompt_ids: list[int], max_new_tokens: int = 512,
-               temperature: float = 0.0, schema: Optional[dict[str, Any]] = None) -> Request:
+               temperature: float = 0.0, schema: Optional[dict[str, Any]] = None,
+               top_p: float = 1.0) -> Request:
         if schema is not None and self.hf_tokenizer is not None \
                 and not self.supports_bpe_grammar:
             raise ValueError(
@@ -207,6 +225,7 @@ class LLMEngine:
                 prompt_ids=list(prompt_ids),
                 max_new_tokens=max(1, min(max_new_tokens, room)),
                 temperature=temperature,
+                top_p=max(0.01, min(1.0, float(top_p))),
                 schema=schema,
                 fsm=JsonFsm(schema) if schema else None,
             )
@@ -223,8 +242,9 @@ class LLMEngine:
 
     def generate(self, prompt_ids: list[int], max_new_tokens: int = 512,
                  temperature: float = 0.0, schema: Optional[dict[str, Any]] = None,
-                 timeout_s: float = 600.0) -> Request:
-        req = self.submit(prompt_ids, max_new_tokens, temperature, schema)
+                 timeout_s: float = 600.0, top_p: float = 1.0) -> Request:
+        req = self.submit(prompt_ids, max_new_tokens, temperature, schema,
+                          top_p=top_p)
         if self._thread is None:
             self.run_until_idle()
         else:
@@ -664,6 +684,7 @@ class LLMEngine:
                 chosen_t = logits.argmax(-1)
             else:
                 temp = max(r.temperature for r in batch)
+                logits = _nucleus_mask(logits, [r.top_p for r in batch])
                 probs = torch.softmax(logits.float() / temp, dim=-1)
                 chosen_t = torch.multinomial(probs, 1).squeeze(-1)
             now = time.time()
@@ -694,6 +715,11 @@ class LLMEngine:
             chosen = ops.masked_greedy(region, mask_d)
         else:
             temp = max(r.temperature for r in batch)
+            if any(r.top_p < 1.0 for r in batch):
+                # nucleus over the grammar-allowed set (mass is computed
+                # AFTER masking so top_p means "of the legal tokens")
+                masked = region.masked_fill(~mask_d, float("-inf"))
+                region = _nucleus_mask(masked, [r.top_p for r in batch])
             chosen = ops.masked_sample(region, mask_d, temperature=temp)
         self.stats["sample_mask_time"] = (self.stats.get("sample_mask_time", 0.0)
                                           + time.time() - tm0)

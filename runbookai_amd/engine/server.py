@@ -131,6 +131,7 @@ class ServingAdapter:
             prompt_ids,
             max_new_tokens=max(1, min(max_tokens, self.max_tokens_cap)),
             temperature=float(body.get("temperature") or 0.0),
+            top_p=float(body.get("top_p") or 1.0),
             schema=schema,
         )
 

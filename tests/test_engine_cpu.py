@@ -457,3 +457,68 @@ class TestCancelFuzz:
             assert eng.model.kv.free_blocks == free0
         finally:
             eng.shutdown()
+
+
+class TestTopP:
+    def test_nucleus_mask_drops_tail(self):
+        import torch
+
+        from runbookai_amd.engine.engine import _nucleus_mask
+
+        logits = torch.tensor([[3.0, 2.0, 1.0, 0.0, -5.0]])
+        out = _nucleus_mask(logits, [0.6])
+        # top token(s) covering 0.6 mass survive; the tail is -inf
+        assert out[0, 0].item() == 3.0
+        assert out[0, 4].item() == float("-inf")
+        kept = (out[0] > float("-inf")).sum().item()
+        assert 1 <= kept < 5
+
+    def test_top_p_one_passthrough(self):
+        import torch
+
+        from runbookai_amd.engine.engine import _nucleus_mask
+
+        logits = torch.randn(3, 16)
+        assert _nucleus_mask(logits, [1.0, 1.0, 1.0]) is logits
+
+    def test_peaked_distribution_collapses_to_argmax(self):
+        import torch
+
+        from runbookai_amd.engine.engine import _nucleus_mask
+
+        # one dominant token (prob ~0.98): any top_p <= 0.98 keeps only it
+        logits = torch.full((1, 8), -3.0)
+        logits[0, 5] = 5.0
+        out = _nucleus_mask(logits, [0.5])
+        kept = (out[0] > float("-inf")).nonzero().flatten().tolist()
+        assert kept == [5]
+
+    def test_top_p_sampling_stays_in_nucleus(self):
+        import torch
+
+        from runbookai_amd.engine.engine import _nucleus_mask
+
+        torch.manual_seed(0)
+        logits = torch.randn(1, 64) * 4
+        out = _nucleus_mask(logits, [0.3])
+        nucleus = set((out[0] > float("-inf")).nonzero().flatten().tolist())
+        probs = torch.softmax(out.float(), dim=-1)
+        for _ in range(50):
+            tok = int(torch.multinomial(probs[0], 1))
+            assert tok in nucleus
+
+    def test_top_p_with_grammar(self):
+        import json as _json
+
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            schema = {"type": "object", "properties": {"a": {"type": "string"}},
+                      "required": ["a"]}
+            req = eng.generate([1, 2], max_new_tokens=64, temperature=0.9,
+                               top_p=0.5, schema=schema)
+            parsed = _json.loads(eng.tokenizer.decode(req.out_ids))
+            assert "a" in parsed
+        finally:
+            eng.shutdown()
