@@ -638,3 +638,60 @@ class TestDecodeGemvLdsCap:
         act = ref.silu_mul(g.contiguous(), u.contiguous())
         expected = act.float() @ w.cpu().float().t() + res.cpu().float()
         assert (out.cpu().float() - expected).abs().max().item() < 2.0
+
+
+class TestEngineGpuServing:
+    """New serving-era engine paths on real hardware: cancellation KV
+    reclaim, top_p nucleus sampling, and the OpenAI adapter end-to-end."""
+
+    def test_cancel_running_frees_kv_on_gpu(self):
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device=DEV, background=False, kv_blocks=256)
+        try:
+            free0 = eng.model.kv.free_blocks
+            req = eng.submit(list(range(64)), max_new_tokens=512)
+            eng.step()
+            assert req in eng.running
+            eng.cancel(req)
+            eng.step()
+            assert req.state == "done"
+            assert eng.model.kv.free_blocks == free0
+        finally:
+            eng.shutdown()
+
+    def test_top_p_with_grammar_on_gpu(self):
+        import json
+
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device=DEV, background=False, kv_blocks=256)
+        try:
+            schema = {"type": "object", "properties": {"s": {"type": "string"}},
+                      "required": ["s"]}
+            req = eng.generate([3, 4, 5], max_new_tokens=64, temperature=0.9,
+                               top_p=0.5, schema=schema)
+            parsed = json.loads(eng.tokenizer.decode(req.out_ids))
+            assert "s" in parsed
+            assert ops.extension_loaded()
+        finally:
+            eng.shutdown()
+
+    def test_openai_adapter_on_gpu(self):
+        import json
+
+        from runbookai_amd.engine.engine import LLMEngine
+        from runbookai_amd.engine.server import ServingAdapter
+
+        eng = LLMEngine(model="tiny", device=DEV, background=False, kv_blocks=256)
+        try:
+            adapter = ServingAdapter(eng, model_name="tiny")
+            out = adapter.chat_completion({
+                "messages": [{"role": "user", "content": "status?"}],
+                "response_format": {"type": "json_object"},
+                "max_tokens": 64})
+            parsed = json.loads(out["choices"][0]["message"]["content"])
+            assert isinstance(parsed, dict)
+            assert out["usage"]["completion_tokens"] > 0
+        finally:
+            eng.shutdown()
