@@ -325,3 +325,41 @@ class TestStopSequences:
         text = "".join(json.loads(f)["choices"][0]["text"]
                        for f in frames[:-1] if f != "[DONE]")
         assert stop_char not in text
+
+
+class TestStopStreamingProperty:
+    def test_incremental_emission_matches_truncation(self):
+        """Chunked safe-emit + final flush reconstructs exactly the
+        truncated text for any chunking of the stream — stop sequences
+        spanning chunk boundaries included."""
+        import random
+
+        from runbookai_amd.engine.server import (
+            _safe_emit_len,
+            _truncate_at_stop,
+        )
+
+        rng = random.Random(3)
+        alphabet = "abSTOP "
+        for trial in range(200):
+            full = "".join(rng.choice(alphabet) for _ in range(rng.randrange(1, 40)))
+            stops = [rng.choice(["STOP", "ab", "P "])]
+            expect, _hit = _truncate_at_stop(full, stops)
+            # simulate the streaming loop over a random chunking
+            emitted = ""
+            seen = ""
+            i = 0
+            while i < len(full):
+                i += rng.randrange(1, 5)
+                seen = full[:min(i, len(full))]
+                cut, hit = _truncate_at_stop(seen, stops)
+                if hit:
+                    emitted += cut[len(emitted):]
+                    break
+                safe = _safe_emit_len(seen, stops) if i < len(full) else len(seen)
+                if safe > len(emitted):
+                    emitted += seen[len(emitted):safe]
+            else:
+                cut, hit = _truncate_at_stop(seen, stops)
+                emitted += (cut if hit else seen)[len(emitted):]
+            assert emitted == expect, (full, stops, emitted, expect)
