@@ -303,3 +303,67 @@ def test_aws_query_lambda_listing_sets_hint():
     orch._update_cloudwatch_hints("aws_query", {}, {
         "results": {"lambda": {"items": [{"FunctionName": "img-resize"}]}}})
     assert orch.inferred_log_group == "/aws/lambda/img-resize"
+
+
+def _remediation_llm(steps):
+    llm = scripted_llm()
+    # override remediation plan with the given steps
+    llm.matchers = [(rx, fn) for rx, fn in llm.matchers]
+    llm.on(r"planning remediation", json.dumps({
+        "summary": "plan", "steps": steps, "rollback": ""}))
+    # matchers are checked in order; prepend the override
+    llm.matchers.insert(0, llm.matchers.pop())
+    return llm
+
+
+def test_skill_step_executes_through_skill_tool():
+    """Reference L348-383: auto-approved skill-mapped steps run via the
+    skill tool with the step's params."""
+    llm = _remediation_llm([
+        {"description": "restart checkout", "risk": "low",
+         "matchingSkill": "restart-service", "params": {"service": "checkout-api"}}])
+    tools = MockToolExecutor(overrides={"skill": {"ok": True}})
+    events = []
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=tools, auto_remediate=True)
+    orch.on(lambda e: events.append(e.type))
+    orch.investigate("checkout latency")
+    skill_calls = [p for t, p in tools.calls
+                   if t == "skill" and p.get("action") == "execute"]
+    assert skill_calls and skill_calls[0]["name"] == "restart-service"
+    assert skill_calls[0]["params"]["service"] == "checkout-api"
+    assert "remediation_step" in events
+
+
+def test_command_only_step_left_for_manual_execution():
+    """Reference L385-417: a step with only a shell command (no tool, no
+    skill) is surfaced as manual, never executed."""
+    llm = _remediation_llm([
+        {"description": "flush the cache", "risk": "low",
+         "command": "redis-cli FLUSHDB"}])
+    tools = MockToolExecutor()
+    events = []
+    orch = InvestigationOrchestrator(llm=llm, tool_executor=tools, auto_remediate=True)
+    orch.on(lambda e: events.append((e.type, e.data)))
+    orch.investigate("stale cache")
+    manual = [d for t, d in events if t == "remediation_manual"]
+    assert manual and manual[0]["description"] == "flush the cache"
+    # the skill tool may be listed for planning context, never executed
+    assert not any(t == "skill" and p.get("action") == "execute"
+                   for t, p in tools.calls)
+
+
+def test_approval_callback_approves_skill_execution():
+    """Reference L419-459: an approving callback lets a high-risk
+    skill-mapped step run."""
+    llm = _remediation_llm([
+        {"description": "rollback deploy", "risk": "high", "requiresApproval": True,
+         "matchingSkill": "rollback-deployment"}])
+    tools = MockToolExecutor(overrides={"skill": {"ok": True}})
+    asked = []
+    orch = InvestigationOrchestrator(
+        llm=llm, tool_executor=tools, auto_remediate=True,
+        approval_callback=lambda step: asked.append(step) or True)
+    orch.investigate("bad deploy")
+    assert asked and asked[0]["risk"] == "high"
+    assert any(t == "skill" and p.get("action") == "execute"
+               for t, p in tools.calls)
