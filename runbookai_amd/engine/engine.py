@@ -167,11 +167,10 @@ class LLMEngine:
         self.stats = {"requests": 0, "prefill_tokens": 0, "decode_tokens": 0,
                       "steps": 0, "prefill_time": 0.0, "decode_time": 0.0,
                       "cached_prefix_tokens": 0}
-        from collections import deque as _deque
-        #: (ttft_s, e2e_s, out_tokens) of recently finished requests
-        self._latency_ring: "_deque" = _deque(maxlen=512)
-        self._stop = False
         from collections import deque
+        #: (ttft_s, e2e_s, out_tokens) of recently finished requests
+        self._latency_ring: "deque" = deque(maxlen=512)
+        self._stop = False
         self._ev_queue: "deque" = deque()      # (phase, ev0, ev1, ev2) pending fold
         self._step_batch: list[Request] = []   # requests in the executing step
         self._thread: Optional[threading.Thread] = None
