@@ -359,6 +359,9 @@ class ServingAdapter:
         if not isinstance(inputs, list) or not inputs or \
                 not all(isinstance(t, str) for t in inputs):
             raise ValueError("input must be a string or array of strings")
+        if len(inputs) > 2048:
+            raise ValueError("input exceeds the 2048-item batch limit")
+        inputs = [t[:32768] for t in inputs]  # bound per-item work
         vecs = self.embedder.embed_texts(inputs)
         return {
             "object": "list",

@@ -363,3 +363,10 @@ class TestStopStreamingProperty:
                 cut, hit = _truncate_at_stop(seen, stops)
                 emitted += (cut if hit else seen)[len(emitted):]
             assert emitted == expect, (full, stops, emitted, expect)
+
+
+class TestEmbeddingsLimits:
+    def test_batch_limit(self, client):
+        r = client.post("/v1/embeddings", json={"input": ["x"] * 2049})
+        assert r.status_code == 400
+        assert "2048" in r.json()["error"]["message"]
