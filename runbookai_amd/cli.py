@@ -129,22 +129,44 @@ def chat(ctx: click.Context, provider: Optional[str]) -> None:
     from .agent.types import AgentConfig
 
     rt = _build_runtime(ctx.obj["config"], None, provider)
-    memory = ConversationMemory(summarize_after_messages=16, llm=rt["llm"])
+    # conversation memory persists across chat sessions (beyond the
+    # reference, whose memory is in-process only)
+    mem_path = os.path.join(".runbook", "chat_memory.json")
+    memory = None
+    if os.path.exists(mem_path):
+        try:
+            with open(mem_path, encoding="utf-8") as f:
+                memory = ConversationMemory.from_json(f.read())
+            memory.llm = rt["llm"]
+            _echo(f"{DIM}restored {memory.stats()['messageCount']} messages, "
+                  f"{memory.stats()['investigationCount']} investigations{RESET}")
+        except (OSError, ValueError, KeyError):
+            memory = None
+    if memory is None:
+        memory = ConversationMemory(summarize_after_messages=16, llm=rt["llm"])
     agent = Agent(llm=rt["llm"], tools=rt["tools"], knowledge_retriever=rt["retriever"],
                   config=AgentConfig(), scratchpad_dir=".runbook/scratchpad",
                   conversation_memory=memory)
     _echo(f"{BOLD}runbook chat{RESET} — type 'exit' to quit")
-    while True:
-        try:
-            query = input(f"{BOLD}> {RESET}").strip()
-        except (EOFError, KeyboardInterrupt):
-            break
-        if query.lower() in ("exit", "quit", ""):
-            if query:
+    try:
+        while True:
+            try:
+                query = input(f"{BOLD}> {RESET}").strip()
+            except (EOFError, KeyboardInterrupt):
                 break
-            continue
-        for event in agent.run(query):
-            _render_agent_event(event)
+            if query.lower() in ("exit", "quit", ""):
+                if query:
+                    break
+                continue
+            for event in agent.run(query):
+                _render_agent_event(event)
+    finally:
+        try:
+            os.makedirs(".runbook", exist_ok=True)
+            with open(mem_path, "w", encoding="utf-8") as f:
+                f.write(memory.to_json())
+        except OSError:
+            pass
 
 
 # -- investigate ----------------------------------------------------------------

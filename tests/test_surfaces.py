@@ -571,3 +571,26 @@ class TestAdapterRegistry:
 
         with _pytest.raises(ValueError):
             create_adapter({"kind": "nope-backend"})
+
+
+class TestChatMemoryPersistence:
+    def test_chat_restores_and_saves_memory(self, runner, tmp_path, monkeypatch):
+        monkeypatch.chdir(tmp_path)
+        from runbookai_amd.agent.conversation_memory import ConversationMemory
+        from runbookai_amd.cli import cli
+
+        # seed a saved memory from a "previous session"
+        prior = ConversationMemory()
+        prior.add_investigation("redis outage", "pool exhausted", ["redis"])
+        os.makedirs(".runbook", exist_ok=True)
+        with open(".runbook/chat_memory.json", "w") as f:
+            f.write(prior.to_json())
+
+        result = runner.invoke(cli, ["chat", "--provider", "mock"],
+                               input="exit\n", obj={})
+        assert result.exit_code == 0
+        assert "restored" in result.output
+        # file survives the session (re-saved on exit)
+        restored = ConversationMemory.from_json(
+            open(".runbook/chat_memory.json").read())
+        assert restored.get_investigations()[0].query == "redis outage"
