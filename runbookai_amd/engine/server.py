@@ -392,11 +392,19 @@ def create_app(engine: Optional[LLMEngine] = None, model: str = "tiny",
     from fastapi import FastAPI
     from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
 
+    owns_engine = engine is None
     if engine is None:
         engine = LLMEngine(model=model, **engine_kwargs)
     adapter = ServingAdapter(engine, model_name=model_name or model)
     app = FastAPI(title="runbookai-amd serving", version="0.2.0")
     app.state.adapter = adapter
+
+    if owns_engine:
+        # engines we constructed get torn down with the server so
+        # in-flight requests fail fast instead of hanging callers
+        @app.on_event("shutdown")
+        def _shutdown_engine() -> None:  # pragma: no cover - uvicorn lifecycle
+            engine.shutdown()
 
     def _error(status: int, message: str) -> JSONResponse:
         return JSONResponse(status_code=status,
