@@ -370,3 +370,27 @@ class TestEmbeddingsLimits:
         r = client.post("/v1/embeddings", json={"input": ["x"] * 2049})
         assert r.status_code == 400
         assert "2048" in r.json()["error"]["message"]
+
+
+class TestFailureIsolation:
+    def test_step_failure_maps_to_500_and_engine_recovers(self):
+        from fastapi.testclient import TestClient
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False, kv_blocks=128)
+        try:
+            client = TestClient(create_app(engine=eng, model_name="tiny"))
+            real_prefill = eng.model.prefill
+
+            def broken(*a, **k):
+                raise RuntimeError("injected device fault")
+
+            eng.model.prefill = broken
+            r = client.post("/v1/completions", json={"prompt": "x", "max_tokens": 4})
+            assert r.status_code == 500
+            assert "injected device fault" in r.json()["error"]["message"]
+            # the fault touched only that request; the engine serves again
+            eng.model.prefill = real_prefill
+            r2 = client.post("/v1/completions", json={"prompt": "x", "max_tokens": 4})
+            assert r2.status_code == 200
+        finally:
+            eng.shutdown()
