@@ -441,6 +441,142 @@ class SimScenario:
         return s
 
     @classmethod
+    def nat_port_exhaustion(cls) -> "SimScenario":
+        """Outbound SNAT port exhaustion on the NAT gateway after a deploy
+        disabled HTTP keep-alives — external calls fail intermittently while
+        everything in-VPC looks healthy."""
+        s = cls(name="nat-port-exhaustion")
+        s.incident = {
+            "id": "PD-EXAMPLE-009",
+            "title": "payment-service reporting intermittent gateway timeouts to the card processor",
+            "status": "triggered",
+            "urgency": "high",
+            "service": "payment-service",
+            "createdAt": "2026-03-02T14:05:00Z",
+        }
+        s.services = [
+            {"name": "payment-service", "status": "degraded", "type": "ecs"},
+            {"name": "fraud-scorer", "status": "degraded", "type": "ecs"},
+            {"name": "checkout-api", "status": "healthy", "type": "ecs"},
+            {"name": "nat-gateway-prod", "status": "saturated", "type": "network"},
+        ]
+        s.alarms = [
+            {"name": "payment-external-errors", "state": "ALARM",
+             "reason": "card-processor call failures > 8% for 10 minutes",
+             "service": "payment-service"},
+            {"name": "nat-gw-ErrorPortAllocation", "state": "ALARM",
+             "reason": "ErrorPortAllocation > 0 (SNAT ports exhausted)",
+             "service": "nat-gateway-prod"},
+            {"name": "payment-cpu", "state": "OK", "reason": "", "service": "payment-service"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-03-02T14:06:11Z", "service": "payment-service", "level": "ERROR",
+             "message": "POST https://api.cardprocessor.example/v2/charge: dial tcp: connect: connection timed out"},
+            {"timestamp": "2026-03-02T14:06:40Z", "service": "fraud-scorer", "level": "ERROR",
+             "message": "GET https://ipinfo.external.example/lookup: i/o timeout after 10s"},
+            {"timestamp": "2026-03-02T14:07:02Z", "service": "payment-service", "level": "WARN",
+             "message": "retrying charge c-55281 (attempt 2/3) — previous attempt timed out"},
+            {"timestamp": "2026-03-02T14:08:15Z", "service": "payment-service", "level": "ERROR",
+             "message": "charge failed after 3 attempts: upstream unreachable (all external calls affected)"},
+            {"timestamp": "2026-03-02T14:09:00Z", "service": "payment-service", "level": "INFO",
+             "message": "in-VPC dependencies healthy: db 2ms, redis 1ms — only egress is failing"},
+        ]
+        s.metrics = {
+            "nat.ErrorPortAllocation": [0, 0, 0, 12, 240, 890, 1450, 1600],
+            "nat.ActiveConnectionCount": [18000, 19500, 21000, 44000, 61000, 64000, 64500, 64512],
+            "payment-service.external_error_rate": [0.2, 0.3, 0.2, 3.1, 8.4, 11.9, 12.5, 12.2],
+        }
+        s.deployments = [
+            {"service": "payment-service", "version": "v2026.03.02-4", "at": "2026-03-02T13:48:00Z",
+             "change": "http client rewrite: per-request connections (keep-alive disabled)"},
+        ]
+        s.pods = [
+            {"name": "payment-service-9d1f", "namespace": "prod", "status": "Running",
+             "restarts": 0, "cpu": "310m", "memory": "640Mi"},
+            {"name": "fraud-scorer-3c77", "namespace": "prod", "status": "Running",
+             "restarts": 0, "cpu": "150m", "memory": "380Mi"},
+        ]
+        s.monitors = [
+            {"name": "external call error rate", "status": "Alert",
+             "query": "avg:payment.external.errors{*} > 5"},
+            {"name": "nat port allocation errors", "status": "Alert",
+             "query": "sum:aws.natgateway.error_port_allocation{*} > 0"},
+        ]
+        s.resources = {
+            "ec2": [{"id": "nat-0a1b2c3d", "type": "natgateway", "status": "available",
+                     "subnet": "private-a", "snatPortsInUse": 64512, "snatPortsMax": 64512}],
+            "ecs": [{"name": "payment-service", "desiredCount": 8, "runningCount": 8,
+                     "taskDefinition": "payment-service:77"}],
+        }
+        return s
+
+    @classmethod
+    def clock_skew_auth(cls) -> "SimScenario":
+        """Auth failures from node clock skew: the time-sync daemon died on
+        one node pool, JWTs mint with future timestamps and verifiers reject
+        them — a 401 spike that is NOT a credential or deploy problem."""
+        s = cls(name="clock-skew-auth")
+        s.incident = {
+            "id": "PD-EXAMPLE-010",
+            "title": "auth-service 401 rate spiked; users intermittently logged out",
+            "status": "triggered",
+            "urgency": "high",
+            "service": "auth-service",
+            "createdAt": "2026-03-05T03:22:00Z",
+        }
+        s.services = [
+            {"name": "auth-service", "status": "degraded", "type": "eks"},
+            {"name": "session-api", "status": "degraded", "type": "eks"},
+            {"name": "user-profile", "status": "healthy", "type": "eks"},
+        ]
+        s.alarms = [
+            {"name": "auth-401-rate", "state": "ALARM",
+             "reason": "401 responses > 6% of auth traffic for 20 minutes",
+             "service": "auth-service"},
+            {"name": "node-clock-sync", "state": "ALARM",
+             "reason": "chronyd inactive on node pool spot-c; offset 47s and drifting",
+             "service": "auth-service"},
+        ]
+        s.log_events = [
+            {"timestamp": "2026-03-05T03:24:10Z", "service": "session-api", "level": "ERROR",
+             "message": "jwt validation failed: token used before issued (iat is in the future)"},
+            {"timestamp": "2026-03-05T03:24:31Z", "service": "auth-service", "level": "WARN",
+             "message": "issued token with iat 2026-03-05T03:25:18Z from pod on node spot-c-14"},
+            {"timestamp": "2026-03-05T03:25:02Z", "service": "session-api", "level": "ERROR",
+             "message": "jwt validation failed: nbf (not before) check failed, rejecting request"},
+            {"timestamp": "2026-03-05T03:26:40Z", "service": "session-api", "level": "INFO",
+             "message": "tokens minted by pods on node pool on-demand-a validate fine"},
+            {"timestamp": "2026-03-05T03:27:12Z", "service": "auth-service", "level": "ERROR",
+             "message": "node spot-c-14: chronyd unit failed 41 minutes ago; system clock offset +47.2s"},
+        ]
+        s.metrics = {
+            "auth-service.401_rate": [0.4, 0.5, 0.4, 2.1, 4.8, 6.3, 6.9, 6.6],
+            "node.spot-c.clock_offset_s": [0.01, 0.02, 0.01, 8.5, 21.0, 34.5, 43.1, 47.2],
+            "session-api.jwt_rejections": [2, 1, 3, 180, 560, 840, 960, 910],
+        }
+        s.deployments = []  # no deploys: the red herring everyone checks first
+        s.pods = [
+            {"name": "auth-service-5k2m", "namespace": "prod", "status": "Running",
+             "restarts": 0, "cpu": "200m", "memory": "500Mi", "node": "spot-c-14"},
+            {"name": "auth-service-8n4q", "namespace": "prod", "status": "Running",
+             "restarts": 0, "cpu": "190m", "memory": "480Mi", "node": "on-demand-a-02"},
+            {"name": "session-api-2j9x", "namespace": "prod", "status": "Running",
+             "restarts": 0, "cpu": "160m", "memory": "350Mi", "node": "on-demand-a-03"},
+        ]
+        s.monitors = [
+            {"name": "auth 401 spike", "status": "Alert",
+             "query": "sum:auth.responses.401{*}.as_rate() > 0.06"},
+            {"name": "node clock offset", "status": "Alert",
+             "query": "max:node.clock.offset{pool:spot-c} > 1"},
+        ]
+        s.resources = {
+            "eks": [{"name": "prod-cluster", "nodePools": [
+                {"name": "spot-c", "nodes": 6, "issue": "chronyd inactive on 3 nodes"},
+                {"name": "on-demand-a", "nodes": 8, "issue": None}]}],
+        }
+        return s
+
+    @classmethod
     def from_fixture(cls, case: dict[str, Any]) -> "SimScenario":
         """Generate a scenario from an eval fixture case: the telemetry
         reflects the expected root cause so a competent agent can find it."""
@@ -493,6 +629,8 @@ _SCENARIOS = {
     "dns-resolution": SimScenario.dns_resolution,
     "queue-backlog": SimScenario.queue_backlog,
     "db-cpu-saturation": SimScenario.db_cpu_saturation,
+    "nat-port-exhaustion": SimScenario.nat_port_exhaustion,
+    "clock-skew-auth": SimScenario.clock_skew_auth,
 }
 
 _current: Optional[SimScenario] = None

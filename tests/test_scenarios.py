@@ -46,6 +46,8 @@ def test_scenario_loads_and_tools_see_it(name):
     ("dns-resolution", "SERVFAIL"),
     ("queue-backlog", "poison message"),
     ("db-cpu-saturation", "Seq Scan"),
+    ("nat-port-exhaustion", "only egress is failing"),
+    ("clock-skew-auth", "token used before issued"),
 ])
 def test_causal_needle_reachable_through_logs(name, needle):
     set_scenario(load_scenario(name))
@@ -92,3 +94,25 @@ def test_orchestrated_investigation_completes_on_each_world(name):
     result = orch.investigate(load_scenario(name).incident["title"])
     assert result.success
     assert "complete" in result.phases_visited
+
+
+def test_nat_scenario_isolates_egress():
+    """The NAT world's in-VPC signals stay healthy: the causal trail must
+    point at SNAT port allocation, not any internal dependency."""
+    set_scenario(load_scenario("nat-port-exhaustion"))
+    reg = ToolRegistry()
+    alarms = reg.execute("cloudwatch_alarms", {"state": "ALARM"})
+    names = [a["name"] for a in alarms["alarms"]]
+    assert any("PortAllocation" in n for n in names)
+    # the deploy that disabled keep-alives is discoverable
+    sc = load_scenario("nat-port-exhaustion")
+    assert any("keep-alive" in d["change"] for d in sc.deployments)
+
+
+def test_clock_skew_scenario_has_no_deploy_red_herring():
+    """Clock-skew causality with an empty deploy history: the answer must
+    come from the chronyd/node-offset trail."""
+    sc = load_scenario("clock-skew-auth")
+    assert sc.deployments == []
+    assert any("chronyd" in e["message"] for e in sc.log_events)
+    assert sc.metrics["node.spot-c.clock_offset_s"][-1] > 40
