@@ -106,9 +106,59 @@ ARCHETYPES: list[dict[str, Any]] = [
      "symptoms": ["cross-zone call failures", "elevated packet loss",
                   "health checks flapping"],
      "services": [["zone-b-fleet", "replicator"], ["cluster-peer", "gossip-mesh"]]},
+    {"name": "egress-port-exhaustion",
+     "keywords": ["nat", "port", "exhausted", "egress"],
+     "extra": [["snat"], ["keep-alive"], ["outbound"]],
+     "symptoms": ["external calls timing out", "in-VPC traffic healthy",
+                  "port allocation errors on the NAT gateway"],
+     "services": [["payment-service", "fraud-scorer"],
+                  ["webhook-sender", "partner-sync"]]},
+    {"name": "clock-skew",
+     "keywords": ["clock", "skew", "offset", "time"],
+     "extra": [["chronyd"], ["ntp"], ["iat"]],
+     "symptoms": ["token validation failures", "401 rate climbing",
+                  "signatures rejected as not yet valid"],
+     "services": [["auth-service", "session-api"], ["token-issuer", "api-edge"]]},
+    {"name": "memory-leak-gradual",
+     "keywords": ["memory", "leak", "growing", "rss"],
+     "extra": [["heap"], ["unbounded cache"], ["native"]],
+     "symptoms": ["rss climbing for hours", "gc pause times rising",
+                  "restarts temporarily fix latency"],
+     "services": [["stream-processor", "metrics-ingest"],
+                  ["graph-api", "cache-warmer"]]},
 ]
 
 NATURAL_TELEMETRY: dict[str, dict] = {
+ "egress-port-exhaustion": {
+  "logs": [
+   "POST https://api.external.example/v2/send: dial tcp: connect: connection timed out",
+   "outbound call failed; retried 3 times ({extra0} ports at ceiling)",
+   "in-VPC dependencies responding normally; only egress affected",
+   "ErrorPortAllocation reported by the NAT gateway"
+  ],
+  "alarm": "NAT gateway port allocation errors > 0 affecting {svc}",
+  "metric": "{svc}.external_error_rate"
+ },
+ "clock-skew": {
+  "logs": [
+   "jwt validation failed: token used before issued (iat in the future)",
+   "nbf check failed rejecting request from {svc}",
+   "{extra0} unit inactive on node; system clock offset +40s and drifting",
+   "tokens minted on the other node pool validate fine"
+  ],
+  "alarm": "node clock offset over 1s affecting {svc}",
+  "metric": "{svc}.jwt_rejections"
+ },
+ "memory-leak-gradual": {
+  "logs": [
+   "rss grew 3.1GB over 6h with flat traffic on {svc}",
+   "gc: full collection took 2100ms, freed 1% of heap",
+   "{extra0} retained size growing monotonically between deploys",
+   "container approaching memory limit; no OOM kill yet"
+  ],
+  "alarm": "memory usage slope positive for 6h on {svc}",
+  "metric": "{svc}.rss_bytes"
+ },
  "conn-pool-exhaustion": {
   "logs": [
    "could not acquire connection from pool after 5000ms ({extra0} pool size 50, in use 50)",
