@@ -288,6 +288,19 @@ def status(ctx: click.Context) -> None:
             _echo("  gpu: none visible (CPU mode)")
     except Exception:  # noqa: BLE001
         _echo("  gpu: torch unavailable")
+    try:
+        from . import ops
+
+        if ops.extension_loaded():
+            _echo(f"  hip extension: {GREEN}loaded{RESET} (gfx950 kernels active)")
+        else:
+            import torch as _t
+
+            mode = "CPU fp32 references" if not _t.cuda.is_available() \
+                else f"{RED}NOT LOADED on a GPU — ops will raise{RESET}"
+            _echo(f"  hip extension: {mode}")
+    except Exception as e:  # noqa: BLE001
+        _echo(f"  hip extension: probe failed ({type(e).__name__})")
     if os.path.isdir(".runbook"):
         from .knowledge.retriever.default import create_retriever
 
