@@ -83,13 +83,46 @@ def _generic(tool: str, args: dict[str, Any], data: Any) -> CompactToolResult:
                              services=_extract_services(data))
 
 
+def _lambda_names(data: Any) -> list[str]:
+    """Lambda function names from result items — FunctionName field first,
+    the ARN's last segment when the name is absent (reference
+    tool-summarizer.test.ts:5-59)."""
+    names: list[str] = []
+
+    def visit(obj: Any, depth: int = 0) -> None:
+        if depth > 4 or len(names) >= 6:
+            return
+        if isinstance(obj, dict):
+            fn = obj.get("FunctionName") or obj.get("functionName")
+            if isinstance(fn, str):
+                names.append(fn)
+            else:
+                arn = obj.get("FunctionArn") or obj.get("functionArn") or obj.get("arn")
+                if isinstance(arn, str) and ":function:" in arn:
+                    names.append(arn.rsplit(":function:", 1)[1].split(":")[0])
+            for v in obj.values():
+                visit(v, depth + 1)
+        elif isinstance(obj, list):
+            for v in obj[:10]:
+                visit(v, depth + 1)
+
+    visit(data)
+    return list(dict.fromkeys(names))[:6]
+
+
 def _sum_aws_query(args: dict[str, Any], data: Any) -> CompactToolResult:
     svc = args.get("service", "multi")
     n = _count_items(data)
     errs = _find_errors(data)
+    fns = _lambda_names(data)
+    summary = f"aws_query[{svc}]: {n} resources"
+    if fns:
+        summary += " — lambda: " + ", ".join(fns[:3])
     return CompactToolResult(
-        summary=f"aws_query[{svc}]: {n} resources",
-        item_count=n, has_errors=errs, services=_extract_services(data),
+        summary=summary,
+        item_count=n, has_errors=errs,
+        services=_extract_services(data) + fns,
+        highlights=[f"lambda function: {f}" for f in fns[:3]],
     )
 
 

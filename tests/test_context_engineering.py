@@ -595,3 +595,21 @@ class TestKnowledgeContextDepth:
         m._queried_services.add("a")
         m.reset()
         assert m.runbook_index and not m.jit_results and not m._queried_services
+
+
+class TestLambdaSummaries:
+    """Reference agent/__tests__/tool-summarizer.test.ts (2 cases)."""
+
+    def test_lambda_names_in_compact_summary(self):
+        out = ToolSummarizer().summarize("aws_query", {"service": "lambda"}, {
+            "items": [{"FunctionName": "checkout-worker", "Runtime": "python3.12"},
+                      {"FunctionName": "img-resize", "Runtime": "nodejs20"}]})
+        assert "checkout-worker" in out.summary
+        assert "img-resize" in out.summary
+        assert "checkout-worker" in out.services
+
+    def test_lambda_name_from_arn_when_missing(self):
+        out = ToolSummarizer().summarize("aws_query", {"service": "lambda"}, {
+            "items": [{"FunctionArn":
+                       "arn:aws:lambda:us-east-1:123:function:billing-fn:7"}]})
+        assert "billing-fn" in out.summary
