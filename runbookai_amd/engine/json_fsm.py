@@ -315,7 +315,15 @@ class JsonFsm:
         if t == "array":
             return [0x5B]
         if t in ("integer", "number"):
-            return _DIGITS
+            # the FIRST digit of an array number item is chosen by the
+            # ARRAY frame — run it through the same digit-level bound
+            # filter a live number frame would apply, or an out-of-range
+            # leading digit would make the bounds unsatisfiable later
+            probe = {"kind": "number", "float": t == "number", "len": 0,
+                     "max_len": 10, "has_dot": False, "lit": "",
+                     "minimum": items.get("minimum"),
+                     "maximum": items.get("maximum")}
+            return self._number_filter(probe, list(_DIGITS))
         if t == "boolean":
             return [ord("t"), ord("f")]
         return [0x22]

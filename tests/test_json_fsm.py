@@ -141,3 +141,52 @@ def test_fsm_step_interface():
     data = json.loads(emitted.decode())
     assert data["x"] in ("a", "b")
     assert fsm.done
+
+
+class TestArrayNumberBounds:
+    """Regression: the array frame chooses a number item's FIRST digit —
+    it must honor the item schema's digit-level bounds (found by the
+    bounds-aware fuzz; before the fix `{"maximum": 0}` items could open
+    with '1' and the unsatisfiable-fallback then let anything through)."""
+
+    def _walk_greedy(self, schema, prefer=b"975310"):
+        fsm = JsonFsm(schema)
+        out = bytearray()
+        for _ in range(4000):
+            if fsm.done:
+                break
+            allowed = fsm.allowed_bytes()
+            if not allowed:
+                break
+            pick = None
+            for p in prefer:  # adversarial: biggest digits first
+                if p in allowed:
+                    pick = p
+                    break
+            if pick is None:
+                pick = sorted(allowed)[0]
+            fsm.advance(pick)
+            if pick != 0:
+                out.append(pick)
+        return json.loads(out.decode())
+
+    def test_array_item_maximum(self):
+        schema = {"type": "array",
+                  "items": {"type": "integer", "minimum": 0, "maximum": 0},
+                  "minItems": 2, "maxItems": 2}
+        vals = self._walk_greedy(schema)
+        assert vals == [0, 0]
+
+    def test_array_item_minimum(self):
+        schema = {"type": "array",
+                  "items": {"type": "integer", "minimum": 7, "maximum": 9},
+                  "minItems": 3, "maxItems": 3}
+        for v in self._walk_greedy(schema):
+            assert 7 <= v <= 9
+
+    def test_array_float_bounds(self):
+        schema = {"type": "array",
+                  "items": {"type": "number", "minimum": 1.0, "maximum": 2.0},
+                  "minItems": 2, "maxItems": 2}
+        for v in self._walk_greedy(schema):
+            assert 1.0 <= v <= 2.0

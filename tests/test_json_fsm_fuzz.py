@@ -23,12 +23,25 @@ KEYS = ["alpha", "beta", "g", "x2", "long_key_name"]
 ENUMS = [["a", "b"], ["branch", "prune", "confirm"], ["low", "medium", "high"]]
 
 
+def _bounded_int(lo: int, span: int) -> dict:
+    return {"type": "integer", "minimum": lo, "maximum": lo + span}
+
+
+def _bounded_num(lo: int, span: int) -> dict:
+    return {"type": "number", "minimum": float(lo), "maximum": float(lo + span)}
+
+
 def leaf_schema() -> st.SearchStrategy:
     return st.one_of(
         st.builds(lambda n: {"type": "string", "maxLength": n},
                   st.integers(min_value=1, max_value=24)),
         st.just({"type": "number"}),
         st.just({"type": "integer"}),
+        # digit-level bound enforcement under fuzz (advisor finding r1)
+        st.builds(_bounded_int, st.integers(min_value=0, max_value=500),
+                  st.integers(min_value=0, max_value=5000)),
+        st.builds(_bounded_num, st.integers(min_value=0, max_value=99),
+                  st.integers(min_value=1, max_value=400)),
         st.just({"type": "boolean"}),
         st.sampled_from([{"enum": e} for e in ENUMS]),
     )
@@ -83,8 +96,16 @@ def check(value, schema) -> None:
         assert len(value) <= int(schema.get("maxLength", 200))
     elif t == "integer":
         assert isinstance(value, int) and not isinstance(value, bool)
+        if "minimum" in schema:
+            assert value >= schema["minimum"], (value, schema)
+        if "maximum" in schema:
+            assert value <= schema["maximum"], (value, schema)
     elif t == "number":
         assert isinstance(value, (int, float)) and not isinstance(value, bool)
+        if "minimum" in schema:
+            assert value >= schema["minimum"], (value, schema)
+        if "maximum" in schema:
+            assert value <= schema["maximum"], (value, schema)
     elif t == "boolean":
         assert isinstance(value, bool)
 
