@@ -616,3 +616,44 @@ def test_mask_row_matches_allowed_tokens(tmp_path):
             break
         tid = rng.choice(allowed)
         masker.advance_token(fsm, masker.token_bytes[tid])
+
+
+def test_bpe_masker_respects_array_number_bounds(tmp_path):
+    """Token-level admission must inherit the digit-level bound filter —
+    including the array-first-digit path fixed in json_fsm."""
+    import json as _json
+    import random
+
+    from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer
+    from runbookai_amd.engine.grammar_bpe import build_masker
+    from runbookai_amd.engine.json_fsm import JsonFsm
+
+    tokenizers = pytest.importorskip("tokenizers")
+    tk = tokenizers.ByteLevelBPETokenizer()
+    tk.train_from_iterator(["0 1 2 3 4 5 6 7 8 9 10 42 99 123 999"] * 30,
+                           vocab_size=400, min_frequency=1,
+                           special_tokens=["<|eot_id|>"])
+    tk.save(str(tmp_path / "tokenizer.json"))
+    masker = build_masker(BpeTokenizer.from_file(str(tmp_path / "tokenizer.json")))
+    schema = {"type": "object",
+              "properties": {"scores": {
+                  "type": "array",
+                  "items": {"type": "integer", "minimum": 3, "maximum": 7},
+                  "minItems": 2, "maxItems": 4}},
+              "required": ["scores"]}
+    for seed in range(6):
+        rng = random.Random(seed)
+        fsm = JsonFsm(schema)
+        out = bytearray()
+        for _ in range(2000):
+            if fsm.done:
+                break
+            allowed = masker.allowed_tokens(fsm)
+            if not allowed:
+                break
+            tid = rng.choice(sorted(allowed))
+            tb = masker.token_bytes[tid]
+            masker.advance_token(fsm, tb)
+            out.extend(tb)
+        data = _json.loads(out.decode())
+        assert all(3 <= v <= 7 for v in data["scores"]), data
