@@ -287,8 +287,14 @@ class JsonFsm:
                 lo_k = p * ten_k
                 if lo_k > hi_b:
                     return False  # grows monotonically with k
-                hi_k = lo_k + ten_k - (0 if is_f else 1)
-                if hi_k >= lo_b:
+                # k appended digits span [p*10^k, (p+1)*10^k - 1] as
+                # integers; fractions extend that to an interval OPEN at
+                # (p+1)*10^k — so the float test must be strict ('14'
+                # reaches 149.99... but never 150)
+                if is_f:
+                    if lo_k + ten_k > lo_b:
+                        return True
+                elif lo_k + ten_k - 1 >= lo_b:
                     return True
             return False
 

@@ -159,3 +159,44 @@ def test_greedy_first_choice_terminates(schema, seed):
             out.append(b)
     assert fsm.done or not fsm.allowed_bytes()
     check(json.loads(out.decode()), schema)
+
+
+@settings(max_examples=80, deadline=None)
+@given(lo=st.integers(min_value=0, max_value=800),
+       span=st.integers(min_value=0, max_value=9000),
+       seed=st.integers(min_value=0, max_value=2**31),
+       as_float=st.booleans(),
+       in_array=st.booleans())
+def test_bounded_numbers_edge_biased_walk(lo, span, seed, as_float, in_array):
+    """Edge-biased policy (prefer the LARGEST allowed digit, sometimes the
+    smallest) over random bounds — hammers the interval arithmetic at its
+    boundaries, in both the property path and the array-first-digit path."""
+    import random
+
+    num = {"type": "number" if as_float else "integer",
+           "minimum": float(lo) if as_float else lo,
+           "maximum": float(lo + span) if as_float else lo + span}
+    if in_array:
+        schema = {"type": "object",
+                  "properties": {"v": {"type": "array", "items": num,
+                                       "minItems": 2, "maxItems": 3}},
+                  "required": ["v"]}
+    else:
+        schema = {"type": "object", "properties": {"v": num}, "required": ["v"]}
+    rng = random.Random(seed)
+    fsm = JsonFsm(schema)
+    out = bytearray()
+    for _ in range(4000):
+        if fsm.done:
+            break
+        allowed = fsm.allowed_bytes()
+        if not allowed:
+            break
+        b = max(allowed) if rng.random() < 0.8 else min(allowed)
+        fsm.advance(b)
+        if b != NUMBER_CLOSE_SENTINEL:
+            out.append(b)
+    data = json.loads(out.decode())
+    vals = data["v"] if isinstance(data["v"], list) else [data["v"]]
+    for v in vals:
+        assert num["minimum"] <= v <= num["maximum"], (v, num)
