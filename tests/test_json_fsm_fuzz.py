@@ -200,3 +200,71 @@ def test_bounded_numbers_edge_biased_walk(lo, span, seed, as_float, in_array):
     vals = data["v"] if isinstance(data["v"], list) else [data["v"]]
     for v in vals:
         assert num["minimum"] <= v <= num["maximum"], (v, num)
+
+
+@settings(max_examples=150, deadline=None)
+@given(lo=st.integers(min_value=0, max_value=999),
+       hi_span=st.integers(min_value=0, max_value=999),
+       lit=st.sampled_from(["", "0", "1", "9", "12", "14", "99", "100", "149",
+                            "150", "2", "29", "3", "30", "7", "70", "701"]),
+       as_float=st.booleans())
+def test_number_filter_matches_bruteforce_oracle(lo, hi_span, lit, as_float):
+    """Oracle check: a digit/close byte is allowed iff brute-force
+    enumeration of ALL completions within max_len finds one in bounds.
+    (This is the component two real bugs hid in — closed-vs-open interval
+    arithmetic is easy to get subtly wrong.)"""
+    hi = lo + hi_span
+    max_len = 4
+    fsm = JsonFsm({"type": "object", "properties": {}, "required": []})
+    frame = {"kind": "number", "float": as_float, "len": len(lit),
+             "max_len": max_len, "has_dot": "." in lit, "lit": lit,
+             "minimum": float(lo) if as_float else lo,
+             "maximum": float(hi) if as_float else hi}
+    if len(lit) > max_len or (lit.startswith("0") and len(lit) > 1):
+        return  # not a reachable literal state
+    candidates = [0x00] + [ord(c) for c in "0123456789"] \
+        + ([0x2E] if as_float and "." not in lit else [])
+    got = set(fsm._number_filter(frame, list(candidates)))
+
+    def completions(s):
+        """All legal JSON number literals extending s — every CHARACTER
+        (dot included) counts toward max_len, exactly as the FSM counts."""
+        out = []
+        if s and not s.endswith("."):
+            out.append(s)
+        if len(s) >= max_len:
+            return out
+        if "." in s:
+            nxt = list("0123456789")
+        elif s == "":
+            nxt = list("0123456789")
+        elif s == "0":
+            nxt = ["."] if as_float else []  # leading-zero rule
+        else:
+            nxt = list("0123456789") + (["."] if as_float else [])
+        for c in nxt:
+            out.extend(completions(s + c))
+        return out
+
+    def in_bounds(s):
+        v = float(s)
+        return lo <= v <= hi
+
+    any_completion = any(in_bounds(c) for c in completions(lit))
+    if not any_completion:
+        # unsatisfiable literal: the filter falls back to unfiltered
+        # (documented behavior) — nothing to cross-check
+        return
+    for b in candidates:
+        if b == 0x00:
+            expect = bool(lit) and not lit.endswith(".") and in_bounds(lit)
+        elif b == 0x2E:
+            nxt = lit + "."
+            expect = any(in_bounds(c) for c in completions(nxt))
+        else:
+            nxt = lit + chr(b)
+            if nxt.startswith("0") and len(nxt) > 1 and "." not in nxt:
+                continue  # filter may or may not allow; fsm forbids later
+            expect = any(in_bounds(c) for c in completions(nxt))
+        assert (b in got) == expect, (chr(b) if b > 1 else "close", lit,
+                                      lo, hi, as_float, sorted(got))
