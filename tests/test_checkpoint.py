@@ -657,3 +657,68 @@ def test_bpe_masker_respects_array_number_bounds(tmp_path):
             out.extend(tb)
         data = _json.loads(out.decode())
         assert all(3 <= v <= 7 for v in data["scores"]), data
+
+
+def test_masker_admission_matches_bruteforce_walk(tmp_path):
+    """Oracle: a token is admitted iff replaying its byte expansion
+    through a CLONE of the FSM accepts every byte (the trie walk and the
+    string fast path must agree with first-principles simulation)."""
+    import copy
+    import random
+
+    from runbookai_amd.engine.bpe_tokenizer import BpeTokenizer
+    from runbookai_amd.engine.grammar_bpe import build_masker
+    from runbookai_amd.engine.json_fsm import JsonFsm
+
+    tokenizers = pytest.importorskip("tokenizers")
+    tk = tokenizers.ByteLevelBPETokenizer()
+    tk.train_from_iterator(
+        ['{"a": 1, "b": "xy"} confidence high 0.9 [3, 4]'] * 40,
+        vocab_size=420, min_frequency=1, special_tokens=["<|eot_id|>"])
+    tk.save(str(tmp_path / "tokenizer.json"))
+    masker = build_masker(BpeTokenizer.from_file(str(tmp_path / "tokenizer.json")))
+
+    def accepts(fsm, tb: bytes) -> bool:
+        from runbookai_amd.engine.json_fsm import NUMBER_CLOSE_SENTINEL
+
+        sim = copy.deepcopy(fsm)
+        for byte in tb:
+            # the number-close sentinel is zero-width: traverse it freely
+            # when the literal byte isn't directly legal (matches the
+            # masker's trie walk)
+            for _ in range(4):
+                allowed = set() if sim.done else set(sim.allowed_bytes())
+                if byte in allowed:
+                    break
+                if NUMBER_CLOSE_SENTINEL in allowed:
+                    sim.advance(NUMBER_CLOSE_SENTINEL)
+                    continue
+                return False
+            else:
+                return False
+            sim.advance(byte)
+        return True
+
+    schema = {"type": "object",
+              "properties": {"k": {"type": "string", "maxLength": 8},
+                             "n": {"type": "integer", "minimum": 2, "maximum": 40},
+                             "e": {"enum": ["low", "high"]}},
+              "required": ["k", "n", "e"]}
+    rng = random.Random(5)
+    fsm = JsonFsm(schema)
+    for step in range(400):
+        if fsm.done:
+            break
+        got = set(masker.allowed_tokens(fsm))
+        # cross-check a sample of the vocab against the simulator
+        sample = rng.sample(sorted(masker.token_bytes), k=60)
+        sample += [t for t in rng.sample(sorted(got), k=min(10, len(got)))
+                   if t in masker.token_bytes] if got else []
+        for tid in sample:
+            tb = masker.token_bytes[tid]
+            assert (tid in got) == accepts(fsm, tb), (
+                step, tid, tb, fsm.allowed_bytes()[:12])
+        if not got:
+            break
+        tid = rng.choice(sorted(got))
+        masker.advance_token(fsm, masker.token_bytes[tid])
