@@ -180,10 +180,12 @@ def chat(ctx: click.Context, provider: Optional[str]) -> None:
 @click.option("--provider", default=None)
 @click.option("--scenario", default=None)
 @click.option("--checkpoint/--no-checkpoint", "do_checkpoint", default=True)
+@click.option("--report", default=None, help="write a markdown report to this path")
 @click.pass_context
 def investigate(ctx: click.Context, incident_id: str, verbose: bool, auto_remediate: bool,
                 learn: bool, apply_runbook_updates: bool, provider: Optional[str],
-                scenario: Optional[str], do_checkpoint: bool) -> None:
+                scenario: Optional[str], do_checkpoint: bool,
+                report: Optional[str]) -> None:
     """Structured hypothesis-driven investigation of an incident."""
     from .agent.orchestrator import InvestigationOrchestrator
     from .session.checkpoint import CheckpointStore, checkpoint_from_machine
@@ -231,6 +233,11 @@ def investigate(ctx: click.Context, incident_id: str, verbose: bool, auto_remedi
         _echo("\n" + result.summary)
         _echo(f"\n{DIM}duration: {result.duration_ms} ms · LLM calls: "
               f"{orch.stats['llm_calls']} · tool calls: {orch.stats['tool_calls']}{RESET}")
+        if report:
+            from .session.report import write_investigation_report
+
+            path = write_investigation_report(report, result, orch)
+            _echo(f"{DIM}report written: {path}{RESET}")
         if learn:
             from .learning.loop import run_learning_loop
 

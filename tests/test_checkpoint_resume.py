@@ -229,3 +229,64 @@ class TestStoreParity:
         from runbookai_amd.session.checkpoint import format_checkpoint_list_markdown
 
         assert "No checkpoints" in format_checkpoint_list_markdown([])
+
+
+class TestInvestigationReport:
+    def _result(self):
+        from runbookai_amd.agent.orchestrator import InvestigationResult
+
+        return InvestigationResult(
+            investigation_id="inv-rep-1",
+            root_cause="redis connection pool exhaustion",
+            confidence="high",
+            summary="evidence: pool exhausted logs and client-count alarm",
+            affected_services=["checkout-api", "redis"],
+            remediation_plan={"summary": "raise pool", "rollback": "revert",
+                              "steps": [{"description": "bump pool to 500",
+                                         "risk": "medium",
+                                         "requiresApproval": False},
+                                        {"description": "rolling restart",
+                                         "risk": "high",
+                                         "requiresApproval": True}]},
+            duration_ms=812,
+            phases_visited=["triage", "hypothesize", "investigate",
+                            "evaluate", "conclude", "complete"],
+            hypotheses=[{"statement": "pool too small", "status": "confirmed",
+                         "confidence": 0.92},
+                        {"statement": "network partition", "status": "pruned",
+                         "confidence": 0.1}],
+            evidence=["pool exhausted log lines"],
+        )
+
+    def test_render_contains_all_sections(self):
+        from runbookai_amd.session.report import render_investigation_report
+
+        md = render_investigation_report(self._result())
+        for needle in ("# Investigation report", "redis connection pool exhaustion",
+                       "## Remediation plan", "requires approval",
+                       "## Hypotheses", "✅", "❌", "## Phase trace",
+                       "triage → hypothesize"):
+            assert needle in md, needle
+
+    def test_write_creates_dirs(self, tmp_path):
+        from runbookai_amd.session.report import write_investigation_report
+
+        path = write_investigation_report(
+            str(tmp_path / "reports" / "inv.md"), self._result())
+        assert open(path).read().startswith("# Investigation report")
+
+    def test_cli_report_flag(self, tmp_path, monkeypatch):
+        import click.testing
+
+        monkeypatch.chdir(tmp_path)
+        from runbookai_amd.cli import cli
+
+        runner = click.testing.CliRunner()
+        result = runner.invoke(cli, [
+            "investigate", "PD-EXAMPLE-001", "--provider", "mock",
+            "--scenario", "redis-conn-exhaustion", "--no-checkpoint",
+            "--report", "out/report.md"], obj={})
+        assert result.exit_code == 0, result.output
+        assert "report written" in result.output
+        md = open(tmp_path / "out" / "report.md").read()
+        assert "# Investigation report" in md
