@@ -259,6 +259,60 @@ def investigate(ctx: click.Context, incident_id: str, verbose: bool, auto_remedi
             _render_agent_event(event)
 
 
+@cli.command()
+@click.argument("session_id", required=False)
+@click.option("--dir", "directory", default=".runbook/scratchpad",
+              help="scratchpad directory")
+@click.option("--full", is_flag=True, help="show full tool results")
+def replay(session_id: Optional[str], directory: str, full: bool) -> None:
+    """Replay a recorded agent session's audit trail (scratchpad JSONL) —
+    the investigation's timeline for debugging and handoff. Without a
+    SESSION_ID, lists recorded sessions."""
+    from .agent.scratchpad import Scratchpad
+
+    if not session_id:
+        if not os.path.isdir(directory):
+            _echo(f"{DIM}(no sessions recorded under {directory}){RESET}")
+            return
+        files = sorted(fn for fn in os.listdir(directory) if fn.endswith(".jsonl"))
+        for fn in files:
+            path = os.path.join(directory, fn)
+            n = sum(1 for _ in open(path, encoding="utf-8"))
+            _echo(f"  {fn[:-6]}  {DIM}({n} entries){RESET}")
+        if not files:
+            _echo(f"{DIM}(no sessions recorded under {directory}){RESET}")
+        return
+    pad = Scratchpad.load(session_id, directory)
+    if not pad.entries:
+        _echo(f"{RED}no recorded session '{session_id}' in {directory}{RESET}")
+        sys.exit(1)
+    for e in pad.entries:
+        d = e.data
+        ts = ""
+        if e.timestamp:
+            import datetime as _dt
+
+            secs = e.timestamp / 1000.0 if e.timestamp > 1e11 else e.timestamp
+            ts = _dt.datetime.fromtimestamp(secs).strftime("%H:%M:%S ")
+        if e.kind == "init":
+            _echo(f"{BOLD}{ts}▶ session start:{RESET} {d.get('query', '')}")
+        elif e.kind == "thinking":
+            _echo(f"{DIM}{ts}💭 {str(d.get('text', ''))[:200]}{RESET}")
+        elif e.kind == "tool_result":
+            _echo(f"{CYAN}{ts}🔧 {d.get('tool')}{RESET} "
+                  f"{DIM}{json.dumps(d.get('args', {}), default=str)[:100]}{RESET}")
+            _echo(f"   → {str(d.get('summary', ''))[:160]} "
+                  f"{DIM}[{d.get('resultId', '')}]{RESET}")
+            if full and d.get("fullResult") is not None:
+                _echo(f"{DIM}{json.dumps(d.get('fullResult'), default=str)[:800]}{RESET}")
+        elif e.kind == "compaction":
+            _echo(f"{DIM}{ts}🧹 compaction: cleared {d.get('cleared', '?')}{RESET}")
+        elif e.kind == "answer":
+            _echo(f"{GREEN}{ts}✅ answer:{RESET} {str(d.get('text', ''))[:400]}")
+        else:
+            _echo(f"{DIM}{ts}{e.kind}: {json.dumps(d, default=str)[:140]}{RESET}")
+
+
 # -- demo / status / init --------------------------------------------------------
 
 @cli.command()

@@ -594,3 +594,33 @@ class TestChatMemoryPersistence:
         restored = ConversationMemory.from_json(
             open(".runbook/chat_memory.json").read())
         assert restored.get_investigations()[0].query == "redis outage"
+
+
+class TestReplay:
+    def test_replay_lists_and_renders(self, runner, tmp_path, monkeypatch):
+        monkeypatch.chdir(tmp_path)
+        from runbookai_amd.agent.scratchpad import Scratchpad
+        from runbookai_amd.cli import cli
+
+        pad = Scratchpad("sess-abc", str(tmp_path / "pads"))
+        pad.append("init", query="why is checkout slow?")
+        pad.append_tool_result("cloudwatch_alarms", {"state": "ALARM"},
+                               "1 alarm firing", {"alarms": [{"name": "x"}]})
+        pad.append("answer", text="redis pool exhausted")
+
+        listing = runner.invoke(cli, ["replay", "--dir", str(tmp_path / "pads")],
+                                obj={})
+        assert "sess-abc" in listing.output
+
+        out = runner.invoke(cli, ["replay", "sess-abc",
+                                  "--dir", str(tmp_path / "pads")], obj={})
+        assert out.exit_code == 0
+        assert "why is checkout slow?" in out.output
+        assert "cloudwatch_alarms" in out.output
+        assert "redis pool exhausted" in out.output
+
+    def test_replay_unknown_session_fails(self, runner, tmp_path):
+        from runbookai_amd.cli import cli
+
+        out = runner.invoke(cli, ["replay", "nope", "--dir", str(tmp_path)], obj={})
+        assert out.exit_code == 1
