@@ -522,3 +522,24 @@ class TestTopP:
             assert "a" in parsed
         finally:
             eng.shutdown()
+
+
+class TestSaturation:
+    def test_64_concurrent_requests_all_complete(self):
+        """Admission at max_batch with a small KV pool: everything queued
+        beyond capacity waits and still completes; the pool ends whole."""
+        from runbookai_amd.engine.engine import LLMEngine
+
+        eng = LLMEngine(model="tiny", device="cpu", background=False,
+                        kv_blocks=256, max_batch=16)
+        try:
+            free0 = eng.model.kv.free_blocks
+            reqs = [eng.submit([i % 200, 3, 5], max_new_tokens=6)
+                    for i in range(64)]
+            eng.run_until_idle()
+            assert all(r.state == "done" and not r.error for r in reqs)
+            assert all(len(r.out_ids) >= 1 for r in reqs)
+            assert eng.model.kv.free_blocks == free0
+            assert eng.stats["requests"] == 64
+        finally:
+            eng.shutdown()
