@@ -1,7 +1,7 @@
 """Causal query builder: hypothesis text -> concrete tool queries.
 
 Parity with reference src/agent/causal-query.ts (484 LoC): 8 keyword-matched
-FAILURE_PATTERNS each with canned tool queries (L30-208);
+FAILURE_PATTERNS (8 reference patterns + auth/egress added for the\nround-2 incident worlds) each with canned tool queries (L30-208);
 generate_queries_for_hypothesis with a generic fallback trio (L241-297);
 is_query_too_broad / suggest_query_refinements anti-patterns (L333-392);
 prioritize_queries dedupe + cap 10 (L397-430); summarize_query_results
@@ -106,6 +106,32 @@ FAILURE_PATTERNS: list[FailurePattern] = [
                "db error evidence", 2),
             _q("datadog", {"action": "metrics", "query": "avg:postgresql.connections{*}"},
                "db connection trend", 3),
+        ],
+    ),
+    FailurePattern(
+        name="auth",
+        keywords=["auth", "401", "403", "jwt", "token", "credential", "unauthorized",
+                  "clock", "skew", "certificate"],
+        queries=[
+            _q("cloudwatch_logs", {"filter": "401 OR unauthorized OR token", "limit": 50},
+               "auth failure evidence", 1),
+            _q("datadog", {"action": "metrics", "query": "sum:auth.responses.401{*}.as_rate()"},
+               "401 rate trend", 2),
+            _q("kubernetes_query", {"action": "nodes"}, "node pool state (clock/cert issues)", 3),
+        ],
+    ),
+    FailurePattern(
+        name="egress",
+        keywords=["egress", "nat", "outbound", "external call", "snat", "third-party",
+                  "upstream unreachable"],
+        queries=[
+            _q("cloudwatch_logs", {"filter": "dial tcp OR i/o timeout OR unreachable",
+                                   "limit": 50}, "outbound failure evidence", 1),
+            _q("datadog", {"action": "metrics",
+                           "query": "sum:aws.natgateway.error_port_allocation{*}"},
+               "NAT port allocation errors", 2),
+            _q("aws_query", {"service": "ec2", "operation": "list"},
+               "NAT gateway / network state", 3),
         ],
     ),
     FailurePattern(

@@ -116,3 +116,20 @@ def test_clock_skew_scenario_has_no_deploy_red_herring():
     assert sc.deployments == []
     assert any("chronyd" in e["message"] for e in sc.log_events)
     assert sc.metrics["node.spot-c.clock_offset_s"][-1] > 40
+
+
+def test_new_worlds_match_causal_patterns():
+    """Hypotheses phrased like the new worlds' root causes must map onto
+    targeted query plans (not just the generic fallback trio)."""
+    from runbookai_amd.agent.causal_query import (
+        generate_queries_for_hypothesis,
+        match_failure_patterns,
+    )
+
+    auth = match_failure_patterns("clock skew causing jwt 401 failures")
+    assert any(p.name == "auth" for p in auth)
+    egress = match_failure_patterns("nat gateway snat port exhaustion blocks outbound calls")
+    assert any(p.name == "egress" for p in egress)
+    qs = generate_queries_for_hypothesis(
+        "NAT port exhaustion", "external calls time out", ["payment-service"])
+    assert any("natgateway" in str(q.params) for q in qs)
