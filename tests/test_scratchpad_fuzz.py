@@ -72,3 +72,36 @@ def test_retry_loop_detection_fires_on_identical_calls(tmp_path_factory,
                                                       {"q": text}))
     if repeats >= 3:
         assert warned, "identical repeated calls never flagged"
+
+
+@settings(max_examples=60, deadline=None)
+@given(n=st.integers(min_value=0, max_value=60),
+       err_mod=st.integers(min_value=2, max_value=7),
+       budget=st.one_of(st.none(), st.integers(min_value=500, max_value=40000)),
+       seed=st.integers(min_value=0, max_value=10_000))
+def test_compaction_plan_is_exact_partition(n, err_mod, budget, seed):
+    """For ANY result population: the plan's keep_full/keep_compact/clear
+    sets are disjoint and cover every result exactly once, caps hold, and
+    the budgeted variant respects its token budget estimate."""
+    import random
+
+    from runbookai_amd.agent.context_compactor import ContextCompactor
+    from runbookai_amd.agent.scratchpad import Scratchpad
+
+    rng = random.Random(seed)
+    pad = Scratchpad(f"fz-{seed}")
+    for i in range(n):
+        pad.append_tool_result(
+            rng.choice(["cloudwatch_logs", "datadog", "aws_query"]),
+            {"i": i}, f"summary {i}", {"i": i}, has_errors=(i % err_mod == 0))
+    c = ContextCompactor()
+    plan = c.compact(pad, query="errors", token_budget=budget)
+    full, compact, clear = set(plan.keep_full), set(plan.keep_compact), set(plan.clear)
+    all_ids = {r.result_id for r in pad.tool_uses}
+    assert full | compact | clear == all_ids
+    assert not (full & compact) and not (full & clear) and not (compact & clear)
+    assert len(full) <= c.config.max_full
+    if budget is None:
+        assert len(compact) <= c.config.max_compact
+    else:
+        assert c.estimated_tokens(plan) <= max(budget, 2200)  # >= one full result
