@@ -8,7 +8,9 @@ knowledge sync/search/add/validate/stats @1250-1553, deploy @1556,
 config @1587, integrations claude @1666-1995, webhook @1998,
 slack-gateway @2056, mcp serve/tools @2182-2206, operability @2209-2350,
 checkpoint list/show/delete @2353-2461. Ink UI is replaced by plain
-ANSI terminal rendering.
+ANSI terminal rendering. Beyond the reference: eval/eval-all, serve
+(OpenAI-compatible endpoint), replay (session audit trails),
+checkpoint resume, investigate --report, persistent chat memory.
 """
 from __future__ import annotations
 
