@@ -125,3 +125,28 @@ def test_orchestrator_all_tools_down(seed):
     d = result.to_dict()
     assert d["phasesVisited"]
     assert isinstance(d["summary"], str)
+
+
+@settings(max_examples=20, deadline=None)
+@given(seed=st.integers(0, 5000), garbage=st.floats(0.2, 0.9))
+def test_chaos_over_real_simulated_worlds(seed, garbage):
+    """Chaotic model output over REAL scenario telemetry (not canned mock
+    tools): the full tool registry resolves against a randomly-picked
+    simulated world and the orchestrator still returns structure."""
+    import random
+
+    from runbookai_amd.providers.simulation import _SCENARIOS, load_scenario, set_scenario
+    from runbookai_amd.tools.registry import ToolRegistry
+
+    name = random.Random(seed).choice(sorted(_SCENARIOS))
+    set_scenario(load_scenario(name))
+    try:
+        llm = ChaoticLlm(seed, garbage)
+        orch = InvestigationOrchestrator(llm=llm, tool_executor=ToolRegistry(),
+                                         max_iterations=2)
+        result = orch.investigate(f"chaos over {name}")
+        d = result.to_dict()
+        assert isinstance(d["rootCause"], str)
+        assert d["phasesVisited"]
+    finally:
+        set_scenario(None)
