@@ -318,6 +318,32 @@ class ServingAdapter:
         messages = body.get("messages") or []
         if not isinstance(messages, list) or not messages:
             raise ValueError("messages must be a non-empty array")
+        if body.get("tools"):
+            # tool decisions come from a multi-stage constrained decode, so
+            # the call arrives as ONE delta (single-chunk streaming form)
+            result = self._chat_with_tools(body, messages)
+            choice = result["choices"][0]
+            delta: dict[str, Any] = {"role": "assistant"}
+            if choice["message"].get("tool_calls"):
+                delta["tool_calls"] = [
+                    {"index": i, **c} for i, c in
+                    enumerate(choice["message"]["tool_calls"])]
+            if choice["message"].get("content"):
+                delta["content"] = choice["message"]["content"]
+            cid = result["id"]
+            yield "data: " + json.dumps({
+                "id": cid, "object": "chat.completion.chunk", "created": _now(),
+                "model": self.model_name,
+                "choices": [{"index": 0, "delta": delta, "finish_reason": None}],
+            }) + "\n\n"
+            yield "data: " + json.dumps({
+                "id": cid, "object": "chat.completion.chunk", "created": _now(),
+                "model": self.model_name,
+                "choices": [{"index": 0, "delta": {},
+                             "finish_reason": choice["finish_reason"]}],
+            }) + "\n\n"
+            yield "data: [DONE]\n\n"
+            return
         schema = self._resolve_schema(body.get("response_format"))
         req = self._submit(self._encode_chat(messages), body, schema)
         cid = _gen_id("chatcmpl")

@@ -394,3 +394,23 @@ class TestFailureIsolation:
             assert r2.status_code == 200
         finally:
             eng.shutdown()
+
+
+class TestStreamingToolCalls:
+    def test_tools_with_stream_single_delta(self, client):
+        with client.stream("POST", "/v1/chat/completions", json={
+                "messages": [{"role": "user", "content": "check alarms"}],
+                "tools": TestToolCalling.TOOLS, "stream": True,
+                "max_tokens": 96}) as r:
+            assert r.status_code == 200
+            body = "".join(r.iter_text())
+        frames = [l[6:] for l in body.splitlines() if l.startswith("data: ")]
+        assert frames[-1] == "[DONE]"
+        chunks = [json.loads(f) for f in frames[:-1]]
+        finish = chunks[-1]["choices"][0]["finish_reason"]
+        assert finish in ("tool_calls", "stop")
+        if finish == "tool_calls":
+            delta = chunks[0]["choices"][0]["delta"]
+            call = delta["tool_calls"][0]
+            assert call["index"] == 0 and call["type"] == "function"
+            json.loads(call["function"]["arguments"])
