@@ -23,9 +23,7 @@ from typing import Any, Optional
 import torch
 
 from .. import ops
-from ..agent.llm_parser import PROMPT_SCHEMAS
 from .json_fsm import NUMBER_CLOSE_SENTINEL, _STRING_BYTES, JsonFsm
-from .kv_cache import PagedKvCache
 from .llama import CONFIGS, LlamaModel
 from .tokenizer import (
     ACTIVE_VOCAB,

@@ -12,7 +12,7 @@ from __future__ import annotations
 import json
 import re
 import sys
-from typing import Any, Optional
+from typing import Any
 
 # Dangerous-command regex list (reference @380-417).
 DANGEROUS_PATTERNS = [
