@@ -157,6 +157,15 @@ def dispatch(payload: dict[str, Any], retriever: Any = None, store: Any = None,
         return handle_post_tool_use(payload, store)
     if event == "Stop":
         return handle_stop(payload, store)
+    if event in ("SubagentStop", "PreCompact"):
+        # recorded for the learning loop's session timeline; never block
+        if store is not None:
+            try:
+                store.append_event(payload.get("session_id", "unknown"),
+                                   {"kind": event.lower(), **payload})
+            except Exception:  # noqa: BLE001
+                pass
+        return {"continue": True}
     return {"continue": True}
 
 

@@ -311,3 +311,13 @@ class TestSessionStoreBackends:
         store.append_event("s3x", {"kind": "stop"})
         assert store.pending_uploads
         assert store.pending_uploads[0]["key"] == "claude-sessions/s3x.jsonl"
+
+
+class TestLifecycleEvents:
+    def test_subagent_stop_and_precompact_recorded(self, tmp_path):
+        store = SessionStore(directory=str(tmp_path))
+        for ev in ("SubagentStop", "PreCompact"):
+            out = dispatch({"hook_event_name": ev, "session_id": "lc"}, store=store)
+            assert out["continue"]
+        kinds = [e["kind"] for e in store.get_session_events("lc")]
+        assert kinds == ["subagentstop", "precompact"]
