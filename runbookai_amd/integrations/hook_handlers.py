@@ -186,4 +186,8 @@ def handle_stdin() -> None:
         store = SessionStore()
     except Exception:  # noqa: BLE001
         pass
-    print(json.dumps(dispatch(payload, retriever, store), default=str))
+    import os as _os
+
+    inject = _os.environ.get("RUNBOOK_HOOKS_NO_CONTEXT", "") not in ("1", "true")
+    print(json.dumps(dispatch(payload, retriever, store,
+                              inject_context=inject), default=str))
