@@ -307,3 +307,44 @@ class TestAwsCliRealism:
         with _pytest.raises(Exception):
             self._r().execute("aws_cli", {"command":
                 "aws ec2 terminate-instances --instance-ids i-1"})
+
+
+class TestGarbageRobustness:
+    """Byte-soup through every text-parsing surface: no exceptions."""
+
+    def _soup(self, rng, n=120):
+        return bytes(rng.randrange(256) for _ in range(rng.randrange(0, n))) \
+            .decode("utf-8", "replace")
+
+    def test_slack_command_parsing(self):
+        import random
+
+        from runbookai_amd.slack.gateway import build_slack_request, parse_command
+
+        rng = random.Random(99)
+        for _ in range(100):
+            p = parse_command(self._soup(rng))
+            req = build_slack_request(p, {"ts": self._soup(rng, 10)})
+            assert "command" in p and "threadTs" in req
+
+    def test_aws_cli_date_preprocessing(self):
+        import random
+
+        from runbookai_amd.tools.registry import ToolRegistry
+
+        rng = random.Random(7)
+        for _ in range(100):
+            out = ToolRegistry.preprocess_date_expressions(
+                "aws ce get-cost " + self._soup(rng))
+            assert isinstance(out, str)
+
+    def test_mermaid_renderer(self):
+        import random
+
+        from runbookai_amd.tools.diagram import mermaid
+
+        rng = random.Random(3)
+        for prefix in ("", "graph TD\n", "sequenceDiagram\n"):
+            for _ in range(60):
+                out = mermaid.render_mermaid(prefix + self._soup(rng))
+                assert isinstance(out, str)
