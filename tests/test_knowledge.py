@@ -504,3 +504,19 @@ def test_google_token_exchange_and_refresh_protocol():
         assert seen[1]["refresh_token"] == "rt-1"
     finally:
         srv.shutdown()
+
+
+class TestGarbageRobustness:
+    def test_sources_and_parsers_survive_byte_soup(self):
+        import random
+
+        from runbookai_amd.knowledge.sources.confluence import html_to_text
+        from runbookai_amd.knowledge.sources.filesystem import chunk_markdown
+
+        rng = random.Random(42)
+        for _ in range(60):
+            n = rng.randrange(0, 300)
+            s = bytes(rng.randrange(256) for _ in range(n)).decode("utf-8", "replace")
+            assert isinstance(html_to_text("<div>" + s + "</div>"), str)
+            chunks = chunk_markdown(s, "t")
+            assert isinstance(chunks, list)
