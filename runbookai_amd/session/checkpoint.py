@@ -19,6 +19,13 @@ from typing import Any, Optional
 MAX_CHECKPOINTS_PER_INVESTIGATION = 50
 
 
+def _as_float(v) -> float:
+    try:
+        return float(v)
+    except (TypeError, ValueError):
+        return 0.0
+
+
 @dataclass
 class InvestigationCheckpoint:
     checkpoint_id: str
@@ -65,7 +72,7 @@ class InvestigationCheckpoint:
             evidence=d.get("evidence", []),
             remediation=d.get("remediation"),
             root_cause=d.get("rootCause", ""),
-            created_at=d.get("createdAt", 0.0),
+            created_at=_as_float(d.get("createdAt", 0.0)),
             label=d.get("label", ""),
             session_id=d.get("sessionId", ""),
         )
@@ -202,7 +209,7 @@ class CheckpointStore:
             try:
                 with open(os.path.join(inv_dir, fn), encoding="utf-8") as f:
                     out.append(InvestigationCheckpoint.from_dict(json.load(f)))
-            except (json.JSONDecodeError, KeyError, OSError):
+            except (json.JSONDecodeError, KeyError, TypeError, OSError):
                 continue
         out.sort(key=lambda c: c.created_at)
         return out

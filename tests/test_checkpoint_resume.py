@@ -290,3 +290,26 @@ class TestInvestigationReport:
         assert "report written" in result.output
         md = open(tmp_path / "out" / "report.md").read()
         assert "# Investigation report" in md
+
+
+def test_malformed_checkpoint_files_do_not_break_listing(tmp_path):
+    """A corrupt createdAt (string) or truncated file must not crash
+    list()/load_latest() sorting."""
+    import json as _json
+
+    from runbookai_amd.session.checkpoint import CheckpointStore, InvestigationCheckpoint
+
+    store = CheckpointStore(base_dir=str(tmp_path))
+    good = InvestigationCheckpoint(checkpoint_id=CheckpointStore.new_id(),
+                                   investigation_id="inv-x")
+    store.save(good)
+    inv_dir = tmp_path / "inv-x"
+    (inv_dir / "badts.json").write_text(_json.dumps({
+        "checkpointId": "badts", "investigationId": "inv-x",
+        "createdAt": "yesterday-ish"}))
+    (inv_dir / "trunc.json").write_text('{"checkpointId": "t"')
+    cps = store.list("inv-x")
+    ids = {c.checkpoint_id for c in cps}
+    assert good.checkpoint_id in ids
+    assert "badts" in ids  # loads with createdAt coerced to 0.0
+    assert store.load_latest("inv-x") is not None
