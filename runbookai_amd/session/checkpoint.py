@@ -291,8 +291,14 @@ def machine_from_checkpoint(cp: InvestigationCheckpoint,
         # nothing mid-flight to continue; restart the pipeline on the
         # restored evidence instead of replaying a terminal state
         m.phase = Phase.TRIAGE
-    m.hypotheses = {h["id"]: Hypothesis.from_dict(h)
-                    for h in cp.hypotheses if h.get("id")}
+    m.hypotheses = {}
+    for h in cp.hypotheses:
+        if not isinstance(h, dict) or not h.get("id"):
+            continue  # corrupt snapshot entries are skipped, not fatal
+        try:
+            m.hypotheses[h["id"]] = Hypothesis.from_dict(h)
+        except (KeyError, TypeError, ValueError):
+            continue
     m.affected_services = list(cp.services)
     m.symptoms = list(cp.symptoms)
     m.started_at = now_ms()

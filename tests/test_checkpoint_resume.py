@@ -313,3 +313,18 @@ def test_malformed_checkpoint_files_do_not_break_listing(tmp_path):
     assert good.checkpoint_id in ids
     assert "badts" in ids  # loads with createdAt coerced to 0.0
     assert store.load_latest("inv-x") is not None
+
+
+def test_rehydration_skips_corrupt_hypothesis_entries():
+    from runbookai_amd.session.checkpoint import (
+        InvestigationCheckpoint,
+        machine_from_checkpoint,
+    )
+
+    cp = InvestigationCheckpoint(
+        checkpoint_id="x", investigation_id="inv-g", phase="bogus-phase",
+        hypotheses=[{"id": "h1", "statement": "real"}, {"nope": 1},
+                    "a-string", {}, None])
+    m = machine_from_checkpoint(cp)
+    assert list(m.hypotheses) == ["h1"]
+    assert m.phase.value == "triage"  # unknown phase resets to triage
