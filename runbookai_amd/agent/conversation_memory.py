@@ -271,15 +271,23 @@ class ConversationMemory:
     def from_json(cls, raw: str) -> "ConversationMemory":
         data = json.loads(raw)
         mem = cls(summarize_after_messages=data.get("summarizeAfter", 16))
-        mem.messages = [Message(role=m["role"], content=m["content"],
-                                timestamp=m.get("timestamp", 0),
-                                id=m.get("id", new_id("msg-")),
-                                metadata=m.get("metadata", {}))
-                        for m in data.get("messages", [])]
-        mem.compressed_summary = data.get("compressedSummary", "")
-        mem.investigations = [
-            InvestigationSummary(query=s["query"], answer_digest=s.get("answerDigest", ""),
-                                 services=s.get("services", []), timestamp=s.get("timestamp", 0))
-            for s in data.get("investigations", [])
-        ]
+        msgs = data.get("messages", [])
+        for m in msgs if isinstance(msgs, list) else []:
+            if not isinstance(m, dict) or "content" not in m:
+                continue  # malformed persisted entries are skipped
+            mem.messages.append(Message(
+                role=str(m.get("role", "user")), content=str(m["content"]),
+                timestamp=m.get("timestamp", 0),
+                id=m.get("id", new_id("msg-")),
+                metadata=m.get("metadata", {}) if isinstance(m.get("metadata"), dict) else {}))
+        mem.compressed_summary = str(data.get("compressedSummary", "") or "")
+        invs = data.get("investigations", [])
+        for s in invs if isinstance(invs, list) else []:
+            if not isinstance(s, dict):
+                continue
+            mem.investigations.append(InvestigationSummary(
+                query=str(s.get("query", "") or ""),
+                answer_digest=str(s.get("answerDigest", "") or ""),
+                services=s.get("services", []) if isinstance(s.get("services"), list) else [],
+                timestamp=s.get("timestamp", 0)))
         return mem

@@ -213,6 +213,8 @@ class Scratchpad:
                         d = json.loads(line)
                     except json.JSONDecodeError:
                         continue
+                    if not isinstance(d, dict):  # corrupt line: skip
+                        continue
                     kind = d.pop("kind", "note")
                     ts = d.pop("timestamp", now_ms())
                     pad.entries.append(ScratchpadEntry(kind=kind, data=d, timestamp=ts))

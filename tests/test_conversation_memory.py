@@ -223,3 +223,21 @@ class TestCompression:
         for i in range(10):
             mem.add_message("assistant", f"step {i}")
         assert "payments-api" in mem.get_context_for_prompt()
+
+
+class TestCorruptPersistence:
+    def test_from_json_tolerates_malformed_entries(self):
+        for raw in ('{"messages": "nope"}',
+                    '{"messages": [{"content": 1}, "x", {"role": "user"}]}',
+                    '{"investigations": [{"query": null}, "y"]}',
+                    '{"investigations": "x", "compressedSummary": null}'):
+            mem = ConversationMemory.from_json(raw)
+            # skipped entries never crash, survivors keep working
+            mem.stats()
+            mem.get_context_for_prompt()
+
+    def test_partial_message_survives(self):
+        mem = ConversationMemory.from_json(
+            '{"messages": [{"content": "kept", "role": "assistant"},'
+            ' {"nope": 1}]}')
+        assert [m.content for m in mem.get_messages()] == ["kept"]
