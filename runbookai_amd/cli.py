@@ -67,7 +67,10 @@ def _build_runtime(config: Config, scenario: Optional[str] = None,
 def cli(ctx: click.Context, config_path: Optional[str]) -> None:
     """Runbook — MI355X-native AI SRE investigation agent."""
     ctx.ensure_object(dict)
-    ctx.obj["config"] = load_config(config_path)
+    try:
+        ctx.obj["config"] = load_config(config_path)
+    except ValueError as e:
+        raise click.ClickException(str(e)) from e
 
 
 # -- ask / chat ----------------------------------------------------------------

@@ -125,7 +125,14 @@ def load_config(path: Optional[str] = None, cwd: Optional[str] = None) -> Config
     for p in candidates:
         if p and os.path.exists(p):
             with open(p, encoding="utf-8") as f:
-                raw = yaml.safe_load(f) or {}
+                try:
+                    raw = yaml.safe_load(f) or {}
+                except yaml.YAMLError as e:
+                    raise ValueError(
+                        f"config file {p} is not valid YAML: "
+                        f"{getattr(e, 'problem', e)}") from e
+            if not isinstance(raw, dict):
+                raise ValueError(f"config file {p} must be a YAML mapping")
             return Config.model_validate(_interpolate_env(raw))
     return Config()
 

@@ -1,4 +1,5 @@
 """services.yaml schema tests."""
+import pytest
 import yaml
 
 from runbookai_amd.config.services import (
@@ -45,3 +46,34 @@ def test_validation_catches_problems():
 def test_missing_file_defaults(tmp_path):
     cfg = load_services_config(runbook_dir=str(tmp_path))
     assert cfg.services == []
+
+
+class TestCorruptConfig:
+    def test_invalid_yaml_is_a_clean_error(self, tmp_path):
+        from runbookai_amd.config.schema import load_config
+
+        p = tmp_path / "cfg.yaml"
+        p.write_text("llm: [broken\n\t")
+        with pytest.raises(ValueError, match="not valid YAML"):
+            load_config(str(p))
+
+    def test_non_mapping_yaml_rejected(self, tmp_path):
+        from runbookai_amd.config.schema import load_config
+
+        p = tmp_path / "cfg.yaml"
+        p.write_text("- just\n- a list\n")
+        with pytest.raises(ValueError, match="mapping"):
+            load_config(str(p))
+
+    def test_cli_reports_config_error_cleanly(self, tmp_path):
+        import click.testing
+
+        from runbookai_amd.cli import cli
+
+        p = tmp_path / "cfg.yaml"
+        p.write_text("llm: [broken\n\t")
+        out = click.testing.CliRunner().invoke(
+            cli, ["--config", str(p), "status"], obj={})
+        assert out.exit_code != 0
+        assert "not valid YAML" in out.output
+        assert "Traceback" not in out.output
