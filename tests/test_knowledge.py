@@ -520,3 +520,22 @@ class TestGarbageRobustness:
             assert isinstance(html_to_text("<div>" + s + "</div>"), str)
             chunks = chunk_markdown(s, "t")
             assert isinstance(chunks, list)
+
+    def test_fts_store_survives_byte_soup(self):
+        import random
+
+        from runbookai_amd.knowledge.store.sqlite_store import KnowledgeStore
+        from runbookai_amd.knowledge.types import KnowledgeDocument
+
+        rng = random.Random(1)
+
+        def soup(n=100):
+            return bytes(rng.randrange(256) for _ in range(rng.randrange(0, n))) \
+                .decode("utf-8", "replace")
+
+        store = KnowledgeStore(db_path=":memory:")
+        for i in range(25):
+            store.upsert_document(KnowledgeDocument(
+                id=f"d{i}", title=soup(40), content=soup(300),
+                services=[soup(10)], symptoms=[soup(10)]))
+            assert isinstance(store.search(soup(30), limit=3), list)
