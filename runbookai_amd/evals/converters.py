@@ -14,9 +14,20 @@ import re
 from typing import Any, Optional
 
 
+def _dict_items(seq: Any) -> list[dict[str, Any]]:
+    """Dataset rows that aren't mappings are skipped, not fatal — public
+    dataset dumps routinely contain stray strings/nulls."""
+    if isinstance(seq, dict):
+        seq = seq.get("cases") or seq.get("records") or seq.get("incidents") or []
+    if not isinstance(seq, list):
+        return []
+    return [r for r in seq if isinstance(r, dict)]
+
+
 def rcaeval_to_fixtures(records: list[dict[str, Any]], pass_threshold: float = 0.7) -> dict[str, Any]:
     """RCAEval record: {case_id, system, fault_type, root_cause_service,
     root_cause_indicator?, description?} -> fixture case."""
+    records = _dict_items(records)
     cases = []
     for r in records:
         service = str(r.get("root_cause_service", r.get("rootCauseService", "unknown")))
@@ -44,6 +55,7 @@ def rcaeval_to_fixtures(records: list[dict[str, Any]], pass_threshold: float = 0
 def rootly_logs_to_fixtures(incidents: list[dict[str, Any]],
                             pass_threshold: float = 0.7) -> dict[str, Any]:
     """Rootly incident log: {title, summary?, services?, cause?, severity?}."""
+    incidents = _dict_items(incidents)
     cases = []
     for inc in incidents:
         title = str(inc.get("title", "incident"))
@@ -69,6 +81,7 @@ def rootly_logs_to_fixtures(incidents: list[dict[str, Any]],
 def tracerca_to_fixtures(records: list[dict[str, Any]],
                          pass_threshold: float = 0.7) -> dict[str, Any]:
     """TraceRCA record: {trace_id?, anomalous_service, latency_ms?, services?}."""
+    records = _dict_items(records)
     cases = []
     for r in records:
         svc = str(r.get("anomalous_service", r.get("service", "unknown")))

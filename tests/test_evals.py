@@ -98,3 +98,28 @@ class TestConverters:
         fx = tracerca_to_fixtures([{"anomalous_service": "payment", "latency_ms": 900,
                                     "services": ["gateway", "payment"]}])
         assert fx["cases"][0]["expected"]["affectedServices"] == ["payment"]
+
+
+class TestConverterRobustness:
+    def test_converters_skip_non_dict_rows(self):
+        from runbookai_amd.evals.converters import (
+            rcaeval_to_fixtures,
+            rootly_logs_to_fixtures,
+            tracerca_to_fixtures,
+        )
+
+        for fn in (rcaeval_to_fixtures, rootly_logs_to_fixtures,
+                   tracerca_to_fixtures):
+            for garbage in ({}, {"cases": "x"}, ["stray", None],
+                            {"incidents": [{"title": None}, "s"]}):
+                out = fn(garbage)
+                assert isinstance(out, dict) and "cases" in out
+
+    def test_converters_still_convert_valid_rows_among_garbage(self):
+        from runbookai_amd.evals.converters import rcaeval_to_fixtures
+
+        out = rcaeval_to_fixtures(["junk", {"case_id": "c1",
+                                            "root_cause_service": "orders-db",
+                                            "fault_type": "cpu"}, None])
+        assert len(out["cases"]) == 1
+        assert out["cases"][0]["id"] == "c1"
