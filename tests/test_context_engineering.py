@@ -613,3 +613,45 @@ class TestLambdaSummaries:
             "items": [{"FunctionArn":
                        "arn:aws:lambda:us-east-1:123:function:billing-fn:7"}]})
         assert "billing-fn" in out.summary
+
+
+class TestGarbageRobustness:
+    """Byte-soup through the context-engineering surfaces: no exceptions."""
+
+    def _soup(self, rng, n=200):
+        return bytes(rng.randrange(256) for _ in range(rng.randrange(0, n))) \
+            .decode("utf-8", "replace")
+
+    def test_memory_extraction(self):
+        import random
+
+        rng = random.Random(9)
+        mem = InvestigationMemory("probe-g")
+        for _ in range(40):
+            mem.extract_from_thinking(self._soup(rng, 400))
+        mem.build_final_summary()
+
+    def test_hypothesis_engine_rendering(self):
+        import random
+
+        from runbookai_amd.agent.hypothesis import HypothesisEngine
+
+        rng = random.Random(9)
+        eng = HypothesisEngine()
+        for _ in range(20):
+            eng.add(self._soup(rng, 60), rationale=self._soup(rng, 30))
+        assert isinstance(eng.to_markdown(), str)
+        assert isinstance(eng.to_tree_data(), list)
+
+    def test_confidence_and_risk_classification(self):
+        import random
+
+        from runbookai_amd.agent.approval import classify_risk
+
+        rng = random.Random(9)
+        for _ in range(60):
+            assert classify_evidence(self._soup(rng)) in (
+                "supporting", "contradicting", "neutral")
+            risk = classify_risk(self._soup(rng, 30), self._soup(rng, 30))
+            assert isinstance(risk, str)
+        assert 0.0 <= calculate_confidence(-5, 999) <= 1.0
