@@ -539,3 +539,20 @@ class TestGarbageRobustness:
                 id=f"d{i}", title=soup(40), content=soup(300),
                 services=[soup(10)], symptoms=[soup(10)]))
             assert isinstance(store.search(soup(30), limit=3), list)
+
+    def test_graph_traversals_terminate_on_cycles(self):
+        from runbookai_amd.knowledge.store.graph_store import ServiceGraph
+
+        g = ServiceGraph()
+        for a, b in [("a", "b"), ("b", "c"), ("c", "a"), ("c", "d"), ("d", "b")]:
+            g.add_dependency(a, b)
+        down = g.downstream("a")
+        up = g.upstream("a")
+        assert set(down) == {"b", "c", "d"}
+        assert set(up) == {"b", "c", "d"}
+        # service-context blast radius over the same cyclic graph
+        from runbookai_amd.agent.service_context import ServiceContextManager
+
+        mgr = ServiceContextManager(g)
+        info = mgr.blast_radius_info("a")
+        assert isinstance(info, dict)
