@@ -655,3 +655,27 @@ class TestGarbageRobustness:
             risk = classify_risk(self._soup(rng, 30), self._soup(rng, 30))
             assert isinstance(risk, str)
         assert 0.0 <= calculate_confidence(-5, 999) <= 1.0
+
+
+class TestToolCacheProperties:
+    def test_lru_never_exceeds_max_and_hits_are_consistent(self):
+        """Random op sequences: size cap holds; a get immediately after a
+        put for a cacheable tool always hits (no eviction of the newest)."""
+        import random
+
+        rng = random.Random(11)
+        c = ToolCache(max_entries=8)
+        for step in range(400):
+            tool = rng.choice(["datadog", "prometheus", "aws_query"])
+            args = {"q": rng.randrange(20)}
+            op = rng.random()
+            if op < 0.6:
+                c.put(tool, args, step)
+                assert c.get(tool, args) == step  # newest never evicted
+            elif op < 0.9:
+                c.get(tool, args)
+            else:
+                c.invalidate(tool_name=tool)
+            assert len(c._store) <= 8
+        stats = c.stats()
+        assert stats["hits"] > 0 and stats["evictions"] >= 0
