@@ -679,3 +679,32 @@ class TestToolCacheProperties:
             assert len(c._store) <= 8
         stats = c.stats()
         assert stats["hits"] > 0 and stats["evictions"] >= 0
+
+
+class TestParallelExecutorUnderFailures:
+    def test_all_calls_accounted_for_with_mixed_failures(self):
+        import random
+        import time as _t
+
+        rng = random.Random(4)
+
+        def make(tool_id):
+            def ex(**kw):
+                if rng.random() < 0.3:
+                    raise RuntimeError(f"boom-{tool_id}")
+                _t.sleep(rng.random() * 0.01)
+                return {"ok": tool_id}
+            return Tool(name=f"t{tool_id}", description="", parameters={},
+                        execute=ex)
+
+        tools = {f"t{i}": make(i) for i in range(6)}
+        ex = ParallelToolExecutor(max_concurrent=4, timeout_s=5)
+        calls = [ToolCall(id=f"c{i}", name=f"t{i % 6}", arguments={})
+                 for i in range(24)]
+        results = ex.execute_all(tools, calls)
+        assert len(results) == 24
+        by_id = {r.call_id if hasattr(r, "call_id") else r.call.id: r
+                 for r in results}
+        assert len(by_id) == 24  # every call produced exactly one result
+        assert any(getattr(r, "error", None) for r in results)
+        assert any(not getattr(r, "error", None) for r in results)
