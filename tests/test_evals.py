@@ -123,3 +123,31 @@ class TestConverterRobustness:
                                             "fault_type": "cpu"}, None])
         assert len(out["cases"]) == 1
         assert out["cases"][0]["id"] == "c1"
+
+
+class TestScoringRobustness:
+    def test_scorer_on_garbage_inputs(self):
+        """Any text through the scorer: scores stay in [0, 1], never raise."""
+        import random
+
+        from runbookai_amd.evals.scoring import score_investigation_result
+
+        rng = random.Random(8)
+
+        def soup(n=120):
+            return bytes(rng.randrange(256) for _ in range(rng.randrange(0, n))) \
+                .decode("utf-8", "replace")
+
+        for _ in range(40):
+            score = score_investigation_result(
+                result={"rootCause": soup(200), "summary": soup(200),
+                        "confidence": rng.choice(
+                            ["low", "medium", "high", "", soup(4)]),
+                        "affectedServices": [soup(10)]},
+                expected={"rootCauseKeywords": [soup(10), soup(10)],
+                          "affectedServices": [soup(12)],
+                          "requiredPhrases": [soup(8)],
+                          "forbiddenPhrases": [soup(8)],
+                          "confidenceAtLeast": rng.choice(
+                              ["low", "medium", "high", soup(5)])})
+            assert 0.0 <= score["overall"] <= 1.0
